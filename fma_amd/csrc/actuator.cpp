@@ -1,0 +1,680 @@
+// Host side of the MI355X sleep/wake actuator (torch extension).
+//
+// Native replacement for the GPU hot path the reference delegates to vLLM's
+// sleep mode (reference README.md:16-26; the controller only calls
+// POST /sleep, POST /wake_up over HTTP — pkg/controller/dual-pods/
+// inference-server.go:1497,1712). Here the actual tensor movement between
+// GPU HBM3E and pinned host DRAM is implemented directly:
+//
+// - DeviceArena: one contiguous device allocation holding every parameter
+//   of a model instance. Preferred backing is HIP virtual memory management
+//   (hipMemAddressReserve + hipMemCreate/hipMemMap): the virtual address is
+//   reserved once and survives sleep, so parameter tensors (views into the
+//   arena) stay valid across sleep/wake while the physical HBM is truly
+//   released while asleep. Fallback is plain hipMalloc with re-binding.
+// - pack_to_host / restore_from_host: for models whose tensors live in
+//   scattered allocations, a descriptor-table gather/scatter HIP kernel
+//   (kernels.hip) coalesces shards chunk-by-chunk through device staging
+//   buffers, double-buffered against hipMemcpyAsync on dedicated copy
+//   streams (or written straight to pinned host by the kernel in direct
+//   mode).
+//
+// All transfers use pinned host memory; chunked round-robin across two copy
+// streams keeps the SDMA engines busy, and the wake-path physical mapping
+// (hipMemCreate/hipMemMap) overlaps in-flight H2D copies.
+
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include <algorithm>
+#include <chrono>
+#include <cstring>
+#include <mutex>
+#include <tuple>
+#include <unordered_map>
+#include <vector>
+
+#include "kernels.h"
+
+namespace {
+
+#define FMA_HIP_CHECK(expr)                                                  \
+  do {                                                                       \
+    hipError_t _e = (expr);                                                  \
+    TORCH_CHECK(_e == hipSuccess, "HIP error at ", __FILE__, ":", __LINE__,  \
+                " — ", hipGetErrorString(_e), " in `" #expr "`");            \
+  } while (0)
+
+using Clock = std::chrono::steady_clock;
+
+double seconds_since(Clock::time_point t0) {
+  return std::chrono::duration<double>(Clock::now() - t0).count();
+}
+
+constexpr int kNumCopyStreams = 2;
+constexpr int64_t kDefaultChunk = 256ll << 20;  // 256 MiB
+
+struct DeviceCtx {
+  int device = -1;
+  hipStream_t kernel_stream = nullptr;
+  hipStream_t copy_streams[kNumCopyStreams] = {nullptr, nullptr};
+  hipEvent_t sync_event = nullptr;
+
+  void init(int dev) {
+    device = dev;
+    FMA_HIP_CHECK(hipSetDevice(dev));
+    FMA_HIP_CHECK(hipStreamCreateWithFlags(&kernel_stream, hipStreamNonBlocking));
+    for (auto& s : copy_streams) {
+      FMA_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+    }
+    FMA_HIP_CHECK(hipEventCreateWithFlags(&sync_event, hipEventDisableTiming));
+  }
+};
+
+DeviceCtx& ctx_for(int device) {
+  static std::mutex mu;
+  static std::unordered_map<int, DeviceCtx> ctxs;
+  std::lock_guard<std::mutex> lk(mu);
+  auto it = ctxs.find(device);
+  if (it == ctxs.end()) {
+    it = ctxs.emplace(device, DeviceCtx{}).first;
+    it->second.init(device);
+  }
+  return it->second;
+}
+
+// Make our private streams order after torch's current stream (parameter
+// init / forward work must be complete before we move bytes).
+void join_torch_stream(DeviceCtx& ctx) {
+  auto torch_stream = c10::hip::getCurrentHIPStream(ctx.device);
+  FMA_HIP_CHECK(hipEventRecord(ctx.sync_event, torch_stream.stream()));
+  FMA_HIP_CHECK(hipStreamWaitEvent(ctx.kernel_stream, ctx.sync_event, 0));
+  for (auto& s : ctx.copy_streams) {
+    FMA_HIP_CHECK(hipStreamWaitEvent(s, ctx.sync_event, 0));
+  }
+}
+
+void sync_pipeline(DeviceCtx& ctx) {
+  FMA_HIP_CHECK(hipStreamSynchronize(ctx.kernel_stream));
+  for (auto& s : ctx.copy_streams) {
+    FMA_HIP_CHECK(hipStreamSynchronize(s));
+  }
+}
+
+void check_host_buffer(const at::Tensor& host, unsigned long long need) {
+  TORCH_CHECK(host.device().is_cpu(), "host buffer must be a CPU tensor");
+  TORCH_CHECK(host.scalar_type() == at::kByte, "host buffer must be uint8");
+  TORCH_CHECK(host.is_contiguous(), "host buffer must be contiguous");
+  TORCH_CHECK(static_cast<unsigned long long>(host.nbytes()) >= need,
+              "host buffer too small: ", host.nbytes(), " < ", need);
+  TORCH_CHECK(host.is_pinned(), "host buffer must be pinned (pin_memory=True)");
+}
+
+unsigned char* host_device_ptr(const at::Tensor& host) {
+  void* dp = nullptr;
+  hipError_t e = hipHostGetDevicePointer(&dp, host.data_ptr(), 0);
+  if (e != hipSuccess || dp == nullptr) return nullptr;
+  return static_cast<unsigned char*>(dp);
+}
+
+int64_t chunk_or_default(int64_t chunk_bytes) {
+  return chunk_bytes > 0 ? chunk_bytes : kDefaultChunk;
+}
+
+void validate_tensors(const std::vector<at::Tensor>& tensors,
+                      const std::vector<int64_t>& offsets) {
+  TORCH_CHECK(tensors.size() == offsets.size(),
+              "tensors and offsets length mismatch");
+  TORCH_CHECK(!tensors.empty(), "empty tensor list");
+  const auto dev = tensors[0].device();
+  TORCH_CHECK(dev.is_cuda(), "tensors must live on the GPU");
+  for (size_t i = 0; i < tensors.size(); ++i) {
+    const auto& t = tensors[i];
+    TORCH_CHECK(t.device() == dev, "tensor ", i, " on a different device");
+    TORCH_CHECK(t.is_contiguous(), "tensor ", i, " must be contiguous");
+    TORCH_CHECK((reinterpret_cast<uintptr_t>(t.data_ptr()) & 0xF) == 0,
+                "tensor ", i, " not 16-byte aligned");
+    TORCH_CHECK((offsets[i] & (FMA_ARENA_ALIGN - 1)) == 0,
+                "offset ", i, " not ", FMA_ARENA_ALIGN, "-byte aligned");
+    TORCH_CHECK(i == 0 || offsets[i] >= offsets[i - 1],
+                "offsets must be non-decreasing");
+  }
+}
+
+int64_t flat_extent(const std::vector<at::Tensor>& tensors,
+                    const std::vector<int64_t>& offsets) {
+  int64_t total = 0;
+  for (size_t i = 0; i < tensors.size(); ++i) {
+    total = std::max(total, offsets[i] + static_cast<int64_t>(tensors[i].nbytes()));
+  }
+  return total;
+}
+
+// Chunked descriptor plan: splits every tensor along flat-space chunk
+// boundaries. Descriptor "flat pointers" are stored as offsets *within the
+// chunk* and retargeted (staging buffer or mapped host base) at launch.
+struct ChunkPlan {
+  int64_t chunk = 0;
+  int64_t total_bytes = 0;
+  int64_t nchunks = 0;
+  // concatenated per-chunk descs, with in-chunk offsets in place of flat ptrs
+  std::vector<FmaCopyDesc> descs;
+  std::vector<unsigned long long> prefix;   // concatenated (n_c+1 each)
+  std::vector<int64_t> desc_lo;             // per chunk: first desc index
+  std::vector<int64_t> prefix_lo;           // per chunk: first prefix index
+  std::vector<unsigned long long> units;    // per chunk: total 16B units
+};
+
+ChunkPlan build_chunk_plan(const std::vector<at::Tensor>& tensors,
+                           const std::vector<int64_t>& offsets,
+                           int64_t chunk, bool pack) {
+  ChunkPlan plan;
+  plan.chunk = chunk;
+  plan.total_bytes = flat_extent(tensors, offsets);
+  plan.nchunks = (plan.total_bytes + chunk - 1) / chunk;
+  std::vector<std::vector<FmaCopyDesc>> per_chunk(plan.nchunks);
+  for (size_t i = 0; i < tensors.size(); ++i) {
+    const int64_t bytes = static_cast<int64_t>(tensors[i].nbytes());
+    if (!bytes) continue;
+    auto* tp = static_cast<unsigned char*>(tensors[i].data_ptr());
+    int64_t pos = offsets[i];
+    int64_t rem = bytes;
+    while (rem > 0) {
+      const int64_t c = pos / chunk;
+      const int64_t in_chunk = pos - c * chunk;
+      const int64_t take = std::min(rem, chunk - in_chunk);
+      FmaCopyDesc d;
+      auto* flat_as_off = reinterpret_cast<unsigned char*>(
+          static_cast<uintptr_t>(in_chunk));
+      d.src = pack ? tp + (bytes - rem) : flat_as_off;
+      d.dst = pack ? flat_as_off : tp + (bytes - rem);
+      d.bytes = static_cast<unsigned long long>(take);
+      per_chunk[c].push_back(d);
+      pos += take;
+      rem -= take;
+    }
+  }
+  for (int64_t c = 0; c < plan.nchunks; ++c) {
+    plan.desc_lo.push_back(static_cast<int64_t>(plan.descs.size()));
+    plan.prefix_lo.push_back(static_cast<int64_t>(plan.prefix.size()));
+    unsigned long long u = 0;
+    plan.prefix.push_back(0);
+    for (const auto& d : per_chunk[c]) {
+      TORCH_CHECK(per_chunk[c].size() <= FMA_MAX_DESCS_PER_LAUNCH,
+                  "too many descriptors in one chunk; raise chunk_bytes");
+      plan.descs.push_back(d);
+      u += (d.bytes + 15ull) >> 4;
+      plan.prefix.push_back(u);
+    }
+    plan.units.push_back(u);
+  }
+  return plan;
+}
+
+// Device copy of the full plan (one upload, freed after sync at the end).
+struct DevBlob {
+  void* blob = nullptr;
+  FmaCopyDesc* descs = nullptr;
+  unsigned long long* prefix = nullptr;
+
+  void upload(const std::vector<FmaCopyDesc>& descs_h,
+              const std::vector<unsigned long long>& prefix_h,
+              hipStream_t stream) {
+    const size_t db = descs_h.size() * sizeof(FmaCopyDesc);
+    const size_t pb = prefix_h.size() * sizeof(unsigned long long);
+    FMA_HIP_CHECK(hipMalloc(&blob, std::max<size_t>(db + pb, 16)));
+    descs = static_cast<FmaCopyDesc*>(blob);
+    prefix = reinterpret_cast<unsigned long long*>(
+        static_cast<unsigned char*>(blob) + db);
+    if (db) {
+      FMA_HIP_CHECK(hipMemcpyAsync(descs, descs_h.data(), db,
+                                   hipMemcpyHostToDevice, stream));
+    }
+    if (pb) {
+      FMA_HIP_CHECK(hipMemcpyAsync(prefix, prefix_h.data(), pb,
+                                   hipMemcpyHostToDevice, stream));
+    }
+  }
+  ~DevBlob() {
+    if (blob) (void)hipFree(blob);
+  }
+};
+
+enum class XferMode : int64_t {
+  kStaged = 0,     // gather/scatter kernel <-> device staging, SDMA to host
+  kDirect = 1,     // kernel reads/writes pinned host directly over PCIe
+  kPerTensor = 2,  // one hipMemcpyAsync per tensor (no kernel) — baseline
+};
+
+// Retarget a chunk's descriptors onto a concrete base pointer.
+std::vector<FmaCopyDesc> retarget(const ChunkPlan& plan, int64_t c,
+                                  unsigned char* base, bool pack) {
+  const int64_t lo = plan.desc_lo[c];
+  const int64_t hi = (c + 1 < plan.nchunks) ? plan.desc_lo[c + 1]
+                                            : static_cast<int64_t>(plan.descs.size());
+  std::vector<FmaCopyDesc> out(plan.descs.begin() + lo, plan.descs.begin() + hi);
+  for (auto& d : out) {
+    if (pack) {
+      d.dst = base + reinterpret_cast<uintptr_t>(d.dst);
+    } else {
+      d.src = base + reinterpret_cast<uintptr_t>(d.src);
+    }
+  }
+  return out;
+}
+
+double pack_to_host(const std::vector<at::Tensor>& tensors,
+                    const std::vector<int64_t>& offsets, at::Tensor host,
+                    int64_t mode_i, int64_t chunk_bytes) {
+  validate_tensors(tensors, offsets);
+  const int device = tensors[0].device().index();
+  auto& ctx = ctx_for(device);
+  FMA_HIP_CHECK(hipSetDevice(device));
+  const auto mode = static_cast<XferMode>(mode_i);
+  const int64_t chunk = chunk_or_default(chunk_bytes);
+  const int64_t total = flat_extent(tensors, offsets);
+  check_host_buffer(host, total);
+  auto* host_ptr = static_cast<unsigned char*>(host.data_ptr());
+
+  const auto t0 = Clock::now();
+  join_torch_stream(ctx);
+
+  if (mode == XferMode::kPerTensor) {
+    for (size_t i = 0; i < tensors.size(); ++i) {
+      const auto bytes = tensors[i].nbytes();
+      if (!bytes) continue;
+      FMA_HIP_CHECK(hipMemcpyAsync(host_ptr + offsets[i], tensors[i].data_ptr(),
+                                   bytes, hipMemcpyDeviceToHost,
+                                   ctx.copy_streams[i % kNumCopyStreams]));
+    }
+    sync_pipeline(ctx);
+    return seconds_since(t0);
+  }
+
+  if (mode == XferMode::kDirect) {
+    auto* hdp = host_device_ptr(host);
+    TORCH_CHECK(hdp, "pinned host memory is not device-mapped; use staged mode");
+    // single launch over the whole flat space, chunk = everything
+    ChunkPlan plan = build_chunk_plan(tensors, offsets, total, /*pack=*/true);
+    auto descs = retarget(plan, 0, hdp, /*pack=*/true);
+    std::vector<unsigned long long> prefix(
+        plan.prefix.begin(), plan.prefix.begin() + descs.size() + 1);
+    DevBlob dev;
+    dev.upload(descs, prefix, ctx.kernel_stream);
+    FMA_HIP_CHECK(fma_launch_batched_copy(dev.descs, dev.prefix,
+                                          static_cast<int>(descs.size()),
+                                          plan.units[0], ctx.kernel_stream));
+    FMA_HIP_CHECK(hipStreamSynchronize(ctx.kernel_stream));
+    return seconds_since(t0);
+  }
+
+  // Staged pipeline.
+  ChunkPlan plan = build_chunk_plan(tensors, offsets, chunk, /*pack=*/true);
+  void* staging[2] = {nullptr, nullptr};
+  FMA_HIP_CHECK(hipMalloc(&staging[0], chunk));
+  FMA_HIP_CHECK(hipMalloc(&staging[1], chunk));
+  hipEvent_t copied[2], packed[2];
+  for (int b = 0; b < 2; ++b) {
+    FMA_HIP_CHECK(hipEventCreateWithFlags(&copied[b], hipEventDisableTiming));
+    FMA_HIP_CHECK(hipEventCreateWithFlags(&packed[b], hipEventDisableTiming));
+  }
+  // one desc/prefix upload for all chunks, with per-chunk slices; the
+  // staging retarget is done on-device by passing the staging base to the
+  // kernel? No: descs carry absolute pointers, so we upload retargeted
+  // copies for both staging buffers once (chunks alternate buffers).
+  std::vector<FmaCopyDesc> all_descs;
+  std::vector<int64_t> launch_desc_lo(plan.nchunks), launch_ndesc(plan.nchunks);
+  for (int64_t c = 0; c < plan.nchunks; ++c) {
+    auto descs =
+        retarget(plan, c, static_cast<unsigned char*>(staging[c & 1]), true);
+    launch_desc_lo[c] = static_cast<int64_t>(all_descs.size());
+    launch_ndesc[c] = static_cast<int64_t>(descs.size());
+    all_descs.insert(all_descs.end(), descs.begin(), descs.end());
+  }
+  DevBlob dev;
+  dev.upload(all_descs, plan.prefix, ctx.kernel_stream);
+
+  for (int64_t c = 0; c < plan.nchunks; ++c) {
+    const int b = static_cast<int>(c & 1);
+    if (c >= 2) {
+      FMA_HIP_CHECK(hipStreamWaitEvent(ctx.kernel_stream, copied[b], 0));
+    }
+    FMA_HIP_CHECK(fma_launch_batched_copy(
+        dev.descs + launch_desc_lo[c], dev.prefix + plan.prefix_lo[c],
+        static_cast<int>(launch_ndesc[c]), plan.units[c], ctx.kernel_stream));
+    FMA_HIP_CHECK(hipEventRecord(packed[b], ctx.kernel_stream));
+    auto cs = ctx.copy_streams[b];
+    FMA_HIP_CHECK(hipStreamWaitEvent(cs, packed[b], 0));
+    const int64_t lo = c * chunk;
+    const int64_t sz = std::min<int64_t>(chunk, plan.total_bytes - lo);
+    FMA_HIP_CHECK(hipMemcpyAsync(host_ptr + lo, staging[b], sz,
+                                 hipMemcpyDeviceToHost, cs));
+    FMA_HIP_CHECK(hipEventRecord(copied[b], cs));
+  }
+  sync_pipeline(ctx);
+  for (int b = 0; b < 2; ++b) {
+    (void)hipEventDestroy(copied[b]);
+    (void)hipEventDestroy(packed[b]);
+    (void)hipFree(staging[b]);
+  }
+  return seconds_since(t0);
+}
+
+double restore_from_host(const std::vector<at::Tensor>& tensors,
+                         const std::vector<int64_t>& offsets, at::Tensor host,
+                         int64_t mode_i, int64_t chunk_bytes) {
+  validate_tensors(tensors, offsets);
+  const int device = tensors[0].device().index();
+  auto& ctx = ctx_for(device);
+  FMA_HIP_CHECK(hipSetDevice(device));
+  const auto mode = static_cast<XferMode>(mode_i);
+  const int64_t chunk = chunk_or_default(chunk_bytes);
+  const int64_t total = flat_extent(tensors, offsets);
+  check_host_buffer(host, total);
+  auto* host_ptr = static_cast<unsigned char*>(host.data_ptr());
+
+  const auto t0 = Clock::now();
+  join_torch_stream(ctx);
+
+  if (mode == XferMode::kPerTensor) {
+    for (size_t i = 0; i < tensors.size(); ++i) {
+      const auto bytes = tensors[i].nbytes();
+      if (!bytes) continue;
+      FMA_HIP_CHECK(hipMemcpyAsync(tensors[i].data_ptr(), host_ptr + offsets[i],
+                                   bytes, hipMemcpyHostToDevice,
+                                   ctx.copy_streams[i % kNumCopyStreams]));
+    }
+    sync_pipeline(ctx);
+    return seconds_since(t0);
+  }
+
+  if (mode == XferMode::kDirect) {
+    auto* hdp = host_device_ptr(host);
+    TORCH_CHECK(hdp, "pinned host memory is not device-mapped; use staged mode");
+    ChunkPlan plan = build_chunk_plan(tensors, offsets, total, /*pack=*/false);
+    auto descs = retarget(plan, 0, hdp, /*pack=*/false);
+    std::vector<unsigned long long> prefix(
+        plan.prefix.begin(), plan.prefix.begin() + descs.size() + 1);
+    DevBlob dev;
+    dev.upload(descs, prefix, ctx.kernel_stream);
+    FMA_HIP_CHECK(fma_launch_batched_copy(dev.descs, dev.prefix,
+                                          static_cast<int>(descs.size()),
+                                          plan.units[0], ctx.kernel_stream));
+    FMA_HIP_CHECK(hipStreamSynchronize(ctx.kernel_stream));
+    return seconds_since(t0);
+  }
+
+  // Staged pipeline: H2D chunk -> scatter kernel, double-buffered.
+  ChunkPlan plan = build_chunk_plan(tensors, offsets, chunk, /*pack=*/false);
+  void* staging[2] = {nullptr, nullptr};
+  FMA_HIP_CHECK(hipMalloc(&staging[0], chunk));
+  FMA_HIP_CHECK(hipMalloc(&staging[1], chunk));
+  hipEvent_t arrived[2], scattered[2];
+  for (int b = 0; b < 2; ++b) {
+    FMA_HIP_CHECK(hipEventCreateWithFlags(&arrived[b], hipEventDisableTiming));
+    FMA_HIP_CHECK(hipEventCreateWithFlags(&scattered[b], hipEventDisableTiming));
+  }
+  std::vector<FmaCopyDesc> all_descs;
+  std::vector<int64_t> launch_desc_lo(plan.nchunks), launch_ndesc(plan.nchunks);
+  for (int64_t c = 0; c < plan.nchunks; ++c) {
+    auto descs =
+        retarget(plan, c, static_cast<unsigned char*>(staging[c & 1]), false);
+    launch_desc_lo[c] = static_cast<int64_t>(all_descs.size());
+    launch_ndesc[c] = static_cast<int64_t>(descs.size());
+    all_descs.insert(all_descs.end(), descs.begin(), descs.end());
+  }
+  DevBlob dev;
+  dev.upload(all_descs, plan.prefix, ctx.kernel_stream);
+
+  for (int64_t c = 0; c < plan.nchunks; ++c) {
+    const int b = static_cast<int>(c & 1);
+    auto cs = ctx.copy_streams[b];
+    if (c >= 2) {
+      FMA_HIP_CHECK(hipStreamWaitEvent(cs, scattered[b], 0));
+    }
+    const int64_t lo = c * chunk;
+    const int64_t sz = std::min<int64_t>(chunk, plan.total_bytes - lo);
+    FMA_HIP_CHECK(hipMemcpyAsync(staging[b], host_ptr + lo, sz,
+                                 hipMemcpyHostToDevice, cs));
+    FMA_HIP_CHECK(hipEventRecord(arrived[b], cs));
+    FMA_HIP_CHECK(hipStreamWaitEvent(ctx.kernel_stream, arrived[b], 0));
+    FMA_HIP_CHECK(fma_launch_batched_copy(
+        dev.descs + launch_desc_lo[c], dev.prefix + plan.prefix_lo[c],
+        static_cast<int>(launch_ndesc[c]), plan.units[c], ctx.kernel_stream));
+    FMA_HIP_CHECK(hipEventRecord(scattered[b], ctx.kernel_stream));
+  }
+  sync_pipeline(ctx);
+  for (int b = 0; b < 2; ++b) {
+    (void)hipEventDestroy(arrived[b]);
+    (void)hipEventDestroy(scattered[b]);
+    (void)hipFree(staging[b]);
+  }
+  return seconds_since(t0);
+}
+
+// ---------------------------------------------------------------------------
+// DeviceArena
+// ---------------------------------------------------------------------------
+
+bool device_supports_vmm(int device) {
+  int v = 0;
+  hipError_t e = hipDeviceGetAttribute(
+      &v, hipDeviceAttributeVirtualMemoryManagementSupported, device);
+  return e == hipSuccess && v != 0;
+}
+
+class DeviceArena {
+ public:
+  DeviceArena(int64_t nbytes, int device, bool try_vmm)
+      : size_(nbytes), device_(device) {
+    TORCH_CHECK(nbytes > 0, "arena size must be positive");
+    FMA_HIP_CHECK(hipSetDevice(device_));
+    vmm_ = try_vmm && device_supports_vmm(device_);
+    if (vmm_) {
+      hipMemAllocationProp prop = alloc_prop();
+      size_t gran = 0;
+      FMA_HIP_CHECK(hipMemGetAllocationGranularity(
+          &gran, &prop, hipMemAllocationGranularityMinimum));
+      granularity_ = std::max<size_t>(gran, 1);
+      phys_chunk_ =
+          ((1ull << 30) + granularity_ - 1) / granularity_ * granularity_;
+      padded_ = (static_cast<size_t>(nbytes) + granularity_ - 1) /
+                granularity_ * granularity_;
+      FMA_HIP_CHECK(
+          hipMemAddressReserve(&base_, padded_, granularity_, nullptr, 0));
+      for (size_t off = 0; off < padded_; off += phys_chunk_) {
+        map_slice(off, std::min(phys_chunk_, padded_ - off));
+      }
+    } else {
+      padded_ = static_cast<size_t>(nbytes);
+      FMA_HIP_CHECK(hipMalloc(&base_, padded_));
+    }
+    mapped_ = true;
+  }
+
+  ~DeviceArena() { release_all(); }
+  DeviceArena(const DeviceArena&) = delete;
+  DeviceArena& operator=(const DeviceArena&) = delete;
+
+  int64_t data_ptr() const { return reinterpret_cast<int64_t>(base_); }
+  bool is_mapped() const { return mapped_; }
+  bool uses_vmm() const { return vmm_; }
+  int64_t size_bytes() const { return size_; }
+  int device() const { return device_; }
+
+  at::Tensor view(int64_t offset, std::vector<int64_t> sizes,
+                  at::ScalarType dtype) {
+    TORCH_CHECK(mapped_, "arena is asleep (not mapped)");
+    int64_t numel = 1;
+    for (auto s : sizes) numel *= s;
+    const int64_t bytes = numel * static_cast<int64_t>(at::elementSize(dtype));
+    TORCH_CHECK(offset >= 0 && offset + bytes <= size_, "view out of bounds");
+    auto options =
+        at::TensorOptions().dtype(dtype).device(at::Device(at::kCUDA, device_));
+    return at::from_blob(static_cast<unsigned char*>(base_) + offset, sizes,
+                         options);
+  }
+
+  // sleep(level=1): D2H into pinned host DRAM, then release physical HBM.
+  double sleep_to(at::Tensor host, int64_t chunk_bytes) {
+    TORCH_CHECK(mapped_, "arena already asleep");
+    check_host_buffer(host, size_);
+    auto& ctx = ctx_for(device_);
+    FMA_HIP_CHECK(hipSetDevice(device_));
+    const int64_t chunk = chunk_or_default(chunk_bytes);
+    auto* host_ptr = static_cast<unsigned char*>(host.data_ptr());
+    auto* dev_ptr = static_cast<unsigned char*>(base_);
+    const auto t0 = Clock::now();
+    join_torch_stream(ctx);
+    for (int64_t off = 0, c = 0; off < size_; off += chunk, ++c) {
+      const int64_t sz = std::min<int64_t>(chunk, size_ - off);
+      FMA_HIP_CHECK(hipMemcpyAsync(host_ptr + off, dev_ptr + off, sz,
+                                   hipMemcpyDeviceToHost,
+                                   ctx.copy_streams[c % kNumCopyStreams]));
+    }
+    sync_pipeline(ctx);
+    unmap_physical();
+    mapped_ = false;
+    return seconds_since(t0);
+  }
+
+  // wake_up: re-acquire physical HBM and copy the arena back; in VMM mode
+  // mapping of slice k+1 overlaps the H2D of slice k and all tensor views
+  // remain valid (same VA).
+  double wake_from(at::Tensor host, int64_t chunk_bytes) {
+    TORCH_CHECK(!mapped_, "arena already awake");
+    check_host_buffer(host, size_);
+    auto& ctx = ctx_for(device_);
+    FMA_HIP_CHECK(hipSetDevice(device_));
+    const int64_t chunk = chunk_or_default(chunk_bytes);
+    auto* host_ptr = static_cast<unsigned char*>(host.data_ptr());
+    const auto t0 = Clock::now();
+    if (vmm_) {
+      int64_t copied = 0, c = 0;
+      for (size_t off = 0; off < padded_; off += phys_chunk_) {
+        const size_t sz = std::min(phys_chunk_, padded_ - off);
+        map_slice(off, sz);
+        const int64_t avail =
+            std::min<int64_t>(static_cast<int64_t>(off + sz), size_);
+        auto* dev_ptr = static_cast<unsigned char*>(base_);
+        while (copied < avail) {
+          const int64_t take = std::min<int64_t>(chunk, avail - copied);
+          FMA_HIP_CHECK(hipMemcpyAsync(dev_ptr + copied, host_ptr + copied,
+                                       take, hipMemcpyHostToDevice,
+                                       ctx.copy_streams[c++ % kNumCopyStreams]));
+          copied += take;
+        }
+      }
+    } else {
+      FMA_HIP_CHECK(hipMalloc(&base_, padded_));  // NOTE: base may change
+      auto* dev_ptr = static_cast<unsigned char*>(base_);
+      for (int64_t off = 0, c = 0; off < size_; off += chunk, ++c) {
+        const int64_t sz = std::min<int64_t>(chunk, size_ - off);
+        FMA_HIP_CHECK(hipMemcpyAsync(dev_ptr + off, host_ptr + off, sz,
+                                     hipMemcpyHostToDevice,
+                                     ctx.copy_streams[c % kNumCopyStreams]));
+      }
+    }
+    sync_pipeline(ctx);
+    mapped_ = true;
+    return seconds_since(t0);
+  }
+
+ private:
+  hipMemAllocationProp alloc_prop() const {
+    hipMemAllocationProp prop{};
+    prop.type = hipMemAllocationTypePinned;
+    prop.location.type = hipMemLocationTypeDevice;
+    prop.location.id = device_;
+    return prop;
+  }
+
+  void map_slice(size_t off, size_t sz) {
+    hipMemAllocationProp prop = alloc_prop();
+    hipMemGenericAllocationHandle_t h{};
+    FMA_HIP_CHECK(hipMemCreate(&h, sz, &prop, 0));
+    FMA_HIP_CHECK(
+        hipMemMap(static_cast<unsigned char*>(base_) + off, sz, 0, h, 0));
+    hipMemAccessDesc acc{};
+    acc.location.type = hipMemLocationTypeDevice;
+    acc.location.id = device_;
+    acc.flags = hipMemAccessFlagsProtReadWrite;
+    FMA_HIP_CHECK(hipMemSetAccess(static_cast<unsigned char*>(base_) + off, sz,
+                                  &acc, 1));
+    handles_.emplace_back(off, sz, h);
+  }
+
+  void unmap_physical() {
+    if (vmm_) {
+      for (auto& [off, sz, h] : handles_) {
+        (void)hipMemUnmap(static_cast<unsigned char*>(base_) + off, sz);
+        (void)hipMemRelease(h);
+      }
+      handles_.clear();
+    } else if (base_) {
+      (void)hipFree(base_);
+      base_ = nullptr;
+    }
+  }
+
+  void release_all() {
+    if (mapped_) unmap_physical();
+    if (vmm_ && base_) {
+      (void)hipMemAddressFree(base_, padded_);
+      base_ = nullptr;
+    }
+    mapped_ = false;
+  }
+
+  int64_t size_ = 0;
+  size_t padded_ = 0;
+  size_t granularity_ = 1;
+  size_t phys_chunk_ = 1ull << 30;
+  int device_ = 0;
+  bool vmm_ = false;
+  bool mapped_ = false;
+  void* base_ = nullptr;
+  std::vector<std::tuple<size_t, size_t, hipMemGenericAllocationHandle_t>>
+      handles_;
+};
+
+std::tuple<int64_t, int64_t> device_mem_info(int device) {
+  FMA_HIP_CHECK(hipSetDevice(device));
+  size_t free_b = 0, total_b = 0;
+  FMA_HIP_CHECK(hipMemGetInfo(&free_b, &total_b));
+  return {static_cast<int64_t>(free_b), static_cast<int64_t>(total_b)};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() =
+      "MI355X-native sleep/wake actuator (HIP pack/scatter + pinned transfers)";
+  m.def("pack_to_host", &pack_to_host,
+        "Gather scattered device tensors into a pinned host buffer",
+        py::arg("tensors"), py::arg("offsets"), py::arg("host"),
+        py::arg("mode") = 0, py::arg("chunk_bytes") = 0);
+  m.def("restore_from_host", &restore_from_host,
+        "Scatter a pinned host buffer back into device tensors",
+        py::arg("tensors"), py::arg("offsets"), py::arg("host"),
+        py::arg("mode") = 0, py::arg("chunk_bytes") = 0);
+  m.def("device_supports_vmm", &device_supports_vmm, py::arg("device"));
+  m.def("device_mem_info", &device_mem_info, py::arg("device"));
+  py::class_<DeviceArena>(m, "DeviceArena")
+      .def(py::init<int64_t, int, bool>(), py::arg("nbytes"), py::arg("device"),
+           py::arg("try_vmm") = true)
+      .def("view", &DeviceArena::view, py::arg("offset"), py::arg("sizes"),
+           py::arg("dtype"))
+      .def("sleep_to", &DeviceArena::sleep_to, py::arg("host"),
+           py::arg("chunk_bytes") = 0)
+      .def("wake_from", &DeviceArena::wake_from, py::arg("host"),
+           py::arg("chunk_bytes") = 0)
+      .def_property_readonly("data_ptr", &DeviceArena::data_ptr)
+      .def_property_readonly("is_mapped", &DeviceArena::is_mapped)
+      .def_property_readonly("uses_vmm", &DeviceArena::uses_vmm)
+      .def_property_readonly("size_bytes", &DeviceArena::size_bytes)
+      .def_property_readonly("device", &DeviceArena::device);
+}

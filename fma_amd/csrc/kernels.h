@@ -1,0 +1,29 @@
+// Shared declarations between the HIP kernels (kernels.hip) and the
+// torch-extension host code (actuator.cpp).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+// One copy descriptor: both pointers must be 16-byte aligned (enforced by
+// the host side; device allocations are 256-B aligned, arena offsets are
+// padded to FMA_ARENA_ALIGN).
+struct FmaCopyDesc {
+  const unsigned char* src;
+  unsigned char* dst;
+  unsigned long long bytes;
+};
+
+// LDS-staged prefix table bounds the descriptor count per launch:
+// (n+1) * 8 B of dynamic LDS, capped well under the 160 KiB/CU budget.
+#define FMA_MAX_DESCS_PER_LAUNCH 8192
+#define FMA_ARENA_ALIGN 256
+
+extern "C" hipError_t fma_launch_batched_copy(const FmaCopyDesc* descs_dev,
+                                              const unsigned long long* prefix_dev,
+                                              int ndesc,
+                                              unsigned long long total_units,
+                                              hipStream_t stream);
+
+extern "C" hipError_t fma_launch_contiguous_copy(const void* src, void* dst,
+                                                 unsigned long long bytes,
+                                                 hipStream_t stream);

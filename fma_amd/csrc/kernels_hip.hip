@@ -1,0 +1,133 @@
+// CDNA4 (gfx950) kernels for the FMA sleep/wake hot path.
+//
+// The job: move every parameter shard of a model between its (scattered)
+// device storages and one contiguous buffer — the "pack" direction feeds a
+// single D2H transfer into pinned host DRAM on sleep(level=1), the "scatter"
+// direction re-materializes tensors after the H2D transfer on wake_up.
+// (Native replacement for what the reference delegates to vLLM's
+// CuMemAllocator offload; reference README.md:16-26,
+// pkg/controller/dual-pods/inference-server.go:1497,1712.)
+//
+// Design notes (MI355X):
+// - Pure streaming copy => memory-bound. 16 B/lane loads/stores (uint4) give
+//   1 KiB per wave-instruction, the coalescing sweet spot; HBM3E ceiling is
+//   ~6.3 TB/s measured (MI355X_MICROARCH.md §HBM).
+// - One flat global index space of 16-byte units across all descriptors;
+//   each thread binary-searches the per-descriptor prefix table (staged in
+//   LDS, ~10 steps for <=4k tensors) and copies one unit per grid-stride
+//   iteration. Tails (<16 B) handled by the unit that owns them.
+// - The same kernel serves D2D (staging), D2H and H2D with pinned host
+//   memory mapped into the device address space — whichever wins on the
+//   hardware is chosen by the host-side pipeline (actuator.cpp).
+// - Grid sized >> 256 workgroups to fill all 8 XCDs; plain linear
+//   blockIdx->unit mapping is already XCD-interleaved for streaming.
+
+#include <hip/hip_runtime.h>
+
+#include "kernels.h"
+
+#ifndef FMA_COPY_BLOCK
+#define FMA_COPY_BLOCK 256
+#endif
+
+namespace {
+
+__global__ __launch_bounds__(FMA_COPY_BLOCK) void batched_copy_kernel(
+    const FmaCopyDesc* __restrict__ descs,
+    const unsigned long long* __restrict__ prefix,  // ndesc+1 entries, 16B units
+    int ndesc,
+    unsigned long long total_units) {
+  // Stage the prefix table in LDS: the binary search per unit touches
+  // ~log2(ndesc) words; from LDS that is ~50 cycles total, invisible next
+  // to the ~900-cycle HBM load it accompanies.
+  extern __shared__ unsigned long long s_prefix[];
+  for (int i = threadIdx.x; i <= ndesc; i += blockDim.x) {
+    s_prefix[i] = prefix[i];
+  }
+  __syncthreads();
+
+  const unsigned long long stride =
+      static_cast<unsigned long long>(gridDim.x) * blockDim.x;
+  for (unsigned long long u =
+           static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
+       u < total_units; u += stride) {
+    // largest d with prefix[d] <= u
+    int lo = 0, hi = ndesc;
+    while (hi - lo > 1) {
+      const int mid = (lo + hi) >> 1;
+      if (s_prefix[mid] <= u) {
+        lo = mid;
+      } else {
+        hi = mid;
+      }
+    }
+    const FmaCopyDesc d = descs[lo];
+    const unsigned long long byte_off = (u - s_prefix[lo]) * 16ull;
+    const unsigned char* __restrict__ s = d.src + byte_off;
+    unsigned char* __restrict__ t = d.dst + byte_off;
+    const unsigned long long rem = d.bytes - byte_off;
+    if (rem >= 16ull) {
+      *reinterpret_cast<uint4*>(t) = *reinterpret_cast<const uint4*>(s);
+    } else {
+      for (unsigned long long i = 0; i < rem; ++i) {
+        t[i] = s[i];
+      }
+    }
+  }
+}
+
+// Simple contiguous copy at max vector width; used for arena <-> staging
+// slices where no descriptor table is needed (both sides contiguous).
+__global__ __launch_bounds__(FMA_COPY_BLOCK) void contiguous_copy_kernel(
+    const unsigned char* __restrict__ src,
+    unsigned char* __restrict__ dst,
+    unsigned long long bytes) {
+  const unsigned long long total_units = bytes >> 4;
+  const unsigned long long stride =
+      static_cast<unsigned long long>(gridDim.x) * blockDim.x;
+  unsigned long long u =
+      static_cast<unsigned long long>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (; u < total_units; u += stride) {
+    reinterpret_cast<uint4*>(dst)[u] = reinterpret_cast<const uint4*>(src)[u];
+  }
+  // tail bytes by thread 0 of block 0
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    for (unsigned long long i = total_units << 4; i < bytes; ++i) {
+      dst[i] = src[i];
+    }
+  }
+}
+
+inline int copy_grid(unsigned long long total_units) {
+  // >=2048 workgroups fills 256 CUs at 8 blocks/CU; cap so tiny copies
+  // don't pay launch-size overhead.
+  unsigned long long blocks = (total_units + FMA_COPY_BLOCK - 1) / FMA_COPY_BLOCK;
+  if (blocks > 4096ull) blocks = 4096ull;
+  if (blocks == 0ull) blocks = 1ull;
+  return static_cast<int>(blocks);
+}
+
+}  // namespace
+
+extern "C" hipError_t fma_launch_batched_copy(const FmaCopyDesc* descs_dev,
+                                              const unsigned long long* prefix_dev,
+                                              int ndesc,
+                                              unsigned long long total_units,
+                                              hipStream_t stream) {
+  if (ndesc <= 0 || total_units == 0) return hipSuccess;
+  if (ndesc > FMA_MAX_DESCS_PER_LAUNCH) return hipErrorInvalidValue;
+  const size_t lds = static_cast<size_t>(ndesc + 1) * sizeof(unsigned long long);
+ hipLaunchKernelGGL(( batched_copy_kernel), dim3(copy_grid(total_units)), dim3(FMA_COPY_BLOCK), lds, stream, 
+      descs_dev, prefix_dev, ndesc, total_units);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t fma_launch_contiguous_copy(const void* src, void* dst,
+                                                 unsigned long long bytes,
+                                                 hipStream_t stream) {
+  if (bytes == 0) return hipSuccess;
+ hipLaunchKernelGGL(( contiguous_copy_kernel), dim3(copy_grid(bytes >> 4)), dim3(FMA_COPY_BLOCK), 0, stream, 
+      static_cast<const unsigned char*>(src), static_cast<unsigned char*>(dst),
+      bytes);
+  return hipGetLastError();
+}

@@ -1,0 +1,294 @@
+"""Llama-architecture model with arena-resident weights.
+
+The serving runtime materializes every parameter as a view into one
+contiguous :class:`~fma_amd.ops.actuation.ArenaActuator` region, which is
+what makes sleep(level=1)/wake_up a pure pinned-transfer (no per-tensor
+gather needed on the hot path). Weights are random-initialized — this stack
+has no network access; BASELINE.md's configs all use random-init weights of
+the named architecture — or loaded from safetensors when a file is present.
+
+Tensor-parallel layout (Megatron-style, one process per GPU over RCCL):
+q/k/v and gate/up are row-sharded (output features / tp), o and down are
+column-sharded (input features / tp) with one all-reduce each per layer;
+embeddings, norms and lm_head are replicated. The reference has no model
+code at all — parallelism there is an opaque ``--tensor-parallel-size``
+passed to vLLM (reference docs/launcher.md:584-595); here it is native.
+
+Inference only: plain tensors, no autograd.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "custom"
+    vocab_size: int = 32768
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    max_seq_len: int = 4096
+    rope_theta: float = 500000.0
+    norm_eps: float = 1e-5
+    dtype: torch.dtype = torch.bfloat16
+
+    @property
+    def head_dim(self) -> int:
+        return self.hidden_size // self.num_heads
+
+    # -- presets ------------------------------------------------------------
+
+    @staticmethod
+    def tiny() -> "LlamaConfig":
+        return LlamaConfig(name="tiny", vocab_size=512, hidden_size=64,
+                           intermediate_size=128, num_layers=2, num_heads=4,
+                           num_kv_heads=2, max_seq_len=256)
+
+    @staticmethod
+    def llama3_8b() -> "LlamaConfig":
+        return LlamaConfig(name="llama-3-8b", vocab_size=128256,
+                           hidden_size=4096, intermediate_size=14336,
+                           num_layers=32, num_heads=32, num_kv_heads=8,
+                           max_seq_len=8192)
+
+    @staticmethod
+    def llama3_70b() -> "LlamaConfig":
+        return LlamaConfig(name="llama-3-70b", vocab_size=128256,
+                           hidden_size=8192, intermediate_size=28672,
+                           num_layers=80, num_heads=64, num_kv_heads=8,
+                           max_seq_len=8192)
+
+    @staticmethod
+    def from_total_gib(gib: float, dtype: torch.dtype = torch.bfloat16
+                       ) -> "LlamaConfig":
+        """Synthetic 70B-shaped config whose total parameter bytes are
+        ~`gib` GiB — used for BASELINE's '64 GiB of tensors' wake metric.
+        Small targets shrink the base dims so layer count stays sane."""
+        if gib >= 16:
+            dims = dict(vocab_size=32768, hidden_size=8192,
+                        intermediate_size=28672, num_heads=64, num_kv_heads=8)
+        elif gib >= 2:
+            dims = dict(vocab_size=32768, hidden_size=4096,
+                        intermediate_size=14336, num_heads=32, num_kv_heads=8)
+        else:
+            dims = dict(vocab_size=4096, hidden_size=1024,
+                        intermediate_size=3584, num_heads=8, num_kv_heads=8)
+        cfg = LlamaConfig(name=f"synthetic-{gib:g}gib", num_layers=1,
+                          max_seq_len=4096, dtype=dtype, **dims)
+        esize = torch.empty(0, dtype=dtype).element_size()
+        fixed = (2 * cfg.vocab_size * cfg.hidden_size + cfg.hidden_size) * esize
+        kv_dim = cfg.num_kv_heads * cfg.head_dim
+        per_layer = esize * (
+            2 * cfg.hidden_size * cfg.hidden_size            # wq, wo
+            + 2 * kv_dim * cfg.hidden_size                   # wk, wv
+            + 3 * cfg.hidden_size * cfg.intermediate_size    # gate, up, down
+            + 2 * cfg.hidden_size)                           # norms
+        target = gib * (1 << 30)
+        cfg.num_layers = max(1, round((target - fixed) / per_layer))
+        return cfg
+
+    @staticmethod
+    def by_name(name: str) -> "LlamaConfig":
+        presets = {
+            "tiny": LlamaConfig.tiny,
+            "llama-3-8b": LlamaConfig.llama3_8b,
+            "llama-3-70b": LlamaConfig.llama3_70b,
+        }
+        if name in presets:
+            return presets[name]()
+        if name.startswith("synthetic-") and name.endswith("gib"):
+            return LlamaConfig.from_total_gib(float(name[len("synthetic-"):-3]))
+        raise KeyError(f"unknown model preset {name!r}")
+
+    # -- parameter schema ---------------------------------------------------
+
+    def param_specs(self, tp_rank: int = 0, tp_size: int = 1
+                    ) -> List[Tuple[str, Tuple[int, ...], torch.dtype]]:
+        """Names/shapes of this rank's parameter shard, in layout order."""
+        assert self.num_heads % tp_size == 0, "heads must divide tp"
+        assert self.num_kv_heads % tp_size == 0 or tp_size <= self.num_kv_heads, \
+            "kv heads must divide tp"
+        h = self.hidden_size
+        hd = self.head_dim
+        q_local = self.num_heads // tp_size * hd
+        kv_local = max(self.num_kv_heads // tp_size, 1) * hd
+        i_local = self.intermediate_size // tp_size
+        d = self.dtype
+        specs: List[Tuple[str, Tuple[int, ...], torch.dtype]] = [
+            ("embed.weight", (self.vocab_size, h), d),
+        ]
+        for li in range(self.num_layers):
+            p = f"layers.{li}."
+            specs += [
+                (p + "attn_norm.weight", (h,), d),
+                (p + "wq.weight", (q_local, h), d),
+                (p + "wk.weight", (kv_local, h), d),
+                (p + "wv.weight", (kv_local, h), d),
+                (p + "wo.weight", (h, q_local), d),
+                (p + "mlp_norm.weight", (h,), d),
+                (p + "w_gate.weight", (i_local, h), d),
+                (p + "w_up.weight", (i_local, h), d),
+                (p + "w_down.weight", (h, i_local), d),
+            ]
+        specs += [
+            ("final_norm.weight", (h,), d),
+            ("lm_head.weight", (self.vocab_size, h), d),
+        ]
+        return specs
+
+    def total_param_bytes(self, tp_rank: int = 0, tp_size: int = 1) -> int:
+        esize = torch.empty(0, dtype=self.dtype).element_size()
+        return sum(esize * math.prod(s) for _, s, _ in
+                   self.param_specs(tp_rank, tp_size))
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+    xf = x.float()
+    xf = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (xf * w.float()).to(x.dtype)
+
+
+def precompute_rope(cfg: LlamaConfig, device) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Host-precomputed cos/sin tables (trig on device turns a memory-bound
+    op VALU-bound — cdna_hip_programming.md Appendix B / element-wise)."""
+    hd = cfg.head_dim
+    inv = 1.0 / (cfg.rope_theta ** (torch.arange(0, hd, 2, dtype=torch.float32) / hd))
+    t = torch.arange(cfg.max_seq_len, dtype=torch.float32)
+    freqs = torch.outer(t, inv)
+    return freqs.cos().to(device), freqs.sin().to(device)
+
+
+def apply_rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+               start_pos: int) -> torch.Tensor:
+    # x: [B, T, heads, head_dim]
+    B, T, H, D = x.shape
+    c = cos[start_pos:start_pos + T].view(1, T, 1, D // 2)
+    s = sin[start_pos:start_pos + T].view(1, T, 1, D // 2)
+    xf = x.float().view(B, T, H, D // 2, 2)
+    x0, x1 = xf[..., 0], xf[..., 1]
+    out = torch.stack((x0 * c - x1 * s, x0 * s + x1 * c), dim=-1)
+    return out.view(B, T, H, D).to(x.dtype)
+
+
+class KVCache:
+    def __init__(self, cfg: LlamaConfig, batch: int, device,
+                 tp_size: int = 1, max_seq: Optional[int] = None):
+        kv_heads = max(cfg.num_kv_heads // tp_size, 1)
+        seq = max_seq or cfg.max_seq_len
+        shape = (cfg.num_layers, 2, batch, seq, kv_heads, cfg.head_dim)
+        self.data = torch.zeros(shape, dtype=cfg.dtype, device=device)
+        self.seq_len = 0
+
+    def free(self) -> None:
+        self.data = None  # type: ignore[assignment]
+
+
+class LlamaModel:
+    """Functional Llama over a dict of (arena-view) tensors."""
+
+    def __init__(self, cfg: LlamaConfig, params: Dict[str, torch.Tensor],
+                 device, tp_rank: int = 0, tp_size: int = 1,
+                 tp_group: Optional[object] = None):
+        self.cfg = cfg
+        self.params = params
+        self.device = device
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        self.tp_group = tp_group
+        self.rope_cos, self.rope_sin = precompute_rope(cfg, device)
+
+    def rebind(self, params: Dict[str, torch.Tensor]) -> None:
+        """Swap in fresh views after a non-VMM wake changed the arena base."""
+        self.params = params
+
+    def init_weights(self, seed: int = 0) -> None:
+        torch.manual_seed(seed + self.tp_rank)
+        for name, p in self.params.items():
+            if p.dim() > 1:
+                p.normal_(0.0, 0.02)  # in-place, device-side RNG: no temps
+            else:
+                p.fill_(1.0)  # norm gains
+
+    def _maybe_all_reduce(self, x: torch.Tensor) -> torch.Tensor:
+        if self.tp_size > 1:
+            dist.all_reduce(x, group=self.tp_group)
+        return x
+
+    @torch.no_grad()
+    def forward(self, tokens: torch.Tensor, cache: Optional[KVCache] = None,
+                start_pos: int = 0) -> torch.Tensor:
+        """tokens [B, T] -> logits [B, T, vocab]."""
+        cfg = self.cfg
+        P = self.params
+        B, T = tokens.shape
+        q_heads = cfg.num_heads // self.tp_size
+        kv_heads = max(cfg.num_kv_heads // self.tp_size, 1)
+        hd = cfg.head_dim
+
+        x = F.embedding(tokens, P["embed.weight"])
+        for li in range(cfg.num_layers):
+            p = f"layers.{li}."
+            h = rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps)
+            q = F.linear(h, P[p + "wq.weight"]).view(B, T, q_heads, hd)
+            k = F.linear(h, P[p + "wk.weight"]).view(B, T, kv_heads, hd)
+            v = F.linear(h, P[p + "wv.weight"]).view(B, T, kv_heads, hd)
+            q = apply_rope(q, self.rope_cos, self.rope_sin, start_pos)
+            k = apply_rope(k, self.rope_cos, self.rope_sin, start_pos)
+            if cache is not None:
+                cache.data[li, 0, :, start_pos:start_pos + T] = k
+                cache.data[li, 1, :, start_pos:start_pos + T] = v
+                k = cache.data[li, 0, :, : start_pos + T]
+                v = cache.data[li, 1, :, : start_pos + T]
+            # SDPA wants [B, heads, T, hd]
+            qh = q.transpose(1, 2)
+            kh = k.transpose(1, 2)
+            vh = v.transpose(1, 2)
+            if kv_heads != q_heads:
+                rep = q_heads // kv_heads
+                kh = kh.repeat_interleave(rep, dim=1)
+                vh = vh.repeat_interleave(rep, dim=1)
+            att = F.scaled_dot_product_attention(
+                qh, kh, vh, is_causal=(T > 1))
+            att = att.transpose(1, 2).reshape(B, T, q_heads * hd)
+            x = x + self._maybe_all_reduce(F.linear(att, P[p + "wo.weight"]))
+
+            h = rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
+            gate = F.linear(h, P[p + "w_gate.weight"])
+            up = F.linear(h, P[p + "w_up.weight"])
+            x = x + self._maybe_all_reduce(
+                F.linear(F.silu(gate) * up, P[p + "w_down.weight"]))
+
+        x = rmsnorm(x, P["final_norm.weight"], cfg.norm_eps)
+        return F.linear(x, P["lm_head.weight"]).float()
+
+    @torch.no_grad()
+    def generate(self, prompt: torch.Tensor, max_new_tokens: int = 16,
+                 cache: Optional[KVCache] = None) -> torch.Tensor:
+        """Greedy decode. prompt [B, T] -> [B, T + max_new_tokens]."""
+        B, T = prompt.shape
+        own_cache = cache is None
+        if own_cache:
+            cache = KVCache(self.cfg, B, self.device, self.tp_size,
+                            max_seq=min(self.cfg.max_seq_len,
+                                        T + max_new_tokens))
+        logits = self.forward(prompt, cache, 0)
+        out = prompt
+        pos = T
+        for _ in range(max_new_tokens):
+            nxt = logits[:, -1].argmax(-1, keepdim=True)
+            out = torch.cat([out, nxt], dim=1)
+            logits = self.forward(nxt, cache, pos)
+            pos += 1
+        if own_cache:
+            cache.free()
+        return out

@@ -1,0 +1,228 @@
+"""Python API over the MI355X actuator extension (fma_amd._C).
+
+Two actuation strategies, both moving model tensors between GPU HBM3E and
+pinned host DRAM (the native replacement for vLLM's sleep level=1 the
+reference drives over HTTP — reference README.md:16-26,
+pkg/controller/dual-pods/inference-server.go:1497,1712):
+
+- :class:`ArenaActuator` — parameters live as views into one contiguous
+  device arena, so sleep/wake is a pure chunked pinned-memcpy (plus, in VMM
+  mode, physical map/unmap at a constant virtual address). This is the fast
+  path used by the serving runtime for models it materializes itself.
+- :class:`PackActuator` — for tensors scattered across the caching
+  allocator: the HIP gather kernel coalesces every shard into one flat
+  space for a single pinned D2H per chunk; wake re-allocates the storages
+  and scatters back.
+
+On machines without a GPU (CI), :class:`FakeArenaActuator` emulates the
+same interface on CPU memory so the serving runtime, launcher and
+controllers are fully testable (the mock seam the reference achieves with
+its CPU vLLM image — reference docs/launcher.md:656-707).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import torch
+
+ARENA_ALIGN = 256
+
+# staged / direct / per_tensor map to XferMode in csrc/actuator.cpp
+MODE_STAGED = 0
+MODE_DIRECT = 1
+MODE_PER_TENSOR = 2
+
+_C = None
+_IMPORT_ERROR: Optional[BaseException] = None
+try:
+    from fma_amd import _C as _C  # type: ignore[attr-defined, no-redef]
+except Exception as e:  # pragma: no cover - exercised only when ext missing
+    _IMPORT_ERROR = e
+
+
+def native_available() -> bool:
+    return _C is not None
+
+
+def require_native():
+    """The HIP extension is mandatory whenever a GPU is present: silently
+    falling back to an eager path on real hardware would invalidate every
+    latency number, so fail loudly instead."""
+    if _C is None:
+        raise ImportError(
+            "fma_amd._C (HIP actuator extension) is not built; run "
+            "`python setup.py build_ext --inplace` (PYTORCH_ROCM_ARCH=gfx950). "
+            f"Original import error: {_IMPORT_ERROR!r}"
+        )
+    return _C
+
+
+def align_up(n: int, a: int = ARENA_ALIGN) -> int:
+    return (n + a - 1) // a * a
+
+
+def plan_layout(specs: Sequence[Tuple[str, Tuple[int, ...], torch.dtype]]
+                ) -> Tuple[Dict[str, Tuple[int, Tuple[int, ...], torch.dtype]], int]:
+    """Assign 256-B-aligned flat offsets to named tensor specs.
+
+    Returns ({name: (offset, shape, dtype)}, total_bytes).
+    """
+    layout: Dict[str, Tuple[int, Tuple[int, ...], torch.dtype]] = {}
+    off = 0
+    for name, shape, dtype in specs:
+        numel = 1
+        for s in shape:
+            numel *= s
+        nbytes = numel * torch.empty(0, dtype=dtype).element_size()
+        layout[name] = (off, tuple(shape), dtype)
+        off += align_up(max(nbytes, 1))
+    return layout, off
+
+
+def alloc_pinned(nbytes: int) -> torch.Tensor:
+    """One pinned host buffer. Allocated once per instance and reused across
+    every sleep/wake cycle (pinning cost stays off the wake path). Plain
+    host memory on GPU-less machines (tests)."""
+    if torch.cuda.is_available():
+        return torch.empty(nbytes, dtype=torch.uint8, pin_memory=True)
+    return torch.empty(nbytes, dtype=torch.uint8)
+
+
+class ArenaActuator:
+    """Contiguous device arena with sleep/wake via one pinned host buffer."""
+
+    def __init__(self, nbytes: int, device: int = 0, try_vmm: bool = True,
+                 chunk_bytes: int = 0):
+        C = require_native()
+        self._arena = C.DeviceArena(nbytes, device, try_vmm)
+        self.device = device
+        self.nbytes = nbytes
+        self.chunk_bytes = chunk_bytes
+        self._base_at_view: int = self._arena.data_ptr
+
+    @property
+    def uses_vmm(self) -> bool:
+        return self._arena.uses_vmm
+
+    @property
+    def is_mapped(self) -> bool:
+        return self._arena.is_mapped
+
+    def view(self, offset: int, shape: Tuple[int, ...], dtype: torch.dtype
+             ) -> torch.Tensor:
+        return self._arena.view(offset, list(shape), dtype)
+
+    def sleep(self, host: torch.Tensor) -> float:
+        return self._arena.sleep_to(host, self.chunk_bytes)
+
+    def wake(self, host: torch.Tensor) -> Tuple[float, bool]:
+        """Returns (seconds, views_invalidated). With VMM backing the VA is
+        constant and views survive; otherwise the caller must re-bind."""
+        t = self._arena.wake_from(host, self.chunk_bytes)
+        invalidated = self._arena.data_ptr != self._base_at_view
+        self._base_at_view = self._arena.data_ptr
+        return t, invalidated
+
+
+class FakeArenaActuator:
+    """CPU emulation of ArenaActuator for GPU-less tests.
+
+    Guarded by FMA_FAKE_GPU=1 (or explicit construction) so a missing
+    extension on a real GPU box can never silently fall through to this.
+    """
+
+    def __init__(self, nbytes: int, device: int = 0, try_vmm: bool = True,
+                 chunk_bytes: int = 0):
+        if not (os.environ.get("FMA_FAKE_GPU") == "1"
+                or not torch.cuda.is_available()):
+            raise RuntimeError("FakeArenaActuator is only for GPU-less machines")
+        self.device = device
+        self.nbytes = nbytes
+        self.chunk_bytes = chunk_bytes
+        self._buf = torch.zeros(nbytes, dtype=torch.uint8)
+        self._asleep = False
+        self.uses_vmm = True  # emulates the constant-VA behavior
+
+    @property
+    def is_mapped(self) -> bool:
+        return not self._asleep
+
+    def view(self, offset: int, shape: Tuple[int, ...], dtype: torch.dtype
+             ) -> torch.Tensor:
+        assert not self._asleep, "arena is asleep"
+        numel = 1
+        for s in shape:
+            numel *= s
+        nbytes = numel * torch.empty(0, dtype=dtype).element_size()
+        return self._buf[offset:offset + nbytes].view(dtype).view(shape)
+
+    def sleep(self, host: torch.Tensor) -> float:
+        assert not self._asleep, "arena already asleep"
+        host[: self.nbytes].copy_(self._buf)
+        self._buf.zero_()  # poison: reads while asleep are wrong by design
+        self._asleep = True
+        return 1e-9
+
+    def wake(self, host: torch.Tensor) -> Tuple[float, bool]:
+        assert self._asleep, "arena already awake"
+        self._buf.copy_(host[: self.nbytes])
+        self._asleep = False
+        return 1e-9, False
+
+
+class PackActuator:
+    """Sleep/wake for tensors scattered across the caching allocator.
+
+    sleep: HIP gather kernel -> staging -> pinned host, then storages are
+    resized to zero and the caching allocator's reserve is returned to the
+    system so another instance can use the HBM.
+    wake: storages re-allocated, pinned host -> staging -> HIP scatter.
+    """
+
+    def __init__(self, tensors: Dict[str, torch.Tensor],
+                 mode: int = MODE_STAGED, chunk_bytes: int = 0):
+        require_native()
+        self.names: List[str] = sorted(tensors.keys())
+        self.tensors = tensors
+        self.mode = mode
+        self.chunk_bytes = chunk_bytes
+        off = 0
+        self.offsets: List[int] = []
+        for n in self.names:
+            self.offsets.append(off)
+            off += align_up(max(tensors[n].nbytes, 1))
+        self.total_bytes = off
+        self.asleep = False
+
+    def _tensor_list(self) -> List[torch.Tensor]:
+        return [self.tensors[n] for n in self.names]
+
+    def sleep(self, host: torch.Tensor) -> float:
+        C = require_native()
+        t = C.pack_to_host(self._tensor_list(), self.offsets, host,
+                           self.mode, self.chunk_bytes)
+        for n in self.names:
+            self.tensors[n].untyped_storage().resize_(0)
+        torch.cuda.empty_cache()
+        self.asleep = True
+        return t
+
+    def wake(self, host: torch.Tensor) -> float:
+        C = require_native()
+        for n in self.names:
+            t = self.tensors[n]
+            t.untyped_storage().resize_(t.numel() * t.element_size())
+        t = C.restore_from_host(self._tensor_list(), self.offsets, host,
+                                self.mode, self.chunk_bytes)
+        self.asleep = False
+        return t
+
+
+def make_arena(nbytes: int, device: int = 0, try_vmm: bool = True,
+               chunk_bytes: int = 0):
+    """Arena factory: native on a GPU machine, fake on CPU-only machines."""
+    if torch.cuda.is_available():
+        return ArenaActuator(nbytes, device, try_vmm, chunk_bytes)
+    return FakeArenaActuator(nbytes, device, try_vmm, chunk_bytes)

@@ -1,0 +1,154 @@
+"""Actuation engine: a model instance with fast sleep(level=1)/wake_up.
+
+This is the MI355X-native replacement for the slice of vLLM the reference
+depends on (sleep mode + a serving endpoint; reference README.md:16-33,
+docs/dual-pods.md:618-627). One engine == one model instance on one GPU
+(or one TP rank of it):
+
+- parameters are views into a contiguous DeviceArena (ops/actuation.py), so
+  ``sleep`` is chunked pinned D2H + physical-HBM release and ``wake_up`` is
+  physical re-map overlapped with pinned H2D — no allocator round-trips, no
+  module re-init on the hot path;
+- the pinned host buffer is allocated once at instance creation (the
+  launcher's create step), keeping hipHostMalloc cost off the wake path;
+- for TP instances every rank restores its own shard locally (no inter-GPU
+  weight traffic over xGMI) and an RCCL barrier gates the transition to
+  ``is_sleeping == False`` so all ranks re-enter serving together.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Dict, Optional
+
+import torch
+import torch.distributed as dist
+
+from fma_amd.models.llama import KVCache, LlamaConfig, LlamaModel
+from fma_amd.ops import actuation
+
+
+class ActuationEngine:
+    AWAKE = "awake"
+    SLEEPING = "sleeping"
+
+    def __init__(self, cfg: LlamaConfig, device_index: int = 0,
+                 tp_rank: int = 0, tp_size: int = 1, tp_group=None,
+                 use_vmm: bool = True, chunk_bytes: int = 0, seed: int = 0,
+                 init_weights: bool = True):
+        self.cfg = cfg
+        self.device_index = device_index
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        self.tp_group = tp_group
+        self.on_gpu = torch.cuda.is_available()
+        self.device = torch.device("cuda", device_index) if self.on_gpu \
+            else torch.device("cpu")
+        if self.on_gpu:
+            torch.cuda.set_device(device_index)
+
+        specs = cfg.param_specs(tp_rank, tp_size)
+        self.layout, self.total_bytes = actuation.plan_layout(specs)
+        t0 = time.perf_counter()
+        self.arena = actuation.make_arena(self.total_bytes, device_index,
+                                          try_vmm=use_vmm,
+                                          chunk_bytes=chunk_bytes)
+        self.params = self._make_views()
+        self.model = LlamaModel(cfg, self.params, self.device,
+                                tp_rank, tp_size, tp_group)
+        if init_weights:
+            self.model.init_weights(seed)
+        self.host = actuation.alloc_pinned(self.total_bytes)
+        if self.on_gpu:
+            torch.cuda.synchronize(self.device)
+        self.create_seconds = time.perf_counter() - t0
+
+        self.state = self.AWAKE
+        self.sleep_count = 0
+        self.wake_count = 0
+        self.last_sleep_seconds: Optional[float] = None
+        self.last_wake_seconds: Optional[float] = None
+
+    # -- lifecycle -----------------------------------------------------------
+
+    def _make_views(self) -> Dict[str, torch.Tensor]:
+        return {name: self.arena.view(off, shape, dtype)
+                for name, (off, shape, dtype) in self.layout.items()}
+
+    def is_sleeping(self) -> bool:
+        return self.state == self.SLEEPING
+
+    def sleep(self, level: int = 1) -> float:
+        """Offload weights to pinned host DRAM and free the HBM.
+
+        level=1 == vLLM semantics: weights survive in host memory, KV/cache
+        state is discarded (reference README.md:16-26). level=2 would drop
+        weights too; we keep the host copy either way (it is the wake
+        source) but level is accepted for wire compatibility.
+        """
+        if self.state == self.SLEEPING:
+            return 0.0
+        t = self.arena.sleep(self.host)
+        if self.on_gpu:
+            # return caching-allocator reserves (activations, KV) so another
+            # instance's wake can claim the HBM
+            torch.cuda.empty_cache()
+        self.state = self.SLEEPING
+        self.sleep_count += 1
+        self.last_sleep_seconds = t
+        return t
+
+    def wake_up(self) -> float:
+        """Restore weights to HBM; all TP ranks synchronize before the
+        instance reports itself awake."""
+        if self.state == self.AWAKE:
+            return 0.0
+        t0 = time.perf_counter()
+        _, invalidated = self.arena.wake(self.host)
+        if invalidated:
+            # non-VMM fallback: arena base moved; re-point the views
+            self.params = self._make_views()
+            self.model.rebind(self.params)
+        if self.tp_size > 1 and dist.is_initialized():
+            dist.barrier(group=self.tp_group)
+        t = time.perf_counter() - t0
+        self.state = self.AWAKE
+        self.wake_count += 1
+        self.last_wake_seconds = t
+        return t
+
+    # -- serving -------------------------------------------------------------
+
+    @torch.no_grad()
+    def generate(self, tokens: torch.Tensor, max_new_tokens: int = 16
+                 ) -> torch.Tensor:
+        if self.state != self.AWAKE:
+            raise RuntimeError("engine is sleeping")
+        return self.model.generate(tokens.to(self.device), max_new_tokens)
+
+    def generate_text(self, prompt: str, max_new_tokens: int = 16) -> str:
+        """Byte-level round trip (no tokenizer assets offline): UTF-8 bytes
+        are the token ids; generated ids map back to bytes."""
+        ids = [b % self.cfg.vocab_size for b in prompt.encode("utf-8")] or [1]
+        toks = torch.tensor([ids], dtype=torch.long, device=self.device)
+        out = self.generate(toks, max_new_tokens)[0, len(ids):]
+        return bytes(int(t) % 256 for t in out.tolist()).decode(
+            "utf-8", errors="replace")
+
+    def new_kv_cache(self, batch: int, max_seq: Optional[int] = None) -> KVCache:
+        return KVCache(self.cfg, batch, self.device, self.tp_size, max_seq)
+
+    def stats(self) -> Dict[str, object]:
+        return {
+            "model": self.cfg.name,
+            "state": self.state,
+            "param_bytes": self.total_bytes,
+            "tp_rank": self.tp_rank,
+            "tp_size": self.tp_size,
+            "uses_vmm": getattr(self.arena, "uses_vmm", False),
+            "sleep_count": self.sleep_count,
+            "wake_count": self.wake_count,
+            "last_sleep_seconds": self.last_sleep_seconds,
+            "last_wake_seconds": self.last_wake_seconds,
+            "create_seconds": self.create_seconds,
+        }

@@ -1,0 +1,22 @@
+import os
+import sys
+
+import pytest
+
+# repo root importable when pytest runs from anywhere
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: needs a real MI355X (run via gpurun / driver)")
+
+
+def pytest_collection_modifyitems(config, items):
+    import torch
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason="no GPU on this machine")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)

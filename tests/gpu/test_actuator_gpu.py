@@ -1,0 +1,176 @@
+"""GPU tests for the HIP actuator: numerics vs plain PyTorch reference.
+
+Every test compares the HIP pack/scatter/arena paths against a plain
+PyTorch (fp32/CPU) reference of the same data movement.
+"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def C():
+    from fma_amd.ops import actuation
+    # native extension is mandatory on a GPU box — no silent fallback
+    return actuation.require_native()
+
+
+def scattered_tensors(device="cuda:0", seed=3):
+    """Mix of sizes incl. non-16B-multiple tails and tiny tensors."""
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    shapes = [(1024, 1024), (17,), (3, 5, 7), (4096, 128), (1,), (333, 9)]
+    dtypes = [torch.bfloat16, torch.float32, torch.bfloat16, torch.float16,
+              torch.float32, torch.bfloat16]
+    ts = []
+    for shape, dt in zip(shapes, dtypes):
+        cpu = torch.randn(shape, generator=g, dtype=torch.float32).to(dt)
+        ts.append(cpu.to(device))
+    return ts
+
+
+def flat_reference(tensors, offsets, total):
+    """Plain PyTorch reference for the pack layout."""
+    buf = torch.zeros(total, dtype=torch.uint8)
+    for t, off in zip(tensors, offsets):
+        raw = t.detach().cpu().contiguous().view(torch.uint8).view(-1)
+        buf[off:off + raw.numel()] = raw
+    return buf
+
+
+@pytest.mark.parametrize("mode", [0, 1, 2], ids=["staged", "direct", "per_tensor"])
+def test_pack_matches_reference(C, mode):
+    from fma_amd.ops.actuation import align_up
+    ts = scattered_tensors()
+    offsets, off = [], 0
+    for t in ts:
+        offsets.append(off)
+        off += align_up(max(t.nbytes, 1))
+    host = torch.empty(off, dtype=torch.uint8, pin_memory=True)
+    host.fill_(0)
+    if mode == 1:
+        try:
+            C.pack_to_host(ts, offsets, host, mode, 1 << 20)
+        except RuntimeError as e:
+            pytest.skip(f"direct host mapping unavailable: {e}")
+    else:
+        C.pack_to_host(ts, offsets, host, mode, 1 << 20)
+    ref = flat_reference(ts, offsets, off)
+    for t, o in zip(ts, offsets):
+        assert torch.equal(host[o:o + t.nbytes], ref[o:o + t.nbytes])
+
+
+@pytest.mark.parametrize("mode", [0, 1, 2], ids=["staged", "direct", "per_tensor"])
+def test_restore_matches_reference(C, mode):
+    from fma_amd.ops.actuation import align_up
+    ts = scattered_tensors()
+    originals = [t.clone() for t in ts]
+    offsets, off = [], 0
+    for t in ts:
+        offsets.append(off)
+        off += align_up(max(t.nbytes, 1))
+    host = torch.empty(off, dtype=torch.uint8, pin_memory=True)
+    C.pack_to_host(ts, offsets, host, 0, 1 << 20)
+    for t in ts:
+        t.zero_()
+    if mode == 1:
+        try:
+            C.restore_from_host(ts, offsets, host, mode, 1 << 20)
+        except RuntimeError as e:
+            pytest.skip(f"direct host mapping unavailable: {e}")
+    else:
+        C.restore_from_host(ts, offsets, host, mode, 1 << 20)
+    torch.cuda.synchronize()
+    for t, o in zip(ts, originals):
+        assert torch.equal(t, o)
+
+
+def test_arena_sleep_wake_bit_exact(C):
+    nbytes = 64 << 20
+    arena = C.DeviceArena(nbytes, 0, True)
+    v = arena.view(0, [nbytes // 2], torch.bfloat16)
+    v.normal_()
+    snap = v.clone()
+    host = torch.empty(nbytes, dtype=torch.uint8, pin_memory=True)
+    t_sleep = arena.sleep_to(host, 8 << 20)
+    assert not arena.is_mapped
+    t_wake = arena.wake_from(host, 8 << 20)
+    assert arena.is_mapped
+    v2 = arena.view(0, [nbytes // 2], torch.bfloat16)
+    assert torch.equal(v2, snap)
+    assert t_sleep > 0 and t_wake > 0
+
+
+def test_arena_vmm_constant_va(C):
+    """With VMM backing, views must survive sleep/wake (constant VA)."""
+    if not C.device_supports_vmm(0):
+        pytest.skip("no VMM support on this device")
+    nbytes = 16 << 20
+    arena = C.DeviceArena(nbytes, 0, True)
+    assert arena.uses_vmm
+    v = arena.view(0, [nbytes // 4], torch.float32)
+    v.fill_(1.25)
+    base0 = arena.data_ptr
+    host = torch.empty(nbytes, dtype=torch.uint8, pin_memory=True)
+    arena.sleep_to(host, 0)
+    arena.wake_from(host, 0)
+    assert arena.data_ptr == base0
+    # the ORIGINAL view (not a fresh one) still reads restored data
+    assert torch.equal(v, torch.full_like(v, 1.25))
+
+
+def test_arena_frees_hbm_during_sleep(C):
+    free0, _ = C.device_mem_info(0)
+    nbytes = 2 << 30
+    arena = C.DeviceArena(nbytes, 0, True)
+    free_mapped, _ = C.device_mem_info(0)
+    assert free0 - free_mapped >= nbytes * 0.9
+    host = torch.empty(nbytes, dtype=torch.uint8, pin_memory=True)
+    arena.sleep_to(host, 0)
+    free_sleeping, _ = C.device_mem_info(0)
+    assert free_sleeping - free_mapped >= nbytes * 0.9, \
+        "sleep did not release physical HBM"
+    arena.wake_from(host, 0)
+    del arena
+
+
+def test_engine_sleep_wake_gpu():
+    from fma_amd.models.llama import LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+    cfg = LlamaConfig.tiny()
+    cfg.dtype = torch.bfloat16
+    eng = ActuationEngine(cfg, 0, seed=11)
+    assert eng.on_gpu
+    toks = torch.randint(0, cfg.vocab_size, (1, 8), device="cuda:0")
+    before = eng.model.forward(toks).clone()
+    eng.sleep()
+    assert eng.is_sleeping()
+    eng.wake_up()
+    after = eng.model.forward(toks)
+    assert torch.equal(before, after)
+    out = eng.generate(toks, 4)
+    assert out.shape == (1, 12)
+
+
+def test_pack_actuator_frees_memory():
+    from fma_amd.ops.actuation import PackActuator, alloc_pinned
+    from fma_amd import _C as C
+    ts = {f"t{i}": torch.randn(1 << 20, device="cuda:0") for i in range(8)}
+    snap = {k: v.clone() for k, v in ts.items()}
+    act = PackActuator(ts)
+    host = alloc_pinned(act.total_bytes)
+    act.sleep(host)
+    assert all(v.untyped_storage().size() == 0 for v in ts.values())
+    act.wake(host)
+    torch.cuda.synchronize()
+    for k in ts:
+        assert torch.equal(ts[k], snap[k])
+
+
+def test_native_extension_is_loaded():
+    """Guard against the silent-eager-fallback failure mode."""
+    import fma_amd._C as C
+    assert hasattr(C, "DeviceArena")
+    import torch
+    assert torch.version.hip is not None

@@ -1,0 +1,125 @@
+"""CPU tests for the actuation engine + llama model (fake arena).
+
+Numerics here establish that sleep->wake is bit-exact: the model's outputs
+before sleep and after wake must be identical, which is the contract the
+controller relies on when it treats a sleeping instance as a checkpoint
+(reference SURVEY §5 'the sleeping instance itself is the checkpoint')."""
+
+import os
+
+import pytest
+import torch
+
+os.environ.setdefault("FMA_FAKE_GPU", "1")
+
+from fma_amd.models.llama import LlamaConfig, LlamaModel  # noqa: E402
+from fma_amd.ops import actuation  # noqa: E402
+from fma_amd.runtime.engine import ActuationEngine  # noqa: E402
+
+
+@pytest.fixture()
+def engine():
+    return ActuationEngine(LlamaConfig.tiny(), seed=7)
+
+
+def test_layout_alignment():
+    cfg = LlamaConfig.tiny()
+    layout, total = actuation.plan_layout(cfg.param_specs())
+    assert total > 0
+    for name, (off, shape, dtype) in layout.items():
+        assert off % actuation.ARENA_ALIGN == 0
+    # offsets strictly increasing in spec order and non-overlapping
+    offs = [v[0] for v in layout.values()]
+    assert offs == sorted(offs)
+
+
+def test_param_count_8b_shape():
+    cfg = LlamaConfig.llama3_8b()
+    total = cfg.total_param_bytes()
+    nparams = total / 2  # bf16
+    assert 7.5e9 < nparams < 8.6e9  # llama-3-8b is ~8.0B params
+
+
+def test_synthetic_config_hits_target_bytes():
+    cfg = LlamaConfig.from_total_gib(64)
+    total = cfg.total_param_bytes()
+    assert abs(total - 64 * (1 << 30)) / (64 * (1 << 30)) < 0.02
+
+
+def test_tp_sharding_splits_params():
+    cfg = LlamaConfig.llama3_8b()
+    full = cfg.total_param_bytes()
+    shard = cfg.total_param_bytes(0, 8)
+    # replicated embed/head keep the shard above full/8
+    assert full / 8 < shard < full / 2
+
+
+def test_forward_shapes(engine):
+    toks = torch.randint(0, engine.cfg.vocab_size, (2, 5))
+    logits = engine.model.forward(toks)
+    assert logits.shape == (2, 5, engine.cfg.vocab_size)
+    assert torch.isfinite(logits).all()
+
+
+def test_generate(engine):
+    toks = torch.randint(0, engine.cfg.vocab_size, (1, 4))
+    out = engine.generate(toks, max_new_tokens=3)
+    assert out.shape == (1, 7)
+
+
+def test_generate_matches_kv_free_forward(engine):
+    """Decode with KV cache must agree with full-context forward."""
+    toks = torch.randint(0, engine.cfg.vocab_size, (1, 6))
+    out = engine.generate(toks, max_new_tokens=2)
+    # recompute: greedy next token from full forward at each step
+    seq = toks
+    for _ in range(2):
+        logits = engine.model.forward(seq, cache=None, start_pos=0)
+        nxt = logits[:, -1].argmax(-1, keepdim=True)
+        seq = torch.cat([seq, nxt], dim=1)
+    assert torch.equal(out, seq)
+
+
+def test_sleep_wake_bit_exact(engine):
+    toks = torch.randint(0, engine.cfg.vocab_size, (1, 4))
+    before = engine.model.forward(toks).clone()
+    snap = {n: p.clone() for n, p in engine.params.items()}
+
+    t_sleep = engine.sleep()
+    assert engine.is_sleeping()
+    assert t_sleep >= 0
+    with pytest.raises(AssertionError):
+        engine.arena.view(0, (1,), torch.uint8)  # asleep arena unreadable
+
+    t_wake = engine.wake_up()
+    assert not engine.is_sleeping()
+    assert t_wake >= 0
+    for n, p in engine.params.items():
+        assert torch.equal(p, snap[n]), f"param {n} corrupted by sleep/wake"
+    after = engine.model.forward(toks)
+    assert torch.equal(before, after)
+
+
+def test_double_sleep_wake_idempotent(engine):
+    engine.sleep()
+    assert engine.sleep() == 0.0
+    engine.wake_up()
+    assert engine.wake_up() == 0.0
+    assert engine.sleep_count == 1 and engine.wake_count == 1
+
+
+def test_generate_while_sleeping_raises(engine):
+    engine.sleep()
+    with pytest.raises(RuntimeError):
+        engine.generate(torch.zeros(1, 1, dtype=torch.long))
+
+
+def test_generate_text_roundtrip(engine):
+    out = engine.generate_text("hello", max_new_tokens=4)
+    assert isinstance(out, str)
+
+
+def test_stats_shape(engine):
+    s = engine.stats()
+    assert s["state"] == "awake"
+    assert s["param_bytes"] == engine.total_bytes
