@@ -43,7 +43,7 @@ def main() -> None:
                     help="total parameter GiB across all ranks")
     ap.add_argument("--mode", choices=["arena", "pack"], default="arena")
     ap.add_argument("--chunk-mb", type=int, default=0)
-    ap.add_argument("--no-vmm", action="store_true")
+    ap.add_argument("--vmm", action="store_true", help="opt into VMM arena (unreliable on ROCm 7.2)")
     args = ap.parse_args()
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -70,7 +70,7 @@ def main() -> None:
     eng = ActuationEngine(
         cfg, device_index=local_rank if on_gpu else 0,
         tp_rank=rank, tp_size=world, tp_group=None,
-        use_vmm=not args.no_vmm, chunk_bytes=args.chunk_mb << 20, seed=1234)
+        use_vmm=args.vmm or None, chunk_bytes=args.chunk_mb << 20, seed=1234)
     log(f"[rank {rank}] engine up: {eng.total_bytes/2**30:.2f} GiB/rank, "
         f"{cfg.num_layers} layers, vmm={eng.stats()['uses_vmm']}, "
         f"create {time.perf_counter()-t0:.1f}s")

@@ -95,6 +95,11 @@ void join_torch_stream(DeviceCtx& ctx) {
   for (auto& s : ctx.copy_streams) {
     FMA_HIP_CHECK(hipStreamWaitEvent(s, ctx.sync_event, 0));
   }
+  // Belt-and-braces: tensors may have been written on a stream the c10 TLS
+  // does not report (per-thread default streams, other libraries). A device
+  // sync here costs microseconds against a multi-second transfer and makes
+  // the actuation independent of the caller's stream discipline.
+  FMA_HIP_CHECK(hipDeviceSynchronize());
 }
 
 void sync_pipeline(DeviceCtx& ctx) {
@@ -485,9 +490,7 @@ class DeviceArena {
                 granularity_ * granularity_;
       FMA_HIP_CHECK(
           hipMemAddressReserve(&base_, padded_, granularity_, nullptr, 0));
-      for (size_t off = 0; off < padded_; off += phys_chunk_) {
-        map_slice(off, std::min(phys_chunk_, padded_ - off));
-      }
+      map_all();
     } else {
       padded_ = static_cast<size_t>(nbytes);
       FMA_HIP_CHECK(hipMalloc(&base_, padded_));
@@ -504,6 +507,7 @@ class DeviceArena {
   bool uses_vmm() const { return vmm_; }
   int64_t size_bytes() const { return size_; }
   int device() const { return device_; }
+  double last_map_seconds() const { return map_seconds_; }
 
   at::Tensor view(int64_t offset, std::vector<int64_t> sizes,
                   at::ScalarType dtype) {
@@ -553,20 +557,19 @@ class DeviceArena {
     auto* host_ptr = static_cast<unsigned char*>(host.data_ptr());
     const auto t0 = Clock::now();
     if (vmm_) {
-      int64_t copied = 0, c = 0;
-      for (size_t off = 0; off < padded_; off += phys_chunk_) {
-        const size_t sz = std::min(phys_chunk_, padded_ - off);
-        map_slice(off, sz);
-        const int64_t avail =
-            std::min<int64_t>(static_cast<int64_t>(off + sz), size_);
-        auto* dev_ptr = static_cast<unsigned char*>(base_);
-        while (copied < avail) {
-          const int64_t take = std::min<int64_t>(chunk, avail - copied);
-          FMA_HIP_CHECK(hipMemcpyAsync(dev_ptr + copied, host_ptr + copied,
-                                       take, hipMemcpyHostToDevice,
-                                       ctx.copy_streams[c++ % kNumCopyStreams]));
-          copied += take;
-        }
+      // hipMemSetAccess on ROCm must span the whole reservation, so the
+      // mapping phase completes before copies start. hipMemCreate/hipMemMap
+      // are fast relative to the PCIe transfer; map_seconds_ records the
+      // split for profiling.
+      const auto tm0 = Clock::now();
+      map_all();
+      map_seconds_ = seconds_since(tm0);
+      auto* dev_ptr = static_cast<unsigned char*>(base_);
+      for (int64_t off = 0, c = 0; off < size_; off += chunk, ++c) {
+        const int64_t sz = std::min<int64_t>(chunk, size_ - off);
+        FMA_HIP_CHECK(hipMemcpyAsync(dev_ptr + off, host_ptr + off, sz,
+                                     hipMemcpyHostToDevice,
+                                     ctx.copy_streams[c % kNumCopyStreams]));
       }
     } else {
       FMA_HIP_CHECK(hipMalloc(&base_, padded_));  // NOTE: base may change
@@ -598,13 +601,29 @@ class DeviceArena {
     FMA_HIP_CHECK(hipMemCreate(&h, sz, &prop, 0));
     FMA_HIP_CHECK(
         hipMemMap(static_cast<unsigned char*>(base_) + off, sz, 0, h, 0));
+    handles_.emplace_back(off, sz, h);
+  }
+
+  // Map every physical slice, then grant access over the WHOLE reservation:
+  // ROCm's hipMemSetAccess rejects sub-ranges of a reservation (observed
+  // "invalid argument" for per-slice calls on multi-slice arenas, gfx950 /
+  // ROCm 7.2), so access is set once here. The mapping re-uses a VA that
+  // may have been cached (L1/TLB) before an unmap, so finish by quiescing
+  // the device and invalidating every CU's L1 with an agent-scope acquire
+  // (`buffer_inv sc1`) — without this, post-wake reads are intermittently
+  // stale (observed on gfx950 / ROCm 7.2).
+  void map_all() {
+    for (size_t off = 0; off < padded_; off += phys_chunk_) {
+      map_slice(off, std::min(phys_chunk_, padded_ - off));
+    }
     hipMemAccessDesc acc{};
     acc.location.type = hipMemLocationTypeDevice;
     acc.location.id = device_;
     acc.flags = hipMemAccessFlagsProtReadWrite;
-    FMA_HIP_CHECK(hipMemSetAccess(static_cast<unsigned char*>(base_) + off, sz,
-                                  &acc, 1));
-    handles_.emplace_back(off, sz, h);
+    FMA_HIP_CHECK(hipMemSetAccess(base_, padded_, &acc, 1));
+    FMA_HIP_CHECK(hipDeviceSynchronize());
+    FMA_HIP_CHECK(fma_launch_cache_invalidate(nullptr));
+    FMA_HIP_CHECK(hipDeviceSynchronize());
   }
 
   void unmap_physical() {
@@ -636,6 +655,7 @@ class DeviceArena {
   int device_ = 0;
   bool vmm_ = false;
   bool mapped_ = false;
+  double map_seconds_ = 0.0;
   void* base_ = nullptr;
   std::vector<std::tuple<size_t, size_t, hipMemGenericAllocationHandle_t>>
       handles_;
@@ -676,5 +696,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def_property_readonly("is_mapped", &DeviceArena::is_mapped)
       .def_property_readonly("uses_vmm", &DeviceArena::uses_vmm)
       .def_property_readonly("size_bytes", &DeviceArena::size_bytes)
-      .def_property_readonly("device", &DeviceArena::device);
+      .def_property_readonly("device", &DeviceArena::device)
+      .def_property_readonly("last_map_seconds", &DeviceArena::last_map_seconds);
 }

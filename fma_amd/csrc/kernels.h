@@ -27,3 +27,5 @@ extern "C" hipError_t fma_launch_batched_copy(const FmaCopyDesc* descs_dev,
 extern "C" hipError_t fma_launch_contiguous_copy(const void* src, void* dst,
                                                  unsigned long long bytes,
                                                  hipStream_t stream);
+
+extern "C" hipError_t fma_launch_cache_invalidate(hipStream_t stream);

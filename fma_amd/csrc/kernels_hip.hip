@@ -131,3 +131,21 @@ extern "C" hipError_t fma_launch_contiguous_copy(const void* src, void* dst,
       bytes);
   return hipGetLastError();
 }
+
+namespace {
+
+// One workgroup per CU issuing an agent-scope acquire: lowers to
+// `buffer_inv sc1`, invalidating that CU's vector L1 (MI355X_MICROARCH.md
+// §Workgroup dispatch table). Used after a VMM unmap/remap cycle so no CU
+// can serve reads at the re-mapped VA from lines cached before the remap.
+__global__ void cache_invalidate_kernel() {
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+}
+
+}  // namespace
+
+extern "C" hipError_t fma_launch_cache_invalidate(hipStream_t stream) {
+  // 2048 workgroups of one wave: >=1 lands on every one of the 256 CUs.
+ hipLaunchKernelGGL(( cache_invalidate_kernel), dim3(2048), dim3(64), 0, stream, );
+  return hipGetLastError();
+}

@@ -93,10 +93,26 @@ def alloc_pinned(nbytes: int) -> torch.Tensor:
 class ArenaActuator:
     """Contiguous device arena with sleep/wake via one pinned host buffer."""
 
-    def __init__(self, nbytes: int, device: int = 0, try_vmm: bool = True,
-                 chunk_bytes: int = 0):
+    def __init__(self, nbytes: int, device: int = 0,
+                 try_vmm: Optional[bool] = None, chunk_bytes: int = 0):
         C = require_native()
-        self._arena = C.DeviceArena(nbytes, device, try_vmm)
+        if try_vmm is None:
+            # VMM (constant-VA remap) measured UNRELIABLE on ROCm 7.2 /
+            # gfx950: after unmap→remap or VA reuse, SDMA reads are
+            # intermittently stale (tools/debug_arena.py: 86-100% corrupt
+            # cycles under swap stress). Default is the plain hipMalloc
+            # arena — views are re-bound on wake, which costs milliseconds.
+            try_vmm = os.environ.get("FMA_TRY_VMM") == "1"
+        if try_vmm:
+            try:
+                self._arena = C.DeviceArena(nbytes, device, True)
+            except RuntimeError as e:
+                import warnings
+                warnings.warn(f"VMM arena failed ({e}); falling back to "
+                              "hipMalloc arena (views re-bound on wake)")
+                self._arena = C.DeviceArena(nbytes, device, False)
+        else:
+            self._arena = C.DeviceArena(nbytes, device, False)
         self.device = device
         self.nbytes = nbytes
         self.chunk_bytes = chunk_bytes
@@ -133,7 +149,7 @@ class FakeArenaActuator:
     extension on a real GPU box can never silently fall through to this.
     """
 
-    def __init__(self, nbytes: int, device: int = 0, try_vmm: bool = True,
+    def __init__(self, nbytes: int, device: int = 0, try_vmm=None,
                  chunk_bytes: int = 0):
         if not (os.environ.get("FMA_FAKE_GPU") == "1"
                 or not torch.cuda.is_available()):
@@ -220,7 +236,7 @@ class PackActuator:
         return t
 
 
-def make_arena(nbytes: int, device: int = 0, try_vmm: bool = True,
+def make_arena(nbytes: int, device: int = 0, try_vmm=None,
                chunk_bytes: int = 0):
     """Arena factory: native on a GPU machine, fake on CPU-only machines."""
     if torch.cuda.is_available():
