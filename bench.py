@@ -43,6 +43,8 @@ def main() -> None:
                     help="total parameter GiB across all ranks")
     ap.add_argument("--mode", choices=["arena", "pack"], default="arena")
     ap.add_argument("--chunk-mb", type=int, default=0)
+    ap.add_argument("--nstreams", type=int, default=2)
+    ap.add_argument("--slab-mb", type=int, default=0, help="slab size MiB (0=default 1 GiB)")
     ap.add_argument("--vmm", action="store_true", help="opt into VMM arena (unreliable on ROCm 7.2)")
     args = ap.parse_args()
 
@@ -70,7 +72,9 @@ def main() -> None:
     eng = ActuationEngine(
         cfg, device_index=local_rank if on_gpu else 0,
         tp_rank=rank, tp_size=world, tp_group=None,
-        use_vmm=args.vmm or None, chunk_bytes=args.chunk_mb << 20, seed=1234)
+        use_vmm=args.vmm or None, chunk_bytes=args.chunk_mb << 20, seed=1234,
+        nstreams=args.nstreams,
+        slab_bytes=(args.slab_mb << 20) if args.slab_mb else None)
     log(f"[rank {rank}] engine up: {eng.total_bytes/2**30:.2f} GiB/rank, "
         f"{cfg.num_layers} layers, vmm={eng.stats()['uses_vmm']}, "
         f"create {time.perf_counter()-t0:.1f}s")

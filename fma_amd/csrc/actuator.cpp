@@ -54,7 +54,7 @@ double seconds_since(Clock::time_point t0) {
   return std::chrono::duration<double>(Clock::now() - t0).count();
 }
 
-constexpr int kNumCopyStreams = 2;
+constexpr int kNumCopyStreams = 4;
 constexpr int64_t kDefaultChunk = 256ll << 20;  // 256 MiB
 
 struct DeviceCtx {
@@ -471,13 +471,37 @@ bool device_supports_vmm(int device) {
   return e == hipSuccess && v != 0;
 }
 
+// One model instance's parameter storage with sleep/wake. Three backings:
+//
+// - slabbed hipMalloc (DEFAULT): the flat space is split into slabs
+//   (boundaries chosen by the Python layout planner so no tensor straddles
+//   one). wake_up overlaps slab k+1's hipMalloc with slab k's in-flight
+//   H2D copies, hiding allocator latency behind the PCIe stream. Slab
+//   pointers change across sleep/wake, so views are re-bound by the caller
+//   (generation counter signals it).
+// - single hipMalloc: slab plan of one entry.
+// - VMM (opt-in, FMA_TRY_VMM=1): constant-VA remap via hipMemAddressReserve
+//   + hipMemCreate/hipMemMap. Measured UNRELIABLE on ROCm 7.2/gfx950
+//   (stale SDMA reads after unmap→remap; tools/debug_arena.py), kept for
+//   future driver versions.
 class DeviceArena {
  public:
-  DeviceArena(int64_t nbytes, int device, bool try_vmm)
+  DeviceArena(int64_t nbytes, int device, bool try_vmm,
+              std::vector<int64_t> slab_sizes = {})
       : size_(nbytes), device_(device) {
     TORCH_CHECK(nbytes > 0, "arena size must be positive");
     FMA_HIP_CHECK(hipSetDevice(device_));
     vmm_ = try_vmm && device_supports_vmm(device_);
+    if (slab_sizes.empty()) slab_sizes.push_back(nbytes);
+    int64_t sum = 0;
+    slab_prefix_.push_back(0);
+    for (auto s : slab_sizes) {
+      TORCH_CHECK(s > 0, "slab sizes must be positive");
+      sum += s;
+      slab_prefix_.push_back(sum);
+    }
+    TORCH_CHECK(sum >= nbytes, "slab sizes cover less than the arena");
+    slab_sizes_ = std::move(slab_sizes);
     if (vmm_) {
       hipMemAllocationProp prop = alloc_prop();
       size_t gran = 0;
@@ -489,11 +513,10 @@ class DeviceArena {
       padded_ = (static_cast<size_t>(nbytes) + granularity_ - 1) /
                 granularity_ * granularity_;
       FMA_HIP_CHECK(
-          hipMemAddressReserve(&base_, padded_, granularity_, nullptr, 0));
+          hipMemAddressReserve(&vmm_base_, padded_, granularity_, nullptr, 0));
       map_all();
     } else {
-      padded_ = static_cast<size_t>(nbytes);
-      FMA_HIP_CHECK(hipMalloc(&base_, padded_));
+      alloc_slabs();
     }
     mapped_ = true;
   }
@@ -502,12 +525,19 @@ class DeviceArena {
   DeviceArena(const DeviceArena&) = delete;
   DeviceArena& operator=(const DeviceArena&) = delete;
 
-  int64_t data_ptr() const { return reinterpret_cast<int64_t>(base_); }
+  int64_t data_ptr() const {
+    return reinterpret_cast<int64_t>(vmm_ ? vmm_base_ : slabs_.empty()
+                                     ? nullptr : slabs_[0]);
+  }
   bool is_mapped() const { return mapped_; }
   bool uses_vmm() const { return vmm_; }
   int64_t size_bytes() const { return size_; }
   int device() const { return device_; }
+  int64_t generation() const { return generation_; }
   double last_map_seconds() const { return map_seconds_; }
+  int64_t num_slabs() const {
+    return static_cast<int64_t>(vmm_ ? 1 : slab_sizes_.size());
+  }
 
   at::Tensor view(int64_t offset, std::vector<int64_t> sizes,
                   at::ScalarType dtype) {
@@ -518,75 +548,127 @@ class DeviceArena {
     TORCH_CHECK(offset >= 0 && offset + bytes <= size_, "view out of bounds");
     auto options =
         at::TensorOptions().dtype(dtype).device(at::Device(at::kCUDA, device_));
-    return at::from_blob(static_cast<unsigned char*>(base_) + offset, sizes,
-                         options);
+    return at::from_blob(resolve(offset, bytes), sizes, options);
   }
 
   // sleep(level=1): D2H into pinned host DRAM, then release physical HBM.
-  double sleep_to(at::Tensor host, int64_t chunk_bytes) {
+  double sleep_to(at::Tensor host, int64_t chunk_bytes, int64_t nstreams) {
     TORCH_CHECK(mapped_, "arena already asleep");
     check_host_buffer(host, size_);
     auto& ctx = ctx_for(device_);
     FMA_HIP_CHECK(hipSetDevice(device_));
     const int64_t chunk = chunk_or_default(chunk_bytes);
+    const int ns = clamp_streams(nstreams);
     auto* host_ptr = static_cast<unsigned char*>(host.data_ptr());
-    auto* dev_ptr = static_cast<unsigned char*>(base_);
     const auto t0 = Clock::now();
     join_torch_stream(ctx);
-    for (int64_t off = 0, c = 0; off < size_; off += chunk, ++c) {
-      const int64_t sz = std::min<int64_t>(chunk, size_ - off);
-      FMA_HIP_CHECK(hipMemcpyAsync(host_ptr + off, dev_ptr + off, sz,
-                                   hipMemcpyDeviceToHost,
-                                   ctx.copy_streams[c % kNumCopyStreams]));
-    }
+    int c = 0;
+    for_each_span([&](unsigned char* dev, int64_t flat, int64_t len) {
+      for (int64_t off = 0; off < len; off += chunk, ++c) {
+        const int64_t sz = std::min<int64_t>(chunk, len - off);
+        FMA_HIP_CHECK(hipMemcpyAsync(host_ptr + flat + off, dev + off, sz,
+                                     hipMemcpyDeviceToHost,
+                                     ctx.copy_streams[c % ns]));
+      }
+    });
     sync_pipeline(ctx);
     unmap_physical();
     mapped_ = false;
     return seconds_since(t0);
   }
 
-  // wake_up: re-acquire physical HBM and copy the arena back; in VMM mode
-  // mapping of slice k+1 overlaps the H2D of slice k and all tensor views
-  // remain valid (same VA).
-  double wake_from(at::Tensor host, int64_t chunk_bytes) {
+  // wake_up: re-acquire physical HBM and copy back. Slabbed path: slab
+  // k+1's hipMalloc overlaps slab k's H2D copies.
+  double wake_from(at::Tensor host, int64_t chunk_bytes, int64_t nstreams) {
     TORCH_CHECK(!mapped_, "arena already awake");
     check_host_buffer(host, size_);
     auto& ctx = ctx_for(device_);
     FMA_HIP_CHECK(hipSetDevice(device_));
     const int64_t chunk = chunk_or_default(chunk_bytes);
+    const int ns = clamp_streams(nstreams);
     auto* host_ptr = static_cast<unsigned char*>(host.data_ptr());
     const auto t0 = Clock::now();
     if (vmm_) {
-      // hipMemSetAccess on ROCm must span the whole reservation, so the
-      // mapping phase completes before copies start. hipMemCreate/hipMemMap
-      // are fast relative to the PCIe transfer; map_seconds_ records the
-      // split for profiling.
       const auto tm0 = Clock::now();
       map_all();
       map_seconds_ = seconds_since(tm0);
-      auto* dev_ptr = static_cast<unsigned char*>(base_);
+      auto* dev_ptr = static_cast<unsigned char*>(vmm_base_);
       for (int64_t off = 0, c = 0; off < size_; off += chunk, ++c) {
         const int64_t sz = std::min<int64_t>(chunk, size_ - off);
         FMA_HIP_CHECK(hipMemcpyAsync(dev_ptr + off, host_ptr + off, sz,
                                      hipMemcpyHostToDevice,
-                                     ctx.copy_streams[c % kNumCopyStreams]));
+                                     ctx.copy_streams[c % ns]));
       }
     } else {
-      FMA_HIP_CHECK(hipMalloc(&base_, padded_));  // NOTE: base may change
-      auto* dev_ptr = static_cast<unsigned char*>(base_);
-      for (int64_t off = 0, c = 0; off < size_; off += chunk, ++c) {
-        const int64_t sz = std::min<int64_t>(chunk, size_ - off);
-        FMA_HIP_CHECK(hipMemcpyAsync(dev_ptr + off, host_ptr + off, sz,
-                                     hipMemcpyHostToDevice,
-                                     ctx.copy_streams[c % kNumCopyStreams]));
+      double alloc_acc = 0.0;
+      int c = 0;
+      for (size_t i = 0; i < slab_sizes_.size(); ++i) {
+        const auto ta = Clock::now();
+        void* p = nullptr;
+        FMA_HIP_CHECK(hipMalloc(&p, slab_sizes_[i]));  // overlaps prior copies
+        alloc_acc += seconds_since(ta);
+        slabs_.push_back(p);
+        const int64_t flat0 = slab_prefix_[i];
+        const int64_t len =
+            std::min<int64_t>(slab_sizes_[i], size_ - flat0);
+        for (int64_t off = 0; off < len; off += chunk, ++c) {
+          const int64_t sz = std::min<int64_t>(chunk, len - off);
+          FMA_HIP_CHECK(hipMemcpyAsync(static_cast<unsigned char*>(p) + off,
+                                       host_ptr + flat0 + off, sz,
+                                       hipMemcpyHostToDevice,
+                                       ctx.copy_streams[c % ns]));
+        }
       }
+      map_seconds_ = alloc_acc;
     }
     sync_pipeline(ctx);
     mapped_ = true;
+    ++generation_;
     return seconds_since(t0);
   }
 
  private:
+  static int clamp_streams(int64_t n) {
+    if (n <= 0) return 2;
+    return static_cast<int>(std::min<int64_t>(n, kNumCopyStreams));
+  }
+
+  unsigned char* resolve(int64_t offset, int64_t bytes) {
+    if (vmm_) return static_cast<unsigned char*>(vmm_base_) + offset;
+    // slab lookup: last prefix <= offset
+    auto it = std::upper_bound(slab_prefix_.begin(), slab_prefix_.end(),
+                               offset);
+    const size_t idx = static_cast<size_t>(it - slab_prefix_.begin()) - 1;
+    TORCH_CHECK(idx < slabs_.size(), "offset beyond mapped slabs");
+    const int64_t in_slab = offset - slab_prefix_[idx];
+    TORCH_CHECK(in_slab + bytes <= slab_sizes_[idx],
+                "tensor straddles a slab boundary (bad layout plan): offset ",
+                offset, " bytes ", bytes, " slab ", idx);
+    return static_cast<unsigned char*>(slabs_[idx]) + in_slab;
+  }
+
+  // Iterate contiguous device spans with their flat offsets.
+  template <typename F>
+  void for_each_span(F&& f) {
+    if (vmm_) {
+      f(static_cast<unsigned char*>(vmm_base_), 0, size_);
+      return;
+    }
+    for (size_t i = 0; i < slabs_.size(); ++i) {
+      const int64_t flat0 = slab_prefix_[i];
+      const int64_t len = std::min<int64_t>(slab_sizes_[i], size_ - flat0);
+      if (len > 0) f(static_cast<unsigned char*>(slabs_[i]), flat0, len);
+    }
+  }
+
+  void alloc_slabs() {
+    for (auto sz : slab_sizes_) {
+      void* p = nullptr;
+      FMA_HIP_CHECK(hipMalloc(&p, sz));
+      slabs_.push_back(p);
+    }
+  }
+
   hipMemAllocationProp alloc_prop() const {
     hipMemAllocationProp prop{};
     prop.type = hipMemAllocationTypePinned;
@@ -600,18 +682,13 @@ class DeviceArena {
     hipMemGenericAllocationHandle_t h{};
     FMA_HIP_CHECK(hipMemCreate(&h, sz, &prop, 0));
     FMA_HIP_CHECK(
-        hipMemMap(static_cast<unsigned char*>(base_) + off, sz, 0, h, 0));
+        hipMemMap(static_cast<unsigned char*>(vmm_base_) + off, sz, 0, h, 0));
     handles_.emplace_back(off, sz, h);
   }
 
-  // Map every physical slice, then grant access over the WHOLE reservation:
-  // ROCm's hipMemSetAccess rejects sub-ranges of a reservation (observed
-  // "invalid argument" for per-slice calls on multi-slice arenas, gfx950 /
-  // ROCm 7.2), so access is set once here. The mapping re-uses a VA that
-  // may have been cached (L1/TLB) before an unmap, so finish by quiescing
-  // the device and invalidating every CU's L1 with an agent-scope acquire
-  // (`buffer_inv sc1`) — without this, post-wake reads are intermittently
-  // stale (observed on gfx950 / ROCm 7.2).
+  // Map every physical slice, then grant access over the WHOLE reservation
+  // (ROCm rejects per-slice hipMemSetAccess), then quiesce + invalidate
+  // every CU's L1 (the remap re-uses a VA that may be cached).
   void map_all() {
     for (size_t off = 0; off < padded_; off += phys_chunk_) {
       map_slice(off, std::min(phys_chunk_, padded_ - off));
@@ -620,7 +697,7 @@ class DeviceArena {
     acc.location.type = hipMemLocationTypeDevice;
     acc.location.id = device_;
     acc.flags = hipMemAccessFlagsProtReadWrite;
-    FMA_HIP_CHECK(hipMemSetAccess(base_, padded_, &acc, 1));
+    FMA_HIP_CHECK(hipMemSetAccess(vmm_base_, padded_, &acc, 1));
     FMA_HIP_CHECK(hipDeviceSynchronize());
     FMA_HIP_CHECK(fma_launch_cache_invalidate(nullptr));
     FMA_HIP_CHECK(hipDeviceSynchronize());
@@ -629,21 +706,21 @@ class DeviceArena {
   void unmap_physical() {
     if (vmm_) {
       for (auto& [off, sz, h] : handles_) {
-        (void)hipMemUnmap(static_cast<unsigned char*>(base_) + off, sz);
+        (void)hipMemUnmap(static_cast<unsigned char*>(vmm_base_) + off, sz);
         (void)hipMemRelease(h);
       }
       handles_.clear();
-    } else if (base_) {
-      (void)hipFree(base_);
-      base_ = nullptr;
+    } else {
+      for (auto* p : slabs_) (void)hipFree(p);
+      slabs_.clear();
     }
   }
 
   void release_all() {
     if (mapped_) unmap_physical();
-    if (vmm_ && base_) {
-      (void)hipMemAddressFree(base_, padded_);
-      base_ = nullptr;
+    if (vmm_ && vmm_base_) {
+      (void)hipMemAddressFree(vmm_base_, padded_);
+      vmm_base_ = nullptr;
     }
     mapped_ = false;
   }
@@ -655,8 +732,12 @@ class DeviceArena {
   int device_ = 0;
   bool vmm_ = false;
   bool mapped_ = false;
+  int64_t generation_ = 0;
   double map_seconds_ = 0.0;
-  void* base_ = nullptr;
+  void* vmm_base_ = nullptr;
+  std::vector<int64_t> slab_sizes_;
+  std::vector<int64_t> slab_prefix_;
+  std::vector<void*> slabs_;
   std::vector<std::tuple<size_t, size_t, hipMemGenericAllocationHandle_t>>
       handles_;
 };
@@ -684,18 +765,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("device_supports_vmm", &device_supports_vmm, py::arg("device"));
   m.def("device_mem_info", &device_mem_info, py::arg("device"));
   py::class_<DeviceArena>(m, "DeviceArena")
-      .def(py::init<int64_t, int, bool>(), py::arg("nbytes"), py::arg("device"),
-           py::arg("try_vmm") = true)
+      .def(py::init<int64_t, int, bool, std::vector<int64_t>>(),
+           py::arg("nbytes"), py::arg("device"), py::arg("try_vmm") = false,
+           py::arg("slab_sizes") = std::vector<int64_t>{})
       .def("view", &DeviceArena::view, py::arg("offset"), py::arg("sizes"),
            py::arg("dtype"))
       .def("sleep_to", &DeviceArena::sleep_to, py::arg("host"),
-           py::arg("chunk_bytes") = 0)
+           py::arg("chunk_bytes") = 0, py::arg("nstreams") = 2)
       .def("wake_from", &DeviceArena::wake_from, py::arg("host"),
-           py::arg("chunk_bytes") = 0)
+           py::arg("chunk_bytes") = 0, py::arg("nstreams") = 2)
       .def_property_readonly("data_ptr", &DeviceArena::data_ptr)
       .def_property_readonly("is_mapped", &DeviceArena::is_mapped)
       .def_property_readonly("uses_vmm", &DeviceArena::uses_vmm)
       .def_property_readonly("size_bytes", &DeviceArena::size_bytes)
       .def_property_readonly("device", &DeviceArena::device)
+      .def_property_readonly("generation", &DeviceArena::generation)
+      .def_property_readonly("num_slabs", &DeviceArena::num_slabs)
       .def_property_readonly("last_map_seconds", &DeviceArena::last_map_seconds);
 }

@@ -63,22 +63,41 @@ def align_up(n: int, a: int = ARENA_ALIGN) -> int:
     return (n + a - 1) // a * a
 
 
-def plan_layout(specs: Sequence[Tuple[str, Tuple[int, ...], torch.dtype]]
-                ) -> Tuple[Dict[str, Tuple[int, Tuple[int, ...], torch.dtype]], int]:
-    """Assign 256-B-aligned flat offsets to named tensor specs.
+#: default slab size for the segmented arena: big enough that per-slab
+#: hipMalloc overhead amortizes, small enough that wake_up overlaps most
+#: allocation time behind in-flight PCIe copies.
+DEFAULT_SLAB_BYTES = 1 << 30
 
-    Returns ({name: (offset, shape, dtype)}, total_bytes).
+
+def plan_layout(specs: Sequence[Tuple[str, Tuple[int, ...], torch.dtype]],
+                slab_bytes: int = 0,
+                ) -> Tuple[Dict[str, Tuple[int, Tuple[int, ...], torch.dtype]],
+                           int, List[int]]:
+    """Assign 256-B-aligned flat offsets to named tensor specs, packing them
+    into slabs of ~slab_bytes so no tensor straddles a slab boundary
+    (slab_bytes=0: one slab).
+
+    Returns ({name: (offset, shape, dtype)}, total_bytes, slab_sizes).
     """
     layout: Dict[str, Tuple[int, Tuple[int, ...], torch.dtype]] = {}
-    off = 0
+    slab_sizes: List[int] = []
+    closed = 0   # total bytes in closed slabs
+    used = 0     # bytes used in the open slab
     for name, shape, dtype in specs:
         numel = 1
         for s in shape:
             numel *= s
-        nbytes = numel * torch.empty(0, dtype=dtype).element_size()
-        layout[name] = (off, tuple(shape), dtype)
-        off += align_up(max(nbytes, 1))
-    return layout, off
+        nbytes = align_up(max(numel * torch.empty(0, dtype=dtype).element_size(), 1))
+        if slab_bytes and used and used + nbytes > slab_bytes:
+            slab_sizes.append(used)
+            closed += used
+            used = 0
+        layout[name] = (closed + used, tuple(shape), dtype)
+        used += nbytes
+    if used:
+        slab_sizes.append(used)
+        closed += used
+    return layout, closed, slab_sizes
 
 
 def alloc_pinned(nbytes: int) -> torch.Tensor:
@@ -94,29 +113,32 @@ class ArenaActuator:
     """Contiguous device arena with sleep/wake via one pinned host buffer."""
 
     def __init__(self, nbytes: int, device: int = 0,
-                 try_vmm: Optional[bool] = None, chunk_bytes: int = 0):
+                 try_vmm: Optional[bool] = None, chunk_bytes: int = 0,
+                 slab_sizes: Optional[List[int]] = None, nstreams: int = 2):
         C = require_native()
         if try_vmm is None:
             # VMM (constant-VA remap) measured UNRELIABLE on ROCm 7.2 /
             # gfx950: after unmap→remap or VA reuse, SDMA reads are
             # intermittently stale (tools/debug_arena.py: 86-100% corrupt
-            # cycles under swap stress). Default is the plain hipMalloc
+            # cycles under swap stress). Default is the slabbed hipMalloc
             # arena — views are re-bound on wake, which costs milliseconds.
             try_vmm = os.environ.get("FMA_TRY_VMM") == "1"
+        slabs = list(slab_sizes or [])
         if try_vmm:
             try:
-                self._arena = C.DeviceArena(nbytes, device, True)
+                self._arena = C.DeviceArena(nbytes, device, True, [])
             except RuntimeError as e:
                 import warnings
                 warnings.warn(f"VMM arena failed ({e}); falling back to "
-                              "hipMalloc arena (views re-bound on wake)")
-                self._arena = C.DeviceArena(nbytes, device, False)
+                              "slabbed hipMalloc arena (views re-bound on wake)")
+                self._arena = C.DeviceArena(nbytes, device, False, slabs)
         else:
-            self._arena = C.DeviceArena(nbytes, device, False)
+            self._arena = C.DeviceArena(nbytes, device, False, slabs)
         self.device = device
         self.nbytes = nbytes
         self.chunk_bytes = chunk_bytes
-        self._base_at_view: int = self._arena.data_ptr
+        self.nstreams = nstreams
+        self._gen_at_view: int = self._arena.generation
 
     @property
     def uses_vmm(self) -> bool:
@@ -126,20 +148,25 @@ class ArenaActuator:
     def is_mapped(self) -> bool:
         return self._arena.is_mapped
 
+    @property
+    def last_map_seconds(self) -> float:
+        return self._arena.last_map_seconds
+
     def view(self, offset: int, shape: Tuple[int, ...], dtype: torch.dtype
              ) -> torch.Tensor:
         return self._arena.view(offset, list(shape), dtype)
 
     def sleep(self, host: torch.Tensor) -> float:
-        return self._arena.sleep_to(host, self.chunk_bytes)
+        return self._arena.sleep_to(host, self.chunk_bytes, self.nstreams)
 
     def wake(self, host: torch.Tensor) -> Tuple[float, bool]:
         """Returns (seconds, views_invalidated). With VMM backing the VA is
         constant and views survive; otherwise the caller must re-bind."""
-        t = self._arena.wake_from(host, self.chunk_bytes)
-        invalidated = self._arena.data_ptr != self._base_at_view
-        self._base_at_view = self._arena.data_ptr
-        return t, invalidated
+        t = self._arena.wake_from(host, self.chunk_bytes, self.nstreams)
+        self._gen_at_view = self._arena.generation
+        # VMM keeps the VA constant; slabbed/plain arenas re-allocate, so
+        # every wake invalidates existing views.
+        return t, not self._arena.uses_vmm
 
 
 class FakeArenaActuator:
@@ -150,7 +177,7 @@ class FakeArenaActuator:
     """
 
     def __init__(self, nbytes: int, device: int = 0, try_vmm=None,
-                 chunk_bytes: int = 0):
+                 chunk_bytes: int = 0, slab_sizes=None, nstreams: int = 2):
         if not (os.environ.get("FMA_FAKE_GPU") == "1"
                 or not torch.cuda.is_available()):
             raise RuntimeError("FakeArenaActuator is only for GPU-less machines")
@@ -237,8 +264,10 @@ class PackActuator:
 
 
 def make_arena(nbytes: int, device: int = 0, try_vmm=None,
-               chunk_bytes: int = 0):
+               chunk_bytes: int = 0, slab_sizes=None, nstreams: int = 2):
     """Arena factory: native on a GPU machine, fake on CPU-only machines."""
     if torch.cuda.is_available():
-        return ArenaActuator(nbytes, device, try_vmm, chunk_bytes)
-    return FakeArenaActuator(nbytes, device, try_vmm, chunk_bytes)
+        return ArenaActuator(nbytes, device, try_vmm, chunk_bytes,
+                             slab_sizes, nstreams)
+    return FakeArenaActuator(nbytes, device, try_vmm, chunk_bytes,
+                             slab_sizes, nstreams)

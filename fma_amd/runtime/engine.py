@@ -35,7 +35,8 @@ class ActuationEngine:
     def __init__(self, cfg: LlamaConfig, device_index: int = 0,
                  tp_rank: int = 0, tp_size: int = 1, tp_group=None,
                  use_vmm=None, chunk_bytes: int = 0, seed: int = 0,
-                 init_weights: bool = True):
+                 init_weights: bool = True, nstreams: int = 2,
+                 slab_bytes: int = None):
         self.cfg = cfg
         self.device_index = device_index
         self.tp_rank = tp_rank
@@ -48,11 +49,16 @@ class ActuationEngine:
             torch.cuda.set_device(device_index)
 
         specs = cfg.param_specs(tp_rank, tp_size)
-        self.layout, self.total_bytes = actuation.plan_layout(specs)
+        if slab_bytes is None:
+            slab_bytes = actuation.DEFAULT_SLAB_BYTES
+        self.layout, self.total_bytes, slab_sizes = actuation.plan_layout(
+            specs, slab_bytes=slab_bytes)
         t0 = time.perf_counter()
         self.arena = actuation.make_arena(self.total_bytes, device_index,
                                           try_vmm=use_vmm,
-                                          chunk_bytes=chunk_bytes)
+                                          chunk_bytes=chunk_bytes,
+                                          slab_sizes=slab_sizes,
+                                          nstreams=nstreams)
         self.params = self._make_views()
         self.model = LlamaModel(cfg, self.params, self.device,
                                 tp_rank, tp_size, tp_group)

@@ -24,7 +24,7 @@ def engine():
 
 def test_layout_alignment():
     cfg = LlamaConfig.tiny()
-    layout, total = actuation.plan_layout(cfg.param_specs())
+    layout, total, slabs = actuation.plan_layout(cfg.param_specs())
     assert total > 0
     for name, (off, shape, dtype) in layout.items():
         assert off % actuation.ARENA_ALIGN == 0
@@ -123,3 +123,28 @@ def test_stats_shape(engine):
     s = engine.stats()
     assert s["state"] == "awake"
     assert s["param_bytes"] == engine.total_bytes
+
+
+def test_plan_layout_slabs_no_straddle():
+    import itertools
+    import math
+
+    cfg = LlamaConfig.llama3_8b()
+    slab_bytes = 64 << 20
+    layout, total, slabs = actuation.plan_layout(cfg.param_specs(),
+                                                 slab_bytes=slab_bytes)
+    assert sum(slabs) == total
+    prefix = [0] + list(itertools.accumulate(slabs))
+    for name, (off, shape, dtype) in layout.items():
+        nbytes = math.prod(shape) * torch.empty(0, dtype=dtype).element_size()
+        si = max(i for i in range(len(prefix)) if prefix[i] <= off)
+        assert off + nbytes <= prefix[si + 1], f"{name} straddles slab {si}"
+
+
+def test_plan_layout_oversized_tensor_gets_own_slab():
+    specs = [("a", (100,), torch.uint8),
+             ("big", (3 << 20,), torch.uint8),  # larger than slab target
+             ("b", (100,), torch.uint8)]
+    layout, total, slabs = actuation.plan_layout(specs, slab_bytes=1 << 20)
+    assert sum(slabs) == total
+    assert len(slabs) == 3  # a | big | b
