@@ -43,7 +43,7 @@ def main() -> None:
                     help="total parameter GiB across all ranks")
     ap.add_argument("--mode", choices=["arena", "pack"], default="arena")
     ap.add_argument("--chunk-mb", type=int, default=0)
-    ap.add_argument("--nstreams", type=int, default=2)
+    ap.add_argument("--nstreams", type=int, default=1)
     ap.add_argument("--slab-mb", type=int, default=0, help="slab size MiB (0=default 1 GiB)")
     ap.add_argument("--vmm", action="store_true", help="opt into VMM arena (unreliable on ROCm 7.2)")
     args = ap.parse_args()

@@ -66,7 +66,7 @@ def align_up(n: int, a: int = ARENA_ALIGN) -> int:
 #: default slab size for the segmented arena: big enough that per-slab
 #: hipMalloc overhead amortizes, small enough that wake_up overlaps most
 #: allocation time behind in-flight PCIe copies.
-DEFAULT_SLAB_BYTES = 1 << 30
+DEFAULT_SLAB_BYTES = 512 << 20
 
 
 def plan_layout(specs: Sequence[Tuple[str, Tuple[int, ...], torch.dtype]],
@@ -114,7 +114,7 @@ class ArenaActuator:
 
     def __init__(self, nbytes: int, device: int = 0,
                  try_vmm: Optional[bool] = None, chunk_bytes: int = 0,
-                 slab_sizes: Optional[List[int]] = None, nstreams: int = 2):
+                 slab_sizes: Optional[List[int]] = None, nstreams: int = 1):
         C = require_native()
         if try_vmm is None:
             # VMM (constant-VA remap) measured UNRELIABLE on ROCm 7.2 /
@@ -177,7 +177,7 @@ class FakeArenaActuator:
     """
 
     def __init__(self, nbytes: int, device: int = 0, try_vmm=None,
-                 chunk_bytes: int = 0, slab_sizes=None, nstreams: int = 2):
+                 chunk_bytes: int = 0, slab_sizes=None, nstreams: int = 1):
         if not (os.environ.get("FMA_FAKE_GPU") == "1"
                 or not torch.cuda.is_available()):
             raise RuntimeError("FakeArenaActuator is only for GPU-less machines")
@@ -264,7 +264,7 @@ class PackActuator:
 
 
 def make_arena(nbytes: int, device: int = 0, try_vmm=None,
-               chunk_bytes: int = 0, slab_sizes=None, nstreams: int = 2):
+               chunk_bytes: int = 0, slab_sizes=None, nstreams: int = 1):
     """Arena factory: native on a GPU machine, fake on CPU-only machines."""
     if torch.cuda.is_available():
         return ArenaActuator(nbytes, device, try_vmm, chunk_bytes,

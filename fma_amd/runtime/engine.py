@@ -35,7 +35,7 @@ class ActuationEngine:
     def __init__(self, cfg: LlamaConfig, device_index: int = 0,
                  tp_rank: int = 0, tp_size: int = 1, tp_group=None,
                  use_vmm=None, chunk_bytes: int = 0, seed: int = 0,
-                 init_weights: bool = True, nstreams: int = 2,
+                 init_weights: bool = True, nstreams: int = 1,
                  slab_bytes: int = None):
         self.cfg = cfg
         self.device_index = device_index
