@@ -1,0 +1,199 @@
+"""Serving HTTP runtime: the process the launcher starts per instance.
+
+Speaks the exact wire surface the dual-pods controller drives on an
+inference server (reference pkg/controller/dual-pods/inference-server.go:
+1497 wake, 1712 sleep, 1985 is_sleeping; SleepState JSON shape
+pkg/api/interface.go:131-135), plus a minimal OpenAI-style completion
+endpoint for smoke traffic:
+
+  GET  /health           -> {"status": "OK"} once serving
+  GET  /is_sleeping      -> {"is_sleeping": bool}
+  POST /sleep?level=1    -> offload weights to pinned host DRAM
+  POST /wake_up          -> restore weights (RCCL barrier gates readiness)
+  GET  /v1/models        -> model card
+  POST /v1/completions   -> {"prompt", "max_tokens"}
+  GET  /stats            -> engine counters (sleep/wake seconds etc.)
+
+CLI options mirror the reference's ModelServerConfig.Options surface
+(reference api/fma/v1alpha1/inferenceserverconfig_types.go:35-62):
+``--model``, ``--port``, ``--tensor-parallel-size``, ``--enable-sleep-mode``
+(always on here), ``--max-model-len``, ``--seed``; unknown options are
+tolerated so reference ISC manifests keep working.
+
+TP instances: this process is rank 0; it forks ranks 1..N-1 before binding
+the port (see fma_amd/parallel/tp.py).
+"""
+
+from __future__ import annotations
+
+import argparse
+import multiprocessing
+import os
+import socket
+import threading
+import time
+from typing import List, Optional
+
+from fastapi import FastAPI, Query
+from fastapi.responses import JSONResponse
+
+from fma_amd.models.llama import LlamaConfig
+
+
+def parse_options(options: str) -> argparse.Namespace:
+    ap = argparse.ArgumentParser(prog="fma-serve", add_help=False)
+    ap.add_argument("--model", default="tiny")
+    ap.add_argument("--port", type=int, default=8000)
+    ap.add_argument("--host", default="0.0.0.0")
+    ap.add_argument("--tensor-parallel-size", type=int, default=1)
+    ap.add_argument("--enable-sleep-mode", action="store_true")
+    ap.add_argument("--max-model-len", type=int, default=None)
+    ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--served-model-name", default=None)
+    ap.add_argument("--start-asleep", action="store_true",
+                    help="immediately sleep after load (pre-warmed instance)")
+    args, unknown = ap.parse_known_args(options.split())
+    args.unknown = unknown
+    return args
+
+
+def _free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _worker_entry(rank: int, world: int, master_port: int,
+                  model_name: str, max_model_len: Optional[int],
+                  seed: int) -> None:
+    # ranks 1..N-1: build the engine shard and serve TP commands forever
+    # (workers stay in the instance's process group so the launcher's
+    # killpg on force-stop reaps them too)
+    from fma_amd.parallel import tp
+    from fma_amd.runtime.engine import ActuationEngine
+
+    ctx = tp.init_tp(rank, world, master_port, list(range(world)))
+    cfg = LlamaConfig.by_name(model_name)
+    if max_model_len:
+        cfg.max_seq_len = max_model_len
+    eng = ActuationEngine(cfg, device_index=ctx.device_index,
+                          tp_rank=rank, tp_size=world,
+                          tp_group=ctx.device_group, seed=seed)
+    tp.run_worker_loop(ctx, eng)
+
+
+class ServingRuntime:
+    """Everything behind the HTTP app: engine (+TP fan-out) and state."""
+
+    def __init__(self, args: argparse.Namespace):
+        from fma_amd.parallel import tp
+        from fma_amd.runtime.engine import ActuationEngine
+
+        self.args = args
+        self.lock = threading.Lock()
+        self.started_at = time.time()
+        world = args.tensor_parallel_size
+        self.workers: List[multiprocessing.Process] = []
+        master_port = _free_port() if world > 1 else 0
+        mp = multiprocessing.get_context("fork")
+        for r in range(1, world):
+            p = mp.Process(target=_worker_entry,
+                           args=(r, world, master_port, args.model,
+                                 args.max_model_len, args.seed),
+                           daemon=True)
+            p.start()
+            self.workers.append(p)
+        ctx = tp.init_tp(0, world, master_port, list(range(world)))
+        cfg = LlamaConfig.by_name(args.model)
+        if args.max_model_len:
+            cfg.max_seq_len = args.max_model_len
+        engine = ActuationEngine(cfg, device_index=ctx.device_index,
+                                 tp_rank=0, tp_size=world,
+                                 tp_group=ctx.device_group, seed=args.seed)
+        self.rt = tp.TPRuntime(ctx, engine) if world > 1 else engine
+        self.model_name = args.served_model_name or args.model
+        if args.start_asleep:
+            self.rt.sleep(1)
+        self.ready = True
+
+    def close(self) -> None:
+        if hasattr(self.rt, "stop"):
+            self.rt.stop()
+        for p in self.workers:
+            p.join(timeout=5)
+
+
+def create_app(runtime: ServingRuntime) -> FastAPI:
+    app = FastAPI(title="fma-amd inference server", version="0.1")
+    rt = runtime.rt
+
+    @app.get("/health")
+    def health():
+        return {"status": "OK"}
+
+    @app.get("/is_sleeping")
+    def is_sleeping():
+        # JSON shape per reference pkg/api/interface.go:131-135
+        return {"is_sleeping": rt.is_sleeping()}
+
+    @app.post("/sleep")
+    def sleep(level: int = Query(default=1)):
+        with runtime.lock:
+            t = rt.sleep(level)
+        return {"status": "ok", "level": level, "seconds": t}
+
+    @app.post("/wake_up")
+    def wake_up():
+        with runtime.lock:
+            t = rt.wake_up()
+        return {"status": "ok", "seconds": t}
+
+    @app.get("/v1/models")
+    def models():
+        return {"object": "list", "data": [{
+            "id": runtime.model_name, "object": "model",
+            "owned_by": "fma-amd",
+        }]}
+
+    @app.post("/v1/completions")
+    def completions(body: dict):
+        prompt = body.get("prompt", "")
+        max_tokens = int(body.get("max_tokens", 16))
+        with runtime.lock:
+            if rt.is_sleeping():
+                return JSONResponse(
+                    {"error": "model is sleeping"}, status_code=409)
+            t0 = time.time()
+            text = rt.generate_text(prompt, max_tokens)
+        return {
+            "id": f"cmpl-{int(t0*1e6)}",
+            "object": "text_completion",
+            "model": runtime.model_name,
+            "choices": [{"index": 0, "text": text,
+                         "finish_reason": "length"}],
+            "usage": {"completion_tokens": max_tokens},
+        }
+
+    @app.get("/stats")
+    def stats():
+        return rt.stats()
+
+    return app
+
+
+def main(argv: Optional[List[str]] = None) -> None:
+    import sys
+
+    import uvicorn
+
+    args = parse_options(" ".join(argv if argv is not None else sys.argv[1:]))
+    runtime = ServingRuntime(args)
+    app = create_app(runtime)
+    try:
+        uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+    finally:
+        runtime.close()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,201 @@
+"""Launcher REST API tests (CPU; child processes are stubbed sleepers)."""
+
+import asyncio
+import os
+import signal
+import time
+
+import pytest
+from fastapi.testclient import TestClient
+
+os.environ.setdefault("FMA_FAKE_GPU", "1")
+os.environ.setdefault("FMA_MOCK_GPU_COUNT", "4")
+
+import fma_amd.launcher.instance as instance_mod  # noqa: E402
+from fma_amd.launcher.broadcaster import (EventBroadcaster,  # noqa: E402
+                                          RevisionTooOld)
+from fma_amd.launcher.gputranslator import GpuTranslator  # noqa: E402
+from fma_amd.launcher.notifier import instances_signature  # noqa: E402
+from fma_amd.launcher.service import InstanceManager, create_app  # noqa: E402
+
+
+def _stub_kickoff(options, env_vars, log_path):
+    """Child body for tests: write a log and wait for SIGTERM."""
+    with open(log_path, "w") as f:
+        f.write("stub server started: " + options + "\n")
+        f.write("0123456789" * 20)
+        f.flush()
+    signal.signal(signal.SIGTERM, lambda *a: os._exit(0))
+    while True:
+        time.sleep(0.2)
+
+
+@pytest.fixture()
+def client(monkeypatch, tmp_path):
+    monkeypatch.setattr(instance_mod, "kickoff", _stub_kickoff)
+    mgr = InstanceManager(GpuTranslator("naive"), str(tmp_path))
+    app = create_app(mgr)
+    with TestClient(app) as c:
+        c.mgr = mgr
+        yield c
+    mgr.stop_all()
+
+
+ROOT = "/v2/vllm/instances"
+
+
+def _mkconfig(**kw):
+    cfg = {"options": "--model tiny --port 9301", "gpu_uuids": ["GPU-0"],
+           "env_vars": {"X": "1"}}
+    cfg.update(kw)
+    return cfg
+
+
+def test_index_and_health(client):
+    assert client.get("/health").json() == {"status": "OK"}
+    info = client.get("/").json()
+    assert "create_instance" in info["endpoints"]
+
+
+def test_create_auto_id_and_get(client):
+    r = client.post(ROOT, json=_mkconfig())
+    assert r.status_code == 201
+    body = r.json()
+    iid = body["instance_id"]
+    assert body["status"] == "running"
+    assert body["revision"] >= 1
+    got = client.get(f"{ROOT}/{iid}").json()
+    assert got["options"] == "--model tiny --port 9301"
+    assert got["gpu_uuids"] == ["GPU-0"]
+
+
+def test_create_named_conflict(client):
+    r = client.put(f"{ROOT}/inst-a", json=_mkconfig())
+    assert r.status_code == 201
+    r2 = client.put(f"{ROOT}/inst-a", json=_mkconfig())
+    assert r2.status_code == 409
+
+
+def test_list_and_delete(client):
+    client.put(f"{ROOT}/one", json=_mkconfig())
+    client.put(f"{ROOT}/two", json=_mkconfig())
+    ls = client.get(ROOT).json()
+    assert ls["total_instances"] == 2
+    assert ls["running_instances"] == 2
+    assert "revision" in ls
+    r = client.delete(f"{ROOT}/one")
+    assert r.status_code == 200
+    assert r.json()["status"] == "stopped"
+    assert client.get(f"{ROOT}/one").status_code == 404
+    ls = client.get(ROOT).json()
+    assert ls["total_instances"] == 1
+    r = client.delete(ROOT)
+    assert r.json()["deleted"] == 1
+
+
+def test_delete_unknown_404(client):
+    assert client.delete(f"{ROOT}/nope").status_code == 404
+
+
+def test_log_plain_and_ranges(client):
+    client.put(f"{ROOT}/lg", json=_mkconfig())
+    time.sleep(0.3)  # let the stub write its log
+    full = client.get(f"{ROOT}/lg/log")
+    assert full.status_code == 200
+    assert b"stub server started" in full.content
+
+    r = client.get(f"{ROOT}/lg/log", headers={"Range": "bytes=0-9"})
+    assert r.status_code == 206
+    assert len(r.content) == 10
+    assert r.headers["content-range"].startswith("bytes 0-9/")
+
+    size = int(r.headers["content-range"].split("/")[1])
+    r = client.get(f"{ROOT}/lg/log", headers={"Range": f"bytes={size}-"})
+    assert r.status_code == 416
+
+    r = client.get(f"{ROOT}/lg/log", headers={"Range": "bytes=-5"})
+    assert r.status_code == 206
+    assert len(r.content) == 5
+
+    r = client.get(f"{ROOT}/lg/log", headers={"Range": "bytes=zz"})
+    assert r.status_code == 400
+
+    assert client.get(f"{ROOT}/none/log").status_code == 404
+
+
+def test_stopped_instance_detected(client):
+    r = client.put(f"{ROOT}/dying", json=_mkconfig())
+    iid = r.json()["instance_id"]
+    inst = client.mgr.instances[iid]
+    os.kill(inst.pid, signal.SIGTERM)
+    deadline = time.time() + 5
+    while time.time() < deadline:
+        st = client.get(f"{ROOT}/{iid}").json()
+        if st["status"] == "stopped":
+            break
+        time.sleep(0.05)
+    assert st["status"] == "stopped"
+    # a STOPPED watch event was recorded
+    evs = client.mgr.broadcaster._events
+    assert any(e["type"] == "STOPPED" and e["instance_id"] == iid for e in evs)
+
+
+def test_watch_410_when_too_old(client):
+    b = client.mgr.broadcaster
+    for i in range(EventBroadcaster.BUFFER_LIMIT + 10):
+        b.append("CREATED", f"x{i}", b.next_revision())
+    r = client.get(f"{ROOT}/watch", params={"since": 1})
+    assert r.status_code == 410
+
+
+def test_broadcaster_watch_replay():
+    b = EventBroadcaster()
+    for i in range(5):
+        b.append("CREATED", f"i{i}", b.next_revision())
+
+    async def take3():
+        out = []
+        async for ev in b.watch(since=2):
+            out.append(ev)
+            if len(out) == 3:
+                break
+        return out
+
+    out = asyncio.run(take3())
+    assert [e["revision"] for e in out] == [3, 4, 5]
+
+
+def test_broadcaster_check_since():
+    b = EventBroadcaster()
+    for i in range(EventBroadcaster.BUFFER_LIMIT + 5):
+        b.append("CREATED", f"i{i}", b.next_revision())
+    with pytest.raises(RevisionTooOld):
+        b.check_since(1)
+    b.check_since(b.revision)  # current cursor fine
+
+
+def test_signature_stable():
+    body1 = {"instances": [{"instance_id": "a", "status": "running"},
+                           {"instance_id": "b", "status": "stopped"}]}
+    body2 = {"instances": list(reversed(body1["instances"]))}
+    assert instances_signature(body1) == instances_signature(body2)
+    body3 = {"instances": [{"instance_id": "a", "status": "stopped"},
+                           {"instance_id": "b", "status": "stopped"}]}
+    assert instances_signature(body1) != instances_signature(body3)
+
+
+def test_gpu_uuid_translation_sets_env(client, monkeypatch):
+    seen = {}
+    orig = instance_mod.ServerInstance.start
+
+    def spy(self):
+        env = dict(self.config.env_vars)
+        if self.config.gpu_uuids:
+            env["HIP_VISIBLE_DEVICES"] = \
+                self.translator.visible_devices_value(self.config.gpu_uuids)
+        seen.update(env)
+        return orig(self)
+
+    monkeypatch.setattr(instance_mod.ServerInstance, "start", spy)
+    client.put(f"{ROOT}/tr", json=_mkconfig(gpu_uuids=["GPU-2", "GPU-0"]))
+    assert seen["HIP_VISIBLE_DEVICES"] == "2,0"

@@ -1,0 +1,65 @@
+"""Serving runtime HTTP surface (CPU, fake arena)."""
+
+import os
+
+import pytest
+from fastapi.testclient import TestClient
+
+os.environ.setdefault("FMA_FAKE_GPU", "1")
+
+from fma_amd.runtime.server import (ServingRuntime, create_app,  # noqa: E402
+                                    parse_options)
+
+
+@pytest.fixture(scope="module")
+def client():
+    rt = ServingRuntime(parse_options("--model tiny --enable-sleep-mode"))
+    app = create_app(rt)
+    with TestClient(app) as c:
+        yield c
+
+
+def test_parse_options_tolerates_unknown():
+    args = parse_options("--model tiny --port 9000 --gpu-memory-utilization 0.9")
+    assert args.model == "tiny"
+    assert args.port == 9000
+    assert "--gpu-memory-utilization" in args.unknown
+
+
+def test_health(client):
+    r = client.get("/health")
+    assert r.status_code == 200
+    assert r.json() == {"status": "OK"}
+
+
+def test_sleep_wake_cycle(client):
+    assert client.get("/is_sleeping").json() == {"is_sleeping": False}
+    r = client.post("/sleep", params={"level": 1})
+    assert r.status_code == 200
+    assert client.get("/is_sleeping").json() == {"is_sleeping": True}
+    # completions rejected while asleep
+    r = client.post("/v1/completions", json={"prompt": "hi"})
+    assert r.status_code == 409
+    r = client.post("/wake_up")
+    assert r.status_code == 200
+    assert client.get("/is_sleeping").json() == {"is_sleeping": False}
+
+
+def test_completions(client):
+    r = client.post("/v1/completions",
+                    json={"prompt": "hello", "max_tokens": 4})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "text_completion"
+    assert len(body["choices"]) == 1
+
+
+def test_models(client):
+    body = client.get("/v1/models").json()
+    assert body["data"][0]["id"] == "tiny"
+
+
+def test_stats(client):
+    body = client.get("/stats").json()
+    assert body["model"] == "tiny"
+    assert body["sleep_count"] >= 1
