@@ -1,0 +1,113 @@
+"""Prometheus metrics, name-compatible with the reference's dashboards.
+
+Metric names, labels and buckets follow the reference exactly so existing
+dashboards/alerts port over (reference pkg/controller/dual-pods/
+controller.go:204-293, docs/metrics.md:14-212; populator metrics
+pkg/controller/launcher-populator/metrics.go:36-310).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+try:
+    from prometheus_client import (REGISTRY, Counter, Gauge, Histogram,
+                                   start_http_server)
+    HAVE_PROM = True
+except Exception:  # pragma: no cover
+    HAVE_PROM = False
+
+_METRICS = {}
+
+
+def _get_or_make(cls, name, doc, labels=(), **kw):
+    if not HAVE_PROM:
+        return _Noop()
+    key = name
+    if key not in _METRICS:
+        _METRICS[key] = cls(name, doc, labelnames=list(labels), **kw)
+    return _METRICS[key]
+
+
+class _Noop:
+    def labels(self, *a, **kw):
+        return self
+
+    def observe(self, *a):
+        pass
+
+    def inc(self, *a):
+        pass
+
+    def dec(self, *a):
+        pass
+
+    def set(self, *a):
+        pass
+
+
+# reference controller.go:268
+ACTUATION_BUCKETS = (0, 1, 3, 5, 7.5, 10, 15, 30, 60, 120, 240, 480, 960, 1920)
+# reference controller.go:275
+LAUNCHER_CREATE_BUCKETS = (0.001, 0.005, 0.01, 0.05, 0.1, 0.5, 1, 2.5, 5)
+# reference controller.go:284
+HTTP_LATENCY_BUCKETS = (0.001, 0.01, 0.1, 0.3, 1, 3, 10, 30, 90, 270, 810)
+
+
+def actuation_seconds():
+    return _get_or_make(
+        Histogram, "fma_actuation_seconds",
+        "requester container start -> readiness relay, by actuation path",
+        ("path", "instancesDeleted", "isc_name"),
+        buckets=ACTUATION_BUCKETS)
+
+
+def launcher_create_seconds():
+    return _get_or_make(
+        Histogram, "fma_launcher_create_seconds",
+        "kube API create-Pod call latency for launchers",
+        buckets=LAUNCHER_CREATE_BUCKETS)
+
+
+def http_latency_seconds():
+    return _get_or_make(
+        Histogram, "fma_http_latency_seconds",
+        "controller -> stub/launcher/server HTTP latency",
+        ("purpose", "method", "status_code"),
+        buckets=HTTP_LATENCY_BUCKETS)
+
+
+def requester_count():
+    return _get_or_make(Gauge, "fma_requester_count",
+                        "server-requesting Pods known to the controller")
+
+
+def isc_count():
+    return _get_or_make(Gauge, "fma_isc_count",
+                        "InferenceServerConfig objects known")
+
+
+def duality():
+    return _get_or_make(
+        Gauge, "fma_duality",
+        "1 per (requester, provider) binding; labels join to GPU metrics",
+        ("requester_name", "provider_name", "node"))
+
+
+def launcher_pod_count():
+    return _get_or_make(
+        Gauge, "fma_launcher_pod_count",
+        "launcher Pods by LauncherConfig and phase",
+        ("lcfg_name", "phase"))
+
+
+def serve_metrics(port: int = 8002) -> None:
+    """Expose /metrics (reference pkg/observability/prom-and-debug.go:34-79
+    serves :8002; pprof has no Python analog — py-spy attaches externally)."""
+    if HAVE_PROM:
+        start_http_server(port)
+
+
+def observe_http(purpose: str, method: str, status: int, seconds: float
+                 ) -> None:
+    http_latency_seconds().labels(purpose, method, str(status)).observe(seconds)

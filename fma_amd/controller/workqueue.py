@@ -1,0 +1,176 @@
+"""Generic typed work-queue framework (reference pkg/controller/generic/
+queue-work.go:35-141 and knows-processed-sync.go:34-103).
+
+Semantics preserved from the reference:
+- rate-limited re-queue with exponential per-item backoff capped at 20 s;
+- an item added while being processed is re-processed afterwards (dirty
+  set), never processed concurrently with itself;
+- KnowsProcessedSync: sentinel items mark the initial batch; a callback
+  fires once every item enqueued before start has been processed (the
+  populator uses this to gate its key workers —
+  reference populator.go:356-363).
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from typing import Callable, Generic, Hashable, List, Optional, TypeVar
+
+T = TypeVar("T", bound=Hashable)
+
+MAX_BACKOFF_SECONDS = 20.0
+BASE_BACKOFF_SECONDS = 0.005
+
+
+class RateLimitingQueue(Generic[T]):
+    """Subset of client-go's workqueue: Add / AddAfter / AddRateLimited /
+    Forget / Get / Done / ShutDown with dirty/processing sets."""
+
+    def __init__(self) -> None:
+        self._cond = threading.Condition()
+        self._queue: List[T] = []
+        self._dirty: set = set()
+        self._processing: set = set()
+        self._failures: dict = {}
+        self._shutdown = False
+        self._timers: List[threading.Timer] = []
+
+    def add(self, item: T) -> None:
+        with self._cond:
+            if self._shutdown or item in self._dirty:
+                return
+            self._dirty.add(item)
+            if item in self._processing:
+                return
+            self._queue.append(item)
+            self._cond.notify()
+
+    def add_after(self, item: T, delay: float) -> None:
+        if delay <= 0:
+            self.add(item)
+            return
+        t = threading.Timer(delay, self.add, args=(item,))
+        t.daemon = True
+        with self._cond:
+            if self._shutdown:
+                return
+            self._timers.append(t)
+        t.start()
+
+    def add_rate_limited(self, item: T) -> None:
+        with self._cond:
+            n = self._failures.get(item, 0)
+            self._failures[item] = n + 1
+        self.add_after(item, min(BASE_BACKOFF_SECONDS * (2 ** n),
+                                 MAX_BACKOFF_SECONDS))
+
+    def forget(self, item: T) -> None:
+        with self._cond:
+            self._failures.pop(item, None)
+
+    def get(self) -> Optional[T]:
+        with self._cond:
+            while not self._queue and not self._shutdown:
+                self._cond.wait(timeout=0.2)
+            if not self._queue:
+                return None
+            item = self._queue.pop(0)
+            self._dirty.discard(item)
+            self._processing.add(item)
+            return item
+
+    def done(self, item: T) -> None:
+        with self._cond:
+            self._processing.discard(item)
+            if item in self._dirty:
+                self._queue.append(item)
+                self._cond.notify()
+
+    def shut_down(self) -> None:
+        with self._cond:
+            self._shutdown = True
+            for t in self._timers:
+                t.cancel()
+            self._cond.notify_all()
+
+    def __len__(self) -> int:
+        with self._cond:
+            return len(self._queue)
+
+
+class QueueAndWorkers(Generic[T]):
+    """Queue + N worker threads running `process(item) -> bool retry`."""
+
+    def __init__(self, name: str, num_workers: int,
+                 process: Callable[[T], bool]):
+        self.name = name
+        self.queue: RateLimitingQueue[T] = RateLimitingQueue()
+        self.num_workers = num_workers
+        self.process = process
+        self.threads: List[threading.Thread] = []
+
+    def start(self) -> None:
+        for i in range(self.num_workers):
+            th = threading.Thread(target=self._worker, daemon=True,
+                                  name=f"{self.name}-worker-{i}")
+            th.start()
+            self.threads.append(th)
+
+    def _worker(self) -> None:
+        while True:
+            item = self.queue.get()
+            if item is None:
+                return
+            try:
+                retry = self.process(item)
+            except Exception:  # noqa: BLE001 - reconcile must not kill worker
+                import traceback
+                traceback.print_exc()
+                retry = True
+            if retry:
+                self.queue.add_rate_limited(item)
+            else:
+                self.queue.forget(item)
+            self.queue.done(item)
+
+    def stop(self) -> None:
+        self.queue.shut_down()
+        for th in self.threads:
+            th.join(timeout=2)
+
+
+class InitialSyncTracker:
+    """Tracks when every item present at start has been processed once."""
+
+    def __init__(self, on_done: Callable[[], None]):
+        self._pending: set = set()
+        self._started = False
+        self._fired = False
+        self._lock = threading.Lock()
+        self._on_done = on_done
+
+    def register(self, item) -> None:
+        with self._lock:
+            if not self._started:
+                self._pending.add(item)
+
+    def start(self) -> None:
+        fire = False
+        with self._lock:
+            self._started = True
+            fire = not self._pending and not self._fired
+            if fire:
+                self._fired = True
+        if fire:
+            self._on_done()
+
+    def mark_processed(self, item) -> None:
+        fire = False
+        with self._lock:
+            self._pending.discard(item)
+            if self._started and not self._pending and not self._fired:
+                self._fired = True
+                fire = True
+        if fire:
+            self._on_done()
