@@ -13,7 +13,7 @@ unchanged; only the cluster-resource identifiers differ (``amd.com/gpu``).
 from __future__ import annotations
 
 import re
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List, Optional, Union
 
 from pydantic import BaseModel, Field, field_validator
 
@@ -97,10 +97,11 @@ class LauncherConfigSpec(BaseModel):
 
 class ResourceRange(BaseModel):
     """Inclusive min/max quantity bounds
-    (reference launcherpopulationpolicy_types.go:104-121)."""
+    (reference launcherpopulationpolicy_types.go:104-121). Quantities may
+    be strings with k8s suffixes ("512Gi") or plain numbers."""
 
-    min: Optional[str] = None
-    max: Optional[str] = None
+    min: Optional[Union[str, int, float]] = None
+    max: Optional[Union[str, int, float]] = None
 
     def contains(self, value: "str | int | float") -> bool:
         v = parse_quantity(value)

@@ -108,8 +108,17 @@ def select_or_reclaim(launchers: List[LauncherView],
                       desired_port: int,
                       last_used: Optional[Dict[str, float]] = None,
                       ) -> SelectionResult:
-    last_used = last_used or {}
+    # LRU times come from the instance views; an explicit map (the
+    # controller's cache, which may know about instances no longer listed)
+    # overrides them.
+    lu: Dict[str, float] = {}
+    for l in launchers:
+        for inst in l.instances:
+            lu[inst.instance_id] = inst.last_used
+    lu.update(last_used or {})
+
     candidate_with_capacity: Optional[LauncherView] = None
+    capacity_deletions: List[Tuple[str, str]] = []
     some_not_ready = False
     best_plan: Optional[_Plan] = None
 
@@ -124,14 +133,21 @@ def select_or_reclaim(launchers: List[LauncherView],
         has_sleeping = False
         port_conflicts: List[str] = []
         others: List[str] = []
+        stopped: List[str] = []  # dead instances: free deletions
         for inst in lv.instances:
             if inst.port is None:
                 # repair: delete the malformed instance, retry fresh
                 return SelectionResult(
                     retry=True, deletions=[(lv.name, inst.instance_id)])
+            if inst.status == "stopped":
+                # a stopped instance only occupies a slot (and, for the
+                # target id, would 409 a named re-create): always delete
+                # (the reference deletes these in syncLauncherInstances
+                # before selection, inference-server.go:2129-2151)
+                stopped.append(inst.instance_id)
+                continue
             if inst.instance_id == target_instance_id:
-                if inst.status != "stopped":
-                    has_sleeping = True
+                has_sleeping = True
                 continue
             if inst.port == desired_port:
                 port_conflicts.append(inst.instance_id)
@@ -140,22 +156,26 @@ def select_or_reclaim(launchers: List[LauncherView],
         if has_sleeping:
             return SelectionResult(launcher=lv, has_sleeping_instance=True)
 
-        total = len(lv.instances)
-        if not port_conflicts and total <= max_others:
+        total_live = len(port_conflicts) + len(others)
+        if not port_conflicts and total_live <= max_others:
             if candidate_with_capacity is None:
                 candidate_with_capacity = lv
+                capacity_deletions = [(lv.name, s) for s in stopped]
             continue
 
-        to_delete = max(total - max_others, 1)
+        to_delete = max(total_live - max_others, 1)
         victims = port_conflicts + pick_instance_victims(
-            others, last_used, to_delete - len(port_conflicts))
-        lru_id, lru_t = _plan_lru(victims, last_used)
-        plan = _Plan(lv, victims, lru_id, lru_t)
+            others, lu, to_delete - len(port_conflicts))
+        if not victims:
+            continue
+        lru_id, lru_t = _plan_lru(victims, lu)
+        plan = _Plan(lv, victims + stopped, lru_id, lru_t)
         if best_plan is None or _cmp_plans(plan, best_plan) < 0:
             best_plan = plan
 
     if candidate_with_capacity is not None:
-        return SelectionResult(launcher=candidate_with_capacity)
+        return SelectionResult(launcher=candidate_with_capacity,
+                               deletions=capacity_deletions)
     if best_plan is not None:
         return SelectionResult(
             launcher=best_plan.launcher,

@@ -1,0 +1,147 @@
+"""Store semantics (preconditions, finalizers, watch) + admission policies."""
+
+import threading
+
+import pytest
+
+from fma_amd.api import contracts as C
+from fma_amd.store import objects as ob
+from fma_amd.store.admission import install_policies
+from fma_amd.store.memstore import (AlreadyExists, Conflict, Invalid,
+                                    MemStore, NotFound)
+
+
+def test_create_get_update_delete():
+    s = MemStore()
+    pod = s.create(ob.new_object("Pod", "p1"))
+    assert ob.uid_of(pod)
+    assert ob.rv_of(pod)
+    with pytest.raises(AlreadyExists):
+        s.create(ob.new_object("Pod", "p1"))
+    pod["spec"]["x"] = 1
+    pod2 = s.update(pod)
+    assert ob.meta(pod2)["generation"] == 2  # spec change bumps generation
+    pod2["status"] = {"phase": "Running"}
+    pod3 = s.update(pod2)
+    assert ob.meta(pod3)["generation"] == 2  # status change does not
+    s.delete("Pod", "p1")
+    with pytest.raises(NotFound):
+        s.get("Pod", "p1")
+
+
+def test_resource_version_conflict():
+    s = MemStore()
+    pod = s.create(ob.new_object("Pod", "p"))
+    stale = ob.deepcopy(pod)
+    s.update(pod)  # bumps RV
+    stale["spec"]["y"] = 2
+    with pytest.raises(Conflict):
+        s.update(stale)
+
+
+def test_preconditioned_delete():
+    s = MemStore()
+    pod = s.create(ob.new_object("Pod", "p"))
+    with pytest.raises(Conflict):
+        s.delete("Pod", "p", expect_uid="wrong")
+    s.delete("Pod", "p", expect_uid=ob.uid_of(pod))
+
+
+def test_finalizer_gated_deletion():
+    s = MemStore()
+    pod = ob.new_object("Pod", "p")
+    pod["metadata"]["finalizers"] = ["x/protect"]
+    pod = s.create(pod)
+    s.delete("Pod", "p")
+    cur = s.get("Pod", "p")  # still there, deleting
+    assert ob.is_deleting(cur)
+    cur["metadata"]["finalizers"] = []
+    s.update(cur)
+    with pytest.raises(NotFound):
+        s.get("Pod", "p")
+
+
+def test_owner_reference_gc():
+    s = MemStore()
+    owner = s.create(ob.new_object("LauncherConfig", "lc",
+                                   spec={"maxInstances": 1}))
+    child = ob.new_object("Pod", "child")
+    ob.meta(child)["ownerReferences"] = [
+        {"kind": "LauncherConfig", "name": "lc", "uid": ob.uid_of(owner)}]
+    s.create(child)
+    s.delete("LauncherConfig", "lc")
+    assert s.try_get("Pod", "child") is None
+
+
+def test_watch_delivers_events():
+    s = MemStore()
+    stop = threading.Event()
+    got = []
+
+    def consume():
+        for ev in s.watch(stop=stop):
+            got.append((ev.type, ob.name_of(ev.obj)))
+            if len(got) >= 3:
+                return
+
+    t = threading.Thread(target=consume, daemon=True)
+    t.start()
+    pod = s.create(ob.new_object("Pod", "w1"))
+    pod["spec"]["z"] = 1
+    s.update(pod)
+    s.delete("Pod", "w1")
+    t.join(timeout=5)
+    stop.set()
+    assert got == [("ADDED", "w1"), ("MODIFIED", "w1"), ("DELETED", "w1")]
+
+
+def test_watch_replays_history():
+    s = MemStore()
+    s.create(ob.new_object("Pod", "a"))
+    s.create(ob.new_object("Pod", "b"))
+    evs = list(s.watch(since=0, timeout=0.05))
+    assert [ob.name_of(e.obj) for e in evs] == ["a", "b"]
+
+
+# -- admission ------------------------------------------------------------
+
+def bound_requester(s):
+    pod = ob.new_object(
+        "Pod", "req",
+        annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: "isc1"},
+        labels={C.DUAL_LABEL: "launcher1"})
+    return s.create(pod, actor="user")
+
+
+def test_policy_blocks_user_touching_managed_metadata():
+    s = MemStore()
+    install_policies(s)
+    pod = s.create(ob.new_object("Pod", "p"), actor="user")
+    pod = s.get("Pod", "p")
+    ob.annotations_of(pod)[C.REQUESTER_ANNOTATION] = "evil"
+    with pytest.raises(Invalid):
+        s.update(pod, actor="user")
+    # the controller may
+    s.update(pod, actor="dual-pods-controller")
+
+
+def test_policy_blocks_isc_change_on_bound_requester():
+    s = MemStore()
+    install_policies(s)
+    pod = bound_requester(s)
+    pod = s.get("Pod", "req")
+    ob.annotations_of(pod)[C.INFERENCE_SERVER_CONFIG_ANNOTATION] = "other"
+    with pytest.raises(Invalid):
+        s.update(pod, actor="user")
+
+
+def test_policy_allows_isc_change_on_unbound_requester():
+    s = MemStore()
+    install_policies(s)
+    pod = ob.new_object(
+        "Pod", "req",
+        annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: "isc1"})
+    s.create(pod, actor="user")
+    pod = s.get("Pod", "req")
+    ob.annotations_of(pod)[C.INFERENCE_SERVER_CONFIG_ANNOTATION] = "other"
+    s.update(pod, actor="user")  # unbound: allowed
