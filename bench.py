@@ -74,7 +74,8 @@ def main() -> None:
         tp_rank=rank, tp_size=world, tp_group=None,
         use_vmm=args.vmm or None, chunk_bytes=args.chunk_mb << 20, seed=1234,
         nstreams=args.nstreams,
-        slab_bytes=(args.slab_mb << 20) if args.slab_mb else None)
+        slab_bytes=(args.slab_mb << 20) if args.slab_mb else None,
+        actuation_mode=args.mode)
     log(f"[rank {rank}] engine up: {eng.total_bytes/2**30:.2f} GiB/rank, "
         f"{cfg.num_layers} layers, vmm={eng.stats()['uses_vmm']}, "
         f"create {time.perf_counter()-t0:.1f}s")

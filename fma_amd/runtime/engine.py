@@ -36,7 +36,8 @@ class ActuationEngine:
                  tp_rank: int = 0, tp_size: int = 1, tp_group=None,
                  use_vmm=None, chunk_bytes: int = 0, seed: int = 0,
                  init_weights: bool = True, nstreams: int = 1,
-                 slab_bytes: int = None):
+                 slab_bytes: int = None, actuation_mode: str = "arena",
+                 pack_xfer_mode: int = 0):
         self.cfg = cfg
         self.device_index = device_index
         self.tp_rank = tp_rank
@@ -51,15 +52,30 @@ class ActuationEngine:
         specs = cfg.param_specs(tp_rank, tp_size)
         if slab_bytes is None:
             slab_bytes = actuation.DEFAULT_SLAB_BYTES
-        self.layout, self.total_bytes, slab_sizes = actuation.plan_layout(
-            specs, slab_bytes=slab_bytes)
+        self.actuation_mode = actuation_mode
         t0 = time.perf_counter()
-        self.arena = actuation.make_arena(self.total_bytes, device_index,
-                                          try_vmm=use_vmm,
-                                          chunk_bytes=chunk_bytes,
-                                          slab_sizes=slab_sizes,
-                                          nstreams=nstreams)
-        self.params = self._make_views()
+        if actuation_mode == "pack":
+            # scattered-tensor mode: parameters live in ordinary caching-
+            # allocator storage; sleep/wake goes through the HIP
+            # gather/scatter kernel (ops.actuation.PackActuator)
+            self.layout, self.total_bytes, _ = actuation.plan_layout(specs)
+            self.params = {
+                name: torch.empty(shape, dtype=dtype, device=self.device)
+                for name, (off, shape, dtype) in self.layout.items()}
+            self.arena = None
+            self.packer = actuation.PackActuator(
+                self.params, mode=pack_xfer_mode, chunk_bytes=chunk_bytes)                 if self.on_gpu else _FakePacker(self.params)
+            self.total_bytes = self.packer.total_bytes
+        else:
+            self.layout, self.total_bytes, slab_sizes = actuation.plan_layout(
+                specs, slab_bytes=slab_bytes)
+            self.arena = actuation.make_arena(self.total_bytes, device_index,
+                                              try_vmm=use_vmm,
+                                              chunk_bytes=chunk_bytes,
+                                              slab_sizes=slab_sizes,
+                                              nstreams=nstreams)
+            self.packer = None
+            self.params = self._make_views()
         self.model = LlamaModel(cfg, self.params, self.device,
                                 tp_rank, tp_size, tp_group)
         if init_weights:
@@ -94,7 +110,10 @@ class ActuationEngine:
         """
         if self.state == self.SLEEPING:
             return 0.0
-        t = self.arena.sleep(self.host)
+        if self.packer is not None:
+            t = self.packer.sleep(self.host)
+        else:
+            t = self.arena.sleep(self.host)
         if self.on_gpu:
             # return caching-allocator reserves (activations, KV) so another
             # instance's wake can claim the HBM
@@ -110,11 +129,14 @@ class ActuationEngine:
         if self.state == self.AWAKE:
             return 0.0
         t0 = time.perf_counter()
-        _, invalidated = self.arena.wake(self.host)
-        if invalidated:
-            # non-VMM fallback: arena base moved; re-point the views
-            self.params = self._make_views()
-            self.model.rebind(self.params)
+        if self.packer is not None:
+            self.packer.wake(self.host)
+        else:
+            _, invalidated = self.arena.wake(self.host)
+            if invalidated:
+                # non-VMM fallback: arena base moved; re-point the views
+                self.params = self._make_views()
+                self.model.rebind(self.params)
         if self.tp_size > 1 and dist.is_initialized():
             dist.barrier(group=self.tp_group)
         t = time.perf_counter() - t0
@@ -152,9 +174,41 @@ class ActuationEngine:
             "tp_rank": self.tp_rank,
             "tp_size": self.tp_size,
             "uses_vmm": getattr(self.arena, "uses_vmm", False),
+            "actuation_mode": self.actuation_mode,
             "sleep_count": self.sleep_count,
             "wake_count": self.wake_count,
             "last_sleep_seconds": self.last_sleep_seconds,
             "last_wake_seconds": self.last_wake_seconds,
             "create_seconds": self.create_seconds,
         }
+
+
+class _FakePacker:
+    """CPU emulation of PackActuator for GPU-less tests."""
+
+    def __init__(self, params):
+        self.params = params
+        self.total_bytes = sum(
+            ((p.nbytes + 255) // 256) * 256 for p in params.values())
+        self.asleep = False
+
+    def sleep(self, host):
+        off = 0
+        for name in sorted(self.params):
+            p = self.params[name]
+            raw = p.contiguous().view(torch.uint8).view(-1)
+            host[off:off + raw.numel()].copy_(raw)
+            p.zero_()
+            off += ((p.nbytes + 255) // 256) * 256
+        self.asleep = True
+        return 1e-9
+
+    def wake(self, host):
+        off = 0
+        for name in sorted(self.params):
+            p = self.params[name]
+            n = p.nbytes
+            p.view(torch.uint8).view(-1).copy_(host[off:off + n])
+            off += ((n + 255) // 256) * 256
+        self.asleep = False
+        return 1e-9

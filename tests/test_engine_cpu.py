@@ -148,3 +148,16 @@ def test_plan_layout_oversized_tensor_gets_own_slab():
     layout, total, slabs = actuation.plan_layout(specs, slab_bytes=1 << 20)
     assert sum(slabs) == total
     assert len(slabs) == 3  # a | big | b
+
+
+def test_pack_mode_engine_bit_exact():
+    eng = ActuationEngine(LlamaConfig.tiny(), seed=3, actuation_mode="pack")
+    toks = torch.randint(0, eng.cfg.vocab_size, (1, 5))
+    before = eng.model.forward(toks).clone()
+    snap = {n: p.clone() for n, p in eng.params.items()}
+    eng.sleep()
+    assert eng.is_sleeping()
+    eng.wake_up()
+    for n, p in eng.params.items():
+        assert torch.equal(p, snap[n]), n
+    assert torch.equal(before, eng.model.forward(toks))
