@@ -13,9 +13,12 @@ struct FmaCopyDesc {
   unsigned long long bytes;
 };
 
-// LDS-staged prefix table bounds the descriptor count per launch:
-// (n+1) * 8 B of dynamic LDS, capped well under the 160 KiB/CU budget.
-#define FMA_MAX_DESCS_PER_LAUNCH 8192
+// LDS-staged prefix + descriptor tables bound the descriptor count per
+// launch: (n+1)*8 + n*24 B of dynamic LDS, kept under the default 64 KiB
+// dynamic-LDS-per-workgroup limit. The chunked planner keeps per-chunk
+// descriptor counts far below this (raise chunk_bytes if a model with
+// thousands of tiny tensors ever trips it).
+#define FMA_MAX_DESCS_PER_LAUNCH 2000
 #define FMA_ARENA_ALIGN 256
 
 extern "C" hipError_t fma_launch_batched_copy(const FmaCopyDesc* descs_dev,
