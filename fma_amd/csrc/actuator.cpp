@@ -460,6 +460,39 @@ double restore_from_host(const std::vector<at::Tensor>& tensors,
   return seconds_since(t0);
 }
 
+// Isolated D2D gather: tensors -> one device buffer via the batched-copy
+// kernel only (no PCIe traffic). Exists so the kernel's own bandwidth can
+// be measured without the pipeline around it (rocprof microbench).
+double gather_d2d(const std::vector<at::Tensor>& tensors,
+                  const std::vector<int64_t>& offsets, at::Tensor out,
+                  int64_t repeats) {
+  validate_tensors(tensors, offsets);
+  TORCH_CHECK(out.is_cuda() && out.scalar_type() == at::kByte &&
+              out.is_contiguous(), "out must be a contiguous uint8 GPU tensor");
+  const int device = tensors[0].device().index();
+  auto& ctx = ctx_for(device);
+  FMA_HIP_CHECK(hipSetDevice(device));
+  const int64_t total = flat_extent(tensors, offsets);
+  TORCH_CHECK(out.nbytes() >= static_cast<size_t>(total), "out too small");
+  ChunkPlan plan = build_chunk_plan(tensors, offsets, total, /*pack=*/true);
+  auto descs = retarget(plan, 0, static_cast<unsigned char*>(out.data_ptr()),
+                        /*pack=*/true);
+  std::vector<unsigned long long> prefix(
+      plan.prefix.begin(), plan.prefix.begin() + descs.size() + 1);
+  DevBlob dev;
+  dev.upload(descs, prefix, ctx.kernel_stream);
+  FMA_HIP_CHECK(hipStreamSynchronize(ctx.kernel_stream));
+  join_torch_stream(ctx);
+  const auto t0 = Clock::now();
+  for (int64_t r = 0; r < std::max<int64_t>(repeats, 1); ++r) {
+    FMA_HIP_CHECK(fma_launch_batched_copy(dev.descs, dev.prefix,
+                                          static_cast<int>(descs.size()),
+                                          plan.units[0], ctx.kernel_stream));
+  }
+  FMA_HIP_CHECK(hipStreamSynchronize(ctx.kernel_stream));
+  return seconds_since(t0) / std::max<int64_t>(repeats, 1);
+}
+
 // ---------------------------------------------------------------------------
 // DeviceArena
 // ---------------------------------------------------------------------------
@@ -763,6 +796,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("tensors"), py::arg("offsets"), py::arg("host"),
         py::arg("mode") = 0, py::arg("chunk_bytes") = 0);
   m.def("device_supports_vmm", &device_supports_vmm, py::arg("device"));
+  m.def("gather_d2d", &gather_d2d,
+        "Batched gather kernel D2D microbench (returns seconds/iteration)",
+        py::arg("tensors"), py::arg("offsets"), py::arg("out"),
+        py::arg("repeats") = 1);
   m.def("device_mem_info", &device_mem_info, py::arg("device"));
   py::class_<DeviceArena>(m, "DeviceArena")
       .def(py::init<int64_t, int, bool, std::vector<int64_t>>(),
