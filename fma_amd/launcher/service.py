@@ -277,7 +277,7 @@ def main() -> None:
 
     ap = argparse.ArgumentParser("fma-launcher")
     ap.add_argument("--port", type=int, default=contracts.LAUNCHER_SERVICE_PORT)
-    ap.add_argument("--host", default="0.0.0.0")
+    ap.add_argument("--host", default=None)
     ap.add_argument("--gpu-mode", default=None,
                     choices=[None, "real", "gpu-map", "naive"])
     ap.add_argument("--log-dir", default="/tmp")
@@ -291,9 +291,11 @@ def main() -> None:
     if torch.cuda.is_available():
         import fma_amd._C  # noqa: F401  HIP actuator must be present
 
+    import os as _os
+    host = args.host or _os.environ.get("FMA_BIND_HOST", "0.0.0.0")
     mgr = InstanceManager(GpuTranslator(args.gpu_mode), args.log_dir)
     app = create_app(mgr)
-    uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
+    uvicorn.run(app, host=host, port=args.port, log_level="warning")
 
 
 if __name__ == "__main__":

@@ -238,12 +238,13 @@ def main() -> None:
     probes = create_probes_app(state)
     spi = create_spi_app(state)
 
+    host = os.environ.get("FMA_BIND_HOST", "0.0.0.0")
     t = threading.Thread(
-        target=lambda: uvicorn.run(probes, host="0.0.0.0", port=probes_port,
+        target=lambda: uvicorn.run(probes, host=host, port=probes_port,
                                    log_level="warning"),
         daemon=True)
     t.start()
-    uvicorn.run(spi, host="0.0.0.0", port=spi_port, log_level="warning")
+    uvicorn.run(spi, host=host, port=spi_port, log_level="warning")
 
 
 if __name__ == "__main__":

@@ -44,7 +44,7 @@ def parse_options(options: str) -> argparse.Namespace:
     ap = argparse.ArgumentParser(prog="fma-serve", add_help=False)
     ap.add_argument("--model", default="tiny")
     ap.add_argument("--port", type=int, default=8000)
-    ap.add_argument("--host", default="0.0.0.0")
+    ap.add_argument("--host", default=None)
     ap.add_argument("--tensor-parallel-size", type=int, default=1)
     ap.add_argument("--enable-sleep-mode", action="store_true")
     ap.add_argument("--max-model-len", type=int, default=None)
@@ -54,6 +54,9 @@ def parse_options(options: str) -> argparse.Namespace:
                     help="immediately sleep after load (pre-warmed instance)")
     args, unknown = ap.parse_known_args(options.split())
     args.unknown = unknown
+    if args.host is None:
+        # node agent / launcher export the Pod's loopback identity
+        args.host = os.environ.get("FMA_BIND_HOST", "0.0.0.0")
     return args
 
 

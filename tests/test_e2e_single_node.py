@@ -1,0 +1,183 @@
+"""End-to-end single-node actuation test (CPU, real processes, real HTTP).
+
+The analog of the reference's kind-without-GPUs e2e
+(reference test/e2e/run-launcher-based.sh + test-cases.sh): an in-process
+cluster store + node agent play apiserver + kubelet; the launcher, serving
+runtime and requester stub run as REAL separate processes on per-Pod
+loopback IPs; the dual-pods controller and launcher-populator run their
+real threads and reconcile over real HTTP.
+
+Flow exercised: LPP/LC -> populator creates a launcher Pod -> node agent
+spawns the launcher process -> requester Pod (ISC annotation) -> controller
+discovers GPUs via the stub SPI, binds the launcher, creates the instance
+(launcher forks the serving runtime), relays readiness -> requester /ready
+goes 200. Then deletion -> sleep + unbind, and a second requester hot-
+starts on the sleeping instance.
+"""
+
+import os
+import sys
+import time
+
+import httpx
+import pytest
+
+from fma_amd.api import contracts as C
+from fma_amd.controller.dualpods.controller import (ControllerConfig,
+                                                    DualPodsController)
+from fma_amd.controller.httpadapter import HttpAdapter
+from fma_amd.controller.populator.populator import LauncherPopulator
+from fma_amd.node.agent import NodeAgent
+from fma_amd.store import objects as ob
+from fma_amd.store.admission import install_policies
+from fma_amd.store.memstore import MemStore
+
+pytestmark = pytest.mark.timeout(180)
+
+ISC_PORT = 8355
+
+
+def wait_for(cond, timeout=60, interval=0.25, desc="condition"):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        v = cond()
+        if v:
+            return v
+        time.sleep(interval)
+    raise AssertionError(f"timed out waiting for {desc}")
+
+
+@pytest.fixture()
+def cluster(tmp_path):
+    store = MemStore()
+    install_policies(store)
+    node = ob.new_object("Node", "node-a", labels={"gpu": "mi355x"})
+    node["status"] = {"allocatable": {C.GPU_RESOURCE_NAME: 8}}
+    store.create(node)
+
+    env = {
+        "PYTHONPATH": os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))),
+        "FMA_FAKE_GPU": "1",
+        "FMA_MOCK_GPU_COUNT": "4",
+        "FMA_GPU_MODE": "naive",
+        "FMA_ACCELERATORS": "GPU-0",
+    }
+    agent = NodeAgent(store, "node-a", node_index=7,
+                      log_dir=str(tmp_path), extra_env=env)
+    agent.start()
+
+    ctl = DualPodsController(store, HttpAdapter(), ControllerConfig())
+    ctl.start()
+    pop = LauncherPopulator(store)
+    pop.start()
+    yield {"store": store, "agent": agent, "ctl": ctl, "pop": pop}
+    ctl.stop()
+    pop.stop()
+    agent.stop()
+
+
+def mk_isc_lc_lpp(store):
+    store.create(ob.new_object(
+        "LauncherConfig", "lc1",
+        spec={"maxInstances": 2, "podTemplate": {"spec": {"containers": [{
+            "name": "launcher",
+            "command": [sys.executable, "-m", "fma_amd.launcher.service"],
+        }]}}}))
+    store.create(ob.new_object(
+        "InferenceServerConfig", "isc1",
+        spec={"modelServerConfig": {
+            "port": ISC_PORT,
+            "options": "--model tiny",
+            "labels": {"llm-d.ai/model": "tiny"}},
+            "launcherConfigName": "lc1"}))
+    store.create(ob.new_object(
+        "LauncherPopulationPolicy", "lpp1",
+        spec={"enhancedNodeSelector": {
+            "labelSelector": {"matchLabels": {"gpu": "mi355x"}}},
+            "countForLauncher": [
+                {"launcherConfigName": "lc1", "launcherCount": 1}]}))
+
+
+def mk_requester(store, name):
+    pod = ob.new_object(
+        "Pod", name,
+        annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: "isc1"},
+        spec={"nodeName": "node-a", "containers": [{
+            "name": "requester",
+            "command": [sys.executable, "-m", "fma_amd.requester.server"],
+        }]})
+    return store.create(pod, actor="user")
+
+
+def launcher_pod(store):
+    pods = [p for p in store.list("Pod")
+            if ob.labels_of(p).get(C.COMPONENT_LABEL) == C.LAUNCHER_COMPONENT]
+    return pods[0] if pods else None
+
+
+def requester_ready(store, agent, name):
+    pod = store.try_get("Pod", name)
+    if pod is None:
+        return False
+    pp = agent.pods.get(name)
+    if pp is None:
+        return False
+    try:
+        r = httpx.get(f"http://{pp.ip}:8080/ready", timeout=2)
+        return r.status_code == 200
+    except httpx.HTTPError:
+        return False
+
+
+def test_full_launcher_based_actuation(cluster):
+    store, agent = cluster["store"], cluster["agent"]
+    mk_isc_lc_lpp(store)
+
+    # populator creates a launcher; agent runs it and it becomes Ready
+    lp = wait_for(lambda: launcher_pod(store), 30, desc="launcher pod")
+    wait_for(lambda: ob.pod_is_ready(store.get("Pod", ob.name_of(lp))),
+             60, desc="launcher Ready")
+
+    # requester arrives; full actuation to readiness
+    mk_requester(store, "req1")
+    wait_for(lambda: requester_ready(store, agent, "req1"), 90,
+             desc="requester /ready 200")
+
+    lp = store.get("Pod", ob.name_of(lp))
+    anns = ob.annotations_of(lp)
+    assert anns[C.REQUESTER_ANNOTATION].endswith(" req1")
+    assert ob.labels_of(lp)[C.SLEEPING_LABEL] == "false"
+    assert ob.labels_of(lp)["llm-d.ai/model"] == "tiny"
+    iid = anns[C.INSTANCE_ID_ANNOTATION]
+
+    # the serving instance answers completions on the launcher's IP
+    lp_ip = lp["status"]["podIP"]
+    r = httpx.post(f"http://{lp_ip}:{ISC_PORT}/v1/completions",
+                   json={"prompt": "hi", "max_tokens": 2}, timeout=10)
+    assert r.status_code == 200
+
+    # notifier signature reflected onto the Pod
+    wait_for(lambda: C.INSTANCE_SIGNATURE_ANNOTATION in ob.annotations_of(
+        store.get("Pod", ob.name_of(lp))), 20, desc="notifier signature")
+
+    # delete the requester: unbind + sleep, launcher survives
+    store.delete("Pod", "req1", actor="user")
+    wait_for(lambda: store.try_get("Pod", "req1") is None, 60,
+             desc="requester gone")
+    lp = store.get("Pod", ob.name_of(lp))
+    assert C.REQUESTER_ANNOTATION not in ob.annotations_of(lp)
+    assert ob.labels_of(lp)[C.SLEEPING_LABEL] == "true"
+    assert "llm-d.ai/model" not in ob.labels_of(lp)
+    r = httpx.get(f"http://{lp_ip}:{ISC_PORT}/is_sleeping", timeout=5)
+    assert r.json() == {"is_sleeping": True}
+
+    # second requester with the same ISC: hot start on the same instance
+    mk_requester(store, "req2")
+    wait_for(lambda: requester_ready(store, agent, "req2"), 60,
+             desc="req2 ready (hot start)")
+    lp = store.get("Pod", ob.name_of(lp))
+    assert ob.annotations_of(lp)[C.REQUESTER_ANNOTATION].endswith(" req2")
+    assert ob.annotations_of(lp)[C.INSTANCE_ID_ANNOTATION] == iid
+    r = httpx.get(f"http://{lp_ip}:{ISC_PORT}/is_sleeping", timeout=5)
+    assert r.json() == {"is_sleeping": False}
