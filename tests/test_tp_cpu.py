@@ -28,16 +28,22 @@ rt = ServingRuntime(parse_options(
 r = rt.rt
 assert not r.is_sleeping()
 toks = torch.randint(0, r.engine.cfg.vocab_size, (1, 6))
-before = r.engine.model.forward(toks).clone()
+# NOTE: every model invocation must go through the TPRuntime command API
+# (a bare rank-0 forward would issue collectives with no partner)
+before = r.generate(toks, max_new_tokens=3).clone()
+param_snap = {n: p.clone() for n, p in r.engine.params.items()}
 r.sleep(1)
 assert r.is_sleeping()
 r.wake_up()  # includes the all-rank barrier
 assert not r.is_sleeping()
-after = r.engine.model.forward(toks)
-assert torch.equal(before, after), "TP sleep/wake corrupted weights"
-out = r.generate(toks, max_new_tokens=3)
-assert out.shape == (1, 9), out.shape
+for n, p in r.engine.params.items():
+    assert torch.equal(p, param_snap[n]), f"rank0 param {n} corrupted"
+after = r.generate(toks, max_new_tokens=3)
+assert torch.equal(before, after), "TP sleep/wake changed decode output"
+assert after.shape == (1, 9), after.shape
 assert r.stats()["tp_size"] == 2
+txt = r.generate_text("hello", 2)
+assert isinstance(txt, str)
 r.stop()
 print("TP_PROBE_OK")
 """
