@@ -19,7 +19,7 @@ import os
 os.environ.setdefault("FMA_FAKE_GPU", "1")
 os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
 import sys
-sys.path.insert(0, {root!r})
+sys.path.insert(0, "__ROOT__")
 import torch
 from fma_amd.runtime.server import ServingRuntime, parse_options
 
@@ -52,7 +52,7 @@ print("TP_PROBE_OK")
 def test_tp_runtime_sleep_wake_generate(tmp_path):
     root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     res = subprocess.run(
-        [sys.executable, "-c", PROBE.format(root=root)],
+        [sys.executable, "-c", PROBE.replace("__ROOT__", root)],
         capture_output=True, text=True, timeout=150,
         env=dict(os.environ, PYTHONPATH=root))
     assert res.returncode == 0, f"stdout={res.stdout}\nstderr={res.stderr}"
