@@ -284,7 +284,14 @@ def main() -> None:
     args = ap.parse_args()
 
     # Pre-import the heavy modules so forked instances skip cold-start —
-    # the core launcher trick (reference launcher.py:39-42).
+    # the core launcher trick (reference launcher.py:39-42) — and pre-mmap
+    # any known model checkpoints so a swapped-in server's weight load
+    # hits the page cache (reference docs/dual-pods.md:599-608 analog).
+    import os as _os2
+    for d in (_os2.environ.get("FMA_PREMAP_MODELS") or "").split(":"):
+        if d and _os2.path.isdir(d):
+            from fma_amd.models.loader import premap_safetensors
+            premap_safetensors(d)
     import torch  # noqa: F401
     import fma_amd.runtime.engine  # noqa: F401
     import fma_amd.runtime.server  # noqa: F401

