@@ -287,6 +287,7 @@ class LauncherPopulator:
         desired, lc_digest = self.policy.snapshot_for_key(node, lc_name)
         pods = self._launchers_for_key(node, lc_name)
         self._record_phases(lc_name, pods)
+        self._schedule_phase_flip(key, pods)
         if desired == HANDS_OFF or lc_digest is None:
             return False
 
@@ -358,6 +359,29 @@ class LauncherPopulator:
             self._expect_stamp[key] = self.clock.time()
         except (NotFound, Conflict):
             pass
+
+    def _schedule_phase_flip(self, key: Tuple[str, str],
+                             pods: List[Dict[str, Any]]) -> None:
+        """Timed re-reconcile exactly at the next stuck-phase transition —
+        no polling sweep (reference metrics.go:303-310,
+        reportStuckLaunchers populator.go:579-626)."""
+        soonest: Optional[float] = None
+        now = self.clock.time()
+        for pod in pods:
+            if ob.annotations_of(pod).get(contracts.REQUESTER_ANNOTATION):
+                continue
+            created = ob.meta(pod).get("creationTimestamp") or now
+            scheduled = bool(ob.pod_node_name(pod))
+            if not scheduled:
+                flip = created + STUCK_SCHEDULING_SECONDS
+            elif not ob.pod_is_ready(pod):
+                flip = created + STUCK_STARTING_SECONDS
+            else:
+                continue
+            if flip > now and (soonest is None or flip < soonest):
+                soonest = flip
+        if soonest is not None:
+            self.key_queue.queue.add_after(key, soonest - now + 0.05)
 
     # -- stuck detection (reference metrics.go:244-310) -------------------
 

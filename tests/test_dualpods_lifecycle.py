@@ -1,0 +1,169 @@
+"""Lifecycle cases mirroring the reference's e2e suite
+(reference test/e2e/test-cases.sh:260-905), run against the fake store.
+"""
+
+import copy
+
+from fma_amd.api import contracts as C
+from fma_amd.controller.dualpods.controller import (ControllerConfig,
+                                                    DualPodsController)
+from fma_amd.controller.dualpods.identity import instance_id
+from fma_amd.store import objects as ob
+
+from tests.test_dualpods_controller import (FakeInstanceServer, FakeLauncher,
+                                            MSC, Stub, drive, infsvr_item,
+                                            mk_world)
+
+
+def add_requester(w, name, isc_name, ip):
+    req = ob.new_object(
+        "Pod", name,
+        annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: isc_name},
+        spec={"nodeName": "node-a", "containers": [{"name": "stub"}]})
+    req = w["store"].create(req)
+    req["status"] = {"phase": "Running", "podIP": ip}
+    w["store"].update(req)
+    w["http"].register(f"{ip}:8081", Stub(["GPU-0"]))
+    return req
+
+
+def test_multiple_instances_share_one_launcher():
+    """ISC-A then ISC-B (different ports) use the same launcher; both
+    instances coexist (one awake, one sleeping); ISC-A hot-starts again
+    (reference test-cases.sh:512 and :560)."""
+    w = mk_world(sleeping_target=False)  # empty launcher, maxInstances=2
+    store, http, launcher = w["store"], w["http"], w["launcher"]
+
+    msc_b = copy.deepcopy(MSC)
+    msc_b["port"] = 8010
+    msc_b["labels"] = {"llm-d.ai/model": "other"}
+    store.create(ob.new_object(
+        "InferenceServerConfig", "isc2",
+        spec={"modelServerConfig": msc_b, "launcherConfigName": "lc1"}))
+
+    def on_create(iid, inst):
+        port = 8000 if "--port 8000" in inst["options"] else 8010
+        http.register(f"10.0.0.2:{port}", FakeInstanceServer(sleeping=False))
+
+    launcher.on_create = on_create
+
+    # requester A uses isc1 (port 8000)
+    drive(w["ctl"], infsvr_item(store))
+    iid_a = w["iid"]
+    assert iid_a in launcher.instances
+
+    # A leaves; instance A sleeps
+    store.delete("Pod", "req1")
+    drive(w["ctl"], infsvr_item(store))
+
+    # requester B uses isc2 (port 8010) on the SAME launcher
+    add_requester(w, "reqB", "isc2", "10.0.0.5")
+    drive(w["ctl"], infsvr_item(store, "reqB"))
+    iid_b = instance_id(msc_b, ["GPU-0"])
+    assert iid_b in launcher.instances
+    assert iid_a in launcher.instances, "sleeping instance A survived"
+    assert len(launcher.instances) == 2
+    lp = store.get("Pod", "launcher1")
+    assert ob.annotations_of(lp)[C.REQUESTER_ANNOTATION].endswith(" reqB")
+
+    # B leaves; A returns -> hot start on the surviving instance A
+    store.delete("Pod", "reqB")
+    drive(w["ctl"], infsvr_item(store, "reqB"))
+    add_requester(w, "reqA2", "isc1", "10.0.0.6")
+    drive(w["ctl"], infsvr_item(store, "reqA2"))
+    lp = store.get("Pod", "launcher1")
+    assert ob.annotations_of(lp)[C.REQUESTER_ANNOTATION].endswith(" reqA2")
+    assert len(launcher.instances) == 2  # no new instance created
+
+
+def test_isc_gc_deletes_obsolete_sleeping_instance():
+    """ISC spec change deletes the now-stale sleeping, unbound instance
+    (reference test-cases.sh:745, instanceGCItem)."""
+    w = mk_world()
+    store, launcher = w["store"], w["launcher"]
+    drive(w["ctl"], infsvr_item(store))
+    store.delete("Pod", "req1")
+    drive(w["ctl"], infsvr_item(store))
+    assert w["iid"] in launcher.instances  # sleeping, unbound
+
+    isc = store.get("InferenceServerConfig", "isc1")
+    isc["spec"]["modelServerConfig"]["options"] = "--model tiny --seed 9"
+    store.update(isc)
+    w["ctl"].reconcile_isc_gc("node-a", "isc1")
+    assert w["iid"] not in launcher.instances, "obsolete instance not GCed"
+
+
+def test_isc_gc_keeps_current_instance():
+    w = mk_world()
+    store, launcher = w["store"], w["launcher"]
+    drive(w["ctl"], infsvr_item(store))
+    store.delete("Pod", "req1")
+    drive(w["ctl"], infsvr_item(store))
+    w["ctl"].reconcile_isc_gc("node-a", "isc1")  # unchanged ISC
+    assert w["iid"] in launcher.instances
+
+
+def test_obsolete_awake_instance_deleted_on_unbind():
+    """If the ISC changed while bound, unbind DELETES the instance instead
+    of sleeping it (reference test-cases.sh:784,
+    maybeDeleteObsoleteInstance)."""
+    w = mk_world()
+    store, launcher = w["store"], w["launcher"]
+    drive(w["ctl"], infsvr_item(store))
+    isc = store.get("InferenceServerConfig", "isc1")
+    isc["spec"]["modelServerConfig"]["options"] = "--model tiny --v2"
+    store.update(isc)
+    store.delete("Pod", "req1")
+    drive(w["ctl"], infsvr_item(store))
+    assert w["iid"] not in launcher.instances, \
+        "obsolete instance should be deleted, not slept"
+    assert w["inst_srv"].sleeps == 0
+
+
+def test_unbound_launcher_sync_deletes_stopped_and_sleeps_awake():
+    """Unbound launcher housekeeping (reference syncLauncherInstances +
+    test-cases.sh:905): stopped instances deleted; an awake instance on
+    an unbound launcher is put back to sleep."""
+    w = mk_world(sleeping_target=False)
+    launcher, http = w["launcher"], w["http"]
+    launcher.instances["Istoppedi"] = {
+        "instance_id": "Istoppedi", "status": "stopped",
+        "options": "--port 9100",
+        "annotations": {"inference-port": "9100"}}
+    awake_srv = FakeInstanceServer(sleeping=False)
+    http.register("10.0.0.2:9200", awake_srv)
+    launcher.instances["Iawakei"] = {
+        "instance_id": "Iawakei", "status": "running",
+        "options": "--port 9200",
+        "annotations": {"inference-port": "9200"}}
+    w["ctl"].reconcile_unbound_launcher("node-a", "launcher1")
+    assert "Istoppedi" not in launcher.instances
+    assert awake_srv.sleeping, "awake instance on unbound launcher not slept"
+
+
+def test_same_node_collision_two_requesters_one_slot():
+    """Two requesters, one single-slot launcher: one binds, the other
+    waits (cold-start creation kicks in) — no double-bind
+    (reference test-cases.sh:396)."""
+    w = mk_world(sleeping_target=False, max_instances=1)
+
+    def on_create(iid, inst):
+        w["http"].register("10.0.0.2:8000",
+                           FakeInstanceServer(sleeping=False))
+
+    w["launcher"].on_create = on_create
+    add_requester(w, "reqX", "isc1", "10.0.0.7")
+    drive(w["ctl"], infsvr_item(w["store"]))
+
+    item_x = infsvr_item(w["store"], "reqX")
+    for _ in range(6):
+        w["ctl"]._process(item_x)
+    lp = w["store"].get("Pod", "launcher1")
+    bound_to = ob.annotations_of(lp)[C.REQUESTER_ANNOTATION]
+    # exactly one of them is bound to launcher1
+    assert bound_to.endswith(" req1") or bound_to.endswith(" reqX")
+    # and no Pod carries a second binding to the same launcher
+    bindings = [ob.annotations_of(p).get(C.REQUESTER_ANNOTATION)
+                for p in w["store"].list("Pod")
+                if ob.annotations_of(p).get(C.REQUESTER_ANNOTATION)]
+    assert len(bindings) == len(set(bindings))
