@@ -89,11 +89,16 @@ def main() -> None:
         if on_gpu:
             torch.cuda.synchronize()
 
+    def alloc_wait():
+        return getattr(eng.arena, "last_alloc_wait_seconds", 0.0) \
+            if eng.arena is not None else 0.0
+
     # warmup
     for i in range(args.warmup):
         ts = eng.sleep()
         tw = eng.wake_up()
-        log(f"[rank {rank}] warmup {i}: sleep {ts:.3f}s wake {tw:.3f}s")
+        log(f"[rank {rank}] warmup {i}: sleep {ts:.3f}s wake {tw:.3f}s "
+            f"(alloc-wait {alloc_wait():.3f}s)")
 
     barrier()
     sync()
@@ -103,6 +108,8 @@ def main() -> None:
     for _ in range(args.steps):
         sleep_times.append(eng.sleep())
         wake_times.append(eng.wake_up())
+        log(f"[rank {rank}] step: sleep {sleep_times[-1]:.3f}s "
+            f"wake {wake_times[-1]:.3f}s (alloc-wait {alloc_wait():.3f}s)")
     barrier()
     sync()
     wall1 = time.perf_counter()

@@ -30,9 +30,11 @@
 #include <torch/extension.h>
 
 #include <algorithm>
+#include <atomic>
 #include <chrono>
 #include <cstring>
 #include <mutex>
+#include <thread>
 #include <tuple>
 #include <unordered_map>
 #include <vector>
@@ -568,6 +570,8 @@ class DeviceArena {
   int device() const { return device_; }
   int64_t generation() const { return generation_; }
   double last_map_seconds() const { return map_seconds_; }
+  double last_alloc_wait_seconds() const { return alloc_wait_seconds_; }
+  void set_alloc_threads(int n) { alloc_threads_ = std::max(1, n); }
   int64_t num_slabs() const {
     return static_cast<int64_t>(vmm_ ? 1 : slab_sizes_.size());
   }
@@ -633,17 +637,50 @@ class DeviceArena {
                                      ctx.copy_streams[c % ns]));
       }
     } else {
-      double alloc_acc = 0.0;
+      // Allocation runs on helper threads AHEAD of the copy loop: on some
+      // hosts hipMalloc commits physical pages at well below the PCIe
+      // rate, and a single thread's allocations then dominate the wake.
+      // alloc_threads_ workers fill slabs_ in order; the main thread
+      // issues each slab's H2D as soon as its pointer lands.
+      const auto ta0 = Clock::now();
+      const size_t n = slab_sizes_.size();
+      std::vector<void*> ptrs(n, nullptr);
+      std::atomic<size_t> next{0};
+      std::atomic<bool> failed{false};
+      auto alloc_worker = [&]() {
+        (void)hipSetDevice(device_);
+        for (;;) {
+          const size_t i = next.fetch_add(1);
+          if (i >= n) return;
+          void* p = nullptr;
+          if (hipMalloc(&p, slab_sizes_[i]) != hipSuccess) {
+            failed.store(true);
+            return;
+          }
+          // store after successful alloc; release ordering pairs with the
+          // acquire load in the copy loop below
+          __atomic_store_n(&ptrs[i], p, __ATOMIC_RELEASE);
+        }
+      };
+      const int nthreads = std::max(1, alloc_threads_);
+      std::vector<std::thread> workers;
+      for (int t = 1; t < nthreads; ++t) workers.emplace_back(alloc_worker);
+      // the calling thread participates too unless it must start copying
+      std::thread self_worker(alloc_worker);
       int c = 0;
-      for (size_t i = 0; i < slab_sizes_.size(); ++i) {
-        const auto ta = Clock::now();
-        void* p = nullptr;
-        FMA_HIP_CHECK(hipMalloc(&p, slab_sizes_[i]));  // overlaps prior copies
-        alloc_acc += seconds_since(ta);
+      double alloc_wait = 0.0;
+      for (size_t i = 0; i < n; ++i) {
+        const auto tw = Clock::now();
+        void* p;
+        while ((p = __atomic_load_n(&ptrs[i], __ATOMIC_ACQUIRE)) == nullptr) {
+          if (failed.load()) break;
+          std::this_thread::yield();
+        }
+        alloc_wait += seconds_since(tw);
+        if (p == nullptr) break;
         slabs_.push_back(p);
         const int64_t flat0 = slab_prefix_[i];
-        const int64_t len =
-            std::min<int64_t>(slab_sizes_[i], size_ - flat0);
+        const int64_t len = std::min<int64_t>(slab_sizes_[i], size_ - flat0);
         for (int64_t off = 0; off < len; off += chunk, ++c) {
           const int64_t sz = std::min<int64_t>(chunk, len - off);
           FMA_HIP_CHECK(hipMemcpyAsync(static_cast<unsigned char*>(p) + off,
@@ -652,7 +689,11 @@ class DeviceArena {
                                        ctx.copy_streams[c % ns]));
         }
       }
-      map_seconds_ = alloc_acc;
+      self_worker.join();
+      for (auto& t : workers) t.join();
+      TORCH_CHECK(!failed.load(), "hipMalloc failed during wake");
+      map_seconds_ = seconds_since(ta0);
+      alloc_wait_seconds_ = alloc_wait;
     }
     sync_pipeline(ctx);
     mapped_ = true;
@@ -767,6 +808,8 @@ class DeviceArena {
   bool mapped_ = false;
   int64_t generation_ = 0;
   double map_seconds_ = 0.0;
+  double alloc_wait_seconds_ = 0.0;
+  int alloc_threads_ = 4;
   void* vmm_base_ = nullptr;
   std::vector<int64_t> slab_sizes_;
   std::vector<int64_t> slab_prefix_;
@@ -818,5 +861,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def_property_readonly("device", &DeviceArena::device)
       .def_property_readonly("generation", &DeviceArena::generation)
       .def_property_readonly("num_slabs", &DeviceArena::num_slabs)
-      .def_property_readonly("last_map_seconds", &DeviceArena::last_map_seconds);
+      .def_property_readonly("last_map_seconds", &DeviceArena::last_map_seconds)
+      .def_property_readonly("last_alloc_wait_seconds",
+                             &DeviceArena::last_alloc_wait_seconds)
+      .def("set_alloc_threads", &DeviceArena::set_alloc_threads,
+           py::arg("n"));
 }

@@ -139,6 +139,8 @@ class ArenaActuator:
         self.chunk_bytes = chunk_bytes
         self.nstreams = nstreams
         self._gen_at_view: int = self._arena.generation
+        threads = int(os.environ.get("FMA_ALLOC_THREADS", "4"))
+        self._arena.set_alloc_threads(threads)
 
     @property
     def uses_vmm(self) -> bool:
@@ -151,6 +153,10 @@ class ArenaActuator:
     @property
     def last_map_seconds(self) -> float:
         return self._arena.last_map_seconds
+
+    @property
+    def last_alloc_wait_seconds(self) -> float:
+        return self._arena.last_alloc_wait_seconds
 
     def view(self, offset: int, shape: Tuple[int, ...], dtype: torch.dtype
              ) -> torch.Tensor:
