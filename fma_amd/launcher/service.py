@@ -70,6 +70,8 @@ class InstanceManager:
 
     def _on_instance_exit(self, inst: ServerInstance,
                           exit_code: Optional[int]) -> None:
+        if inst.instance_id not in self.instances:
+            return  # exit caused by deletion: DELETED already tells the story
         rev = self.broadcaster.next_revision()
         self.broadcaster.append("STOPPED", inst.instance_id, rev,
                                 {"exit_code": exit_code})

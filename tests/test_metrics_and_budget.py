@@ -1,0 +1,83 @@
+"""Metrics registry contents + accelerator-memory sleep budget."""
+
+import pytest
+
+from fma_amd.api import contracts as C
+from fma_amd.controller import metrics
+from fma_amd.controller.dualpods.controller import (ControllerConfig,
+                                                    DualPodsController)
+from fma_amd.store import objects as ob
+
+from tests.test_dualpods_controller import (FakeInstanceServer, drive,
+                                            infsvr_item, mk_world)
+
+
+def test_metric_names_match_reference():
+    # touching the factories registers them; names per reference
+    # controller.go:204-293 / metrics.go:79-95
+    metrics.actuation_seconds()
+    metrics.launcher_create_seconds()
+    metrics.http_latency_seconds()
+    metrics.requester_count()
+    metrics.isc_count()
+    metrics.duality()
+    metrics.launcher_pod_count()
+    if not metrics.HAVE_PROM:
+        pytest.skip("prometheus_client unavailable")
+    from prometheus_client import REGISTRY
+    names = {m.name for m in REGISTRY.collect()}
+    for want in ("fma_actuation_seconds", "fma_launcher_create_seconds",
+                 "fma_http_latency_seconds", "fma_requester_count",
+                 "fma_isc_count", "fma_duality", "fma_launcher_pod_count"):
+        assert want in names, f"{want} not registered"
+
+
+def test_actuation_histogram_observes_on_relay():
+    if not metrics.HAVE_PROM:
+        pytest.skip("prometheus_client unavailable")
+    from prometheus_client import REGISTRY
+    before = REGISTRY.get_sample_value(
+        "fma_actuation_seconds_count",
+        {"path": "hot", "instancesDeleted": "0", "isc_name": "isc1"}) or 0
+    w = mk_world()
+    drive(w["ctl"], infsvr_item(w["store"]))
+    after = REGISTRY.get_sample_value(
+        "fma_actuation_seconds_count",
+        {"path": "hot", "instancesDeleted": "0", "isc_name": "isc1"}) or 0
+    assert after == before + 1
+
+
+def test_accel_memory_budget_defers_wake():
+    """Wake waits while accelerator memory is above the sleeping budget
+    (reference accelMemoryIsLowEnough, inference-server.go:1991-2014)."""
+    w = mk_world()
+    w["ctl"].cfg.accelerator_sleeping_memory_limit_mib = 1  # 1 MiB budget
+
+    # stub reports 2 GiB in use on the GPU -> wake deferred
+    class BusyStub:
+        def __call__(self, method, path, json, params):
+            if path == "/v1/dual-pods/accelerators":
+                return 200, ["GPU-0"]
+            if path == "/v1/dual-pods/accelerator-memory-usage":
+                return 200, {"GPU-0": 2 << 30}
+            if path == "/v1/become-ready":
+                return 200, {}
+            return 404, {}
+
+    busy = BusyStub()
+    w["http"].register("10.0.0.1:8081", busy)
+    item = infsvr_item(w["store"])
+    for _ in range(5):
+        w["ctl"]._process(item)
+    assert w["inst_srv"].wakes == 0, "wake should wait for memory budget"
+
+    # memory drains -> wake proceeds
+    class FreeStub(BusyStub):
+        def __call__(self, method, path, json, params):
+            if path == "/v1/dual-pods/accelerator-memory-usage":
+                return 200, {"GPU-0": 0}
+            return super().__call__(method, path, json, params)
+
+    w["http"].register("10.0.0.1:8081", FreeStub())
+    drive(w["ctl"], item)
+    assert w["inst_srv"].wakes == 1
