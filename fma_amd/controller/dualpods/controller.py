@@ -553,8 +553,28 @@ class DualPodsController:
             except (Conflict, NotFound):
                 pass
 
+    def _configure_proxy(self, requester: Dict[str, Any],
+                         sdata: ServerData) -> None:
+        """Point the requester's TCP reverse proxy at the serving endpoint
+        so traffic addressed to the requester Pod reaches the provider
+        (the reference's release-0.7 feature; stub side
+        pkg/server/requester/proxy/server.go:39-217). 409 == already
+        configured, which is fine (configure-once semantics)."""
+        provider = self._find_provider_for(ob.uid_of(requester),
+                                           ob.name_of(requester))
+        if provider is None or not sdata.port:
+            return
+        ip = ob.pod_ip(provider)
+        if not ip:
+            return
+        self.http.request(
+            "PUT", self._stub_url(requester) + contracts.PROXY_CONFIG_PATH,
+            purpose="proxy-config",
+            json={"address": ip, "port": int(sdata.port)})
+
     def _relay_readiness(self, requester: Dict[str, Any], sdata: ServerData
                          ) -> bool:
+        self._configure_proxy(requester, sdata)
         r = self.http.request(
             "POST", self._stub_url(requester) + contracts.BECOME_READY_PATH,
             purpose="become-ready")
@@ -659,6 +679,7 @@ class DualPodsController:
         if not ip or not ob.pod_is_ready(provider):
             return RETRY
         port = _direct_server_port(provider)
+        sdata.port = port  # for the proxy configuration at relay time
         base = f"http://{ip}:{port}"
         sr = self.http.request("GET", base + contracts.IS_SLEEPING_PATH,
                                purpose="query-sleeping")

@@ -120,15 +120,21 @@ class NodeAgent:
         container. Containers without an explicit command are skipped
         (images do not run here — commands name python modules)."""
         containers = ob.pod_containers(pod)
+        # role priority: a provider's inference-server outranks the stub
+        # container it inherited from the requester spec
+        priority = (contracts.INFERENCE_SERVER_CONTAINER, "launcher",
+                    "requester", "main", "stub")
         main = None
-        for c in containers:
-            if c.get("name") in ("launcher", contracts.INFERENCE_SERVER_CONTAINER,
-                                 "requester", "main", "stub"):
-                main = c
+        for name in priority:
+            for c in containers:
+                if c.get("name") == name and c.get("command"):
+                    main = c
+                    break
+            if main:
                 break
-        if main is None and containers:
-            main = containers[0]
-        if not main or not main.get("command"):
+        if main is None:
+            main = next((c for c in containers if c.get("command")), None)
+        if not main:
             return None, None, None
         env = {e["name"]: str(e.get("value", ""))
                for e in main.get("env", []) if "name" in e}

@@ -90,6 +90,7 @@ class Stub:
         self.gpus = gpus
         self.ready_calls = 0
         self.unready_calls = 0
+        self.proxy_target = None
 
     def __call__(self, method, path, json, params):
         if path == "/v1/dual-pods/accelerators":
@@ -102,6 +103,11 @@ class Stub:
             return 200, {}
         if path == "/v1/dual-pods/accelerator-memory-usage":
             return 200, {g: 0 for g in self.gpus}
+        if path == "/v1/proxy/config" and method == "PUT":
+            if self.proxy_target is not None:
+                return 409, {}
+            self.proxy_target = json
+            return 200, {"status": "ok"}
         return 404, {}
 
 
@@ -197,6 +203,8 @@ def test_hot_start_binds_and_wakes():
     assert w["inst_srv"].wakes == 1
     assert not w["inst_srv"].sleeping
     assert w["stub"].ready_calls >= 1
+    # proxy pointed at the serving endpoint (release-0.7 feature)
+    assert w["stub"].proxy_target == {"address": "10.0.0.2", "port": 8000}
     req = w["store"].get("Pod", "req1")
     assert REQUESTER_FINALIZER in ob.finalizers_of(req)
     assert ob.labels_of(req)[C.DUAL_LABEL] == "launcher1"

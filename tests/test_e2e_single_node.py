@@ -181,3 +181,58 @@ def test_full_launcher_based_actuation(cluster):
     assert ob.annotations_of(lp)[C.INSTANCE_ID_ANNOTATION] == iid
     r = httpx.get(f"http://{lp_ip}:{ISC_PORT}/is_sleeping", timeout=5)
     assert r.json() == {"is_sleeping": False}
+
+
+DIRECT_PATCH = """
+spec:
+  containers:
+  - name: inference-server
+    command: ["{python}", "-m", "fma_amd.runtime.server",
+              "--model", "tiny", "--port", "8361"]
+    ports:
+    - containerPort: 8361
+"""
+
+
+def test_direct_path_actuation(cluster):
+    """Direct (launcher-less) provider: server-patch -> nominal Pod ->
+    node agent runs the serving runtime -> readiness relay + proxy config
+    (the reference's Milestone-2 flow, test/e2e/run.sh)."""
+    store, agent = cluster["store"], cluster["agent"]
+    cm = ob.new_object("ConfigMap", C.GPU_MAP_CONFIGMAP)
+    cm["data"] = {"node-a": '{"GPU-0": 0}'}
+    store.create(cm)
+
+    patch = DIRECT_PATCH.format(python=sys.executable)
+    pod = ob.new_object(
+        "Pod", "dreq1",
+        annotations={C.SERVER_PATCH_ANNOTATION: patch},
+        spec={"nodeName": "node-a", "containers": [
+            {"name": "requester",
+             "command": [sys.executable, "-m", "fma_amd.requester.server"]},
+            {"name": "inference-server"}]})
+    store.create(pod, actor="user")
+
+    wait_for(lambda: requester_ready(store, agent, "dreq1"), 90,
+             desc="direct requester ready")
+    provider = store.get("Pod", "dreq1-server")
+    assert ob.annotations_of(provider)[C.REQUESTER_ANNOTATION].endswith(
+        " dreq1")
+    prov_ip = provider["status"]["podIP"]
+    r = httpx.get(f"http://{prov_ip}:8361/is_sleeping", timeout=5)
+    assert r.json() == {"is_sleeping": False}
+
+    # proxy on the requester points at the provider's serving endpoint
+    stub_ip = agent.pods["dreq1"].ip
+    r = httpx.get(f"http://{stub_ip}:8081/v1/proxy/config", timeout=5)
+    assert r.status_code == 200
+    assert r.json() == {"address": prov_ip, "port": 8361}
+
+    # delete requester -> provider slept, kept as sleeper
+    store.delete("Pod", "dreq1", actor="user")
+    wait_for(lambda: store.try_get("Pod", "dreq1") is None, 60,
+             desc="direct requester gone")
+    provider = store.get("Pod", "dreq1-server")
+    assert ob.labels_of(provider)[C.SLEEPING_LABEL] == "true"
+    r = httpx.get(f"http://{prov_ip}:8361/is_sleeping", timeout=5)
+    assert r.json() == {"is_sleeping": True}
