@@ -60,6 +60,7 @@ class ServerData:
     readiness_relayed: bool = False
     deleted_instances: Set[str] = field(default_factory=set)
     instance_last_used: Dict[str, float] = field(default_factory=dict)
+    log_position: int = 0
 
 
 @dataclass
@@ -514,7 +515,36 @@ class DualPodsController:
                 return RETRY
         self._apply_bound_labels(provider, isc)
         sdata.instance_last_used[sdata.instance_id] = self.clock.time()
+        self._relay_instance_log(requester, provider, sdata)
         return self._relay_readiness(requester, sdata)
+
+    def _relay_instance_log(self, requester: Dict[str, Any],
+                            provider: Dict[str, Any],
+                            sdata: ServerData) -> None:
+        """Pipe new server-log bytes to the requester's /v1/set-log sink so
+        the requesting Pod surfaces its server's output (the consumer of
+        the SPI's SetLogPath contract, reference pkg/spi/interface.go:52-66;
+        position-deduplicated on the stub side, so re-relays are
+        harmless)."""
+        iid = sdata.instance_id
+        if not iid:
+            return
+        client = self._launcher_client(provider)
+        r = self.http.request(
+            "GET", f"{client.base}/v2/vllm/instances/{iid}/log",
+            purpose="log-fetch",
+            headers={"Range": f"bytes={sdata.log_position}-"})
+        if r.status not in (200, 206) or not isinstance(r.body, str) \
+                or not r.body:
+            return
+        chunk = r.body
+        sr = self.http.request(
+            "POST", self._stub_url(requester) + contracts.SET_LOG_PATH,
+            purpose="log-relay",
+            params={contracts.LOG_START_POS_PARAM: sdata.log_position},
+            content=chunk.encode("utf-8", "ignore"))
+        if sr.ok:
+            sdata.log_position += len(chunk.encode("utf-8", "ignore"))
 
     def _accel_memory_low_enough(self, requester: Dict[str, Any],
                                  sdata: ServerData) -> bool:

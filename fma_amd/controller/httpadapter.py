@@ -44,14 +44,16 @@ class HttpAdapter:
         self._observe = observe
 
     def request(self, method: str, url: str, *, purpose: str = "",
-                json: Any = None, params: Optional[Dict[str, Any]] = None
-                ) -> HttpResult:
+                json: Any = None, params: Optional[Dict[str, Any]] = None,
+                headers: Optional[Dict[str, str]] = None,
+                content: Optional[bytes] = None) -> HttpResult:
         start = time.time()
         logger.debug("http call", extra={"httpCallStartTime": start,
                                          "method": method, "url": url,
                                          "purpose": purpose})
         try:
-            r = self._client.request(method, url, json=json, params=params)
+            r = self._client.request(method, url, json=json, params=params,
+                                     headers=headers, content=content)
             status = r.status_code
             try:
                 body = r.json()
@@ -80,8 +82,9 @@ class FakeHttp:
         self.routes[hostport] = handler
 
     def request(self, method: str, url: str, *, purpose: str = "",
-                json: Any = None, params: Optional[Dict[str, Any]] = None
-                ) -> HttpResult:
+                json: Any = None, params: Optional[Dict[str, Any]] = None,
+                headers: Optional[Dict[str, str]] = None,
+                content: Optional[bytes] = None) -> HttpResult:
         self.calls.append((method, url, purpose))
         assert url.startswith("http://"), url
         rest = url[len("http://"):]
@@ -91,9 +94,14 @@ class FakeHttp:
         if handler is None:
             return HttpResult(0, f"connection refused: {hostport}")
         if callable(handler) and not hasattr(handler, "request"):
-            status, body = handler(method, path, json, params)
+            # 4-arg fake handlers predate headers/content; pass content as
+            # json payload for simplicity
+            status, body = handler(method, path,
+                                   json if content is None else content,
+                                   params)
             return HttpResult(status, body)
-        r = handler.request(method, path, json=json, params=params)
+        r = handler.request(method, path, json=json, params=params,
+                            headers=headers, content=content)
         try:
             body = r.json()
         except Exception:

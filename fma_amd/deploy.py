@@ -60,6 +60,23 @@ def main() -> None:
 
     agent = NodeAgent(store, args.node_name, node_index=1)
     agent.start()
+
+    # mini-scheduler: in the single-node stack, Pods created without a
+    # nodeName are bound to this node (kube-scheduler's role)
+    stop = threading.Event()
+
+    def schedule_loop():
+        for ev in store.watch(kinds=["Pod"], stop=stop):
+            if ev.type in ("ADDED", "MODIFIED"):
+                pod = store.try_get("Pod", ev.obj["metadata"]["name"])
+                if pod is not None and not pod.get("spec", {}).get("nodeName"):
+                    pod.setdefault("spec", {})["nodeName"] = args.node_name
+                    try:
+                        store.update(pod, actor="system")
+                    except Exception:
+                        pass
+
+    threading.Thread(target=schedule_loop, daemon=True).start()
     ctl = DualPodsController(
         store, HttpAdapter(observe=metrics.observe_http),
         ControllerConfig(sleeper_limit=args.sleeper_limit))
@@ -76,6 +93,7 @@ def main() -> None:
         uvicorn.run(app, host="127.0.0.1", port=args.store_port,
                     log_level="warning")
     finally:
+        stop.set()
         ctl.stop()
         pop.stop()
         agent.stop()

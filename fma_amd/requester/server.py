@@ -79,7 +79,13 @@ class LogSink:
                 raise ValueError(f"startPos {start_pos} beyond size {size}")
             new_from = size - start_pos
             if new_from < len(chunk):
-                self.buf.extend(chunk[new_from:])
+                fresh = chunk[new_from:]
+                self.buf.extend(fresh)
+                # surface relayed server output on the requester's own
+                # stdout so the requesting Pod's log shows its server
+                import sys
+                sys.stdout.write(fresh.decode("utf-8", "replace"))
+                sys.stdout.flush()
 
     def contents(self) -> bytes:
         with self.lock:

@@ -161,6 +161,12 @@ def test_full_launcher_based_actuation(cluster):
     wait_for(lambda: C.INSTANCE_SIGNATURE_ANNOTATION in ob.annotations_of(
         store.get("Pod", ob.name_of(lp))), 20, desc="notifier signature")
 
+    # server log relayed to the requester's stdout via /v1/set-log
+    req_log = os.path.join(agent.log_dir, "pod-req1.log")
+    wait_for(lambda: os.path.exists(req_log) and
+             "[kickoff]" in open(req_log, errors="replace").read(), 30,
+             desc="server log relayed to requester")
+
     # delete the requester: unbind + sleep, launcher survives
     store.delete("Pod", "req1", actor="user")
     wait_for(lambda: store.try_get("Pod", "req1") is None, 60,
@@ -236,3 +242,35 @@ def test_direct_path_actuation(cluster):
     assert ob.labels_of(provider)[C.SLEEPING_LABEL] == "true"
     r = httpx.get(f"http://{prov_ip}:8361/is_sleeping", timeout=5)
     assert r.json() == {"is_sleeping": True}
+
+
+def test_controller_restart_recovers_live_binding(cluster):
+    """Kill the controller mid-flight and start a fresh one: the binding
+    and serving state recover purely from Pod metadata
+    (reference test-cases.sh:720)."""
+    store, agent = cluster["store"], cluster["agent"]
+    mk_isc_lc_lpp(store)
+    lp = wait_for(lambda: launcher_pod(store), 30, desc="launcher pod")
+    wait_for(lambda: ob.pod_is_ready(store.get("Pod", ob.name_of(lp))),
+             60, desc="launcher Ready")
+    mk_requester(store, "req1")
+    wait_for(lambda: requester_ready(store, agent, "req1"), 90,
+             desc="requester ready")
+
+    cluster["ctl"].stop()
+    ctl2 = DualPodsController(store, HttpAdapter(), ControllerConfig())
+    ctl2.start()
+    try:
+        # the recovered controller can still unbind cleanly...
+        store.delete("Pod", "req1", actor="user")
+        wait_for(lambda: store.try_get("Pod", "req1") is None, 60,
+                 desc="requester gone via recovered controller")
+        lp2 = store.get("Pod", ob.name_of(lp))
+        assert C.REQUESTER_ANNOTATION not in ob.annotations_of(lp2)
+        assert ob.labels_of(lp2)[C.SLEEPING_LABEL] == "true"
+        # ...and hot-start the next requester
+        mk_requester(store, "req9")
+        wait_for(lambda: requester_ready(store, agent, "req9"), 60,
+                 desc="req9 hot start via recovered controller")
+    finally:
+        ctl2.stop()
