@@ -217,7 +217,7 @@ class NodeAgent:
             return False
 
     def _status_loop(self) -> None:
-        while not self._stop.wait(0.5):
+        while not self._stop.wait(0.15):
             for name, pp in list(self.pods.items()):
                 exited = pp.proc.poll() is not None
                 ready = False if exited else self._probe_ready(pp)
