@@ -45,6 +45,9 @@ PROVIDER_FINALIZER = "dual-pods.llm-d.ai/provider-protection"
 
 RETRY = True
 DONE = False
+#: fixed re-queue delay for known wait states (server booting, pod
+#: starting) — the reference's retryAfter (inference-server.go:512)
+WAIT = 1.0
 
 
 @dataclass
@@ -266,7 +269,7 @@ class DualPodsController:
         if sdata.gpus is None:
             gpus = self._query_gpus(requester)
             if gpus is None:
-                return RETRY
+                return WAIT  # stub not answering yet
             sdata.gpus = gpus
         metrics.requester_count().set(sum(
             1 for p in self.store.list("Pod", self.ns)
@@ -379,11 +382,11 @@ class DualPodsController:
             if r.ok or r.status == 404:
                 sdata.deleted_instances.add(iid)
         if sel.retry:
-            return RETRY
+            return WAIT  # launchers not ready / repairs pending
         if sel.launcher is None:
             created = self._create_launcher(node, isc)
             sdata.needed_new_launcher = True
-            return RETRY if created else DONE
+            return WAIT if created else DONE
         if not sel.has_sleeping_instance:
             sdata.needed_new_instance = True
         return self._bind(requester, sel.launcher.pod, sdata, isc)
@@ -486,9 +489,9 @@ class DualPodsController:
             if cr.status not in (201, 409):
                 return RETRY
             sdata.needed_new_instance = True
-            return RETRY
+            return WAIT  # instance process is booting
         if not r.ok:
-            return RETRY
+            return WAIT
         if r.body.get("status") == contracts.INSTANCE_STATUS_STOPPED:
             # bound instance died: delete the requester so its owner
             # re-creates it (reference :454-507)
@@ -505,7 +508,7 @@ class DualPodsController:
         sr = self.http.request("GET", base + contracts.IS_SLEEPING_PATH,
                                purpose="query-sleeping")
         if not sr.ok:
-            return RETRY  # server still starting
+            return WAIT  # server still starting
         if sr.body.get("is_sleeping"):
             if not self._accel_memory_low_enough(requester, sdata):
                 return RETRY
@@ -707,7 +710,7 @@ class DualPodsController:
                           ) -> bool:
         ip = ob.pod_ip(provider)
         if not ip or not ob.pod_is_ready(provider):
-            return RETRY
+            return WAIT  # provider pod still starting
         port = _direct_server_port(provider)
         sdata.port = port  # for the proxy configuration at relay time
         base = f"http://{ip}:{port}"

@@ -100,7 +100,12 @@ class RateLimitingQueue(Generic[T]):
 
 
 class QueueAndWorkers(Generic[T]):
-    """Queue + N worker threads running `process(item) -> bool retry`."""
+    """Queue + N worker threads running `process(item)`.
+
+    process returns False (done), True (retry with exponential backoff) or
+    a float (re-queue after exactly that many seconds — the reference's
+    processResult.retryAfter for known wait states, e.g. a server that is
+    still booting; inference-server.go:448-452, :512)."""
 
     def __init__(self, name: str, num_workers: int,
                  process: Callable[[T], bool]):
@@ -128,7 +133,11 @@ class QueueAndWorkers(Generic[T]):
                 import traceback
                 traceback.print_exc()
                 retry = True
-            if retry:
+            if isinstance(retry, (int, float)) and not isinstance(retry, bool) \
+                    and retry > 0:
+                self.queue.forget(item)  # a scheduled wait is not a failure
+                self.queue.add_after(item, float(retry))
+            elif retry:
                 self.queue.add_rate_limited(item)
             else:
                 self.queue.forget(item)
