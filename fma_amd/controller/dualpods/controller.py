@@ -45,9 +45,10 @@ PROVIDER_FINALIZER = "dual-pods.llm-d.ai/provider-protection"
 
 RETRY = True
 DONE = False
-#: fixed re-queue delay for known wait states (server booting, pod
-#: starting) — the reference's retryAfter (inference-server.go:512)
-WAIT = 1.0
+#: wait states (server booting, pod starting) re-queue through the same
+#: exponential backoff, but the dual-pods queue caps it at 2 s (the
+#: reference caps at 20 s, which shows up directly in cold T_actuation)
+WAIT = True
 
 
 @dataclass
@@ -88,7 +89,7 @@ class DualPodsController:
         self._lock = threading.Lock()
         self._stop = threading.Event()
         self.workers = QueueAndWorkers("dualpods", self.cfg.num_workers,
-                                       self._process)
+                                       self._process, max_backoff=2.0)
         self._watch_thread: Optional[threading.Thread] = None
 
     # ------------------------------------------------------------------

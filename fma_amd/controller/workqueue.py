@@ -27,7 +27,8 @@ class RateLimitingQueue(Generic[T]):
     """Subset of client-go's workqueue: Add / AddAfter / AddRateLimited /
     Forget / Get / Done / ShutDown with dirty/processing sets."""
 
-    def __init__(self) -> None:
+    def __init__(self, max_backoff: float = MAX_BACKOFF_SECONDS) -> None:
+        self.max_backoff = max_backoff
         self._cond = threading.Condition()
         self._queue: List[T] = []
         self._dirty: set = set()
@@ -63,7 +64,7 @@ class RateLimitingQueue(Generic[T]):
             n = self._failures.get(item, 0)
             self._failures[item] = n + 1
         self.add_after(item, min(BASE_BACKOFF_SECONDS * (2 ** n),
-                                 MAX_BACKOFF_SECONDS))
+                                 self.max_backoff))
 
     def forget(self, item: T) -> None:
         with self._cond:
@@ -108,9 +109,10 @@ class QueueAndWorkers(Generic[T]):
     still booting; inference-server.go:448-452, :512)."""
 
     def __init__(self, name: str, num_workers: int,
-                 process: Callable[[T], bool]):
+                 process: Callable[[T], bool],
+                 max_backoff: float = MAX_BACKOFF_SECONDS):
         self.name = name
-        self.queue: RateLimitingQueue[T] = RateLimitingQueue()
+        self.queue: RateLimitingQueue[T] = RateLimitingQueue(max_backoff)
         self.num_workers = num_workers
         self.process = process
         self.threads: List[threading.Thread] = []

@@ -125,6 +125,9 @@ class ServingRuntime:
             loader.load_into_params(checkpoint, engine.params)
         self.rt = tp.TPRuntime(ctx, engine) if world > 1 else engine
         self.model_name = args.served_model_name or args.model
+        print(f"[serve] engine up in {engine.create_seconds:.2f}s "
+              f"({engine.total_bytes/2**30:.2f} GiB, "
+              f"pid={os.getpid()}, world={world})", flush=True)
         if args.start_asleep:
             self.rt.sleep(1)
         self.ready = True
@@ -199,8 +202,11 @@ def main(argv: Optional[List[str]] = None) -> None:
 
     import uvicorn
 
+    t0 = time.time()
     args = parse_options(" ".join(argv if argv is not None else sys.argv[1:]))
+    print(f"[serve] parsing done at +{time.time()-t0:.2f}s", flush=True)
     runtime = ServingRuntime(args)
+    print(f"[serve] runtime ready at +{time.time()-t0:.2f}s", flush=True)
     app = create_app(runtime)
     try:
         uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
