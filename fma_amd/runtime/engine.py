@@ -54,6 +54,7 @@ class ActuationEngine:
             slab_bytes = actuation.DEFAULT_SLAB_BYTES
         self.actuation_mode = actuation_mode
         t0 = time.perf_counter()
+        self.timing = {}
         if actuation_mode == "pack":
             # scattered-tensor mode: parameters live in ordinary caching-
             # allocator storage; sleep/wake goes through the HIP
@@ -76,11 +77,18 @@ class ActuationEngine:
                                               nstreams=nstreams)
             self.packer = None
             self.params = self._make_views()
+        self.timing["arena_s"] = time.perf_counter() - t0
+        t1 = time.perf_counter()
         self.model = LlamaModel(cfg, self.params, self.device,
                                 tp_rank, tp_size, tp_group)
         if init_weights:
             self.model.init_weights(seed)
+        if self.on_gpu:
+            torch.cuda.synchronize(self.device)
+        self.timing["init_s"] = time.perf_counter() - t1
+        t2 = time.perf_counter()
         self.host = actuation.alloc_pinned(self.total_bytes)
+        self.timing["pin_s"] = time.perf_counter() - t2
         if self.on_gpu:
             torch.cuda.synchronize(self.device)
         self.create_seconds = time.perf_counter() - t0

@@ -127,7 +127,8 @@ class ServingRuntime:
         self.model_name = args.served_model_name or args.model
         print(f"[serve] engine up in {engine.create_seconds:.2f}s "
               f"({engine.total_bytes/2**30:.2f} GiB, "
-              f"pid={os.getpid()}, world={world})", flush=True)
+              f"pid={os.getpid()}, world={world}, "
+              f"timing={getattr(engine, 'timing', {})})", flush=True)
         if args.start_asleep:
             self.rt.sleep(1)
         self.ready = True
