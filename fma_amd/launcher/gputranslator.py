@@ -79,9 +79,9 @@ class GpuTranslator:
                     os.environ.get("FMA_GPU_MAP_FILE"):
                 mode = self.MODE_GPU_MAP
             else:
-                import torch
-                mode = self.MODE_REAL if torch.cuda.is_available() \
-                    else self.MODE_NAIVE
+                # must not initialize HIP (the launcher forks GPU children)
+                from fma_amd.utils.gpus import gpu_present
+                mode = self.MODE_REAL if gpu_present() else self.MODE_NAIVE
         self.mode = mode
         self._map = gpu_map or self._load_map()
 
@@ -90,10 +90,10 @@ class GpuTranslator:
             m = _amdsmi_map() or _rocm_smi_map()
             if m:
                 return m
-            # GPUs are visible to torch but no SMI identity source: fall
-            # back to positional ids (still unique on one node)
-            import torch
-            n = torch.cuda.device_count()
+            # no SMI identity source: positional ids (no HIP init — the
+            # launcher forks GPU children)
+            from fma_amd.utils.gpus import gpu_count
+            n = gpu_count()
             if n == 0:
                 raise GpuTranslationError("real mode but no GPUs visible")
             return {f"GPU-{i}": i for i in range(n)}

@@ -297,8 +297,12 @@ def main() -> None:
     import torch  # noqa: F401
     import fma_amd.runtime.engine  # noqa: F401
     import fma_amd.runtime.server  # noqa: F401
-    if torch.cuda.is_available():
-        import fma_amd._C  # noqa: F401  HIP actuator must be present
+    # CRITICAL: do NOT touch torch.cuda here — the launcher forks serving
+    # instances, and a child forked from a HIP-initialized parent launches
+    # kernels ~1000x slower (measured: 108 s weight init for 15 GiB).
+    from fma_amd.utils.gpus import gpu_present
+    if gpu_present():
+        import fma_amd._C  # noqa: F401  HIP actuator must be importable
 
     import os as _os
     host = args.host or _os.environ.get("FMA_BIND_HOST", "0.0.0.0")
