@@ -360,7 +360,6 @@ class DualPodsController:
                         if inst["instance_id"] in sdata.deleted_instances:
                             continue
                         port = _instance_port(inst)
-                        views.append  # keep lint quiet
                         lv.instances.append(InstanceView(
                             instance_id=inst["instance_id"],
                             status=inst.get("status", "stopped"),
@@ -482,6 +481,7 @@ class DualPodsController:
     def _run_bound_launcher(self, node: str, requester: Dict[str, Any],
                             provider: Dict[str, Any], sdata: ServerData,
                             isc: Dict[str, Any]) -> bool:
+        self._warn_outdated_routing(provider, isc)
         client = self._launcher_client(provider)
         r = client.get_instance(sdata.instance_id)
         if r.status == 404:
@@ -549,6 +549,28 @@ class DualPodsController:
             content=chunk.encode("utf-8", "ignore"))
         if sr.ok:
             sdata.log_position += len(chunk.encode("utf-8", "ignore"))
+
+    def _warn_outdated_routing(self, provider: Dict[str, Any],
+                               isc: Dict[str, Any]) -> None:
+        """Routing metadata is frozen at bind time; if the ISC changed
+        since, surface it as an Event (reference docs/dual-pods.md:687-717:
+        OutdatedRoutingMetadata)."""
+        import json
+        stored = ob.annotations_of(provider).get(
+            contracts.ISC_ROUTING_METADATA_ANNOTATION)
+        msc_now = isc["spec"]["modelServerConfig"]
+        fresh = json.dumps({"labels": msc_now.get("labels", {}),
+                            "annotations": msc_now.get("annotations", {})},
+                           sort_keys=True)
+        if stored is not None and stored != fresh:
+            from fma_amd.controller.events import (REASON_OUTDATED_ROUTING,
+                                                   record_event)
+            record_event(
+                self.store, provider, REASON_OUTDATED_ROUTING,
+                f"bound provider {ob.name_of(provider)} carries routing "
+                f"metadata older than InferenceServerConfig "
+                f"{ob.name_of(isc)}; it will refresh at next bind",
+                actor="dual-pods-controller", namespace=self.ns)
 
     def _accel_memory_low_enough(self, requester: Dict[str, Any],
                                  sdata: ServerData) -> bool:
@@ -934,7 +956,7 @@ class DualPodsController:
                         "POST", f"http://{ip}:{port}" + contracts.SLEEP_PATH,
                         purpose="sleep", params={"level": 1})
                     any_awake = True
-        want = "false" if any_awake else "true"
+        # every instance on an unbound launcher is (now) sleeping
         if ob.labels_of(pod).get(contracts.SLEEPING_LABEL) != "true":
             cur = self.store.try_get("Pod", pod_name, self.ns)
             if cur is not None:
@@ -943,7 +965,6 @@ class DualPodsController:
                     self.store.update(cur, actor="dual-pods-controller")
                 except (Conflict, NotFound):
                     pass
-        _ = want
         return DONE
 
     def reconcile_isc_gc(self, node: str, isc_name: str) -> bool:

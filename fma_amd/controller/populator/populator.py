@@ -420,6 +420,12 @@ class LauncherPopulator:
                         self.store.update(cur, actor="launcher-populator")
                     except (Conflict, NotFound):
                         pass
+                    from fma_amd.controller.events import (
+                        REASON_LAUNCHER_STUCK, record_event)
+                    record_event(
+                        self.store, pod, REASON_LAUNCHER_STUCK,
+                        f"launcher Pod {ob.name_of(pod)} is {phase}",
+                        actor="launcher-populator", namespace=self.ns)
         for phase in ("bound", "unbound", "stale", "stuck_scheduling",
                       "stuck_starting"):
             metrics.launcher_pod_count().labels(lc_name, phase).set(

@@ -6,12 +6,12 @@
 // inference-server.go:1497,1712). Here the actual tensor movement between
 // GPU HBM3E and pinned host DRAM is implemented directly:
 //
-// - DeviceArena: one contiguous device allocation holding every parameter
-//   of a model instance. Preferred backing is HIP virtual memory management
-//   (hipMemAddressReserve + hipMemCreate/hipMemMap): the virtual address is
-//   reserved once and survives sleep, so parameter tensors (views into the
-//   arena) stay valid across sleep/wake while the physical HBM is truly
-//   released while asleep. Fallback is plain hipMalloc with re-binding.
+// - DeviceArena: the parameter store of one model instance. Default
+//   backing is a SLABBED hipMalloc arena (layout planned so no tensor
+//   straddles a slab): wake overlaps threaded slab allocation with the
+//   H2D stream, sleep truly frees HBM. HIP VMM constant-VA backing is
+//   implemented but opt-in (FMA_TRY_VMM=1): ROCm 7.2 serves stale SDMA
+//   reads after unmap/remap (see tools/debug_arena.py).
 // - pack_to_host / restore_from_host: for models whose tensors live in
 //   scattered allocations, a descriptor-table gather/scatter HIP kernel
 //   (kernels.hip) coalesces shards chunk-by-chunk through device staging

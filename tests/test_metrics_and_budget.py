@@ -81,3 +81,34 @@ def test_accel_memory_budget_defers_wake():
     w["http"].register("10.0.0.1:8081", FreeStub())
     drive(w["ctl"], item)
     assert w["inst_srv"].wakes == 1
+
+
+def test_launcher_stuck_event_recorded():
+    from tests.test_populator import (FakeClock, drain_key, launcher_pods,
+                                      mk_lc, mk_lpp, mk_node, mk_pop)
+    from fma_amd.store.memstore import MemStore as MS
+    store = MS()
+    clock = FakeClock(2000.0)
+    mk_node(store, "n1", labels={"gpu": "x"})
+    mk_lc(store)
+    mk_lpp(store, "p1", count=1, match_labels={"gpu": "x"})
+    pop = mk_pop(store, clock)
+    drain_key(pop, ("n1", "lc1"))
+    pod = launcher_pods(store, "n1")[0]
+    created = ob.meta(store.get("Pod", ob.name_of(pod)))["creationTimestamp"]
+    clock.t = created + 500
+    drain_key(pop, ("n1", "lc1"))
+    events = store.list("Event")
+    assert any(e["reason"] == "LauncherStuck" for e in events)
+
+
+def test_outdated_routing_event_recorded():
+    w = mk_world()
+    drive(w["ctl"], infsvr_item(w["store"]))
+    isc = w["store"].get("InferenceServerConfig", "isc1")
+    isc["spec"]["modelServerConfig"]["labels"] = {"llm-d.ai/model": "v2"}
+    w["store"].update(isc)
+    # re-reconcile the (still bound) server against the changed ISC
+    w["ctl"]._process(infsvr_item(w["store"]))
+    events = w["store"].list("Event")
+    assert any(e["reason"] == "OutdatedRoutingMetadata" for e in events)
