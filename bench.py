@@ -42,7 +42,7 @@ def main() -> None:
     ap.add_argument("--gib", type=float, default=64.0,
                     help="total parameter GiB across all ranks")
     ap.add_argument("--mode", choices=["arena", "pack"], default="arena")
-    ap.add_argument("--pack-xfer", type=int, default=0,
+    ap.add_argument("--pack-xfer", type=int, default=None,
                     help="pack transfer mode: 0 staged, 1 direct-kernel, 2 per-tensor")
     ap.add_argument("--chunk-mb", type=int, default=0)
     ap.add_argument("--nstreams", type=int, default=1)

@@ -37,7 +37,7 @@ class ActuationEngine:
                  use_vmm=None, chunk_bytes: int = 0, seed: int = 0,
                  init_weights: bool = True, nstreams: int = 1,
                  slab_bytes: int = None, actuation_mode: str = "arena",
-                 pack_xfer_mode: int = 0):
+                 pack_xfer_mode=None):
         self.cfg = cfg
         self.device_index = device_index
         self.tp_rank = tp_rank
