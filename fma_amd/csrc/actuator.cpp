@@ -588,6 +588,32 @@ class DeviceArena {
     return at::from_blob(resolve(offset, bytes), sizes, options);
   }
 
+  // Bulk host->device refill of a MAPPED arena (checkpoint loading: the
+  // caller fills the pinned buffer on CPU threads, then this streams it
+  // into the slabs at PCIe rate). No allocation, no state change.
+  double load_from(at::Tensor host, int64_t chunk_bytes, int64_t nstreams) {
+    TORCH_CHECK(mapped_, "arena must be awake to load into");
+    check_host_buffer(host, size_);
+    auto& ctx = ctx_for(device_);
+    FMA_HIP_CHECK(hipSetDevice(device_));
+    const int64_t chunk = chunk_or_default(chunk_bytes);
+    const int ns = clamp_streams(nstreams);
+    auto* host_ptr = static_cast<unsigned char*>(host.data_ptr());
+    const auto t0 = Clock::now();
+    join_torch_stream(ctx);
+    int c = 0;
+    for_each_span([&](unsigned char* dev, int64_t flat, int64_t len) {
+      for (int64_t off = 0; off < len; off += chunk, ++c) {
+        const int64_t sz = std::min<int64_t>(chunk, len - off);
+        FMA_HIP_CHECK(hipMemcpyAsync(dev + off, host_ptr + flat + off, sz,
+                                     hipMemcpyHostToDevice,
+                                     ctx.copy_streams[c % ns]));
+      }
+    });
+    sync_pipeline(ctx);
+    return seconds_since(t0);
+  }
+
   // sleep(level=1): D2H into pinned host DRAM, then release physical HBM.
   double sleep_to(at::Tensor host, int64_t chunk_bytes, int64_t nstreams) {
     TORCH_CHECK(mapped_, "arena already asleep");
@@ -851,6 +877,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("view", &DeviceArena::view, py::arg("offset"), py::arg("sizes"),
            py::arg("dtype"))
       .def("sleep_to", &DeviceArena::sleep_to, py::arg("host"),
+           py::arg("chunk_bytes") = 0, py::arg("nstreams") = 2)
+      .def("load_from", &DeviceArena::load_from, py::arg("host"),
            py::arg("chunk_bytes") = 0, py::arg("nstreams") = 2)
       .def("wake_from", &DeviceArena::wake_from, py::arg("host"),
            py::arg("chunk_bytes") = 0, py::arg("nstreams") = 2)

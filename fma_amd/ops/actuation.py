@@ -165,6 +165,10 @@ class ArenaActuator:
     def sleep(self, host: torch.Tensor) -> float:
         return self._arena.sleep_to(host, self.chunk_bytes, self.nstreams)
 
+    def load_from(self, host: torch.Tensor) -> float:
+        """Refill a mapped arena from the pinned buffer (checkpoint load)."""
+        return self._arena.load_from(host, self.chunk_bytes, self.nstreams)
+
     def wake(self, host: torch.Tensor) -> Tuple[float, bool]:
         """Returns (seconds, views_invalidated). With VMM backing the VA is
         constant and views survive; otherwise the caller must re-bind."""
@@ -212,6 +216,11 @@ class FakeArenaActuator:
         host[: self.nbytes].copy_(self._buf)
         self._buf.zero_()  # poison: reads while asleep are wrong by design
         self._asleep = True
+        return 1e-9
+
+    def load_from(self, host: torch.Tensor) -> float:
+        assert not self._asleep, "arena must be awake to load into"
+        self._buf.copy_(host[: self.nbytes])
         return 1e-9
 
     def wake(self, host: torch.Tensor) -> Tuple[float, bool]:

@@ -153,6 +153,49 @@ class ActuationEngine:
         self.last_wake_seconds = t
         return t
 
+    def load_checkpoint(self, path: str, cpu_threads: int = 8) -> float:
+        """Fast checkpoint load: safetensors -> pinned host buffer (parallel
+        CPU copies from the mmap) -> one pipelined H2D into the arena.
+        Beats tensor-by-tensor pageable copies (~10-20 GB/s) by staging
+        through the pinned buffer at the PCIe rate.
+        """
+        import concurrent.futures
+        import time as _t
+
+        if self.actuation_mode != "arena" or self.arena is None:
+            from fma_amd.models import loader
+            t0 = _t.perf_counter()
+            loader.load_into_params(path, self.params)
+            return _t.perf_counter() - t0
+
+        t0 = _t.perf_counter()
+        from fma_amd.models import loader
+        host_np = self.host.numpy()
+
+        def stage(item):
+            name, tensor = item
+            if name not in self.layout:
+                raise KeyError(f"checkpoint tensor {name!r} unknown")
+            off, shape, dtype = self.layout[name]
+            if tuple(tensor.shape) != tuple(shape):
+                raise ValueError(f"shape mismatch for {name}")
+            raw = tensor.to(dtype).contiguous().view(torch.uint8).view(-1)
+            host_np[off:off + raw.numel()] = raw.numpy()
+            return name
+
+        seen = set()
+        with concurrent.futures.ThreadPoolExecutor(cpu_threads) as ex:
+            for name in ex.map(stage, loader.iter_safetensors(path)):
+                seen.add(name)
+        missing = set(self.layout) - seen
+        if missing:
+            raise KeyError(f"checkpoint missing parameters: "
+                           f"{sorted(missing)[:5]}...")
+        self.arena.load_from(self.host)
+        if self.on_gpu:
+            torch.cuda.synchronize(self.device)
+        return _t.perf_counter() - t0
+
     # -- serving -------------------------------------------------------------
 
     @torch.no_grad()

@@ -121,8 +121,9 @@ class ServingRuntime:
                                  tp_group=ctx.device_group, seed=args.seed,
                                  init_weights=checkpoint is None)
         if checkpoint is not None:
-            from fma_amd.models import loader
-            loader.load_into_params(checkpoint, engine.params)
+            t_load = engine.load_checkpoint(checkpoint)
+            print(f"[serve] checkpoint {checkpoint} loaded in {t_load:.2f}s",
+                  flush=True)
         self.rt = tp.TPRuntime(ctx, engine) if world > 1 else engine
         self.model_name = args.served_model_name or args.model
         print(f"[serve] engine up in {engine.create_seconds:.2f}s "

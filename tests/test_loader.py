@@ -53,3 +53,31 @@ def test_server_loads_checkpoint_dir(tmp_path):
     toks = torch.randint(0, src.cfg.vocab_size, (1, 4))
     assert torch.equal(src.model.forward(toks),
                        rt.rt.model.forward(toks))
+
+
+def test_fast_load_checkpoint_into_arena(tmp_path):
+    src = ActuationEngine(LlamaConfig.tiny(), seed=31)
+    ckpt = str(tmp_path / "fast-ckpt")
+    loader.save_params(src.params, ckpt, src.cfg)
+    dst = ActuationEngine(LlamaConfig.tiny(), seed=77, init_weights=False)
+    t = dst.load_checkpoint(ckpt)
+    assert t >= 0
+    toks = torch.randint(0, src.cfg.vocab_size, (1, 6))
+    assert torch.equal(src.model.forward(toks), dst.model.forward(toks))
+    # still sleep/wake clean afterwards
+    before = dst.model.forward(toks).clone()
+    dst.sleep()
+    dst.wake_up()
+    assert torch.equal(before, dst.model.forward(toks))
+
+
+def test_fast_load_missing_param_raises(tmp_path):
+    src = ActuationEngine(LlamaConfig.tiny(), seed=1)
+    ckpt = str(tmp_path / "partial")
+    params = dict(src.params)
+    params.pop("final_norm.weight")
+    loader.save_params(params, ckpt, src.cfg)
+    dst = ActuationEngine(LlamaConfig.tiny(), seed=2, init_weights=False)
+    import pytest as _pytest
+    with _pytest.raises(KeyError):
+        dst.load_checkpoint(ckpt)
