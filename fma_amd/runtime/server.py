@@ -112,6 +112,12 @@ class ServingRuntime:
             from fma_amd.models import loader
             checkpoint = args.model
             cfg = loader.config_from_dir(checkpoint) or LlamaConfig.tiny()
+            if world > 1:
+                raise NotImplementedError(
+                    "checkpoint loading with --tensor-parallel-size > 1 "
+                    "needs shard-aware slicing (checkpoints store unsharded "
+                    "tensors); load a TP-sharded checkpoint per rank or use "
+                    "random-init models for TP instances")
         else:
             cfg = LlamaConfig.by_name(args.model)
         if args.max_model_len:

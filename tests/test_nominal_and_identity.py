@@ -1,0 +1,74 @@
+"""Strategic merge, template rendering, identity determinism units."""
+
+import pytest
+
+from fma_amd.controller.dualpods import nominal
+from fma_amd.controller.dualpods.identity import (instance_id, nominal_hash,
+                                                  template_hash)
+
+
+def test_render_template_substitutes_provider_data():
+    out = nominal.render_template(
+        "vol: {{.LocalVolume}} on {{ .NodeName }}",
+        {"NodeName": "n1", "LocalVolume": "pvc-a"})
+    assert out == "vol: pvc-a on n1"
+
+
+def test_render_template_unknown_field_raises():
+    with pytest.raises(nominal.NominalError):
+        nominal.render_template("{{.Nope}}", {"NodeName": "n1"})
+
+
+def test_strategic_merge_named_lists_merge_by_name():
+    base = {"containers": [
+        {"name": "a", "image": "a:1", "env": [{"name": "X", "value": "1"}]},
+        {"name": "b", "image": "b:1"}]}
+    patch = {"containers": [
+        {"name": "a", "image": "a:2"},
+        {"name": "c", "image": "c:1"}]}
+    out = nominal.strategic_merge(base, patch)
+    by_name = {c["name"]: c for c in out["containers"]}
+    assert by_name["a"]["image"] == "a:2"
+    assert by_name["a"]["env"] == [{"name": "X", "value": "1"}]  # kept
+    assert by_name["b"]["image"] == "b:1"
+    assert "c" in by_name
+    assert [c["name"] for c in out["containers"]] == ["a", "b", "c"]
+
+
+def test_strategic_merge_none_deletes():
+    assert nominal.strategic_merge({"a": 1, "b": 2}, {"a": None}) == {"b": 2}
+
+
+def test_strategic_merge_plain_list_replaces():
+    assert nominal.strategic_merge({"l": [1, 2]}, {"l": [3]}) == {"l": [3]}
+
+
+def test_instance_id_deterministic_and_gpu_sensitive():
+    msc = {"port": 8000, "options": "--model x"}
+    a = instance_id(msc, ["GPU-0"])
+    assert a == instance_id(dict(msc), ["GPU-0"])  # dict order irrelevant
+    assert a != instance_id(msc, ["GPU-1"])
+    assert a != instance_id({**msc, "options": "--model y"}, ["GPU-0"])
+    assert a.startswith("I") and a.endswith("i")
+
+
+def test_nominal_hash_covers_spec_gpus_node():
+    spec = {"containers": [{"name": "inference-server"}]}
+    h = nominal_hash(spec, ["GPU-0"], "n1")
+    assert h == nominal_hash(dict(spec), ["GPU-0"], "n1")
+    assert h != nominal_hash(spec, ["GPU-1"], "n1")
+    assert h != nominal_hash(spec, ["GPU-0"], "n2")
+
+
+def test_template_hash_stable_under_canonicalization():
+    from fma_amd.controller.populator.podtemplate import canonicalize_template
+    t1 = {"spec": {"volumes": [{"name": "b"}, {"name": "a"}],
+                   "containers": [{"name": "launcher",
+                                   "ports": [{"containerPort": 9},
+                                             {"containerPort": 1}]}]}}
+    t2 = {"spec": {"volumes": [{"name": "a"}, {"name": "b"}],
+                   "containers": [{"name": "launcher",
+                                   "ports": [{"containerPort": 1},
+                                             {"containerPort": 9}]}]}}
+    assert template_hash(canonicalize_template(t1)) == \
+        template_hash(canonicalize_template(t2))
