@@ -21,7 +21,7 @@ from typing import Dict, List
 from fma_amd.api import contracts
 from fma_amd.store import objects as ob
 from fma_amd.store.client import StoreClient
-from fma_amd.store.memstore import Conflict, MemStore, NotFound
+from fma_amd.store.memstore import Conflict, NotFound
 
 ALLOCS_CONFIGMAP = "gpu-allocs"
 

@@ -32,8 +32,8 @@ from fma_amd.api import contracts
 from fma_amd.controller import metrics
 from fma_amd.controller.dualpods import nominal as nominal_mod
 from fma_amd.controller.dualpods.identity import instance_id as compute_iid
-from fma_amd.controller.dualpods.selection import (InstanceView, LauncherView,
-                                                   SelectionResult,
+from fma_amd.controller.dualpods.selection import (InstanceView,
+                                                   LauncherView,
                                                    select_or_reclaim)
 from fma_amd.controller.httpadapter import LauncherClient
 from fma_amd.controller.workqueue import QueueAndWorkers

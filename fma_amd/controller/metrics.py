@@ -11,8 +11,7 @@ from __future__ import annotations
 from typing import Optional
 
 try:
-    from prometheus_client import (REGISTRY, Counter, Gauge, Histogram,
-                                   start_http_server)
+    from prometheus_client import Gauge, Histogram, start_http_server
     HAVE_PROM = True
 except Exception:  # pragma: no cover
     HAVE_PROM = False

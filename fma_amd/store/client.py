@@ -14,8 +14,8 @@ from typing import Any, Dict, Iterator, List, Optional
 import httpx
 
 from fma_amd.store import objects as ob
-from fma_amd.store.memstore import (AlreadyExists, ApiError, Conflict,
-                                    Invalid, NotFound, WatchEvent)
+from fma_amd.store.memstore import (ApiError, Conflict, Invalid,
+                                    NotFound, WatchEvent)
 
 
 def _raise_for(code: int, message: str) -> None:

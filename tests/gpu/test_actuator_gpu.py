@@ -222,3 +222,19 @@ def test_model_swap_two_engines():
         b.sleep()
         a.wake_up()
         assert torch.equal(outa, a.model.forward(toks))
+
+
+def test_checkpoint_fast_load_gpu(tmp_path):
+    """Fast checkpoint load (pinned staging + pipelined H2D) is bit-exact
+    on the GPU."""
+    from fma_amd.models import loader
+    from fma_amd.models.llama import LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+    cfg = LlamaConfig.tiny()
+    src = ActuationEngine(cfg, 0, seed=41)
+    ckpt = str(tmp_path / "gck")
+    loader.save_params(src.params, ckpt, cfg)
+    dst = ActuationEngine(cfg, 0, seed=55, init_weights=False)
+    dst.load_checkpoint(ckpt)
+    toks = torch.randint(0, cfg.vocab_size, (1, 6), device="cuda:0")
+    assert torch.equal(src.model.forward(toks), dst.model.forward(toks))
