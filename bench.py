@@ -133,7 +133,7 @@ def main() -> None:
         value = mean_wake
         result = {
             "metric": "wake_up latency (s) + time-to-ready after swap, "
-                      "64 GiB model",
+                      "64 GiB model @ 1/2/4/8 MI355X",
             "value": round(value, 4),
             "unit": "s",
             "n_gpus": n_gpus,
