@@ -238,3 +238,24 @@ def test_checkpoint_fast_load_gpu(tmp_path):
     dst.load_checkpoint(ckpt)
     toks = torch.randint(0, cfg.vocab_size, (1, 6), device="cuda:0")
     assert torch.equal(src.model.forward(toks), dst.model.forward(toks))
+
+
+def test_hipgraph_decode_matches_eager():
+    """Captured-graph decode replays produce exactly the eager tokens."""
+    from fma_amd.models.decode_graph import StaticDecoder
+    from fma_amd.models.llama import LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+    eng = ActuationEngine(LlamaConfig.tiny(), 0, seed=17)
+    torch.manual_seed(5)
+    prompt = torch.randint(0, eng.cfg.vocab_size, (1, 6), device="cuda:0")
+    eager = eng.model.generate(prompt, max_new_tokens=8)
+
+    dec = StaticDecoder(eng.model, batch=1, max_seq=32)
+    dec.capture()
+    assert dec.graph is not None
+    graphed = dec.generate(prompt, max_new_tokens=8)
+    assert torch.equal(eager, graphed)
+    # reusable for a second prompt without re-capture
+    prompt2 = torch.randint(0, eng.cfg.vocab_size, (1, 3), device="cuda:0")
+    eager2 = eng.model.generate(prompt2, max_new_tokens=4)
+    assert torch.equal(dec.generate(prompt2, max_new_tokens=4), eager2)
