@@ -67,6 +67,19 @@ def main():
         r.wake_up()
         post_wake_ttft.append(first_token_seconds())
 
+    graphed_tps = None
+    eng = r.engine if hasattr(r, "engine") else r
+    if eng.on_gpu and eng.tp_size == 1:
+        from fma_amd.models.decode_graph import StaticDecoder
+        dec = StaticDecoder(eng.model, 1, args.prompt_len + args.new_tokens + 2)
+        dec.capture()
+        dec.generate(toks, args.new_tokens)  # warm
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        dec.generate(toks, args.new_tokens)
+        torch.cuda.synchronize()
+        graphed_tps = args.new_tokens / (time.perf_counter() - t0)
+
     print(json.dumps({
         "metric": "T_first_token / decode throughput",
         "model": args.model,
@@ -74,6 +87,7 @@ def main():
         "steady_ttft_s": round(steady_ttft, 4),
         "post_wake_ttft_s": round(min(post_wake_ttft), 4),
         "decode_tok_s": round(steady_tps, 2),
+        "decode_tok_s_hipgraph": round(graphed_tps, 2) if graphed_tps else None,
     }))
     if hasattr(r, "stop"):
         r.stop()
