@@ -844,6 +844,24 @@ class DeviceArena {
       handles_;
 };
 
+// Batch-1 bf16 GEMV on the torch current stream (composes with eager ops
+// without extra synchronization). Returns fp32 [M].
+at::Tensor gemv_bf16(const at::Tensor& W, const at::Tensor& x) {
+  TORCH_CHECK(W.is_cuda() && x.is_cuda(), "gemv_bf16 needs GPU tensors");
+  TORCH_CHECK(W.scalar_type() == at::kBFloat16 &&
+              x.scalar_type() == at::kBFloat16, "gemv_bf16 is bf16-only");
+  TORCH_CHECK(W.dim() == 2 && W.is_contiguous(), "W must be [M,K] contiguous");
+  const int64_t M = W.size(0), K = W.size(1);
+  TORCH_CHECK(x.numel() == K && x.is_contiguous(), "x must be K elements");
+  TORCH_CHECK((K & 7) == 0, "K must be a multiple of 8");
+  auto y = at::empty({M}, W.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream(W.device().index());
+  FMA_HIP_CHECK(fma_launch_gemv_bf16(
+      W.data_ptr(), x.data_ptr(), y.data_ptr<float>(),
+      static_cast<int>(M), static_cast<int>(K), stream.stream()));
+  return y;
+}
+
 std::tuple<int64_t, int64_t> device_mem_info(int device) {
   FMA_HIP_CHECK(hipSetDevice(device));
   size_t free_b = 0, total_b = 0;
@@ -865,6 +883,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("tensors"), py::arg("offsets"), py::arg("host"),
         py::arg("mode") = 0, py::arg("chunk_bytes") = 0);
   m.def("device_supports_vmm", &device_supports_vmm, py::arg("device"));
+  m.def("gemv_bf16", &gemv_bf16, "Batch-1 bf16 GEMV (fp32 out)",
+        py::arg("W"), py::arg("x"));
   m.def("gather_d2d", &gather_d2d,
         "Batched gather kernel D2D microbench (returns seconds/iteration)",
         py::arg("tensors"), py::arg("offsets"), py::arg("out"),

@@ -32,3 +32,7 @@ extern "C" hipError_t fma_launch_contiguous_copy(const void* src, void* dst,
                                                  hipStream_t stream);
 
 extern "C" hipError_t fma_launch_cache_invalidate(hipStream_t stream);
+
+extern "C" hipError_t fma_launch_gemv_bf16(const void* W, const void* x,
+                                           float* y, int M, int K,
+                                           hipStream_t stream);

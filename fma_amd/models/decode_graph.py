@@ -24,6 +24,8 @@ from typing import Optional
 import torch
 import torch.nn.functional as F
 
+from fma_amd.ops.linear import fast_linear
+
 from fma_amd.models.llama import LlamaModel, rmsnorm
 
 
@@ -77,9 +79,9 @@ class StaticDecoder:
         for li in range(cfg.num_layers):
             p = f"layers.{li}."
             h = rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps)
-            q = F.linear(h, P[p + "wq.weight"]).view(B, 1, q_heads, hd)
-            k = F.linear(h, P[p + "wk.weight"]).view(B, 1, kv_heads, hd)
-            v = F.linear(h, P[p + "wv.weight"]).view(B, 1, kv_heads, hd)
+            q = fast_linear(h, P[p + "wq.weight"]).view(B, 1, q_heads, hd)
+            k = fast_linear(h, P[p + "wk.weight"]).view(B, 1, kv_heads, hd)
+            v = fast_linear(h, P[p + "wv.weight"]).view(B, 1, kv_heads, hd)
             q = _rope1(q, cos, sin)
             k = _rope1(k, cos, sin)
             # static cache write at pos
@@ -94,14 +96,14 @@ class StaticDecoder:
             att = F.scaled_dot_product_attention(
                 q.transpose(1, 2), kh, vh, attn_mask=mask.to(q.dtype))
             att = att.transpose(1, 2).reshape(B, 1, q_heads * hd)
-            x = x + F.linear(att, P[p + "wo.weight"])
+            x = x + fast_linear(att, P[p + "wo.weight"])
             h = rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
-            gate = F.linear(h, P[p + "w_gate.weight"])
-            up = F.linear(h, P[p + "w_up.weight"])
-            x = x + F.linear(F.silu(gate) * up, P[p + "w_down.weight"])
+            gate = fast_linear(h, P[p + "w_gate.weight"])
+            up = fast_linear(h, P[p + "w_up.weight"])
+            x = x + fast_linear(F.silu(gate) * up, P[p + "w_down.weight"])
 
         x = rmsnorm(x, P["final_norm.weight"], cfg.norm_eps)
-        logits = F.linear(x[:, 0], P["lm_head.weight"]).float()
+        logits = fast_linear(x[:, 0], P["lm_head.weight"]).float()
         self.logits.copy_(logits)
         nxt = logits.argmax(-1, keepdim=True)
         self.pos.add_(1)

@@ -27,6 +27,8 @@ import torch
 import torch.distributed as dist
 import torch.nn.functional as F
 
+from fma_amd.ops.linear import fast_linear
+
 
 @dataclass
 class LlamaConfig:
@@ -239,9 +241,9 @@ class LlamaModel:
         for li in range(cfg.num_layers):
             p = f"layers.{li}."
             h = rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps)
-            q = F.linear(h, P[p + "wq.weight"]).view(B, T, q_heads, hd)
-            k = F.linear(h, P[p + "wk.weight"]).view(B, T, kv_heads, hd)
-            v = F.linear(h, P[p + "wv.weight"]).view(B, T, kv_heads, hd)
+            q = fast_linear(h, P[p + "wq.weight"]).view(B, T, q_heads, hd)
+            k = fast_linear(h, P[p + "wk.weight"]).view(B, T, kv_heads, hd)
+            v = fast_linear(h, P[p + "wv.weight"]).view(B, T, kv_heads, hd)
             q = apply_rope(q, self.rope_cos, self.rope_sin, start_pos)
             k = apply_rope(k, self.rope_cos, self.rope_sin, start_pos)
             if cache is not None:
@@ -260,16 +262,16 @@ class LlamaModel:
             att = F.scaled_dot_product_attention(
                 qh, kh, vh, is_causal=(T > 1))
             att = att.transpose(1, 2).reshape(B, T, q_heads * hd)
-            x = x + self._maybe_all_reduce(F.linear(att, P[p + "wo.weight"]))
+            x = x + self._maybe_all_reduce(fast_linear(att, P[p + "wo.weight"]))
 
             h = rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
-            gate = F.linear(h, P[p + "w_gate.weight"])
-            up = F.linear(h, P[p + "w_up.weight"])
+            gate = fast_linear(h, P[p + "w_gate.weight"])
+            up = fast_linear(h, P[p + "w_up.weight"])
             x = x + self._maybe_all_reduce(
-                F.linear(F.silu(gate) * up, P[p + "w_down.weight"]))
+                fast_linear(F.silu(gate) * up, P[p + "w_down.weight"]))
 
         x = rmsnorm(x, P["final_norm.weight"], cfg.norm_eps)
-        return F.linear(x, P["lm_head.weight"]).float()
+        return fast_linear(x, P["lm_head.weight"]).float()
 
     @torch.no_grad()
     def generate(self, prompt: torch.Tensor, max_new_tokens: int = 16,

@@ -1,0 +1,41 @@
+"""Linear dispatch: hand-written GEMV for batch-1 decode, hipBLASLt else.
+
+Batch-1 decode reads every weight byte once per token — pure streaming.
+hipBLASLt's batch-1 GEMV measured ~1 TB/s effective on llama shapes;
+csrc/kernels.hip's wave-per-row GEMV targets the streaming ceiling. The
+dispatch is shape-based: single-token inputs on bf16 weights take the
+kernel, everything else (prefill, batches) stays on hipBLASLt GEMMs.
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+import torch.nn.functional as F
+
+from fma_amd.ops import actuation
+
+_ENABLED = None
+
+
+def gemv_enabled() -> bool:
+    global _ENABLED
+    if _ENABLED is None:
+        _ENABLED = (torch.cuda.is_available()
+                    and actuation.native_available()
+                    and os.environ.get("FMA_DISABLE_GEMV") != "1")
+    return _ENABLED
+
+
+def fast_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """F.linear(x, w) with the decode-GEMV fast path."""
+    if (gemv_enabled() and x.is_cuda
+            and w.dtype == torch.bfloat16 and x.dtype == torch.bfloat16
+            and x.numel() == x.shape[-1]          # B*T == 1
+            and (w.shape[1] & 7) == 0
+            and w.shape[1] * 2 <= 64 * 1024       # x fits in dynamic LDS
+            and w.is_contiguous()):
+        y = actuation._C.gemv_bf16(w, x.reshape(-1).contiguous())
+        return y.to(torch.bfloat16).view(*x.shape[:-1], w.shape[0])
+    return F.linear(x, w)
