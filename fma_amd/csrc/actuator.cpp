@@ -845,8 +845,10 @@ class DeviceArena {
 };
 
 // Batch-1 bf16 GEMV on the torch current stream (composes with eager ops
-// without extra synchronization). Returns fp32 [M].
-at::Tensor gemv_bf16(const at::Tensor& W, const at::Tensor& x) {
+// without extra synchronization). fp32 out by default; out_bf16 fuses the
+// down-convert into the store (saves one elementwise launch per linear).
+at::Tensor gemv_bf16(const at::Tensor& W, const at::Tensor& x,
+                     bool out_bf16) {
   TORCH_CHECK(W.is_cuda() && x.is_cuda(), "gemv_bf16 needs GPU tensors");
   TORCH_CHECK(W.scalar_type() == at::kBFloat16 &&
               x.scalar_type() == at::kBFloat16, "gemv_bf16 is bf16-only");
@@ -854,12 +856,68 @@ at::Tensor gemv_bf16(const at::Tensor& W, const at::Tensor& x) {
   const int64_t M = W.size(0), K = W.size(1);
   TORCH_CHECK(x.numel() == K && x.is_contiguous(), "x must be K elements");
   TORCH_CHECK((K & 7) == 0, "K must be a multiple of 8");
-  auto y = at::empty({M}, W.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream(W.device().index());
+  if (out_bf16) {
+    auto y = at::empty({M}, W.options());
+    FMA_HIP_CHECK(fma_launch_gemv_bf16_out16(
+        W.data_ptr(), x.data_ptr(), y.data_ptr(),
+        static_cast<int>(M), static_cast<int>(K), stream.stream()));
+    return y;
+  }
+  auto y = at::empty({M}, W.options().dtype(at::kFloat));
   FMA_HIP_CHECK(fma_launch_gemv_bf16(
       W.data_ptr(), x.data_ptr(), y.data_ptr<float>(),
       static_cast<int>(M), static_cast<int>(K), stream.stream()));
   return y;
+}
+
+at::Tensor rmsnorm1_bf16(const at::Tensor& x, const at::Tensor& w,
+                         double eps) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
+              x.is_contiguous(), "x must be contiguous bf16 GPU");
+  const int64_t H = x.numel();
+  TORCH_CHECK(w.numel() == H && w.is_contiguous(), "w must match x");
+  TORCH_CHECK((H & 7) == 0, "H must be a multiple of 8");
+  auto y = at::empty_like(x);
+  auto stream = c10::hip::getCurrentHIPStream(x.device().index());
+  FMA_HIP_CHECK(fma_launch_rmsnorm1_bf16(
+      x.data_ptr(), w.data_ptr(), y.data_ptr(), static_cast<int>(H),
+      static_cast<float>(eps), stream.stream()));
+  return y;
+}
+
+at::Tensor silu_mul_bf16(const at::Tensor& g, const at::Tensor& u) {
+  TORCH_CHECK(g.is_cuda() && g.scalar_type() == at::kBFloat16 &&
+              g.is_contiguous() && u.is_contiguous(), "bf16 contiguous only");
+  TORCH_CHECK(g.numel() == u.numel(), "shape mismatch");
+  TORCH_CHECK((g.numel() & 7) == 0, "N must be a multiple of 8");
+  auto y = at::empty_like(g);
+  auto stream = c10::hip::getCurrentHIPStream(g.device().index());
+  FMA_HIP_CHECK(fma_launch_silu_mul_bf16(
+      g.data_ptr(), u.data_ptr(), y.data_ptr(),
+      static_cast<int>(g.numel()), stream.stream()));
+  return y;
+}
+
+// In-place single-position RoPE: q viewed as [heads, hd]; cos/sin one row.
+at::Tensor& rope1_bf16_(at::Tensor& q, const at::Tensor& cos_row,
+                        const at::Tensor& sin_row, int64_t heads,
+                        int64_t head_dim) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+              q.is_contiguous(), "q must be contiguous bf16 GPU");
+  TORCH_CHECK(q.numel() == heads * head_dim, "q size mismatch");
+  TORCH_CHECK(cos_row.scalar_type() == at::kFloat &&
+              sin_row.scalar_type() == at::kFloat, "cos/sin must be fp32");
+  TORCH_CHECK(cos_row.numel() == head_dim / 2 &&
+              sin_row.numel() == head_dim / 2, "cos/sin must be hd/2");
+  TORCH_CHECK(cos_row.is_contiguous() && sin_row.is_contiguous(),
+              "cos/sin must be contiguous");
+  auto stream = c10::hip::getCurrentHIPStream(q.device().index());
+  FMA_HIP_CHECK(fma_launch_rope1_bf16(
+      q.data_ptr(), cos_row.data_ptr<float>(), sin_row.data_ptr<float>(),
+      static_cast<int>(heads), static_cast<int>(head_dim / 2),
+      stream.stream()));
+  return q;
 }
 
 std::tuple<int64_t, int64_t> device_mem_info(int device) {
@@ -883,8 +941,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("tensors"), py::arg("offsets"), py::arg("host"),
         py::arg("mode") = 0, py::arg("chunk_bytes") = 0);
   m.def("device_supports_vmm", &device_supports_vmm, py::arg("device"));
-  m.def("gemv_bf16", &gemv_bf16, "Batch-1 bf16 GEMV (fp32 out)",
-        py::arg("W"), py::arg("x"));
+  m.def("gemv_bf16", &gemv_bf16, "Batch-1 bf16 GEMV",
+        py::arg("W"), py::arg("x"), py::arg("out_bf16") = false);
+  m.def("rmsnorm1_bf16", &rmsnorm1_bf16, py::arg("x"), py::arg("w"),
+        py::arg("eps"));
+  m.def("silu_mul_bf16", &silu_mul_bf16, py::arg("g"), py::arg("u"));
+  m.def("rope1_bf16_", &rope1_bf16_, py::arg("q"), py::arg("cos_row"),
+        py::arg("sin_row"), py::arg("heads"), py::arg("head_dim"));
   m.def("gather_d2d", &gather_d2d,
         "Batched gather kernel D2D microbench (returns seconds/iteration)",
         py::arg("tensors"), py::arg("offsets"), py::arg("out"),

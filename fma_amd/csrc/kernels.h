@@ -36,3 +36,16 @@ extern "C" hipError_t fma_launch_cache_invalidate(hipStream_t stream);
 extern "C" hipError_t fma_launch_gemv_bf16(const void* W, const void* x,
                                            float* y, int M, int K,
                                            hipStream_t stream);
+
+extern "C" hipError_t fma_launch_gemv_bf16_out16(const void* W, const void* x,
+                                                 void* y, int M, int K,
+                                                 hipStream_t stream);
+extern "C" hipError_t fma_launch_rmsnorm1_bf16(const void* x, const void* w,
+                                               void* y, int H, float eps,
+                                               hipStream_t stream);
+extern "C" hipError_t fma_launch_silu_mul_bf16(const void* g, const void* u,
+                                               void* y, int N,
+                                               hipStream_t stream);
+extern "C" hipError_t fma_launch_rope1_bf16(void* q, const float* cos_row,
+                                            const float* sin_row, int heads,
+                                            int half_hd, hipStream_t stream);

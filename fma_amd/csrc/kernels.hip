@@ -193,11 +193,33 @@ __device__ inline float bf16_to_f32(unsigned short h) {
   return c.f;
 }
 
-template <bool kUseLds>
+__device__ inline unsigned short f32_to_bf16(float f) {
+  union {
+    float f;
+    unsigned int u;
+  } c;
+  c.f = f;
+  if ((c.u & 0x7fffffffu) > 0x7f800000u) {  // NaN: quiet, torch-compatible
+    return static_cast<unsigned short>((c.u >> 16) | 0x0040u);
+  }
+  const unsigned int lsb = (c.u >> 16) & 1u;
+  c.u += 0x7fffu + lsb;  // round to nearest even
+  return static_cast<unsigned short>(c.u >> 16);
+}
+
+
+__device__ inline void gemv_store(float* y, int row, float acc) {
+  y[row] = acc;
+}
+__device__ inline void gemv_store(unsigned short* y, int row, float acc) {
+  y[row] = f32_to_bf16(acc);
+}
+
+template <bool kUseLds, typename OutT>
 __global__ __launch_bounds__(256) void gemv_bf16_kernel(
     const unsigned short* __restrict__ W,  // [M, K] row-major bf16
     const unsigned short* __restrict__ x,  // [K] bf16
-    float* __restrict__ y,                 // [M] fp32
+    OutT* __restrict__ y,                  // [M] fp32 or bf16
     int M, int K) {
   // x staged in LDS when it fits without hurting occupancy; for wide K
   // (w_down shapes) every wave reads the same x slices, which the L2
@@ -240,7 +262,7 @@ __global__ __launch_bounds__(256) void gemv_bf16_kernel(
     for (int off = 32; off > 0; off >>= 1) {
       acc += __shfl_down(acc, off, 64);
     }
-    if (lane == 0) y[row] = acc;
+    if (lane == 0) gemv_store(y, row, acc);
   }
 }
 
@@ -257,13 +279,163 @@ extern "C" hipError_t fma_launch_gemv_bf16(const void* W, const void* x,
   if (blocks > 2048) blocks = 2048;
   const size_t lds = static_cast<size_t>(K) * sizeof(unsigned short);
   if (lds <= 32 * 1024) {  // >= 5 blocks/CU with x staged
-    gemv_bf16_kernel<true><<<blocks, block, lds, stream>>>(
+    gemv_bf16_kernel<true, float><<<blocks, block, lds, stream>>>(
         static_cast<const unsigned short*>(W),
         static_cast<const unsigned short*>(x), y, M, K);
   } else {
-    gemv_bf16_kernel<false><<<blocks, block, 0, stream>>>(
+    gemv_bf16_kernel<false, float><<<blocks, block, 0, stream>>>(
         static_cast<const unsigned short*>(W),
         static_cast<const unsigned short*>(x), y, M, K);
   }
+  return hipGetLastError();
+}
+
+// ---------------------------------------------------------------------------
+// Fused decode elementwise kernels. Profiling the eager decode showed more
+// GPU time in fragmented elementwise kernels (rmsnorm = 5 launches, rope,
+// silu*up, bf16 converts) than in the GEMVs that do the actual weight
+// reads; each fused op here is one launch, fp32 math, RNE bf16 out.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+// One block: y = x * rsqrt(mean(x^2) + eps) * w  (single token, H <= 64k)
+__global__ __launch_bounds__(256) void rmsnorm1_bf16_kernel(
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ w,
+    unsigned short* __restrict__ y, int H, float eps) {
+  __shared__ float red[256];
+  float ss = 0.0f;
+  const int vec_h = H >> 3;
+  for (int i = threadIdx.x; i < vec_h; i += blockDim.x) {
+    const uint4 xv = reinterpret_cast<const uint4*>(x)[i];
+    const unsigned short* xh = reinterpret_cast<const unsigned short*>(&xv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float v = bf16_to_f32(xh[j]);
+      ss = fmaf(v, v, ss);
+    }
+  }
+  red[threadIdx.x] = ss;
+  __syncthreads();
+  for (int s = 128; s > 0; s >>= 1) {
+    if (threadIdx.x < s) red[threadIdx.x] += red[threadIdx.x + s];
+    __syncthreads();
+  }
+  const float scale = rsqrtf(red[0] / H + eps);
+  for (int i = threadIdx.x; i < vec_h; i += blockDim.x) {
+    const uint4 xv = reinterpret_cast<const uint4*>(x)[i];
+    const uint4 wv = reinterpret_cast<const uint4*>(w)[i];
+    const unsigned short* xh = reinterpret_cast<const unsigned short*>(&xv);
+    const unsigned short* wh = reinterpret_cast<const unsigned short*>(&wv);
+    uint4 ov;
+    unsigned short* oh = reinterpret_cast<unsigned short*>(&ov);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      oh[j] = f32_to_bf16(bf16_to_f32(xh[j]) * scale * bf16_to_f32(wh[j]));
+    }
+    reinterpret_cast<uint4*>(y)[i] = ov;
+  }
+}
+
+// y = silu(g) * u, elementwise over N bf16 values (N % 8 == 0)
+__global__ __launch_bounds__(256) void silu_mul_bf16_kernel(
+    const unsigned short* __restrict__ g,
+    const unsigned short* __restrict__ u,
+    unsigned short* __restrict__ y, int N) {
+  const int vec_n = N >> 3;
+  const int stride = gridDim.x * blockDim.x;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < vec_n; i += stride) {
+    const uint4 gv = reinterpret_cast<const uint4*>(g)[i];
+    const uint4 uv = reinterpret_cast<const uint4*>(u)[i];
+    const unsigned short* gh = reinterpret_cast<const unsigned short*>(&gv);
+    const unsigned short* uh = reinterpret_cast<const unsigned short*>(&uv);
+    uint4 ov;
+    unsigned short* oh = reinterpret_cast<unsigned short*>(&ov);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float gf = bf16_to_f32(gh[j]);
+      const float s = gf / (1.0f + __expf(-gf));
+      oh[j] = f32_to_bf16(s * bf16_to_f32(uh[j]));
+    }
+    reinterpret_cast<uint4*>(y)[i] = ov;
+  }
+}
+
+// In-place RoPE on one token: q [heads, hd] as (x0,x1) pairs; cos/sin [hd/2]
+// fp32 (one row of the host-precomputed table).
+__global__ __launch_bounds__(256) void rope1_bf16_kernel(
+    unsigned short* __restrict__ q, const float* __restrict__ cos_row,
+    const float* __restrict__ sin_row, int heads, int half_hd) {
+  const int total = heads * half_hd;
+  const int stride = gridDim.x * blockDim.x;
+  for (int p = blockIdx.x * blockDim.x + threadIdx.x; p < total; p += stride) {
+    const int d = p % half_hd;
+    const float c = cos_row[d];
+    const float s = sin_row[d];
+    const float x0 = bf16_to_f32(q[2 * p]);
+    const float x1 = bf16_to_f32(q[2 * p + 1]);
+    q[2 * p] = f32_to_bf16(x0 * c - x1 * s);
+    q[2 * p + 1] = f32_to_bf16(x0 * s + x1 * c);
+  }
+}
+
+}  // namespace
+
+extern "C" hipError_t fma_launch_gemv_bf16_out16(const void* W, const void* x,
+                                                 void* y, int M, int K,
+                                                 hipStream_t stream) {
+  if ((K & 7) != 0) return hipErrorInvalidValue;
+  const int block = 256;
+  const int waves_per_block = block / 64;
+  int blocks = (M + waves_per_block - 1) / waves_per_block;
+  if (blocks > 2048) blocks = 2048;
+  const size_t lds = static_cast<size_t>(K) * sizeof(unsigned short);
+  if (lds <= 32 * 1024) {
+    gemv_bf16_kernel<true, unsigned short><<<blocks, block, lds, stream>>>(
+        static_cast<const unsigned short*>(W),
+        static_cast<const unsigned short*>(x),
+        static_cast<unsigned short*>(y), M, K);
+  } else {
+    gemv_bf16_kernel<false, unsigned short><<<blocks, block, 0, stream>>>(
+        static_cast<const unsigned short*>(W),
+        static_cast<const unsigned short*>(x),
+        static_cast<unsigned short*>(y), M, K);
+  }
+  return hipGetLastError();
+}
+
+extern "C" hipError_t fma_launch_rmsnorm1_bf16(const void* x, const void* w,
+                                               void* y, int H, float eps,
+                                               hipStream_t stream) {
+  if ((H & 7) != 0) return hipErrorInvalidValue;
+  rmsnorm1_bf16_kernel<<<1, 256, 0, stream>>>(
+      static_cast<const unsigned short*>(x),
+      static_cast<const unsigned short*>(w),
+      static_cast<unsigned short*>(y), H, eps);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t fma_launch_silu_mul_bf16(const void* g, const void* u,
+                                               void* y, int N,
+                                               hipStream_t stream) {
+  if ((N & 7) != 0) return hipErrorInvalidValue;
+  int blocks = ((N >> 3) + 255) / 256;
+  if (blocks > 1024) blocks = 1024;
+  silu_mul_bf16_kernel<<<blocks, 256, 0, stream>>>(
+      static_cast<const unsigned short*>(g),
+      static_cast<const unsigned short*>(u),
+      static_cast<unsigned short*>(y), N);
+  return hipGetLastError();
+}
+
+extern "C" hipError_t fma_launch_rope1_bf16(void* q, const float* cos_row,
+                                            const float* sin_row, int heads,
+                                            int half_hd, hipStream_t stream) {
+  const int total = heads * half_hd;
+  int blocks = (total + 255) / 256;
+  if (blocks > 1024) blocks = 1024;
+  rope1_bf16_kernel<<<blocks, 256, 0, stream>>>(
+      static_cast<unsigned short*>(q), cos_row, sin_row, heads, half_hd);
   return hipGetLastError();
 }

@@ -24,6 +24,8 @@ from typing import Optional
 import torch
 import torch.nn.functional as F
 
+from fma_amd.ops.decode_ops import (fast_rmsnorm, fast_rope1,
+                                    fast_silu_mul)
 from fma_amd.ops.linear import fast_linear
 
 from fma_amd.models.llama import LlamaModel, rmsnorm
@@ -78,12 +80,16 @@ class StaticDecoder:
 
         for li in range(cfg.num_layers):
             p = f"layers.{li}."
-            h = rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps)
+            h = fast_rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps)
             q = fast_linear(h, P[p + "wq.weight"]).view(B, 1, q_heads, hd)
             k = fast_linear(h, P[p + "wk.weight"]).view(B, 1, kv_heads, hd)
             v = fast_linear(h, P[p + "wv.weight"]).view(B, 1, kv_heads, hd)
-            q = _rope1(q, cos, sin)
-            k = _rope1(k, cos, sin)
+            if q.is_cuda and q.dtype == torch.bfloat16:
+                q = fast_rope1(q, cos[0], sin[0])
+                k = fast_rope1(k, cos[0], sin[0])
+            else:
+                q = _rope1(q, cos, sin)
+                k = _rope1(k, cos, sin)
             # static cache write at pos
             self.cache[li, 0].index_copy_(1, self.pos.view(1), k)
             self.cache[li, 1].index_copy_(1, self.pos.view(1), v)
@@ -97,12 +103,13 @@ class StaticDecoder:
                 q.transpose(1, 2), kh, vh, attn_mask=mask.to(q.dtype))
             att = att.transpose(1, 2).reshape(B, 1, q_heads * hd)
             x = x + fast_linear(att, P[p + "wo.weight"])
-            h = rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
+            h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
             gate = fast_linear(h, P[p + "w_gate.weight"])
             up = fast_linear(h, P[p + "w_up.weight"])
-            x = x + fast_linear(F.silu(gate) * up, P[p + "w_down.weight"])
+            x = x + fast_linear(fast_silu_mul(gate, up),
+                                P[p + "w_down.weight"])
 
-        x = rmsnorm(x, P["final_norm.weight"], cfg.norm_eps)
+        x = fast_rmsnorm(x, P["final_norm.weight"], cfg.norm_eps)
         logits = fast_linear(x[:, 0], P["lm_head.weight"]).float()
         self.logits.copy_(logits)
         nxt = logits.argmax(-1, keepdim=True)

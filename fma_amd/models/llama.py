@@ -27,6 +27,8 @@ import torch
 import torch.distributed as dist
 import torch.nn.functional as F
 
+from fma_amd.ops.decode_ops import (fast_rmsnorm, fast_rope1,
+                                    fast_silu_mul)
 from fma_amd.ops.linear import fast_linear
 
 
@@ -238,14 +240,25 @@ class LlamaModel:
         hd = cfg.head_dim
 
         x = F.embedding(tokens, P["embed.weight"])
+        # fused single-token decode path (one kernel per elementwise op)
+        decode1 = (B == 1 and T == 1 and x.is_cuda
+                   and x.dtype == torch.bfloat16)
         for li in range(cfg.num_layers):
             p = f"layers.{li}."
-            h = rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps)
+            h = fast_rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps) \
+                if decode1 else rmsnorm(x, P[p + "attn_norm.weight"],
+                                        cfg.norm_eps)
             q = fast_linear(h, P[p + "wq.weight"]).view(B, T, q_heads, hd)
             k = fast_linear(h, P[p + "wk.weight"]).view(B, T, kv_heads, hd)
             v = fast_linear(h, P[p + "wv.weight"]).view(B, T, kv_heads, hd)
-            q = apply_rope(q, self.rope_cos, self.rope_sin, start_pos)
-            k = apply_rope(k, self.rope_cos, self.rope_sin, start_pos)
+            if decode1:
+                q = fast_rope1(q, self.rope_cos[start_pos],
+                               self.rope_sin[start_pos])
+                k = fast_rope1(k, self.rope_cos[start_pos],
+                               self.rope_sin[start_pos])
+            else:
+                q = apply_rope(q, self.rope_cos, self.rope_sin, start_pos)
+                k = apply_rope(k, self.rope_cos, self.rope_sin, start_pos)
             if cache is not None:
                 cache.data[li, 0, :, start_pos:start_pos + T] = k
                 cache.data[li, 1, :, start_pos:start_pos + T] = v
@@ -264,13 +277,17 @@ class LlamaModel:
             att = att.transpose(1, 2).reshape(B, T, q_heads * hd)
             x = x + self._maybe_all_reduce(fast_linear(att, P[p + "wo.weight"]))
 
-            h = rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
+            h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps) \
+                if decode1 else rmsnorm(x, P[p + "mlp_norm.weight"],
+                                        cfg.norm_eps)
             gate = fast_linear(h, P[p + "w_gate.weight"])
             up = fast_linear(h, P[p + "w_up.weight"])
+            act = fast_silu_mul(gate, up) if decode1 else F.silu(gate) * up
             x = x + self._maybe_all_reduce(
-                fast_linear(F.silu(gate) * up, P[p + "w_down.weight"]))
+                fast_linear(act, P[p + "w_down.weight"]))
 
-        x = rmsnorm(x, P["final_norm.weight"], cfg.norm_eps)
+        x = fast_rmsnorm(x, P["final_norm.weight"], cfg.norm_eps) \
+            if decode1 else rmsnorm(x, P["final_norm.weight"], cfg.norm_eps)
         return fast_linear(x, P["lm_head.weight"]).float()
 
     @torch.no_grad()

@@ -35,6 +35,7 @@ def fast_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
             and x.numel() == x.shape[-1]          # B*T == 1
             and (w.shape[1] & 7) == 0
             and w.is_contiguous()):
-        y = actuation._C.gemv_bf16(w, x.reshape(-1).contiguous())
-        return y.to(torch.bfloat16).view(*x.shape[:-1], w.shape[0])
+        y = actuation._C.gemv_bf16(w, x.reshape(-1).contiguous(),
+                                   True)  # bf16 out, convert fused
+        return y.view(*x.shape[:-1], w.shape[0])
     return F.linear(x, w)

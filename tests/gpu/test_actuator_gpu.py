@@ -285,3 +285,58 @@ def test_fast_linear_dispatch_matches_hipblaslt():
     xb = torch.randn(2, 3, 256, dtype=torch.bfloat16, device="cuda:0")
     assert torch.equal(fast_linear(xb, W),
                        torch.nn.functional.linear(xb, W))
+
+
+def test_fused_decode_ops_match_references():
+    """Fused rmsnorm / silu*up / rope vs plain PyTorch fp32 references."""
+    import fma_amd._C as C
+    H = 4096
+    x = torch.randn(H, dtype=torch.bfloat16, device="cuda:0")
+    w = torch.randn(H, dtype=torch.bfloat16, device="cuda:0")
+    ref = (x.float() * torch.rsqrt(x.float().pow(2).mean() + 1e-5)
+           * w.float()).to(torch.bfloat16)
+    out = C.rmsnorm1_bf16(x, w, 1e-5)
+    assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2)
+
+    g = torch.randn(H, dtype=torch.bfloat16, device="cuda:0")
+    u = torch.randn(H, dtype=torch.bfloat16, device="cuda:0")
+    ref = (torch.nn.functional.silu(g.float()) * u.float()).to(torch.bfloat16)
+    out = C.silu_mul_bf16(g, u)
+    assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2)
+
+    heads, hd = 32, 128
+    q = torch.randn(heads * hd, dtype=torch.bfloat16, device="cuda:0")
+    cos = torch.randn(hd // 2, dtype=torch.float32, device="cuda:0")
+    sin = torch.randn(hd // 2, dtype=torch.float32, device="cuda:0")
+    qf = q.float().view(heads, hd // 2, 2)
+    x0, x1 = qf[..., 0], qf[..., 1]
+    ref = torch.stack((x0 * cos - x1 * sin, x0 * sin + x1 * cos),
+                      dim=-1).reshape(-1).to(torch.bfloat16)
+    C.rope1_bf16_(q, cos, sin, heads, hd)
+    assert torch.allclose(q.float(), ref.float(), atol=2e-2, rtol=2e-2)
+
+
+def test_fused_decode_path_matches_eager_prefill_decode():
+    """GPU: single-token fused forward continues a prefilled cache with the
+    same tokens the all-eager (batched, non-fused) path would produce."""
+    import os
+    from fma_amd.models.llama import LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+    eng = ActuationEngine(LlamaConfig.tiny(), 0, seed=23)
+    torch.manual_seed(9)
+    prompt = torch.randint(0, eng.cfg.vocab_size, (1, 6), device="cuda:0")
+    fused = eng.model.generate(prompt, max_new_tokens=6)
+    os.environ["FMA_DISABLE_FUSED_OPS"] = "1"
+    os.environ["FMA_DISABLE_GEMV"] = "1"
+    import fma_amd.ops.decode_ops as dops
+    import fma_amd.ops.linear as lin
+    dops._ENABLED = None
+    lin._ENABLED = None
+    try:
+        eager = eng.model.generate(prompt, max_new_tokens=6)
+    finally:
+        del os.environ["FMA_DISABLE_FUSED_OPS"]
+        del os.environ["FMA_DISABLE_GEMV"]
+        dops._ENABLED = None
+        lin._ENABLED = None
+    assert torch.equal(fused, eager), (fused, eager)
