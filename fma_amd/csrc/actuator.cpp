@@ -920,6 +920,29 @@ at::Tensor& rope1_bf16_(at::Tensor& q, const at::Tensor& cos_row,
   return q;
 }
 
+// Single-token GQA decode attention over the [S, kvH, hd] cache slices.
+at::Tensor attn_decode_bf16(const at::Tensor& q, const at::Tensor& k,
+                            const at::Tensor& v, int64_t t) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+              q.is_contiguous(), "q must be contiguous bf16 [qH, hd]");
+  TORCH_CHECK(k.dim() == 3 && k.is_contiguous() && v.is_contiguous(),
+              "k/v must be contiguous [S, kvH, hd]");
+  TORCH_CHECK(k.scalar_type() == at::kBFloat16 &&
+              v.scalar_type() == at::kBFloat16, "bf16 only");
+  const int64_t q_heads = q.size(0), hd = q.size(1);
+  const int64_t kv_heads = k.size(1);
+  TORCH_CHECK(k.size(2) == hd && v.sizes() == k.sizes(), "shape mismatch");
+  TORCH_CHECK(t >= 1 && t <= k.size(0), "t out of cache bounds");
+  auto out = at::empty_like(q);
+  auto stream = c10::hip::getCurrentHIPStream(q.device().index());
+  FMA_HIP_CHECK(fma_launch_attn_decode_bf16(
+      q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+      static_cast<int>(t), static_cast<int>(q_heads),
+      static_cast<int>(kv_heads), static_cast<int>(hd),
+      static_cast<long long>(kv_heads * hd), stream.stream()));
+  return out;
+}
+
 std::tuple<int64_t, int64_t> device_mem_info(int device) {
   FMA_HIP_CHECK(hipSetDevice(device));
   size_t free_b = 0, total_b = 0;
@@ -946,6 +969,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm1_bf16", &rmsnorm1_bf16, py::arg("x"), py::arg("w"),
         py::arg("eps"));
   m.def("silu_mul_bf16", &silu_mul_bf16, py::arg("g"), py::arg("u"));
+  m.def("attn_decode_bf16", &attn_decode_bf16, py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("t"));
   m.def("rope1_bf16_", &rope1_bf16_, py::arg("q"), py::arg("cos_row"),
         py::arg("sin_row"), py::arg("heads"), py::arg("head_dim"));
   m.def("gather_d2d", &gather_d2d,

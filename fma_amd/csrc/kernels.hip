@@ -439,3 +439,112 @@ extern "C" hipError_t fma_launch_rope1_bf16(void* q, const float* cos_row,
       static_cast<unsigned short*>(q), cos_row, sin_row, heads, half_hd);
   return hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// Single-token GQA decode attention (flash-style online softmax).
+//
+// out[qh] = softmax(q[qh] . K[kv(qh), :t] / sqrt(hd)) @ V[kv(qh), :t]
+//
+// One workgroup (4 waves) per query head; each wave walks seq positions
+// (4 at a time per block), lanes split head_dim (2 elements each -> the
+// 256 B K/V row reads are fully coalesced and the per-wave accumulator
+// is just 2 VGPRs/lane). Replaces the eager path's repeat_interleave
+// copies (2 full-cache copies per layer) and ~6 launches per layer.
+// Layout: K/V rows may be strided (cache is [S, kvH, hd]).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+__global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
+    const unsigned short* __restrict__ q,   // [qH, hd]
+    const unsigned short* __restrict__ K,   // rows: K + s*k_stride + kvh*hd
+    const unsigned short* __restrict__ V,
+    unsigned short* __restrict__ out,       // [qH, hd]
+    int t, int q_heads, int kv_heads, int hd,
+    long long k_stride /* elements between seq positions */) {
+  const int qh = blockIdx.x;
+  if (qh >= q_heads) return;
+  const int kvh = qh / (q_heads / kv_heads);
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves = blockDim.x >> 6;
+  const int per_lane = hd / 64;  // 2 for hd=128
+
+  // q slice for this head, lane's components (fp32)
+  float qreg[4];
+  const float scale = rsqrtf(static_cast<float>(hd));
+  for (int j = 0; j < per_lane; ++j) {
+    qreg[j] = bf16_to_f32(q[qh * hd + lane * per_lane + j]) * scale;
+  }
+
+  float m = -1e30f, l = 0.0f;
+  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+
+  const unsigned short* kbase = K + static_cast<long long>(kvh) * hd;
+  const unsigned short* vbase = V + static_cast<long long>(kvh) * hd;
+  for (int s = wave; s < t; s += waves) {
+    const unsigned short* krow = kbase + s * k_stride;
+    float dot = 0.0f;
+    for (int j = 0; j < per_lane; ++j) {
+      dot = fmaf(qreg[j], bf16_to_f32(krow[lane * per_lane + j]), dot);
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      dot += __shfl_down(dot, off, 64);
+    }
+    dot = __shfl(dot, 0, 64);  // broadcast score
+    const float m_new = fmaxf(m, dot);
+    const float alpha = __expf(m - m_new);
+    const float w = __expf(dot - m_new);
+    l = l * alpha + w;
+    const unsigned short* vrow = vbase + s * k_stride;
+    for (int j = 0; j < per_lane; ++j) {
+      acc[j] = acc[j] * alpha + w * bf16_to_f32(vrow[lane * per_lane + j]);
+    }
+    m = m_new;
+  }
+
+  // merge the 4 waves' (m, l, acc) via LDS
+  __shared__ float s_m[4], s_l[4], s_acc[4][128 * 2];
+  if (lane == 0) {
+    s_m[wave] = m;
+    s_l[wave] = l;
+  }
+  for (int j = 0; j < per_lane; ++j) {
+    s_acc[wave][lane * per_lane + j] = acc[j];
+  }
+  __syncthreads();
+  if (wave == 0) {
+    float m_tot = -1e30f;
+    for (int wv = 0; wv < waves; ++wv) m_tot = fmaxf(m_tot, s_m[wv]);
+    float l_tot = 0.0f;
+    float out_acc[4] = {0.f, 0.f, 0.f, 0.f};
+    for (int wv = 0; wv < waves; ++wv) {
+      const float alpha = __expf(s_m[wv] - m_tot);
+      l_tot += s_l[wv] * alpha;
+      for (int j = 0; j < per_lane; ++j) {
+        out_acc[j] += s_acc[wv][lane * per_lane + j] * alpha;
+      }
+    }
+    const float inv = 1.0f / l_tot;
+    for (int j = 0; j < per_lane; ++j) {
+      out[qh * hd + lane * per_lane + j] = f32_to_bf16(out_acc[j] * inv);
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" hipError_t fma_launch_attn_decode_bf16(
+    const void* q, const void* K, const void* V, void* out, int t,
+    int q_heads, int kv_heads, int hd, long long k_stride,
+    hipStream_t stream) {
+  if (hd > 256 || (hd & 63) != 0) return hipErrorInvalidValue;
+  if (q_heads % kv_heads != 0) return hipErrorInvalidValue;
+  attn_decode_bf16_kernel<<<q_heads, 256, 0, stream>>>(
+      static_cast<const unsigned short*>(q),
+      static_cast<const unsigned short*>(K),
+      static_cast<const unsigned short*>(V),
+      static_cast<unsigned short*>(out), t, q_heads, kv_heads, hd, k_stride);
+  return hipGetLastError();
+}

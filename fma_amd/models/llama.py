@@ -27,8 +27,8 @@ import torch
 import torch.distributed as dist
 import torch.nn.functional as F
 
-from fma_amd.ops.decode_ops import (fast_rmsnorm, fast_rope1,
-                                    fast_silu_mul)
+from fma_amd.ops.decode_ops import (fast_attn_decode, fast_rmsnorm,
+                                    fast_rope1, fast_silu_mul)
 from fma_amd.ops.linear import fast_linear
 
 
@@ -264,17 +264,24 @@ class LlamaModel:
                 cache.data[li, 1, :, start_pos:start_pos + T] = v
                 k = cache.data[li, 0, :, : start_pos + T]
                 v = cache.data[li, 1, :, : start_pos + T]
-            # SDPA wants [B, heads, T, hd]
-            qh = q.transpose(1, 2)
-            kh = k.transpose(1, 2)
-            vh = v.transpose(1, 2)
-            if kv_heads != q_heads:
-                rep = q_heads // kv_heads
-                kh = kh.repeat_interleave(rep, dim=1)
-                vh = vh.repeat_interleave(rep, dim=1)
-            att = F.scaled_dot_product_attention(
-                qh, kh, vh, is_causal=(T > 1))
-            att = att.transpose(1, 2).reshape(B, T, q_heads * hd)
+            att = None
+            if decode1 and cache is not None:
+                # fused GQA decode attention over the raw cache (no
+                # repeat_interleave copies, one launch)
+                att = fast_attn_decode(q, cache.data[li, 0, 0],
+                                       cache.data[li, 1, 0], start_pos + 1)
+            if att is None:
+                # SDPA wants [B, heads, T, hd]
+                qh = q.transpose(1, 2)
+                kh = k.transpose(1, 2)
+                vh = v.transpose(1, 2)
+                if kv_heads != q_heads:
+                    rep = q_heads // kv_heads
+                    kh = kh.repeat_interleave(rep, dim=1)
+                    vh = vh.repeat_interleave(rep, dim=1)
+                att = F.scaled_dot_product_attention(
+                    qh, kh, vh, is_causal=(T > 1))
+                att = att.transpose(1, 2).reshape(B, T, q_heads * hd)
             x = x + self._maybe_all_reduce(fast_linear(att, P[p + "wo.weight"]))
 
             h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps) \

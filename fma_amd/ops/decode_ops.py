@@ -64,3 +64,19 @@ def fast_rope1(x: torch.Tensor, cos_row: torch.Tensor, sin_row: torch.Tensor
     x0, x1 = xf[..., 0], xf[..., 1]
     out = torch.stack((x0 * c - x1 * s, x0 * s + x1 * c), dim=-1)
     return out.view(B, T, heads, hd).to(x.dtype)
+
+
+def fast_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
+                     v_cache: torch.Tensor, t: int):
+    """q [1,1,qH,hd]; k/v_cache [S,kvH,hd] (one layer, batch 1). Returns
+    [1,1,qH*hd] or None when the fused kernel does not apply."""
+    if not available():
+        return None
+    _, _, qh, hd = q.shape
+    if (hd % 64 != 0 or hd > 256 or not q.is_cuda
+            or q.dtype != torch.bfloat16
+            or qh % k_cache.shape[1] != 0):
+        return None
+    out = actuation._C.attn_decode_bf16(
+        q.reshape(qh, hd).contiguous(), k_cache, v_cache, t)
+    return out.view(1, 1, qh * hd)

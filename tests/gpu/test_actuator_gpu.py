@@ -340,3 +340,46 @@ def test_fused_decode_path_matches_eager_prefill_decode():
         dops._ENABLED = None
         lin._ENABLED = None
     assert torch.equal(fused, eager), (fused, eager)
+
+
+def test_attn_decode_matches_sdpa_reference():
+    import fma_amd._C as C
+    torch.manual_seed(7)
+    for (qH, kvH, hd, S, t) in ((32, 8, 128, 256, 100), (64, 8, 128, 64, 64),
+                                (8, 8, 64, 32, 1), (16, 2, 128, 512, 511)):
+        q = torch.randn(qH, hd, dtype=torch.bfloat16, device="cuda:0")
+        k = torch.randn(S, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
+        v = torch.randn(S, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
+        out = C.attn_decode_bf16(q, k, v, t)
+        # plain fp32 reference
+        rep = qH // kvH
+        kf = k[:t].float().permute(1, 0, 2).repeat_interleave(rep, 0)
+        vf = v[:t].float().permute(1, 0, 2).repeat_interleave(rep, 0)
+        scores = (q.float().unsqueeze(1) * kf).sum(-1) / (hd ** 0.5)
+        w = torch.softmax(scores, dim=-1)
+        ref = (w.unsqueeze(-1) * vf).sum(1)
+        assert torch.allclose(out.float(), ref, atol=2e-2, rtol=2e-2), \
+            (qH, kvH, hd, S, t)
+
+
+def test_decode_with_fused_attention_matches_nonfused():
+    import os
+    from fma_amd.models.llama import LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+    # head_dim 128 so the fused attention applies
+    cfg = LlamaConfig(name="midi", vocab_size=1024, hidden_size=512,
+                      intermediate_size=1024, num_layers=2, num_heads=4,
+                      num_kv_heads=2, max_seq_len=128)
+    eng = ActuationEngine(cfg, 0, seed=29)
+    torch.manual_seed(11)
+    prompt = torch.randint(0, cfg.vocab_size, (1, 5), device="cuda:0")
+    fused = eng.model.generate(prompt, max_new_tokens=6)
+    os.environ["FMA_DISABLE_FUSED_OPS"] = "1"
+    import fma_amd.ops.decode_ops as dops
+    dops._ENABLED = None
+    try:
+        plain = eng.model.generate(prompt, max_new_tokens=6)
+    finally:
+        del os.environ["FMA_DISABLE_FUSED_OPS"]
+        dops._ENABLED = None
+    assert torch.equal(fused, plain)
