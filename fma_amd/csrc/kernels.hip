@@ -455,6 +455,8 @@ extern "C" hipError_t fma_launch_rope1_bf16(void* q, const float* cos_row,
 
 namespace {
 
+#define FMA_ATTN_MAX_T 8192
+
 __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
     const unsigned short* __restrict__ q,   // [qH, hd]
     const unsigned short* __restrict__ K,   // rows: K + s*k_stride + kvh*hd
@@ -462,73 +464,105 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
     unsigned short* __restrict__ out,       // [qH, hd]
     int t, int q_heads, int kv_heads, int hd,
     long long k_stride /* elements between seq positions */) {
+  // Two passes so no step serializes on the previous one:
+  //  1. scores: one THREAD per seq position (full-hd dot, q read from
+  //     LDS) -> s_scores; block-reduce max and exp-sum;
+  //  2. output: waves split positions, lanes split hd, weights read from
+  //     LDS — the only loop-carried dependency is the FMA accumulator.
+  __shared__ float s_scores[FMA_ATTN_MAX_T];
+  __shared__ float s_q[256];
+  __shared__ float s_red[256];
+  __shared__ float s_wacc[4][256];
+
   const int qh = blockIdx.x;
   if (qh >= q_heads) return;
   const int kvh = qh / (q_heads / kv_heads);
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int waves = blockDim.x >> 6;
-  const int per_lane = hd / 64;  // 2 for hd=128
-
-  // q slice for this head, lane's components (fp32)
-  float qreg[4];
+  const int per_lane = hd / 64;
   const float scale = rsqrtf(static_cast<float>(hd));
-  for (int j = 0; j < per_lane; ++j) {
-    qreg[j] = bf16_to_f32(q[qh * hd + lane * per_lane + j]) * scale;
-  }
 
-  float m = -1e30f, l = 0.0f;
-  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int i = threadIdx.x; i < hd; i += blockDim.x) {
+    s_q[i] = bf16_to_f32(q[qh * hd + i]) * scale;
+  }
+  __syncthreads();
 
   const unsigned short* kbase = K + static_cast<long long>(kvh) * hd;
   const unsigned short* vbase = V + static_cast<long long>(kvh) * hd;
-  for (int s = wave; s < t; s += waves) {
-    const unsigned short* krow = kbase + s * k_stride;
-    float dot = 0.0f;
-    for (int j = 0; j < per_lane; ++j) {
-      dot = fmaf(qreg[j], bf16_to_f32(krow[lane * per_lane + j]), dot);
-    }
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1) {
-      dot += __shfl_down(dot, off, 64);
-    }
-    dot = __shfl(dot, 0, 64);  // broadcast score
-    const float m_new = fmaxf(m, dot);
-    const float alpha = __expf(m - m_new);
-    const float w = __expf(dot - m_new);
-    l = l * alpha + w;
-    const unsigned short* vrow = vbase + s * k_stride;
-    for (int j = 0; j < per_lane; ++j) {
-      acc[j] = acc[j] * alpha + w * bf16_to_f32(vrow[lane * per_lane + j]);
-    }
-    m = m_new;
-  }
 
-  // merge the 4 waves' (m, l, acc) via LDS
-  __shared__ float s_m[4], s_l[4], s_acc[4][128 * 2];
-  if (lane == 0) {
-    s_m[wave] = m;
-    s_l[wave] = l;
+  // pass 1: scores
+  float local_max = -1e30f;
+  for (int s0 = threadIdx.x; s0 < t; s0 += blockDim.x) {
+    const unsigned short* krow = kbase + s0 * k_stride;
+    float dot = 0.0f;
+    for (int i8 = 0; i8 < hd; i8 += 8) {
+      const uint4 kv8 = *reinterpret_cast<const uint4*>(krow + i8);
+      const unsigned short* kh = reinterpret_cast<const unsigned short*>(&kv8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        dot = fmaf(s_q[i8 + j], bf16_to_f32(kh[j]), dot);
+      }
+    }
+    s_scores[s0] = dot;
+    local_max = fmaxf(local_max, dot);
+  }
+  s_red[threadIdx.x] = local_max;
+  __syncthreads();
+  for (int r = 128; r > 0; r >>= 1) {
+    if (threadIdx.x < r) {
+      s_red[threadIdx.x] = fmaxf(s_red[threadIdx.x], s_red[threadIdx.x + r]);
+    }
+    __syncthreads();
+  }
+  const float m = s_red[0];
+  __syncthreads();
+  float local_sum = 0.0f;
+  for (int s0 = threadIdx.x; s0 < t; s0 += blockDim.x) {
+    const float w = __expf(s_scores[s0] - m);
+    s_scores[s0] = w;
+    local_sum += w;
+  }
+  s_red[threadIdx.x] = local_sum;
+  __syncthreads();
+  for (int r = 128; r > 0; r >>= 1) {
+    if (threadIdx.x < r) s_red[threadIdx.x] += s_red[threadIdx.x + r];
+    __syncthreads();
+  }
+  const float inv_l = 1.0f / s_red[0];
+
+  // pass 2: weighted V accumulation, 4-deep unroll for FMA-latency ILP
+  float acc0[4] = {0.f, 0.f, 0.f, 0.f};
+  float acc1[4] = {0.f, 0.f, 0.f, 0.f};
+  int s0 = wave;
+  for (; s0 + waves < t; s0 += 2 * waves) {
+    const unsigned short* vrow_a = vbase + s0 * k_stride;
+    const unsigned short* vrow_b = vbase + (s0 + waves) * k_stride;
+    const float wa = s_scores[s0];
+    const float wb = s_scores[s0 + waves];
+    for (int j = 0; j < per_lane; ++j) {
+      acc0[j] = fmaf(wa, bf16_to_f32(vrow_a[lane * per_lane + j]), acc0[j]);
+      acc1[j] = fmaf(wb, bf16_to_f32(vrow_b[lane * per_lane + j]), acc1[j]);
+    }
+  }
+  if (s0 < t) {
+    const unsigned short* vrow = vbase + s0 * k_stride;
+    const float w = s_scores[s0];
+    for (int j = 0; j < per_lane; ++j) {
+      acc0[j] = fmaf(w, bf16_to_f32(vrow[lane * per_lane + j]), acc0[j]);
+    }
   }
   for (int j = 0; j < per_lane; ++j) {
-    s_acc[wave][lane * per_lane + j] = acc[j];
+    s_wacc[wave][lane * per_lane + j] = acc0[j] + acc1[j];
   }
   __syncthreads();
   if (wave == 0) {
-    float m_tot = -1e30f;
-    for (int wv = 0; wv < waves; ++wv) m_tot = fmaxf(m_tot, s_m[wv]);
-    float l_tot = 0.0f;
-    float out_acc[4] = {0.f, 0.f, 0.f, 0.f};
-    for (int wv = 0; wv < waves; ++wv) {
-      const float alpha = __expf(s_m[wv] - m_tot);
-      l_tot += s_l[wv] * alpha;
-      for (int j = 0; j < per_lane; ++j) {
-        out_acc[j] += s_acc[wv][lane * per_lane + j] * alpha;
-      }
-    }
-    const float inv = 1.0f / l_tot;
     for (int j = 0; j < per_lane; ++j) {
-      out[qh * hd + lane * per_lane + j] = f32_to_bf16(out_acc[j] * inv);
+      float v_out = 0.0f;
+      for (int wv = 0; wv < waves; ++wv) {
+        v_out += s_wacc[wv][lane * per_lane + j];
+      }
+      out[qh * hd + lane * per_lane + j] = f32_to_bf16(v_out * inv_l);
     }
   }
 }
@@ -541,6 +575,7 @@ extern "C" hipError_t fma_launch_attn_decode_bf16(
     hipStream_t stream) {
   if (hd > 256 || (hd & 63) != 0) return hipErrorInvalidValue;
   if (q_heads % kv_heads != 0) return hipErrorInvalidValue;
+  if (t > FMA_ATTN_MAX_T) return hipErrorInvalidValue;
   attn_decode_bf16_kernel<<<q_heads, 256, 0, stream>>>(
       static_cast<const unsigned short*>(q),
       static_cast<const unsigned short*>(K),
