@@ -935,11 +935,21 @@ at::Tensor attn_decode_bf16(const at::Tensor& q, const at::Tensor& k,
   TORCH_CHECK(t >= 1 && t <= k.size(0), "t out of cache bounds");
   auto out = at::empty_like(q);
   auto stream = c10::hip::getCurrentHIPStream(q.device().index());
+  const int chunks = fma_attn_decode_chunks(static_cast<int>(t),
+                                            static_cast<int>(q_heads));
+  at::Tensor partials;
+  float* pptr = nullptr;
+  if (chunks > 1) {
+    partials = at::empty({q_heads, chunks, hd + 2},
+                         q.options().dtype(at::kFloat));
+    pptr = partials.data_ptr<float>();
+  }
   FMA_HIP_CHECK(fma_launch_attn_decode_bf16(
       q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
       static_cast<int>(t), static_cast<int>(q_heads),
       static_cast<int>(kv_heads), static_cast<int>(hd),
-      static_cast<long long>(kv_heads * hd), stream.stream()));
+      static_cast<long long>(kv_heads * hd), pptr, chunks,
+      stream.stream()));
   return out;
 }
 

@@ -50,7 +50,8 @@ extern "C" hipError_t fma_launch_rope1_bf16(void* q, const float* cos_row,
                                             const float* sin_row, int heads,
                                             int half_hd, hipStream_t stream);
 
+extern "C" int fma_attn_decode_chunks(int t, int q_heads);
 extern "C" hipError_t fma_launch_attn_decode_bf16(
     const void* q, const void* K, const void* V, void* out, int t,
     int q_heads, int kv_heads, int hd, long long k_stride,
-    hipStream_t stream);
+    float* partials, int chunks, hipStream_t stream);

@@ -457,25 +457,30 @@ namespace {
 
 #define FMA_ATTN_MAX_T 8192
 
+// Split-sequence ("flash-decode") layout: grid.x = q_heads, grid.y =
+// seq chunks, so long contexts fill all 256 CUs instead of q_heads
+// blocks. Each block emits an (m, l, acc[hd]) partial; a tiny combine
+// kernel merges chunks per head. chunks == 1 writes the output directly.
 __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
     const unsigned short* __restrict__ q,   // [qH, hd]
     const unsigned short* __restrict__ K,   // rows: K + s*k_stride + kvh*hd
     const unsigned short* __restrict__ V,
-    unsigned short* __restrict__ out,       // [qH, hd]
+    unsigned short* __restrict__ out,       // [qH, hd] (chunks == 1)
+    float* __restrict__ partials,           // [qH, chunks, hd + 2]
     int t, int q_heads, int kv_heads, int hd,
-    long long k_stride /* elements between seq positions */) {
-  // Two passes so no step serializes on the previous one:
-  //  1. scores: one THREAD per seq position (full-hd dot, q read from
-  //     LDS) -> s_scores; block-reduce max and exp-sum;
-  //  2. output: waves split positions, lanes split hd, weights read from
-  //     LDS — the only loop-carried dependency is the FMA accumulator.
-  __shared__ float s_scores[FMA_ATTN_MAX_T];
+    long long k_stride) {
+  __shared__ float s_scores[FMA_ATTN_MAX_T / 4];
   __shared__ float s_q[256];
   __shared__ float s_red[256];
   __shared__ float s_wacc[4][256];
 
   const int qh = blockIdx.x;
-  if (qh >= q_heads) return;
+  const int chunk = blockIdx.y;
+  const int chunks = gridDim.y;
+  const int span = (t + chunks - 1) / chunks;
+  const int s_begin = chunk * span;
+  const int s_end = min(s_begin + span, t);
+  const int n = s_end - s_begin;
   const int kvh = qh / (q_heads / kv_heads);
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -487,14 +492,29 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
     s_q[i] = bf16_to_f32(q[qh * hd + i]) * scale;
   }
   __syncthreads();
+  if (n <= 0) {
+    if (chunks > 1 && threadIdx.x == 0) {
+      float* p = partials + (static_cast<long long>(qh) * chunks + chunk) *
+                                (hd + 2);
+      p[hd] = -1e30f;
+      p[hd + 1] = 0.0f;
+    }
+    if (chunks > 1) {
+      for (int i = threadIdx.x; i < hd; i += blockDim.x) {
+        partials[(static_cast<long long>(qh) * chunks + chunk) * (hd + 2) +
+                 i] = 0.0f;
+      }
+    }
+    return;
+  }
 
   const unsigned short* kbase = K + static_cast<long long>(kvh) * hd;
   const unsigned short* vbase = V + static_cast<long long>(kvh) * hd;
 
-  // pass 1: scores
+  // pass 1: scores for this chunk
   float local_max = -1e30f;
-  for (int s0 = threadIdx.x; s0 < t; s0 += blockDim.x) {
-    const unsigned short* krow = kbase + s0 * k_stride;
+  for (int s0 = threadIdx.x; s0 < n; s0 += blockDim.x) {
+    const unsigned short* krow = kbase + (s_begin + s0) * k_stride;
     float dot = 0.0f;
     for (int i8 = 0; i8 < hd; i8 += 8) {
       const uint4 kv8 = *reinterpret_cast<const uint4*>(krow + i8);
@@ -518,7 +538,7 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
   const float m = s_red[0];
   __syncthreads();
   float local_sum = 0.0f;
-  for (int s0 = threadIdx.x; s0 < t; s0 += blockDim.x) {
+  for (int s0 = threadIdx.x; s0 < n; s0 += blockDim.x) {
     const float w = __expf(s_scores[s0] - m);
     s_scores[s0] = w;
     local_sum += w;
@@ -529,15 +549,15 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
     if (threadIdx.x < r) s_red[threadIdx.x] += s_red[threadIdx.x + r];
     __syncthreads();
   }
-  const float inv_l = 1.0f / s_red[0];
+  const float l = s_red[0];
 
-  // pass 2: weighted V accumulation, 4-deep unroll for FMA-latency ILP
+  // pass 2: weighted V accumulation (FMA-only dependency, 2-deep unroll)
   float acc0[4] = {0.f, 0.f, 0.f, 0.f};
   float acc1[4] = {0.f, 0.f, 0.f, 0.f};
   int s0 = wave;
-  for (; s0 + waves < t; s0 += 2 * waves) {
-    const unsigned short* vrow_a = vbase + s0 * k_stride;
-    const unsigned short* vrow_b = vbase + (s0 + waves) * k_stride;
+  for (; s0 + waves < n; s0 += 2 * waves) {
+    const unsigned short* vrow_a = vbase + (s_begin + s0) * k_stride;
+    const unsigned short* vrow_b = vbase + (s_begin + s0 + waves) * k_stride;
     const float wa = s_scores[s0];
     const float wb = s_scores[s0 + waves];
     for (int j = 0; j < per_lane; ++j) {
@@ -545,8 +565,8 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
       acc1[j] = fmaf(wb, bf16_to_f32(vrow_b[lane * per_lane + j]), acc1[j]);
     }
   }
-  if (s0 < t) {
-    const unsigned short* vrow = vbase + s0 * k_stride;
+  if (s0 < n) {
+    const unsigned short* vrow = vbase + (s_begin + s0) * k_stride;
     const float w = s_scores[s0];
     for (int j = 0; j < per_lane; ++j) {
       acc0[j] = fmaf(w, bf16_to_f32(vrow[lane * per_lane + j]), acc0[j]);
@@ -557,29 +577,92 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
   }
   __syncthreads();
   if (wave == 0) {
-    for (int j = 0; j < per_lane; ++j) {
-      float v_out = 0.0f;
-      for (int wv = 0; wv < waves; ++wv) {
-        v_out += s_wacc[wv][lane * per_lane + j];
+    if (chunks == 1) {
+      const float inv_l = 1.0f / l;
+      for (int j = 0; j < per_lane; ++j) {
+        float v_out = 0.0f;
+        for (int wv = 0; wv < waves; ++wv) {
+          v_out += s_wacc[wv][lane * per_lane + j];
+        }
+        out[qh * hd + lane * per_lane + j] = f32_to_bf16(v_out * inv_l);
       }
-      out[qh * hd + lane * per_lane + j] = f32_to_bf16(v_out * inv_l);
+    } else {
+      float* p = partials + (static_cast<long long>(qh) * chunks + chunk) *
+                                (hd + 2);
+      for (int j = 0; j < per_lane; ++j) {
+        float v_out = 0.0f;
+        for (int wv = 0; wv < waves; ++wv) {
+          v_out += s_wacc[wv][lane * per_lane + j];
+        }
+        p[lane * per_lane + j] = v_out;
+      }
+      if (lane == 0) {
+        p[hd] = m;
+        p[hd + 1] = l;
+      }
     }
+  }
+}
+
+// Merge chunk partials: one wave per q head.
+__global__ __launch_bounds__(64) void attn_decode_combine_kernel(
+    const float* __restrict__ partials,  // [qH, chunks, hd + 2]
+    unsigned short* __restrict__ out,    // [qH, hd]
+    int chunks, int hd) {
+  const int qh = blockIdx.x;
+  const int lane = threadIdx.x;
+  const int per_lane = hd / 64;
+  const float* base = partials + static_cast<long long>(qh) * chunks * (hd + 2);
+  float m_tot = -1e30f;
+  for (int c = 0; c < chunks; ++c) m_tot = fmaxf(m_tot, base[c * (hd + 2) + hd]);
+  float l_tot = 0.0f;
+  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int c = 0; c < chunks; ++c) {
+    const float* p = base + c * (hd + 2);
+    const float alpha = __expf(p[hd] - m_tot);
+    l_tot += p[hd + 1] * alpha;
+    for (int j = 0; j < per_lane; ++j) {
+      acc[j] += p[lane * per_lane + j] * alpha;
+    }
+  }
+  const float inv = 1.0f / l_tot;
+  for (int j = 0; j < per_lane; ++j) {
+    out[qh * hd + lane * per_lane + j] = f32_to_bf16(acc[j] * inv);
   }
 }
 
 }  // namespace
 
+extern "C" int fma_attn_decode_chunks(int t, int q_heads) {
+  // fill the chip: aim for >= 256 blocks, chunk >= 256 positions, and
+  // keep per-chunk scores within the LDS window
+  int chunks = 256 / q_heads;
+  if (chunks < 1) chunks = 1;
+  const int max_by_span = (t + 255) / 256;
+  if (chunks > max_by_span) chunks = max_by_span;
+  const int min_by_lds = (t + FMA_ATTN_MAX_T / 4 - 1) / (FMA_ATTN_MAX_T / 4);
+  if (chunks < min_by_lds) chunks = min_by_lds;
+  return chunks;
+}
+
 extern "C" hipError_t fma_launch_attn_decode_bf16(
     const void* q, const void* K, const void* V, void* out, int t,
     int q_heads, int kv_heads, int hd, long long k_stride,
-    hipStream_t stream) {
+    float* partials, int chunks, hipStream_t stream) {
   if (hd > 256 || (hd & 63) != 0) return hipErrorInvalidValue;
   if (q_heads % kv_heads != 0) return hipErrorInvalidValue;
   if (t > FMA_ATTN_MAX_T) return hipErrorInvalidValue;
- hipLaunchKernelGGL(( attn_decode_bf16_kernel), dim3(q_heads), dim3(256), 0, stream, 
+  if (chunks > 1 && partials == nullptr) return hipErrorInvalidValue;
+  dim3 grid(q_heads, chunks);
+ hipLaunchKernelGGL(( attn_decode_bf16_kernel), dim3(grid), dim3(256), 0, stream, 
       static_cast<const unsigned short*>(q),
       static_cast<const unsigned short*>(K),
       static_cast<const unsigned short*>(V),
-      static_cast<unsigned short*>(out), t, q_heads, kv_heads, hd, k_stride);
+      static_cast<unsigned short*>(out), partials, t, q_heads, kv_heads, hd,
+      k_stride);
+  if (chunks > 1) {
+   hipLaunchKernelGGL(( attn_decode_combine_kernel), dim3(q_heads), dim3(64), 0, stream, 
+        partials, static_cast<unsigned short*>(out), chunks, hd);
+  }
   return hipGetLastError();
 }
