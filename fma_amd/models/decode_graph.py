@@ -142,14 +142,25 @@ class StaticDecoder:
 
     @torch.no_grad()
     def prefill(self, prompt: torch.Tensor) -> None:
-        """Run the prompt through the step loop (eager or graph)."""
+        """Batched eager prefill into the decoder's cache; ends in the same
+        state the step loop would reach (pos=T, first prediction staged)."""
+        from types import SimpleNamespace
+
         B, T = prompt.shape
         assert B == self.batch and T < self.max_seq
         self.pos.zero_()
         self.cache.zero_()
         self.out_tokens.zero_()
-        for t in range(T):
-            self.input_tok.copy_(prompt[:, t:t + 1])
+        if T > 1:
+            cache_view = SimpleNamespace(data=self.cache)
+            logits = self.model.forward(prompt, cache_view, 0)
+            nxt = logits[:, -1].argmax(-1, keepdim=True)
+            self.pos.fill_(T)
+            self.out_tokens[:, T:T + 1] = nxt
+            self.input_tok.copy_(nxt)
+            self.logits.copy_(logits[:, -1])
+        else:
+            self.input_tok.copy_(prompt[:, 0:1])
             self._one()
 
     def _one(self) -> None:
