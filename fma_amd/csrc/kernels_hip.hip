@@ -651,7 +651,10 @@ extern "C" hipError_t fma_launch_attn_decode_bf16(
     float* partials, int chunks, hipStream_t stream) {
   if (hd > 256 || (hd & 63) != 0) return hipErrorInvalidValue;
   if (q_heads % kv_heads != 0) return hipErrorInvalidValue;
-  if (t > FMA_ATTN_MAX_T) return hipErrorInvalidValue;
+  // any t: the chunk heuristic bounds each chunk's scores to the LDS
+  // window (FMA_ATTN_MAX_T/4)
+  if ((t + chunks - 1) / chunks > FMA_ATTN_MAX_T / 4)
+    return hipErrorInvalidValue;
   if (chunks > 1 && partials == nullptr) return hipErrorInvalidValue;
   dim3 grid(q_heads, chunks);
  hipLaunchKernelGGL(( attn_decode_bf16_kernel), dim3(grid), dim3(256), 0, stream, 

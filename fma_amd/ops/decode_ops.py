@@ -73,7 +73,7 @@ def fast_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
     if not available():
         return None
     _, _, qh, hd = q.shape
-    if (hd % 64 != 0 or hd > 256 or t > 8192 or not q.is_cuda
+    if (hd % 64 != 0 or hd > 256 or t > (1 << 17) or not q.is_cuda
             or q.dtype != torch.bfloat16
             or qh % k_cache.shape[1] != 0):
         return None
