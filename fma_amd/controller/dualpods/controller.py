@@ -332,9 +332,31 @@ class DualPodsController:
         uid, name = ob.uid_of(requester), ob.name_of(requester)
         provider = self._find_provider_for(uid, name)
         if provider is None:
+            if self._was_bound(requester):
+                # the provider vanished while bound: mirror the deletion so
+                # the requester's owner re-creates it (reference
+                # inference-server.go:257-290, docs/dual-pods.md:750-769)
+                return self._mirror_provider_deletion(requester)
             return self._bind_to_launcher(node, requester, sdata, isc, lc_name)
         return self._run_bound_launcher(node, requester, provider, sdata,
                                         isc)
+
+    def _was_bound(self, requester: Dict[str, Any]) -> bool:
+        """The dual label on the requester is the durable record that a
+        binding existed (survives controller restarts)."""
+        return contracts.DUAL_LABEL in ob.labels_of(requester)
+
+    def _mirror_provider_deletion(self, requester: Dict[str, Any]) -> bool:
+        self.http.request(
+            "POST", self._stub_url(requester) + contracts.BECOME_UNREADY_PATH,
+            purpose="become-unready")
+        try:
+            self.store.delete("Pod", ob.name_of(requester), self.ns,
+                              actor="dual-pods-controller",
+                              expect_uid=ob.uid_of(requester))
+        except (NotFound, Conflict):
+            pass
+        return RETRY  # continue into the deletion flow (finalizer release)
 
     def _launcher_views(self, pods: List[Dict[str, Any]], sdata: ServerData
                         ) -> Tuple[List[LauncherView], bool]:
@@ -676,6 +698,8 @@ class DualPodsController:
                           sdata: ServerData) -> bool:
         uid, name = ob.uid_of(requester), ob.name_of(requester)
         provider = self._find_provider_for(uid, name)
+        if provider is None and self._was_bound(requester):
+            return self._mirror_provider_deletion(requester)
         patch = ob.annotations_of(requester)[contracts.SERVER_PATCH_ANNOTATION]
         indices = self._gpu_indices(node, sdata.gpus or [])
         if indices is None:

@@ -167,3 +167,33 @@ def test_same_node_collision_two_requesters_one_slot():
                 for p in w["store"].list("Pod")
                 if ob.annotations_of(p).get(C.REQUESTER_ANNOTATION)]
     assert len(bindings) == len(set(bindings))
+
+
+def test_exogenous_provider_deletion_mirrors_to_requester():
+    """Bound provider vanishes -> the requester is deleted so its owner
+    re-creates it (reference inference-server.go:257-290)."""
+    w = mk_world()
+    store = w["store"]
+    drive(w["ctl"], infsvr_item(store))
+    req = store.get("Pod", "req1")
+    assert C.DUAL_LABEL in ob.labels_of(req)
+
+    # force-remove the bound launcher (finalizer cleared, then delete)
+    lp = store.get("Pod", "launcher1")
+    ob.finalizers_of(lp).clear()
+    store.update(lp, actor="dual-pods-controller")
+    store.delete("Pod", "launcher1", actor="system")
+    assert store.try_get("Pod", "launcher1") is None
+
+    drive(w["ctl"], infsvr_item(store))
+    assert store.try_get("Pod", "req1") is None, \
+        "requester should mirror the provider's deletion"
+    # and NO replacement launcher was silently re-bound for it
+    assert w["stub"].unready_calls >= 1
+
+
+def test_unbound_requester_not_mirrored():
+    """A requester that was never bound cold-starts instead of dying."""
+    w = mk_world(with_launcher=False)
+    w["ctl"]._process(infsvr_item(w["store"]))
+    assert w["store"].try_get("Pod", "req1") is not None
