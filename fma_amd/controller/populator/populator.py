@@ -308,6 +308,8 @@ class LauncherPopulator:
                 deleting.append(pod)
             elif ob.annotations_of(pod).get(contracts.REQUESTER_ANNOTATION):
                 bound.append(pod)
+            elif ob.pod_phase(pod) == "Failed":
+                stale.append(pod)  # crashed unbound launcher: replace it
             elif ob.annotations_of(pod).get(
                     contracts.LAUNCHER_TEMPLATE_HASH_ANNOTATION) != \
                     lc_digest.template_hash:

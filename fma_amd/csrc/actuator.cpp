@@ -848,7 +848,8 @@ class DeviceArena {
 // without extra synchronization). fp32 out by default; out_bf16 fuses the
 // down-convert into the store (saves one elementwise launch per linear).
 at::Tensor gemv_bf16(const at::Tensor& W, const at::Tensor& x,
-                     bool out_bf16) {
+                     bool out_bf16,
+                     const c10::optional<at::Tensor>& residual) {
   TORCH_CHECK(W.is_cuda() && x.is_cuda(), "gemv_bf16 needs GPU tensors");
   TORCH_CHECK(W.scalar_type() == at::kBFloat16 &&
               x.scalar_type() == at::kBFloat16, "gemv_bf16 is bf16-only");
@@ -858,12 +859,20 @@ at::Tensor gemv_bf16(const at::Tensor& W, const at::Tensor& x,
   TORCH_CHECK((K & 7) == 0, "K must be a multiple of 8");
   auto stream = c10::hip::getCurrentHIPStream(W.device().index());
   if (out_bf16) {
+    const void* rptr = nullptr;
+    if (residual.has_value()) {
+      const auto& r = residual.value();
+      TORCH_CHECK(r.scalar_type() == at::kBFloat16 && r.numel() == M &&
+                  r.is_contiguous(), "residual must be contiguous bf16 [M]");
+      rptr = r.data_ptr();
+    }
     auto y = at::empty({M}, W.options());
     FMA_HIP_CHECK(fma_launch_gemv_bf16_out16(
-        W.data_ptr(), x.data_ptr(), y.data_ptr(),
+        W.data_ptr(), x.data_ptr(), y.data_ptr(), rptr,
         static_cast<int>(M), static_cast<int>(K), stream.stream()));
     return y;
   }
+  TORCH_CHECK(!residual.has_value(), "residual needs out_bf16=True");
   auto y = at::empty({M}, W.options().dtype(at::kFloat));
   FMA_HIP_CHECK(fma_launch_gemv_bf16(
       W.data_ptr(), x.data_ptr(), y.data_ptr<float>(),
@@ -975,7 +984,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("mode") = 0, py::arg("chunk_bytes") = 0);
   m.def("device_supports_vmm", &device_supports_vmm, py::arg("device"));
   m.def("gemv_bf16", &gemv_bf16, "Batch-1 bf16 GEMV",
-        py::arg("W"), py::arg("x"), py::arg("out_bf16") = false);
+        py::arg("W"), py::arg("x"), py::arg("out_bf16") = false,
+        py::arg("residual") = py::none());
   m.def("rmsnorm1_bf16", &rmsnorm1_bf16, py::arg("x"), py::arg("w"),
         py::arg("eps"));
   m.def("silu_mul_bf16", &silu_mul_bf16, py::arg("g"), py::arg("u"));

@@ -38,7 +38,8 @@ extern "C" hipError_t fma_launch_gemv_bf16(const void* W, const void* x,
                                            hipStream_t stream);
 
 extern "C" hipError_t fma_launch_gemv_bf16_out16(const void* W, const void* x,
-                                                 void* y, int M, int K,
+                                                 void* y, const void* residual,
+                                                 int M, int K,
                                                  hipStream_t stream);
 extern "C" hipError_t fma_launch_rmsnorm1_bf16(const void* x, const void* w,
                                                void* y, int H, float eps,

@@ -220,6 +220,7 @@ __global__ __launch_bounds__(256) void gemv_bf16_kernel(
     const unsigned short* __restrict__ W,  // [M, K] row-major bf16
     const unsigned short* __restrict__ x,  // [K] bf16
     OutT* __restrict__ y,                  // [M] fp32 or bf16
+    const unsigned short* __restrict__ r,  // optional residual [M] bf16
     int M, int K) {
   // x staged in LDS when it fits without hurting occupancy; for wide K
   // (w_down shapes) every wave reads the same x slices, which the L2
@@ -262,7 +263,10 @@ __global__ __launch_bounds__(256) void gemv_bf16_kernel(
     for (int off = 32; off > 0; off >>= 1) {
       acc += __shfl_down(acc, off, 64);
     }
-    if (lane == 0) gemv_store(y, row, acc);
+    if (lane == 0) {
+      if (r != nullptr) acc += bf16_to_f32(r[row]);
+      gemv_store(y, row, acc);
+    }
   }
 }
 
@@ -281,11 +285,11 @@ extern "C" hipError_t fma_launch_gemv_bf16(const void* W, const void* x,
   if (lds <= 32 * 1024) {  // >= 5 blocks/CU with x staged
     gemv_bf16_kernel<true, float><<<blocks, block, lds, stream>>>(
         static_cast<const unsigned short*>(W),
-        static_cast<const unsigned short*>(x), y, M, K);
+        static_cast<const unsigned short*>(x), y, nullptr, M, K);
   } else {
     gemv_bf16_kernel<false, float><<<blocks, block, 0, stream>>>(
         static_cast<const unsigned short*>(W),
-        static_cast<const unsigned short*>(x), y, M, K);
+        static_cast<const unsigned short*>(x), y, nullptr, M, K);
   }
   return hipGetLastError();
 }
@@ -383,7 +387,8 @@ __global__ __launch_bounds__(256) void rope1_bf16_kernel(
 }  // namespace
 
 extern "C" hipError_t fma_launch_gemv_bf16_out16(const void* W, const void* x,
-                                                 void* y, int M, int K,
+                                                 void* y, const void* residual,
+                                                 int M, int K,
                                                  hipStream_t stream) {
   if ((K & 7) != 0) return hipErrorInvalidValue;
   const int block = 256;
@@ -395,12 +400,14 @@ extern "C" hipError_t fma_launch_gemv_bf16_out16(const void* W, const void* x,
     gemv_bf16_kernel<true, unsigned short><<<blocks, block, lds, stream>>>(
         static_cast<const unsigned short*>(W),
         static_cast<const unsigned short*>(x),
-        static_cast<unsigned short*>(y), M, K);
+        static_cast<unsigned short*>(y),
+        static_cast<const unsigned short*>(residual), M, K);
   } else {
     gemv_bf16_kernel<false, unsigned short><<<blocks, block, 0, stream>>>(
         static_cast<const unsigned short*>(W),
         static_cast<const unsigned short*>(x),
-        static_cast<unsigned short*>(y), M, K);
+        static_cast<unsigned short*>(y),
+        static_cast<const unsigned short*>(residual), M, K);
   }
   return hipGetLastError();
 }

@@ -29,7 +29,7 @@ import torch.nn.functional as F
 
 from fma_amd.ops.decode_ops import (fast_attn_decode, fast_rmsnorm,
                                     fast_rope1, fast_silu_mul)
-from fma_amd.ops.linear import fast_linear
+from fma_amd.ops.linear import fast_linear, fast_linear_residual
 
 
 @dataclass
@@ -282,7 +282,11 @@ class LlamaModel:
                 att = F.scaled_dot_product_attention(
                     qh, kh, vh, is_causal=(T > 1))
                 att = att.transpose(1, 2).reshape(B, T, q_heads * hd)
-            x = x + self._maybe_all_reduce(fast_linear(att, P[p + "wo.weight"]))
+            if decode1 and self.tp_size == 1:
+                x = fast_linear_residual(att, P[p + "wo.weight"], x)
+            else:
+                x = x + self._maybe_all_reduce(
+                    fast_linear(att, P[p + "wo.weight"]))
 
             h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps) \
                 if decode1 else rmsnorm(x, P[p + "mlp_norm.weight"],
@@ -290,8 +294,11 @@ class LlamaModel:
             gate = fast_linear(h, P[p + "w_gate.weight"])
             up = fast_linear(h, P[p + "w_up.weight"])
             act = fast_silu_mul(gate, up) if decode1 else F.silu(gate) * up
-            x = x + self._maybe_all_reduce(
-                fast_linear(act, P[p + "w_down.weight"]))
+            if decode1 and self.tp_size == 1:
+                x = fast_linear_residual(act, P[p + "w_down.weight"], x)
+            else:
+                x = x + self._maybe_all_reduce(
+                    fast_linear(act, P[p + "w_down.weight"]))
 
         x = fast_rmsnorm(x, P["final_norm.weight"], cfg.norm_eps) \
             if decode1 else rmsnorm(x, P["final_norm.weight"], cfg.norm_eps)

@@ -39,3 +39,18 @@ def fast_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
                                    True)  # bf16 out, convert fused
         return y.view(*x.shape[:-1], w.shape[0])
     return F.linear(x, w)
+
+
+def fast_linear_residual(x: torch.Tensor, w: torch.Tensor,
+                         residual: torch.Tensor) -> torch.Tensor:
+    """residual + F.linear(x, w) with the add fused into the GEMV store."""
+    if (gemv_enabled() and x.is_cuda
+            and w.dtype == torch.bfloat16 and x.dtype == torch.bfloat16
+            and x.numel() == x.shape[-1]
+            and residual.numel() == w.shape[0]
+            and (w.shape[1] & 7) == 0
+            and w.is_contiguous()):
+        y = actuation._C.gemv_bf16(w, x.reshape(-1).contiguous(), True,
+                                   residual.reshape(-1).contiguous())
+        return y.view(residual.shape)
+    return residual + F.linear(x, w)

@@ -383,3 +383,13 @@ def test_decode_with_fused_attention_matches_nonfused():
         del os.environ["FMA_DISABLE_FUSED_OPS"]
         dops._ENABLED = None
     assert torch.equal(fused, plain)
+
+
+def test_gemv_residual_fusion_matches():
+    import fma_amd._C as C
+    W = torch.randn(512, 256, dtype=torch.bfloat16, device="cuda:0")
+    x = torch.randn(256, dtype=torch.bfloat16, device="cuda:0")
+    r = torch.randn(512, dtype=torch.bfloat16, device="cuda:0")
+    fused = C.gemv_bf16(W, x, True, r)
+    ref = (r.float() + W.float() @ x.float())
+    assert torch.allclose(fused.float(), ref, atol=3e-2, rtol=3e-2)

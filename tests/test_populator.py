@@ -244,3 +244,20 @@ def test_lc_deletion_garbage_collects_launchers():
     store.delete("LauncherConfig", "lc1")
     # owner-reference GC in the store removes the launcher pods
     assert launcher_pods(store) == []
+
+
+def test_failed_launcher_replaced():
+    store = MemStore()
+    mk_node(store, "n1", labels={"gpu": "x"})
+    mk_lc(store)
+    mk_lpp(store, "p1", count=1, match_labels={"gpu": "x"})
+    pop = mk_pop(store)
+    drain_key(pop, ("n1", "lc1"))
+    pod = launcher_pods(store, "n1")[0]
+    cur = store.get("Pod", ob.name_of(pod))
+    cur.setdefault("status", {})["phase"] = "Failed"
+    store.update(cur, actor="node-agent", subresource="status")
+    drain_key(pop, ("n1", "lc1"))
+    pods = launcher_pods(store, "n1")
+    assert len(pods) == 1
+    assert ob.uid_of(pods[0]) != ob.uid_of(pod), "failed launcher not replaced"
