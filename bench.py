@@ -42,8 +42,8 @@ def main() -> None:
     ap.add_argument("--gib", type=float, default=64.0,
                     help="total parameter GiB across all ranks")
     ap.add_argument("--mode", choices=["arena", "pack"], default="arena")
-    ap.add_argument("--pack-xfer", type=int, default=None,
-                    help="pack transfer mode: 0 staged, 1 direct-kernel, 2 per-tensor")
+    ap.add_argument("--pack-xfer", type=int, default=0,
+                    help="pack transfer mode: 0 staged-kernel (default), 1 direct-kernel, 2 per-tensor, -1 auto")
     ap.add_argument("--chunk-mb", type=int, default=0)
     ap.add_argument("--nstreams", type=int, default=1)
     ap.add_argument("--slab-mb", type=int, default=0, help="slab size MiB (0=default 1 GiB)")
@@ -77,7 +77,8 @@ def main() -> None:
         use_vmm=args.vmm or None, chunk_bytes=args.chunk_mb << 20, seed=1234,
         nstreams=args.nstreams,
         slab_bytes=(args.slab_mb << 20) if args.slab_mb else None,
-        actuation_mode=args.mode, pack_xfer_mode=args.pack_xfer)
+        actuation_mode=args.mode,
+        pack_xfer_mode=None if args.pack_xfer < 0 else args.pack_xfer)
     log(f"[rank {rank}] engine up: {eng.total_bytes/2**30:.2f} GiB/rank, "
         f"{cfg.num_layers} layers, vmm={eng.stats()['uses_vmm']}, "
         f"create {time.perf_counter()-t0:.1f}s")

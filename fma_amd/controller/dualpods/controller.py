@@ -28,6 +28,8 @@ import time
 from dataclasses import dataclass, field
 from typing import Any, Dict, List, Optional, Set, Tuple
 
+import logging
+
 from fma_amd.api import contracts
 from fma_amd.controller import metrics
 from fma_amd.controller.dualpods import nominal as nominal_mod
@@ -42,6 +44,8 @@ from fma_amd.store.memstore import Conflict, MemStore, NotFound
 
 REQUESTER_FINALIZER = "dual-pods.llm-d.ai/requester-protection"
 PROVIDER_FINALIZER = "dual-pods.llm-d.ai/provider-protection"
+
+logger = logging.getLogger("fma.dualpods")
 
 RETRY = True
 DONE = False
@@ -464,6 +468,8 @@ class DualPodsController:
                               expect_rv=ob.rv_of(cur))
         except Conflict:
             return RETRY
+        logger.info("bound requester %s to launcher %s (instance %s)",
+                    name, ob.name_of(cur), sdata.instance_id)
         # FYI labels on the requester (reference :1431-1484)
         self._apply_requester_fyi(requester, ob.name_of(cur),
                                   sdata.instance_id)
@@ -539,6 +545,10 @@ class DualPodsController:
                                    purpose="wake")
             if not wr.ok:
                 return RETRY
+            logger.info("woke instance %s on %s (%.3fs)", sdata.instance_id,
+                        ob.name_of(provider),
+                        float(wr.body.get("seconds", 0) or 0)
+                        if isinstance(wr.body, dict) else 0.0)
         self._apply_bound_labels(provider, isc)
         sdata.instance_last_used[sdata.instance_id] = self.clock.time()
         self._relay_instance_log(requester, provider, sdata)
@@ -880,6 +890,8 @@ class DualPodsController:
             return RETRY
         except NotFound:
             pass
+        logger.info("unbound provider %s (instance slept or deleted)",
+                    ob.name_of(cur))
         if requester is not None:
             metrics.duality().labels(
                 ob.name_of(requester), ob.name_of(cur),
