@@ -274,3 +274,53 @@ def test_controller_restart_recovers_live_binding(cluster):
                  desc="req9 hot start via recovered controller")
     finally:
         ctl2.stop()
+
+
+def test_four_concurrent_actuations_on_one_node(cluster):
+    """Config #5 shape at the control-plane level: the populator holds 4
+    launchers on the node; 4 requesters with 4 different ISCs all reach
+    Ready concurrently."""
+    store, agent = cluster["store"], cluster["agent"]
+    store.create(ob.new_object(
+        "LauncherConfig", "lc1",
+        spec={"maxInstances": 2, "podTemplate": {"spec": {"containers": [{
+            "name": "launcher",
+            "command": [sys.executable, "-m", "fma_amd.launcher.service"],
+        }]}}}))
+    for i in range(4):
+        store.create(ob.new_object(
+            "InferenceServerConfig", f"isc-c{i}",
+            spec={"modelServerConfig": {
+                "port": 8380 + i, "options": "--model tiny"},
+                "launcherConfigName": "lc1"}))
+    store.create(ob.new_object(
+        "LauncherPopulationPolicy", "lpp4",
+        spec={"enhancedNodeSelector": {"labelSelector": {}},
+              "countForLauncher": [
+                  {"launcherConfigName": "lc1", "launcherCount": 4}]}))
+
+    wait_for(lambda: sum(
+        1 for p in store.list("Pod")
+        if ob.labels_of(p).get(C.COMPONENT_LABEL) == C.LAUNCHER_COMPONENT
+        and ob.pod_is_ready(p)) >= 4, 90, desc="4 launchers ready")
+
+    for i in range(4):
+        pod = ob.new_object(
+            "Pod", f"creq{i}",
+            annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: f"isc-c{i}"},
+            spec={"nodeName": "node-a", "containers": [{
+                "name": "requester",
+                "command": [sys.executable, "-m",
+                            "fma_amd.requester.server"]}]})
+        store.create(pod, actor="user")
+
+    for i in range(4):
+        wait_for(lambda i=i: requester_ready(store, agent, f"creq{i}"), 120,
+                 desc=f"creq{i} ready")
+    # four distinct launchers are bound
+    bound = [ob.annotations_of(p).get(C.REQUESTER_ANNOTATION)
+             for p in store.list("Pod")
+             if ob.labels_of(p).get(C.COMPONENT_LABEL) ==
+             C.LAUNCHER_COMPONENT
+             and ob.annotations_of(p).get(C.REQUESTER_ANNOTATION)]
+    assert len(bound) == 4 and len(set(bound)) == 4
