@@ -1,0 +1,84 @@
+"""Whole-product smoke: deploy.py + CLI + manifests as real processes."""
+
+import os
+import socket
+import subprocess
+import sys
+import time
+
+import httpx
+import pytest
+
+pytestmark = pytest.mark.timeout(180)
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def free_port():
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def test_deploy_cli_end_to_end(tmp_path):
+    port = free_port()
+    env = dict(os.environ, PYTHONPATH=ROOT, FMA_FAKE_GPU="1",
+               FMA_GPU_MODE="naive", FMA_ACCELERATORS="GPU-0")
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "fma_amd.deploy", "--node-name", "node-1",
+         "--store-port", str(port), "--metrics-port", str(free_port()),
+         "--gpus", "2"],
+        env=env, cwd=ROOT, stdout=open(tmp_path / "deploy.log", "wb"),
+        stderr=subprocess.STDOUT, start_new_session=True)
+    base = f"http://127.0.0.1:{port}"
+    try:
+        deadline = time.time() + 60
+        up = False
+        while time.time() < deadline:
+            try:
+                if httpx.get(base + "/healthz", timeout=2).status_code == 200:
+                    up = True
+                    break
+            except httpx.HTTPError:
+                time.sleep(0.3)
+        assert up, open(tmp_path / "deploy.log").read().decode("utf-8",
+                                                               "replace")
+
+        def cli(*args):
+            return subprocess.run(
+                [sys.executable, "-m", "fma_amd.cli", "--store-url", base,
+                 *args],
+                env=env, cwd=ROOT, capture_output=True, text=True, timeout=60)
+
+        r = cli("apply", "-f", "manifests/example.yaml")
+        assert r.returncode == 0, r.stderr
+        assert "created" in r.stdout
+
+        # requester becomes Ready through the whole stack
+        deadline = time.time() + 90
+        ready = False
+        while time.time() < deadline:
+            r = cli("get", "pod", "my-model-request", "-o", "json")
+            if r.returncode == 0 and '"type": "Ready", "status": "True"' in \
+                    r.stdout.replace("'", '"'):
+                ready = True
+                break
+            if r.returncode == 0 and '"status": "True"' in r.stdout and \
+                    '"Ready"' in r.stdout:
+                ready = True
+                break
+            time.sleep(0.5)
+        assert ready, cli("get", "pods").stdout + \
+            open(tmp_path / "deploy.log").read().decode("utf-8", "replace")[-2000:]
+
+        r = cli("get", "pods")
+        assert "my-model-request" in r.stdout
+        r = cli("delete", "pod", "my-model-request")
+        assert r.returncode == 0
+    finally:
+        import signal
+        try:
+            os.killpg(proc.pid, signal.SIGTERM)
+        except ProcessLookupError:
+            pass
+        proc.wait(timeout=15)
