@@ -393,3 +393,23 @@ def test_gemv_residual_fusion_matches():
     fused = C.gemv_bf16(W, x, True, r)
     ref = (r.float() + W.float() @ x.float())
     assert torch.allclose(fused.float(), ref, atol=3e-2, rtol=3e-2)
+
+
+def test_no_hbm_leak_across_engine_lifecycles():
+    """Create/sleep/wake/destroy engines repeatedly: free HBM returns to
+    the starting level (guards arena/staging/pinned leak paths)."""
+    from fma_amd.models.llama import LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+    import fma_amd._C as C
+    import gc
+    free0, _ = C.device_mem_info(0)
+    for i in range(6):
+        eng = ActuationEngine(LlamaConfig.from_total_gib(1), 0, seed=i)
+        eng.sleep()
+        eng.wake_up()
+        del eng
+        gc.collect()
+    torch.cuda.empty_cache()
+    free1, _ = C.device_mem_info(0)
+    assert free0 - free1 < (256 << 20), \
+        f"leaked {(free0-free1)/2**20:.0f} MiB across lifecycles"
