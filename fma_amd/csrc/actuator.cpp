@@ -962,6 +962,32 @@ at::Tensor attn_decode_bf16(const at::Tensor& q, const at::Tensor& k,
   return out;
 }
 
+// Causal GQA prefill attention on MFMA: Q [T, qH, hd] against the
+// [S, kvH, hd] cache slices holding keys [0, pos0 + T).
+at::Tensor attn_prefill_bf16(const at::Tensor& q, const at::Tensor& k,
+                             const at::Tensor& v, int64_t pos0) {
+  TORCH_CHECK(q.dim() == 3 && q.is_cuda() &&
+              q.scalar_type() == at::kBFloat16 && q.is_contiguous(),
+              "q must be contiguous bf16 [T, qH, hd]");
+  TORCH_CHECK(k.dim() == 3 && k.is_contiguous() && v.is_contiguous(),
+              "k/v must be contiguous [S, kvH, hd]");
+  TORCH_CHECK(k.scalar_type() == at::kBFloat16 &&
+              v.scalar_type() == at::kBFloat16, "bf16 only");
+  const int64_t T = q.size(0), q_heads = q.size(1), hd = q.size(2);
+  const int64_t kv_heads = k.size(1);
+  TORCH_CHECK(hd == 64 || hd == 128, "hd must be 64 or 128");
+  TORCH_CHECK(k.size(2) == hd && v.sizes() == k.sizes(), "shape mismatch");
+  TORCH_CHECK(q_heads % kv_heads == 0, "q_heads % kv_heads != 0");
+  TORCH_CHECK(pos0 >= 0 && pos0 + T <= k.size(0), "keys out of cache");
+  auto out = at::empty_like(q);
+  auto stream = c10::hip::getCurrentHIPStream(q.device().index());
+  FMA_HIP_CHECK(fma_launch_attn_prefill_bf16(
+      q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+      static_cast<int>(T), static_cast<int>(pos0), static_cast<int>(q_heads),
+      static_cast<int>(kv_heads), static_cast<int>(hd), stream.stream()));
+  return out;
+}
+
 std::tuple<int64_t, int64_t> device_mem_info(int device) {
   FMA_HIP_CHECK(hipSetDevice(device));
   size_t free_b = 0, total_b = 0;
@@ -989,6 +1015,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm1_bf16", &rmsnorm1_bf16, py::arg("x"), py::arg("w"),
         py::arg("eps"));
   m.def("silu_mul_bf16", &silu_mul_bf16, py::arg("g"), py::arg("u"));
+  m.def("attn_prefill_bf16", &attn_prefill_bf16, py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("pos0"));
   m.def("attn_decode_bf16", &attn_decode_bf16, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("t"));
   m.def("rope1_bf16_", &rope1_bf16_, py::arg("q"), py::arg("cos_row"),

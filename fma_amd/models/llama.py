@@ -27,8 +27,8 @@ import torch
 import torch.distributed as dist
 import torch.nn.functional as F
 
-from fma_amd.ops.decode_ops import (fast_attn_decode, fast_rmsnorm,
-                                    fast_rope1, fast_silu_mul)
+from fma_amd.ops.decode_ops import (fast_attn_decode, fast_attn_prefill,
+                                    fast_rmsnorm, fast_rope1, fast_silu_mul)
 from fma_amd.ops.linear import fast_linear, fast_linear_residual
 
 
@@ -270,6 +270,15 @@ class LlamaModel:
                 # repeat_interleave copies, one launch)
                 att = fast_attn_decode(q, cache.data[li, 0, 0],
                                        cache.data[li, 1, 0], start_pos + 1)
+            if att is None and B == 1 and T > 1 and x.is_cuda \
+                    and x.dtype == torch.bfloat16:
+                # causal GQA prefill on MFMA matrix cores (one launch,
+                # reads the cache layout directly)
+                if cache is not None:
+                    att = fast_attn_prefill(q, cache.data[li, 0, 0],
+                                            cache.data[li, 1, 0], start_pos)
+                elif start_pos == 0:
+                    att = fast_attn_prefill(q, k[0], v[0], 0)
             if att is None:
                 # SDPA wants [B, heads, T, hd]
                 qh = q.transpose(1, 2)

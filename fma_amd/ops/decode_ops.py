@@ -66,6 +66,24 @@ def fast_rope1(x: torch.Tensor, cos_row: torch.Tensor, sin_row: torch.Tensor
     return out.view(B, T, heads, hd).to(x.dtype)
 
 
+def fast_attn_prefill(q: torch.Tensor, k_cache: torch.Tensor,
+                      v_cache: torch.Tensor, pos0: int):
+    """Causal GQA prefill on MFMA matrix cores. q [1,T,qH,hd]; k/v_cache
+    [S,kvH,hd] holding keys [0, pos0+T). Returns [1,T,qH*hd] or None when
+    the kernel does not apply (caller falls back to SDPA)."""
+    if not available() or os.environ.get("FMA_DISABLE_MFMA_PREFILL") == "1":
+        return None
+    b, t, qh, hd = q.shape
+    if (b != 1 or hd not in (64, 128) or not q.is_cuda
+            or q.dtype != torch.bfloat16
+            or qh % k_cache.shape[1] != 0
+            or pos0 + t > k_cache.shape[0]):
+        return None
+    out = actuation._C.attn_prefill_bf16(
+        q.reshape(t, qh, hd).contiguous(), k_cache, v_cache, pos0)
+    return out.view(1, t, qh * hd)
+
+
 def fast_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                      v_cache: torch.Tensor, t: int):
     """q [1,1,qH,hd]; k/v_cache [S,kvH,hd] (one layer, batch 1). Returns

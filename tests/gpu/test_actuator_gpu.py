@@ -413,3 +413,63 @@ def test_no_hbm_leak_across_engine_lifecycles():
     free1, _ = C.device_mem_info(0)
     assert free0 - free1 < (256 << 20), \
         f"leaked {(free0-free1)/2**20:.0f} MiB across lifecycles"
+
+
+def test_attn_prefill_mfma_matches_fp32_reference():
+    """MFMA prefill attention (causal, GQA) vs a plain fp32 reference,
+    including tail tiles (T % 32 != 0), a chunked-prefill offset (pos0>0)
+    and both supported head dims."""
+    import fma_amd._C as C
+    torch.manual_seed(13)
+    for (qH, kvH, hd, S, T, pos0) in (
+            (32, 8, 128, 256, 256, 0),   # full tiles, llama-8B shape
+            (16, 2, 128, 200, 100, 33),  # chunked prefill mid-stream
+            (8, 8, 64, 96, 33, 0),       # hd=64, tail tile
+            (4, 1, 128, 40, 1, 7),       # single query row (degenerate)
+    ):
+        q = torch.randn(T, qH, hd, dtype=torch.bfloat16, device="cuda:0")
+        k = torch.randn(S, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
+        v = torch.randn(S, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
+        out = C.attn_prefill_bf16(q, k, v, pos0)
+        # plain fp32 reference with an explicit causal mask
+        t_kv = pos0 + T
+        rep = qH // kvH
+        kf = k[:t_kv].float().permute(1, 0, 2).repeat_interleave(rep, 0)
+        vf = v[:t_kv].float().permute(1, 0, 2).repeat_interleave(rep, 0)
+        qf = q.float().permute(1, 0, 2)                    # [qH, T, hd]
+        scores = qf @ kf.transpose(1, 2) / (hd ** 0.5)     # [qH, T, t_kv]
+        qpos = pos0 + torch.arange(T, device="cuda:0")
+        kpos = torch.arange(t_kv, device="cuda:0")
+        scores.masked_fill_(kpos[None, None, :] > qpos[None, :, None],
+                            float("-inf"))
+        ref = (torch.softmax(scores, dim=-1) @ vf).permute(1, 0, 2)
+        assert torch.allclose(out.float(), ref, atol=3e-2, rtol=3e-2), \
+            (qH, kvH, hd, S, T, pos0,
+             (out.float() - ref).abs().max().item())
+
+
+def test_prefill_mfma_vs_sdpa_path_logits_close():
+    """Full-model prefill logits with the MFMA attention path vs the SDPA
+    path (bf16 rounding-level agreement), exercised through the real
+    cache-writing forward."""
+    import os
+    from fma_amd.models.llama import KVCache, LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+    cfg = LlamaConfig(name="pfx", vocab_size=512, hidden_size=512,
+                      intermediate_size=768, num_layers=2, num_heads=4,
+                      num_kv_heads=2, max_seq_len=128)
+    eng = ActuationEngine(cfg, 0, seed=31)
+    torch.manual_seed(3)
+    prompt = torch.randint(0, cfg.vocab_size, (1, 40), device="cuda:0")
+    cache = KVCache(cfg, 1, "cuda:0", 1, 128)
+    mfma = eng.model.forward(prompt, cache, 0).float()
+    os.environ["FMA_DISABLE_MFMA_PREFILL"] = "1"
+    try:
+        cache2 = KVCache(cfg, 1, "cuda:0", 1, 128)
+        sdpa = eng.model.forward(prompt, cache2, 0).float()
+    finally:
+        del os.environ["FMA_DISABLE_MFMA_PREFILL"]
+    assert torch.allclose(mfma, sdpa, atol=8e-2, rtol=8e-2), \
+        (mfma - sdpa).abs().max().item()
+    cache.free()
+    cache2.free()
