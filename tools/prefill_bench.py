@@ -6,9 +6,13 @@ full-model prefill (tiny model) with FMA_DISABLE_MFMA_PREFILL toggled.
 """
 
 import argparse
+import os
+import sys
 import time
 
-import torch
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch  # noqa: E402
 
 
 def bench_op(fn, iters):
