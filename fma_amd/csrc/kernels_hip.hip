@@ -712,12 +712,15 @@ __device__ inline float wave32_sum(float v) {
   return v;
 }
 
+template <int HD>
 __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
     const unsigned short* __restrict__ Q,  // [T, qH, hd]
     const unsigned short* __restrict__ K,  // [S, kvH, hd]
     const unsigned short* __restrict__ V,  // [S, kvH, hd]
     unsigned short* __restrict__ O,        // [T, qH, hd]
-    int T, int pos0, int q_heads, int kv_heads, int hd) {
+    int T, int pos0, int q_heads, int kv_heads) {
+  constexpr int kNblk = HD / 32;    // 2 (hd=64) or 4 (hd=128)
+  constexpr int kKsteps = HD / 16;  // QK^T k-steps
   const int qh = blockIdx.x;
   const int tile = blockIdx.y;
   const int r0 = tile * 32;
@@ -726,42 +729,45 @@ __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
   const int l = threadIdx.x;
   const int half = l >> 5;       // 0 or 1
   const int lane32 = l & 31;
-  const int nblk = hd / 32;      // 2 (hd=64) or 4 (hd=128)
-  const int ksteps = hd / 16;    // QK^T k-steps
-  const float scale = rsqrtf(static_cast<float>(hd));
+  const float scale = rsqrtf(static_cast<float>(HD));
 
   __shared__ __bf16 s_p[32 * 32];
-  __shared__ __bf16 s_v[32 * 128];
+  __shared__ __bf16 s_v[32 * HD];  // transposed: [hd][key]
 
   // Q fragments, kept in registers for the whole key loop.
   // A-layout: lane holds A[lane32][8*half + i] per 16-k step.
-  bf16x8_t qf[8];
+  bf16x8_t qf[kKsteps];
   const int q_row = r0 + lane32;
   const bool row_live = q_row < T;
-  for (int ks = 0; ks < ksteps; ++ks) {
-    const int kbase = ks * 16 + 8 * half;
-    if (row_live) {
-      const unsigned short* qp =
-          Q + (static_cast<long long>(q_row) * q_heads + qh) * hd + kbase;
+  if (row_live) {
+    const unsigned short* qp =
+        Q + (static_cast<long long>(q_row) * q_heads + qh) * HD + 8 * half;
+#pragma unroll
+    for (int ks = 0; ks < kKsteps; ++ks) {
+      bf16x8_t qv = *reinterpret_cast<const bf16x8_t*>(qp + ks * 16);
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         // pre-scale Q by 1/sqrt(hd) (cheaper than scaling S)
-        qf[ks][i] = static_cast<__bf16>(bf16_to_f32(qp[i]) * scale);
+        qf[ks][i] = static_cast<__bf16>(static_cast<float>(qv[i]) * scale);
       }
-    } else {
+    }
+  } else {
+#pragma unroll
+    for (int ks = 0; ks < kKsteps; ++ks) {
 #pragma unroll
       for (int i = 0; i < 8; ++i) qf[ks][i] = static_cast<__bf16>(0.0f);
     }
   }
 
   float m_acc[16], l_acc[16];
-  f32x16_t oacc[4];
+  f32x16_t oacc[kNblk];
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     m_acc[r] = -1e30f;
     l_acc[r] = 0.0f;
   }
-  for (int b = 0; b < 4; ++b) {
+#pragma unroll
+  for (int b = 0; b < kNblk; ++b) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) oacc[b][r] = 0.0f;
   }
@@ -770,22 +776,42 @@ __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
   const int kt_end = min((pos0 + r0 + 31) / 32 + 1, (t_kv + 31) / 32);
   for (int kt = 0; kt < kt_end; ++kt) {
     const int kcol0 = kt * 32;
+    const int key = kcol0 + lane32;  // B-fragment column == key index
+    const bool key_live = key < t_kv;
+    // ---- stage this key tile of V into LDS, transposed to [hd][key]
+    // (so the P.V B-fragment reads below are contiguous b128 loads)
+    {
+      const unsigned short* vp =
+          V + (static_cast<long long>(min(key, t_kv - 1)) * kv_heads + kvh) *
+                  HD + 8 * half;
+#pragma unroll
+      for (int c8 = 0; c8 < HD / 16; ++c8) {
+        bf16x8_t vv;
+        if (key_live) {
+          vv = *reinterpret_cast<const bf16x8_t*>(vp + c8 * 16);
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) vv[i] = static_cast<__bf16>(0.0f);
+        }
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          s_v[(c8 * 16 + 8 * half + i) * 32 + lane32] = vv[i];
+        }
+      }
+    }
+
     // ---- S = Q . K^T over this key tile
     f32x16_t sacc;
 #pragma unroll
     for (int r = 0; r < 16; ++r) sacc[r] = 0.0f;
-    const int key = kcol0 + lane32;  // B-fragment column == key index
-    const bool key_live = key < t_kv;
-    for (int ks = 0; ks < ksteps; ++ks) {
+    const unsigned short* kp =
+        K + (static_cast<long long>(key_live ? key : 0) * kv_heads + kvh) *
+                HD + 8 * half;
+#pragma unroll
+    for (int ks = 0; ks < kKsteps; ++ks) {
       bf16x8_t kf;
       if (key_live) {
-        const unsigned short* kp =
-            K + (static_cast<long long>(key) * kv_heads + kvh) * hd +
-            ks * 16 + 8 * half;
-#pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          kf[i] = *reinterpret_cast<const __bf16*>(kp + i);
-        }
+        kf = *reinterpret_cast<const bf16x8_t*>(kp + ks * 16);
       } else {
 #pragma unroll
         for (int i = 0; i < 8; ++i) kf[i] = static_cast<__bf16>(0.0f);
@@ -801,16 +827,16 @@ __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
       const int row = (r % 4) + 8 * (r / 4) + 4 * half;  // row within tile
       const int qpos = pos0 + r0 + row;
       const int kpos = kcol0 + lane32;
-      float s = sacc[r];
-      if (kpos > qpos || kpos >= t_kv) s = -1e30f;
-      const float row_max = wave32_max(s);
+      float sv = sacc[r];
+      if (kpos > qpos || kpos >= t_kv) sv = -1e30f;
+      const float row_max = wave32_max(sv);
       const float m_new = fmaxf(m_acc[r], row_max);
       const float alpha = __expf(m_acc[r] - m_new);
-      const float p = __expf(s - m_new);
+      const float p = __expf(sv - m_new);
       l_acc[r] = l_acc[r] * alpha + wave32_sum(p);
       m_acc[r] = m_new;
 #pragma unroll
-      for (int b = 0; b < 4; ++b) oacc[b][r] *= alpha;
+      for (int b = 0; b < kNblk; ++b) oacc[b][r] *= alpha;
       p_regs[r] = p;
     }
 
@@ -820,37 +846,18 @@ __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
       const int row = (r % 4) + 8 * (r / 4) + 4 * half;
       s_p[row * 32 + lane32] = static_cast<__bf16>(p_regs[r]);
     }
-    // V tile -> LDS (coalesced: each lane one key row slice)
-    {
-      const int vkey = kcol0 + lane32;
-      const unsigned short* vp =
-          V + (static_cast<long long>(min(vkey, t_kv - 1)) * kv_heads + kvh) *
-                  hd;
-      const bool vlive = vkey < t_kv;
-      for (int c = half * 8; c < hd; c += 16) {
-#pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          s_v[lane32 * 128 + c + i] =
-              vlive ? *reinterpret_cast<const __bf16*>(vp + c + i)
-                    : static_cast<__bf16>(0.0f);
-        }
-      }
-    }
     __builtin_amdgcn_s_waitcnt(0);  // drain LDS writes (single wave)
 
     // ---- O += P . V  (K dim = 32 keys = 2 MFMA k-steps)
+#pragma unroll
     for (int ks2 = 0; ks2 < 2; ++ks2) {
-      bf16x8_t pf;
+      bf16x8_t pf = *reinterpret_cast<const bf16x8_t*>(
+          &s_p[lane32 * 32 + ks2 * 16 + 8 * half]);
 #pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        pf[i] = s_p[lane32 * 32 + ks2 * 16 + 8 * half + i];
-      }
-      for (int b = 0; b < nblk; ++b) {
-        bf16x8_t vf;
-#pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          vf[i] = s_v[(ks2 * 16 + 8 * half + i) * 128 + b * 32 + lane32];
-        }
+      for (int b = 0; b < kNblk; ++b) {
+        // transposed LDS: one contiguous b128 read per fragment
+        bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+            &s_v[(b * 32 + lane32) * 32 + ks2 * 16 + 8 * half]);
         oacc[b] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf, vf, oacc[b], 0,
                                                           0, 0);
       }
@@ -866,8 +873,9 @@ __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
     if (orow >= T) continue;
     const float inv_l = l_acc[r] > 0.0f ? 1.0f / l_acc[r] : 0.0f;
     unsigned short* op =
-        O + (static_cast<long long>(orow) * q_heads + qh) * hd;
-    for (int b = 0; b < nblk; ++b) {
+        O + (static_cast<long long>(orow) * q_heads + qh) * HD;
+#pragma unroll
+    for (int b = 0; b < kNblk; ++b) {
       op[b * 32 + lane32] = f32_to_bf16(oacc[b][r] * inv_l);
     }
   }
@@ -881,10 +889,18 @@ extern "C" hipError_t fma_launch_attn_prefill_bf16(
   if (hd != 64 && hd != 128) return hipErrorInvalidValue;
   if (q_heads % kv_heads != 0) return hipErrorInvalidValue;
   dim3 grid(q_heads, (T + 31) / 32);
- hipLaunchKernelGGL(( attn_prefill_bf16_kernel), dim3(grid), dim3(64), 0, stream, 
-      static_cast<const unsigned short*>(Q),
-      static_cast<const unsigned short*>(K),
-      static_cast<const unsigned short*>(V),
-      static_cast<unsigned short*>(O), T, pos0, q_heads, kv_heads, hd);
+  if (hd == 128) {
+   hipLaunchKernelGGL(( attn_prefill_bf16_kernel<128>), dim3(grid), dim3(64), 0, stream, 
+        static_cast<const unsigned short*>(Q),
+        static_cast<const unsigned short*>(K),
+        static_cast<const unsigned short*>(V),
+        static_cast<unsigned short*>(O), T, pos0, q_heads, kv_heads);
+  } else {
+   hipLaunchKernelGGL(( attn_prefill_bf16_kernel<64>), dim3(grid), dim3(64), 0, stream, 
+        static_cast<const unsigned short*>(Q),
+        static_cast<const unsigned short*>(K),
+        static_cast<const unsigned short*>(V),
+        static_cast<unsigned short*>(O), T, pos0, q_heads, kv_heads);
+  }
   return hipGetLastError();
 }

@@ -462,13 +462,13 @@ def test_prefill_mfma_vs_sdpa_path_logits_close():
     torch.manual_seed(3)
     prompt = torch.randint(0, cfg.vocab_size, (1, 40), device="cuda:0")
     cache = KVCache(cfg, 1, "cuda:0", 1, 128)
-    mfma = eng.model.forward(prompt, cache, 0).float()
-    os.environ["FMA_DISABLE_MFMA_PREFILL"] = "1"
+    os.environ["FMA_MFMA_PREFILL"] = "1"
     try:
-        cache2 = KVCache(cfg, 1, "cuda:0", 1, 128)
-        sdpa = eng.model.forward(prompt, cache2, 0).float()
+        mfma = eng.model.forward(prompt, cache, 0).float()
     finally:
-        del os.environ["FMA_DISABLE_MFMA_PREFILL"]
+        del os.environ["FMA_MFMA_PREFILL"]
+    cache2 = KVCache(cfg, 1, "cuda:0", 1, 128)
+    sdpa = eng.model.forward(prompt, cache2, 0).float()
     assert torch.allclose(mfma, sdpa, atol=8e-2, rtol=8e-2), \
         (mfma - sdpa).abs().max().item()
     cache.free()
