@@ -712,6 +712,109 @@ __device__ inline float wave32_sum(float v) {
   return v;
 }
 
+// s_waitcnt immediate that waits lgkmcnt(0) only (vmcnt/expcnt left
+// outstanding): gfx9/CDNA encoding vmcnt[3:0]|expcnt[6:4]|lgkmcnt[11:8]
+// with vmcnt high bits [15:14].
+#define FMA_WAIT_LGKM0 0xC07F
+
+template <int HD>
+struct PrefillTileBufs {
+  bf16x8_t kf[HD / 16];  // K B-fragments for one 32-key tile
+  bf16x8_t vv[HD / 16];  // V rows (one key per lane) for LDS staging
+};
+
+// Issue the global loads for key tile kt into `buf` (no waits here; the
+// consumer's first use synchronizes via the compiler's vmcnt tracking).
+template <int HD>
+__device__ __forceinline__ void prefill_load_tile(
+    const unsigned short* __restrict__ K,
+    const unsigned short* __restrict__ V, int kcol0, int t_kv, int kv_heads,
+    int kvh, int lane32, int half, PrefillTileBufs<HD>& buf) {
+  const int key = kcol0 + lane32;
+  const bool live = key < t_kv;
+  const long long row = static_cast<long long>(live ? key : 0) * kv_heads +
+                        kvh;
+  const unsigned short* kp = K + row * HD + 8 * half;
+  const unsigned short* vp = V + row * HD + 8 * half;
+#pragma unroll
+  for (int ks = 0; ks < HD / 16; ++ks) {
+    if (live) {
+      buf.kf[ks] = *reinterpret_cast<const bf16x8_t*>(kp + ks * 16);
+      buf.vv[ks] = *reinterpret_cast<const bf16x8_t*>(vp + ks * 16);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        buf.kf[ks][i] = static_cast<__bf16>(0.0f);
+        buf.vv[ks][i] = static_cast<__bf16>(0.0f);
+      }
+    }
+  }
+}
+
+// One 32-key tile: S = Q.K^T (MFMA), online-softmax update, P and V
+// through LDS, O += P.V (MFMA). Wave-synchronous (single wave per
+// workgroup), so LDS ordering needs only lgkmcnt waits.
+template <int HD>
+__device__ __forceinline__ void prefill_process_tile(
+    const PrefillTileBufs<HD>& buf, const bf16x8_t (&qf)[HD / 16],
+    __bf16* s_p, __bf16* s_v, int kcol0, int pos0, int r0, int t_kv,
+    int lane32, int half, float (&m_acc)[16], float (&l_acc)[16],
+    f32x16_t (&oacc)[HD / 32]) {
+  constexpr int kNblk = HD / 32;
+  // stage V transposed ([hd][key]) so P.V fragment reads are b128
+#pragma unroll
+  for (int ks = 0; ks < HD / 16; ++ks) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      s_v[(ks * 16 + 8 * half + i) * 32 + lane32] = buf.vv[ks][i];
+    }
+  }
+
+  f32x16_t sacc;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) sacc[r] = 0.0f;
+#pragma unroll
+  for (int ks = 0; ks < HD / 16; ++ks) {
+    sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf[ks], buf.kf[ks], sacc,
+                                                   0, 0, 0);
+  }
+
+  // online softmax; P goes straight to LDS in A-fragment source layout
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r % 4) + 8 * (r / 4) + 4 * half;  // row within tile
+    const int qpos = pos0 + r0 + row;
+    const int kpos = kcol0 + lane32;
+    float sv = sacc[r];
+    if (kpos > qpos || kpos >= t_kv) sv = -1e30f;
+    const float row_max = wave32_max(sv);
+    const float m_new = fmaxf(m_acc[r], row_max);
+    const float alpha = __expf(m_acc[r] - m_new);
+    const float p = __expf(sv - m_new);
+    l_acc[r] = l_acc[r] * alpha + wave32_sum(p);
+    m_acc[r] = m_new;
+#pragma unroll
+    for (int b = 0; b < kNblk; ++b) oacc[b][r] *= alpha;
+    s_p[row * 32 + lane32] = static_cast<__bf16>(p);
+  }
+  __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);  // P + V writes visible
+
+  // O += P . V  (K dim = 32 keys = 2 MFMA k-steps)
+#pragma unroll
+  for (int ks2 = 0; ks2 < 2; ++ks2) {
+    bf16x8_t pf = *reinterpret_cast<const bf16x8_t*>(
+        &s_p[lane32 * 32 + ks2 * 16 + 8 * half]);
+#pragma unroll
+    for (int b = 0; b < kNblk; ++b) {
+      bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+          &s_v[(b * 32 + lane32) * 32 + ks2 * 16 + 8 * half]);
+      oacc[b] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf, vf, oacc[b], 0,
+                                                        0, 0);
+    }
+  }
+  __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);  // reads done before reuse
+}
+
 template <int HD>
 __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
     const unsigned short* __restrict__ Q,  // [T, qH, hd]
@@ -719,15 +822,14 @@ __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
     const unsigned short* __restrict__ V,  // [S, kvH, hd]
     unsigned short* __restrict__ O,        // [T, qH, hd]
     int T, int pos0, int q_heads, int kv_heads) {
-  constexpr int kNblk = HD / 32;    // 2 (hd=64) or 4 (hd=128)
-  constexpr int kKsteps = HD / 16;  // QK^T k-steps
+  constexpr int kNblk = HD / 32;
   const int qh = blockIdx.x;
   const int tile = blockIdx.y;
   const int r0 = tile * 32;
   if (r0 >= T) return;
   const int kvh = qh / (q_heads / kv_heads);
   const int l = threadIdx.x;
-  const int half = l >> 5;       // 0 or 1
+  const int half = l >> 5;  // 0 or 1
   const int lane32 = l & 31;
   const float scale = rsqrtf(static_cast<float>(HD));
 
@@ -736,14 +838,14 @@ __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
 
   // Q fragments, kept in registers for the whole key loop.
   // A-layout: lane holds A[lane32][8*half + i] per 16-k step.
-  bf16x8_t qf[kKsteps];
+  bf16x8_t qf[HD / 16];
   const int q_row = r0 + lane32;
   const bool row_live = q_row < T;
   if (row_live) {
     const unsigned short* qp =
         Q + (static_cast<long long>(q_row) * q_heads + qh) * HD + 8 * half;
 #pragma unroll
-    for (int ks = 0; ks < kKsteps; ++ks) {
+    for (int ks = 0; ks < HD / 16; ++ks) {
       bf16x8_t qv = *reinterpret_cast<const bf16x8_t*>(qp + ks * 16);
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
@@ -753,7 +855,7 @@ __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
     }
   } else {
 #pragma unroll
-    for (int ks = 0; ks < kKsteps; ++ks) {
+    for (int ks = 0; ks < HD / 16; ++ks) {
 #pragma unroll
       for (int i = 0; i < 8; ++i) qf[ks][i] = static_cast<__bf16>(0.0f);
     }
@@ -772,97 +874,28 @@ __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
     for (int r = 0; r < 16; ++r) oacc[b][r] = 0.0f;
   }
 
-  const int t_kv = pos0 + T;                       // causal key horizon
+  const int t_kv = pos0 + T;  // causal key horizon
   const int kt_end = min((pos0 + r0 + 31) / 32 + 1, (t_kv + 31) / 32);
-  for (int kt = 0; kt < kt_end; ++kt) {
-    const int kcol0 = kt * 32;
-    const int key = kcol0 + lane32;  // B-fragment column == key index
-    const bool key_live = key < t_kv;
-    // ---- stage this key tile of V into LDS, transposed to [hd][key]
-    // (so the P.V B-fragment reads below are contiguous b128 loads)
-    {
-      const unsigned short* vp =
-          V + (static_cast<long long>(min(key, t_kv - 1)) * kv_heads + kvh) *
-                  HD + 8 * half;
-#pragma unroll
-      for (int c8 = 0; c8 < HD / 16; ++c8) {
-        bf16x8_t vv;
-        if (key_live) {
-          vv = *reinterpret_cast<const bf16x8_t*>(vp + c8 * 16);
-        } else {
-#pragma unroll
-          for (int i = 0; i < 8; ++i) vv[i] = static_cast<__bf16>(0.0f);
-        }
-#pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          s_v[(c8 * 16 + 8 * half + i) * 32 + lane32] = vv[i];
-        }
-      }
-    }
 
-    // ---- S = Q . K^T over this key tile
-    f32x16_t sacc;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) sacc[r] = 0.0f;
-    const unsigned short* kp =
-        K + (static_cast<long long>(key_live ? key : 0) * kv_heads + kvh) *
-                HD + 8 * half;
-#pragma unroll
-    for (int ks = 0; ks < kKsteps; ++ks) {
-      bf16x8_t kf;
-      if (key_live) {
-        kf = *reinterpret_cast<const bf16x8_t*>(kp + ks * 16);
-      } else {
-#pragma unroll
-        for (int i = 0; i < 8; ++i) kf[i] = static_cast<__bf16>(0.0f);
-      }
-      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf[ks], kf, sacc, 0, 0,
-                                                     0);
+  // Software-pipelined key loop: tile kt+1's global loads are in flight
+  // while tile kt computes. Two explicit register buffers (A/B) keep
+  // every index compile-time — dynamic indexing would spill to scratch.
+  PrefillTileBufs<HD> bufA, bufB;
+  prefill_load_tile<HD>(K, V, 0, t_kv, kv_heads, kvh, lane32, half, bufA);
+  for (int kt = 0; kt < kt_end; kt += 2) {
+    if (kt + 1 < kt_end) {
+      prefill_load_tile<HD>(K, V, (kt + 1) * 32, t_kv, kv_heads, kvh,
+                            lane32, half, bufB);
     }
-
-    // ---- online softmax update (row stats across lane32 within a half)
-    float p_regs[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = (r % 4) + 8 * (r / 4) + 4 * half;  // row within tile
-      const int qpos = pos0 + r0 + row;
-      const int kpos = kcol0 + lane32;
-      float sv = sacc[r];
-      if (kpos > qpos || kpos >= t_kv) sv = -1e30f;
-      const float row_max = wave32_max(sv);
-      const float m_new = fmaxf(m_acc[r], row_max);
-      const float alpha = __expf(m_acc[r] - m_new);
-      const float p = __expf(sv - m_new);
-      l_acc[r] = l_acc[r] * alpha + wave32_sum(p);
-      m_acc[r] = m_new;
-#pragma unroll
-      for (int b = 0; b < kNblk; ++b) oacc[b][r] *= alpha;
-      p_regs[r] = p;
+    prefill_process_tile<HD>(bufA, qf, s_p, s_v, kt * 32, pos0, r0, t_kv,
+                             lane32, half, m_acc, l_acc, oacc);
+    if (kt + 1 >= kt_end) break;
+    if (kt + 2 < kt_end) {
+      prefill_load_tile<HD>(K, V, (kt + 2) * 32, t_kv, kv_heads, kvh,
+                            lane32, half, bufA);
     }
-
-    // ---- P (C/D layout) -> LDS -> A-fragment layout
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int row = (r % 4) + 8 * (r / 4) + 4 * half;
-      s_p[row * 32 + lane32] = static_cast<__bf16>(p_regs[r]);
-    }
-    __builtin_amdgcn_s_waitcnt(0);  // drain LDS writes (single wave)
-
-    // ---- O += P . V  (K dim = 32 keys = 2 MFMA k-steps)
-#pragma unroll
-    for (int ks2 = 0; ks2 < 2; ++ks2) {
-      bf16x8_t pf = *reinterpret_cast<const bf16x8_t*>(
-          &s_p[lane32 * 32 + ks2 * 16 + 8 * half]);
-#pragma unroll
-      for (int b = 0; b < kNblk; ++b) {
-        // transposed LDS: one contiguous b128 read per fragment
-        bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
-            &s_v[(b * 32 + lane32) * 32 + ks2 * 16 + 8 * half]);
-        oacc[b] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf, vf, oacc[b], 0,
-                                                          0, 0);
-      }
-    }
-    __builtin_amdgcn_s_waitcnt(0);  // LDS reads done before next overwrite
+    prefill_process_tile<HD>(bufB, qf, s_p, s_v, (kt + 1) * 32, pos0, r0,
+                             t_kv, lane32, half, m_acc, l_acc, oacc);
   }
 
   // ---- epilogue: O /= l, store rows < T

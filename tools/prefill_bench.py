@@ -29,11 +29,21 @@ def bench_op(fn, iters):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=50)
+    ap.add_argument("--t", type=int, default=0,
+                    help="single T, MFMA kernel only (for profiling)")
     args = ap.parse_args()
     import fma_amd._C as C
 
     qH, kvH, hd = 32, 8, 128
     rep = qH // kvH
+    if args.t:
+        T = args.t
+        q = torch.randn(T, qH, hd, dtype=torch.bfloat16, device="cuda:0")
+        k = torch.randn(T, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
+        v = torch.randn(T, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
+        t_ms = bench_op(lambda: C.attn_prefill_bf16(q, k, v, 0), args.iters)
+        print(f"T={T} mfma_ms={t_ms:.3f}")
+        return
     print(f"{'T':>6} {'mfma_ms':>9} {'sdpa_ms':>9} {'speedup':>8} {'max_err':>9}")
     for T in (128, 512, 1024, 2048, 4096):
         q = torch.randn(T, qH, hd, dtype=torch.bfloat16, device="cuda:0")
