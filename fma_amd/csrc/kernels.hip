@@ -816,7 +816,8 @@ __device__ __forceinline__ void prefill_process_tile(
 }
 
 template <int HD>
-__global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
+__global__ __launch_bounds__(64)
+__attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
     const unsigned short* __restrict__ Q,  // [T, qH, hd]
     const unsigned short* __restrict__ K,  // [S, kvH, hd]
     const unsigned short* __restrict__ V,  // [S, kvH, hd]
@@ -877,25 +878,15 @@ __global__ __launch_bounds__(64) void attn_prefill_bf16_kernel(
   const int t_kv = pos0 + T;  // causal key horizon
   const int kt_end = min((pos0 + r0 + 31) / 32 + 1, (t_kv + 31) / 32);
 
-  // Software-pipelined key loop: tile kt+1's global loads are in flight
-  // while tile kt computes. Two explicit register buffers (A/B) keep
-  // every index compile-time — dynamic indexing would spill to scratch.
-  PrefillTileBufs<HD> bufA, bufB;
-  prefill_load_tile<HD>(K, V, 0, t_kv, kv_heads, kvh, lane32, half, bufA);
-  for (int kt = 0; kt < kt_end; kt += 2) {
-    if (kt + 1 < kt_end) {
-      prefill_load_tile<HD>(K, V, (kt + 1) * 32, t_kv, kv_heads, kvh,
-                            lane32, half, bufB);
-    }
-    prefill_process_tile<HD>(bufA, qf, s_p, s_v, kt * 32, pos0, r0, t_kv,
+  // Occupancy (2+ waves/SIMD, amdgpu_waves_per_eu below) hides the
+  // global-load and LDS latency; PMC profiling showed prefetch double-
+  // buffering bought nothing at occupancy 1 while costing 128 AGPRs.
+  PrefillTileBufs<HD> buf;
+  for (int kt = 0; kt < kt_end; ++kt) {
+    prefill_load_tile<HD>(K, V, kt * 32, t_kv, kv_heads, kvh, lane32, half,
+                          buf);
+    prefill_process_tile<HD>(buf, qf, s_p, s_v, kt * 32, pos0, r0, t_kv,
                              lane32, half, m_acc, l_acc, oacc);
-    if (kt + 1 >= kt_end) break;
-    if (kt + 2 < kt_end) {
-      prefill_load_tile<HD>(K, V, (kt + 2) * 32, t_kv, kv_heads, kvh,
-                            lane32, half, bufA);
-    }
-    prefill_process_tile<HD>(bufB, qf, s_p, s_v, (kt + 1) * 32, pos0, r0,
-                             t_kv, lane32, half, m_acc, l_acc, oacc);
   }
 
   // ---- epilogue: O /= l, store rows < T
