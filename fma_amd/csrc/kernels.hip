@@ -717,103 +717,9 @@ __device__ inline float wave32_sum(float v) {
 // with vmcnt high bits [15:14].
 #define FMA_WAIT_LGKM0 0xC07F
 
-template <int HD>
-struct PrefillTileBufs {
-  bf16x8_t kf[HD / 16];  // K B-fragments for one 32-key tile
-  bf16x8_t vv[HD / 16];  // V rows (one key per lane) for LDS staging
-};
-
-// Issue the global loads for key tile kt into `buf` (no waits here; the
-// consumer's first use synchronizes via the compiler's vmcnt tracking).
-template <int HD>
-__device__ __forceinline__ void prefill_load_tile(
-    const unsigned short* __restrict__ K,
-    const unsigned short* __restrict__ V, int kcol0, int t_kv, int kv_heads,
-    int kvh, int lane32, int half, PrefillTileBufs<HD>& buf) {
-  const int key = kcol0 + lane32;
-  const bool live = key < t_kv;
-  const long long row = static_cast<long long>(live ? key : 0) * kv_heads +
-                        kvh;
-  const unsigned short* kp = K + row * HD + 8 * half;
-  const unsigned short* vp = V + row * HD + 8 * half;
-#pragma unroll
-  for (int ks = 0; ks < HD / 16; ++ks) {
-    if (live) {
-      buf.kf[ks] = *reinterpret_cast<const bf16x8_t*>(kp + ks * 16);
-      buf.vv[ks] = *reinterpret_cast<const bf16x8_t*>(vp + ks * 16);
-    } else {
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        buf.kf[ks][i] = static_cast<__bf16>(0.0f);
-        buf.vv[ks][i] = static_cast<__bf16>(0.0f);
-      }
-    }
-  }
-}
-
-// One 32-key tile: S = Q.K^T (MFMA), online-softmax update, P and V
-// through LDS, O += P.V (MFMA). Wave-synchronous (single wave per
-// workgroup), so LDS ordering needs only lgkmcnt waits.
-template <int HD>
-__device__ __forceinline__ void prefill_process_tile(
-    const PrefillTileBufs<HD>& buf, const bf16x8_t (&qf)[HD / 16],
-    __bf16* s_p, __bf16* s_v, int kcol0, int pos0, int r0, int t_kv,
-    int lane32, int half, float (&m_acc)[16], float (&l_acc)[16],
-    f32x16_t (&oacc)[HD / 32]) {
-  constexpr int kNblk = HD / 32;
-  // stage V transposed ([hd][key]) so P.V fragment reads are b128
-#pragma unroll
-  for (int ks = 0; ks < HD / 16; ++ks) {
-#pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      s_v[(ks * 16 + 8 * half + i) * 32 + lane32] = buf.vv[ks][i];
-    }
-  }
-
-  f32x16_t sacc;
-#pragma unroll
-  for (int r = 0; r < 16; ++r) sacc[r] = 0.0f;
-#pragma unroll
-  for (int ks = 0; ks < HD / 16; ++ks) {
-    sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf[ks], buf.kf[ks], sacc,
-                                                   0, 0, 0);
-  }
-
-  // online softmax; P goes straight to LDS in A-fragment source layout
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int row = (r % 4) + 8 * (r / 4) + 4 * half;  // row within tile
-    const int qpos = pos0 + r0 + row;
-    const int kpos = kcol0 + lane32;
-    float sv = sacc[r];
-    if (kpos > qpos || kpos >= t_kv) sv = -1e30f;
-    const float row_max = wave32_max(sv);
-    const float m_new = fmaxf(m_acc[r], row_max);
-    const float alpha = __expf(m_acc[r] - m_new);
-    const float p = __expf(sv - m_new);
-    l_acc[r] = l_acc[r] * alpha + wave32_sum(p);
-    m_acc[r] = m_new;
-#pragma unroll
-    for (int b = 0; b < kNblk; ++b) oacc[b][r] *= alpha;
-    s_p[row * 32 + lane32] = static_cast<__bf16>(p);
-  }
-  __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);  // P + V writes visible
-
-  // O += P . V  (K dim = 32 keys = 2 MFMA k-steps)
-#pragma unroll
-  for (int ks2 = 0; ks2 < 2; ++ks2) {
-    bf16x8_t pf = *reinterpret_cast<const bf16x8_t*>(
-        &s_p[lane32 * 32 + ks2 * 16 + 8 * half]);
-#pragma unroll
-    for (int b = 0; b < kNblk; ++b) {
-      bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
-          &s_v[(b * 32 + lane32) * 32 + ks2 * 16 + 8 * half]);
-      oacc[b] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf, vf, oacc[b], 0,
-                                                        0, 0);
-    }
-  }
-  __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);  // reads done before reuse
-}
+// Row stride of the f32 score scratch: 33 makes every per-row transpose
+// access land in a distinct LDS bank (32 would put a whole row in one).
+#define FMA_SROW 33
 
 template <int HD>
 __global__ __launch_bounds__(64)
@@ -834,8 +740,11 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
   const int lane32 = l & 31;
   const float scale = rsqrtf(static_cast<float>(HD));
 
-  __shared__ __bf16 s_p[32 * 32];
-  __shared__ __bf16 s_v[32 * HD];  // transposed: [hd][key]
+  __shared__ __bf16 s_p[32 * 32];          // P tile, A-fragment source
+  __shared__ __bf16 s_v[2][32 * HD];       // V, transposed [hd][key], 2 bufs
+  __shared__ float s_s[32 * FMA_SROW];     // scores/probs transpose scratch
+  __shared__ float s_stat[32 * 2];         // per-row (m_new, alpha)
+  __shared__ float s_l[32];                // per-row l at epilogue
 
   // Q fragments, kept in registers for the whole key loop.
   // A-layout: lane holds A[lane32][8*half + i] per 16-k step.
@@ -862,13 +771,14 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
     }
   }
 
-  float m_acc[16], l_acc[16];
+  // Per-ROW softmax stats live in "transpose layout": lane l keeps the
+  // running max and sum for row l%32 (both halves redundantly). The C/D
+  // per-register layout picks them up from LDS each tile. This replaces
+  // 10 ds_bpermute reduction chains per register (160 per tile) with a
+  // handful of conflict-free LDS transposes — and 2 VGPRs instead of 32.
+  float m_row = -1e30f, l_row = 0.0f;
+
   f32x16_t oacc[kNblk];
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    m_acc[r] = -1e30f;
-    l_acc[r] = 0.0f;
-  }
 #pragma unroll
   for (int b = 0; b < kNblk; ++b) {
 #pragma unroll
@@ -877,25 +787,133 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
 
   const int t_kv = pos0 + T;  // causal key horizon
   const int kt_end = min((pos0 + r0 + 31) / 32 + 1, (t_kv + 31) / 32);
-
-  // Occupancy (2+ waves/SIMD, amdgpu_waves_per_eu below) hides the
-  // global-load and LDS latency; PMC profiling showed prefetch double-
-  // buffering bought nothing at occupancy 1 while costing 128 AGPRs.
-  PrefillTileBufs<HD> buf;
   for (int kt = 0; kt < kt_end; ++kt) {
-    prefill_load_tile<HD>(K, V, kt * 32, t_kv, kv_heads, kvh, lane32, half,
-                          buf);
-    prefill_process_tile<HD>(buf, qf, s_p, s_v, kt * 32, pos0, r0, t_kv,
-                             lane32, half, m_acc, l_acc, oacc);
+    const int kcol0 = kt * 32;
+    const int key = kcol0 + lane32;
+    const bool key_live = key < t_kv;
+    const long long krow =
+        static_cast<long long>(key_live ? key : 0) * kv_heads + kvh;
+
+    // ---- stage this tile's V into the alternating LDS buffer,
+    // transposed to [hd][key] (the PV reads then are contiguous b128).
+    // Previous tile's PV reads hit the OTHER buffer, so no extra wait.
+    __bf16* vdst = s_v[kt & 1];
+    {
+      const unsigned short* vp = V + krow * HD + 8 * half;
+#pragma unroll
+      for (int c8 = 0; c8 < HD / 16; ++c8) {
+        bf16x8_t vv;
+        if (key_live) {
+          vv = *reinterpret_cast<const bf16x8_t*>(vp + c8 * 16);
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) vv[i] = static_cast<__bf16>(0.0f);
+        }
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          vdst[(c8 * 16 + 8 * half + i) * 32 + lane32] = vv[i];
+        }
+      }
+    }
+
+    // ---- S = Q . K^T over this key tile
+    f32x16_t sacc;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) sacc[r] = 0.0f;
+    const unsigned short* kp = K + krow * HD + 8 * half;
+#pragma unroll
+    for (int ks = 0; ks < HD / 16; ++ks) {
+      bf16x8_t kf;
+      if (key_live) {
+        kf = *reinterpret_cast<const bf16x8_t*>(kp + ks * 16);
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) kf[i] = static_cast<__bf16>(0.0f);
+      }
+      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf[ks], kf, sacc, 0, 0,
+                                                     0);
+    }
+
+    // ---- phase A: masked scores into the f32 transpose scratch
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r % 4) + 8 * (r / 4) + 4 * half;
+      const int qpos = pos0 + r0 + row;
+      if (kcol0 + lane32 > qpos || kcol0 + lane32 >= t_kv) sacc[r] = -1e30f;
+      s_s[row * FMA_SROW + lane32] = sacc[r];
+    }
+    __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);
+
+    // ---- phase B: per-row max (lane l reduces row l%32, cols 16h..)
+    {
+      float m_part = -1e30f;
+#pragma unroll
+      for (int i = 0; i < 16; ++i) {
+        m_part = fmaxf(m_part, s_s[lane32 * FMA_SROW + 16 * half + i]);
+      }
+      const float m_tile = fmaxf(m_part, __shfl_xor(m_part, 32));
+      const float m_new = fmaxf(m_row, m_tile);
+      const float alpha = __expf(m_row - m_new);
+      m_row = m_new;
+      l_row *= alpha;
+      if (half == 0) {
+        s_stat[lane32 * 2 + 0] = m_new;
+        s_stat[lane32 * 2 + 1] = alpha;
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);
+
+    // ---- phase C: P = exp(S - m) into s_p (MFMA layout) and s_s
+    // (transpose layout, for the row sums); rescale O by alpha
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r % 4) + 8 * (r / 4) + 4 * half;
+      const float m_new = s_stat[row * 2 + 0];
+      const float alpha = s_stat[row * 2 + 1];
+      const float p = __expf(sacc[r] - m_new);
+#pragma unroll
+      for (int b = 0; b < kNblk; ++b) oacc[b][r] *= alpha;
+      s_p[row * 32 + lane32] = static_cast<__bf16>(p);
+      s_s[row * FMA_SROW + lane32] = p;
+    }
+    __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);
+
+    // ---- phase D: row sums from the transpose scratch
+    {
+      float sum = 0.0f;
+#pragma unroll
+      for (int i = 0; i < 16; ++i) {
+        sum += s_s[lane32 * FMA_SROW + 16 * half + i];
+      }
+      l_row += sum + __shfl_xor(sum, 32);
+    }
+
+    // ---- O += P . V  (K dim = 32 keys = 2 MFMA k-steps)
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      bf16x8_t pf = *reinterpret_cast<const bf16x8_t*>(
+          &s_p[lane32 * 32 + ks2 * 16 + 8 * half]);
+#pragma unroll
+      for (int b = 0; b < kNblk; ++b) {
+        bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+            &vdst[(b * 32 + lane32) * 32 + ks2 * 16 + 8 * half]);
+        oacc[b] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf, vf, oacc[b], 0,
+                                                          0, 0);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);  // s_p reads done
   }
 
   // ---- epilogue: O /= l, store rows < T
+  if (half == 0) s_l[lane32] = l_row;
+  __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int row = (r % 4) + 8 * (r / 4) + 4 * half;
     const int orow = r0 + row;
     if (orow >= T) continue;
-    const float inv_l = l_acc[r] > 0.0f ? 1.0f / l_acc[r] : 0.0f;
+    const float lr = s_l[row];
+    const float inv_l = lr > 0.0f ? 1.0f / lr : 0.0f;
     unsigned short* op =
         O + (static_cast<long long>(orow) * q_heads + qh) * HD;
 #pragma unroll
