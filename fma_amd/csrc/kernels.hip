@@ -721,6 +721,54 @@ __device__ inline float wave32_sum(float v) {
 // access land in a distinct LDS bank (32 would put a whole row in one).
 #define FMA_SROW 33
 
+// Load one 32-key tile of K as B-fragments (8 bf16 per lane per k-step).
+template <int HD>
+__device__ __forceinline__ void prefill_load_k(
+    const unsigned short* __restrict__ K, int kcol0, int t_kv, int kv_heads,
+    int kvh, int lane32, int half, bf16x8_t (&kf)[HD / 16]) {
+  const int key = kcol0 + lane32;
+  const bool live = key < t_kv;
+  const unsigned short* kp =
+      K + (static_cast<long long>(live ? key : 0) * kv_heads + kvh) * HD +
+      8 * half;
+#pragma unroll
+  for (int ks = 0; ks < HD / 16; ++ks) {
+    if (live) {
+      kf[ks] = *reinterpret_cast<const bf16x8_t*>(kp + ks * 16);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) kf[ks][i] = static_cast<__bf16>(0.0f);
+    }
+  }
+}
+
+// Stage one 32-key tile of V into LDS, transposed to [hd][key] so the
+// P.V B-fragment reads are contiguous b128 loads.
+template <int HD>
+__device__ __forceinline__ void prefill_stage_v(
+    const unsigned short* __restrict__ V, int kcol0, int t_kv, int kv_heads,
+    int kvh, int lane32, int half, __bf16* vdst) {
+  const int key = kcol0 + lane32;
+  const bool live = key < t_kv;
+  const unsigned short* vp =
+      V + (static_cast<long long>(live ? key : 0) * kv_heads + kvh) * HD +
+      8 * half;
+#pragma unroll
+  for (int c8 = 0; c8 < HD / 16; ++c8) {
+    bf16x8_t vv;
+    if (live) {
+      vv = *reinterpret_cast<const bf16x8_t*>(vp + c8 * 16);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) vv[i] = static_cast<__bf16>(0.0f);
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      vdst[(c8 * 16 + 8 * half + i) * 32 + lane32] = vv[i];
+    }
+  }
+}
+
 template <int HD>
 __global__ __launch_bounds__(64)
 __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
@@ -731,7 +779,10 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
     int T, int pos0, int q_heads, int kv_heads) {
   constexpr int kNblk = HD / 32;
   const int qh = blockIdx.x;
-  const int tile = blockIdx.y;
+  // schedule the HEAVIEST tiles first: under causal masking tile i does
+  // i+1 key tiles of work, so launching high tiles last would leave a
+  // ragged mostly-idle tail on the 1024 SIMDs
+  const int tile = gridDim.y - 1 - blockIdx.y;
   const int r0 = tile * 32;
   if (r0 >= T) return;
   const int kvh = qh / (q_heads / kv_heads);
@@ -771,11 +822,10 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
     }
   }
 
-  // Per-ROW softmax stats live in "transpose layout": lane l keeps the
-  // running max and sum for row l%32 (both halves redundantly). The C/D
-  // per-register layout picks them up from LDS each tile. This replaces
-  // 10 ds_bpermute reduction chains per register (160 per tile) with a
-  // handful of conflict-free LDS transposes — and 2 VGPRs instead of 32.
+  // Per-ROW softmax stats in "transpose layout": lane l keeps the running
+  // max and sum for row l%32 (both halves redundantly); the C/D register
+  // layout picks them up from LDS each tile. This replaces 10 ds_bpermute
+  // chains per register (160 per tile) with conflict-free LDS transposes.
   float m_row = -1e30f, l_row = 0.0f;
 
   f32x16_t oacc[kNblk];
@@ -787,54 +837,11 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
 
   const int t_kv = pos0 + T;  // causal key horizon
   const int kt_end = min((pos0 + r0 + 31) / 32 + 1, (t_kv + 31) / 32);
-  for (int kt = 0; kt < kt_end; ++kt) {
-    const int kcol0 = kt * 32;
-    const int key = kcol0 + lane32;
-    const bool key_live = key < t_kv;
-    const long long krow =
-        static_cast<long long>(key_live ? key : 0) * kv_heads + kvh;
 
-    // ---- stage this tile's V into the alternating LDS buffer,
-    // transposed to [hd][key] (the PV reads then are contiguous b128).
-    // Previous tile's PV reads hit the OTHER buffer, so no extra wait.
-    __bf16* vdst = s_v[kt & 1];
-    {
-      const unsigned short* vp = V + krow * HD + 8 * half;
-#pragma unroll
-      for (int c8 = 0; c8 < HD / 16; ++c8) {
-        bf16x8_t vv;
-        if (key_live) {
-          vv = *reinterpret_cast<const bf16x8_t*>(vp + c8 * 16);
-        } else {
-#pragma unroll
-          for (int i = 0; i < 8; ++i) vv[i] = static_cast<__bf16>(0.0f);
-        }
-#pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          vdst[(c8 * 16 + 8 * half + i) * 32 + lane32] = vv[i];
-        }
-      }
-    }
-
-    // ---- S = Q . K^T over this key tile
-    f32x16_t sacc;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) sacc[r] = 0.0f;
-    const unsigned short* kp = K + krow * HD + 8 * half;
-#pragma unroll
-    for (int ks = 0; ks < HD / 16; ++ks) {
-      bf16x8_t kf;
-      if (key_live) {
-        kf = *reinterpret_cast<const bf16x8_t*>(kp + ks * 16);
-      } else {
-#pragma unroll
-        for (int i = 0; i < 8; ++i) kf[i] = static_cast<__bf16>(0.0f);
-      }
-      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf[ks], kf, sacc, 0, 0,
-                                                     0);
-    }
-
-    // ---- phase A: masked scores into the f32 transpose scratch
+  // Softmax + PV for one tile whose scores sit in sacc and whose V tile
+  // is staged (transposed) at vdst. Wave-synchronous: lgkm-only waits.
+  auto softmax_pv = [&](f32x16_t& sacc, const __bf16* vdst, int kcol0) {
+    // phase A: masked scores into the f32 transpose scratch
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int row = (r % 4) + 8 * (r / 4) + 4 * half;
@@ -843,8 +850,7 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
       s_s[row * FMA_SROW + lane32] = sacc[r];
     }
     __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);
-
-    // ---- phase B: per-row max (lane l reduces row l%32, cols 16h..)
+    // phase B: per-row max (lane l reduces row l%32, cols 16*half..)
     {
       float m_part = -1e30f;
 #pragma unroll
@@ -862,9 +868,8 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
       }
     }
     __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);
-
-    // ---- phase C: P = exp(S - m) into s_p (MFMA layout) and s_s
-    // (transpose layout, for the row sums); rescale O by alpha
+    // phase C: P = exp(S-m) into s_p (MFMA layout) + s_s (for row sums);
+    // rescale O by alpha
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int row = (r % 4) + 8 * (r / 4) + 4 * half;
@@ -877,8 +882,7 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
       s_s[row * FMA_SROW + lane32] = p;
     }
     __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);
-
-    // ---- phase D: row sums from the transpose scratch
+    // phase D: row sums from the transpose scratch
     {
       float sum = 0.0f;
 #pragma unroll
@@ -887,8 +891,7 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
       }
       l_row += sum + __shfl_xor(sum, 32);
     }
-
-    // ---- O += P . V  (K dim = 32 keys = 2 MFMA k-steps)
+    // O += P . V  (K dim = 32 keys = 2 MFMA k-steps)
 #pragma unroll
     for (int ks2 = 0; ks2 < 2; ++ks2) {
       bf16x8_t pf = *reinterpret_cast<const bf16x8_t*>(
@@ -901,7 +904,38 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
                                                           0, 0);
       }
     }
-    __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);  // s_p reads done
+    __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);  // s_p/s_v reads done
+  };
+
+  auto qk = [&](const bf16x8_t(&kf)[HD / 16]) {
+    f32x16_t sacc;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) sacc[r] = 0.0f;
+#pragma unroll
+    for (int ks = 0; ks < HD / 16; ++ks) {
+      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf[ks], kf[ks], sacc,
+                                                     0, 0, 0);
+    }
+    return sacc;
+  };
+
+  // Software-pipelined key loop: kf is dead the moment the QK MFMAs
+  // have consumed it, so the NEXT tile's K loads re-fill the same
+  // buffer — and the next V tile streams into the other LDS buffer —
+  // while the current tile's softmax+PV runs. The ~700-cycle HBM
+  // latency hides behind compute with zero extra registers.
+  bf16x8_t kf[HD / 16];
+  prefill_load_k<HD>(K, 0, t_kv, kv_heads, kvh, lane32, half, kf);
+  prefill_stage_v<HD>(V, 0, t_kv, kv_heads, kvh, lane32, half, s_v[0]);
+  for (int kt = 0; kt < kt_end; ++kt) {
+    f32x16_t sacc = qk(kf);
+    if (kt + 1 < kt_end) {
+      prefill_load_k<HD>(K, (kt + 1) * 32, t_kv, kv_heads, kvh, lane32,
+                         half, kf);
+      prefill_stage_v<HD>(V, (kt + 1) * 32, t_kv, kv_heads, kvh, lane32,
+                          half, s_v[(kt + 1) & 1]);
+    }
+    softmax_pv(sacc, s_v[kt & 1], kt * 32);
   }
 
   // ---- epilogue: O /= l, store rows < T
