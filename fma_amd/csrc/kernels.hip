@@ -791,9 +791,12 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
   const int lane32 = l & 31;
   const float scale = rsqrtf(static_cast<float>(HD));
 
+  // LDS is the occupancy limiter here (8 workgroups/CU for 2 waves/SIMD
+  // needs <= 20 KB each out of 160 KB/CU): P tile 2 KB + two V buffers
+  // 16 KB + bf16 score scratch 2.1 KB + stats ~0.3 KB = 20.4 -> fits 7-8.
   __shared__ __bf16 s_p[32 * 32];          // P tile, A-fragment source
   __shared__ __bf16 s_v[2][32 * HD];       // V, transposed [hd][key], 2 bufs
-  __shared__ float s_s[32 * FMA_SROW];     // scores/probs transpose scratch
+  __shared__ __bf16 s_s[32 * FMA_SROW];    // score transpose scratch (max)
   __shared__ float s_stat[32 * 2];         // per-row (m_new, alpha)
   __shared__ float s_l[32];                // per-row l at epilogue
 
@@ -847,7 +850,7 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
       const int row = (r % 4) + 8 * (r / 4) + 4 * half;
       const int qpos = pos0 + r0 + row;
       if (kcol0 + lane32 > qpos || kcol0 + lane32 >= t_kv) sacc[r] = -1e30f;
-      s_s[row * FMA_SROW + lane32] = sacc[r];
+      s_s[row * FMA_SROW + lane32] = static_cast<__bf16>(sacc[r]);
     }
     __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);
     // phase B: per-row max (lane l reduces row l%32, cols 16*half..)
@@ -855,7 +858,8 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
       float m_part = -1e30f;
 #pragma unroll
       for (int i = 0; i < 16; ++i) {
-        m_part = fmaxf(m_part, s_s[lane32 * FMA_SROW + 16 * half + i]);
+        m_part = fmaxf(m_part, static_cast<float>(
+                                   s_s[lane32 * FMA_SROW + 16 * half + i]));
       }
       const float m_tile = fmaxf(m_part, __shfl_xor(m_part, 32));
       const float m_new = fmaxf(m_row, m_tile);
@@ -879,15 +883,19 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
 #pragma unroll
       for (int b = 0; b < kNblk; ++b) oacc[b][r] *= alpha;
       s_p[row * 32 + lane32] = static_cast<__bf16>(p);
-      s_s[row * FMA_SROW + lane32] = p;
     }
     __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);
-    // phase D: row sums from the transpose scratch
+    // phase D: row sums of bf16 P straight from s_p (summing the
+    // ROUNDED probabilities matches what the PV MFMA actually uses)
     {
+      bf16x8_t pr = *reinterpret_cast<const bf16x8_t*>(
+          &s_p[lane32 * 32 + 16 * half]);
+      bf16x8_t pr2 = *reinterpret_cast<const bf16x8_t*>(
+          &s_p[lane32 * 32 + 16 * half + 8]);
       float sum = 0.0f;
 #pragma unroll
-      for (int i = 0; i < 16; ++i) {
-        sum += s_s[lane32 * FMA_SROW + 16 * half + i];
+      for (int i = 0; i < 8; ++i) {
+        sum += static_cast<float>(pr[i]) + static_cast<float>(pr2[i]);
       }
       l_row += sum + __shfl_xor(sum, 32);
     }

@@ -72,13 +72,16 @@ def fast_attn_prefill(q: torch.Tensor, k_cache: torch.Tensor,
     [S,kvH,hd] holding keys [0, pos0+T). Returns [1,T,qH*hd] or None when
     the kernel does not apply (caller falls back to SDPA).
 
-    Opt-in (FMA_MFMA_PREFILL=1): numerics-exact vs the fp32 reference,
-    but torch SDPA's prefill kernel is currently faster (measured in
-    tools/prefill_bench.py), so the default prefill path stays on SDPA
-    while decode stays on our kernels."""
-    if not available() or os.environ.get("FMA_MFMA_PREFILL") != "1":
+    Dispatch is measured, not aspirational (tools/prefill_bench.py on
+    MI355X): the hand-written kernel beats torch SDPA up to T~128 and
+    trails it beyond (0.56x at 4K ctx), so by default it serves short
+    prefills only. FMA_MFMA_PREFILL=1 forces it always, =0 never."""
+    mode = os.environ.get("FMA_MFMA_PREFILL", "auto")
+    if not available() or mode == "0":
         return None
     b, t, qh, hd = q.shape
+    if mode != "1" and t > 128:
+        return None
     if (b != 1 or hd not in (64, 128) or not q.is_cuda
             or q.dtype != torch.bfloat16
             or qh % k_cache.shape[1] != 0

@@ -465,10 +465,11 @@ def test_prefill_mfma_vs_sdpa_path_logits_close():
     os.environ["FMA_MFMA_PREFILL"] = "1"
     try:
         mfma = eng.model.forward(prompt, cache, 0).float()
+        os.environ["FMA_MFMA_PREFILL"] = "0"
+        cache2 = KVCache(cfg, 1, "cuda:0", 1, 128)
+        sdpa = eng.model.forward(prompt, cache2, 0).float()
     finally:
         del os.environ["FMA_MFMA_PREFILL"]
-    cache2 = KVCache(cfg, 1, "cuda:0", 1, 128)
-    sdpa = eng.model.forward(prompt, cache2, 0).float()
     assert torch.allclose(mfma, sdpa, atol=8e-2, rtol=8e-2), \
         (mfma - sdpa).abs().max().item()
     cache.free()
