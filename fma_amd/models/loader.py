@@ -95,13 +95,46 @@ def save_params(params: Dict[str, torch.Tensor], path: str,
             }, f)
 
 
+# Megatron-style sharding of the llama parameter set (models/llama.py
+# param_specs): column-parallel weights split output rows, row-parallel
+# weights split input columns, everything else is replicated.
+_COL_PARALLEL = {"wq", "wk", "wv", "w_gate", "w_up"}
+_ROW_PARALLEL = {"wo", "w_down"}
+
+
+def shard_slice(name: str, tensor: torch.Tensor, tp_rank: int,
+                tp_size: int) -> torch.Tensor:
+    """Slice one full (unsharded) checkpoint tensor down to the Megatron
+    shard that rank `tp_rank` of `tp_size` holds. Views, no copies."""
+    if tp_size <= 1:
+        return tensor
+    parts = name.split(".")
+    leaf = parts[-2] if len(parts) >= 2 else ""
+    if leaf in _COL_PARALLEL:
+        n = tensor.shape[0]
+        if n % tp_size != 0:
+            raise ValueError(
+                f"{name}: dim0 {n} not divisible by tp_size {tp_size} "
+                "(tp_size > num_kv_heads is unsupported for checkpoints)")
+        step = n // tp_size
+        return tensor[tp_rank * step:(tp_rank + 1) * step]
+    if leaf in _ROW_PARALLEL:
+        n = tensor.shape[1]
+        if n % tp_size != 0:
+            raise ValueError(f"{name}: dim1 {n} not divisible by tp_size")
+        step = n // tp_size
+        return tensor[:, tp_rank * step:(tp_rank + 1) * step]
+    return tensor
+
+
 def load_into_params(path: str, params: Dict[str, torch.Tensor],
-                     strict: bool = True) -> int:
+                     strict: bool = True, tp_rank: int = 0,
+                     tp_size: int = 1) -> int:
     """Copy checkpoint tensors into existing (arena-view) parameters.
 
-    Returns the number of tensors loaded. TP sharding is not resolved
-    here — shard-aware names must already match (rank-sharded checkpoints
-    carry the shard in their filename/layout).
+    Returns the number of tensors loaded. Checkpoints store the full
+    (unsharded) tensors; with tp_size > 1 each rank slices out its
+    Megatron shard (shard_slice) before copying.
     """
     loaded = 0
     seen = set()
@@ -112,9 +145,10 @@ def load_into_params(path: str, params: Dict[str, torch.Tensor],
                                "matching parameter")
             continue
         p = params[name]
+        tensor = shard_slice(name, tensor, tp_rank, tp_size)
         if tuple(tensor.shape) != tuple(p.shape):
             raise ValueError(f"shape mismatch for {name}: checkpoint "
-                             f"{tuple(tensor.shape)} vs param "
+                             f"shard {tuple(tensor.shape)} vs param "
                              f"{tuple(p.shape)}")
         p.copy_(tensor.to(p.dtype))
         seen.add(name)

@@ -165,7 +165,9 @@ class ActuationEngine:
         if self.actuation_mode != "arena" or self.arena is None:
             from fma_amd.models import loader
             t0 = _t.perf_counter()
-            loader.load_into_params(path, self.params)
+            loader.load_into_params(path, self.params,
+                                    tp_rank=self.tp_rank,
+                                    tp_size=self.tp_size)
             return _t.perf_counter() - t0
 
         t0 = _t.perf_counter()
@@ -177,6 +179,9 @@ class ActuationEngine:
             if name not in self.layout:
                 raise KeyError(f"checkpoint tensor {name!r} unknown")
             off, shape, dtype = self.layout[name]
+            tensor = loader.shard_slice(tensor=tensor, name=name,
+                                        tp_rank=self.tp_rank,
+                                        tp_size=self.tp_size)
             if tuple(tensor.shape) != tuple(shape):
                 raise ValueError(f"shape mismatch for {name}")
             raw = tensor.to(dtype).contiguous().view(torch.uint8).view(-1)

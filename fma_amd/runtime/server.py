@@ -76,12 +76,20 @@ def _worker_entry(rank: int, world: int, master_port: int,
     from fma_amd.runtime.engine import ActuationEngine
 
     ctx = tp.init_tp(rank, world, master_port, list(range(world)))
-    cfg = LlamaConfig.by_name(model_name)
+    checkpoint = model_name if os.path.isdir(model_name) else None
+    if checkpoint:
+        from fma_amd.models import loader
+        cfg = loader.config_from_dir(checkpoint) or LlamaConfig.tiny()
+    else:
+        cfg = LlamaConfig.by_name(model_name)
     if max_model_len:
         cfg.max_seq_len = max_model_len
     eng = ActuationEngine(cfg, device_index=ctx.device_index,
                           tp_rank=rank, tp_size=world,
-                          tp_group=ctx.device_group, seed=seed)
+                          tp_group=ctx.device_group, seed=seed,
+                          init_weights=checkpoint is None)
+    if checkpoint:
+        eng.load_checkpoint(checkpoint)
     tp.run_worker_loop(ctx, eng)
 
 
@@ -112,12 +120,9 @@ class ServingRuntime:
             from fma_amd.models import loader
             checkpoint = args.model
             cfg = loader.config_from_dir(checkpoint) or LlamaConfig.tiny()
-            if world > 1:
-                raise NotImplementedError(
-                    "checkpoint loading with --tensor-parallel-size > 1 "
-                    "needs shard-aware slicing (checkpoints store unsharded "
-                    "tensors); load a TP-sharded checkpoint per rank or use "
-                    "random-init models for TP instances")
+            # TP > 1: each rank slices its Megatron shard out of the full
+            # checkpoint (loader.shard_slice); workers load their own in
+            # _worker_entry before entering the command loop
         else:
             cfg = LlamaConfig.by_name(args.model)
         if args.max_model_len:

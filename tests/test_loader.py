@@ -81,3 +81,85 @@ def test_fast_load_missing_param_raises(tmp_path):
     import pytest as _pytest
     with _pytest.raises(KeyError):
         dst.load_checkpoint(ckpt)
+
+
+# -- TP-sharded checkpoint loading -------------------------------------------
+
+def _tp_cfg():
+    return LlamaConfig(name="tpc", vocab_size=64, hidden_size=64,
+                       intermediate_size=96, num_layers=2, num_heads=4,
+                       num_kv_heads=2, max_seq_len=32)
+
+
+def test_shard_slice_reassembles_full_tensor():
+    torch.manual_seed(5)
+    full = {
+        "layers.0.wq.weight": torch.randn(16, 8),      # col-parallel (dim0)
+        "layers.0.wo.weight": torch.randn(8, 16),      # row-parallel (dim1)
+        "layers.0.attn_norm.weight": torch.randn(8),   # replicated
+        "embed.weight": torch.randn(10, 8),            # replicated
+    }
+    for name, t in full.items():
+        shards = [loader.shard_slice(name, t, r, 2) for r in range(2)]
+        if "wq" in name:
+            assert torch.equal(torch.cat(shards, dim=0), t)
+        elif "wo" in name:
+            assert torch.equal(torch.cat(shards, dim=1), t)
+        else:
+            assert all(torch.equal(s, t) for s in shards)
+
+
+def test_shard_slice_rejects_indivisible():
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        loader.shard_slice("layers.0.wk.weight", torch.randn(3, 8), 0, 2)
+
+
+def test_load_into_params_tp_sharded(tmp_path):
+    """A full checkpoint loads into each rank's Megatron shard; the
+    shards stitched back together equal the full parameters."""
+    cfg = _tp_cfg()
+    src = ActuationEngine(cfg, seed=9)
+    ckpt = str(tmp_path / "full-ckpt")
+    loader.save_params(src.params, ckpt, cfg)
+
+    col = {"wq", "wk", "wv", "w_gate", "w_up"}
+    row = {"wo", "w_down"}
+    for tp_size in (2,):
+        shards = []
+        for r in range(tp_size):
+            params = {n: torch.zeros(s, dtype=d)
+                      for n, s, d in cfg.param_specs(r, tp_size)}
+            n = loader.load_into_params(ckpt, params,
+                                        tp_rank=r, tp_size=tp_size)
+            assert n == len(params)
+            shards.append(params)
+        for name, fullp in src.params.items():
+            leaf = name.split(".")[-2]
+            if leaf in col:
+                glued = torch.cat([s[name] for s in shards], dim=0)
+            elif leaf in row:
+                glued = torch.cat([s[name] for s in shards], dim=1)
+            else:
+                glued = shards[0][name]
+            assert torch.equal(glued, fullp), name
+
+
+def test_engine_tp_rank_load_checkpoint(tmp_path):
+    """An engine constructed as TP rank 1 of 2 loads its shard straight
+    from a full checkpoint through the arena staging path."""
+    cfg = _tp_cfg()
+    src = ActuationEngine(cfg, seed=13)
+    ckpt = str(tmp_path / "full2")
+    loader.save_params(src.params, ckpt, cfg)
+    eng = ActuationEngine(cfg, seed=1, tp_rank=1, tp_size=2,
+                          init_weights=False)
+    eng.load_checkpoint(ckpt)
+    q_local = cfg.num_heads // 2 * cfg.head_dim
+    assert torch.equal(eng.params["layers.0.wq.weight"],
+                       src.params["layers.0.wq.weight"][q_local:])
+    i_local = cfg.intermediate_size // 2
+    assert torch.equal(eng.params["layers.1.w_down.weight"],
+                       src.params["layers.1.w_down.weight"][:, i_local:])
+    assert torch.equal(eng.params["final_norm.weight"],
+                       src.params["final_norm.weight"])
