@@ -20,6 +20,7 @@ def main():
     ap.add_argument("--debug-accelerator-memory", type=int, default=None,
                     help="MiB budget per sleeping accelerator")
     ap.add_argument("--metrics-port", type=int, default=8002)
+    ap.add_argument("--debug-port", type=int, default=8003)
     args = ap.parse_args()
     store = StoreClient(args.store_url, actor="dual-pods-controller")
     ctl = DualPodsController(
@@ -32,6 +33,7 @@ def main():
                 args.debug_accelerator_memory * args.sleeper_limit
                 if args.debug_accelerator_memory else None)))
     metrics.serve_metrics(args.metrics_port)
+    metrics.serve_debug(args.debug_port)
     ctl.start()
     try:
         while True:

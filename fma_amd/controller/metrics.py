@@ -107,6 +107,55 @@ def serve_metrics(port: int = 8002) -> None:
         start_http_server(port)
 
 
+def serve_debug(port: int = 8003):
+    """The reference's debug listener analog (prom-and-debug.go:68-79
+    serves Go /debug/pprof on :8003). Go's pprof has no direct Python
+    equivalent; this serves the operational 90% — "where is it stuck" —
+    as /debug/threads (a live stack dump of every thread) plus
+    /debug/vars (gc + thread counts). Returns the server (daemon thread).
+    """
+    import gc
+    import http.server
+    import json
+    import sys
+    import threading
+    import traceback
+
+    class Handler(http.server.BaseHTTPRequestHandler):
+        def do_GET(self):  # noqa: N802
+            if self.path.startswith("/debug/threads"):
+                names = {t.ident: t.name for t in threading.enumerate()}
+                out = []
+                for tid, frame in sys._current_frames().items():
+                    out.append(f"--- thread {tid} ({names.get(tid, '?')})")
+                    out.extend(x.rstrip()
+                               for x in traceback.format_stack(frame))
+                body = ("\n".join(out) + "\n").encode()
+                ctype = "text/plain"
+            elif self.path.startswith("/debug/vars"):
+                body = json.dumps({
+                    "threads": threading.active_count(),
+                    "gc": gc.get_count(),
+                }).encode()
+                ctype = "application/json"
+            else:
+                self.send_error(404)
+                return
+            self.send_response(200)
+            self.send_header("Content-Type", ctype)
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+        def log_message(self, *a):  # quiet
+            pass
+
+    srv = http.server.ThreadingHTTPServer(("0.0.0.0", port), Handler)
+    threading.Thread(target=srv.serve_forever, daemon=True,
+                     name="fma-debug-http").start()
+    return srv
+
+
 def observe_http(purpose: str, method: str, status: int, seconds: float
                  ) -> None:
     http_latency_seconds().labels(purpose, method, str(status)).observe(seconds)

@@ -13,12 +13,14 @@ def main():
     ap.add_argument("--namespace", default="default")
     ap.add_argument("--key-workers", type=int, default=4)
     ap.add_argument("--metrics-port", type=int, default=8004)
+    ap.add_argument("--debug-port", type=int, default=8005)
     args = ap.parse_args()
     pop = LauncherPopulator(StoreClient(args.store_url,
                                         actor="launcher-populator"),
                             namespace=args.namespace,
                             key_workers=args.key_workers)
     metrics.serve_metrics(args.metrics_port)
+    metrics.serve_debug(args.debug_port)
     pop.start()
     try:
         while True:

@@ -42,6 +42,7 @@ def main() -> None:
     ap.add_argument("--node-name", default="node-1")
     ap.add_argument("--store-port", type=int, default=8081)
     ap.add_argument("--metrics-port", type=int, default=8002)
+    ap.add_argument("--debug-port", type=int, default=8003)
     ap.add_argument("--sleeper-limit", type=int, default=1)
     ap.add_argument("--gpus", type=int, default=None,
                     help="GPUs to advertise on the Node (default: detect)")
@@ -84,6 +85,7 @@ def main() -> None:
     pop = LauncherPopulator(store)
     pop.start()
     metrics.serve_metrics(args.metrics_port)
+    metrics.serve_debug(args.debug_port)
 
     app = create_app(store)
     print(f"fma-amd single-node stack up: store http://127.0.0.1:"

@@ -112,3 +112,32 @@ def test_outdated_routing_event_recorded():
     w["ctl"]._process(infsvr_item(w["store"]))
     events = w["store"].list("Event")
     assert any(e["reason"] == "OutdatedRoutingMetadata" for e in events)
+
+
+def test_debug_endpoint_serves_thread_stacks():
+    """The reference's :8003 debug listener analog
+    (pkg/observability/prom-and-debug.go:68-79): /debug/threads dumps
+    live stacks, /debug/vars serves counters, unknown paths 404."""
+    import json
+    import urllib.request
+
+    from fma_amd.controller import metrics as M
+
+    srv = M.serve_debug(0)
+    port = srv.server_address[1]
+    try:
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/debug/threads", timeout=5).read()
+        assert b"--- thread" in body
+        assert b"test_debug_endpoint" in body  # this very frame
+        v = json.loads(urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/debug/vars", timeout=5).read())
+        assert v["threads"] >= 1
+        try:
+            urllib.request.urlopen(
+                f"http://127.0.0.1:{port}/nope", timeout=5)
+            assert False, "expected 404"
+        except urllib.error.HTTPError as e:
+            assert e.code == 404
+    finally:
+        srv.shutdown()
