@@ -20,6 +20,7 @@ from __future__ import annotations
 import statistics
 import time
 from dataclasses import dataclass, field
+from enum import Enum
 from typing import Any, Dict, List, Optional
 
 
@@ -175,6 +176,18 @@ class DualPodsBenchmark:
                 self.ops.delete_requester(name)
         return report
 
+    def run_new_variant(self, iscs: List[str], n: int = 2
+                        ) -> List[BenchReport]:
+        """Introduce model variants one after another; each variant gets
+        its own baseline pass tagged variant-<isc> (reference
+        scenarios.py:271-330 run_new_variant_scenario)."""
+        reports = []
+        for isc in iscs:
+            r = self.run_baseline(isc, n=n)
+            r.scenario = f"variant-{isc}"
+            reports.append(r)
+        return reports
+
     def run_scaling(self, isc: str, n: int = 4) -> BenchReport:
         """N concurrent requesters (reference scenarios.py scaling)."""
         report = BenchReport("scaling")
@@ -190,3 +203,80 @@ class DualPodsBenchmark:
         for name in names:
             self.ops.delete_requester(name)
         return report
+
+
+class ScenarioStatus(Enum):
+    """Reference benchmark_diagnostics.py:24 (SUCCESS/FAILURE)."""
+    SUCCESS = 1
+    FAILURE = 2
+
+
+@dataclass
+class ScenarioResult:
+    """Outcome + context of one scenario run, the diagnosis input
+    (reference benchmark_diagnostics.py ScenarioResult:56-71)."""
+    status: ScenarioStatus
+    report: Optional[BenchReport] = None
+    failed_requester: str = ""
+    unready: List[str] = field(default_factory=list)
+    error: str = ""
+
+
+class BenchmarkDiagnosis:
+    """Collect post-mortem state for a failing scenario before exiting
+    (reference benchmark_diagnostics.py BenchmarkDiagnosis:73-180: dumps
+    dual-pods controller logs and failing pod descriptions via kubectl;
+    here the store IS the cluster, so the dump is the controller's log
+    records plus the JSON of every Pod involved)."""
+
+    def __init__(self, store=None, log_records: Optional[List[str]] = None):
+        self.store = store
+        self.log_records = log_records or []
+
+    def collect_diagnostics(self, result: ScenarioResult,
+                            out_dir: str) -> List[str]:
+        import json
+        import os
+        os.makedirs(out_dir, exist_ok=True)
+        written = []
+
+        def dump(fname: str, text: str) -> None:
+            p = os.path.join(out_dir, fname)
+            with open(p, "w") as f:
+                f.write(text)
+            written.append(p)
+
+        dump("scenario-result.json", json.dumps({
+            "status": result.status.name,
+            "failed_requester": result.failed_requester,
+            "unready": result.unready,
+            "error": result.error,
+            "summary": result.report.summary() if result.report else None,
+        }, indent=2))
+        if self.log_records:
+            dump("dual-pods-controller.log", "\n".join(self.log_records))
+        if self.store is not None:
+            pods = {}
+            for pod in self.store.list("Pod"):
+                name = pod["metadata"]["name"]
+                if (name == result.failed_requester or name in result.unready
+                        or "launcher" in name):
+                    pods[name] = pod
+            dump("pods.json", json.dumps(pods, indent=2, default=str))
+        return written
+
+
+def query_gpu_usage() -> Dict[str, Any]:
+    """Accelerator memory in use, per GPU (reference benchmark_base.py:330
+    query_gpu_usage shells nvidia-smi; MI355X uses rocm-smi)."""
+    import json
+    import subprocess
+    try:
+        out = subprocess.run(
+            ["rocm-smi", "--showmeminfo", "vram", "--json"],
+            capture_output=True, text=True, timeout=20)
+        if out.returncode != 0:
+            return {}
+        return json.loads(out.stdout or "{}")
+    except (OSError, ValueError, subprocess.SubprocessError):
+        return {}
