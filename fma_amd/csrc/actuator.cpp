@@ -981,10 +981,22 @@ at::Tensor attn_prefill_bf16(const at::Tensor& q, const at::Tensor& k,
   TORCH_CHECK(pos0 >= 0 && pos0 + T <= k.size(0), "keys out of cache");
   auto out = at::empty_like(q);
   auto stream = c10::hip::getCurrentHIPStream(q.device().index());
+  const int chunks = fma_attn_prefill_chunks(
+      static_cast<int>(T), static_cast<int>(pos0),
+      static_cast<int>(q_heads));
+  at::Tensor partials;
+  float* pptr = nullptr;
+  if (chunks > 1) {
+    const int64_t tiles = (T + 31) / 32;
+    partials = at::empty({q_heads, tiles, chunks, 32, hd + 2},
+                         q.options().dtype(at::kFloat));
+    pptr = partials.data_ptr<float>();
+  }
   FMA_HIP_CHECK(fma_launch_attn_prefill_bf16(
       q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
       static_cast<int>(T), static_cast<int>(pos0), static_cast<int>(q_heads),
-      static_cast<int>(kv_heads), static_cast<int>(hd), stream.stream()));
+      static_cast<int>(kv_heads), static_cast<int>(hd), pptr, chunks,
+      stream.stream()));
   return out;
 }
 

@@ -57,6 +57,9 @@ extern "C" hipError_t fma_launch_attn_decode_bf16(
     int q_heads, int kv_heads, int hd, long long k_stride,
     float* partials, int chunks, hipStream_t stream);
 
+extern "C" int fma_attn_prefill_chunks(int T, int pos0, int q_heads);
+
 extern "C" hipError_t fma_launch_attn_prefill_bf16(
     const void* Q, const void* K, const void* V, void* O, int T, int pos0,
-    int q_heads, int kv_heads, int hd, hipStream_t stream);
+    int q_heads, int kv_heads, int hd, float* partials, int chunks,
+    hipStream_t stream);

@@ -776,7 +776,13 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
     const unsigned short* __restrict__ K,  // [S, kvH, hd]
     const unsigned short* __restrict__ V,  // [S, kvH, hd]
     unsigned short* __restrict__ O,        // [T, qH, hd]
+    float* __restrict__ partials,          // [qH,tiles,chunks,32,hd+2]|null
     int T, int pos0, int q_heads, int kv_heads) {
+  // Split-sequence mode (gridDim.z > 1): small prefills launch too few
+  // waves to fill 1024 SIMDs (T=512 -> 512 waves), so chunk blockIdx.z
+  // takes key tiles {z, z+chunks, ...} (strided keeps the causal load
+  // balanced) and writes unnormalized (m, l, O) partials; the combine
+  // kernel below merges them. Same recipe as attn_decode.
   constexpr int kNblk = HD / 32;
   const int qh = blockIdx.x;
   // schedule the HEAVIEST tiles first: under causal masking tile i does
@@ -932,18 +938,46 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
   // buffer — and the next V tile streams into the other LDS buffer —
   // while the current tile's softmax+PV runs. The ~700-cycle HBM
   // latency hides behind compute with zero extra registers.
+  const int chunk = blockIdx.z;
+  const int chunks = gridDim.z;
   bf16x8_t kf[HD / 16];
-  prefill_load_k<HD>(K, 0, t_kv, kv_heads, kvh, lane32, half, kf);
-  prefill_stage_v<HD>(V, 0, t_kv, kv_heads, kvh, lane32, half, s_v[0]);
-  for (int kt = 0; kt < kt_end; ++kt) {
+  if (chunk < kt_end) {
+    prefill_load_k<HD>(K, chunk * 32, t_kv, kv_heads, kvh, lane32, half,
+                       kf);
+    prefill_stage_v<HD>(V, chunk * 32, t_kv, kv_heads, kvh, lane32, half,
+                        s_v[0]);
+  }
+  int par = 0;  // ping-pong parity of the V LDS buffer
+  for (int kt = chunk; kt < kt_end; kt += chunks, par ^= 1) {
     f32x16_t sacc = qk(kf);
-    if (kt + 1 < kt_end) {
-      prefill_load_k<HD>(K, (kt + 1) * 32, t_kv, kv_heads, kvh, lane32,
-                         half, kf);
-      prefill_stage_v<HD>(V, (kt + 1) * 32, t_kv, kv_heads, kvh, lane32,
-                          half, s_v[(kt + 1) & 1]);
+    if (kt + chunks < kt_end) {
+      prefill_load_k<HD>(K, (kt + chunks) * 32, t_kv, kv_heads, kvh,
+                         lane32, half, kf);
+      prefill_stage_v<HD>(V, (kt + chunks) * 32, t_kv, kv_heads, kvh,
+                          lane32, half, s_v[par ^ 1]);
     }
-    softmax_pv(sacc, s_v[kt & 1], kt * 32);
+    softmax_pv(sacc, s_v[par], kt * 32);
+  }
+
+  if (partials != nullptr) {
+    // unnormalized partial: [m, l, O(hd)] per row
+    const int tiles = gridDim.y;
+    float* base = partials +
+        ((static_cast<long long>(qh) * tiles + tile) * chunks + chunk) *
+            (32 * (HD + 2));
+    if (half == 0) {
+      base[lane32 * (HD + 2) + 0] = m_row;
+      base[lane32 * (HD + 2) + 1] = l_row;
+    }
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r % 4) + 8 * (r / 4) + 4 * half;
+#pragma unroll
+      for (int b = 0; b < kNblk; ++b) {
+        base[row * (HD + 2) + 2 + b * 32 + lane32] = oacc[b][r];
+      }
+    }
+    return;
   }
 
   // ---- epilogue: O /= l, store rows < T
@@ -965,26 +999,82 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
   }
 }
 
+// Merge the per-chunk (m, l, O) partials: log-sum-exp reweighting, then
+// normalize and store bf16 (reference for the math: the decode combine
+// kernel above — identical recipe, prefill shapes).
+__global__ void attn_prefill_combine_kernel(
+    const float* __restrict__ partials,  // [qH, tiles, chunks, 32, hd+2]
+    unsigned short* __restrict__ O,      // [T, qH, hd]
+    int T, int q_heads, int hd, int chunks) {
+  const int qh = blockIdx.x;
+  const int tile = blockIdx.y;
+  const int tiles = gridDim.y;
+  const int col = threadIdx.x;  // hd threads
+  const float* tbase = partials +
+      (static_cast<long long>(qh) * tiles + tile) * chunks *
+          (32 * (hd + 2));
+  for (int row = 0; row < 32; ++row) {
+    const int orow = tile * 32 + row;
+    if (orow >= T) continue;
+    float m = -1e30f;
+    for (int c = 0; c < chunks; ++c) {
+      m = fmaxf(m, tbase[(c * 32 + row) * (hd + 2) + 0]);
+    }
+    float l = 0.0f, acc = 0.0f;
+    for (int c = 0; c < chunks; ++c) {
+      const float* rb = tbase + (c * 32 + row) * (hd + 2);
+      const float w = __expf(rb[0] - m);
+      l += rb[1] * w;
+      acc += rb[2 + col] * w;
+    }
+    const float inv_l = l > 0.0f ? 1.0f / l : 0.0f;
+    O[(static_cast<long long>(orow) * q_heads + qh) * hd + col] =
+        f32_to_bf16(acc * inv_l);
+  }
+}
+
 }  // namespace
+
+extern "C" int fma_attn_prefill_chunks(int T, int pos0, int q_heads) {
+  // enough waves to fill 1024 SIMDs at occupancy 2; each chunk should
+  // still have a few key tiles of work on the biggest rows
+  const int tiles = (T + 31) / 32;
+  const long long waves = static_cast<long long>(q_heads) * tiles;
+  if (waves >= 1536) return 1;
+  int chunks = static_cast<int>(2048 / (waves > 0 ? waves : 1));
+  if (chunks < 1) chunks = 1;
+  if (chunks > 8) chunks = 8;
+  const int key_tiles = (pos0 + T + 31) / 32;
+  if (chunks > key_tiles) chunks = key_tiles;
+  return chunks;
+}
 
 extern "C" hipError_t fma_launch_attn_prefill_bf16(
     const void* Q, const void* K, const void* V, void* O, int T, int pos0,
-    int q_heads, int kv_heads, int hd, hipStream_t stream) {
+    int q_heads, int kv_heads, int hd, float* partials, int chunks,
+    hipStream_t stream) {
   if (hd != 64 && hd != 128) return hipErrorInvalidValue;
   if (q_heads % kv_heads != 0) return hipErrorInvalidValue;
-  dim3 grid(q_heads, (T + 31) / 32);
+  if (chunks > 1 && partials == nullptr) return hipErrorInvalidValue;
+  dim3 grid(q_heads, (T + 31) / 32, chunks);
+  float* par = chunks > 1 ? partials : nullptr;
   if (hd == 128) {
    hipLaunchKernelGGL(( attn_prefill_bf16_kernel<128>), dim3(grid), dim3(64), 0, stream, 
         static_cast<const unsigned short*>(Q),
         static_cast<const unsigned short*>(K),
         static_cast<const unsigned short*>(V),
-        static_cast<unsigned short*>(O), T, pos0, q_heads, kv_heads);
+        static_cast<unsigned short*>(O), par, T, pos0, q_heads, kv_heads);
   } else {
    hipLaunchKernelGGL(( attn_prefill_bf16_kernel<64>), dim3(grid), dim3(64), 0, stream, 
         static_cast<const unsigned short*>(Q),
         static_cast<const unsigned short*>(K),
         static_cast<const unsigned short*>(V),
-        static_cast<unsigned short*>(O), T, pos0, q_heads, kv_heads);
+        static_cast<unsigned short*>(O), par, T, pos0, q_heads, kv_heads);
+  }
+  if (chunks > 1) {
+    dim3 cgrid(q_heads, (T + 31) / 32);
+   hipLaunchKernelGGL(( attn_prefill_combine_kernel), dim3(cgrid), dim3(hd), 0, stream, 
+        partials, static_cast<unsigned short*>(O), T, q_heads, hd, chunks);
   }
   return hipGetLastError();
 }
