@@ -32,6 +32,7 @@
 #include <algorithm>
 #include <atomic>
 #include <chrono>
+#include <cstdlib>
 #include <cstring>
 #include <mutex>
 #include <thread>
@@ -965,7 +966,8 @@ at::Tensor attn_decode_bf16(const at::Tensor& q, const at::Tensor& k,
 // Causal GQA prefill attention on MFMA: Q [T, qH, hd] against the
 // [S, kvH, hd] cache slices holding keys [0, pos0 + T).
 at::Tensor attn_prefill_bf16(const at::Tensor& q, const at::Tensor& k,
-                             const at::Tensor& v, int64_t pos0) {
+                             const at::Tensor& v, int64_t pos0,
+                             int64_t chunks_arg) {
   TORCH_CHECK(q.dim() == 3 && q.is_cuda() &&
               q.scalar_type() == at::kBFloat16 && q.is_contiguous(),
               "q must be contiguous bf16 [T, qH, hd]");
@@ -981,9 +983,17 @@ at::Tensor attn_prefill_bf16(const at::Tensor& q, const at::Tensor& k,
   TORCH_CHECK(pos0 >= 0 && pos0 + T <= k.size(0), "keys out of cache");
   auto out = at::empty_like(q);
   auto stream = c10::hip::getCurrentHIPStream(q.device().index());
-  const int chunks = fma_attn_prefill_chunks(
-      static_cast<int>(T), static_cast<int>(pos0),
-      static_cast<int>(q_heads));
+  int chunks = static_cast<int>(chunks_arg);
+  if (chunks <= 0) {
+    const char* env = std::getenv("FMA_PREFILL_CHUNKS");
+    chunks = env ? std::atoi(env)
+                 : fma_attn_prefill_chunks(static_cast<int>(T),
+                                           static_cast<int>(pos0),
+                                           static_cast<int>(q_heads));
+    if (chunks < 1) chunks = 1;
+  }
+  const int max_chunks = static_cast<int>((pos0 + T + 31) / 32);
+  if (chunks > max_chunks) chunks = max_chunks;
   at::Tensor partials;
   float* pptr = nullptr;
   if (chunks > 1) {
@@ -1028,7 +1038,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("eps"));
   m.def("silu_mul_bf16", &silu_mul_bf16, py::arg("g"), py::arg("u"));
   m.def("attn_prefill_bf16", &attn_prefill_bf16, py::arg("q"), py::arg("k"),
-        py::arg("v"), py::arg("pos0"));
+        py::arg("v"), py::arg("pos0"), py::arg("chunks") = 0);
   m.def("attn_decode_bf16", &attn_decode_bf16, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("t"));
   m.def("rope1_bf16_", &rope1_bf16_, py::arg("q"), py::arg("cos_row"),

@@ -1036,17 +1036,14 @@ __global__ void attn_prefill_combine_kernel(
 }  // namespace
 
 extern "C" int fma_attn_prefill_chunks(int T, int pos0, int q_heads) {
-  // enough waves to fill 1024 SIMDs at occupancy 2; each chunk should
-  // still have a few key tiles of work on the biggest rows
-  const int tiles = (T + 31) / 32;
-  const long long waves = static_cast<long long>(q_heads) * tiles;
-  if (waves >= 1536) return 1;
-  int chunks = static_cast<int>(2048 / (waves > 0 ? waves : 1));
-  if (chunks < 1) chunks = 1;
-  if (chunks > 8) chunks = 8;
-  const int key_tiles = (pos0 + T + 31) / 32;
-  if (chunks > key_tiles) chunks = key_tiles;
-  return chunks;
+  // Measured on MI355X (tools/prefill_bench.py): splitting the key range
+  // never beat the plain path at llama shapes — the partials traffic
+  // (qH*tiles*chunks*32*(hd+2) f32) plus the combine launch outweigh the
+  // extra occupancy even at T=128 (0.039 ms split vs 0.016 plain). The
+  // mechanism stays (correctness-tested) for shapes where it may pay
+  // (tiny qH, huge pos0): force via FMA_PREFILL_CHUNKS or the binding.
+  (void)T; (void)pos0; (void)q_heads;
+  return 1;
 }
 
 extern "C" hipError_t fma_launch_attn_prefill_bf16(

@@ -431,6 +431,10 @@ def test_attn_prefill_mfma_matches_fp32_reference():
         k = torch.randn(S, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
         v = torch.randn(S, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
         out = C.attn_prefill_bf16(q, k, v, pos0)
+        # split-sequence path (partials + combine) must agree too
+        out_split = C.attn_prefill_bf16(q, k, v, pos0, chunks=3)
+        assert torch.allclose(out_split.float(), out.float(),
+                              atol=2e-2, rtol=2e-2)
         # plain fp32 reference with an explicit causal mask
         t_kv = pos0 + T
         rep = qH // kvH
