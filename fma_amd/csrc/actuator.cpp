@@ -958,8 +958,44 @@ at::Tensor attn_decode_bf16(const at::Tensor& q, const at::Tensor& k,
       q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
       static_cast<int>(t), static_cast<int>(q_heads),
       static_cast<int>(kv_heads), static_cast<int>(hd),
-      static_cast<long long>(kv_heads * hd), pptr, chunks,
+      static_cast<long long>(kv_heads * hd), pptr, chunks, nullptr,
       stream.stream()));
+  return out;
+}
+
+// hipGraph-capturable decode attention: the sequence length is read from
+// a device int32 scalar at kernel run time, so one captured launch
+// geometry (sized for max_t) replays correctly as the cache grows.
+at::Tensor attn_decode_bf16_graph(const at::Tensor& q, const at::Tensor& k,
+                                  const at::Tensor& v,
+                                  const at::Tensor& t_dev, int64_t max_t) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+              q.is_contiguous(), "q must be contiguous bf16 [qH, hd]");
+  TORCH_CHECK(k.dim() == 3 && k.is_contiguous() && v.is_contiguous(),
+              "k/v must be contiguous [S, kvH, hd]");
+  TORCH_CHECK(t_dev.is_cuda() && t_dev.scalar_type() == at::kInt &&
+              t_dev.numel() == 1, "t_dev must be a device int32 scalar");
+  const int64_t q_heads = q.size(0), hd = q.size(1);
+  const int64_t kv_heads = k.size(1);
+  TORCH_CHECK(k.size(2) == hd && v.sizes() == k.sizes(), "shape mismatch");
+  TORCH_CHECK(max_t >= 1 && max_t <= k.size(0), "max_t out of cache bounds");
+  auto out = at::empty_like(q);
+  auto stream = c10::hip::getCurrentHIPStream(q.device().index());
+  const int chunks = fma_attn_decode_chunks(static_cast<int>(max_t),
+                                            static_cast<int>(q_heads));
+  at::Tensor partials;
+  float* pptr = nullptr;
+  if (chunks > 1) {
+    partials = at::empty({q_heads, chunks, hd + 2},
+                         q.options().dtype(at::kFloat));
+    pptr = partials.data_ptr<float>();
+  }
+  FMA_HIP_CHECK(fma_launch_attn_decode_bf16(
+      q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+      static_cast<int>(max_t), static_cast<int>(q_heads),
+      static_cast<int>(kv_heads), static_cast<int>(hd),
+      static_cast<long long>(kv_heads * hd), pptr, chunks,
+      t_dev.data_ptr<int>(), stream.stream()));
   return out;
 }
 
@@ -1037,6 +1073,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm1_bf16", &rmsnorm1_bf16, py::arg("x"), py::arg("w"),
         py::arg("eps"));
   m.def("silu_mul_bf16", &silu_mul_bf16, py::arg("g"), py::arg("u"));
+  m.def("attn_decode_bf16_graph", &attn_decode_bf16_graph, py::arg("q"),
+        py::arg("k"), py::arg("v"), py::arg("t_dev"), py::arg("max_t"));
   m.def("attn_prefill_bf16", &attn_prefill_bf16, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("pos0"), py::arg("chunks") = 0);
   m.def("attn_decode_bf16", &attn_decode_bf16, py::arg("q"), py::arg("k"),

@@ -55,7 +55,7 @@ extern "C" int fma_attn_decode_chunks(int t, int q_heads);
 extern "C" hipError_t fma_launch_attn_decode_bf16(
     const void* q, const void* K, const void* V, void* out, int t,
     int q_heads, int kv_heads, int hd, long long k_stride,
-    float* partials, int chunks, hipStream_t stream);
+    float* partials, int chunks, const int* t_dev, hipStream_t stream);
 
 extern "C" int fma_attn_prefill_chunks(int T, int pos0, int q_heads);
 

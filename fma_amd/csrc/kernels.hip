@@ -474,8 +474,13 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
     const unsigned short* __restrict__ V,
     unsigned short* __restrict__ out,       // [qH, hd] (chunks == 1)
     float* __restrict__ partials,           // [qH, chunks, hd + 2]
-    int t, int q_heads, int kv_heads, int hd,
+    const int* __restrict__ t_dev,          // runtime t (hipGraph mode)|null
+    int t_arg, int q_heads, int kv_heads, int hd,
     long long k_stride) {
+  // hipGraph mode: the sequence length is read from device memory so a
+  // captured graph replays correctly as the KV cache grows (t_arg is
+  // then only the upper bound the launch geometry was sized for)
+  const int t = t_dev != nullptr ? *t_dev : t_arg;
   __shared__ float s_scores[FMA_ATTN_MAX_T / 4];
   __shared__ float s_q[256];
   __shared__ float s_red[256];
@@ -655,11 +660,11 @@ extern "C" int fma_attn_decode_chunks(int t, int q_heads) {
 extern "C" hipError_t fma_launch_attn_decode_bf16(
     const void* q, const void* K, const void* V, void* out, int t,
     int q_heads, int kv_heads, int hd, long long k_stride,
-    float* partials, int chunks, hipStream_t stream) {
+    float* partials, int chunks, const int* t_dev, hipStream_t stream) {
   if (hd > 256 || (hd & 63) != 0) return hipErrorInvalidValue;
   if (q_heads % kv_heads != 0) return hipErrorInvalidValue;
   // any t: the chunk heuristic bounds each chunk's scores to the LDS
-  // window (FMA_ATTN_MAX_T/4)
+  // window (FMA_ATTN_MAX_T/4); with t_dev, t is the static upper bound
   if ((t + chunks - 1) / chunks > FMA_ATTN_MAX_T / 4)
     return hipErrorInvalidValue;
   if (chunks > 1 && partials == nullptr) return hipErrorInvalidValue;
@@ -668,8 +673,8 @@ extern "C" hipError_t fma_launch_attn_decode_bf16(
       static_cast<const unsigned short*>(q),
       static_cast<const unsigned short*>(K),
       static_cast<const unsigned short*>(V),
-      static_cast<unsigned short*>(out), partials, t, q_heads, kv_heads, hd,
-      k_stride);
+      static_cast<unsigned short*>(out), partials, t_dev, t, q_heads,
+      kv_heads, hd, k_stride);
   if (chunks > 1) {
     attn_decode_combine_kernel<<<q_heads, 64, 0, stream>>>(
         partials, static_cast<unsigned short*>(out), chunks, hd);

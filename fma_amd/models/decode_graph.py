@@ -24,7 +24,8 @@ from typing import Optional
 import torch
 import torch.nn.functional as F
 
-from fma_amd.ops.decode_ops import (fast_rmsnorm, fast_rope1,
+from fma_amd.ops.decode_ops import (available as _fused_available,
+                                    fast_rmsnorm, fast_rope1,
                                     fast_silu_mul)
 from fma_amd.ops.linear import fast_linear
 
@@ -53,6 +54,9 @@ class StaticDecoder:
                                   device=dev)
         self.out_tokens = torch.zeros((batch, max_seq), dtype=torch.long,
                                       device=dev)
+        # runtime sequence length for the graph-safe attention kernel
+        # (device int32 read by the kernel itself at replay time)
+        self.t_i32 = torch.ones((1,), dtype=torch.int32, device=dev)
         self.graph: Optional[torch.cuda.CUDAGraph] = None
 
     # -- the static step (graph-capturable) --------------------------------
@@ -71,6 +75,16 @@ class StaticDecoder:
         x = F.embedding(self.input_tok, P["embed.weight"])  # [B,1,H]
         cos = m.rope_cos.index_select(0, self.pos.view(1))  # [1, hd/2]
         sin = m.rope_sin.index_select(0, self.pos.view(1))
+        # graph-safe fused attention: the kernel reads t from this device
+        # scalar, so the captured launch (sized for max_seq) replays
+        # correctly at every position — no full-window SDPA, no
+        # repeat_interleave materialization (those cost a 15 GiB model
+        # ~0.6 ms/token: 192 -> 173 tok/s measured)
+        fused_attn = (B == 1 and x.is_cuda and x.dtype == torch.bfloat16
+                      and hd % 64 == 0 and hd <= 256
+                      and q_heads % kv_heads == 0 and _fused_available())
+        if fused_attn:
+            self.t_i32.copy_((self.pos + 1).to(torch.int32))
         # additive mask over the full window: position j attends iff j <= pos
         neg = -1e9  # large finite: stays finite in bf16
         mask = torch.where(self.positions <= self.pos,
@@ -93,15 +107,23 @@ class StaticDecoder:
             # static cache write at pos
             self.cache[li, 0].index_copy_(1, self.pos.view(1), k)
             self.cache[li, 1].index_copy_(1, self.pos.view(1), v)
-            kh = self.cache[li, 0].transpose(1, 2)  # [B, kvH, S, hd]
-            vh = self.cache[li, 1].transpose(1, 2)
-            if kv_heads != q_heads:
-                rep = q_heads // kv_heads
-                kh = kh.repeat_interleave(rep, dim=1)
-                vh = vh.repeat_interleave(rep, dim=1)
-            att = F.scaled_dot_product_attention(
-                q.transpose(1, 2), kh, vh, attn_mask=mask.to(q.dtype))
-            att = att.transpose(1, 2).reshape(B, 1, q_heads * hd)
+            if fused_attn:
+                from fma_amd.ops import actuation
+                att = actuation._C.attn_decode_bf16_graph(
+                    q.reshape(q_heads, hd).contiguous(),
+                    self.cache[li, 0, 0], self.cache[li, 1, 0],
+                    self.t_i32, self.max_seq)
+                att = att.view(B, 1, q_heads * hd)
+            else:
+                kh = self.cache[li, 0].transpose(1, 2)  # [B, kvH, S, hd]
+                vh = self.cache[li, 1].transpose(1, 2)
+                if kv_heads != q_heads:
+                    rep = q_heads // kv_heads
+                    kh = kh.repeat_interleave(rep, dim=1)
+                    vh = vh.repeat_interleave(rep, dim=1)
+                att = F.scaled_dot_product_attention(
+                    q.transpose(1, 2), kh, vh, attn_mask=mask.to(q.dtype))
+                att = att.transpose(1, 2).reshape(B, 1, q_heads * hd)
             x = x + fast_linear(att, P[p + "wo.weight"])
             h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
             gate = fast_linear(h, P[p + "w_gate.weight"])

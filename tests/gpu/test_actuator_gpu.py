@@ -478,3 +478,21 @@ def test_prefill_mfma_vs_sdpa_path_logits_close():
         (mfma - sdpa).abs().max().item()
     cache.free()
     cache2.free()
+
+
+def test_attn_decode_graph_variant_matches_static_t():
+    """The device-t (hipGraph-safe) decode attention equals the static-t
+    kernel at every position, launched with one max_t geometry."""
+    import fma_amd._C as C
+    torch.manual_seed(23)
+    qH, kvH, hd, S = 32, 8, 128, 512
+    q = torch.randn(qH, hd, dtype=torch.bfloat16, device="cuda:0")
+    k = torch.randn(S, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
+    v = torch.randn(S, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
+    t_dev = torch.zeros(1, dtype=torch.int32, device="cuda:0")
+    for t in (1, 7, 100, 511, 512):
+        t_dev.fill_(t)
+        got = C.attn_decode_bf16_graph(q, k, v, t_dev, S)
+        want = C.attn_decode_bf16(q, k, v, t)
+        assert torch.allclose(got.float(), want.float(),
+                              atol=2e-2, rtol=2e-2), t
