@@ -496,3 +496,32 @@ def test_attn_decode_graph_variant_matches_static_t():
         want = C.attn_decode_bf16(q, k, v, t)
         assert torch.allclose(got.float(), want.float(),
                               atol=2e-2, rtol=2e-2), t
+
+
+def test_engine_graph_decode_serving_path():
+    """engine.generate uses the captured hipGraph decoder (batch-1),
+    tokens match eager exactly, and sleep drops / wake rebuilds it
+    (captured pointers die with the arena slabs)."""
+    import os
+    from fma_amd.models.llama import LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+    cfg = LlamaConfig(name="gsrv", vocab_size=512, hidden_size=512,
+                      intermediate_size=768, num_layers=2, num_heads=4,
+                      num_kv_heads=2, max_seq_len=128)
+    eng = ActuationEngine(cfg, 0, seed=41)
+    toks = torch.randint(0, cfg.vocab_size, (1, 10), device="cuda:0")
+    graphed = eng.generate(toks, 6)
+    assert eng.stats()["graph_decode"] is True
+    os.environ["FMA_GRAPH_DECODE"] = "0"
+    try:
+        eager = eng.model.generate(toks, 6)
+    finally:
+        del os.environ["FMA_GRAPH_DECODE"]
+    assert torch.equal(graphed, eager)
+
+    eng.sleep()
+    assert eng.stats()["graph_decode"] is False, "decoder survived sleep"
+    eng.wake_up()
+    again = eng.generate(toks, 6)
+    assert torch.equal(again, eager), "post-wake graph decode diverged"
+    assert eng.stats()["graph_decode"] is True
