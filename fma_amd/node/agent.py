@@ -44,6 +44,11 @@ class PodProcess:
         self.ip = ip
         self.proc = proc
         self.probe = probe
+        self.restarts = 0
+        self.next_restart = 0.0
+        # k8s default is Always; "Never" lets a crash surface as Failed
+        self.restart_policy = (pod.get("spec") or {}).get(
+            "restartPolicy", "Always")
         self.notifier: Optional[PodNotifier] = None
         self.notifier_thread: Optional[threading.Thread] = None
 
@@ -151,6 +156,14 @@ class NodeAgent:
         full_env["POD_IP"] = ip
         full_env["NODE_NAME"] = self.node
         full_env["POD_NAME"] = ob.name_of(pod)
+        # unique per-pod marker inherited by EVERY descendant: the kill
+        # path reaps by scanning /proc for it, the way a kubelet's cgroup
+        # kill takes the whole tree. Without this, a crashed launcher
+        # orphans its instance processes (they run in their own process
+        # groups), which keep the model ports — and on real GPUs the
+        # HBM — alive forever.
+        full_env["FMA_POD_TREE"] = \
+            f"{self.node}/{ob.name_of(pod)}/{ob.uid_of(pod)}"
         log_path = os.path.join(self.log_dir,
                                 f"pod-{ob.name_of(pod)}.log")
         logf = open(log_path, "ab")
@@ -158,6 +171,10 @@ class NodeAgent:
                                 stderr=subprocess.STDOUT,
                                 start_new_session=True)
         pp = PodProcess(pod, ip, proc, probe)
+        # kept for restartPolicy:Always respawns (kubelet semantics)
+        pp.cmd = list(cmd)
+        pp.full_env = full_env
+        pp.log_path = log_path
         self.pods[ob.name_of(pod)] = pp
         self._patch_status(pp, phase="Running", ready=False, ip=ip)
         if ob.labels_of(pod).get(contracts.COMPONENT_LABEL) == \
@@ -199,6 +216,44 @@ class NodeAgent:
                     os.killpg(pp.proc.pid, signal.SIGKILL)
                 except (ProcessLookupError, PermissionError):
                     pass
+        self._reap_pod_tree(pp.pod_name, pp.uid)
+
+    def _reap_pod_tree(self, pod_name: str, uid: str) -> int:
+        """Kill every process carrying this pod's FMA_POD_TREE marker —
+        the cgroup-kill analog. Catches instance processes that outlived
+        a crashed launcher (they setpgrp into their own groups, so the
+        launcher's killpg never reaches them). Exact-PID kills only, by
+        environment marker, never by name pattern."""
+        marker = f"{self.node}/{pod_name}/{uid}"
+        needle = ("FMA_POD_TREE=" + marker).encode()
+        victims = []
+        me = os.getpid()
+        for ent in os.listdir("/proc"):
+            if not ent.isdigit() or int(ent) == me:
+                continue
+            try:
+                with open(f"/proc/{ent}/environ", "rb") as f:
+                    if needle in f.read().split(b"\0"):
+                        victims.append(int(ent))
+            except OSError:
+                continue
+        for pid in victims:
+            try:
+                os.kill(pid, signal.SIGTERM)
+            except (ProcessLookupError, PermissionError):
+                pass
+        if victims:
+            deadline = time.time() + 5
+            while time.time() < deadline:
+                if not any(os.path.exists(f"/proc/{p}") for p in victims):
+                    break
+                time.sleep(0.1)
+            for pid in victims:
+                try:
+                    os.kill(pid, signal.SIGKILL)
+                except (ProcessLookupError, PermissionError):
+                    pass
+        return len(victims)
 
     # -- status --------------------------------------------------------------
 
@@ -220,10 +275,45 @@ class NodeAgent:
         while not self._stop.wait(0.15):
             for name, pp in list(self.pods.items()):
                 exited = pp.proc.poll() is not None
+                restartable = pp.restart_policy != "Never"
+                if exited and pp.proc.returncode and restartable:
+                    # restartPolicy Always (the k8s default the reference
+                    # relies on: a crashed launcher container restarts in
+                    # place and the dual-pods controller re-creates its
+                    # instance). Reap the pod's whole process tree first
+                    # so orphaned instances release their ports/HBM, then
+                    # respawn with CrashLoop-style backoff. The Pod stays
+                    # Running/NotReady through the gap, like kubelet.
+                    now = time.time()
+                    if pp.next_restart == 0.0:
+                        pp.next_restart = now + min(
+                            5.0, 0.5 * (2 ** pp.restarts))
+                        self._patch_status(pp, phase="Running", ready=False,
+                                           ip=pp.ip)
+                        continue
+                    if now < pp.next_restart:
+                        continue
+                    self._reap_pod_tree(pp.pod_name, pp.uid)
+                    self._respawn(pp)
+                    continue
                 ready = False if exited else self._probe_ready(pp)
                 phase = "Failed" if exited and pp.proc.returncode else \
                     ("Succeeded" if exited else "Running")
                 self._patch_status(pp, phase=phase, ready=ready, ip=pp.ip)
+
+    def _respawn(self, pp: PodProcess) -> None:
+        logf = open(pp.log_path, "ab")
+        logf.write(f"[agent] restart #{pp.restarts + 1} of "
+                   f"{pp.pod_name}\n".encode())
+        try:
+            pp.proc = subprocess.Popen(pp.cmd, env=pp.full_env, stdout=logf,
+                                       stderr=subprocess.STDOUT,
+                                       start_new_session=True)
+        except OSError:
+            return
+        pp.restarts += 1
+        pp.next_restart = 0.0
+        self._patch_status(pp, phase="Running", ready=False, ip=pp.ip)
 
     def _patch_status(self, pp: PodProcess, phase: str, ready: bool,
                       ip: str) -> None:

@@ -324,3 +324,53 @@ def test_four_concurrent_actuations_on_one_node(cluster):
              C.LAUNCHER_COMPONENT
              and ob.annotations_of(p).get(C.REQUESTER_ANNOTATION)]
     assert len(bound) == 4 and len(set(bound)) == 4
+
+
+def test_launcher_crash_recovery(cluster):
+    """Chaos: SIGKILL the real launcher process mid-service. The node
+    agent reaps the pod's whole process tree (orphaned instances would
+    otherwise squat on the server port) and restarts the container in
+    place — restartPolicy Always, the k8s semantics the reference relies
+    on (utils/pod-helper.go:44 treats restarts+unready as 'in trouble').
+    The dual-pods controller then re-creates the missing instance on the
+    restarted launcher and the SAME requester becomes ready again."""
+    store, agent = cluster["store"], cluster["agent"]
+    mk_isc_lc_lpp(store)
+    lp0 = wait_for(lambda: launcher_pod(store), 30, desc="launcher pod")
+    wait_for(lambda: ob.pod_is_ready(store.get("Pod", ob.name_of(lp0))),
+             60, desc="launcher Ready")
+    mk_requester(store, "cr1")
+    wait_for(lambda: requester_ready(store, agent, "cr1"), 90,
+             desc="first actuation ready")
+    lname = ob.name_of(launcher_pod(store))
+    uid0 = ob.uid_of(store.get("Pod", lname))
+
+    agent.pods[lname].proc.kill()  # chaos
+
+    # the launcher container restarts in place (same Pod object)...
+    wait_for(lambda: agent.pods[lname].restarts >= 1, 30,
+             desc="launcher restarted by the agent")
+    # ...and the controller re-creates the instance on it: the SAME
+    # requester's model answers completions again. (The requester stub
+    # keeps its last readiness through the gap — the relay is one-way,
+    # like the reference's — so the proof of recovery is the server
+    # actually serving, not the stale stub bit.)
+    lp = store.get("Pod", lname)
+    assert ob.uid_of(lp) == uid0, "pod was replaced, not restarted"
+    lp_ip = lp["status"]["podIP"]
+
+    def completion_ok():
+        try:
+            r = httpx.post(f"http://{lp_ip}:{ISC_PORT}/v1/completions",
+                           json={"prompt": "hi", "max_tokens": 2},
+                           timeout=5)
+            return r.status_code == 200
+        except httpx.HTTPError:
+            return False
+
+    wait_for(completion_ok, 120,
+             desc="completions served again after crash recovery")
+    lp = store.get("Pod", lname)
+    assert ob.annotations_of(lp)[C.REQUESTER_ANNOTATION].endswith(" cr1")
+    assert store.try_get("Pod", "cr1") is not None
+    assert requester_ready(store, agent, "cr1")

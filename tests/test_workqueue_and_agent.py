@@ -147,10 +147,11 @@ def test_agent_reports_failed_phase(agent):
     store, a = agent
     pod = ob.new_object(
         "Pod", "crash",
-        spec={"nodeName": "n1", "containers": [{
-            "name": "main",
-            "command": [sys.executable, "-c", "import sys; sys.exit(3)"],
-        }]})
+        spec={"nodeName": "n1", "restartPolicy": "Never",
+              "containers": [{
+                  "name": "main",
+                  "command": [sys.executable, "-c", "import sys; sys.exit(3)"],
+              }]})
     store.create(pod)
     assert wait(lambda: store.get("Pod", "crash")["status"].get("phase")
                 == "Failed")
