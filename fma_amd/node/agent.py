@@ -321,13 +321,20 @@ class NodeAgent:
         if cur is None or ob.uid_of(cur) != pp.uid:
             return
         status = cur.setdefault("status", {})
+        old_restarts = (status.get("containerStatuses") or
+                        [{}])[0].get("restartCount", 0)
         changed = (status.get("phase") != phase or
                    status.get("podIP") != ip or
+                   old_restarts != pp.restarts or
                    ob.pod_is_ready(cur) != ready)
         if not changed:
             return
         status["phase"] = phase
         status["podIP"] = ip
+        # kubelet-style restart accounting (PodIsInTrouble in the
+        # reference, utils/pod-helper.go:44, keys off this)
+        status["containerStatuses"] = [{"restartCount": pp.restarts,
+                                        "ready": ready}]
         ob.set_pod_ready(cur, ready)
         status.setdefault("startTime", ob.meta(cur).get("creationTimestamp"))
         try:

@@ -374,3 +374,6 @@ def test_launcher_crash_recovery(cluster):
     assert ob.annotations_of(lp)[C.REQUESTER_ANNOTATION].endswith(" cr1")
     assert store.try_get("Pod", "cr1") is not None
     assert requester_ready(store, agent, "cr1")
+    # kubelet-style restart accounting surfaced on the Pod
+    lp = store.get("Pod", lname)
+    assert lp["status"]["containerStatuses"][0]["restartCount"] >= 1
