@@ -15,7 +15,6 @@ import argparse
 import json
 import os
 import random
-import threading
 from typing import Dict, List
 
 from fma_amd.api import contracts

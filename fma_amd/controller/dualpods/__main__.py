@@ -1,6 +1,5 @@
 """dual-pods-controller entry point (reference cmd/dual-pods-controller)."""
 import argparse
-import os
 import time
 
 from fma_amd.controller.dualpods.controller import (ControllerConfig,

@@ -17,7 +17,7 @@ from __future__ import annotations
 import json as jsonlib
 import logging
 import time
-from typing import Any, Callable, Dict, Optional, Tuple
+from typing import Any, Callable, Dict, Optional
 
 logger = logging.getLogger("fma.http")
 

@@ -20,7 +20,6 @@ from __future__ import annotations
 
 import argparse
 import threading
-import time
 
 from fma_amd.controller import metrics
 from fma_amd.controller.dualpods.controller import (ControllerConfig,

@@ -20,7 +20,7 @@ Inference only: plain tensors, no autograd.
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Tuple
 
 import torch
