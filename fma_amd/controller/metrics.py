@@ -8,7 +8,6 @@ pkg/controller/launcher-populator/metrics.go:36-310).
 
 from __future__ import annotations
 
-from typing import Optional
 
 try:
     from prometheus_client import Gauge, Histogram, start_http_server
