@@ -155,3 +155,39 @@ def test_agent_reports_failed_phase(agent):
     store.create(pod)
     assert wait(lambda: store.get("Pod", "crash")["status"].get("phase")
                 == "Failed")
+
+
+def test_reap_pod_tree_kills_marked_orphans(tmp_path):
+    """_reap_pod_tree must kill processes carrying the pod's
+    FMA_POD_TREE marker even when they detached into their own process
+    group and their parent is gone (the crashed-launcher orphan case:
+    an orphan holding the server port blocks every later instance)."""
+    import os
+    import subprocess
+    import time as _time
+
+    from fma_amd.node.agent import NodeAgent
+    from fma_amd.store.memstore import MemStore
+
+    agent = NodeAgent(MemStore(), "nX", node_index=9,
+                      log_dir=str(tmp_path))
+    marker_env = dict(os.environ,
+                      FMA_POD_TREE="nX/podX/uid-123")
+    orphan = subprocess.Popen(
+        [os.sys.executable, "-c", "import time; time.sleep(300)"],
+        env=marker_env, start_new_session=True)
+    bystander = subprocess.Popen(
+        [os.sys.executable, "-c", "import time; time.sleep(300)"],
+        start_new_session=True)
+    try:
+        n = agent._reap_pod_tree("podX", "uid-123")
+        assert n == 1
+        deadline = _time.time() + 10
+        while _time.time() < deadline and orphan.poll() is None:
+            _time.sleep(0.1)
+        assert orphan.poll() is not None, "marked orphan survived"
+        assert bystander.poll() is None, "unmarked process was killed!"
+    finally:
+        for p in (orphan, bystander):
+            if p.poll() is None:
+                p.kill()
