@@ -191,3 +191,41 @@ def test_reap_pod_tree_kills_marked_orphans(tmp_path):
         for p in (orphan, bystander):
             if p.poll() is None:
                 p.kill()
+
+
+def test_crashloop_restarts_with_backoff(tmp_path):
+    """A crash-looping pod (restartPolicy Always) is restarted with
+    growing backoff and its restartCount surfaces in containerStatuses,
+    kubelet-style (the reference's PodIsInTrouble,
+    utils/pod-helper.go:44, keys off restarts + unready)."""
+    import sys
+    import time as _time
+
+    from fma_amd.api import contracts as C  # noqa: F401
+    from fma_amd.node.agent import NodeAgent
+    from fma_amd.store import objects as ob
+    from fma_amd.store.memstore import MemStore
+
+    store = MemStore()
+    agent = NodeAgent(store, "n1", node_index=11, log_dir=str(tmp_path))
+    agent.start()
+    try:
+        pod = ob.new_object(
+            "Pod", "looper",
+            spec={"nodeName": "n1", "containers": [{
+                "name": "main",
+                "command": [sys.executable, "-c", "import sys; sys.exit(5)"],
+            }]})
+        store.create(pod)
+        deadline = _time.time() + 30
+        while _time.time() < deadline:
+            pp = agent.pods.get("looper")
+            if pp is not None and pp.restarts >= 2:
+                break
+            _time.sleep(0.1)
+        assert pp is not None and pp.restarts >= 2, "no restarts happened"
+        cur = store.get("Pod", "looper")
+        assert cur["status"]["phase"] == "Running"  # not Failed: it loops
+        assert cur["status"]["containerStatuses"][0]["restartCount"] >= 1
+    finally:
+        agent.stop()
