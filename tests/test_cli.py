@@ -35,3 +35,26 @@ def test_apply_get_delete(tmp_path, capsys):
     assert "my-model-request" in capsys.readouterr().out
     client.delete("Pod", "my-model-request")
     assert store.try_get("Pod", "my-model-request") is None
+
+
+def test_kubernetes_manifests_parse_and_cover_rbac():
+    """manifests/kubernetes/controllers.yaml (the Helm-chart analog,
+    reference charts/fma-controllers) must stay parseable and keep the
+    RBAC surface the controllers need."""
+    import os
+
+    import yaml
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    docs = [d for d in yaml.safe_load_all(
+        open(os.path.join(root, "manifests/kubernetes/controllers.yaml")))
+        if d]
+    kinds = sorted(d["kind"] for d in docs)
+    assert kinds == ["ClusterRole", "ClusterRoleBinding", "Deployment",
+                     "Deployment", "Role", "RoleBinding", "ServiceAccount"]
+    role = next(d for d in docs if d["kind"] == "Role")
+    pod_rule = next(r for r in role["rules"]
+                    if "pods" in r.get("resources", []))
+    assert set(pod_rule["verbs"]) >= {"create", "delete", "patch", "watch"}
+    crd_rule = next(r for r in role["rules"]
+                    if "inferenceserverconfigs" in r.get("resources", []))
+    assert "watch" in crd_rule["verbs"]
