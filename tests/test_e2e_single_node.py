@@ -377,3 +377,46 @@ def test_launcher_crash_recovery(cluster):
     # kubelet-style restart accounting surfaced on the Pod
     lp = store.get("Pod", lname)
     assert lp["status"]["containerStatuses"][0]["restartCount"] >= 1
+
+
+def test_direct_provider_crash_recovery(cluster):
+    """Chaos on the direct path: SIGKILL the serving-runtime process.
+    The agent restarts the container in place; the restarted server
+    boots fresh and the controller's bound reconcile re-drives it to
+    serving (is_sleeping False) for the same requester."""
+    store, agent = cluster["store"], cluster["agent"]
+    cm = ob.new_object("ConfigMap", C.GPU_MAP_CONFIGMAP)
+    cm["data"] = {"node-a": '{"GPU-0": 0}'}
+    store.create(cm)
+    patch = DIRECT_PATCH.format(python=sys.executable)
+    pod = ob.new_object(
+        "Pod", "xreq1",
+        annotations={C.SERVER_PATCH_ANNOTATION: patch},
+        spec={"nodeName": "node-a", "containers": [
+            {"name": "requester",
+             "command": [sys.executable, "-m", "fma_amd.requester.server"]},
+            {"name": "inference-server"}]})
+    store.create(pod, actor="user")
+    wait_for(lambda: requester_ready(store, agent, "xreq1"), 90,
+             desc="direct requester ready")
+    provider = store.get("Pod", "xreq1-server")
+    prov_ip = provider["status"]["podIP"]
+    uid0 = ob.uid_of(provider)
+
+    agent.pods["xreq1-server"].proc.kill()  # chaos
+
+    wait_for(lambda: agent.pods["xreq1-server"].restarts >= 1, 30,
+             desc="provider restarted by the agent")
+
+    def serving_again():
+        try:
+            r = httpx.get(f"http://{prov_ip}:8361/is_sleeping", timeout=5)
+            return r.status_code == 200 and r.json()["is_sleeping"] is False
+        except httpx.HTTPError:
+            return False
+
+    wait_for(serving_again, 120, desc="direct provider serving again")
+    provider = store.get("Pod", "xreq1-server")
+    assert ob.uid_of(provider) == uid0, "provider replaced, not restarted"
+    assert ob.annotations_of(provider)[C.REQUESTER_ANNOTATION].endswith(
+        " xreq1")
