@@ -93,7 +93,8 @@ class DualPodsController:
         self._lock = threading.Lock()
         self._stop = threading.Event()
         self.workers = QueueAndWorkers("dualpods", self.cfg.num_workers,
-                                       self._process, max_backoff=2.0)
+                                       self._process, max_backoff=2.0,
+                                       metrics_name="dualpods")
         self._watch_thread: Optional[threading.Thread] = None
 
     # ------------------------------------------------------------------

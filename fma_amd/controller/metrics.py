@@ -10,7 +10,8 @@ from __future__ import annotations
 
 
 try:
-    from prometheus_client import Gauge, Histogram, start_http_server
+    from prometheus_client import (Counter, Gauge, Histogram,
+                                   start_http_server)
     HAVE_PROM = True
 except Exception:  # pragma: no cover
     HAVE_PROM = False
@@ -104,6 +105,39 @@ def serve_metrics(port: int = 8002) -> None:
     serves :8002; pprof has no Python analog — py-spy attaches externally)."""
     if HAVE_PROM:
         start_http_server(port)
+
+
+def queue_adds_total():
+    return _get_or_make(
+        Counter, "fma_dpc_innerqueue_adds_total",
+        "items added to the dual-pods inner queue (reference: k8s "
+        "workqueue metrics, docs/metrics.md)", ("name",))
+
+
+def queue_depth():
+    return _get_or_make(
+        Gauge, "fma_dpc_innerqueue_depth",
+        "current depth of the dual-pods inner queue", ("name",))
+
+
+def queue_retries_total():
+    return _get_or_make(
+        Counter, "fma_dpc_innerqueue_retries_total",
+        "rate-limited retries on the dual-pods inner queue", ("name",))
+
+
+def queue_queue_duration_seconds():
+    return _get_or_make(
+        Histogram, "fma_dpc_innerqueue_queue_duration_seconds",
+        "time items wait in the inner queue before processing", ("name",),
+        buckets=(.001, .01, .1, 1, 10, 60))
+
+
+def queue_work_duration_seconds():
+    return _get_or_make(
+        Histogram, "fma_dpc_innerqueue_work_duration_seconds",
+        "time spent processing inner-queue items", ("name",),
+        buckets=(.001, .01, .1, 1, 10, 60))
 
 
 def serve_debug(port: int = 8003):

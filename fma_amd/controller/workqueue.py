@@ -27,7 +27,8 @@ class RateLimitingQueue(Generic[T]):
     """Subset of client-go's workqueue: Add / AddAfter / AddRateLimited /
     Forget / Get / Done / ShutDown with dirty/processing sets."""
 
-    def __init__(self, max_backoff: float = MAX_BACKOFF_SECONDS) -> None:
+    def __init__(self, max_backoff: float = MAX_BACKOFF_SECONDS,
+                 metrics_name: str = "") -> None:
         self.max_backoff = max_backoff
         self._cond = threading.Condition()
         self._queue: List[T] = []
@@ -36,15 +37,27 @@ class RateLimitingQueue(Generic[T]):
         self._failures: dict = {}
         self._shutdown = False
         self._timers: List[threading.Timer] = []
+        # the reference exports k8s workqueue metrics for its inner
+        # queues (docs/metrics.md fma_dpc_innerqueue_*); emit the same
+        # family when a name is given
+        self._mname = metrics_name
+        self._enqueued_at: dict = {}
 
     def add(self, item: T) -> None:
         with self._cond:
             if self._shutdown or item in self._dirty:
                 return
             self._dirty.add(item)
+            if self._mname:
+                from fma_amd.controller import metrics as _m
+                _m.queue_adds_total().labels(self._mname).inc()
+                self._enqueued_at.setdefault(item, time.monotonic())
             if item in self._processing:
                 return
             self._queue.append(item)
+            if self._mname:
+                from fma_amd.controller import metrics as _m
+                _m.queue_depth().labels(self._mname).set(len(self._queue))
             self._cond.notify()
 
     def add_after(self, item: T, delay: float) -> None:
@@ -63,6 +76,9 @@ class RateLimitingQueue(Generic[T]):
         with self._cond:
             n = self._failures.get(item, 0)
             self._failures[item] = n + 1
+        if self._mname:
+            from fma_amd.controller import metrics as _m
+            _m.queue_retries_total().labels(self._mname).inc()
         self.add_after(item, min(BASE_BACKOFF_SECONDS * (2 ** n),
                                  self.max_backoff))
 
@@ -79,6 +95,13 @@ class RateLimitingQueue(Generic[T]):
             item = self._queue.pop(0)
             self._dirty.discard(item)
             self._processing.add(item)
+            if self._mname:
+                from fma_amd.controller import metrics as _m
+                _m.queue_depth().labels(self._mname).set(len(self._queue))
+                t0 = self._enqueued_at.pop(item, None)
+                if t0 is not None:
+                    _m.queue_queue_duration_seconds().labels(
+                        self._mname).observe(time.monotonic() - t0)
             return item
 
     def done(self, item: T) -> None:
@@ -110,9 +133,12 @@ class QueueAndWorkers(Generic[T]):
 
     def __init__(self, name: str, num_workers: int,
                  process: Callable[[T], bool],
-                 max_backoff: float = MAX_BACKOFF_SECONDS):
+                 max_backoff: float = MAX_BACKOFF_SECONDS,
+                 metrics_name: str = ""):
         self.name = name
-        self.queue: RateLimitingQueue[T] = RateLimitingQueue(max_backoff)
+        self.queue: RateLimitingQueue[T] = RateLimitingQueue(
+            max_backoff, metrics_name=metrics_name)
+        self._mname = metrics_name
         self.num_workers = num_workers
         self.process = process
         self.threads: List[threading.Thread] = []
@@ -129,12 +155,17 @@ class QueueAndWorkers(Generic[T]):
             item = self.queue.get()
             if item is None:
                 return
+            t0 = time.monotonic()
             try:
                 retry = self.process(item)
             except Exception:  # noqa: BLE001 - reconcile must not kill worker
                 import traceback
                 traceback.print_exc()
                 retry = True
+            if self._mname:
+                from fma_amd.controller import metrics as _m
+                _m.queue_work_duration_seconds().labels(
+                    self._mname).observe(time.monotonic() - t0)
             if isinstance(retry, (int, float)) and not isinstance(retry, bool) \
                     and retry > 0:
                 self.queue.forget(item)  # a scheduled wait is not a failure

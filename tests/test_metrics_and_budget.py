@@ -141,3 +141,42 @@ def test_debug_endpoint_serves_thread_stacks():
             assert e.code == 404
     finally:
         srv.shutdown()
+
+
+def test_innerqueue_metrics_emitted():
+    """The reference exports k8s workqueue metrics for its inner queue
+    (docs/metrics.md fma_dpc_innerqueue_*); our RateLimitingQueue emits
+    the same family when named."""
+    import time
+
+    from fma_amd.controller import metrics as M
+    from fma_amd.controller.workqueue import QueueAndWorkers
+
+    if not M.HAVE_PROM:
+        import pytest
+        pytest.skip("prometheus_client unavailable")
+    processed = []
+
+    def proc(item):
+        processed.append(item)
+        return item == "retry-once" and len(processed) < 3
+
+    q = QueueAndWorkers("t", 1, proc, metrics_name="dpc-test")
+    q.start()
+    q.queue.add("a")
+    q.queue.add("retry-once")
+    deadline = time.time() + 10
+    while time.time() < deadline and len(processed) < 3:
+        time.sleep(0.05)
+    q.stop()
+    from prometheus_client import REGISTRY
+    adds = REGISTRY.get_sample_value(
+        "fma_dpc_innerqueue_adds_total", {"name": "dpc-test"})
+    retries = REGISTRY.get_sample_value(
+        "fma_dpc_innerqueue_retries_total", {"name": "dpc-test"})
+    work = REGISTRY.get_sample_value(
+        "fma_dpc_innerqueue_work_duration_seconds_count",
+        {"name": "dpc-test"})
+    assert adds and adds >= 2
+    assert retries and retries >= 1
+    assert work and work >= 2
