@@ -96,6 +96,36 @@ def main() -> None:
         return getattr(eng.arena, "last_alloc_wait_seconds", 0.0) \
             if eng.arena is not None else 0.0
 
+    # Box health probe: raw pinned H2D bandwidth + hipMalloc commit rate.
+    # Wake is link-bound (~56 GB/s on a healthy PCIe Gen5 x16 host); some
+    # pool hosts have degraded H2D or slow page commit (seen: 20 GB/s ->
+    # 3.1 s instead of 1.22 s for 64 GiB). Reporting both makes a slow
+    # box self-evident in the result rather than looking like a
+    # regression.
+    h2d_gbps = alloc_gbps = None
+    if on_gpu and rank == 0:
+        probe = min(1 << 30, eng.total_bytes or (1 << 30))
+        host_buf = torch.empty(probe, dtype=torch.uint8, pin_memory=True)
+        dev_buf = torch.empty(probe, dtype=torch.uint8, device="cuda")
+        torch.cuda.synchronize()
+        t0p = time.perf_counter()
+        dev_buf.copy_(host_buf, non_blocking=True)
+        torch.cuda.synchronize()
+        h2d_gbps = probe / (time.perf_counter() - t0p) / 1e9
+        del dev_buf
+        torch.cuda.empty_cache()
+        t0p = time.perf_counter()
+        dev_buf = torch.empty(probe, dtype=torch.uint8, device="cuda")
+        dev_buf.fill_(1)  # force page commit
+        torch.cuda.synchronize()
+        alloc_gbps = probe / (time.perf_counter() - t0p) / 1e9
+        del dev_buf, host_buf
+        torch.cuda.empty_cache()
+        log(f"[rank 0] box probe: pinned H2D {h2d_gbps:.1f} GB/s, "
+            f"alloc+commit {alloc_gbps:.1f} GB/s"
+            + ("  << DEGRADED HOST (healthy: ~56 GB/s H2D)"
+               if h2d_gbps < 40 else ""))
+
     # warmup
     for i in range(args.warmup):
         ts = eng.sleep()
@@ -155,6 +185,8 @@ def main() -> None:
                 "mode": args.mode,
                 "vmm": bool(eng.stats()["uses_vmm"]),
                 "mean_sleep_s": round(mean_sleep, 4),
+                "h2d_gbps": round(h2d_gbps, 1) if h2d_gbps else None,
+                "alloc_gbps": round(alloc_gbps, 1) if alloc_gbps else None,
                 "global_batch": None,
                 "seq_len": None,
             },
