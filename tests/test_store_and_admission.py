@@ -182,3 +182,50 @@ def test_store_concurrent_crud_stress():
     # watch history is strictly increasing
     revs = [e.revision for e in s.watch(since=0, timeout=0.05)]
     assert revs == sorted(revs) and len(revs) == len(set(revs))
+
+
+# -- property fuzz: the CEL-policy analogs hold under arbitrary attack ------
+
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+from fma_amd.store.admission import (FMA_ACTORS, PROTECTED_ANNOTATIONS,  # noqa: E402
+                                     PROTECTED_LABELS)
+
+_texts = st.text(
+    alphabet=st.characters(whitelist_categories=("Ll", "Lu", "Nd")),
+    min_size=1, max_size=12)
+
+
+@settings(max_examples=150, deadline=None)
+@given(key=st.sampled_from(PROTECTED_ANNOTATIONS + PROTECTED_LABELS),
+       value=_texts,
+       actor=st.one_of(_texts, st.sampled_from(sorted(FMA_ACTORS))))
+def test_protected_fields_fuzz(key, value, actor):
+    """Reference CEL policy (fma-immutable-fields.yaml): any change to a
+    protected annotation/label by a non-FMA actor is rejected; FMA
+    service accounts always pass (property-tested over random keys,
+    values and actor names)."""
+    import copy
+
+    from fma_amd.store import objects as ob
+    from fma_amd.store.admission import install_policies
+    from fma_amd.store.memstore import Invalid, MemStore
+
+    store = MemStore()
+    install_policies(store)
+    pod = store.create(ob.new_object("Pod", "t1", spec={}),
+                       actor="dual-pods-controller")
+    mutated = copy.deepcopy(pod)
+    if key in PROTECTED_LABELS:
+        ob.labels_of(mutated)[key] = value
+    else:
+        ob.annotations_of(mutated)[key] = value
+    if actor in FMA_ACTORS:
+        store.update(mutated, actor=actor)  # must not raise
+    else:
+        try:
+            store.update(mutated, actor=actor)
+            raise AssertionError(
+                f"{actor!r} mutated protected {key!r} unchallenged")
+        except Invalid:
+            pass
