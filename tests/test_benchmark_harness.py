@@ -73,3 +73,25 @@ def test_diagnosis_collects_failure_state(tmp_path):
 def test_query_gpu_usage_no_gpu_is_empty_or_dict():
     from fma_amd.benchmark.harness import query_gpu_usage
     assert isinstance(query_gpu_usage(), dict)
+
+
+def test_autoscale_demo_runs(tmp_path):
+    """tools/demo_autoscale.py (the reference's demo-fma-hpa analog)
+    completes a 1->2->1 scale cycle through the real single-node stack
+    and reports per-replica actuation latencies."""
+    import json
+    import os
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    res = subprocess.run(
+        [sys.executable, "tools/demo_autoscale.py", "--curve", "1,2,1",
+         "--log-dir", str(tmp_path)],
+        capture_output=True, text=True, timeout=240, cwd=root,
+        env=dict(os.environ, FMA_FAKE_GPU="1"))
+    assert res.returncode == 0, res.stderr[-2000:]
+    out = json.loads(res.stdout)
+    assert out["scale_ups"] == 2
+    assert out["first_s"] > 0
+    downs = [e for e in out["events"] if e["event"] == "scale-down"]
+    assert len(downs) == 1  # tick 2 scales 2 -> 1 (teardown is unrecorded)
