@@ -8,7 +8,10 @@ load curve, every scale-up is a real actuation (launcher bind + wake or
 cold start), every scale-down is a real unbind+sleep.
 
 Usage:  python tools/demo_autoscale.py [--curve 1,3,4,2,1] [--max 4]
-CPU-only friendly (FMA_FAKE_GPU); on a GPU box the actuations move HBM.
+Runs with the fake arena by default (FMA_FAKE_GPU=1, works anywhere);
+real-GPU actuation of the same control flow is covered by the e2e suite
+(tests/test_e2e_single_node.py on an MI355X box). On a single-GPU box
+keep --max 1 — the naive GPU translator assigns one device per replica.
 """
 
 import argparse
