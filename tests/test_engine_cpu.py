@@ -161,3 +161,50 @@ def test_pack_mode_engine_bit_exact():
     for n, p in eng.params.items():
         assert torch.equal(p, snap[n]), n
     assert torch.equal(before, eng.model.forward(toks))
+
+
+# -- property fuzz: layout planner invariants --------------------------------
+
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+
+@settings(max_examples=200, deadline=None)
+@given(shapes=st.lists(
+    st.tuples(st.integers(1, 64), st.integers(1, 64)), min_size=1,
+    max_size=24),
+    slab_mb=st.sampled_from([0, 1, 4]))
+def test_plan_layout_invariants(shapes, slab_mb):
+    """For ANY tensor list: offsets 256-B aligned, tensors never overlap,
+    never straddle a slab boundary, and slab sizes sum to the total."""
+    import torch
+
+    from fma_amd.ops.actuation import ARENA_ALIGN, plan_layout
+
+    specs = [(f"t{i}", s, torch.bfloat16) for i, s in enumerate(shapes)]
+    slab_bytes = slab_mb << 20
+    layout, total, slab_sizes = plan_layout(specs, slab_bytes)
+    assert sum(slab_sizes) == total
+    slab_edges = []
+    acc = 0
+    for sz in slab_sizes:
+        acc += sz
+        slab_edges.append(acc)
+        if slab_bytes:
+            assert sz <= max(slab_bytes,
+                             max(2 * a * b for _, (a, b), _ in specs))
+    prev_end = 0
+    for name, (sh, _), _ in ((n, (s, d), d) for n, s, d in specs):
+        off, shape, dtype = layout[name]
+        nbytes = 2 * shape[0] * shape[1]
+        assert off % ARENA_ALIGN == 0
+        assert off >= prev_end, "overlap with previous tensor"
+        # never straddles a slab edge
+        start = 0
+        for edge in slab_edges:
+            if off < edge:
+                assert off + nbytes <= edge, \
+                    f"{name} straddles slab edge {edge}"
+                break
+            start = edge
+        prev_end = off + nbytes
+    assert prev_end <= total
