@@ -72,3 +72,31 @@ def test_template_hash_stable_under_canonicalization():
                                              {"containerPort": 9}]}]}}
     assert template_hash(canonicalize_template(t1)) == \
         template_hash(canonicalize_template(t2))
+
+
+# -- property fuzz: instance identity ---------------------------------------
+
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+_opts = st.text(st.characters(whitelist_categories=("Ll", "Nd"),
+                              whitelist_characters=" -"), max_size=30)
+
+
+@settings(max_examples=150, deadline=None)
+@given(opts=_opts, port=st.integers(1024, 65535),
+       gpus=st.lists(st.sampled_from(["GPU-0", "GPU-1", "GPU-2"]),
+                     min_size=1, max_size=3, unique=True))
+def test_instance_id_properties(opts, port, gpus):
+    """Identity is a pure function of (ModelServerConfig, GPU list):
+    deterministic across calls, I...i shaped (reference
+    inference-server.go:1016-1058), and sensitive to every input."""
+    from fma_amd.controller.dualpods.identity import instance_id
+
+    msc = {"port": port, "options": opts}
+    a = instance_id(msc, gpus)
+    assert a == instance_id(dict(msc), list(gpus))
+    assert a.startswith("I") and a.endswith("i") and len(a) > 10
+    assert a != instance_id({"port": port, "options": opts + "x"}, gpus)
+    assert a != instance_id({"port": (port % 65534) + 1, "options": opts},
+                            gpus)
+    assert a != instance_id(msc, gpus + ["GPU-9"])
