@@ -229,3 +229,28 @@ def test_protected_fields_fuzz(key, value, actor):
                 f"{actor!r} mutated protected {key!r} unchallenged")
         except Invalid:
             pass
+
+
+@settings(max_examples=200, deadline=None)
+@given(n=st.integers(0, 10**12),
+       suffix=st.sampled_from([None, "m", "k", "M", "G", "Ki", "Mi", "Gi",
+                               "Ti"]))
+def test_parse_quantity_properties(n, suffix):
+    """k8s quantity parsing (reference resource.Quantity subset): value
+    scales by the suffix factor, ordering is preserved, garbage raises."""
+    import pytest
+
+    from fma_amd.api.types import parse_quantity
+
+    factors = {None: 1, "m": 1e-3, "k": 1e3, "M": 1e6, "G": 1e9,
+               "Ki": 2**10, "Mi": 2**20, "Gi": 2**30, "Ti": 2**40}
+    text = f"{n}{suffix or ''}"
+    v = parse_quantity(text)
+    assert v == n * factors[suffix]
+    # ordering across representations
+    assert parse_quantity(f"{n + 1}{suffix or ''}") > v or n * factors[
+        suffix] == (n + 1) * factors[suffix]
+    # ints/floats pass through
+    assert parse_quantity(n) == float(n)
+    with pytest.raises(ValueError):
+        parse_quantity(f"{n}Zz")
