@@ -254,3 +254,48 @@ def test_parse_quantity_properties(n, suffix):
     assert parse_quantity(n) == float(n)
     with pytest.raises(ValueError):
         parse_quantity(f"{n}Zz")
+
+
+def test_memstore_concurrent_updates_one_winner_per_rv():
+    """Optimistic concurrency under real thread contention: N threads
+    race to update the same object with the SAME expected RV — exactly
+    one wins per round, everyone else gets Conflict (the apiserver
+    semantic every reconcile loop here leans on)."""
+    import threading
+
+    from fma_amd.store import objects as ob
+    from fma_amd.store.memstore import Conflict, MemStore
+
+    store = MemStore()
+    store.create(ob.new_object("Pod", "contended", spec={}))
+    rounds = 20
+    threads = 8
+    wins = []
+    for _ in range(rounds):
+        cur = store.get("Pod", "contended")
+        rv = ob.rv_of(cur)
+        barrier = threading.Barrier(threads)
+        results = [None] * threads
+
+        def attempt(i):
+            import copy
+            mine = copy.deepcopy(cur)
+            ob.annotations_of(mine)["winner"] = f"t{i}"
+            barrier.wait()
+            try:
+                store.update(mine, expect_rv=rv)
+                results[i] = "win"
+            except Conflict:
+                results[i] = "lose"
+
+        ts = [threading.Thread(target=attempt, args=(i,))
+              for i in range(threads)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join()
+        assert results.count("win") == 1, results
+        wins.append(results.index("win"))
+    # the final object reflects the last winner exactly
+    final = store.get("Pod", "contended")
+    assert ob.annotations_of(final)["winner"] == f"t{wins[-1]}"
