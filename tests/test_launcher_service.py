@@ -199,3 +199,33 @@ def test_gpu_uuid_translation_sets_env(client, monkeypatch):
     monkeypatch.setattr(instance_mod.ServerInstance, "start", spy)
     client.put(f"{ROOT}/tr", json=_mkconfig(gpu_uuids=["GPU-2", "GPU-0"]))
     assert seen["HIP_VISIBLE_DEVICES"] == "2,0"
+
+
+def test_log_range_fuzz(client):
+    """RFC 9110 byte-range semantics over many random ranges: 206 slices
+    match python slicing of the full body; start==size is 416; suffix
+    ranges return the tail (reference launcher.py log endpoint +
+    docs/launcher.md range table)."""
+    import random
+
+    client.put(f"{ROOT}/rf", json=_mkconfig())
+    time.sleep(0.3)
+    full = client.get(f"{ROOT}/rf/log").content
+    size = len(full)
+    assert size > 10
+    rng = random.Random(7)
+    for _ in range(40):
+        a = rng.randrange(0, size + 3)
+        b = rng.randrange(a, size + 5)
+        r = client.get(f"{ROOT}/rf/log",
+                       headers={"Range": f"bytes={a}-{b}"})
+        if a >= size:
+            assert r.status_code == 416, (a, b, size)
+        else:
+            assert r.status_code == 206, (a, b, size)
+            assert r.content == full[a:b + 1], (a, b)
+    for n in (1, 3, size, size + 10):
+        r = client.get(f"{ROOT}/rf/log", headers={"Range": f"bytes=-{n}"})
+        assert r.status_code == 206
+        expected = full[-n:] if n <= size else full
+        assert r.content == expected, n
