@@ -229,3 +229,30 @@ def test_log_range_fuzz(client):
         assert r.status_code == 206
         expected = full[-n:] if n <= size else full
         assert r.content == expected, n
+
+
+def test_broadcaster_buffer_wrap_fuzz():
+    """Revision semantics after buffer eviction, fuzzed: for any number
+    of appended events (beyond BUFFER_LIMIT), check_since accepts any
+    `since` within the buffer and 410s anything that predates it —
+    never a silent gap (reference launcher.py watch + 410 Gone)."""
+    import random
+
+    from fma_amd.launcher.broadcaster import EventBroadcaster
+
+    rng = random.Random(11)
+    for _ in range(10):
+        b = EventBroadcaster()
+        n = rng.randrange(1, b.BUFFER_LIMIT * 2 + 7)
+        for i in range(n):
+            b.append("MODIFIED", f"I{i}", b.next_revision())
+        oldest = b.oldest_buffered_revision
+        assert oldest == max(1, n - b.BUFFER_LIMIT + 1)
+        # inside the buffer (or exactly one before): fine
+        b.check_since(oldest - 1)
+        b.check_since(n)
+        if oldest > 2:
+            with pytest.raises(RevisionTooOld):
+                b.check_since(oldest - 2)
+        # since=0 (fresh client) never raises
+        b.check_since(0)
