@@ -40,7 +40,8 @@ from fma_amd.controller.dualpods.selection import (InstanceView,
 from fma_amd.controller.httpadapter import LauncherClient
 from fma_amd.controller.workqueue import QueueAndWorkers
 from fma_amd.store import objects as ob
-from fma_amd.store.memstore import Conflict, MemStore, NotFound
+from fma_amd.store.memstore import (Conflict, MemStore, NotFound,
+                                    RevisionTooOld)
 
 REQUESTER_FINALIZER = "dual-pods.llm-d.ai/requester-protection"
 PROVIDER_FINALIZER = "dual-pods.llm-d.ai/provider-protection"
@@ -119,7 +120,18 @@ class DualPodsController:
             self._enqueue_isc_gc(ob.name_of(isc))
 
     def _watch_loop(self) -> None:
-        since = self.store.list_revision()
+        # Informer loop with 410 recovery: when the bounded event history
+        # evicts past our cursor, re-LIST (resync) and resume from the
+        # list revision, as a Kubernetes reflector would.
+        while not self._stop.is_set():
+            since = self.store.list_revision()
+            try:
+                self._watch_once(since)
+                return  # stop was set
+            except RevisionTooOld:
+                self.resync()
+
+    def _watch_once(self, since: int) -> None:
         # replay everything from before start too (resync covered it)
         for ev in self.store.watch(since=since, stop=self._stop):
             if ev.kind == "Pod":

@@ -32,7 +32,8 @@ from fma_amd.controller.populator.podtemplate import (
     build_node_independent_template, build_launcher_pod, specialize_to_node)
 from fma_amd.controller.workqueue import InitialSyncTracker, QueueAndWorkers
 from fma_amd.store import objects as ob
-from fma_amd.store.memstore import Conflict, MemStore, NotFound
+from fma_amd.store.memstore import (Conflict, MemStore, NotFound,
+                                    RevisionTooOld)
 
 HANDS_OFF = -1  # LC missing/malformed: do not create or delete
 
@@ -131,7 +132,30 @@ class LauncherPopulator:
         self.key_queue.stop()
 
     def _watch_loop(self) -> None:
-        since = self.store.list_revision()
+        # Informer loop with 410 recovery: on history eviction, re-enqueue
+        # a full digest pass and resume from the current list revision.
+        while not self._stop.is_set():
+            since = self.store.list_revision()
+            try:
+                self._watch_once(since)
+                return  # stop was set
+            except RevisionTooOld:
+                self._requeue_all()
+
+    def _requeue_all(self) -> None:
+        for kind, tag in (("LauncherConfig", "lc"),
+                          ("LauncherPopulationPolicy", "lpp"),
+                          ("Node", "node")):
+            for o in self.store.list(kind, self.ns):
+                self.digest_queue.queue.add((tag, ob.name_of(o)))
+        if self._keys_started.is_set():
+            with self.policy.lock:
+                keys = {(n, lc) for n, per in self.policy.desired.items()
+                        for lc in per}
+            for key in keys:
+                self.key_queue.queue.add(key)
+
+    def _watch_once(self, since: int) -> None:
         for ev in self.store.watch(since=since, stop=self._stop):
             if ev.kind == "LauncherPopulationPolicy":
                 self.digest_queue.queue.add(("lpp", ob.name_of(ev.obj)))

@@ -50,6 +50,7 @@ class InstanceManager:
 
     def attach_loop(self, loop: asyncio.AbstractEventLoop) -> None:
         self.loop = loop
+        self.broadcaster.attach_loop(loop)
 
     # -- operations ----------------------------------------------------------
 
@@ -209,6 +210,12 @@ def create_app(manager: Optional[InstanceManager] = None) -> FastAPI:
                     if await request.is_disconnected():
                         return
                     yield json.dumps(ev) + "\n"
+            except RevisionTooOld as e:
+                # Slow consumer overtaken by buffer eviction: emit one
+                # terminal error line (status is already sent) and close so
+                # the client re-LISTs, as with a fresh 410.
+                yield json.dumps({"error": str(e), "code": 410}) + "\n"
+                return
             except asyncio.CancelledError:  # pragma: no cover
                 return
 

@@ -33,7 +33,7 @@ import httpx
 from fma_amd.api import contracts
 from fma_amd.launcher.notifier import PodNotifier
 from fma_amd.store import objects as ob
-from fma_amd.store.memstore import Conflict, NotFound
+from fma_amd.store.memstore import Conflict, NotFound, RevisionTooOld
 
 
 class PodProcess:
@@ -90,10 +90,15 @@ class NodeAgent:
     # -- reconcile -----------------------------------------------------------
 
     def _watch_loop(self) -> None:
-        self._sync_all()
-        for ev in self.store.watch(since=self.store.list_revision(),
-                                   kinds=["Pod"], stop=self._stop):
-            self._handle(ev.type, ev.obj)
+        while not self._stop.is_set():
+            self._sync_all()
+            try:
+                for ev in self.store.watch(since=self.store.list_revision(),
+                                           kinds=["Pod"], stop=self._stop):
+                    self._handle(ev.type, ev.obj)
+                return  # stop was set
+            except RevisionTooOld:
+                continue  # re-LIST and resume from the list revision
 
     def _sync_all(self) -> None:
         for pod in self.store.list("Pod", self.ns):

@@ -15,7 +15,7 @@ import httpx
 
 from fma_amd.store import objects as ob
 from fma_amd.store.memstore import (ApiError, Conflict, Invalid,
-                                    NotFound, WatchEvent)
+                                    NotFound, RevisionTooOld, WatchEvent)
 
 
 def _raise_for(code: int, message: str) -> None:
@@ -23,6 +23,8 @@ def _raise_for(code: int, message: str) -> None:
         raise NotFound(message)
     if code == 409:
         raise Conflict(message)
+    if code == 410:
+        raise RevisionTooOld(message)
     if code == 422:
         raise Invalid(message)
     raise ApiError(code, message)
@@ -137,6 +139,8 @@ class StoreClient:
                         if not line:
                             continue
                         d = json.loads(line)
+                        if d.get("code") == 410:
+                            raise RevisionTooOld(d.get("error", "revision too old"))
                         params["since"] = max(params["since"], d["revision"])
                         yield WatchEvent(d["revision"], d["type"], d["kind"],
                                          d["object"])

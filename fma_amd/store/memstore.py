@@ -55,6 +55,15 @@ class Invalid(ApiError):
         super().__init__(422, message)
 
 
+class RevisionTooOld(ApiError):
+    """Watch cursor predates the bounded event history (events were
+    evicted unseen): the watcher must re-LIST and resume from the list
+    revision. Mirrors Kubernetes' 410 Gone on stale resourceVersions."""
+
+    def __init__(self, message: str):
+        super().__init__(410, message)
+
+
 class WatchEvent:
     __slots__ = ("revision", "type", "kind", "obj")
 
@@ -264,13 +273,20 @@ class MemStore:
         """Yield events with revision > since; blocks for new ones.
 
         Generator exits when ``stop`` is set or ``timeout`` elapses with no
-        new events (informers loop around it).
+        new events (informers loop around it). Raises RevisionTooOld when
+        history eviction overtakes the cursor (the watcher must re-LIST and
+        resume from the list revision, as with Kubernetes 410 Gone).
         """
         cursor = since
         while True:
             batch: List[WatchEvent] = []
             with self._lock:
                 while True:
+                    if cursor and self._history \
+                            and cursor + 1 < self._history[0].revision:
+                        raise RevisionTooOld(
+                            f"watch cursor {cursor} predates history; "
+                            f"oldest is {self._history[0].revision}")
                     batch = [e for e in self._history if e.revision > cursor
                              and (kinds is None or e.kind in kinds)]
                     if batch:

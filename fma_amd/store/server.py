@@ -25,7 +25,7 @@ from typing import Optional
 from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, StreamingResponse
 
-from fma_amd.store.memstore import ApiError, MemStore
+from fma_amd.store.memstore import ApiError, MemStore, RevisionTooOld
 
 
 def create_app(store: Optional[MemStore] = None) -> FastAPI:
@@ -102,10 +102,16 @@ def create_app(store: Optional[MemStore] = None) -> FastAPI:
             while True:
                 if await request.is_disconnected():
                     return
-                batch = await loop.run_in_executor(
-                    None, lambda: list(st.watch(since=cursor,
-                                                kinds=kind_list,
-                                                timeout=1.0)))
+                try:
+                    batch = await loop.run_in_executor(
+                        None, lambda: list(st.watch(since=cursor,
+                                                    kinds=kind_list,
+                                                    timeout=1.0)))
+                except RevisionTooOld as e:
+                    # Terminal line: status already streamed, so signal 410
+                    # in-band; the client raises and its caller re-LISTs.
+                    yield json.dumps({"error": e.message, "code": 410}) + "\n"
+                    return
                 for ev in batch:
                     cursor = max(cursor, ev.revision)
                     yield json.dumps({

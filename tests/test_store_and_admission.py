@@ -299,3 +299,37 @@ def test_memstore_concurrent_updates_one_winner_per_rv():
     # the final object reflects the last winner exactly
     final = store.get("Pod", "contended")
     assert ob.annotations_of(final)["winner"] == f"t{wins[-1]}"
+
+
+def test_watch_raises_revision_too_old_on_history_eviction():
+    """A watch cursor that falls behind the trimmed event history raises
+    RevisionTooOld (410) so the consumer re-LISTs, instead of silently
+    skipping evicted events (ADVICE round-1, memstore.watch)."""
+    from fma_amd.store.memstore import MemStore, RevisionTooOld
+
+    st = MemStore()
+    st._history_cap = 50
+    first = st.create({"kind": "Pod", "metadata": {"name": "p0"}})
+    start_rev = st.list_revision()
+    for i in range(100):
+        st.create({"kind": "Pod", "metadata": {"name": f"p{i+1}"}})
+    # history was trimmed past start_rev
+    with pytest.raises(RevisionTooOld):
+        for _ in st.watch(since=start_rev, timeout=0.1):
+            pass
+    del first
+
+
+def test_watch_current_cursor_survives_eviction():
+    """A caught-up watcher is NOT disturbed by eviction of events it
+    already consumed."""
+    from fma_amd.store.memstore import MemStore
+
+    st = MemStore()
+    st._history_cap = 50
+    for i in range(100):
+        st.create({"kind": "Pod", "metadata": {"name": f"q{i}"}})
+    cursor = st.list_revision()
+    st.create({"kind": "Pod", "metadata": {"name": "fresh"}})
+    evs = list(st.watch(since=cursor, timeout=0.1))
+    assert [e.obj["metadata"]["name"] for e in evs] == ["fresh"]
