@@ -81,6 +81,11 @@ LAUNCHER_COMPONENT = "launcher"
 LAUNCHER_CONFIG_NAME_LABEL = "dual-pods.llm-d.ai/launcher-config-name"
 NODE_NAME_LABEL = "dual-pods.llm-d.ai/node-name"
 LAUNCHER_CONFIG_HASH_ANNOTATION = "dual-pods.llm-d.ai/launcher-config-hash"
+#: nominal-provider hash of a direct (launcher-less) provider Pod, used for
+#: sleeper lookup (reference pkg/controller/dual-pods/controller.go:102);
+#: distinct from LAUNCHER_CONFIG_HASH_ANNOTATION, which is reserved for
+#: launcher-based providers (reference inference-server.go:687)
+NOMINAL_ANNOTATION = "dual-pods.llm-d.ai/nominal"
 LAUNCHER_TEMPLATE_HASH_ANNOTATION = (
     "dual-pods.llm-d.ai/launcher-populator-template-hash"
 )
@@ -102,6 +107,11 @@ PROXY_CONFIG_PATH = "/v1/proxy/config"
 
 PROBES_PORT_DEFAULT = 8080
 SPI_PORT_DEFAULT = 8081
+#: fixed TCP reverse-proxy listen port (reference cmd/requester/main.go
+#: --proxy-port default 8082): the controller does not consume the
+#: listen_port returned by PUT /v1/proxy/config, so in-cluster traffic
+#: must find the proxy at a known port
+PROXY_PORT_DEFAULT = 8082
 
 # ---------------------------------------------------------------------------
 # Inference-server sleep contract (reference pkg/api/interface.go:131-135 and

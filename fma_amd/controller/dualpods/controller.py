@@ -685,7 +685,7 @@ class DualPodsController:
             sdata.readiness_relayed = True
             path = "cold" if sdata.needed_new_launcher else \
                 ("warm" if sdata.needed_new_instance else "hot")
-            start = ob.meta(requester).get("creationTimestamp") or \
+            start = ob.pod_container_start_time(requester) or \
                 self.clock.time()
             isc_name = ob.annotations_of(requester).get(
                 contracts.INFERENCE_SERVER_CONFIG_ANNOTATION, "")
@@ -745,7 +745,7 @@ class DualPodsController:
         sleeper = None
         for pod in self.store.list("Pod", self.ns):
             if ob.annotations_of(pod).get(
-                    contracts.LAUNCHER_CONFIG_HASH_ANNOTATION) == nom_hash \
+                    contracts.NOMINAL_ANNOTATION) == nom_hash \
                     and not ob.annotations_of(pod).get(
                         contracts.REQUESTER_ANNOTATION) \
                     and not ob.is_deleting(pod):
@@ -804,23 +804,33 @@ class DualPodsController:
         return self._relay_readiness(requester, sdata)
 
     def _enforce_sleeper_budget(self, node: str, sdata: ServerData) -> None:
-        """Per-GPU cap on sleeping direct providers: delete the oldest over
-        the limit (reference enforceSleeperBudget, :1354-1428)."""
+        """Per-GPU cap on sleeping direct providers: for each GPU the new
+        provider will occupy, delete the oldest sleepers using that GPU
+        beyond sleeper_limit (reference enforceSleeperBudget,
+        inference-server.go:1354-1428, which loops over GPUIndices; a
+        node-wide count would evict 8x too aggressively on an 8-GPU node)."""
+        # unbound sleeping direct providers on this node, by occupied GPU
         sleepers = []
         for pod in self.store.list("Pod", self.ns):
             if ob.pod_node_name(pod) != node:
                 continue
             if ob.annotations_of(pod).get(contracts.REQUESTER_ANNOTATION):
                 continue
+            if ob.is_deleting(pod):
+                continue
             if ob.labels_of(pod).get(contracts.SLEEPING_LABEL) == "true" and \
-                    ob.annotations_of(pod).get(
-                        contracts.LAUNCHER_CONFIG_HASH_ANNOTATION):
+                    ob.annotations_of(pod).get(contracts.NOMINAL_ANNOTATION):
                 sleepers.append(pod)
-        excess = len(sleepers) - self.cfg.sleeper_limit + 1
-        if excess <= 0:
-            return
         sleepers.sort(key=lambda p: ob.meta(p).get("creationTimestamp") or 0)
-        for pod in sleepers[:excess]:
+        victims: Dict[str, Dict[str, Any]] = {}
+        for gpu in sdata.gpus or []:
+            on_gpu = [p for p in sleepers
+                      if gpu in ob.annotations_of(p).get(
+                          contracts.ACCELERATORS_ANNOTATION, "").split(",")]
+            excess = len(on_gpu) - self.cfg.sleeper_limit
+            for pod in on_gpu[:max(excess, 0)]:
+                victims[ob.uid_of(pod)] = pod
+        for pod in victims.values():
             try:
                 self.store.delete("Pod", ob.name_of(pod), self.ns,
                                   actor="dual-pods-controller",

@@ -144,5 +144,10 @@ def build_nominal_provider(
         res[contracts.GPU_RESOURCE_NAME] = "0"
 
     h = nominal_hash(spec, gpu_uuids, node_name)
-    ob.annotations_of(pod)[contracts.LAUNCHER_CONFIG_HASH_ANNOTATION] = h
+    anns = ob.annotations_of(pod)
+    anns[contracts.NOMINAL_ANNOTATION] = h
+    # record the GPUs this provider occupies so the per-GPU sleeper budget
+    # can count sleepers by accelerator (reference indexes providers by GPU,
+    # controller.go:129-159; annotation key pkg/api/interface.go)
+    anns[contracts.ACCELERATORS_ANNOTATION] = ",".join(gpu_uuids)
     return pod, h

@@ -45,6 +45,7 @@ class PodProcess:
         self.proc = proc
         self.probe = probe
         self.restarts = 0
+        self.started_at = time.time()
         self.next_restart = 0.0
         # k8s default is Always; "Never" lets a crash surface as Failed
         self.restart_policy = (pod.get("spec") or {}).get(
@@ -317,6 +318,7 @@ class NodeAgent:
         except OSError:
             return
         pp.restarts += 1
+        pp.started_at = time.time()
         pp.next_restart = 0.0
         self._patch_status(pp, phase="Running", ready=False, ip=pp.ip)
 
@@ -338,8 +340,11 @@ class NodeAgent:
         status["podIP"] = ip
         # kubelet-style restart accounting (PodIsInTrouble in the
         # reference, utils/pod-helper.go:44, keys off this)
-        status["containerStatuses"] = [{"restartCount": pp.restarts,
-                                        "ready": ready}]
+        status["containerStatuses"] = [
+            {"restartCount": pp.restarts, "ready": ready,
+             # actuation latency is measured from container start, not Pod
+             # creation (reference inference-server.go:574-591)
+             "state": {"running": {"startedAt": pp.started_at}}}]
         ob.set_pod_ready(cur, ready)
         status.setdefault("startTime", ob.meta(cur).get("creationTimestamp"))
         try:

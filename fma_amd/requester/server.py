@@ -237,7 +237,9 @@ def create_spi_app(state: RequesterState) -> FastAPI:
 def main() -> None:
     import uvicorn
 
-    state = RequesterState()
+    proxy_port = int(os.environ.get("PROXY_PORT",
+                                    contracts.PROXY_PORT_DEFAULT))
+    state = RequesterState(proxy_listen_port=proxy_port)
     probes_port = int(os.environ.get("PROBES_PORT",
                                      contracts.PROBES_PORT_DEFAULT))
     spi_port = int(os.environ.get("SPI_PORT", contracts.SPI_PORT_DEFAULT))

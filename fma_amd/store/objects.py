@@ -138,6 +138,25 @@ def pod_containers(pod: Dict[str, Any]) -> List[Dict[str, Any]]:
     return pod.get("spec", {}).get("containers", [])
 
 
+def pod_container_start_time(pod: Dict[str, Any]) -> Optional[float]:
+    """Start time of the Pod's running container(s): the origin for the
+    fma_actuation_seconds histogram (reference inference-server.go:574-591
+    reads ContainerStatuses state.running.startedAt). Falls back to
+    status.startTime, then creationTimestamp."""
+    best = None
+    for cs in pod.get("status", {}).get("containerStatuses", []):
+        t = ((cs.get("state") or {}).get("running") or {}).get("startedAt")
+        if isinstance(t, (int, float)):
+            best = t if best is None else min(best, t)
+    if best is not None:
+        return best
+    t = pod.get("status", {}).get("startTime")
+    if isinstance(t, (int, float)):
+        return t
+    t = meta(pod).get("creationTimestamp")
+    return t if isinstance(t, (int, float)) else None
+
+
 def find_container(pod: Dict[str, Any], name: str) -> Optional[Dict[str, Any]]:
     for c in pod_containers(pod):
         if c.get("name") == name:

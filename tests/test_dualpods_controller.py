@@ -443,5 +443,5 @@ def test_direct_sleeper_reuse():
     assert not srv.sleeping  # woken, not re-created
     providers = [p for p in w["store"].list("Pod")
                  if ob.annotations_of(p).get(
-                     C.LAUNCHER_CONFIG_HASH_ANNOTATION)]
+                     C.NOMINAL_ANNOTATION)]
     assert len(providers) == 1

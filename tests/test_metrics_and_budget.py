@@ -180,3 +180,50 @@ def test_innerqueue_metrics_emitted():
     assert adds and adds >= 2
     assert retries and retries >= 1
     assert work and work >= 2
+
+
+def _mk_sleeper(store, name, gpus, ts):
+    pod = ob.new_object(
+        "Pod", name,
+        labels={C.SLEEPING_LABEL: "true"},
+        annotations={C.NOMINAL_ANNOTATION: "h-" + name,
+                     C.ACCELERATORS_ANNOTATION: ",".join(gpus)},
+        spec={"nodeName": "node-a",
+              "containers": [{"name": "inference-server"}]})
+    pod = store.create(pod)
+    ob.meta(pod)["creationTimestamp"] = ts
+    return store.update(pod)
+
+
+def test_sleeper_budget_is_per_gpu():
+    """The budget counts sleepers per GPU index, not per node (reference
+    enforceSleeperBudget inference-server.go:1354-1428): a new provider on
+    GPU-0 must not evict the only sleeper on GPU-1."""
+    from fma_amd.controller.dualpods.controller import ServerData
+    from fma_amd.store.memstore import MemStore
+
+    store = MemStore()
+    ctl = DualPodsController(store, None, ControllerConfig(sleeper_limit=1))
+    _mk_sleeper(store, "s-g0-old", ["GPU-0"], "2026-01-01T00:00:00Z")
+    _mk_sleeper(store, "s-g0-new", ["GPU-0"], "2026-01-02T00:00:00Z")
+    _mk_sleeper(store, "s-g1", ["GPU-1"], "2026-01-01T00:00:00Z")
+
+    sdata = ServerData(uid="u1", requester_name="r1", gpus=["GPU-0"])
+    ctl._enforce_sleeper_budget("node-a", sdata)
+
+    names = {ob.name_of(p) for p in store.list("Pod")}
+    # GPU-0 had 2 sleepers with limit 1: the oldest goes; GPU-1 untouched
+    assert names == {"s-g0-new", "s-g1"}
+
+
+def test_sleeper_budget_within_limit_no_eviction():
+    from fma_amd.controller.dualpods.controller import ServerData
+    from fma_amd.store.memstore import MemStore
+
+    store = MemStore()
+    ctl = DualPodsController(store, None, ControllerConfig(sleeper_limit=2))
+    _mk_sleeper(store, "s1", ["GPU-0"], "2026-01-01T00:00:00Z")
+    _mk_sleeper(store, "s2", ["GPU-0"], "2026-01-02T00:00:00Z")
+    sdata = ServerData(uid="u1", requester_name="r1", gpus=["GPU-0"])
+    ctl._enforce_sleeper_budget("node-a", sdata)
+    assert len(store.list("Pod")) == 2

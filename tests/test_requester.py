@@ -88,3 +88,38 @@ def test_proxy_configure_once_and_forward():
 def test_proxy_bad_body():
     _, _, spi = make_clients()
     assert spi.put("/v1/proxy/config", json={"address": "x"}).status_code == 400
+
+
+def test_main_pins_proxy_port(monkeypatch):
+    """main() must pin the proxy listener to PROXY_PORT (default 8082,
+    reference cmd/requester/main.go --proxy-port): the controller ignores
+    the listen_port returned by PUT /v1/proxy/config, so an ephemeral port
+    would be unreachable in a real deployment (ADVICE round-1)."""
+    import fma_amd.requester.server as rs
+    from fma_amd.api import contracts
+
+    captured = {}
+
+    def fake_run(app, **kw):
+        captured.setdefault("apps", []).append(app)
+
+    class FakeUvicorn:
+        run = staticmethod(fake_run)
+
+    state_holder = {}
+    orig_state = rs.RequesterState
+
+    def capture_state(proxy_listen_port=0):
+        st = orig_state(proxy_listen_port)
+        state_holder["state"] = st
+        return st
+
+    monkeypatch.setattr(rs, "RequesterState", capture_state)
+    monkeypatch.setitem(__import__("sys").modules, "uvicorn", FakeUvicorn)
+    monkeypatch.setenv("PROXY_PORT", "18982")
+    rs.main()
+    assert state_holder["state"].proxy.listen_port == 18982
+    monkeypatch.delenv("PROXY_PORT")
+    rs.main()
+    assert state_holder["state"].proxy.listen_port == \
+        contracts.PROXY_PORT_DEFAULT
