@@ -38,8 +38,9 @@ from fma_amd.controller.dualpods.selection import (InstanceView,
                                                    LauncherView,
                                                    select_or_reclaim)
 from fma_amd.controller.httpadapter import LauncherClient
-from fma_amd.controller.workqueue import QueueAndWorkers
+from fma_amd.controller.workqueue import NodeQueueAndWorkers
 from fma_amd.store import objects as ob
+from fma_amd.store.indexes import install_pod_indexes
 from fma_amd.store.memstore import (Conflict, MemStore, NotFound,
                                     RevisionTooOld)
 
@@ -85,6 +86,7 @@ class DualPodsController:
     def __init__(self, store: MemStore, http, config: ControllerConfig = None,
                  clock=time):
         self.store = store
+        install_pod_indexes(store)  # O(1) lookups (ref controller.go:129-159)
         self.http = http
         self.cfg = config or ControllerConfig()
         self.clock = clock
@@ -93,9 +95,13 @@ class DualPodsController:
         self.node_locks: Dict[str, threading.Lock] = {}
         self._lock = threading.Lock()
         self._stop = threading.Event()
-        self.workers = QueueAndWorkers("dualpods", self.cfg.num_workers,
-                                       self._process, max_backoff=2.0,
-                                       metrics_name="dualpods")
+        # two-level per-node queue (reference controller.go:404-424):
+        # the node is element [1] of every item tuple; one worker drains
+        # one node at a time, ready items oldest-first
+        self.workers = NodeQueueAndWorkers(
+            "dualpods", self.cfg.num_workers, self._process,
+            node_of=lambda item: item[1], max_backoff=2.0,
+            metrics_name="dualpods")
         self._watch_thread: Optional[threading.Thread] = None
 
     # ------------------------------------------------------------------
@@ -142,11 +148,10 @@ class DualPodsController:
                 if ev.type in ("MODIFIED", "DELETED"):
                     self._enqueue_isc_gc(ob.name_of(ev.obj))
                 # requesters referencing it may now be actionable
-                for pod in self.store.list("Pod", self.ns):
-                    if ob.annotations_of(pod).get(
-                            contracts.INFERENCE_SERVER_CONFIG_ANNOTATION) == \
-                            ob.name_of(ev.obj):
-                        self._enqueue_for(pod)
+                for pod in self.store.index_get(
+                        "Pod", "inferenceserverconfig",
+                        ob.name_of(ev.obj), self.ns):
+                    self._enqueue_for(pod)
 
     # classification (reference careAbout, controller.go:580-593)
     def classify(self, pod: Dict[str, Any]) -> str:
@@ -207,21 +212,13 @@ class DualPodsController:
     def _find_provider_for(self, uid: str, name: str
                            ) -> Optional[Dict[str, Any]]:
         want = f"{uid} {name}"
-        for pod in self.store.list("Pod", self.ns):
-            if ob.annotations_of(pod).get(
-                    contracts.REQUESTER_ANNOTATION) == want:
-                return pod
-        return None
+        hits = self.store.index_get("Pod", "requester", want, self.ns)
+        return hits[0] if hits else None
 
     def _launchers_on_node(self, node: str, lc_name: Optional[str] = None
                            ) -> List[Dict[str, Any]]:
         out = []
-        for pod in self.store.list(
-                "Pod", self.ns,
-                label_selector={contracts.COMPONENT_LABEL:
-                                contracts.LAUNCHER_COMPONENT}):
-            if ob.pod_node_name(pod) != node:
-                continue
+        for pod in self.store.index_get("Pod", "launcherNode", node, self.ns):
             if lc_name and ob.labels_of(pod).get(
                     contracts.LAUNCHER_CONFIG_NAME_LABEL) != lc_name:
                 continue
@@ -743,11 +740,8 @@ class DualPodsController:
 
         # find a matching sleeper by nominal hash (reference :624-642)
         sleeper = None
-        for pod in self.store.list("Pod", self.ns):
-            if ob.annotations_of(pod).get(
-                    contracts.NOMINAL_ANNOTATION) == nom_hash \
-                    and not ob.annotations_of(pod).get(
-                        contracts.REQUESTER_ANNOTATION) \
+        for pod in self.store.index_get("Pod", "nominal", nom_hash, self.ns):
+            if not ob.annotations_of(pod).get(contracts.REQUESTER_ANNOTATION) \
                     and not ob.is_deleting(pod):
                 sleeper = pod
                 break
@@ -809,24 +803,20 @@ class DualPodsController:
         beyond sleeper_limit (reference enforceSleeperBudget,
         inference-server.go:1354-1428, which loops over GPUIndices; a
         node-wide count would evict 8x too aggressively on an 8-GPU node)."""
-        # unbound sleeping direct providers on this node, by occupied GPU
-        sleepers = []
-        for pod in self.store.list("Pod", self.ns):
-            if ob.pod_node_name(pod) != node:
-                continue
-            if ob.annotations_of(pod).get(contracts.REQUESTER_ANNOTATION):
-                continue
-            if ob.is_deleting(pod):
-                continue
-            if ob.labels_of(pod).get(contracts.SLEEPING_LABEL) == "true" and \
-                    ob.annotations_of(pod).get(contracts.NOMINAL_ANNOTATION):
-                sleepers.append(pod)
-        sleepers.sort(key=lambda p: ob.meta(p).get("creationTimestamp") or 0)
+        def is_sleeper(p):
+            return (ob.pod_node_name(p) == node
+                    and not ob.annotations_of(p).get(
+                        contracts.REQUESTER_ANNOTATION)
+                    and not ob.is_deleting(p)
+                    and ob.labels_of(p).get(contracts.SLEEPING_LABEL) == "true"
+                    and ob.annotations_of(p).get(contracts.NOMINAL_ANNOTATION))
+
         victims: Dict[str, Dict[str, Any]] = {}
         for gpu in sdata.gpus or []:
-            on_gpu = [p for p in sleepers
-                      if gpu in ob.annotations_of(p).get(
-                          contracts.ACCELERATORS_ANNOTATION, "").split(",")]
+            on_gpu = [p for p in self.store.index_get("Pod", "gpu", gpu,
+                                                      self.ns)
+                      if is_sleeper(p)]
+            on_gpu.sort(key=lambda p: ob.meta(p).get("creationTimestamp") or 0)
             excess = len(on_gpu) - self.cfg.sleeper_limit
             for pod in on_gpu[:max(excess, 0)]:
                 victims[ob.uid_of(pod)] = pod

@@ -216,3 +216,167 @@ class InitialSyncTracker:
                 fire = True
         if fire:
             self._on_done()
+
+
+class _NodeItemState:
+    __slots__ = ("add_time", "process_after")
+
+    def __init__(self, add_time: float, process_after: float):
+        self.add_time = add_time
+        self.process_after = process_after
+
+
+class TwoLevelQueue(Generic[T]):
+    """Two-level per-node queue (reference pkg/controller/dual-pods/
+    controller.go:404-424 nodeData, add/addAfter/takeReadyItems/
+    earliestPending :1044-1097; nodeItem.process drains ready items
+    oldest-first, inference-server.go:92-143).
+
+    The outer rate-limited queue holds node names; each node keeps a local
+    map of items with their add-time (for oldest-first drain) and
+    process-after (for timed retries). The outer queue's dirty/processing
+    sets guarantee a node is handled by at most one worker at a time, and
+    per-item backoff is tracked per node — so a hot node retries on its own
+    clock and cannot starve other nodes of worker time.
+    """
+
+    def __init__(self, node_of: Callable[[T], str],
+                 max_backoff: float = MAX_BACKOFF_SECONDS,
+                 metrics_name: str = "") -> None:
+        self.node_of = node_of
+        self.max_backoff = max_backoff
+        self.outer: RateLimitingQueue[str] = RateLimitingQueue(
+            max_backoff, metrics_name=metrics_name)
+        self._mu = threading.Lock()
+        self._nodes: dict = {}       # node -> {item: _NodeItemState}
+        self._failures: dict = {}    # (node, item) -> consecutive failures
+
+    def add(self, item: T, delay: float = 0.0) -> None:
+        node = self.node_of(item)
+        now = time.monotonic()
+        with self._mu:
+            d = self._nodes.setdefault(node, {})
+            st = d.get(item)
+            if st is None:
+                d[item] = _NodeItemState(now, now + delay)
+            else:
+                # keep the original add time (oldest-first is by first
+                # enqueue) but never push process-after later
+                st.process_after = min(st.process_after, now + delay)
+        if delay <= 0:
+            self.outer.add(node)
+        else:
+            self.outer.add_after(node, delay)
+
+    def add_after(self, item: T, delay: float) -> None:
+        self.add(item, delay)
+
+    def add_rate_limited(self, item: T) -> None:
+        node = self.node_of(item)
+        with self._mu:
+            n = self._failures.get((node, item), 0)
+            self._failures[(node, item)] = n + 1
+        self.add(item, min(BASE_BACKOFF_SECONDS * (2 ** n),
+                           self.max_backoff))
+
+    def forget(self, item: T) -> None:
+        node = self.node_of(item)
+        with self._mu:
+            self._failures.pop((node, item), None)
+
+    def take_ready(self, node: str) -> List[T]:
+        """Pop every item on the node whose process-after has passed,
+        oldest first (reference takeReadyItems + the drain loop)."""
+        now = time.monotonic()
+        with self._mu:
+            d = self._nodes.get(node)
+            if not d:
+                return []
+            ready = [i for i, st in d.items() if st.process_after <= now]
+            ready.sort(key=lambda i: d[i].add_time)
+            for i in ready:
+                d.pop(i)
+            if not d:
+                self._nodes.pop(node, None)
+            return ready
+
+    def earliest_pending(self, node: str) -> Optional[float]:
+        """Seconds until the node's next not-yet-ready item (reference
+        earliestPending), or None if the node has no pending items."""
+        now = time.monotonic()
+        with self._mu:
+            d = self._nodes.get(node)
+            if not d:
+                return None
+            return max(min(st.process_after for st in d.values()) - now, 0.0)
+
+    def shut_down(self) -> None:
+        self.outer.shut_down()
+
+    def __len__(self) -> int:
+        with self._mu:
+            return sum(len(d) for d in self._nodes.values())
+
+
+class NodeQueueAndWorkers(Generic[T]):
+    """TwoLevelQueue + N workers; one worker drains one node at a time.
+
+    `process(item)` keeps the flat-item signature (the node is embedded in
+    the item; `node_of` extracts it) and the same return protocol as
+    QueueAndWorkers: False done, True exponential retry, float = re-queue
+    after exactly that many seconds."""
+
+    def __init__(self, name: str, num_workers: int,
+                 process: Callable[[T], bool],
+                 node_of: Callable[[T], str],
+                 max_backoff: float = MAX_BACKOFF_SECONDS,
+                 metrics_name: str = ""):
+        self.name = name
+        self.queue: TwoLevelQueue[T] = TwoLevelQueue(
+            node_of, max_backoff, metrics_name=metrics_name)
+        self._mname = metrics_name
+        self.num_workers = num_workers
+        self.process = process
+        self.threads: List[threading.Thread] = []
+
+    def start(self) -> None:
+        for i in range(self.num_workers):
+            th = threading.Thread(target=self._worker, daemon=True,
+                                  name=f"{self.name}-worker-{i}")
+            th.start()
+            self.threads.append(th)
+
+    def _worker(self) -> None:
+        while True:
+            node = self.queue.outer.get()
+            if node is None:
+                return
+            for item in self.queue.take_ready(node):
+                t0 = time.monotonic()
+                try:
+                    retry = self.process(item)
+                except Exception:  # noqa: BLE001 - keep the worker alive
+                    import traceback
+                    traceback.print_exc()
+                    retry = True
+                if self._mname:
+                    from fma_amd.controller import metrics as _m
+                    _m.queue_work_duration_seconds().labels(
+                        self._mname).observe(time.monotonic() - t0)
+                if isinstance(retry, (int, float)) \
+                        and not isinstance(retry, bool) and retry > 0:
+                    self.queue.forget(item)  # scheduled wait, not a failure
+                    self.queue.add(item, float(retry))
+                elif retry:
+                    self.queue.add_rate_limited(item)
+                else:
+                    self.queue.forget(item)
+            delay = self.queue.earliest_pending(node)
+            self.queue.outer.done(node)
+            if delay is not None:
+                self.queue.outer.add_after(node, delay)
+
+    def stop(self) -> None:
+        self.queue.shut_down()
+        for th in self.threads:
+            th.join(timeout=2)

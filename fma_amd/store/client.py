@@ -119,6 +119,24 @@ class StoreClient:
     def _hdr(self, actor: Optional[str]) -> Optional[Dict[str, str]]:
         return {"X-FMA-Actor": actor} if actor else None
 
+    # -- indexes ------------------------------------------------------------
+
+    def add_index(self, kind: str, name: str, fn) -> None:
+        """No-op: the store server pre-installs the standard indexes
+        (store/server.py create_app); registration is server-side."""
+
+    def index_get(self, kind: str, index_name: str, key: str,
+                  namespace: Optional[str] = "default"
+                  ) -> List[Dict[str, Any]]:
+        params: Dict[str, Any] = {"key": key}
+        if namespace is not None:
+            params["namespace"] = namespace
+        r = self._client.get(f"{self.base}/index/{kind}/{index_name}",
+                             params=params)
+        if r.status_code != 200:
+            raise KeyError(r.json().get("error", r.text))
+        return r.json()["items"]
+
     # -- watch --------------------------------------------------------------
 
     def watch(self, since: int = 0, kinds: Optional[List[str]] = None,

@@ -92,6 +92,10 @@ class MemStore:
         self._history: List[WatchEvent] = []
         self._history_cap = 100_000
         self._admission: List[AdmissionHook] = []
+        #: (kind, index_name) -> (fn, {key: {(ns, name): None}})
+        self._indexes: Dict[Tuple[str, str], Tuple[Callable, Dict]] = {}
+        #: (kind, index_name) -> {(ns, name): [keys]}  (for removal)
+        self._indexed_keys: Dict[Tuple[str, str], Dict] = {}
 
     # -- admission ----------------------------------------------------------
 
@@ -104,6 +108,66 @@ class MemStore:
         for hook in self._admission:
             hook(op, old, new, actor)
 
+    # -- indexes -------------------------------------------------------------
+
+    def add_index(self, kind: str, name: str,
+                  fn: Callable[[Dict[str, Any]], List[str]]) -> None:
+        """Register an index (idempotent): fn(obj) -> list of keys. Mirrors
+        the reference's informer indexers (controller.go:129-159); here
+        the index lives store-side and is maintained on every mutation."""
+        with self._lock:
+            if (kind, name) in self._indexes:
+                return
+            idx: Dict[str, Dict] = {}
+            rev: Dict = {}
+            for (k, ns, nm), obj in self._objects.items():
+                if k != kind:
+                    continue
+                keys = fn(obj) or []
+                if keys:
+                    rev[(ns, nm)] = keys
+                    for key in keys:
+                        idx.setdefault(key, {})[(ns, nm)] = None
+            self._indexes[(kind, name)] = (fn, idx)
+            self._indexed_keys[(kind, name)] = rev
+
+    def index_get(self, kind: str, index_name: str, key: str,
+                  namespace: Optional[str] = "default"
+                  ) -> List[Dict[str, Any]]:
+        """All objects whose index keys include ``key`` — a dict hit, not
+        a scan. Raises KeyError for an unregistered index."""
+        with self._lock:
+            if (kind, index_name) not in self._indexes:
+                raise KeyError(f"no index {index_name!r} on {kind}")
+            _, idx = self._indexes[(kind, index_name)]
+            out = []
+            for (ns, nm) in idx.get(key, {}):
+                if namespace is not None and ns != namespace:
+                    continue
+                obj = self._objects.get((kind, ns, nm))
+                if obj is not None:
+                    out.append(ob.deepcopy(obj))
+            return out
+
+    def _index_event(self, type_: str, kind: str, obj: Dict[str, Any]) -> None:
+        ident = (ob.namespace_of(obj), ob.name_of(obj))
+        for (k, iname), (fn, idx) in self._indexes.items():
+            if k != kind:
+                continue
+            rev = self._indexed_keys[(k, iname)]
+            for key in rev.pop(ident, []):
+                bucket = idx.get(key)
+                if bucket is not None:
+                    bucket.pop(ident, None)
+                    if not bucket:
+                        idx.pop(key, None)
+            if type_ != "DELETED":
+                keys = fn(obj) or []
+                if keys:
+                    rev[ident] = keys
+                    for key in keys:
+                        idx.setdefault(key, {})[ident] = None
+
     # -- core CRUD ----------------------------------------------------------
 
     def _emit(self, type_: str, kind: str, obj: Dict[str, Any]) -> None:
@@ -112,6 +176,7 @@ class MemStore:
         self._history.append(ev)
         if len(self._history) > self._history_cap:
             del self._history[: self._history_cap // 10]
+        self._index_event(type_, kind, obj)
         self._lock.notify_all()
 
     def create(self, obj: Dict[str, Any], actor: str = "system") -> Dict[str, Any]:

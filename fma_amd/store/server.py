@@ -32,6 +32,10 @@ def create_app(store: Optional[MemStore] = None) -> FastAPI:
     app = FastAPI(title="fma-amd cluster store")
     st = store or MemStore()
     app.state.store = st
+    # standard Pod indexes are served here so StoreClient.index_get is a
+    # store-side dict hit (reference informer indexers, controller.go:129-159)
+    from fma_amd.store.indexes import install_pod_indexes
+    install_pod_indexes(st)
 
     def actor_of(request: Request) -> str:
         return request.headers.get("X-FMA-Actor", "anonymous")
@@ -91,6 +95,14 @@ def create_app(store: Optional[MemStore] = None) -> FastAPI:
             return {"status": "ok"}
         except ApiError as e:
             return err(e)
+
+    @app.get("/index/{kind}/{iname}")
+    def index_get(kind: str, iname: str, key: str,
+                  namespace: Optional[str] = "default"):
+        try:
+            return {"items": st.index_get(kind, iname, key, namespace)}
+        except KeyError as e:
+            return JSONResponse({"error": str(e)}, status_code=404)
 
     @app.get("/watch")
     async def watch(request: Request, since: int = 0, kinds: str = ""):

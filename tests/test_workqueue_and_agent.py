@@ -229,3 +229,160 @@ def test_crashloop_restarts_with_backoff(tmp_path):
         assert cur["status"]["containerStatuses"][0]["restartCount"] >= 1
     finally:
         agent.stop()
+
+
+# ---------------------------------------------------------------------------
+# Two-level per-node queue (reference controller.go:404-424, 1044-1097)
+# ---------------------------------------------------------------------------
+
+
+def test_two_level_queue_oldest_first_drain():
+    from fma_amd.controller.workqueue import TwoLevelQueue
+
+    q = TwoLevelQueue(node_of=lambda it: it[0])
+    q.add(("n1", "c"))
+    q.add(("n1", "a"))
+    q.add(("n1", "b"))
+    q.add(("n1", "c"))  # re-add keeps original add time (stays first)
+    assert q.take_ready("n1") == [("n1", "c"), ("n1", "a"), ("n1", "b")]
+    assert q.take_ready("n1") == []
+
+
+def test_two_level_queue_process_after():
+    import time as _t
+
+    from fma_amd.controller.workqueue import TwoLevelQueue
+
+    q = TwoLevelQueue(node_of=lambda it: it[0])
+    q.add(("n1", "later"), delay=0.25)
+    q.add(("n1", "now"))
+    assert q.take_ready("n1") == [("n1", "now")]
+    pending = q.earliest_pending("n1")
+    assert pending is not None and 0 < pending <= 0.25
+    _t.sleep(0.3)
+    assert q.take_ready("n1") == [("n1", "later")]
+    assert q.earliest_pending("n1") is None
+
+
+def test_hot_node_cannot_starve_other_nodes():
+    """One node whose item always fails (exponential per-node backoff)
+    must not delay another node's items, even with a single worker."""
+    import threading
+    import time as _t
+
+    from fma_amd.controller.workqueue import NodeQueueAndWorkers
+
+    done = threading.Event()
+    calls = []
+
+    def process(item):
+        calls.append(item)
+        if item[0] == "hot":
+            return True  # always retry
+        done.set()
+        return False
+
+    w = NodeQueueAndWorkers("t", 1, process, node_of=lambda it: it[0])
+    w.start()
+    try:
+        w.queue.add(("hot", "spinner"))
+        _t.sleep(0.05)  # let the hot node fail a few times first
+        w.queue.add(("cold", "one-shot"))
+        assert done.wait(timeout=2.0), \
+            f"cold item starved; calls={calls[:10]}"
+    finally:
+        w.stop()
+
+
+def test_node_serialized_but_nodes_parallel():
+    """Items of one node never run concurrently; items of different nodes
+    do (two workers, a barrier that only two nodes together can pass)."""
+    import threading
+
+    from fma_amd.controller.workqueue import NodeQueueAndWorkers
+
+    barrier = threading.Barrier(2, timeout=5)
+    in_flight = {}
+    overlap = []
+    mu = threading.Lock()
+
+    def process(item):
+        node = item[0]
+        with mu:
+            if in_flight.get(node):
+                overlap.append(item)
+            in_flight[node] = True
+        try:
+            barrier.wait()  # needs BOTH nodes in flight at once
+        except threading.BrokenBarrierError:
+            pass
+        with mu:
+            in_flight[node] = False
+        return False
+
+    w = NodeQueueAndWorkers("t", 2, process, node_of=lambda it: it[0])
+    w.start()
+    try:
+        w.queue.add(("na", 1))
+        w.queue.add(("nb", 1))
+        import time as _t
+        deadline = _t.time() + 3
+        while _t.time() < deadline and len(w.queue.outer) + len(w.queue):
+            _t.sleep(0.02)
+        assert not overlap
+    finally:
+        w.stop()
+
+
+def test_store_index_lookup_matches_scan():
+    from fma_amd.api import contracts as C
+    from fma_amd.store import objects as ob
+    from fma_amd.store.indexes import install_pod_indexes
+    from fma_amd.store.memstore import MemStore
+
+    st = MemStore()
+    install_pod_indexes(st)
+    p1 = st.create(ob.new_object(
+        "Pod", "prov1",
+        annotations={C.REQUESTER_ANNOTATION: "u1 req1",
+                     C.ACCELERATORS_ANNOTATION: "GPU-0,GPU-1"},
+        spec={"nodeName": "node-a", "containers": []}))
+    st.create(ob.new_object(
+        "Pod", "req1", annotations={
+            C.INFERENCE_SERVER_CONFIG_ANNOTATION: "isc1"},
+        spec={"nodeName": "node-a", "containers": []}))
+
+    assert [ob.name_of(p) for p in
+            st.index_get("Pod", "requester", "u1 req1")] == ["prov1"]
+    assert [ob.name_of(p) for p in
+            st.index_get("Pod", "gpu", "GPU-1")] == ["prov1"]
+    assert [ob.name_of(p) for p in
+            st.index_get("Pod", "inferenceserverconfig", "isc1")] == ["req1"]
+
+    # update re-indexes: annotation change moves the pod between keys
+    ob.annotations_of(p1)[C.REQUESTER_ANNOTATION] = "u2 req2"
+    p1 = st.update(p1)
+    assert st.index_get("Pod", "requester", "u1 req1") == []
+    assert [ob.name_of(p) for p in
+            st.index_get("Pod", "requester", "u2 req2")] == ["prov1"]
+
+    # delete drops index entries
+    st.delete("Pod", "prov1")
+    assert st.index_get("Pod", "requester", "u2 req2") == []
+    assert st.index_get("Pod", "gpu", "GPU-0") == []
+
+
+def test_index_registered_late_backfills():
+    from fma_amd.api import contracts as C
+    from fma_amd.store import objects as ob
+    from fma_amd.store.memstore import MemStore
+
+    st = MemStore()
+    st.create(ob.new_object(
+        "Pod", "early", annotations={C.NOMINAL_ANNOTATION: "h1"},
+        spec={"containers": []}))
+    st.add_index("Pod", "nominal",
+                 lambda p: [ob.annotations_of(p).get(C.NOMINAL_ANNOTATION)]
+                 if ob.annotations_of(p).get(C.NOMINAL_ANNOTATION) else [])
+    assert [ob.name_of(p) for p in
+            st.index_get("Pod", "nominal", "h1")] == ["early"]
