@@ -48,6 +48,8 @@ def main() -> None:
     ap.add_argument("--nstreams", type=int, default=1)
     ap.add_argument("--slab-mb", type=int, default=0, help="slab size MiB (0=default 1 GiB)")
     ap.add_argument("--vmm", action="store_true", help="opt into VMM arena (unreliable on ROCm 7.2)")
+    ap.add_argument("--no-swap", action="store_true",
+                    help="skip the model-swap time-to-ready measurement")
     args = ap.parse_args()
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
@@ -60,11 +62,31 @@ def main() -> None:
     on_gpu = torch.cuda.is_available()
     dist_on = world > 1
     if dist_on:
+        from datetime import timedelta
+
         import torch.distributed as dist
         backend = "nccl" if on_gpu else "gloo"
+        # RCCL preflight: dmabuf IPC is the only mode the pool's driver
+        # supports; a hung rank should fail the job in minutes, not hang
+        # the box until the driver's limit kills it
+        os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+        os.environ.setdefault("NCCL_DEBUG", "WARN")
+        os.environ.setdefault("TORCH_NCCL_HEARTBEAT_TIMEOUT_SEC", "180")
         if on_gpu:
+            ngpu = torch.cuda.device_count()
+            if local_rank >= ngpu:
+                log(f"[rank {rank}] FATAL: LOCAL_RANK {local_rank} but only "
+                    f"{ngpu} visible GPUs")
+                sys.exit(2)
             torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+        try:
+            dist.init_process_group(backend=backend, rank=rank,
+                                    world_size=world,
+                                    timeout=timedelta(seconds=180))
+        except Exception as e:  # noqa: BLE001 - surface, don't hang
+            log(f"[rank {rank}] FATAL: init_process_group({backend}) "
+                f"failed: {e}")
+            sys.exit(2)
     n_gpus = world if dist_on else args.gpus
     assert n_gpus == world or world == 1, \
         f"--gpus {args.gpus} vs WORLD_SIZE {world} mismatch"
@@ -103,7 +125,7 @@ def main() -> None:
     # box self-evident in the result rather than looking like a
     # regression.
     h2d_gbps = alloc_gbps = None
-    if on_gpu and rank == 0:
+    if on_gpu:
         probe = min(1 << 30, eng.total_bytes or (1 << 30))
         host_buf = torch.empty(probe, dtype=torch.uint8, pin_memory=True)
         dev_buf = torch.empty(probe, dtype=torch.uint8, device="cuda")
@@ -121,10 +143,18 @@ def main() -> None:
         alloc_gbps = probe / (time.perf_counter() - t0p) / 1e9
         del dev_buf, host_buf
         torch.cuda.empty_cache()
-        log(f"[rank 0] box probe: pinned H2D {h2d_gbps:.1f} GB/s, "
+        log(f"[rank {rank}] box probe: pinned H2D {h2d_gbps:.1f} GB/s, "
             f"alloc+commit {alloc_gbps:.1f} GB/s"
             + ("  << DEGRADED HOST (healthy: ~56 GB/s H2D)"
                if h2d_gbps < 40 else ""))
+        if dist_on:
+            # report the WORST rank's link: wake is gated by the slowest
+            import torch.distributed as dist
+            dev = torch.device("cuda", local_rank)
+            worst = torch.tensor([h2d_gbps, alloc_gbps],
+                                 dtype=torch.float64, device=dev)
+            dist.all_reduce(worst, op=dist.ReduceOp.MIN)
+            h2d_gbps, alloc_gbps = worst.tolist()
 
     # warmup
     for i in range(args.warmup):
@@ -147,6 +177,34 @@ def main() -> None:
     sync()
     wall1 = time.perf_counter()
 
+    # time-to-ready after swap (the second half of the metric string):
+    # a second same-shaped model is parked asleep in pinned host DRAM;
+    # one swap = sleep(A) -> wake(B) -> B ready (the launcher model-swap,
+    # BASELINE config #3, measured at flagship size). Engine B is built
+    # after the timed region so its random-init cost stays out of both.
+    swap_ready_s = None
+    if not args.no_swap:
+        eng_b = ActuationEngine(
+            cfg, device_index=local_rank if on_gpu else 0,
+            tp_rank=rank, tp_size=world, tp_group=None,
+            use_vmm=args.vmm or None, chunk_bytes=args.chunk_mb << 20,
+            seed=4321, nstreams=args.nstreams,
+            slab_bytes=(args.slab_mb << 20) if args.slab_mb else None,
+            actuation_mode=args.mode,
+            pack_xfer_mode=None if args.pack_xfer < 0 else args.pack_xfer)
+        eng_b.sleep()  # park B in host DRAM
+        barrier()
+        sync()
+        t0s = time.perf_counter()
+        eng.sleep()
+        eng_b.wake_up()
+        barrier()
+        sync()
+        swap_ready_s = time.perf_counter() - t0s
+        log(f"[rank {rank}] swap A->B ready in {swap_ready_s:.3f}s")
+        eng_b.sleep()
+        eng.wake_up()  # restore A so repeated runs see identical state
+
     ms_per_step = (wall1 - wall0) / args.steps * 1000.0
     mean_wake = sum(wake_times) / len(wake_times)
     mean_sleep = sum(sleep_times) / len(sleep_times)
@@ -154,10 +212,13 @@ def main() -> None:
     if dist_on:
         import torch.distributed as dist
         dev = torch.device("cuda", local_rank) if on_gpu else torch.device("cpu")
-        agg = torch.tensor([ms_per_step, mean_wake, mean_sleep], dtype=torch.float64,
-                           device=dev)
+        agg = torch.tensor([ms_per_step, mean_wake, mean_sleep,
+                            swap_ready_s or 0.0],
+                           dtype=torch.float64, device=dev)
         dist.all_reduce(agg, op=dist.ReduceOp.MAX)
-        ms_per_step, mean_wake, mean_sleep = agg.tolist()
+        ms_per_step, mean_wake, mean_sleep, swap_max = agg.tolist()
+        if swap_ready_s is not None:
+            swap_ready_s = swap_max
 
     if rank == 0:
         baseline_s = 3.0  # reference: ~3 s wake for 64 GiB (README.md:24-25)
@@ -185,6 +246,8 @@ def main() -> None:
                 "mode": args.mode,
                 "vmm": bool(eng.stats()["uses_vmm"]),
                 "mean_sleep_s": round(mean_sleep, 4),
+                "swap_ready_s": round(swap_ready_s, 4)
+                if swap_ready_s is not None else None,
                 "h2d_gbps": round(h2d_gbps, 1) if h2d_gbps else None,
                 "alloc_gbps": round(alloc_gbps, 1) if alloc_gbps else None,
                 "global_batch": None,
