@@ -42,8 +42,10 @@ def main() -> None:
     ap.add_argument("--gib", type=float, default=64.0,
                     help="total parameter GiB across all ranks")
     ap.add_argument("--mode", choices=["arena", "pack"], default="arena")
-    ap.add_argument("--pack-xfer", type=int, default=0,
-                    help="pack transfer mode: 0 staged-kernel (default), 1 direct-kernel, 2 per-tensor, -1 auto")
+    ap.add_argument("--pack-xfer", type=int, default=-1,
+                    help="pack transfer mode: -1 auto (default; measured "
+                         "fastest per direction), 0 staged-kernel, "
+                         "1 direct-kernel, 2 per-tensor")
     ap.add_argument("--chunk-mb", type=int, default=0)
     ap.add_argument("--nstreams", type=int, default=1)
     ap.add_argument("--slab-mb", type=int, default=0, help="slab size MiB (0=default 1 GiB)")

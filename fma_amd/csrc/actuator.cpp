@@ -25,9 +25,12 @@
 
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
+#include <ATen/native/hip/Resize.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
+
+#include <condition_variable>
 
 #include <algorithm>
 #include <atomic>
@@ -65,6 +68,15 @@ struct DeviceCtx {
   hipStream_t kernel_stream = nullptr;
   hipStream_t copy_streams[kNumCopyStreams] = {nullptr, nullptr};
   hipEvent_t sync_event = nullptr;
+  // Persistent staging for the pack/restore pipelines: allocated on first
+  // use, reused across every sleep/wake cycle so the wake path never pays
+  // hipMalloc for staging. Calls on one device are serialized by the
+  // engine (one actuation at a time per GPU), so no further locking.
+  void* staging[2] = {nullptr, nullptr};
+  int64_t staging_bytes = 0;
+  // pinned bounce buffer for descriptor uploads (grown on demand)
+  void* pinned_descs = nullptr;
+  size_t pinned_descs_bytes = 0;
 
   void init(int dev) {
     device = dev;
@@ -74,6 +86,34 @@ struct DeviceCtx {
       FMA_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
     }
     FMA_HIP_CHECK(hipEventCreateWithFlags(&sync_event, hipEventDisableTiming));
+  }
+
+  void ensure_staging(int64_t chunk) {
+    if (staging_bytes >= chunk && staging[0]) return;
+    for (auto& s : staging) {
+      if (s) (void)hipFree(s);
+      FMA_HIP_CHECK(hipMalloc(&s, chunk));
+    }
+    staging_bytes = chunk;
+  }
+
+  // Release the staging buffers (sleep must hand ALL HBM back; 2 chunks
+  // of staging on a parked GPU would undercut the memory budget).
+  void release_staging() {
+    for (auto& s : staging) {
+      if (s) (void)hipFree(s);
+      s = nullptr;
+    }
+    staging_bytes = 0;
+  }
+
+  unsigned char* ensure_pinned_descs(size_t bytes) {
+    if (pinned_descs_bytes < bytes) {
+      if (pinned_descs) (void)hipHostFree(pinned_descs);
+      FMA_HIP_CHECK(hipHostMalloc(&pinned_descs, bytes, hipHostMallocDefault));
+      pinned_descs_bytes = bytes;
+    }
+    return static_cast<unsigned char*>(pinned_descs);
   }
 };
 
@@ -276,7 +316,12 @@ std::vector<FmaCopyDesc> retarget(const ChunkPlan& plan, int64_t c,
 
 double pack_to_host(const std::vector<at::Tensor>& tensors,
                     const std::vector<int64_t>& offsets, at::Tensor host,
-                    int64_t mode_i, int64_t chunk_bytes) {
+                    int64_t mode_i, int64_t chunk_bytes,
+                    int64_t nstreams = 1) {
+  // one SDMA stream by default: concurrent H2D/D2H on multiple copy
+  // streams measurably UNDERPERFORMS a single engine on MI355X (43 vs 54
+  // GiB/s staged, tools/pack_probe.py) — the link is serial anyway
+  const int ns = std::max<int>(1, std::min<int64_t>(nstreams, kNumCopyStreams));
   validate_tensors(tensors, offsets);
   const int device = tensors[0].device().index();
   auto& ctx = ctx_for(device);
@@ -296,7 +341,7 @@ double pack_to_host(const std::vector<at::Tensor>& tensors,
       if (!bytes) continue;
       FMA_HIP_CHECK(hipMemcpyAsync(host_ptr + offsets[i], tensors[i].data_ptr(),
                                    bytes, hipMemcpyDeviceToHost,
-                                   ctx.copy_streams[i % kNumCopyStreams]));
+                                   ctx.copy_streams[i % ns]));
     }
     sync_pipeline(ctx);
     return seconds_since(t0);
@@ -321,9 +366,8 @@ double pack_to_host(const std::vector<at::Tensor>& tensors,
 
   // Staged pipeline.
   ChunkPlan plan = build_chunk_plan(tensors, offsets, chunk, /*pack=*/true);
-  void* staging[2] = {nullptr, nullptr};
-  FMA_HIP_CHECK(hipMalloc(&staging[0], chunk));
-  FMA_HIP_CHECK(hipMalloc(&staging[1], chunk));
+  ctx.ensure_staging(chunk);
+  void* const* staging = ctx.staging;
   hipEvent_t copied[2], packed[2];
   for (int b = 0; b < 2; ++b) {
     FMA_HIP_CHECK(hipEventCreateWithFlags(&copied[b], hipEventDisableTiming));
@@ -354,7 +398,7 @@ double pack_to_host(const std::vector<at::Tensor>& tensors,
         dev.descs + launch_desc_lo[c], dev.prefix + plan.prefix_lo[c],
         static_cast<int>(launch_ndesc[c]), plan.units[c], ctx.kernel_stream));
     FMA_HIP_CHECK(hipEventRecord(packed[b], ctx.kernel_stream));
-    auto cs = ctx.copy_streams[b];
+    auto cs = ctx.copy_streams[b % ns];
     FMA_HIP_CHECK(hipStreamWaitEvent(cs, packed[b], 0));
     const int64_t lo = c * chunk;
     const int64_t sz = std::min<int64_t>(chunk, plan.total_bytes - lo);
@@ -366,14 +410,18 @@ double pack_to_host(const std::vector<at::Tensor>& tensors,
   for (int b = 0; b < 2; ++b) {
     (void)hipEventDestroy(copied[b]);
     (void)hipEventDestroy(packed[b]);
-    (void)hipFree(staging[b]);
   }
+  // sleep hands all HBM back: staging would undercut the sleeping-memory
+  // budget the controller enforces
+  ctx.release_staging();
   return seconds_since(t0);
 }
 
 double restore_from_host(const std::vector<at::Tensor>& tensors,
                          const std::vector<int64_t>& offsets, at::Tensor host,
-                         int64_t mode_i, int64_t chunk_bytes) {
+                         int64_t mode_i, int64_t chunk_bytes,
+                         int64_t nstreams = 1) {
+  const int ns = std::max<int>(1, std::min<int64_t>(nstreams, kNumCopyStreams));
   validate_tensors(tensors, offsets);
   const int device = tensors[0].device().index();
   auto& ctx = ctx_for(device);
@@ -393,7 +441,7 @@ double restore_from_host(const std::vector<at::Tensor>& tensors,
       if (!bytes) continue;
       FMA_HIP_CHECK(hipMemcpyAsync(tensors[i].data_ptr(), host_ptr + offsets[i],
                                    bytes, hipMemcpyHostToDevice,
-                                   ctx.copy_streams[i % kNumCopyStreams]));
+                                   ctx.copy_streams[i % ns]));
     }
     sync_pipeline(ctx);
     return seconds_since(t0);
@@ -417,9 +465,8 @@ double restore_from_host(const std::vector<at::Tensor>& tensors,
 
   // Staged pipeline: H2D chunk -> scatter kernel, double-buffered.
   ChunkPlan plan = build_chunk_plan(tensors, offsets, chunk, /*pack=*/false);
-  void* staging[2] = {nullptr, nullptr};
-  FMA_HIP_CHECK(hipMalloc(&staging[0], chunk));
-  FMA_HIP_CHECK(hipMalloc(&staging[1], chunk));
+  ctx.ensure_staging(chunk);
+  void* const* staging = ctx.staging;
   hipEvent_t arrived[2], scattered[2];
   for (int b = 0; b < 2; ++b) {
     FMA_HIP_CHECK(hipEventCreateWithFlags(&arrived[b], hipEventDisableTiming));
@@ -439,7 +486,7 @@ double restore_from_host(const std::vector<at::Tensor>& tensors,
 
   for (int64_t c = 0; c < plan.nchunks; ++c) {
     const int b = static_cast<int>(c & 1);
-    auto cs = ctx.copy_streams[b];
+    auto cs = ctx.copy_streams[b % ns];
     if (c >= 2) {
       FMA_HIP_CHECK(hipStreamWaitEvent(cs, scattered[b], 0));
     }
@@ -458,8 +505,235 @@ double restore_from_host(const std::vector<at::Tensor>& tensors,
   for (int b = 0; b < 2; ++b) {
     (void)hipEventDestroy(arrived[b]);
     (void)hipEventDestroy(scattered[b]);
-    (void)hipFree(staging[b]);
   }
+  return seconds_since(t0);
+}
+
+// restore_from_host with storage re-allocation overlapped into the H2D
+// pipeline. The plain wake path pays the full caching-allocator re-commit
+// of every storage BEFORE the first byte moves (≈0.3 s serial on a 64 GiB
+// model); here a background thread resizes storages in flat-offset order
+// while chunks stream, so wake ≈ max(alloc rate, PCIe link rate) — the
+// same overlap the arena path gets from its threaded slab allocation.
+//
+// Descriptor pointers are only known after a tensor's storage exists, so
+// the per-chunk descriptor tables are built lazily: every chunk has its
+// own slice of one device blob, uploaded (via a persistent pinned bounce
+// buffer) on the kernel stream right before its scatter launch.
+double restore_from_host_overlapped(std::vector<at::Tensor> tensors,
+                                    const std::vector<int64_t>& offsets,
+                                    at::Tensor host, int64_t mode_i,
+                                    int64_t chunk_bytes,
+                                    int64_t nstreams = 1) {
+  const int ns = std::max<int>(1, std::min<int64_t>(nstreams, kNumCopyStreams));
+  TORCH_CHECK(tensors.size() == offsets.size() && !tensors.empty(),
+              "tensors/offsets mismatch or empty");
+  const auto dev0 = tensors[0].device();
+  TORCH_CHECK(dev0.is_cuda(), "tensors must live on the GPU");
+  const size_t n = tensors.size();
+  std::vector<size_t> need(n);
+  for (size_t i = 0; i < n; ++i) {
+    const auto& t = tensors[i];
+    TORCH_CHECK(t.device() == dev0, "tensor ", i, " on a different device");
+    TORCH_CHECK(t.is_contiguous(), "tensor ", i, " must be contiguous");
+    TORCH_CHECK((offsets[i] & (FMA_ARENA_ALIGN - 1)) == 0,
+                "offset ", i, " not aligned");
+    TORCH_CHECK(i == 0 || offsets[i] >= offsets[i - 1],
+                "offsets must be non-decreasing");
+    need[i] = t.nbytes();
+  }
+  const int device = dev0.index();
+  auto& ctx = ctx_for(device);
+  FMA_HIP_CHECK(hipSetDevice(device));
+  const auto mode = static_cast<XferMode>(mode_i);
+  const int64_t chunk = chunk_or_default(chunk_bytes);
+  int64_t total = 0;
+  for (size_t i = 0; i < n; ++i) {
+    total = std::max(total, offsets[i] + static_cast<int64_t>(need[i]));
+  }
+  check_host_buffer(host, total);
+  auto* host_ptr = static_cast<unsigned char*>(host.data_ptr());
+
+  const auto t0 = Clock::now();
+  join_torch_stream(ctx);
+  if (mode == XferMode::kStaged) {
+    ctx.ensure_staging(chunk);
+  }
+
+  // ---- background allocator: storages materialize in flat order -------
+  std::mutex mu;
+  std::condition_variable cv;
+  long ready = -1;  // highest tensor index whose storage is committed
+  bool failed = false;
+  std::string fail_msg;
+  std::thread alloc_thread([&] {
+    if (hipSetDevice(device) != hipSuccess) {
+      std::lock_guard<std::mutex> lk(mu);
+      failed = true;
+      fail_msg = "hipSetDevice failed in alloc thread";
+      cv.notify_all();
+      return;
+    }
+    for (size_t i = 0; i < n; ++i) {
+      try {
+        auto* si = tensors[i].storage().unsafeGetStorageImpl();
+        if (si->nbytes() < need[i]) {
+          at::native::resize_bytes_cuda(si, need[i]);
+        }
+      } catch (const std::exception& e) {
+        std::lock_guard<std::mutex> lk(mu);
+        failed = true;
+        fail_msg = e.what();
+        cv.notify_all();
+        return;
+      }
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        ready = static_cast<long>(i);
+      }
+      cv.notify_all();
+    }
+  });
+  auto wait_ready = [&](long idx) {
+    std::unique_lock<std::mutex> lk(mu);
+    cv.wait(lk, [&] { return failed || ready >= idx; });
+    return !failed;
+  };
+  auto fail_out = [&]() {
+    alloc_thread.join();
+    (void)hipDeviceSynchronize();
+    TORCH_CHECK(false, "storage allocation failed during overlapped wake: ",
+                fail_msg);
+    return 0.0;  // unreachable
+  };
+
+  if (mode == XferMode::kPerTensor) {
+    for (size_t i = 0; i < n; ++i) {
+      if (!need[i]) continue;
+      if (!wait_ready(static_cast<long>(i))) return fail_out();
+      FMA_HIP_CHECK(hipMemcpyAsync(tensors[i].data_ptr(),
+                                   host_ptr + offsets[i], need[i],
+                                   hipMemcpyHostToDevice,
+                                   ctx.copy_streams[i % ns]));
+    }
+    sync_pipeline(ctx);
+    alloc_thread.join();
+    return seconds_since(t0);
+  }
+  TORCH_CHECK(mode == XferMode::kStaged,
+              "overlapped restore supports staged/per-tensor modes");
+
+  // ---- per-chunk symbolic descriptor skeletons -------------------------
+  struct SymDesc {
+    int ti;                        // tensor index
+    unsigned long long t_off;      // byte offset within the tensor
+    unsigned long long c_off;      // byte offset within the chunk
+    unsigned long long bytes;
+  };
+  const int64_t nchunks = (total + chunk - 1) / chunk;
+  std::vector<std::vector<SymDesc>> sym(nchunks);
+  std::vector<long> max_ti(nchunks, -1);
+  for (size_t i = 0; i < n; ++i) {
+    int64_t rem = static_cast<int64_t>(need[i]);
+    if (!rem) continue;
+    int64_t pos = offsets[i];
+    while (rem > 0) {
+      const int64_t c = pos / chunk;
+      const int64_t in_chunk = pos - c * chunk;
+      const int64_t take = std::min(rem, chunk - in_chunk);
+      sym[c].push_back({static_cast<int>(i),
+                        static_cast<unsigned long long>(need[i] - rem),
+                        static_cast<unsigned long long>(in_chunk),
+                        static_cast<unsigned long long>(take)});
+      max_ti[c] = std::max(max_ti[c], static_cast<long>(i));
+      pos += take;
+      rem -= take;
+    }
+  }
+  for (int64_t c = 0; c < nchunks; ++c) {
+    TORCH_CHECK(sym[c].size() <= FMA_MAX_DESCS_PER_LAUNCH,
+                "too many descriptors in one chunk; raise chunk_bytes");
+  }
+  // prefix sums (byte counts only — pointer-independent) in one blob
+  std::vector<unsigned long long> prefix_h;
+  std::vector<int64_t> prefix_lo(nchunks), desc_lo(nchunks);
+  std::vector<unsigned long long> units(nchunks, 0);
+  int64_t total_descs = 0;
+  for (int64_t c = 0; c < nchunks; ++c) {
+    desc_lo[c] = total_descs;
+    prefix_lo[c] = static_cast<int64_t>(prefix_h.size());
+    unsigned long long u = 0;
+    prefix_h.push_back(0);
+    for (const auto& d : sym[c]) {
+      u += (d.bytes + 15ull) >> 4;
+      prefix_h.push_back(u);
+    }
+    units[c] = u;
+    total_descs += static_cast<int64_t>(sym[c].size());
+  }
+  // device blob: all chunks' descs + the full prefix array
+  const size_t db = static_cast<size_t>(total_descs) * sizeof(FmaCopyDesc);
+  DevBlob dev;
+  dev.upload({}, prefix_h, ctx.kernel_stream);  // placeholder — see below
+  // DevBlob::upload sized for empty descs; allocate our own desc area:
+  FmaCopyDesc* descs_dev = nullptr;
+  FMA_HIP_CHECK(hipMalloc(&descs_dev, std::max<size_t>(db, 16)));
+  auto* pinned =
+      reinterpret_cast<FmaCopyDesc*>(ctx.ensure_pinned_descs(db ? db : 16));
+
+  hipEvent_t arrived[2], scattered[2];
+  for (int b = 0; b < 2; ++b) {
+    FMA_HIP_CHECK(hipEventCreateWithFlags(&arrived[b], hipEventDisableTiming));
+    FMA_HIP_CHECK(hipEventCreateWithFlags(&scattered[b], hipEventDisableTiming));
+  }
+  void* const* staging = ctx.staging;
+
+  bool bail = false;
+  for (int64_t c = 0; c < nchunks && !bail; ++c) {
+    const int b = static_cast<int>(c & 1);
+    auto cs = ctx.copy_streams[b % ns];
+    // storages for every tensor in this chunk must exist before we can
+    // resolve descriptor pointers (and before the scatter writes them)
+    if (max_ti[c] >= 0 && !wait_ready(max_ti[c])) {
+      bail = true;
+      break;
+    }
+    // fill this chunk's descriptor slice with concrete pointers
+    for (size_t j = 0; j < sym[c].size(); ++j) {
+      const auto& d = sym[c][j];
+      pinned[desc_lo[c] + j] = FmaCopyDesc{
+          static_cast<unsigned char*>(staging[b]) + d.c_off,
+          static_cast<unsigned char*>(tensors[d.ti].data_ptr()) + d.t_off,
+          d.bytes};
+    }
+    if (!sym[c].empty()) {
+      FMA_HIP_CHECK(hipMemcpyAsync(descs_dev + desc_lo[c],
+                                   pinned + desc_lo[c],
+                                   sym[c].size() * sizeof(FmaCopyDesc),
+                                   hipMemcpyHostToDevice, ctx.kernel_stream));
+    }
+    if (c >= 2) {
+      FMA_HIP_CHECK(hipStreamWaitEvent(cs, scattered[b], 0));
+    }
+    const int64_t lo = c * chunk;
+    const int64_t sz = std::min<int64_t>(chunk, total - lo);
+    FMA_HIP_CHECK(hipMemcpyAsync(staging[b], host_ptr + lo, sz,
+                                 hipMemcpyHostToDevice, cs));
+    FMA_HIP_CHECK(hipEventRecord(arrived[b], cs));
+    FMA_HIP_CHECK(hipStreamWaitEvent(ctx.kernel_stream, arrived[b], 0));
+    FMA_HIP_CHECK(fma_launch_batched_copy(
+        descs_dev + desc_lo[c], dev.prefix + prefix_lo[c],
+        static_cast<int>(sym[c].size()), units[c], ctx.kernel_stream));
+    FMA_HIP_CHECK(hipEventRecord(scattered[b], ctx.kernel_stream));
+  }
+  sync_pipeline(ctx);
+  for (int b = 0; b < 2; ++b) {
+    (void)hipEventDestroy(arrived[b]);
+    (void)hipEventDestroy(scattered[b]);
+  }
+  (void)hipFree(descs_dev);
+  if (bail) return fail_out();
+  alloc_thread.join();
   return seconds_since(t0);
 }
 
@@ -1061,12 +1335,24 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_to_host", &pack_to_host,
         "Gather scattered device tensors into a pinned host buffer",
         py::arg("tensors"), py::arg("offsets"), py::arg("host"),
-        py::arg("mode") = 0, py::arg("chunk_bytes") = 0);
+        py::arg("mode") = 0, py::arg("chunk_bytes") = 0,
+        py::arg("nstreams") = 1);
   m.def("restore_from_host", &restore_from_host,
         "Scatter a pinned host buffer back into device tensors",
         py::arg("tensors"), py::arg("offsets"), py::arg("host"),
-        py::arg("mode") = 0, py::arg("chunk_bytes") = 0);
+        py::arg("mode") = 0, py::arg("chunk_bytes") = 0,
+        py::arg("nstreams") = 1);
+  m.def("restore_from_host_overlapped", &restore_from_host_overlapped,
+        "restore_from_host that re-allocates the (released) storages on a "
+        "background thread, overlapped with the H2D pipeline",
+        py::arg("tensors"), py::arg("offsets"), py::arg("host"),
+        py::arg("mode") = 0, py::arg("chunk_bytes") = 0,
+        py::arg("nstreams") = 1,
+        py::call_guard<py::gil_scoped_release>());
   m.def("device_supports_vmm", &device_supports_vmm, py::arg("device"));
+  m.def("_release_staging", [](int device) {
+    ctx_for(device).release_staging();
+  }, py::arg("device"), "debug: drop the persistent staging buffers");
   m.def("gemv_bf16", &gemv_bf16, "Batch-1 bf16 GEMV",
         py::arg("W"), py::arg("x"), py::arg("out_bf16") = false,
         py::arg("residual") = py::none());

@@ -244,14 +244,20 @@ class PackActuator:
         require_native()
         self.names: List[str] = sorted(tensors.keys())
         self.tensors = tensors
+        mean = (sum(t.nbytes for t in tensors.values())
+                / max(len(tensors), 1))
         if mode is None:
-            # measured on MI355X (16 GiB model): per-tensor hipMemcpyAsync
-            # wins when shards are large (0.299 s vs 0.354 s staged); the
-            # gather kernel pays off for many small tensors, where
-            # per-copy overhead would dominate
-            mean = (sum(t.nbytes for t in tensors.values())
-                    / max(len(tensors), 1))
+            # measured on MI355X (tools/cycle_probe.py): the gather kernel
+            # pays off for many small tensors (per-copy SDMA overhead would
+            # dominate); for large shards per-tensor SDMA wins — and for
+            # SLEEP specifically, re-allocated caching-allocator memory
+            # reads back through the kernel at ~34 GiB/s (TLB-unfriendly
+            # page granularity after empty_cache) while SDMA reads stay at
+            # ~52 GiB/s, so sleep prefers SDMA once shards are big enough
             mode = MODE_PER_TENSOR if mean >= (16 << 20) else MODE_STAGED
+            self.sleep_mode = mode
+        else:
+            self.sleep_mode = mode
         self.mode = mode
         self.chunk_bytes = chunk_bytes
         off = 0
@@ -268,7 +274,7 @@ class PackActuator:
     def sleep(self, host: torch.Tensor) -> float:
         C = require_native()
         t = C.pack_to_host(self._tensor_list(), self.offsets, host,
-                           self.mode, self.chunk_bytes)
+                           self.sleep_mode, self.chunk_bytes)
         for n in self.names:
             self.tensors[n].untyped_storage().resize_(0)
         torch.cuda.empty_cache()
@@ -277,11 +283,20 @@ class PackActuator:
 
     def wake(self, host: torch.Tensor) -> float:
         C = require_native()
-        for n in self.names:
-            t = self.tensors[n]
-            t.untyped_storage().resize_(t.numel() * t.element_size())
-        t = C.restore_from_host(self._tensor_list(), self.offsets, host,
-                                self.mode, self.chunk_bytes)
+        if self.mode in (MODE_STAGED, MODE_PER_TENSOR) and \
+                hasattr(C, "restore_from_host_overlapped"):
+            # storage re-allocation happens on a background thread inside
+            # the call, overlapped with the H2D pipeline (the up-front
+            # resize loop cost ~0.3 s serial on a 64 GiB model)
+            t = C.restore_from_host_overlapped(
+                self._tensor_list(), self.offsets, host, self.mode,
+                self.chunk_bytes)
+        else:
+            for n in self.names:
+                tt = self.tensors[n]
+                tt.untyped_storage().resize_(tt.numel() * tt.element_size())
+            t = C.restore_from_host(self._tensor_list(), self.offsets, host,
+                                    self.mode, self.chunk_bytes)
         self.asleep = False
         return t
 

@@ -525,3 +525,49 @@ def test_engine_graph_decode_serving_path():
     again = eng.generate(toks, 6)
     assert torch.equal(again, eager), "post-wake graph decode diverged"
     assert eng.stats()["graph_decode"] is True
+
+
+@pytest.mark.parametrize("mode", [0, 2], ids=["staged", "per_tensor"])
+def test_restore_overlapped_matches_reference(C, mode):
+    """restore_from_host_overlapped re-allocates RELEASED storages on a
+    background thread while chunks stream; result must be bit-exact and
+    every storage fully re-committed."""
+    from fma_amd.ops.actuation import align_up
+    ts = scattered_tensors(seed=11)
+    originals = [t.clone() for t in ts]
+    offsets, off = [], 0
+    for t in ts:
+        offsets.append(off)
+        off += align_up(max(t.nbytes, 1))
+    host = torch.empty(off, dtype=torch.uint8, pin_memory=True)
+    C.pack_to_host(ts, offsets, host, 0, 1 << 20)
+    for t in ts:
+        t.untyped_storage().resize_(0)   # truly released, as after sleep
+    torch.cuda.empty_cache()
+    C.restore_from_host_overlapped(ts, offsets, host, mode, 1 << 20)
+    torch.cuda.synchronize()
+    for t, o in zip(ts, originals):
+        assert t.untyped_storage().nbytes() == t.numel() * t.element_size()
+        assert torch.equal(t, o)
+
+
+def test_restore_overlapped_many_cycles(C):
+    """Repeated sleep/overlapped-wake cycles stay bit-exact (staging and
+    pinned descriptor buffers are persistent and reused)."""
+    from fma_amd.ops.actuation import align_up
+    ts = scattered_tensors(seed=13)
+    originals = [t.clone() for t in ts]
+    offsets, off = [], 0
+    for t in ts:
+        offsets.append(off)
+        off += align_up(max(t.nbytes, 1))
+    host = torch.empty(off, dtype=torch.uint8, pin_memory=True)
+    for cycle in range(4):
+        C.pack_to_host(ts, offsets, host, 0, 1 << 20)
+        for t in ts:
+            t.untyped_storage().resize_(0)
+        torch.cuda.empty_cache()
+        C.restore_from_host_overlapped(ts, offsets, host, 0, 1 << 20)
+        torch.cuda.synchronize()
+        for t, o in zip(ts, originals):
+            assert torch.equal(t, o), f"cycle {cycle} corrupt"
