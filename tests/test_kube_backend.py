@@ -1,0 +1,248 @@
+"""Kubernetes backend: wire-protocol, CEL admission, watch semantics.
+
+The FMA stack's deployment story on a real cluster: KubeStore speaks the
+Kubernetes REST protocol; here it runs against the in-tree apiserver
+double (fma_amd/store/kubeapiserver.py), which evaluates the SHIPPED
+ValidatingAdmissionPolicy YAML (manifests/validating-admission-policies/)
+through the CEL evaluator — the analog of the reference's kind e2e + CEL
+policy checks (reference test/e2e/test-cases.sh:317,
+config/validating-admission-policies/*.yaml).
+"""
+
+import threading
+import time
+
+import httpx
+import pytest
+
+from fma_amd.api import contracts as C
+from fma_amd.store import objects as ob
+from fma_amd.store.kubeapiserver import create_app
+from fma_amd.store.kubestore import KubeStore
+from fma_amd.store.memstore import (AlreadyExists, Conflict, Invalid,
+                                    MemStore, NotFound, RevisionTooOld)
+
+from tests.test_live_servers import ServerThread, free_port
+
+pytestmark = pytest.mark.timeout(120)
+
+
+@pytest.fixture()
+def kube():
+    store = MemStore()
+    port = free_port()
+    with ServerThread(create_app(store), port):
+        base = f"http://127.0.0.1:{port}"
+        yield {
+            "base": base,
+            "mem": store,
+            "user": KubeStore(base, actor="user"),
+            "ctl": KubeStore(base, actor="dual-pods-controller"),
+        }
+
+
+def mk_pod(name, annotations=None, labels=None):
+    return ob.new_object("Pod", name, annotations=annotations or {},
+                         labels=labels or {},
+                         spec={"nodeName": "node-a", "containers": []})
+
+
+# ---------------------------------------------------------------------------
+# wire protocol
+# ---------------------------------------------------------------------------
+
+
+def test_crud_roundtrip_core_and_crd(kube):
+    ks = kube["user"]
+    pod = ks.create(mk_pod("p1", labels={"a": "b"}))
+    assert ob.uid_of(pod) and ob.rv_of(pod)
+    assert ks.get("Pod", "p1")["metadata"]["labels"] == {"a": "b"}
+    with pytest.raises(AlreadyExists):
+        ks.create(mk_pod("p1"))
+
+    isc = ks.create(ob.new_object(
+        "InferenceServerConfig", "isc1",
+        spec={"modelServerConfig": {"port": 8000},
+              "launcherConfigName": "lc1"}))
+    assert isc["spec"]["launcherConfigName"] == "lc1"
+    # raw path sanity: the CRD lives under the k8s group path
+    r = httpx.get(kube["base"] + "/apis/fma.llm-d.ai/v1alpha1/namespaces/"
+                  "default/inferenceserverconfigs/isc1")
+    assert r.status_code == 200
+
+    node = ks.create(ob.new_object("Node", "node-a"))
+    r = httpx.get(kube["base"] + "/api/v1/nodes/node-a")
+    assert r.status_code == 200 and ob.uid_of(r.json()) == ob.uid_of(node)
+
+    ks.delete("Pod", "p1")
+    with pytest.raises(NotFound):
+        ks.get("Pod", "p1")
+
+
+def test_optimistic_concurrency_conflict(kube):
+    ks = kube["user"]
+    pod = ks.create(mk_pod("p2"))
+    stale_rv = ob.rv_of(pod)
+    pod2 = ks.get("Pod", "p2")
+    pod2["metadata"]["labels"] = {"x": "1"}
+    ks.update(pod2)
+    # stale update loses
+    pod["metadata"]["labels"] = {"x": "2"}
+    with pytest.raises(Conflict):
+        ks.update(pod, expect_rv=stale_rv)
+
+
+def test_delete_preconditions(kube):
+    ks = kube["user"]
+    pod = ks.create(mk_pod("p3"))
+    with pytest.raises(Conflict):
+        ks.delete("Pod", "p3", expect_uid="wrong-uid")
+    ks.delete("Pod", "p3", expect_uid=ob.uid_of(pod))
+    assert ks.try_get("Pod", "p3") is None
+
+
+def test_finalizer_defers_deletion(kube):
+    ks, ctl = kube["user"], kube["ctl"]
+    pod = mk_pod("p4")
+    pod["metadata"]["finalizers"] = ["dual-pods.llm-d.ai/test"]
+    pod = ks.create(pod)
+    ks.delete("Pod", "p4")
+    cur = ks.get("Pod", "p4")
+    assert ob.is_deleting(cur)
+    cur["metadata"]["finalizers"] = []
+    ctl.update(cur)
+    assert ks.try_get("Pod", "p4") is None
+
+
+def test_label_selector_list(kube):
+    ks = kube["user"]
+    ks.create(mk_pod("l1", labels={C.COMPONENT_LABEL: C.LAUNCHER_COMPONENT}))
+    ks.create(mk_pod("l2", labels={C.COMPONENT_LABEL: "other"}))
+    got = ks.list("Pod", label_selector={
+        C.COMPONENT_LABEL: C.LAUNCHER_COMPONENT})
+    assert [ob.name_of(p) for p in got] == ["l1"]
+
+
+def test_status_subresource(kube):
+    ks = kube["user"]
+    pod = ks.create(mk_pod("p5"))
+    pod["status"] = {"phase": "Running", "podIP": "10.0.0.9"}
+    updated = ks.update(pod, subresource="status")
+    assert updated["status"]["podIP"] == "10.0.0.9"
+    # generation untouched by status writes
+    assert updated["metadata"]["generation"] == \
+        pod["metadata"]["generation"]
+
+
+# ---------------------------------------------------------------------------
+# shipped CEL admission policies
+# ---------------------------------------------------------------------------
+
+
+def test_vap_denies_user_mutation_of_fma_metadata(kube):
+    """The deny rule comes from the SHIPPED YAML artifact, evaluated by
+    the CEL engine — not a parallel Python re-implementation."""
+    ks, ctl = kube["user"], kube["ctl"]
+    pod = ctl.create(mk_pod(
+        "prov", annotations={C.REQUESTER_ANNOTATION: "u1 req1"},
+        labels={C.DUAL_LABEL: "req1"}))
+
+    hacked = ob.deepcopy(pod)
+    hacked["metadata"]["annotations"][C.REQUESTER_ANNOTATION] = "u2 evil"
+    with pytest.raises(Invalid) as ei:
+        ks.update(hacked)
+    assert "fma-immutable-fields" in str(ei.value)
+
+    # the controller service account may do the same mutation
+    cur = ctl.get("Pod", "prov")
+    cur["metadata"]["annotations"][C.REQUESTER_ANNOTATION] = "u2 req2"
+    ctl.update(cur)
+
+    # non-protected metadata stays freely editable by users
+    cur = ks.get("Pod", "prov")
+    cur["metadata"]["annotations"]["my-note"] = "hello"
+    ks.update(cur)
+
+
+def test_vap_denies_isc_switch_on_bound_requester(kube):
+    ks, ctl = kube["user"], kube["ctl"]
+    ks.create(mk_pod(
+        "req", annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: "isc-a"}))
+    # unbound: switching the ISC is allowed
+    cur = ks.get("Pod", "req")
+    cur["metadata"]["annotations"][C.INFERENCE_SERVER_CONFIG_ANNOTATION] = \
+        "isc-b"
+    ks.update(cur)
+    # bind it (as the controller would)
+    cur = ctl.get("Pod", "req")
+    cur["metadata"]["labels"][C.DUAL_LABEL] = "launcher1"
+    ctl.update(cur)
+    # bound: switching is denied for users
+    cur = ks.get("Pod", "req")
+    cur["metadata"]["annotations"][C.INFERENCE_SERVER_CONFIG_ANNOTATION] = \
+        "isc-c"
+    with pytest.raises(Invalid) as ei:
+        ks.update(cur)
+    assert "fma-bound-serverreqpod" in str(ei.value)
+
+
+# ---------------------------------------------------------------------------
+# watch
+# ---------------------------------------------------------------------------
+
+
+def test_watch_multiplexes_kinds_and_resumes(kube):
+    ks = kube["user"]
+    token = ks.list_revision()
+    got = []
+    stop = threading.Event()
+
+    def consume():
+        for ev in ks.watch(since=token, stop=stop):
+            got.append((ev.type, ev.kind, ob.name_of(ev.obj)))
+            if len(got) >= 3:
+                return
+
+    th = threading.Thread(target=consume, daemon=True)
+    th.start()
+    time.sleep(0.3)
+    ks.create(mk_pod("w1"))
+    ks.create(ob.new_object("LauncherConfig", "lcw",
+                            spec={"maxInstances": 1, "podTemplate": {}}))
+    ks.delete("Pod", "w1")
+    th.join(timeout=20)
+    assert not th.is_alive(), f"watch stalled; got {got}"
+    kinds = {(t, k, n) for t, k, n in got}
+    assert ("ADDED", "Pod", "w1") in kinds
+    assert ("ADDED", "LauncherConfig", "lcw") in kinds
+    assert ("DELETED", "Pod", "w1") in kinds
+    stop.set()
+
+
+def test_watch_unknown_token_raises_revision_too_old(kube):
+    ks = kube["user"]
+    with pytest.raises(RevisionTooOld):
+        for _ in ks.watch(since=999999):
+            pass
+
+
+def test_watch_only_requested_kinds(kube):
+    ks = kube["user"]
+    token = ks.list_revision()
+    stop = threading.Event()
+    got = []
+
+    def consume():
+        for ev in ks.watch(since=token, kinds=["Pod"], stop=stop):
+            got.append((ev.kind, ob.name_of(ev.obj)))
+            if len(got) >= 1:
+                return
+
+    th = threading.Thread(target=consume, daemon=True)
+    th.start()
+    time.sleep(0.3)
+    ks.create(ob.new_object("ConfigMap", "cm-ignored"))
+    ks.create(mk_pod("w2"))
+    th.join(timeout=20)
+    assert got == [("Pod", "w2")]
+    stop.set()

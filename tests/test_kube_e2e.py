@@ -1,0 +1,152 @@
+"""Full-stack e2e over the KUBERNETES wire protocol.
+
+Same real-process actuation flow as tests/test_e2e_single_node.py, but
+every component (both controllers + the node agent) talks to the
+apiserver double through KubeStore — i.e., through the Kubernetes REST
+protocol with the shipped CEL admission policies enforced. This is the
+analog of the reference's kind e2e (reference
+test/e2e/run-launcher-based.sh driving test-cases.sh): scenarios covered
+here map to launcher-based pod creation (:260), wake fast path (:465),
+controller restart recovery (:720) and deletion/unbinding (:836).
+"""
+
+import os
+import sys
+import time
+
+import httpx
+import pytest
+
+from fma_amd.api import contracts as C
+from fma_amd.controller.dualpods.controller import (ControllerConfig,
+                                                    DualPodsController)
+from fma_amd.controller.httpadapter import HttpAdapter
+from fma_amd.controller.populator.populator import LauncherPopulator
+from fma_amd.node.agent import NodeAgent
+from fma_amd.store import objects as ob
+from fma_amd.store.kubeapiserver import create_app
+from fma_amd.store.kubestore import KubeStore
+from fma_amd.store.memstore import Invalid, MemStore
+
+from tests.test_e2e_single_node import (ISC_PORT, launcher_pod,
+                                        mk_isc_lc_lpp, mk_requester,
+                                        requester_ready, wait_for)
+from tests.test_live_servers import ServerThread, free_port
+
+pytestmark = pytest.mark.timeout(300)
+
+
+@pytest.fixture()
+def kube_cluster(tmp_path):
+    mem = MemStore()
+    port = free_port()
+    with ServerThread(create_app(mem), port):
+        base = f"http://127.0.0.1:{port}"
+        admin = KubeStore(base, actor="dual-pods-controller")
+        node = ob.new_object("Node", "node-a", labels={"gpu": "mi355x"})
+        node["status"] = {"allocatable": {C.GPU_RESOURCE_NAME: 8}}
+        admin.create(node)
+
+        env = {
+            "PYTHONPATH": os.path.dirname(os.path.dirname(
+                os.path.abspath(__file__))),
+            "FMA_FAKE_GPU": "1",
+            "FMA_MOCK_GPU_COUNT": "4",
+            "FMA_GPU_MODE": "naive",
+            "FMA_ACCELERATORS": "GPU-0",
+        }
+        agent = NodeAgent(KubeStore(base, actor="node-agent"), "node-a",
+                          node_index=9, log_dir=str(tmp_path),
+                          extra_env=env)
+        agent.start()
+        ctl = DualPodsController(KubeStore(base, actor="dual-pods-controller"),
+                                 HttpAdapter(), ControllerConfig())
+        ctl.start()
+        pop = LauncherPopulator(KubeStore(base, actor="launcher-populator"))
+        pop.start()
+        user = KubeStore(base, actor="user")
+        try:
+            yield {"base": base, "store": user, "agent": agent,
+                   "ctl": ctl, "pop": pop}
+        finally:
+            ctl.stop()
+            pop.stop()
+            agent.stop()
+
+
+def test_kube_full_actuation_hot_start_and_vap(kube_cluster):
+    store, agent = kube_cluster["store"], kube_cluster["agent"]
+    mk_isc_lc_lpp(store)
+
+    lp = wait_for(lambda: launcher_pod(store), 60, desc="launcher pod")
+    wait_for(lambda: ob.pod_is_ready(store.get("Pod", ob.name_of(lp))),
+             90, desc="launcher Ready")
+
+    mk_requester(store, "kreq1")
+    wait_for(lambda: requester_ready(store, agent, "kreq1"), 120,
+             desc="requester /ready 200 over the kube protocol")
+
+    lp = store.get("Pod", ob.name_of(lp))
+    anns = ob.annotations_of(lp)
+    assert anns[C.REQUESTER_ANNOTATION].endswith(" kreq1")
+    assert ob.labels_of(lp)[C.SLEEPING_LABEL] == "false"
+    iid = anns[C.INSTANCE_ID_ANNOTATION]
+
+    # the serving instance answers on the launcher's IP
+    lp_ip = lp["status"]["podIP"]
+    r = httpx.post(f"http://{lp_ip}:{ISC_PORT}/v1/completions",
+                   json={"prompt": "hi", "max_tokens": 2}, timeout=10)
+    assert r.status_code == 200
+
+    # CEL VAP: a user cannot clear the binding annotation on the provider
+    hacked = store.get("Pod", ob.name_of(lp))
+    hacked["metadata"]["annotations"][C.REQUESTER_ANNOTATION] = "u hack"
+    with pytest.raises(Invalid):
+        store.update(hacked)
+
+    # delete requester -> sleep + unbind; launcher survives with the
+    # sleeping instance
+    store.delete("Pod", "kreq1")
+    wait_for(lambda: store.try_get("Pod", "kreq1") is None, 90,
+             desc="requester gone")
+    lp = store.get("Pod", ob.name_of(lp))
+    assert C.REQUESTER_ANNOTATION not in ob.annotations_of(lp)
+    assert ob.labels_of(lp)[C.SLEEPING_LABEL] == "true"
+
+    # hot start: same-ISC requester rebinds the sleeping instance
+    mk_requester(store, "kreq2")
+    wait_for(lambda: requester_ready(store, agent, "kreq2"), 90,
+             desc="kreq2 hot start")
+    lp = store.get("Pod", ob.name_of(lp))
+    assert ob.annotations_of(lp)[C.INSTANCE_ID_ANNOTATION] == iid
+    r = httpx.get(f"http://{lp_ip}:{ISC_PORT}/is_sleeping", timeout=5)
+    assert r.json() == {"is_sleeping": False}
+
+
+def test_kube_controller_restart_recovery(kube_cluster):
+    """Controller restart recovers bindings purely from Pod metadata read
+    back over the kube protocol (reference test-cases.sh:720)."""
+    store, agent = kube_cluster["store"], kube_cluster["agent"]
+    base = kube_cluster["base"]
+    mk_isc_lc_lpp(store)
+    lp = wait_for(lambda: launcher_pod(store), 60, desc="launcher pod")
+    wait_for(lambda: ob.pod_is_ready(store.get("Pod", ob.name_of(lp))),
+             90, desc="launcher Ready")
+    mk_requester(store, "rreq1")
+    wait_for(lambda: requester_ready(store, agent, "rreq1"), 120,
+             desc="requester ready")
+
+    kube_cluster["ctl"].stop()
+    ctl2 = DualPodsController(
+        KubeStore(base, actor="dual-pods-controller"), HttpAdapter(),
+        ControllerConfig())
+    ctl2.start()
+    try:
+        store.delete("Pod", "rreq1")
+        wait_for(lambda: store.try_get("Pod", "rreq1") is None, 90,
+                 desc="requester gone via recovered controller")
+        lp2 = store.get("Pod", ob.name_of(lp))
+        assert C.REQUESTER_ANNOTATION not in ob.annotations_of(lp2)
+        assert ob.labels_of(lp2)[C.SLEEPING_LABEL] == "true"
+    finally:
+        ctl2.stop()
