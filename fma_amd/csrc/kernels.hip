@@ -24,6 +24,8 @@
 
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+
 #include "kernels.h"
 
 #ifndef FMA_COPY_BLOCK
@@ -523,20 +525,52 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
   const unsigned short* kbase = K + static_cast<long long>(kvh) * hd;
   const unsigned short* vbase = V + static_cast<long long>(kvh) * hd;
 
-  // pass 1: scores for this chunk
+  // pass 1: scores, wave-cooperative. One wave covers one K row with a
+  // CONTIGUOUS 256 B load (lane -> hd/64 consecutive elements) and
+  // reduces the dot across lanes — the old thread-per-row layout made
+  // every wavefront touch 64 rows at k_stride apart, collapsing HBM
+  // efficiency ~10x at long t (the 4K-decode deficit in round 1).
+  // 4-row groups per wave so the four K-row loads issue together (one
+  // HBM latency per group, not per row) before the reduce chains run.
   float local_max = -1e30f;
-  for (int s0 = threadIdx.x; s0 < n; s0 += blockDim.x) {
-    const unsigned short* krow = kbase + (s_begin + s0) * k_stride;
-    float dot = 0.0f;
-    for (int i8 = 0; i8 < hd; i8 += 8) {
-      const uint4 kv8 = *reinterpret_cast<const uint4*>(krow + i8);
-      const unsigned short* kh = reinterpret_cast<const unsigned short*>(&kv8);
+  const int gstep = waves * 4;
+  const int full = (n / gstep) * gstep;  // complete 4-row groups
+  for (int g0 = wave * 4; g0 < full; g0 += gstep) {
+    float dot[4];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        dot = fmaf(s_q[i8 + j], bf16_to_f32(kh[j]), dot);
+    for (int g = 0; g < 4; ++g) {
+      const unsigned short* krow = kbase + (s_begin + g0 + g) * k_stride;
+      float d = 0.0f;
+#pragma unroll
+      for (int j = 0; j < 4 && j < per_lane; ++j) {
+        d = fmaf(s_q[lane * per_lane + j],
+                 bf16_to_f32(krow[lane * per_lane + j]), d);
       }
+      dot[g] = d;
     }
-    s_scores[s0] = dot;
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) {
+        dot[g] += __shfl_xor(dot[g], off, 64);
+      }
+      if (lane == 0) s_scores[g0 + g] = dot[g];
+      local_max = fmaxf(local_max, dot[g]);
+    }
+  }
+  for (int r0 = full + wave; r0 < n; r0 += waves) {
+    const unsigned short* krow = kbase + (s_begin + r0) * k_stride;
+    float dot = 0.0f;
+#pragma unroll
+    for (int j = 0; j < 4 && j < per_lane; ++j) {
+      dot = fmaf(s_q[lane * per_lane + j],
+                 bf16_to_f32(krow[lane * per_lane + j]), dot);
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      dot += __shfl_xor(dot, off, 64);
+    }
+    if (lane == 0) s_scores[r0] = dot;
     local_max = fmaxf(local_max, dot);
   }
   s_red[threadIdx.x] = local_max;
@@ -616,20 +650,27 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
   }
 }
 
-// Merge chunk partials: one wave per q head.
-__global__ __launch_bounds__(64) void attn_decode_combine_kernel(
+// Merge chunk partials: 4 waves per q head, each covering a quarter of
+// the chunks (a single wave serialized 32 chunk reads and became half
+// the 4K attention cost once the main kernel went wide).
+__global__ __launch_bounds__(256) void attn_decode_combine_kernel(
     const float* __restrict__ partials,  // [qH, chunks, hd + 2]
     unsigned short* __restrict__ out,    // [qH, hd]
     int chunks, int hd) {
   const int qh = blockIdx.x;
-  const int lane = threadIdx.x;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
   const int per_lane = hd / 64;
+  __shared__ float s_acc[4][256];
+  __shared__ float s_l[4];
   const float* base = partials + static_cast<long long>(qh) * chunks * (hd + 2);
   float m_tot = -1e30f;
-  for (int c = 0; c < chunks; ++c) m_tot = fmaxf(m_tot, base[c * (hd + 2) + hd]);
+  for (int c = 0; c < chunks; ++c) {
+    m_tot = fmaxf(m_tot, base[c * (hd + 2) + hd]);
+  }
   float l_tot = 0.0f;
   float acc[4] = {0.f, 0.f, 0.f, 0.f};
-  for (int c = 0; c < chunks; ++c) {
+  for (int c = wave; c < chunks; c += 4) {
     const float* p = base + c * (hd + 2);
     const float alpha = __expf(p[hd] - m_tot);
     l_tot += p[hd + 1] * alpha;
@@ -637,20 +678,34 @@ __global__ __launch_bounds__(64) void attn_decode_combine_kernel(
       acc[j] += p[lane * per_lane + j] * alpha;
     }
   }
-  const float inv = 1.0f / l_tot;
   for (int j = 0; j < per_lane; ++j) {
-    out[qh * hd + lane * per_lane + j] = f32_to_bf16(acc[j] * inv);
+    s_acc[wave][lane * per_lane + j] = acc[j];
+  }
+  if (lane == 0) s_l[wave] = l_tot;
+  __syncthreads();
+  if (wave == 0) {
+    const float inv =
+        1.0f / (s_l[0] + s_l[1] + s_l[2] + s_l[3]);
+    for (int j = 0; j < per_lane; ++j) {
+      const int e = lane * per_lane + j;
+      const float v = s_acc[0][e] + s_acc[1][e] + s_acc[2][e] + s_acc[3][e];
+      out[qh * hd + e] = f32_to_bf16(v * inv);
+    }
   }
 }
 
 }  // namespace
 
 extern "C" int fma_attn_decode_chunks(int t, int q_heads) {
-  // fill the chip: aim for >= 256 blocks, chunk >= 256 positions, and
-  // keep per-chunk scores within the LDS window
-  int chunks = 256 / q_heads;
+  // OVERFILL the chip: ~1024 blocks puts 4 workgroups on every CU
+  // (LDS/VGPR allow it), and that occupancy is what hides the HBM
+  // latency of the per-row K/V streams — at 256 blocks the kernel ran
+  // at occupancy 1 and ~2% of peak at t=4K (gpurun_out/decode4k_stats).
+  // Keep chunks >= 128 positions so the partials+combine overhead stays
+  // amortized, and within the LDS score window.
+  int chunks = 1024 / q_heads;
   if (chunks < 1) chunks = 1;
-  const int max_by_span = (t + 255) / 256;
+  const int max_by_span = (t + 127) / 128;
   if (chunks > max_by_span) chunks = max_by_span;
   const int min_by_lds = (t + FMA_ATTN_MAX_T / 4 - 1) / (FMA_ATTN_MAX_T / 4);
   if (chunks < min_by_lds) chunks = min_by_lds;
@@ -676,7 +731,7 @@ extern "C" hipError_t fma_launch_attn_decode_bf16(
       static_cast<unsigned short*>(out), partials, t_dev, t, q_heads,
       kv_heads, hd, k_stride);
   if (chunks > 1) {
-    attn_decode_combine_kernel<<<q_heads, 64, 0, stream>>>(
+    attn_decode_combine_kernel<<<q_heads, 256, 0, stream>>>(
         partials, static_cast<unsigned short*>(out), chunks, hd);
   }
   return hipGetLastError();
@@ -748,18 +803,19 @@ __device__ __forceinline__ void prefill_load_k(
 }
 
 // Stage one 32-key tile of V into LDS, transposed to [hd][key] so the
-// P.V B-fragment reads are contiguous b128 loads.
+// P.V B-fragment reads are contiguous b128 loads. [c8_lo, c8_hi) selects
+// the 16-channel chunks this wave stages (multi-wave workgroups split
+// the tile across waves — each stages HD/(16*NW) chunks).
 template <int HD>
 __device__ __forceinline__ void prefill_stage_v(
     const unsigned short* __restrict__ V, int kcol0, int t_kv, int kv_heads,
-    int kvh, int lane32, int half, __bf16* vdst) {
+    int kvh, int lane32, int half, __bf16* vdst, int c8_lo, int c8_hi) {
   const int key = kcol0 + lane32;
   const bool live = key < t_kv;
   const unsigned short* vp =
       V + (static_cast<long long>(live ? key : 0) * kv_heads + kvh) * HD +
       8 * half;
-#pragma unroll
-  for (int c8 = 0; c8 < HD / 16; ++c8) {
+  for (int c8 = c8_lo; c8 < c8_hi; ++c8) {
     bf16x8_t vv;
     if (live) {
       vv = *reinterpret_cast<const bf16x8_t*>(vp + c8 * 16);
@@ -774,9 +830,18 @@ __device__ __forceinline__ void prefill_stage_v(
   }
 }
 
-template <int HD>
-__global__ __launch_bounds__(64)
-__attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
+// NW = waves per workgroup. NW=4 covers the 4 q-heads of one GQA group
+// (or 4 group-mates) with ONE shared V staging pipeline: global V reads
+// and LDS fill drop 4x, and at ~34 KB LDS per 4-wave WG the CU fits 4
+// workgroups = 4 waves/SIMD (the 1-wave shape is LDS-bound at 2).
+// PIPE (NW>1 only): software-pipelined K prefetch + double V buffer at
+// occupancy 2, vs the streaming single-buffer shape that trades ILP for
+// lower LDS. Both are spill-free at waves_per_eu(2); A/B via
+// FMA_PREFILL_PIPE.
+template <int HD, int NW, bool PIPE = false>
+__global__ __launch_bounds__(64 * NW)
+__attribute__((amdgpu_waves_per_eu(2)))
+void attn_prefill_bf16_kernel(
     const unsigned short* __restrict__ Q,  // [T, qH, hd]
     const unsigned short* __restrict__ K,  // [S, kvH, hd]
     const unsigned short* __restrict__ V,  // [S, kvH, hd]
@@ -789,27 +854,41 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
   // balanced) and writes unnormalized (m, l, O) partials; the combine
   // kernel below merges them. Same recipe as attn_decode.
   constexpr int kNblk = HD / 32;
-  const int qh = blockIdx.x;
+  const int wid = NW > 1 ? (threadIdx.x >> 6) : 0;
+  const int qh = blockIdx.x * NW + wid;
   // schedule the HEAVIEST tiles first: under causal masking tile i does
   // i+1 key tiles of work, so launching high tiles last would leave a
   // ragged mostly-idle tail on the 1024 SIMDs
   const int tile = gridDim.y - 1 - blockIdx.y;
   const int r0 = tile * 32;
-  if (r0 >= T) return;
+  if (r0 >= T) return;  // uniform across the WG (depends on blockIdx only)
   const int kvh = qh / (q_heads / kv_heads);
-  const int l = threadIdx.x;
+  const int l = threadIdx.x & 63;
   const int half = l >> 5;  // 0 or 1
   const int lane32 = l & 31;
   const float scale = rsqrtf(static_cast<float>(HD));
 
-  // LDS is the occupancy limiter here (8 workgroups/CU for 2 waves/SIMD
-  // needs <= 20 KB each out of 160 KB/CU): P tile 2 KB + two V buffers
-  // 16 KB + bf16 score scratch 2.1 KB + stats ~0.3 KB = 20.4 -> fits 7-8.
-  __shared__ __bf16 s_p[32 * 32];          // P tile, A-fragment source
-  __shared__ __bf16 s_v[2][32 * HD];       // V, transposed [hd][key], 2 bufs
-  __shared__ __bf16 s_s[32 * FMA_SROW];    // score transpose scratch (max)
-  __shared__ float s_stat[32 * 2];         // per-row (m_new, alpha)
-  __shared__ float s_l[32];                // per-row l at epilogue
+  // NW=1: P 2 KB + two V bufs 16 KB + score scratch 2.1 KB + stats
+  // ~0.3 KB = 20.4 KB -> 7-8 WGs/CU (2 waves/SIMD). NW=4: the V bufs are
+  // SHARED (16 KB once), per-wave scratch x4 -> ~34 KB per 4-wave WG ->
+  // 4 WGs/CU = 4 waves/SIMD.
+  // NW=1 keeps the software-pipelined double V buffer; NW=4 trades the
+  // prefetch for occupancy (4 waves/SIMD needs <= 128 VGPRs, and the
+  // prefetch's long live ranges were pushing 242) and uses ONE V buffer
+  constexpr int kVBufs = (NW > 1 && !PIPE) ? 1 : 2;
+  __shared__ __bf16 s_p_all[NW][32 * 32];        // P tile, A-frag source
+  __shared__ __bf16 s_v[kVBufs][32 * HD];        // V [hd][key], SHARED
+  __shared__ __bf16 s_s_all[NW][32 * FMA_SROW];  // score transpose scratch
+  __shared__ float s_stat_all[NW][32 * 2];       // per-row (m_new, alpha)
+  __shared__ float s_l_all[NW][32];              // per-row l at epilogue
+  __bf16* const s_p = s_p_all[wid];
+  __bf16* const s_s = s_s_all[wid];
+  float* const s_stat = s_stat_all[wid];
+  float* const s_l = s_l_all[wid];
+  // this wave's share of the cooperative V staging
+  constexpr int kC8PerWave = (HD / 16) / NW;
+  const int c8_lo = wid * kC8PerWave;
+  const int c8_hi = c8_lo + kC8PerWave;
 
   // Q fragments, kept in registers for the whole key loop.
   // A-layout: lane holds A[lane32][8*half + i] per 16-k step.
@@ -945,23 +1024,87 @@ __attribute__((amdgpu_waves_per_eu(2))) void attn_prefill_bf16_kernel(
   // latency hides behind compute with zero extra registers.
   const int chunk = blockIdx.z;
   const int chunks = gridDim.z;
-  bf16x8_t kf[HD / 16];
-  if (chunk < kt_end) {
-    prefill_load_k<HD>(K, chunk * 32, t_kv, kv_heads, kvh, lane32, half,
-                       kf);
-    prefill_stage_v<HD>(V, chunk * 32, t_kv, kv_heads, kvh, lane32, half,
-                        s_v[0]);
-  }
-  int par = 0;  // ping-pong parity of the V LDS buffer
-  for (int kt = chunk; kt < kt_end; kt += chunks, par ^= 1) {
-    f32x16_t sacc = qk(kf);
-    if (kt + chunks < kt_end) {
-      prefill_load_k<HD>(K, (kt + chunks) * 32, t_kv, kv_heads, kvh,
-                         lane32, half, kf);
-      prefill_stage_v<HD>(V, (kt + chunks) * 32, t_kv, kv_heads, kvh,
-                          lane32, half, s_v[par ^ 1]);
+  if constexpr (NW == 1) {
+    // Software-pipelined key loop: kf is dead once the QK MFMAs consumed
+    // it, so the NEXT tile's K loads re-fill the same buffer — and the
+    // next V tile streams into the other LDS buffer — while the current
+    // tile's softmax+PV runs.
+    bf16x8_t kf[HD / 16];
+    if (chunk < kt_end) {
+      prefill_load_k<HD>(K, chunk * 32, t_kv, kv_heads, kvh, lane32, half,
+                         kf);
+      prefill_stage_v<HD>(V, chunk * 32, t_kv, kv_heads, kvh, lane32, half,
+                          s_v[0], c8_lo, c8_hi);
     }
-    softmax_pv(sacc, s_v[par], kt * 32);
+    int par = 0;  // ping-pong parity of the V LDS buffer
+    for (int kt = chunk; kt < kt_end; kt += chunks, par ^= 1) {
+      f32x16_t sacc = qk(kf);
+      if (kt + chunks < kt_end) {
+        prefill_load_k<HD>(K, (kt + chunks) * 32, t_kv, kv_heads, kvh,
+                           lane32, half, kf);
+        prefill_stage_v<HD>(V, (kt + chunks) * 32, t_kv, kv_heads, kvh,
+                            lane32, half, s_v[(par ^ 1) % kVBufs], c8_lo,
+                            c8_hi);
+      }
+      softmax_pv(sacc, s_v[par % kVBufs], kt * 32);
+    }
+  } else if constexpr (PIPE) {
+    // Multi-wave pipelined: same K-prefetch + V double buffer as NW==1,
+    // with one barrier per tile for the shared V staging.
+    bf16x8_t kf[HD / 16];
+    if (chunk < kt_end) {
+      prefill_load_k<HD>(K, chunk * 32, t_kv, kv_heads, kvh, lane32, half,
+                         kf);
+      prefill_stage_v<HD>(V, chunk * 32, t_kv, kv_heads, kvh, lane32, half,
+                          s_v[0], c8_lo, c8_hi);
+    }
+    int par = 0;
+    for (int kt = chunk; kt < kt_end; kt += chunks, par ^= 1) {
+      // (a) all waves' shares of s_v[par] (staged last iteration) are
+      // visible; (b) all waves finished reading s_v[par^1] before it is
+      // re-staged below
+      __syncthreads();
+      f32x16_t sacc = qk(kf);
+      if (kt + chunks < kt_end) {
+        prefill_load_k<HD>(K, (kt + chunks) * 32, t_kv, kv_heads, kvh,
+                           lane32, half, kf);
+        prefill_stage_v<HD>(V, (kt + chunks) * 32, t_kv, kv_heads, kvh,
+                            lane32, half, s_v[(par ^ 1) % kVBufs], c8_lo,
+                            c8_hi);
+      }
+      softmax_pv(sacc, s_v[par % kVBufs], kt * 32);
+    }
+  } else {
+    // Multi-wave streaming: no software prefetch (its live ranges cost
+    // ~100 VGPRs), K fragments streamed per 16-k step. One shared V
+    // buffer, two barriers per tile.
+    for (int kt = chunk; kt < kt_end; kt += chunks) {
+      const int key = kt * 32 + lane32;
+      const bool klive = key < t_kv;
+      const unsigned short* kp =
+          K + (static_cast<long long>(klive ? key : 0) * kv_heads + kvh) *
+              HD + 8 * half;
+      prefill_stage_v<HD>(V, kt * 32, t_kv, kv_heads, kvh, lane32, half,
+                          s_v[0], c8_lo, c8_hi);
+      f32x16_t sacc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) sacc[r] = 0.0f;
+#pragma unroll
+      for (int ks = 0; ks < HD / 16; ++ks) {
+        bf16x8_t kfs;
+        if (klive) {
+          kfs = *reinterpret_cast<const bf16x8_t*>(kp + ks * 16);
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) kfs[i] = static_cast<__bf16>(0.0f);
+        }
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf[ks], kfs, sacc,
+                                                       0, 0, 0);
+      }
+      __syncthreads();  // all waves' V shares staged
+      softmax_pv(sacc, s_v[0], kt * 32);
+      __syncthreads();  // all waves done reading before re-stage
+    }
   }
 
   if (partials != nullptr) {
@@ -1058,20 +1201,45 @@ extern "C" hipError_t fma_launch_attn_prefill_bf16(
   if (hd != 64 && hd != 128) return hipErrorInvalidValue;
   if (q_heads % kv_heads != 0) return hipErrorInvalidValue;
   if (chunks > 1 && partials == nullptr) return hipErrorInvalidValue;
-  dim3 grid(q_heads, (T + 31) / 32, chunks);
+  // 4-wave workgroups when 4 consecutive q-heads share one kv-head
+  // (llama GQA: group 4/8); else the 1-wave shape
+  const int group = q_heads / kv_heads;
+  int nw = (q_heads % 4 == 0 && group % 4 == 0) ? 4 : 1;
+  if (const char* e = getenv("FMA_PREFILL_NW")) {
+    const int forced = atoi(e);
+    if (forced == 1 || (forced == 4 && q_heads % 4 == 0 && group % 4 == 0))
+      nw = forced;
+  }
+  dim3 grid(q_heads / nw, (T + 31) / 32, chunks);
   float* par = chunks > 1 ? partials : nullptr;
-  if (hd == 128) {
-    attn_prefill_bf16_kernel<128><<<grid, 64, 0, stream>>>(
-        static_cast<const unsigned short*>(Q),
-        static_cast<const unsigned short*>(K),
-        static_cast<const unsigned short*>(V),
-        static_cast<unsigned short*>(O), par, T, pos0, q_heads, kv_heads);
+  const auto* Qp = static_cast<const unsigned short*>(Q);
+  const auto* Kp = static_cast<const unsigned short*>(K);
+  const auto* Vp = static_cast<const unsigned short*>(V);
+  auto* Op = static_cast<unsigned short*>(O);
+  const char* pe = getenv("FMA_PREFILL_PIPE");
+  const bool pipe = pe != nullptr && atoi(pe) != 0;
+  if (hd == 128 && nw == 4) {
+    if (pipe) {
+      attn_prefill_bf16_kernel<128, 4, true><<<grid, 256, 0, stream>>>(
+          Qp, Kp, Vp, Op, par, T, pos0, q_heads, kv_heads);
+    } else {
+      attn_prefill_bf16_kernel<128, 4><<<grid, 256, 0, stream>>>(
+          Qp, Kp, Vp, Op, par, T, pos0, q_heads, kv_heads);
+    }
+  } else if (hd == 128) {
+    attn_prefill_bf16_kernel<128, 1><<<grid, 64, 0, stream>>>(
+        Qp, Kp, Vp, Op, par, T, pos0, q_heads, kv_heads);
+  } else if (nw == 4) {
+    if (pipe) {
+      attn_prefill_bf16_kernel<64, 4, true><<<grid, 256, 0, stream>>>(
+          Qp, Kp, Vp, Op, par, T, pos0, q_heads, kv_heads);
+    } else {
+      attn_prefill_bf16_kernel<64, 4><<<grid, 256, 0, stream>>>(
+          Qp, Kp, Vp, Op, par, T, pos0, q_heads, kv_heads);
+    }
   } else {
-    attn_prefill_bf16_kernel<64><<<grid, 64, 0, stream>>>(
-        static_cast<const unsigned short*>(Q),
-        static_cast<const unsigned short*>(K),
-        static_cast<const unsigned short*>(V),
-        static_cast<unsigned short*>(O), par, T, pos0, q_heads, kv_heads);
+    attn_prefill_bf16_kernel<64, 1><<<grid, 64, 0, stream>>>(
+        Qp, Kp, Vp, Op, par, T, pos0, q_heads, kv_heads);
   }
   if (chunks > 1) {
     dim3 cgrid(q_heads, (T + 31) / 32);
