@@ -486,11 +486,16 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
   __shared__ float s_scores[FMA_ATTN_MAX_T / 4];
   __shared__ float s_q[256];
   __shared__ float s_red[256];
-  __shared__ float s_wacc[4][256];
+  __shared__ float s_wacc16[16][256];
 
-  const int qh = blockIdx.x;
-  const int chunk = blockIdx.y;
-  const int chunks = gridDim.y;
+  // chunk on X, head on Y: consecutive blockIdx (round-robin over the 8
+  // XCDs) then maps the SAME key chunk of every q-head to the SAME XCD
+  // whenever gridDim.x % 8 == 0 ((c + chunks*qh) % 8 == c % 8), so the 4
+  // q-heads of a GQA group hit their kv-head's K/V tile in that XCD's L2
+  // instead of re-reading HBM 4x.
+  const int qh = blockIdx.y;
+  const int chunk = blockIdx.x;
+  const int chunks = gridDim.x;
   const int span = (t + chunks - 1) / chunks;
   const int s_begin = chunk * span;
   const int s_end = min(s_begin + span, t);
@@ -530,48 +535,33 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
   // reduces the dot across lanes — the old thread-per-row layout made
   // every wavefront touch 64 rows at k_stride apart, collapsing HBM
   // efficiency ~10x at long t (the 4K-decode deficit in round 1).
-  // 4-row groups per wave so the four K-row loads issue together (one
-  // HBM latency per group, not per row) before the reduce chains run.
+  // 16 lanes per row, 4 rows per wave in parallel: one contiguous
+  // hd/16-element load per lane covers the row, the dot reduces in 4
+  // shfl steps within the 16-lane group (vs 6 across the full wave), and
+  // the four rows' loads issue under one HBM latency.
   float local_max = -1e30f;
-  const int gstep = waves * 4;
-  const int full = (n / gstep) * gstep;  // complete 4-row groups
-  for (int g0 = wave * 4; g0 < full; g0 += gstep) {
-    float dot[4];
-#pragma unroll
-    for (int g = 0; g < 4; ++g) {
-      const unsigned short* krow = kbase + (s_begin + g0 + g) * k_stride;
-      float d = 0.0f;
-#pragma unroll
-      for (int j = 0; j < 4 && j < per_lane; ++j) {
-        d = fmaf(s_q[lane * per_lane + j],
-                 bf16_to_f32(krow[lane * per_lane + j]), d);
+  const int g16 = lane >> 4;        // which of the wave's 4 rows
+  const int lane16 = lane & 15;     // position within the row
+  const int epl = hd / 16;          // elements per lane (8 at hd=128)
+  for (int r0 = wave * 4; r0 < n + 3; r0 += waves * 4) {
+    const int row = r0 + g16;
+    float d = 0.0f;
+    if (row < n) {
+      const unsigned short* krow = kbase + (s_begin + row) * k_stride;
+#pragma unroll 8
+      for (int j = 0; j < epl; ++j) {
+        d = fmaf(s_q[lane16 * epl + j],
+                 bf16_to_f32(krow[lane16 * epl + j]), d);
       }
-      dot[g] = d;
     }
 #pragma unroll
-    for (int g = 0; g < 4; ++g) {
-#pragma unroll
-      for (int off = 32; off > 0; off >>= 1) {
-        dot[g] += __shfl_xor(dot[g], off, 64);
-      }
-      if (lane == 0) s_scores[g0 + g] = dot[g];
-      local_max = fmaxf(local_max, dot[g]);
+    for (int off = 8; off > 0; off >>= 1) {
+      d += __shfl_xor(d, off, 16);
     }
-  }
-  for (int r0 = full + wave; r0 < n; r0 += waves) {
-    const unsigned short* krow = kbase + (s_begin + r0) * k_stride;
-    float dot = 0.0f;
-#pragma unroll
-    for (int j = 0; j < 4 && j < per_lane; ++j) {
-      dot = fmaf(s_q[lane * per_lane + j],
-                 bf16_to_f32(krow[lane * per_lane + j]), dot);
+    if (row < n) {
+      if (lane16 == 0) s_scores[row] = d;
+      local_max = fmaxf(local_max, d);
     }
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1) {
-      dot += __shfl_xor(dot, off, 64);
-    }
-    if (lane == 0) s_scores[r0] = dot;
-    local_max = fmaxf(local_max, dot);
   }
   s_red[threadIdx.x] = local_max;
   __syncthreads();
@@ -597,29 +587,25 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
   }
   const float l = s_red[0];
 
-  // pass 2: weighted V accumulation (FMA-only dependency, 2-deep unroll)
-  float acc0[4] = {0.f, 0.f, 0.f, 0.f};
-  float acc1[4] = {0.f, 0.f, 0.f, 0.f};
-  int s0 = wave;
-  for (; s0 + waves < n; s0 += 2 * waves) {
-    const unsigned short* vrow_a = vbase + (s_begin + s0) * k_stride;
-    const unsigned short* vrow_b = vbase + (s_begin + s0 + waves) * k_stride;
-    const float wa = s_scores[s0];
-    const float wb = s_scores[s0 + waves];
-    for (int j = 0; j < per_lane; ++j) {
-      acc0[j] = fmaf(wa, bf16_to_f32(vrow_a[lane * per_lane + j]), acc0[j]);
-      acc1[j] = fmaf(wb, bf16_to_f32(vrow_b[lane * per_lane + j]), acc1[j]);
+  // pass 2: weighted V accumulation with the same 16-lane row mapping
+  // (one contiguous load per lane per row); the 16 row-groups' partial
+  // sums merge through LDS.
+  float acc[16];
+#pragma unroll
+  for (int j = 0; j < 16; ++j) acc[j] = 0.0f;
+  for (int r0 = wave * 4; r0 < n + 3; r0 += waves * 4) {
+    const int row = r0 + g16;
+    if (row >= n) continue;
+    const unsigned short* vrow = vbase + (s_begin + row) * k_stride;
+    const float w = s_scores[row];
+#pragma unroll 8
+    for (int j = 0; j < epl; ++j) {
+      acc[j] = fmaf(w, bf16_to_f32(vrow[lane16 * epl + j]), acc[j]);
     }
   }
-  if (s0 < n) {
-    const unsigned short* vrow = vbase + (s_begin + s0) * k_stride;
-    const float w = s_scores[s0];
-    for (int j = 0; j < per_lane; ++j) {
-      acc0[j] = fmaf(w, bf16_to_f32(vrow[lane * per_lane + j]), acc0[j]);
-    }
-  }
-  for (int j = 0; j < per_lane; ++j) {
-    s_wacc[wave][lane * per_lane + j] = acc0[j] + acc1[j];
+#pragma unroll 8
+  for (int j = 0; j < epl; ++j) {
+    s_wacc16[wave * 4 + g16][lane16 * epl + j] = acc[j];
   }
   __syncthreads();
   if (wave == 0) {
@@ -627,8 +613,9 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
       const float inv_l = 1.0f / l;
       for (int j = 0; j < per_lane; ++j) {
         float v_out = 0.0f;
-        for (int wv = 0; wv < waves; ++wv) {
-          v_out += s_wacc[wv][lane * per_lane + j];
+#pragma unroll
+        for (int g = 0; g < 16; ++g) {
+          v_out += s_wacc16[g][lane * per_lane + j];
         }
         out[qh * hd + lane * per_lane + j] = f32_to_bf16(v_out * inv_l);
       }
@@ -637,8 +624,9 @@ __global__ __launch_bounds__(256) void attn_decode_bf16_kernel(
                                 (hd + 2);
       for (int j = 0; j < per_lane; ++j) {
         float v_out = 0.0f;
-        for (int wv = 0; wv < waves; ++wv) {
-          v_out += s_wacc[wv][lane * per_lane + j];
+#pragma unroll
+        for (int g = 0; g < 16; ++g) {
+          v_out += s_wacc16[g][lane * per_lane + j];
         }
         p[lane * per_lane + j] = v_out;
       }
@@ -703,12 +691,25 @@ extern "C" int fma_attn_decode_chunks(int t, int q_heads) {
   // at occupancy 1 and ~2% of peak at t=4K (gpurun_out/decode4k_stats).
   // Keep chunks >= 128 positions so the partials+combine overhead stays
   // amortized, and within the LDS score window.
-  int chunks = 1024 / q_heads;
+  int chunks = 512 / q_heads;  // ~512 blocks: measured best (24.7 us at
+                               // t=4K vs 26.4 at 1024; tools/
+                               // decode_kernel_bench.py sweep)
   if (chunks < 1) chunks = 1;
   const int max_by_span = (t + 127) / 128;
   if (chunks > max_by_span) chunks = max_by_span;
   const int min_by_lds = (t + FMA_ATTN_MAX_T / 4 - 1) / (FMA_ATTN_MAX_T / 4);
   if (chunks < min_by_lds) chunks = min_by_lds;
+  // multiples of 8 keep the chunk->XCD mapping head-invariant (see the
+  // grid comment in the kernel): same key chunk -> same XCD L2
+  if (chunks > 8 && chunks % 8) chunks += 8 - chunks % 8;
+  if (const char* e = getenv("FMA_DECODE_CHUNKS")) {
+    const int forced = atoi(e);
+    if (forced >= 1) {
+      const int lds_min =
+          (t + FMA_ATTN_MAX_T / 4 - 1) / (FMA_ATTN_MAX_T / 4);
+      chunks = forced < lds_min ? lds_min : forced;
+    }
+  }
   return chunks;
 }
 
@@ -723,7 +724,7 @@ extern "C" hipError_t fma_launch_attn_decode_bf16(
   if ((t + chunks - 1) / chunks > FMA_ATTN_MAX_T / 4)
     return hipErrorInvalidValue;
   if (chunks > 1 && partials == nullptr) return hipErrorInvalidValue;
-  dim3 grid(q_heads, chunks);
+  dim3 grid(chunks, q_heads);
   attn_decode_bf16_kernel<<<grid, 256, 0, stream>>>(
       static_cast<const unsigned short*>(q),
       static_cast<const unsigned short*>(K),
