@@ -120,8 +120,9 @@ class LlamaConfig:
                     ) -> List[Tuple[str, Tuple[int, ...], torch.dtype]]:
         """Names/shapes of this rank's parameter shard, in layout order."""
         assert self.num_heads % tp_size == 0, "heads must divide tp"
-        assert self.num_kv_heads % tp_size == 0 or tp_size <= self.num_kv_heads, \
-            "kv heads must divide tp"
+        assert (self.num_kv_heads % tp_size == 0
+                or tp_size % self.num_kv_heads == 0), \
+            "tp must divide kv heads or be a multiple (KV replication)"
         h = self.hidden_size
         hd = self.head_dim
         q_local = self.num_heads // tp_size * hd
@@ -216,10 +217,22 @@ class LlamaModel:
         self.params = params
 
     def init_weights(self, seed: int = 0) -> None:
+        import zlib
         torch.manual_seed(seed + self.tp_rank)
+        kv_rep = max(self.tp_size // self.cfg.num_kv_heads, 1)
         for name, p in self.params.items():
             if p.dim() > 1:
-                p.normal_(0.0, 0.02)  # in-place, device-side RNG: no temps
+                if kv_rep > 1 and (".wk." in name or ".wv." in name):
+                    # KV replication (tp > kv heads): ranks sharing a
+                    # logical kv head must generate identical weights
+                    group = self.tp_rank * self.cfg.num_kv_heads \
+                        // self.tp_size
+                    g = torch.Generator(device=p.device)
+                    g.manual_seed((seed * 1000003 + group) * 131071
+                                  + zlib.crc32(name.encode()))
+                    p.normal_(0.0, 0.02, generator=g)
+                else:
+                    p.normal_(0.0, 0.02)  # in-place device RNG: no temps
             else:
                 p.fill_(1.0)  # norm gains
 

@@ -103,19 +103,35 @@ _ROW_PARALLEL = {"wo", "w_down"}
 
 
 def shard_slice(name: str, tensor: torch.Tensor, tp_rank: int,
-                tp_size: int) -> torch.Tensor:
+                tp_size: int,
+                local_rows: "int | None" = None) -> torch.Tensor:
     """Slice one full (unsharded) checkpoint tensor down to the Megatron
-    shard that rank `tp_rank` of `tp_size` holds. Views, no copies."""
+    shard that rank `tp_rank` of `tp_size` holds. Views, no copies.
+
+    ``local_rows`` (the destination shard's dim0, known to every caller
+    from the parameter layout) enables KV-head REPLICATION for
+    tp_size > num_kv_heads: when dim0 splits into fewer groups than
+    ranks, ranks r in group g = r * groups // tp_size share logical
+    head(s) g — each replica reads the same checkpoint rows."""
     if tp_size <= 1:
         return tensor
     parts = name.split(".")
     leaf = parts[-2] if len(parts) >= 2 else ""
     if leaf in _COL_PARALLEL:
         n = tensor.shape[0]
+        if local_rows is not None and local_rows * tp_size != n:
+            # destination holds MORE than n/tp rows: KV replication
+            if n % local_rows == 0 and n // local_rows < tp_size:
+                groups = n // local_rows   # == num_kv_heads for wk/wv
+                g = tp_rank * groups // tp_size
+                return tensor[g * local_rows:(g + 1) * local_rows]
+            raise ValueError(
+                f"{name}: dim0 {n} cannot satisfy local shard of "
+                f"{local_rows} rows at tp_size {tp_size}")
         if n % tp_size != 0:
             raise ValueError(
                 f"{name}: dim0 {n} not divisible by tp_size {tp_size} "
-                "(tp_size > num_kv_heads is unsupported for checkpoints)")
+                "and no replication shape given")
         step = n // tp_size
         return tensor[tp_rank * step:(tp_rank + 1) * step]
     if leaf in _ROW_PARALLEL:
@@ -145,7 +161,8 @@ def load_into_params(path: str, params: Dict[str, torch.Tensor],
                                "matching parameter")
             continue
         p = params[name]
-        tensor = shard_slice(name, tensor, tp_rank, tp_size)
+        tensor = shard_slice(name, tensor, tp_rank, tp_size,
+                             local_rows=p.shape[0] if p.dim() else None)
         if tuple(tensor.shape) != tuple(p.shape):
             raise ValueError(f"shape mismatch for {name}: checkpoint "
                              f"shard {tuple(tensor.shape)} vs param "

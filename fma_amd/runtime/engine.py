@@ -186,7 +186,9 @@ class ActuationEngine:
             off, shape, dtype = self.layout[name]
             tensor = loader.shard_slice(tensor=tensor, name=name,
                                         tp_rank=self.tp_rank,
-                                        tp_size=self.tp_size)
+                                        tp_size=self.tp_size,
+                                        local_rows=shape[0] if shape
+                                        else None)
             if tuple(tensor.shape) != tuple(shape):
                 raise ValueError(f"shape mismatch for {name}")
             raw = tensor.to(dtype).contiguous().view(torch.uint8).view(-1)

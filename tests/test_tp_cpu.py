@@ -113,3 +113,79 @@ def test_tp_checkpoint_sharded_load_matches_full_model(tmp_path):
         env=dict(os.environ, PYTHONPATH=root))
     assert res.returncode == 0, f"stdout={res.stdout}\nstderr={res.stderr}"
     assert "TP_CKPT_OK" in res.stdout
+
+
+KVREP_PROBE = r"""
+# tp_size 4 > num_kv_heads 2: each pair of ranks REPLICATES one logical
+# kv head (Megatron grouping); decode must still match the full model.
+import os
+os.environ.setdefault("FMA_FAKE_GPU", "1")
+os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+import sys
+sys.path.insert(0, "__ROOT__")
+import torch
+from fma_amd.runtime.server import ServingRuntime, parse_options
+
+ref = torch.load(os.path.join("__TMP__", "ref.pt"), weights_only=True)
+toks = torch.load(os.path.join("__TMP__", "toks.pt"), weights_only=True)
+rt = ServingRuntime(parse_options(
+    "--model " + os.path.join("__TMP__", "kvrep-ckpt")
+    + " --tensor-parallel-size 4 --seed 4"))
+out = rt.rt.generate(toks, max_new_tokens=3)
+assert torch.equal(out, ref), (out, ref)
+# sleep/wake through the replicated shards stays bit-stable
+before = rt.rt.generate(toks, max_new_tokens=2).clone()
+rt.rt.sleep(1)
+rt.rt.wake_up()
+assert torch.equal(rt.rt.generate(toks, max_new_tokens=2), before)
+rt.rt.stop()
+print("TP_KVREP_OK")
+"""
+
+
+def test_tp_exceeding_kv_heads_replicates(tmp_path):
+    """tp_size > num_kv_heads (VERDICT round-1 gap): ranks sharing a
+    logical kv head load identical replicated wk/wv slices, and TP=4
+    decode over kv_heads=2 is token-identical to the full model."""
+    import torch
+    from fma_amd.models import loader
+    from fma_amd.models.llama import LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+
+    os.environ.setdefault("FMA_FAKE_GPU", "1")
+    cfg = LlamaConfig(name="kvrep", vocab_size=64, hidden_size=64,
+                      intermediate_size=96, num_layers=2, num_heads=4,
+                      num_kv_heads=2, max_seq_len=32)
+    src = ActuationEngine(cfg, seed=23)
+    loader.save_params(src.params, str(tmp_path / "kvrep-ckpt"), cfg)
+    toks = torch.randint(0, cfg.vocab_size, (1, 5),
+                         generator=torch.Generator().manual_seed(5))
+    torch.save(toks, str(tmp_path / "toks.pt"))
+    torch.save(src.model.generate(toks, max_new_tokens=3),
+               str(tmp_path / "ref.pt"))
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    probe = KVREP_PROBE.replace("__ROOT__", root).replace(
+        "__TMP__", str(tmp_path))
+    res = subprocess.run(
+        [sys.executable, "-c", probe],
+        capture_output=True, text=True, timeout=150,
+        env=dict(os.environ, PYTHONPATH=root))
+    assert res.returncode == 0, f"stdout={res.stdout}\nstderr={res.stderr}"
+    assert "TP_KVREP_OK" in res.stdout
+
+
+def test_shard_slice_kv_replication_unit():
+    import torch
+    from fma_amd.models.loader import shard_slice
+
+    full = torch.arange(4 * 8, dtype=torch.float32).view(4, 8)  # 2 heads*2
+    # tp=4, kv rows 4 (2 heads of dim0=2): local_rows=2 per rank,
+    # ranks 0,1 -> head 0 rows, ranks 2,3 -> head 1 rows
+    for r, expect_lo in [(0, 0), (1, 0), (2, 2), (3, 2)]:
+        got = shard_slice("layers.0.wk.weight", full, r, 4, local_rows=2)
+        assert torch.equal(got, full[expect_lo:expect_lo + 2]), r
+    # without replication info, non-divisible raises
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        shard_slice("layers.0.wk.weight", full, 0, 3)
