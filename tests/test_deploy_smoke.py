@@ -82,3 +82,38 @@ def test_deploy_cli_end_to_end(tmp_path):
         except ProcessLookupError:
             pass
         proc.wait(timeout=15)
+
+
+def test_dockerfiles_reference_valid_modules():
+    """Every container CMD must point at an importable module, and the
+    4-image strategy (controller/requester/launcher-gpu/launcher-cpu)
+    exists (reference dockerfiles/* builds 4 images incl. a CPU launcher
+    for GPU-less e2e)."""
+    import importlib
+    import os
+    import re
+
+    root = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "deploy", "docker")
+    files = sorted(os.listdir(root))
+    assert files == ["Dockerfile", "Dockerfile.controller",
+                     "Dockerfile.launcher.cpu", "Dockerfile.requester"]
+    for fn in files:
+        text = open(os.path.join(root, fn)).read()
+        for mod in re.findall(r'"-m", "([\w.]+)"', text):
+            importlib.import_module(mod)
+        assert "PYTHONPATH=/app" in text
+
+
+def test_ci_workflow_lists_real_paths():
+    import os
+
+    import yaml
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    wf = yaml.safe_load(open(os.path.join(
+        root, ".github", "workflows", "ci.yml")))
+    jobs = wf["jobs"]
+    assert {"cpu-tests", "kube-e2e", "hip-build", "images"} <= set(jobs)
+    for df in jobs["images"]["strategy"]["matrix"]["dockerfile"]:
+        assert os.path.exists(os.path.join(root, df)), df
