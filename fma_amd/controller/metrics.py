@@ -144,19 +144,53 @@ def serve_debug(port: int = 8003):
     """The reference's debug listener analog (prom-and-debug.go:68-79
     serves Go /debug/pprof on :8003). Go's pprof has no direct Python
     equivalent; this serves the operational 90% — "where is it stuck" —
-    as /debug/threads (a live stack dump of every thread) plus
-    /debug/vars (gc + thread counts). Returns the server (daemon thread).
+    as /debug/threads (a live stack dump of every thread), /debug/vars
+    (gc + thread counts), and /debug/profile?seconds=S&hz=H — a sampling
+    CPU profiler emitting flamegraph "folded" lines (the same role as
+    /debug/pprof/profile). Returns the server (daemon thread).
     """
     import gc
     import http.server
     import json
     import sys
     import threading
+    import time
     import traceback
+    from collections import Counter as _Counter
+    from urllib.parse import parse_qs, urlparse
+
+    def sample_profile(seconds: float, hz: float) -> str:
+        """Folded-stack samples across every thread (skipping ours)."""
+        me = threading.get_ident()
+        agg: _Counter = _Counter()
+        names = {t.ident: t.name for t in threading.enumerate()}
+        deadline = time.monotonic() + seconds
+        period = 1.0 / max(hz, 1.0)
+        while time.monotonic() < deadline:
+            for tid, frame in sys._current_frames().items():
+                if tid == me:
+                    continue
+                stack = []
+                f = frame
+                while f is not None:
+                    stack.append(f"{f.f_code.co_name} "
+                                 f"({f.f_code.co_filename.rsplit('/', 1)[-1]}"
+                                 f":{f.f_lineno})")
+                    f = f.f_back
+                key = names.get(tid, "?") + ";" + ";".join(reversed(stack))
+                agg[key] += 1
+            time.sleep(period)
+        return "".join(f"{k} {v}\n" for k, v in agg.most_common())
 
     class Handler(http.server.BaseHTTPRequestHandler):
         def do_GET(self):  # noqa: N802
-            if self.path.startswith("/debug/threads"):
+            if self.path.startswith("/debug/profile"):
+                q = parse_qs(urlparse(self.path).query)
+                seconds = min(float(q.get("seconds", ["5"])[0]), 60.0)
+                hz = min(float(q.get("hz", ["100"])[0]), 1000.0)
+                body = sample_profile(seconds, hz).encode()
+                ctype = "text/plain"
+            elif self.path.startswith("/debug/threads"):
                 names = {t.ident: t.name for t in threading.enumerate()}
                 out = []
                 for tid, frame in sys._current_frames().items():

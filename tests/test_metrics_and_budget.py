@@ -227,3 +227,40 @@ def test_sleeper_budget_within_limit_no_eviction():
     sdata = ServerData(uid="u1", requester_name="r1", gpus=["GPU-0"])
     ctl._enforce_sleeper_budget("node-a", sdata)
     assert len(store.list("Pod")) == 2
+
+
+def test_debug_profile_endpoint_samples_stacks():
+    """/debug/profile is the pprof-profile analog: folded flamegraph
+    lines sampled across live threads."""
+    import threading
+    import time
+
+    import httpx
+
+    from fma_amd.controller import metrics as M
+
+    stop = threading.Event()
+
+    def busy_loop_marker():
+        while not stop.is_set():
+            sum(range(200))
+            time.sleep(0.001)
+
+    th = threading.Thread(target=busy_loop_marker, name="busy-marker",
+                          daemon=True)
+    th.start()
+    srv = M.serve_debug(0)
+    try:
+        port = srv.server_address[1]
+        r = httpx.get(
+            f"http://127.0.0.1:{port}/debug/profile?seconds=0.4&hz=200",
+            timeout=10)
+        assert r.status_code == 200
+        assert "busy_loop_marker" in r.text
+        # folded format: "name;frame;frame N"
+        line = next(l for l in r.text.splitlines()
+                    if "busy_loop_marker" in l)
+        assert line.rsplit(" ", 1)[1].isdigit()
+    finally:
+        stop.set()
+        srv.shutdown()
