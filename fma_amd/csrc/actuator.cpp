@@ -1155,6 +1155,41 @@ at::Tensor gemv_bf16(const at::Tensor& W, const at::Tensor& x,
   return y;
 }
 
+std::vector<at::Tensor> gemv_multi_bf16(const at::Tensor& x,
+                                        std::vector<at::Tensor> ws) {
+  TORCH_CHECK(!ws.empty() && ws.size() <= 3,
+              "gemv_multi_bf16 takes 1-3 weight matrices");
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
+              x.is_contiguous(), "x must be contiguous bf16");
+  const int64_t K = x.numel();
+  TORCH_CHECK((K & 7) == 0, "K must be a multiple of 8");
+  std::vector<at::Tensor> ys;
+  for (const auto& w : ws) {
+    TORCH_CHECK(w.is_cuda() && w.scalar_type() == at::kBFloat16 &&
+                w.dim() == 2 && w.is_contiguous() && w.size(1) == K,
+                "weights must be contiguous bf16 [M, K]");
+    ys.push_back(at::empty({w.size(0)}, w.options()));
+  }
+  while (ws.size() < 3) {
+    ws.push_back(at::Tensor());
+  }
+  auto wptr = [&](size_t i) -> const void* {
+    return ws[i].defined() ? ws[i].data_ptr() : nullptr;
+  };
+  auto yptr = [&](size_t i) -> void* {
+    return i < ys.size() ? ys[i].data_ptr() : nullptr;
+  };
+  auto mdim = [&](size_t i) -> int {
+    return ws[i].defined() ? static_cast<int>(ws[i].size(0)) : 0;
+  };
+  auto stream = c10::hip::getCurrentHIPStream(x.device().index());
+  FMA_HIP_CHECK(fma_launch_gemv_multi_bf16(
+      wptr(0), mdim(0), yptr(0), wptr(1), mdim(1), yptr(1), wptr(2),
+      mdim(2), yptr(2), x.data_ptr(), static_cast<int>(K),
+      stream.stream()));
+  return ys;
+}
+
 at::Tensor rmsnorm1_bf16(const at::Tensor& x, const at::Tensor& w,
                          double eps) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
@@ -1353,6 +1388,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("_release_staging", [](int device) {
     ctx_for(device).release_staging();
   }, py::arg("device"), "debug: drop the persistent staging buffers");
+  m.def("gemv_multi_bf16", &gemv_multi_bf16,
+        "1-3 batch-1 bf16 GEMVs sharing one x in a single launch",
+        py::arg("x"), py::arg("weights"));
   m.def("gemv_bf16", &gemv_bf16, "Batch-1 bf16 GEMV",
         py::arg("W"), py::arg("x"), py::arg("out_bf16") = false,
         py::arg("residual") = py::none());

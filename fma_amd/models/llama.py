@@ -29,7 +29,8 @@ import torch.nn.functional as F
 
 from fma_amd.ops.decode_ops import (fast_attn_decode, fast_attn_prefill,
                                     fast_rmsnorm, fast_rope1, fast_silu_mul)
-from fma_amd.ops.linear import fast_linear, fast_linear_residual
+from fma_amd.ops.linear import (fast_linear, fast_linear_multi,
+                                fast_linear_residual)
 
 
 @dataclass
@@ -261,9 +262,18 @@ class LlamaModel:
             h = fast_rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps) \
                 if decode1 else rmsnorm(x, P[p + "attn_norm.weight"],
                                         cfg.norm_eps)
-            q = fast_linear(h, P[p + "wq.weight"]).view(B, T, q_heads, hd)
-            k = fast_linear(h, P[p + "wk.weight"]).view(B, T, kv_heads, hd)
-            v = fast_linear(h, P[p + "wv.weight"]).view(B, T, kv_heads, hd)
+            if decode1:
+                # qkv in ONE launch (shared x, virtual row space)
+                q, k, v = fast_linear_multi(
+                    h, (P[p + "wq.weight"], P[p + "wk.weight"],
+                        P[p + "wv.weight"]))
+            else:
+                q = fast_linear(h, P[p + "wq.weight"])
+                k = fast_linear(h, P[p + "wk.weight"])
+                v = fast_linear(h, P[p + "wv.weight"])
+            q = q.view(B, T, q_heads, hd)
+            k = k.view(B, T, kv_heads, hd)
+            v = v.view(B, T, kv_heads, hd)
             if decode1:
                 q = fast_rope1(q, self.rope_cos[start_pos],
                                self.rope_sin[start_pos])
@@ -313,8 +323,12 @@ class LlamaModel:
             h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps) \
                 if decode1 else rmsnorm(x, P[p + "mlp_norm.weight"],
                                         cfg.norm_eps)
-            gate = fast_linear(h, P[p + "w_gate.weight"])
-            up = fast_linear(h, P[p + "w_up.weight"])
+            if decode1:
+                gate, up = fast_linear_multi(
+                    h, (P[p + "w_gate.weight"], P[p + "w_up.weight"]))
+            else:
+                gate = fast_linear(h, P[p + "w_gate.weight"])
+                up = fast_linear(h, P[p + "w_up.weight"])
             act = fast_silu_mul(gate, up) if decode1 else F.silu(gate) * up
             if decode1 and self.tp_size == 1:
                 x = fast_linear_residual(act, P[p + "w_down.weight"], x)

@@ -41,6 +41,21 @@ def fast_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return F.linear(x, w)
 
 
+def fast_linear_multi(x, weights):
+    """1-3 projections of the same single-token x in ONE kernel launch
+    (qkv, gate+up). Falls back to per-weight F.linear off the fast path."""
+    if (gemv_enabled() and x.is_cuda
+            and x.dtype == torch.bfloat16
+            and x.numel() == x.shape[-1]
+            and all(w.dtype == torch.bfloat16 and w.is_contiguous()
+                    and (w.shape[1] & 7) == 0 for w in weights)):
+        ys = actuation._C.gemv_multi_bf16(x.reshape(-1).contiguous(),
+                                          list(weights))
+        return [y.view(*x.shape[:-1], w.shape[0])
+                for y, w in zip(ys, weights)]
+    return [F.linear(x, w) for w in weights]
+
+
 def fast_linear_residual(x: torch.Tensor, w: torch.Tensor,
                          residual: torch.Tensor) -> torch.Tensor:
     """residual + F.linear(x, w) with the add fused into the GEMV store."""

@@ -48,6 +48,7 @@ def main():
         return dt
 
     def decode_tps():
+        """End-to-end rate of generate() (prefill + decode)."""
         eng = r.engine if hasattr(r, "engine") else r
         t0 = time.perf_counter()
         eng.generate(toks, max_new_tokens=args.new_tokens)
@@ -55,10 +56,32 @@ def main():
             torch.cuda.synchronize()
         return args.new_tokens / (time.perf_counter() - t0)
 
+    def pure_decode_tps():
+        """Steady-state single-token rate: prefill once, then time only
+        the decode steps against the warm KV cache."""
+        eng = r.engine if hasattr(r, "engine") else r
+        cache = eng.new_kv_cache(1, args.prompt_len + args.new_tokens + 2)
+        logits = eng.model.forward(toks.to(eng.device), cache, 0)
+        nxt = logits[:, -1:].argmax(-1)
+        if eng.on_gpu:
+            torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        pos = args.prompt_len
+        for _ in range(args.new_tokens):
+            logits = eng.model.forward(nxt, cache, pos)
+            nxt = logits[:, -1:].argmax(-1)
+            pos += 1
+        if eng.on_gpu:
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        cache.free()
+        return args.new_tokens / dt
+
     # steady state
     first_token_seconds()
     steady_ttft = min(first_token_seconds() for _ in range(args.reps))
     steady_tps = max(decode_tps() for _ in range(args.reps))
+    steady_pure_tps = max(pure_decode_tps() for _ in range(args.reps))
 
     # right after a wake (allocator cold, caches dropped)
     post_wake_ttft = []
@@ -87,6 +110,7 @@ def main():
         "steady_ttft_s": round(steady_ttft, 4),
         "post_wake_ttft_s": round(min(post_wake_ttft), 4),
         "decode_tok_s": round(steady_tps, 2),
+        "decode_tok_s_pure": round(steady_pure_tps, 2),
         "decode_tok_s_hipgraph": round(graphed_tps, 2) if graphed_tps else None,
     }))
     if hasattr(r, "stop"):
