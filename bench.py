@@ -67,7 +67,8 @@ def main() -> None:
         from datetime import timedelta
 
         import torch.distributed as dist
-        backend = "nccl" if on_gpu else "gloo"
+        backend = os.environ.get("FMA_DIST_BACKEND") or (
+            "nccl" if on_gpu else "gloo")
         # RCCL preflight: dmabuf IPC is the only mode the pool's driver
         # supports; a hung rank should fail the job in minutes, not hang
         # the box until the driver's limit kills it
@@ -77,9 +78,12 @@ def main() -> None:
         if on_gpu:
             ngpu = torch.cuda.device_count()
             if local_rank >= ngpu:
-                log(f"[rank {rank}] FATAL: LOCAL_RANK {local_rank} but only "
-                    f"{ngpu} visible GPUs")
-                sys.exit(2)
+                if backend == "gloo":
+                    local_rank = local_rank % ngpu  # rehearsal on 1 GPU
+                else:
+                    log(f"[rank {rank}] FATAL: LOCAL_RANK {local_rank} "
+                        f"but only {ngpu} visible GPUs")
+                    sys.exit(2)
             torch.cuda.set_device(local_rank)
         try:
             dist.init_process_group(backend=backend, rank=rank,
