@@ -150,3 +150,86 @@ def test_kube_controller_restart_recovery(kube_cluster):
         assert ob.labels_of(lp2)[C.SLEEPING_LABEL] == "true"
     finally:
         ctl2.stop()
+
+
+def test_kube_instance_cap_and_second_launcher(kube_cluster):
+    """Per-launcher instance cap over the kube protocol: with
+    maxInstances=1 and two different-ISC requesters, the stack creates a
+    second launcher rather than over-filling the first (reference
+    test-cases.sh:633 'per-launcher instance cap' + :396 same-node
+    collision handling)."""
+    store, agent = kube_cluster["store"], kube_cluster["agent"]
+    store.create(ob.new_object(
+        "LauncherConfig", "lc1",
+        spec={"maxInstances": 1, "podTemplate": {"spec": {"containers": [{
+            "name": "launcher",
+            "command": [__import__("sys").executable, "-m",
+                        "fma_amd.launcher.service"],
+        }]}}}))
+    for i, port in enumerate((8372, 8373)):
+        store.create(ob.new_object(
+            "InferenceServerConfig", f"isc-cap{i}",
+            spec={"modelServerConfig": {"port": port,
+                                        "options": "--model tiny"},
+                  "launcherConfigName": "lc1"}))
+    store.create(ob.new_object(
+        "LauncherPopulationPolicy", "lpp-cap",
+        spec={"enhancedNodeSelector": {"labelSelector": {}},
+              "countForLauncher": [
+                  {"launcherConfigName": "lc1", "launcherCount": 2}]}))
+    wait_for(lambda: sum(
+        1 for p in store.list("Pod")
+        if ob.labels_of(p).get(C.COMPONENT_LABEL) == C.LAUNCHER_COMPONENT
+        and ob.pod_is_ready(p)) >= 2, 120, desc="2 launchers ready")
+
+    for i in range(2):
+        store.create(ob.new_object(
+            "Pod", f"capreq{i}",
+            annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: f"isc-cap{i}"},
+            spec={"nodeName": "node-a", "containers": [{
+                "name": "requester",
+                "command": [__import__("sys").executable, "-m",
+                            "fma_amd.requester.server"]}]}))
+    for i in range(2):
+        wait_for(lambda i=i: requester_ready(store, agent, f"capreq{i}"),
+                 120, desc=f"capreq{i} ready")
+    bound = {ob.annotations_of(p)[C.REQUESTER_ANNOTATION]
+             for p in store.list("Pod")
+             if ob.labels_of(p).get(C.COMPONENT_LABEL) ==
+             C.LAUNCHER_COMPONENT
+             and ob.annotations_of(p).get(C.REQUESTER_ANNOTATION)}
+    assert len(bound) == 2  # two launchers, one instance each (cap=1)
+
+
+def test_kube_stopped_instance_recovers(kube_cluster):
+    """Stopped-instance recovery over the kube protocol (reference
+    test-cases.sh:905): kill the serving instance process; the sentinel
+    exit surfaces through the launcher, the controller deletes the
+    requester Pod (its owner would re-create it in a Deployment)."""
+    import signal
+
+    store, agent = kube_cluster["store"], kube_cluster["agent"]
+    mk_isc_lc_lpp(store)
+    lp = wait_for(lambda: launcher_pod(store), 60, desc="launcher pod")
+    wait_for(lambda: ob.pod_is_ready(store.get("Pod", ob.name_of(lp))),
+             90, desc="launcher Ready")
+    mk_requester(store, "sreq1")
+    wait_for(lambda: requester_ready(store, agent, "sreq1"), 120,
+             desc="requester ready")
+
+    # SIGKILL the forked serving-instance process: it is the launcher
+    # process's child (the wire state carries no pid, as in the
+    # reference; the e2e reaches around the API like test-cases.sh does
+    # with kubectl exec)
+    import psutil
+
+    lname = ob.name_of(lp)
+    launcher_proc = psutil.Process(agent.pods[lname].proc.pid)
+    kids = wait_for(lambda: launcher_proc.children() or None, 30,
+                    desc="instance child process")
+    os.kill(kids[0].pid, signal.SIGKILL)
+
+    # controller reacts: requester Pod deleted (reference deletes it so
+    # the owning ReplicaSet re-creates a fresh one)
+    wait_for(lambda: store.try_get("Pod", "sreq1") is None, 120,
+             desc="requester deleted after instance death")
