@@ -1155,8 +1155,9 @@ at::Tensor gemv_bf16(const at::Tensor& W, const at::Tensor& x,
   return y;
 }
 
-std::vector<at::Tensor> gemv_multi_bf16(const at::Tensor& x,
-                                        std::vector<at::Tensor> ws) {
+std::vector<at::Tensor> gemv_multi_bf16(
+    const at::Tensor& x, std::vector<at::Tensor> ws,
+    const c10::optional<at::Tensor>& norm_w, double norm_eps) {
   TORCH_CHECK(!ws.empty() && ws.size() <= 3,
               "gemv_multi_bf16 takes 1-3 weight matrices");
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
@@ -1182,11 +1183,20 @@ std::vector<at::Tensor> gemv_multi_bf16(const at::Tensor& x,
   auto mdim = [&](size_t i) -> int {
     return ws[i].defined() ? static_cast<int>(ws[i].size(0)) : 0;
   };
+  const void* nw = nullptr;
+  if (norm_w.has_value()) {
+    const auto& n = norm_w.value();
+    TORCH_CHECK(n.scalar_type() == at::kBFloat16 && n.numel() == K &&
+                n.is_contiguous(), "norm_w must be contiguous bf16 [K]");
+    TORCH_CHECK(K * 2 <= 32 * 1024,
+                "norm fusion needs x to fit the LDS stage (K <= 16384)");
+    nw = n.data_ptr();
+  }
   auto stream = c10::hip::getCurrentHIPStream(x.device().index());
   FMA_HIP_CHECK(fma_launch_gemv_multi_bf16(
       wptr(0), mdim(0), yptr(0), wptr(1), mdim(1), yptr(1), wptr(2),
-      mdim(2), yptr(2), x.data_ptr(), static_cast<int>(K),
-      stream.stream()));
+      mdim(2), yptr(2), x.data_ptr(), static_cast<int>(K), nw,
+      static_cast<float>(norm_eps), stream.stream()));
   return ys;
 }
 
@@ -1389,8 +1399,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     ctx_for(device).release_staging();
   }, py::arg("device"), "debug: drop the persistent staging buffers");
   m.def("gemv_multi_bf16", &gemv_multi_bf16,
-        "1-3 batch-1 bf16 GEMVs sharing one x in a single launch",
-        py::arg("x"), py::arg("weights"));
+        "1-3 batch-1 bf16 GEMVs sharing one x in a single launch; "
+        "optional fused rmsnorm of x (bit-identical to rmsnorm1_bf16)",
+        py::arg("x"), py::arg("weights"), py::arg("norm_w") = py::none(),
+        py::arg("norm_eps") = 0.0);
   m.def("gemv_bf16", &gemv_bf16, "Batch-1 bf16 GEMV",
         py::arg("W"), py::arg("x"), py::arg("out_bf16") = false,
         py::arg("residual") = py::none());

@@ -300,7 +300,12 @@ extern "C" hipError_t fma_launch_gemv_bf16(const void* W, const void* x,
 // gate+up): one virtual row space M0+M1+M2, per-row segment lookup.
 // Kills 2 launch gaps per use and fills the chip even when the small
 // KV projections (M=1024 rows) alone could not.
-template <bool kUseLds>
+// kNorm folds the rmsnorm producing x into this launch: every block
+// stages RAW x, computes sum-of-squares with EXACTLY rmsnorm1's
+// accumulation pattern (bit-identical inv_rms), then normalizes its
+// staged copy in place — one launch instead of rmsnorm + gemv, and the
+// norm result never round-trips HBM. Requires the LDS-staged path.
+template <bool kUseLds, bool kNorm = false>
 __global__ __launch_bounds__(256) void gemv_multi_bf16_kernel(
     const unsigned short* __restrict__ W0, int M0,
     unsigned short* __restrict__ y0,
@@ -308,8 +313,11 @@ __global__ __launch_bounds__(256) void gemv_multi_bf16_kernel(
     unsigned short* __restrict__ y1,
     const unsigned short* __restrict__ W2, int M2,
     unsigned short* __restrict__ y2,
-    const unsigned short* __restrict__ x, int K) {
+    const unsigned short* __restrict__ x, int K,
+    const unsigned short* __restrict__ norm_w, float norm_eps) {
+  static_assert(!kNorm || kUseLds, "norm fusion needs the LDS path");
   extern __shared__ unsigned short s_x[];
+  __shared__ float s_red[256];
   const unsigned short* xsrc = x;
   if (kUseLds) {
     for (int i = threadIdx.x; i * 8 < K; i += blockDim.x) {
@@ -318,6 +326,36 @@ __global__ __launch_bounds__(256) void gemv_multi_bf16_kernel(
     }
     __syncthreads();
     xsrc = s_x;
+  }
+  if (kNorm) {
+    // sum of squares: same per-thread element mapping + tree as
+    // rmsnorm1_bf16_kernel so inv_rms is bit-identical to the unfused
+    // two-kernel sequence
+    float ss = 0.0f;
+    const int vec_h = K >> 3;
+    for (int i = threadIdx.x; i < vec_h; i += blockDim.x) {
+      const uint4 xv = reinterpret_cast<const uint4*>(x)[i];
+      const unsigned short* xh =
+          reinterpret_cast<const unsigned short*>(&xv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float v = bf16_to_f32(xh[j]);
+        ss = fmaf(v, v, ss);
+      }
+    }
+    s_red[threadIdx.x] = ss;
+    __syncthreads();
+    for (int r = 128; r > 0; r >>= 1) {
+      if (threadIdx.x < r) s_red[threadIdx.x] += s_red[threadIdx.x + r];
+      __syncthreads();
+    }
+    const float inv_rms =
+        rsqrtf(s_red[0] / static_cast<float>(K) + norm_eps);
+    for (int i = threadIdx.x; i < K; i += blockDim.x) {
+      s_x[i] = f32_to_bf16(bf16_to_f32(s_x[i]) * inv_rms *
+                           bf16_to_f32(norm_w[i]));
+    }
+    __syncthreads();
   }
   const int lane = threadIdx.x & 63;
   const int wave_in_block = threadIdx.x >> 6;
@@ -361,7 +399,7 @@ __global__ __launch_bounds__(256) void gemv_multi_bf16_kernel(
 extern "C" hipError_t fma_launch_gemv_multi_bf16(
     const void* W0, int M0, void* y0, const void* W1, int M1, void* y1,
     const void* W2, int M2, void* y2, const void* x, int K,
-    hipStream_t stream) {
+    const void* norm_w, float norm_eps, hipStream_t stream) {
   if ((K & 7) != 0) return hipErrorInvalidValue;
   const int block = 256;
   const int waves_per_block = block / 64;
@@ -369,13 +407,26 @@ extern "C" hipError_t fma_launch_gemv_multi_bf16(
   int blocks = (M + waves_per_block - 1) / waves_per_block;
   if (blocks > 2048) blocks = 2048;
   const size_t lds = static_cast<size_t>(K) * sizeof(unsigned short);
+  if (norm_w != nullptr) {
+    if (lds > 32 * 1024) return hipErrorInvalidValue;
+    gemv_multi_bf16_kernel<true, true><<<blocks, block, lds, stream>>>(
+        static_cast<const unsigned short*>(W0), M0,
+        static_cast<unsigned short*>(y0),
+        static_cast<const unsigned short*>(W1), M1,
+        static_cast<unsigned short*>(y1),
+        static_cast<const unsigned short*>(W2), M2,
+        static_cast<unsigned short*>(y2),
+        static_cast<const unsigned short*>(x), K,
+        static_cast<const unsigned short*>(norm_w), norm_eps);
+    return hipGetLastError();
+  }
 #define FMA_GEMVM_ARGS                                                    static_cast<const unsigned short*>(W0), M0,                                 static_cast<unsigned short*>(y0),                                       static_cast<const unsigned short*>(W1), M1,                             static_cast<unsigned short*>(y1),                                       static_cast<const unsigned short*>(W2), M2,                             static_cast<unsigned short*>(y2),                                       static_cast<const unsigned short*>(x), K
   if (lds <= 32 * 1024) {
     gemv_multi_bf16_kernel<true><<<blocks, block, lds, stream>>>(
-        FMA_GEMVM_ARGS);
+        FMA_GEMVM_ARGS, nullptr, 0.0f);
   } else {
     gemv_multi_bf16_kernel<false><<<blocks, block, 0, stream>>>(
-        FMA_GEMVM_ARGS);
+        FMA_GEMVM_ARGS, nullptr, 0.0f);
   }
 #undef FMA_GEMVM_ARGS
   return hipGetLastError();

@@ -259,15 +259,19 @@ class LlamaModel:
                    and x.dtype == torch.bfloat16)
         for li in range(cfg.num_layers):
             p = f"layers.{li}."
-            h = fast_rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps) \
-                if decode1 else rmsnorm(x, P[p + "attn_norm.weight"],
-                                        cfg.norm_eps)
             if decode1:
-                # qkv in ONE launch (shared x, virtual row space)
+                # qkv in ONE launch. (A norm-fused variant exists —
+                # gemv_multi_bf16(norm_w=...) — but measured SLOWER:
+                # every block redundantly re-reads x and reduces the sum
+                # of squares, costing more than the one saved launch:
+                # 228 -> 220 tok/s short ctx. Selection is measured.)
+                h = fast_rmsnorm(x, P[p + "attn_norm.weight"],
+                                 cfg.norm_eps)
                 q, k, v = fast_linear_multi(
                     h, (P[p + "wq.weight"], P[p + "wk.weight"],
                         P[p + "wv.weight"]))
             else:
+                h = rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps)
                 q = fast_linear(h, P[p + "wq.weight"])
                 k = fast_linear(h, P[p + "wk.weight"])
                 v = fast_linear(h, P[p + "wv.weight"])
@@ -320,13 +324,13 @@ class LlamaModel:
                 x = x + self._maybe_all_reduce(
                     fast_linear(att, P[p + "wo.weight"]))
 
-            h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps) \
-                if decode1 else rmsnorm(x, P[p + "mlp_norm.weight"],
-                                        cfg.norm_eps)
             if decode1:
+                h = fast_rmsnorm(x, P[p + "mlp_norm.weight"],
+                                 cfg.norm_eps)
                 gate, up = fast_linear_multi(
                     h, (P[p + "w_gate.weight"], P[p + "w_up.weight"]))
             else:
+                h = rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
                 gate = fast_linear(h, P[p + "w_gate.weight"])
                 up = fast_linear(h, P[p + "w_up.weight"])
             act = fast_silu_mul(gate, up) if decode1 else F.silu(gate) * up

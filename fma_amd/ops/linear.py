@@ -41,18 +41,33 @@ def fast_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return F.linear(x, w)
 
 
-def fast_linear_multi(x, weights):
+def fast_linear_multi(x, weights, norm=None):
     """1-3 projections of the same single-token x in ONE kernel launch
-    (qkv, gate+up). Falls back to per-weight F.linear off the fast path."""
+    (qkv, gate+up), optionally with the rmsnorm producing the GEMV input
+    fused in (norm = (weight, eps); x is then the RAW residual stream).
+    Falls back to eager rmsnorm + per-weight F.linear off the fast path."""
+    K = x.shape[-1]
     if (gemv_enabled() and x.is_cuda
             and x.dtype == torch.bfloat16
-            and x.numel() == x.shape[-1]
+            and x.numel() == K
+            and (norm is None or (norm[0].is_contiguous()
+                                  and norm[0].dtype == torch.bfloat16
+                                  and K * 2 <= 32 * 1024))
             and all(w.dtype == torch.bfloat16 and w.is_contiguous()
                     and (w.shape[1] & 7) == 0 for w in weights)):
-        ys = actuation._C.gemv_multi_bf16(x.reshape(-1).contiguous(),
-                                          list(weights))
+        if norm is None:
+            ys = actuation._C.gemv_multi_bf16(x.reshape(-1).contiguous(),
+                                              list(weights))
+        else:
+            ys = actuation._C.gemv_multi_bf16(
+                x.reshape(-1).contiguous(), list(weights),
+                norm[0].reshape(-1), float(norm[1]))
         return [y.view(*x.shape[:-1], w.shape[0])
                 for y, w in zip(ys, weights)]
+    if norm is not None:
+        xf = x.float()
+        xf = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + norm[1])
+        x = (xf * norm[0].float()).to(x.dtype)
     return [F.linear(x, w) for w in weights]
 
 
