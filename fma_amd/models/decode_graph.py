@@ -28,7 +28,8 @@ import torch.nn.functional as F
 from fma_amd.ops.decode_ops import (available as _fused_available,
                                     fast_rmsnorm, fast_rope1,
                                     fast_silu_mul)
-from fma_amd.ops.linear import fast_linear
+from fma_amd.ops.linear import (fast_linear, fast_linear_multi,
+                                fast_linear_residual)
 
 from fma_amd.models.llama import LlamaModel, rmsnorm
 
@@ -96,9 +97,12 @@ class StaticDecoder:
         for li in range(cfg.num_layers):
             p = f"layers.{li}."
             h = fast_rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps)
-            q = fast_linear(h, P[p + "wq.weight"]).view(B, 1, q_heads, hd)
-            k = fast_linear(h, P[p + "wk.weight"]).view(B, 1, kv_heads, hd)
-            v = fast_linear(h, P[p + "wv.weight"]).view(B, 1, kv_heads, hd)
+            q, k, v = fast_linear_multi(
+                h, (P[p + "wq.weight"], P[p + "wk.weight"],
+                    P[p + "wv.weight"]))
+            q = q.view(B, 1, q_heads, hd)
+            k = k.view(B, 1, kv_heads, hd)
+            v = v.view(B, 1, kv_heads, hd)
             if q.is_cuda and q.dtype == torch.bfloat16:
                 q = fast_rope1(q, cos[0], sin[0])
                 k = fast_rope1(k, cos[0], sin[0])
@@ -125,12 +129,21 @@ class StaticDecoder:
                 att = F.scaled_dot_product_attention(
                     q.transpose(1, 2), kh, vh, attn_mask=mask.to(q.dtype))
                 att = att.transpose(1, 2).reshape(B, 1, q_heads * hd)
-            x = x + fast_linear(att, P[p + "wo.weight"])
-            h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
-            gate = fast_linear(h, P[p + "w_gate.weight"])
-            up = fast_linear(h, P[p + "w_up.weight"])
-            x = x + fast_linear(fast_silu_mul(gate, up),
-                                P[p + "w_down.weight"])
+            if x.is_cuda and x.dtype == torch.bfloat16 \
+                    and x.numel() == x.shape[-1]:
+                x = fast_linear_residual(att, P[p + "wo.weight"], x)
+                h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
+                gate, up = fast_linear_multi(
+                    h, (P[p + "w_gate.weight"], P[p + "w_up.weight"]))
+                x = fast_linear_residual(fast_silu_mul(gate, up),
+                                         P[p + "w_down.weight"], x)
+            else:
+                x = x + fast_linear(att, P[p + "wo.weight"])
+                h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
+                gate = fast_linear(h, P[p + "w_gate.weight"])
+                up = fast_linear(h, P[p + "w_up.weight"])
+                x = x + fast_linear(fast_silu_mul(gate, up),
+                                    P[p + "w_down.weight"])
 
         x = fast_rmsnorm(x, P["final_norm.weight"], cfg.norm_eps)
         logits = fast_linear(x[:, 0], P["lm_head.weight"]).float()
