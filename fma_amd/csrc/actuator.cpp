@@ -1229,6 +1229,48 @@ at::Tensor silu_mul_bf16(const at::Tensor& g, const at::Tensor& u) {
 }
 
 // In-place single-position RoPE: q viewed as [heads, hd]; cos/sin one row.
+void rope_qkv_store_bf16_(at::Tensor& q, const at::Tensor& k,
+                          const at::Tensor& v, at::Tensor& kcache,
+                          at::Tensor& vcache, const at::Tensor& cos_tab,
+                          const at::Tensor& sin_tab,
+                          const c10::optional<at::Tensor>& pos_dev,
+                          int64_t pos) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 &&
+              q.is_contiguous() && q.dim() == 2, "q must be bf16 [qH, hd]");
+  TORCH_CHECK(k.is_contiguous() && v.is_contiguous() &&
+              k.scalar_type() == at::kBFloat16 &&
+              v.scalar_type() == at::kBFloat16 && k.dim() == 2 &&
+              v.sizes() == k.sizes(), "k/v must be bf16 [kvH, hd]");
+  TORCH_CHECK(kcache.is_contiguous() && vcache.is_contiguous() &&
+              kcache.dim() == 3 && kcache.sizes() == vcache.sizes() &&
+              kcache.size(1) == k.size(0) && kcache.size(2) == k.size(1),
+              "caches must be contiguous [S, kvH, hd]");
+  const int64_t hd = q.size(1);
+  TORCH_CHECK((hd & 1) == 0 && k.size(1) == hd, "hd must be even");
+  TORCH_CHECK(cos_tab.scalar_type() == at::kFloat &&
+              sin_tab.scalar_type() == at::kFloat &&
+              cos_tab.is_contiguous() && sin_tab.is_contiguous() &&
+              cos_tab.dim() == 2 && cos_tab.size(1) == hd / 2,
+              "cos/sin tables must be contiguous fp32 [S, hd/2]");
+  const int* pd = nullptr;
+  if (pos_dev.has_value()) {
+    const auto& t = pos_dev.value();
+    TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kInt &&
+                t.numel() == 1, "pos_dev must be a device int32 scalar");
+    pd = t.data_ptr<int>();
+  } else {
+    TORCH_CHECK(pos >= 0 && pos < kcache.size(0), "pos out of cache");
+    TORCH_CHECK(pos < cos_tab.size(0), "pos beyond rope table");
+  }
+  auto stream = c10::hip::getCurrentHIPStream(q.device().index());
+  FMA_HIP_CHECK(fma_launch_rope_qkv_store_bf16(
+      q.data_ptr(), k.data_ptr(), v.data_ptr(), kcache.data_ptr(),
+      vcache.data_ptr(), cos_tab.data_ptr<float>(),
+      sin_tab.data_ptr<float>(), pd, static_cast<int>(pos),
+      static_cast<int>(q.size(0)), static_cast<int>(k.size(0)),
+      static_cast<int>(hd / 2), stream.stream()));
+}
+
 at::Tensor& rope1_bf16_(at::Tensor& q, const at::Tensor& cos_row,
                         const at::Tensor& sin_row, int64_t heads,
                         int64_t head_dim) {
@@ -1415,6 +1457,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("v"), py::arg("pos0"), py::arg("chunks") = 0);
   m.def("attn_decode_bf16", &attn_decode_bf16, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("t"));
+  m.def("rope_qkv_store_bf16_", &rope_qkv_store_bf16_,
+        "RoPE(q) in place + RoPE(k)->cache row + v->cache row, one launch",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("kcache"),
+        py::arg("vcache"), py::arg("cos_tab"), py::arg("sin_tab"),
+        py::arg("pos_dev") = py::none(), py::arg("pos") = 0);
   m.def("rope1_bf16_", &rope1_bf16_, py::arg("q"), py::arg("cos_row"),
         py::arg("sin_row"), py::arg("heads"), py::arg("head_dim"));
   m.def("gather_d2d", &gather_d2d,

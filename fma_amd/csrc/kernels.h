@@ -52,6 +52,11 @@ extern "C" hipError_t fma_launch_rmsnorm1_bf16(const void* x, const void* w,
 extern "C" hipError_t fma_launch_silu_mul_bf16(const void* g, const void* u,
                                                void* y, int N,
                                                hipStream_t stream);
+extern "C" hipError_t fma_launch_rope_qkv_store_bf16(
+    void* q, const void* k, const void* v, void* kcache, void* vcache,
+    const float* cos_tab, const float* sin_tab, const int* pos_dev,
+    int pos, int q_heads, int kv_heads, int half_hd, hipStream_t stream);
+
 extern "C" hipError_t fma_launch_rope1_bf16(void* q, const float* cos_row,
                                             const float* sin_row, int heads,
                                             int half_hd, hipStream_t stream);

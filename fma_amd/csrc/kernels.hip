@@ -585,6 +585,79 @@ extern "C" hipError_t fma_launch_rope1_bf16(void* q, const float* cos_row,
   return hipGetLastError();
 }
 
+namespace {
+
+// One launch for the whole decode attention pre-step: RoPE(q) in place,
+// RoPE(k) written into its KV-cache row, v copied into its row — folds
+// 4 tiny launch-bound kernels (2 rope + 2 index_copy) into one. The
+// position (cos/sin table row + cache row) is read from a device int32
+// so a captured hipGraph replays correctly as the cache grows.
+__global__ __launch_bounds__(256) void rope_qkv_store_bf16_kernel(
+    unsigned short* __restrict__ q,        // [qH, hd], roped in place
+    const unsigned short* __restrict__ k,  // [kvH, hd]
+    const unsigned short* __restrict__ v,  // [kvH, hd]
+    unsigned short* __restrict__ kcache,   // [S, kvH, hd] base
+    unsigned short* __restrict__ vcache,
+    const float* __restrict__ cos_tab,     // [S, hd/2]
+    const float* __restrict__ sin_tab,
+    const int* __restrict__ pos_dev,       // device position (or null)
+    int pos_arg, int q_heads, int kv_heads, int half_hd) {
+  const int pos = pos_dev != nullptr ? *pos_dev : pos_arg;
+  const float* cos_row = cos_tab + static_cast<long long>(pos) * half_hd;
+  const float* sin_row = sin_tab + static_cast<long long>(pos) * half_hd;
+  const long long row = static_cast<long long>(pos) * kv_heads *
+                        (2 * half_hd);
+  unsigned short* kdst = kcache + row;
+  unsigned short* vdst = vcache + row;
+  const int qtotal = q_heads * half_hd;
+  const int kvtotal = kv_heads * half_hd;
+  const int stride = gridDim.x * blockDim.x;
+  for (int p = blockIdx.x * blockDim.x + threadIdx.x;
+       p < qtotal + 2 * kvtotal; p += stride) {
+    if (p < qtotal) {
+      const int d = p % half_hd;
+      const float c = cos_row[d];
+      const float sn = sin_row[d];
+      const float x0 = bf16_to_f32(q[2 * p]);
+      const float x1 = bf16_to_f32(q[2 * p + 1]);
+      q[2 * p] = f32_to_bf16(x0 * c - x1 * sn);
+      q[2 * p + 1] = f32_to_bf16(x0 * sn + x1 * c);
+    } else if (p < qtotal + kvtotal) {
+      const int pk = p - qtotal;
+      const int d = pk % half_hd;
+      const float c = cos_row[d];
+      const float sn = sin_row[d];
+      const float x0 = bf16_to_f32(k[2 * pk]);
+      const float x1 = bf16_to_f32(k[2 * pk + 1]);
+      kdst[2 * pk] = f32_to_bf16(x0 * c - x1 * sn);
+      kdst[2 * pk + 1] = f32_to_bf16(x0 * sn + x1 * c);
+    } else {
+      const int pv = p - qtotal - kvtotal;
+      reinterpret_cast<uint*>(vdst)[pv] =
+          reinterpret_cast<const uint*>(v)[pv];
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" hipError_t fma_launch_rope_qkv_store_bf16(
+    void* q, const void* k, const void* v, void* kcache, void* vcache,
+    const float* cos_tab, const float* sin_tab, const int* pos_dev,
+    int pos, int q_heads, int kv_heads, int half_hd, hipStream_t stream) {
+  const int total = (q_heads + 2 * kv_heads) * half_hd;
+  int blocks = (total + 255) / 256;
+  if (blocks > 512) blocks = 512;
+  rope_qkv_store_bf16_kernel<<<blocks, 256, 0, stream>>>(
+      static_cast<unsigned short*>(q),
+      static_cast<const unsigned short*>(k),
+      static_cast<const unsigned short*>(v),
+      static_cast<unsigned short*>(kcache),
+      static_cast<unsigned short*>(vcache), cos_tab, sin_tab, pos_dev,
+      pos, q_heads, kv_heads, half_hd);
+  return hipGetLastError();
+}
+
 // ---------------------------------------------------------------------------
 // Single-token GQA decode attention (flash-style online softmax).
 //

@@ -26,6 +26,7 @@ import torch
 import torch.nn.functional as F
 
 from fma_amd.ops.decode_ops import (available as _fused_available,
+                                    fast_rope_qkv_store,
                                     fast_rmsnorm, fast_rope1,
                                     fast_silu_mul)
 from fma_amd.ops.linear import (fast_linear, fast_linear_multi,
@@ -59,6 +60,7 @@ class StaticDecoder:
         # runtime sequence length for the graph-safe attention kernel
         # (device int32 read by the kernel itself at replay time)
         self.t_i32 = torch.ones((1,), dtype=torch.int32, device=dev)
+        self.pos_i32 = torch.zeros((1,), dtype=torch.int32, device=dev)
         self.graph: Optional[torch.cuda.CUDAGraph] = None
 
     # -- the static step (graph-capturable) --------------------------------
@@ -87,6 +89,7 @@ class StaticDecoder:
                       and q_heads % kv_heads == 0 and _fused_available())
         if fused_attn:
             self.t_i32.copy_((self.pos + 1).to(torch.int32))
+        self.pos_i32.copy_(self.pos.to(torch.int32).view(1))
         # additive mask over the full window: position j attends iff j <= pos
         neg = -1e9  # large finite: stays finite in bf16
         mask = torch.where(self.positions <= self.pos,
@@ -103,15 +106,24 @@ class StaticDecoder:
             q = q.view(B, 1, q_heads, hd)
             k = k.view(B, 1, kv_heads, hd)
             v = v.view(B, 1, kv_heads, hd)
-            if q.is_cuda and q.dtype == torch.bfloat16:
-                q = fast_rope1(q, cos[0], sin[0])
-                k = fast_rope1(k, cos[0], sin[0])
-            else:
-                q = _rope1(q, cos, sin)
-                k = _rope1(k, cos, sin)
-            # static cache write at pos
-            self.cache[li, 0].index_copy_(1, self.pos.view(1), k)
-            self.cache[li, 1].index_copy_(1, self.pos.view(1), v)
+            stored = False
+            if B == 1 and q.is_cuda and q.dtype == torch.bfloat16:
+                # one graph-safe launch: RoPE(q) + RoPE(k)->cache +
+                # v->cache, position read from the device scalar at
+                # replay time
+                stored = fast_rope_qkv_store(
+                    q, k, v, self.cache[li, 0, 0], self.cache[li, 1, 0],
+                    m.rope_cos, m.rope_sin, pos_dev=self.pos_i32)
+            if not stored:
+                if q.is_cuda and q.dtype == torch.bfloat16:
+                    q = fast_rope1(q, cos[0], sin[0])
+                    k = fast_rope1(k, cos[0], sin[0])
+                else:
+                    q = _rope1(q, cos, sin)
+                    k = _rope1(k, cos, sin)
+                # static cache write at pos
+                self.cache[li, 0].index_copy_(1, self.pos.view(1), k)
+                self.cache[li, 1].index_copy_(1, self.pos.view(1), v)
             if fused_attn:
                 from fma_amd.ops import actuation
                 att = actuation._C.attn_decode_bf16_graph(

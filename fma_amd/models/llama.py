@@ -28,7 +28,8 @@ import torch.distributed as dist
 import torch.nn.functional as F
 
 from fma_amd.ops.decode_ops import (fast_attn_decode, fast_attn_prefill,
-                                    fast_rmsnorm, fast_rope1, fast_silu_mul)
+                                    fast_rmsnorm, fast_rope1,
+                                    fast_rope_qkv_store, fast_silu_mul)
 from fma_amd.ops.linear import (fast_linear, fast_linear_multi,
                                 fast_linear_residual)
 
@@ -278,17 +279,27 @@ class LlamaModel:
             q = q.view(B, T, q_heads, hd)
             k = k.view(B, T, kv_heads, hd)
             v = v.view(B, T, kv_heads, hd)
-            if decode1:
-                q = fast_rope1(q, self.rope_cos[start_pos],
-                               self.rope_sin[start_pos])
-                k = fast_rope1(k, self.rope_cos[start_pos],
-                               self.rope_sin[start_pos])
-            else:
-                q = apply_rope(q, self.rope_cos, self.rope_sin, start_pos)
-                k = apply_rope(k, self.rope_cos, self.rope_sin, start_pos)
+            stored = False
+            if decode1 and cache is not None:
+                # RoPE(q) + RoPE(k)->cache + v->cache in ONE launch
+                stored = fast_rope_qkv_store(
+                    q, k, v, cache.data[li, 0, 0], cache.data[li, 1, 0],
+                    self.rope_cos, self.rope_sin, pos=start_pos)
+            if not stored:
+                if decode1:
+                    q = fast_rope1(q, self.rope_cos[start_pos],
+                                   self.rope_sin[start_pos])
+                    k = fast_rope1(k, self.rope_cos[start_pos],
+                                   self.rope_sin[start_pos])
+                else:
+                    q = apply_rope(q, self.rope_cos, self.rope_sin,
+                                   start_pos)
+                    k = apply_rope(k, self.rope_cos, self.rope_sin,
+                                   start_pos)
             if cache is not None:
-                cache.data[li, 0, :, start_pos:start_pos + T] = k
-                cache.data[li, 1, :, start_pos:start_pos + T] = v
+                if not stored:
+                    cache.data[li, 0, :, start_pos:start_pos + T] = k
+                    cache.data[li, 1, :, start_pos:start_pos + T] = v
                 k = cache.data[li, 0, :, : start_pos + T]
                 v = cache.data[li, 1, :, : start_pos + T]
             att = None

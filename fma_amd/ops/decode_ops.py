@@ -66,6 +66,27 @@ def fast_rope1(x: torch.Tensor, cos_row: torch.Tensor, sin_row: torch.Tensor
     return out.view(B, T, heads, hd).to(x.dtype)
 
 
+def fast_rope_qkv_store(q, k, v, kcache, vcache, cos_tab, sin_tab,
+                        pos=None, pos_dev=None) -> bool:
+    """RoPE(q in place) + RoPE(k)->cache + v->cache in ONE launch
+    (replaces 2 rope + 2 cache-write kernels on the decode path).
+    q [1,1,qH,hd]; k/v [1,1,kvH,hd]; k/vcache [S,kvH,hd]. Returns False
+    when the fused kernel does not apply (caller uses the unfused ops)."""
+    if not available():
+        return False
+    if not (q.is_cuda and q.dtype == torch.bfloat16 and q.is_contiguous()
+            and k.is_contiguous() and v.is_contiguous()
+            and kcache.is_contiguous() and vcache.is_contiguous()):
+        return False
+    qh, hd = q.shape[-2], q.shape[-1]
+    kvh = k.shape[-2]
+    actuation._C.rope_qkv_store_bf16_(
+        q.reshape(qh, hd), k.reshape(kvh, hd), v.reshape(kvh, hd),
+        kcache, vcache, cos_tab, sin_tab, pos_dev,
+        0 if pos is None else int(pos))
+    return True
+
+
 def fast_attn_prefill(q: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, pos0: int):
     """Causal GQA prefill on MFMA matrix cores. q [1,T,qH,hd]; k/v_cache

@@ -612,3 +612,42 @@ def test_decode1_qkv_fusion_token_equality(C):
         os.environ.pop("FMA_DISABLE_GEMV")
         lin._ENABLED = None
     assert torch.equal(fused, eager)
+
+
+def test_rope_qkv_store_matches_unfused(C):
+    """Fused RoPE(q)+RoPE(k)->cache+v->cache vs the separate fp32
+    reference ops, host-pos and device-pos (hipGraph) variants."""
+    torch.manual_seed(9)
+    qh, kvh, hd, S = 8, 2, 128, 64
+    cos = torch.randn(S, hd // 2, device="cuda")
+    sin = torch.randn(S, hd // 2, device="cuda")
+
+    def rope_ref(x, pos):
+        xf = x.float().view(-1, hd // 2, 2)
+        c = cos[pos].view(1, -1)
+        s = sin[pos].view(1, -1)
+        x0, x1 = xf[..., 0], xf[..., 1]
+        return torch.stack((x0 * c - x1 * s, x0 * s + x1 * c),
+                           dim=-1).view(x.shape).to(torch.bfloat16)
+
+    for use_dev in (False, True):
+        pos = 7
+        q = torch.randn(qh, hd, dtype=torch.bfloat16, device="cuda")
+        k = torch.randn(kvh, hd, dtype=torch.bfloat16, device="cuda")
+        v = torch.randn(kvh, hd, dtype=torch.bfloat16, device="cuda")
+        kc = torch.zeros(S, kvh, hd, dtype=torch.bfloat16, device="cuda")
+        vc = torch.zeros_like(kc)
+        q_ref = rope_ref(q.clone(), pos)
+        k_ref = rope_ref(k.clone(), pos)
+        if use_dev:
+            pd = torch.tensor([pos], dtype=torch.int32, device="cuda")
+            C.rope_qkv_store_bf16_(q, k, v, kc, vc, cos, sin, pd, 0)
+        else:
+            C.rope_qkv_store_bf16_(q, k, v, kc, vc, cos, sin, None, pos)
+        torch.cuda.synchronize()
+        assert torch.allclose(q.float(), q_ref.float(), atol=0.02,
+                              rtol=0.02)
+        assert torch.allclose(kc[pos].float(), k_ref.float(), atol=0.02,
+                              rtol=0.02)
+        assert torch.equal(vc[pos], v)
+        assert kc.abs().sum() == kc[pos].abs().sum()  # only that row
