@@ -1229,6 +1229,32 @@ at::Tensor silu_mul_bf16(const at::Tensor& g, const at::Tensor& u) {
 }
 
 // In-place single-position RoPE: q viewed as [heads, hd]; cos/sin one row.
+at::Tensor gemv_silu_bf16(const at::Tensor& W, const at::Tensor& gate,
+                          const at::Tensor& up,
+                          const c10::optional<at::Tensor>& residual) {
+  TORCH_CHECK(W.is_cuda() && W.scalar_type() == at::kBFloat16 &&
+              W.dim() == 2 && W.is_contiguous(), "W must be bf16 [M,K]");
+  const int64_t M = W.size(0), K = W.size(1);
+  TORCH_CHECK(gate.scalar_type() == at::kBFloat16 && gate.numel() == K &&
+              gate.is_contiguous() && up.scalar_type() == at::kBFloat16 &&
+              up.numel() == K && up.is_contiguous(),
+              "gate/up must be contiguous bf16 [K]");
+  TORCH_CHECK((K & 7) == 0, "K must be a multiple of 8");
+  const void* rptr = nullptr;
+  if (residual.has_value()) {
+    const auto& r = residual.value();
+    TORCH_CHECK(r.scalar_type() == at::kBFloat16 && r.numel() == M &&
+                r.is_contiguous(), "residual must be contiguous bf16 [M]");
+    rptr = r.data_ptr();
+  }
+  auto y = at::empty({M}, W.options());
+  auto stream = c10::hip::getCurrentHIPStream(W.device().index());
+  FMA_HIP_CHECK(fma_launch_gemv_silu_bf16_out16(
+      W.data_ptr(), gate.data_ptr(), up.data_ptr(), y.data_ptr(), rptr,
+      static_cast<int>(M), static_cast<int>(K), stream.stream()));
+  return y;
+}
+
 void rope_qkv_store_bf16_(at::Tensor& q, const at::Tensor& k,
                           const at::Tensor& v, at::Tensor& kcache,
                           at::Tensor& vcache, const at::Tensor& cos_tab,
@@ -1457,6 +1483,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("v"), py::arg("pos0"), py::arg("chunks") = 0);
   m.def("attn_decode_bf16", &attn_decode_bf16, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("t"));
+  m.def("gemv_silu_bf16", &gemv_silu_bf16,
+        "y = W @ (silu(gate)*up) [+ residual], activation fused into the "
+        "GEMV's input stage",
+        py::arg("W"), py::arg("gate"), py::arg("up"),
+        py::arg("residual") = py::none());
   m.def("rope_qkv_store_bf16_", &rope_qkv_store_bf16_,
         "RoPE(q) in place + RoPE(k)->cache row + v->cache row, one launch",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("kcache"),

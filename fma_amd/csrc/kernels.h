@@ -46,6 +46,10 @@ extern "C" hipError_t fma_launch_gemv_bf16_out16(const void* W, const void* x,
                                                  void* y, const void* residual,
                                                  int M, int K,
                                                  hipStream_t stream);
+extern "C" hipError_t fma_launch_gemv_silu_bf16_out16(
+    const void* W, const void* gate, const void* up, void* y,
+    const void* residual, int M, int K, hipStream_t stream);
+
 extern "C" hipError_t fma_launch_rmsnorm1_bf16(const void* x, const void* w,
                                                void* y, int H, float eps,
                                                hipStream_t stream);

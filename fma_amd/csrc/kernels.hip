@@ -217,13 +217,18 @@ __device__ inline void gemv_store(unsigned short* y, int row, float acc) {
   y[row] = f32_to_bf16(acc);
 }
 
-template <bool kUseLds, typename OutT>
+// kSilu: x is given as the PAIR (g, u); the GEMV input is
+// silu(g)*u, computed during the LDS stage (or on the fly when x does
+// not fit LDS) with exactly silu_mul_bf16_kernel's fp32 math — the
+// separate activation launch disappears.
+template <bool kUseLds, typename OutT, bool kSilu = false>
 __global__ __launch_bounds__(256) void gemv_bf16_kernel(
     const unsigned short* __restrict__ W,  // [M, K] row-major bf16
-    const unsigned short* __restrict__ x,  // [K] bf16
+    const unsigned short* __restrict__ x,  // [K] bf16 (kSilu: gate)
     OutT* __restrict__ y,                  // [M] fp32 or bf16
     const unsigned short* __restrict__ r,  // optional residual [M] bf16
-    int M, int K) {
+    int M, int K,
+    const unsigned short* __restrict__ x2 = nullptr) {  // kSilu: up
   // x staged in LDS when it fits without hurting occupancy; for wide K
   // (w_down shapes) every wave reads the same x slices, which the L2
   // broadcasts — 57 KB of LDS would cap residency at 2 blocks/CU and
@@ -232,8 +237,26 @@ __global__ __launch_bounds__(256) void gemv_bf16_kernel(
   const unsigned short* xsrc = x;
   if (kUseLds) {
     for (int i = threadIdx.x; i * 8 < K; i += blockDim.x) {
-      reinterpret_cast<uint4*>(s_x)[i] =
-          reinterpret_cast<const uint4*>(x)[i];
+      if (kSilu) {
+        const uint4 gv = reinterpret_cast<const uint4*>(x)[i];
+        const uint4 uv = reinterpret_cast<const uint4*>(x2)[i];
+        const unsigned short* gh =
+            reinterpret_cast<const unsigned short*>(&gv);
+        const unsigned short* uh =
+            reinterpret_cast<const unsigned short*>(&uv);
+        uint4 ov;
+        unsigned short* oh = reinterpret_cast<unsigned short*>(&ov);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float gf = bf16_to_f32(gh[j]);
+          const float sg = gf / (1.0f + __expf(-gf));
+          oh[j] = f32_to_bf16(sg * bf16_to_f32(uh[j]));
+        }
+        reinterpret_cast<uint4*>(s_x)[i] = ov;
+      } else {
+        reinterpret_cast<uint4*>(s_x)[i] =
+            reinterpret_cast<const uint4*>(x)[i];
+      }
     }
     __syncthreads();
     xsrc = s_x;
@@ -255,9 +278,23 @@ __global__ __launch_bounds__(256) void gemv_bf16_kernel(
       const uint4 xv = reinterpret_cast<const uint4*>(xsrc)[i];
       const unsigned short* wh = reinterpret_cast<const unsigned short*>(&wv);
       const unsigned short* xh = reinterpret_cast<const unsigned short*>(&xv);
+      if (kSilu && !kUseLds) {
+        const uint4 uv = reinterpret_cast<const uint4*>(x2)[i];
+        const unsigned short* uh =
+            reinterpret_cast<const unsigned short*>(&uv);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        acc = fmaf(bf16_to_f32(wh[j]), bf16_to_f32(xh[j]), acc);
+        for (int j = 0; j < 8; ++j) {
+          const float gf = bf16_to_f32(xh[j]);
+          const float sg = gf / (1.0f + __expf(-gf));
+          const float xv2 = bf16_to_f32(
+              f32_to_bf16(sg * bf16_to_f32(uh[j])));
+          acc = fmaf(bf16_to_f32(wh[j]), xv2, acc);
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          acc = fmaf(bf16_to_f32(wh[j]), bf16_to_f32(xh[j]), acc);
+        }
       }
     }
     // wave-wide reduction
@@ -546,6 +583,35 @@ extern "C" hipError_t fma_launch_gemv_bf16_out16(const void* W, const void* x,
         static_cast<const unsigned short*>(x),
         static_cast<unsigned short*>(y),
         static_cast<const unsigned short*>(residual), M, K);
+  }
+  return hipGetLastError();
+}
+
+extern "C" hipError_t fma_launch_gemv_silu_bf16_out16(
+    const void* W, const void* gate, const void* up, void* y,
+    const void* residual, int M, int K, hipStream_t stream) {
+  if ((K & 7) != 0) return hipErrorInvalidValue;
+  const int block = 256;
+  const int waves_per_block = block / 64;
+  int blocks = (M + waves_per_block - 1) / waves_per_block;
+  if (blocks > 2048) blocks = 2048;
+  const size_t lds = static_cast<size_t>(K) * sizeof(unsigned short);
+  if (lds <= 32 * 1024) {
+    gemv_bf16_kernel<true, unsigned short, true>
+        <<<blocks, block, lds, stream>>>(
+            static_cast<const unsigned short*>(W),
+            static_cast<const unsigned short*>(gate),
+            static_cast<unsigned short*>(y),
+            static_cast<const unsigned short*>(residual), M, K,
+            static_cast<const unsigned short*>(up));
+  } else {
+    gemv_bf16_kernel<false, unsigned short, true>
+        <<<blocks, block, 0, stream>>>(
+            static_cast<const unsigned short*>(W),
+            static_cast<const unsigned short*>(gate),
+            static_cast<unsigned short*>(y),
+            static_cast<const unsigned short*>(residual), M, K,
+            static_cast<const unsigned short*>(up));
   }
   return hipGetLastError();
 }

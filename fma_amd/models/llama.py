@@ -30,8 +30,8 @@ import torch.nn.functional as F
 from fma_amd.ops.decode_ops import (fast_attn_decode, fast_attn_prefill,
                                     fast_rmsnorm, fast_rope1,
                                     fast_rope_qkv_store, fast_silu_mul)
-from fma_amd.ops.linear import (fast_linear, fast_linear_multi,
-                                fast_linear_residual)
+from fma_amd.ops.linear import (fast_down_proj, fast_linear,
+                                fast_linear_multi, fast_linear_residual)
 
 
 @dataclass
@@ -344,6 +344,12 @@ class LlamaModel:
                 h = rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps)
                 gate = fast_linear(h, P[p + "w_gate.weight"])
                 up = fast_linear(h, P[p + "w_up.weight"])
+            # NOTE: a silu-fused w_down GEMV exists (gemv_silu_bf16 /
+            # fast_down_proj) but measured SLOWER: every GEMV block
+            # recomputes silu over its x stage (239.6 -> 228 tok/s), and
+            # on wide-K models whose x exceeds the LDS stage it
+            # recomputes per dot element (70B-shape 74.7 -> 60.5).
+            # Selection is measured; the separate activation launch wins.
             act = fast_silu_mul(gate, up) if decode1 else F.silu(gate) * up
             if decode1 and self.tp_size == 1:
                 x = fast_linear_residual(act, P[p + "w_down.weight"], x)

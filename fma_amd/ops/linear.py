@@ -71,6 +71,23 @@ def fast_linear_multi(x, weights, norm=None):
     return [F.linear(x, w) for w in weights]
 
 
+def fast_down_proj(gate, up, w, residual):
+    """residual + F.linear(silu(gate)*up, w) with the activation fused
+    into the GEMV input stage (one launch for the whole MLP tail)."""
+    if (gemv_enabled() and gate.is_cuda
+            and gate.dtype == torch.bfloat16
+            and gate.numel() == gate.shape[-1]
+            and up.is_contiguous()
+            and residual.numel() == w.shape[0]
+            and (w.shape[1] & 7) == 0 and w.is_contiguous()):
+        y = actuation._C.gemv_silu_bf16(
+            w, gate.reshape(-1).contiguous(), up.reshape(-1).contiguous(),
+            residual.reshape(-1).contiguous())
+        return y.view(residual.shape)
+    act = F.silu(gate) * up
+    return residual + F.linear(act, w)
+
+
 def fast_linear_residual(x: torch.Tensor, w: torch.Tensor,
                          residual: torch.Tensor) -> torch.Tensor:
     """residual + F.linear(x, w) with the add fused into the GEMV store."""

@@ -651,3 +651,19 @@ def test_rope_qkv_store_matches_unfused(C):
                               rtol=0.02)
         assert torch.equal(vc[pos], v)
         assert kc.abs().sum() == kc[pos].abs().sum()  # only that row
+
+
+def test_gemv_silu_matches_unfused(C):
+    """silu-fused GEMV vs eager silu+linear+residual fp32 reference."""
+    torch.manual_seed(12)
+    M, K = 512, 2048
+    w = torch.randn(M, K, dtype=torch.bfloat16, device="cuda")
+    g = torch.randn(K, dtype=torch.bfloat16, device="cuda")
+    u = torch.randn(K, dtype=torch.bfloat16, device="cuda")
+    r = torch.randn(M, dtype=torch.bfloat16, device="cuda")
+    y = C.gemv_silu_bf16(w, g, u, r)
+    gf = g.float()
+    act = (gf / (1 + torch.exp(-gf)) * u.float()).to(torch.bfloat16)
+    ref = (w.float() @ act.float() + r.float())
+    assert torch.allclose(y.float(), ref, atol=0.2, rtol=0.05), \
+        (y.float() - ref).abs().max()
