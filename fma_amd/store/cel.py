@@ -147,8 +147,6 @@ class _Parser:
             if op == "==":
                 return lambda env: _unwrap(left(env)) == _unwrap(right(env))
             return lambda env: _unwrap(left(env)) != _unwrap(right(env))
-        if op == "in":  # pragma: no cover - ident 'in' handled below
-            pass
         if self.peek() == ("ident", "in"):
             self.next()
             right = self.unary_expr()

@@ -296,6 +296,9 @@ class KubeStore:
                         "GET", self.base + path,
                         params={"watch": "1", "resourceVersion": cursor},
                         timeout=httpx.Timeout(5.0, read=None)) as r:
+                    if r.status_code == 410:
+                        out.put(RevisionTooOld(f"{kind} watch expired"))
+                        return
                     if r.status_code != 200:
                         out.put(ApiError(r.status_code, f"watch {kind}"))
                         return
@@ -311,8 +314,9 @@ class KubeStore:
                                 out.put(RevisionTooOld(
                                     f"{kind} watch expired"))
                                 return
-                            out.put(ApiError(code or 500, str(ev)))
-                            return
+                            # transient server-side error: reconnect from
+                            # the cursor like any dropped stream
+                            break
                         obj = ev["object"]
                         new_rv = ob.rv_of(obj)
                         if new_rv:
