@@ -150,7 +150,14 @@ class _Parser:
         if self.peek() == ("ident", "in"):
             self.next()
             right = self.unary_expr()
-            return lambda env: _unwrap(left(env)) in _unwrap(right(env))
+
+            def _in(env):
+                a, b = _unwrap(left(env)), _unwrap(right(env))
+                try:
+                    return a in b
+                except TypeError as e:
+                    raise CelError(f"'in' type error: {e}") from e
+            return _in
         return left
 
     def unary_expr(self):
@@ -311,7 +318,10 @@ def _get_index(obj: Any, key: Any) -> Any:
             return obj[key]
         raise CelError(f"no such key {key!r}")
     if isinstance(obj, list):
-        return obj[key]
+        try:
+            return obj[key]
+        except (TypeError, IndexError) as e:
+            raise CelError(f"index error: {e}") from e
     raise CelError(f"cannot index {type(obj).__name__}")
 
 
@@ -323,16 +333,19 @@ def _method(recv: Any, name: str, args: List[Any]) -> Any:
     if name == "optMap":  # pragma: no cover - not used by our policies
         raise CelError("optMap unsupported")
     recv = _unwrap(recv)
-    if name == "matches":
-        return re.search(args[0], recv) is not None
-    if name == "startsWith":
-        return recv.startswith(args[0])
-    if name == "endsWith":
-        return recv.endswith(args[0])
-    if name == "contains":
-        return args[0] in recv
-    if name == "size":
-        return len(recv)
+    try:
+        if name == "matches":
+            return re.search(args[0], recv) is not None
+        if name == "startsWith":
+            return recv.startswith(args[0])
+        if name == "endsWith":
+            return recv.endswith(args[0])
+        if name == "contains":
+            return args[0] in recv
+        if name == "size":
+            return len(recv)
+    except (TypeError, AttributeError, re.error) as e:
+        raise CelError(f"method {name!r} type error: {e}") from e
     raise CelError(f"unknown method {name!r}")
 
 

@@ -72,3 +72,57 @@ def test_shipped_policy_files_compile():
     assert set(pols.policies) == {"fma-immutable-fields",
                                   "fma-bound-serverreqpod"}
     assert len(pols.bound) == 2
+
+
+def test_cel_fuzz_never_crashes_unexpectedly():
+    """Property: arbitrary token soup either evaluates or raises CelError
+    — never an uncontrolled exception (hypothesis-style fuzz without a
+    GPU-worth of runtime)."""
+    import random
+
+    from fma_amd.store.cel import CelError
+
+    random.seed(42)
+    atoms = ["object", "oldObject", "request", ".", ".?", "(", ")", "[",
+             "]", "'k'", '"v"', "==", "!=", "||", "&&", "!", "orValue",
+             "matches", "has", "metadata", "annotations", "labels", "in",
+             "true", "false", "null", "1", ",", "size"]
+    env = {"object": {"metadata": {"labels": {"k": "v"}}},
+           "oldObject": {"metadata": {}},
+           "request": {"userInfo": {"username": "u"}}}
+    crashes = 0
+    for _ in range(3000):
+        expr = " ".join(random.choice(atoms)
+                        for _ in range(random.randint(1, 12)))
+        try:
+            evaluate(expr, env)
+        except CelError:
+            pass
+        except RecursionError:
+            pass
+        except Exception as e:  # noqa: BLE001
+            crashes += 1
+            if crashes <= 3:
+                print(f"UNEXPECTED {type(e).__name__} for {expr!r}: {e}")
+    assert crashes == 0
+
+
+def test_cel_valid_expression_fuzz_consistent():
+    """Property: generated WELL-FORMED policy-style expressions evaluate
+    to the same result as an equivalent Python evaluation."""
+    import random
+
+    random.seed(7)
+    keys = ["a", "b", "c"]
+    for _ in range(500):
+        ann = {k: random.choice(["", "x", "y"])
+               for k in random.sample(keys, random.randint(0, 3))}
+        old_ann = {k: random.choice(["", "x", "y"])
+                   for k in random.sample(keys, random.randint(0, 3))}
+        k = random.choice(keys)
+        expr = (f"oldObject.metadata.?annotations['{k}'].orValue('') == "
+                f"object.metadata.?annotations['{k}'].orValue('')")
+        env = {"object": {"metadata": {"annotations": ann}},
+               "oldObject": {"metadata": {"annotations": old_ann}}}
+        expect = old_ann.get(k, "") == ann.get(k, "")
+        assert evaluate(expr, env) is expect, (expr, env)

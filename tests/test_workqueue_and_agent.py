@@ -386,3 +386,59 @@ def test_index_registered_late_backfills():
                  if ob.annotations_of(p).get(C.NOMINAL_ANNOTATION) else [])
     assert [ob.name_of(p) for p in
             st.index_get("Pod", "nominal", "h1")] == ["early"]
+
+
+def test_two_level_queue_no_item_lost_fuzz():
+    """Property: under random add/add_after/add_rate_limited from several
+    threads, every item is eventually processed at least once."""
+    import random
+    import threading
+    import time as _t
+
+    from fma_amd.controller.workqueue import NodeQueueAndWorkers
+
+    processed = set()
+    mu = threading.Lock()
+
+    def process(item):
+        with mu:
+            processed.add(item)
+        return False
+
+    w = NodeQueueAndWorkers("fuzz", 3, process, node_of=lambda it: it[0])
+    w.start()
+    all_items = []
+    try:
+        rng = random.Random(11)
+
+        def feeder(tid):
+            r = random.Random(tid)
+            for i in range(120):
+                item = (f"n{r.randint(0, 5)}", tid, i)
+                with mu:
+                    all_items.append(item)
+                mode = r.random()
+                if mode < 0.6:
+                    w.queue.add(item)
+                elif mode < 0.85:
+                    w.queue.add_after(item, r.random() * 0.05)
+                else:
+                    w.queue.add_rate_limited(item)
+
+        threads = [threading.Thread(target=feeder, args=(t,))
+                   for t in range(4)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        deadline = _t.time() + 20
+        while _t.time() < deadline:
+            with mu:
+                if set(all_items) <= processed:
+                    break
+            _t.sleep(0.05)
+        with mu:
+            missing = set(all_items) - processed
+        assert not missing, f"{len(missing)} items never processed"
+    finally:
+        w.stop()
