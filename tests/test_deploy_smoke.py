@@ -117,3 +117,82 @@ def test_ci_workflow_lists_real_paths():
     assert {"cpu-tests", "kube-e2e", "hip-build", "images"} <= set(jobs)
     for df in jobs["images"]["strategy"]["matrix"]["dockerfile"]:
         assert os.path.exists(os.path.join(root, df)), df
+
+
+def _render_helm_subset(chart_dir, values):
+    """Minimal renderer for the template subset the chart uses:
+    {{ .Values.a.b }}, {{- if .Values.a.b }} / {{- end }},
+    {{ .Files.Get "path" }}. Enough to validate the shipped chart
+    renders to parseable Kubernetes YAML without helm in the image."""
+    import os
+    import re
+
+    def lookup(path):
+        cur = values
+        for part in path.split(".")[2:]:  # strip leading .Values
+            cur = cur[part]
+        return cur
+
+    rendered = {}
+    tdir = os.path.join(chart_dir, "templates")
+    for fn in sorted(os.listdir(tdir)):
+        out_lines = []
+        stack = [True]
+        for line in open(os.path.join(tdir, fn)):
+            m = re.match(r"\s*\{\{-? if (\.Values[.\w]+) \}\}", line)
+            if m:
+                stack.append(stack[-1] and bool(lookup(m.group(1))))
+                continue
+            if re.match(r"\s*\{\{-? end \}\}", line):
+                stack.pop()
+                continue
+            if not stack[-1]:
+                continue
+            line = re.sub(
+                r"\{\{ \.Files\.Get \"([^\"]+)\" \}\}",
+                lambda m: open(os.path.join(chart_dir, m.group(1))).read(),
+                line)
+            line = re.sub(r"\{\{ (\.Values[.\w]+) \}\}",
+                          lambda m: str(lookup(m.group(1))), line)
+            out_lines.append(line)
+        rendered[fn] = "".join(out_lines)
+    return rendered
+
+
+def test_helm_chart_renders_to_valid_kubernetes_yaml():
+    import os
+
+    import yaml
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    chart = os.path.join(root, "deploy", "charts", "fma-amd")
+    values = yaml.safe_load(open(os.path.join(chart, "values.yaml")))
+    assert yaml.safe_load(open(os.path.join(chart, "Chart.yaml")))["name"] \
+        == "fma-amd"
+
+    rendered = _render_helm_subset(chart, values)
+    kinds = []
+    for fn, text in rendered.items():
+        for doc in yaml.safe_load_all(text):
+            if doc:
+                kinds.append(doc["kind"])
+                assert "metadata" in doc, fn
+    assert "Deployment" in kinds and kinds.count("Deployment") == 2
+    assert "ServiceAccount" in kinds and "ClusterRole" in kinds
+    assert kinds.count("ValidatingAdmissionPolicy") == 2
+    assert kinds.count("ValidatingAdmissionPolicyBinding") == 2
+
+    # the ServiceAccount name must satisfy the VAP exemption pattern
+    import re as _re
+    sa = next(d for d in yaml.safe_load_all(rendered["rbac.yaml"])
+              if d and d["kind"] == "ServiceAccount")
+    username = (f"system:serviceaccount:{values['namespace']}:"
+                f"{sa['metadata']['name']}")
+    assert _re.match(
+        r"^system:serviceaccount:[^:]+:[^:]*-fma-controllers$", username)
+
+    # toggles prune their sections
+    values2 = dict(values, admissionPolicies={"enabled": False})
+    rendered2 = _render_helm_subset(chart, values2)
+    assert not any(d for d in yaml.safe_load_all(
+        rendered2["admission-policies.yaml"]) if d)
