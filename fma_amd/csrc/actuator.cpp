@@ -1157,7 +1157,8 @@ at::Tensor gemv_bf16(const at::Tensor& W, const at::Tensor& x,
 
 std::vector<at::Tensor> gemv_multi_bf16(
     const at::Tensor& x, std::vector<at::Tensor> ws,
-    const c10::optional<at::Tensor>& norm_w, double norm_eps) {
+    const c10::optional<at::Tensor>& norm_w, double norm_eps,
+    const c10::optional<std::vector<at::Tensor>>& biases) {
   TORCH_CHECK(!ws.empty() && ws.size() <= 3,
               "gemv_multi_bf16 takes 1-3 weight matrices");
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
@@ -1183,6 +1184,19 @@ std::vector<at::Tensor> gemv_multi_bf16(
   auto mdim = [&](size_t i) -> int {
     return ws[i].defined() ? static_cast<int>(ws[i].size(0)) : 0;
   };
+  const void* bptr[3] = {nullptr, nullptr, nullptr};
+  if (biases.has_value()) {
+    const auto& bs = biases.value();
+    TORCH_CHECK(bs.size() == ys.size(),
+                "biases must match the number of weights");
+    for (size_t i = 0; i < bs.size(); ++i) {
+      TORCH_CHECK(bs[i].scalar_type() == at::kBFloat16 &&
+                  bs[i].is_contiguous() &&
+                  bs[i].numel() == ys[i].numel(),
+                  "bias ", i, " must be contiguous bf16 [M]");
+      bptr[i] = bs[i].data_ptr();
+    }
+  }
   const void* nw = nullptr;
   if (norm_w.has_value()) {
     const auto& n = norm_w.value();
@@ -1196,7 +1210,8 @@ std::vector<at::Tensor> gemv_multi_bf16(
   FMA_HIP_CHECK(fma_launch_gemv_multi_bf16(
       wptr(0), mdim(0), yptr(0), wptr(1), mdim(1), yptr(1), wptr(2),
       mdim(2), yptr(2), x.data_ptr(), static_cast<int>(K), nw,
-      static_cast<float>(norm_eps), stream.stream()));
+      static_cast<float>(norm_eps), bptr[0], bptr[1], bptr[2],
+      stream.stream()));
   return ys;
 }
 
@@ -1470,7 +1485,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "1-3 batch-1 bf16 GEMVs sharing one x in a single launch; "
         "optional fused rmsnorm of x (bit-identical to rmsnorm1_bf16)",
         py::arg("x"), py::arg("weights"), py::arg("norm_w") = py::none(),
-        py::arg("norm_eps") = 0.0);
+        py::arg("norm_eps") = 0.0, py::arg("biases") = py::none());
   m.def("gemv_bf16", &gemv_bf16, "Batch-1 bf16 GEMV",
         py::arg("W"), py::arg("x"), py::arg("out_bf16") = false,
         py::arg("residual") = py::none());

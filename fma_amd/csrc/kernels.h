@@ -40,7 +40,8 @@ extern "C" hipError_t fma_launch_gemv_bf16(const void* W, const void* x,
 extern "C" hipError_t fma_launch_gemv_multi_bf16(
     const void* W0, int M0, void* y0, const void* W1, int M1, void* y1,
     const void* W2, int M2, void* y2, const void* x, int K,
-    const void* norm_w, float norm_eps, hipStream_t stream);
+    const void* norm_w, float norm_eps, const void* b0, const void* b1,
+    const void* b2, hipStream_t stream);
 
 extern "C" hipError_t fma_launch_gemv_bf16_out16(const void* W, const void* x,
                                                  void* y, const void* residual,

@@ -351,7 +351,10 @@ __global__ __launch_bounds__(256) void gemv_multi_bf16_kernel(
     const unsigned short* __restrict__ W2, int M2,
     unsigned short* __restrict__ y2,
     const unsigned short* __restrict__ x, int K,
-    const unsigned short* __restrict__ norm_w, float norm_eps) {
+    const unsigned short* __restrict__ norm_w, float norm_eps,
+    const unsigned short* __restrict__ b0 = nullptr,  // optional biases
+    const unsigned short* __restrict__ b1 = nullptr,
+    const unsigned short* __restrict__ b2 = nullptr) {
   static_assert(!kNorm || kUseLds, "norm fusion needs the LDS path");
   extern __shared__ unsigned short s_x[];
   __shared__ float s_red[256];
@@ -403,14 +406,15 @@ __global__ __launch_bounds__(256) void gemv_multi_bf16_kernel(
   const int M = M0 + M1 + M2;
   for (int row = global_wave; row < M; row += total_waves) {
     const unsigned short* W;
+    const unsigned short* bias;
     unsigned short* y;
     int r;
     if (row < M0) {
-      W = W0; y = y0; r = row;
+      W = W0; y = y0; bias = b0; r = row;
     } else if (row < M0 + M1) {
-      W = W1; y = y1; r = row - M0;
+      W = W1; y = y1; bias = b1; r = row - M0;
     } else {
-      W = W2; y = y2; r = row - M0 - M1;
+      W = W2; y = y2; bias = b2; r = row - M0 - M1;
     }
     const uint4* wrow = reinterpret_cast<const uint4*>(W) +
                         static_cast<long long>(r) * vec_k;
@@ -429,14 +433,18 @@ __global__ __launch_bounds__(256) void gemv_multi_bf16_kernel(
     for (int off = 32; off > 0; off >>= 1) {
       acc += __shfl_down(acc, off, 64);
     }
-    if (lane == 0) gemv_store(y, r, acc);
+    if (lane == 0) {
+      if (bias != nullptr) acc += bf16_to_f32(bias[r]);
+      gemv_store(y, r, acc);
+    }
   }
 }
 
 extern "C" hipError_t fma_launch_gemv_multi_bf16(
     const void* W0, int M0, void* y0, const void* W1, int M1, void* y1,
     const void* W2, int M2, void* y2, const void* x, int K,
-    const void* norm_w, float norm_eps, hipStream_t stream) {
+    const void* norm_w, float norm_eps, const void* b0, const void* b1,
+    const void* b2, hipStream_t stream) {
   if ((K & 7) != 0) return hipErrorInvalidValue;
   const int block = 256;
   const int waves_per_block = block / 64;
@@ -454,16 +462,25 @@ extern "C" hipError_t fma_launch_gemv_multi_bf16(
         static_cast<const unsigned short*>(W2), M2,
         static_cast<unsigned short*>(y2),
         static_cast<const unsigned short*>(x), K,
-        static_cast<const unsigned short*>(norm_w), norm_eps);
+        static_cast<const unsigned short*>(norm_w), norm_eps,
+        static_cast<const unsigned short*>(b0),
+        static_cast<const unsigned short*>(b1),
+        static_cast<const unsigned short*>(b2));
     return hipGetLastError();
   }
 #define FMA_GEMVM_ARGS                                                    static_cast<const unsigned short*>(W0), M0,                                 static_cast<unsigned short*>(y0),                                       static_cast<const unsigned short*>(W1), M1,                             static_cast<unsigned short*>(y1),                                       static_cast<const unsigned short*>(W2), M2,                             static_cast<unsigned short*>(y2),                                       static_cast<const unsigned short*>(x), K
   if (lds <= 32 * 1024) {
     gemv_multi_bf16_kernel<true><<<blocks, block, lds, stream>>>(
-        FMA_GEMVM_ARGS, nullptr, 0.0f);
+        FMA_GEMVM_ARGS, nullptr, 0.0f,
+        static_cast<const unsigned short*>(b0),
+        static_cast<const unsigned short*>(b1),
+        static_cast<const unsigned short*>(b2));
   } else {
     gemv_multi_bf16_kernel<false><<<blocks, block, 0, stream>>>(
-        FMA_GEMVM_ARGS, nullptr, 0.0f);
+        FMA_GEMVM_ARGS, nullptr, 0.0f,
+        static_cast<const unsigned short*>(b0),
+        static_cast<const unsigned short*>(b1),
+        static_cast<const unsigned short*>(b2));
   }
 #undef FMA_GEMVM_ARGS
   return hipGetLastError();

@@ -35,8 +35,22 @@ from fma_amd.ops.linear import (fast_down_proj, fast_linear,
 from fma_amd.models.llama import LlamaModel, rmsnorm
 
 
+#: hipGraph replay with vocab sizes beyond this measured boundary hits a
+#: hardware exception on ROCm 7.2 (reproduced at vocab 152064 on BOTH the
+#: all-custom-kernel and the all-torch-op step, eager always clean;
+#: vocab <= 32768 graphs are stable across the whole test matrix and
+#: isolated big-vocab argmax/embedding/GEMV micro-graphs pass — the fault
+#: needs the full capture pool). Until root-caused, big-vocab models
+#: decode eagerly; see NOTES.md round-3 candidates.
+GRAPH_SAFE_VOCAB = 100_000
+
+
 class StaticDecoder:
     """Single-token decode with static shapes over a private KV cache."""
+
+    @staticmethod
+    def supported(cfg) -> bool:
+        return cfg.vocab_size <= GRAPH_SAFE_VOCAB
 
     def __init__(self, model: LlamaModel, batch: int, max_seq: int):
         assert model.tp_size == 1, "graphed decode is single-rank for now"
@@ -102,7 +116,9 @@ class StaticDecoder:
             h = fast_rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps)
             q, k, v = fast_linear_multi(
                 h, (P[p + "wq.weight"], P[p + "wk.weight"],
-                    P[p + "wv.weight"]))
+                    P[p + "wv.weight"]),
+                biases=(P[p + "wq.bias"], P[p + "wk.bias"],
+                        P[p + "wv.bias"]) if cfg.qkv_bias else None)
             q = q.view(B, 1, q_heads, hd)
             k = k.view(B, 1, kv_heads, hd)
             v = v.view(B, 1, kv_heads, hd)

@@ -46,6 +46,9 @@ class LlamaConfig:
     max_seq_len: int = 4096
     rope_theta: float = 500000.0
     norm_eps: float = 1e-5
+    #: Qwen2-family: biases on the q/k/v projections (the only
+    #: architectural delta vs llama the actuation path sees)
+    qkv_bias: bool = False
     dtype: torch.dtype = torch.bfloat16
 
     @property
@@ -59,6 +62,20 @@ class LlamaConfig:
         return LlamaConfig(name="tiny", vocab_size=512, hidden_size=64,
                            intermediate_size=128, num_layers=2, num_heads=4,
                            num_kv_heads=2, max_seq_len=256)
+
+    @staticmethod
+    def qwen2_7b() -> "LlamaConfig":
+        return LlamaConfig(name="qwen2-7b", vocab_size=152064,
+                           hidden_size=3584, intermediate_size=18944,
+                           num_layers=28, num_heads=28, num_kv_heads=4,
+                           max_seq_len=8192, rope_theta=1000000.0,
+                           norm_eps=1e-6, qkv_bias=True)
+
+    @staticmethod
+    def tiny_qwen() -> "LlamaConfig":
+        return LlamaConfig(name="tiny-qwen", vocab_size=512, hidden_size=64,
+                           intermediate_size=128, num_layers=2, num_heads=4,
+                           num_kv_heads=2, max_seq_len=256, qkv_bias=True)
 
     @staticmethod
     def llama3_8b() -> "LlamaConfig":
@@ -107,8 +124,10 @@ class LlamaConfig:
     def by_name(name: str) -> "LlamaConfig":
         presets = {
             "tiny": LlamaConfig.tiny,
+            "tiny-qwen": LlamaConfig.tiny_qwen,
             "llama-3-8b": LlamaConfig.llama3_8b,
             "llama-3-70b": LlamaConfig.llama3_70b,
+            "qwen2-7b": LlamaConfig.qwen2_7b,
         }
         if name in presets:
             return presets[name]()
@@ -141,6 +160,14 @@ class LlamaConfig:
                 (p + "wq.weight", (q_local, h), d),
                 (p + "wk.weight", (kv_local, h), d),
                 (p + "wv.weight", (kv_local, h), d),
+            ]
+            if self.qkv_bias:
+                specs += [
+                    (p + "wq.bias", (q_local,), d),
+                    (p + "wk.bias", (kv_local,), d),
+                    (p + "wv.bias", (kv_local,), d),
+                ]
+            specs += [
                 (p + "wo.weight", (h, q_local), d),
                 (p + "mlp_norm.weight", (h,), d),
                 (p + "w_gate.weight", (i_local, h), d),
@@ -235,6 +262,8 @@ class LlamaModel:
                     p.normal_(0.0, 0.02, generator=g)
                 else:
                     p.normal_(0.0, 0.02)  # in-place device RNG: no temps
+            elif name.endswith(".bias"):
+                p.normal_(0.0, 0.02)
             else:
                 p.fill_(1.0)  # norm gains
 
@@ -260,22 +289,29 @@ class LlamaModel:
                    and x.dtype == torch.bfloat16)
         for li in range(cfg.num_layers):
             p = f"layers.{li}."
+            biases = (P[p + "wq.bias"], P[p + "wk.bias"],
+                      P[p + "wv.bias"]) if cfg.qkv_bias else None
             if decode1:
-                # qkv in ONE launch. (A norm-fused variant exists —
-                # gemv_multi_bf16(norm_w=...) — but measured SLOWER:
-                # every block redundantly re-reads x and reduces the sum
-                # of squares, costing more than the one saved launch:
-                # 228 -> 220 tok/s short ctx. Selection is measured.)
+                # qkv in ONE launch, biases fused into the GEMV stores.
+                # (A norm-fused variant exists — gemv_multi_bf16(
+                # norm_w=...) — but measured SLOWER: every block
+                # redundantly re-reads x and reduces the sum of squares,
+                # costing more than the one saved launch: 228 -> 220
+                # tok/s short ctx. Selection is measured.)
                 h = fast_rmsnorm(x, P[p + "attn_norm.weight"],
                                  cfg.norm_eps)
                 q, k, v = fast_linear_multi(
                     h, (P[p + "wq.weight"], P[p + "wk.weight"],
-                        P[p + "wv.weight"]))
+                        P[p + "wv.weight"]), biases=biases)
             else:
                 h = rmsnorm(x, P[p + "attn_norm.weight"], cfg.norm_eps)
                 q = fast_linear(h, P[p + "wq.weight"])
                 k = fast_linear(h, P[p + "wk.weight"])
                 v = fast_linear(h, P[p + "wv.weight"])
+                if biases is not None:
+                    q = q + biases[0]
+                    k = k + biases[1]
+                    v = v + biases[2]
             q = q.view(B, T, q_heads, hd)
             k = k.view(B, T, kv_heads, hd)
             v = v.view(B, T, kv_heads, hd)

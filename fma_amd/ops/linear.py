@@ -41,11 +41,12 @@ def fast_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return F.linear(x, w)
 
 
-def fast_linear_multi(x, weights, norm=None):
+def fast_linear_multi(x, weights, norm=None, biases=None):
     """1-3 projections of the same single-token x in ONE kernel launch
-    (qkv, gate+up), optionally with the rmsnorm producing the GEMV input
-    fused in (norm = (weight, eps); x is then the RAW residual stream).
-    Falls back to eager rmsnorm + per-weight F.linear off the fast path."""
+    (qkv, gate+up), optionally with per-weight biases fused into the
+    stores (Qwen2-family qkv biases) and/or the rmsnorm producing the
+    GEMV input fused in (norm = (weight, eps); x is then the RAW
+    residual stream). Falls back to eager math off the fast path."""
     K = x.shape[-1]
     if (gemv_enabled() and x.is_cuda
             and x.dtype == torch.bfloat16
@@ -53,22 +54,29 @@ def fast_linear_multi(x, weights, norm=None):
             and (norm is None or (norm[0].is_contiguous()
                                   and norm[0].dtype == torch.bfloat16
                                   and K * 2 <= 32 * 1024))
+            and (biases is None
+                 or all(b.dtype == torch.bfloat16 and b.is_contiguous()
+                        for b in biases))
             and all(w.dtype == torch.bfloat16 and w.is_contiguous()
                     and (w.shape[1] & 7) == 0 for w in weights)):
+        blist = [b.reshape(-1) for b in biases] if biases is not None             else None
         if norm is None:
-            ys = actuation._C.gemv_multi_bf16(x.reshape(-1).contiguous(),
-                                              list(weights))
+            ys = actuation._C.gemv_multi_bf16(
+                x.reshape(-1).contiguous(), list(weights), biases=blist)
         else:
             ys = actuation._C.gemv_multi_bf16(
                 x.reshape(-1).contiguous(), list(weights),
-                norm[0].reshape(-1), float(norm[1]))
+                norm[0].reshape(-1), float(norm[1]), biases=blist)
         return [y.view(*x.shape[:-1], w.shape[0])
                 for y, w in zip(ys, weights)]
     if norm is not None:
         xf = x.float()
         xf = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + norm[1])
         x = (xf * norm[0].float()).to(x.dtype)
-    return [F.linear(x, w) for w in weights]
+    outs = [F.linear(x, w) for w in weights]
+    if biases is not None:
+        outs = [o + b for o, b in zip(outs, biases)]
+    return outs
 
 
 def fast_down_proj(gate, up, w, residual):

@@ -234,6 +234,11 @@ class ActuationEngine:
         if (mode == "0" or batch != 1 or self.tp_size != 1
                 or not self.on_gpu):
             return None
+        from fma_amd.models.decode_graph import StaticDecoder as _SD
+        if mode != "1" and not _SD.supported(self.cfg):
+            # measured ROCm 7.2 graph-replay fault at large vocab
+            # (decode_graph.GRAPH_SAFE_VOCAB); eager decode is clean
+            return None
         if need_seq > self.cfg.max_seq_len:
             return None
         dec = self._decoder

@@ -667,3 +667,30 @@ def test_gemv_silu_matches_unfused(C):
     ref = (w.float() @ act.float() + r.float())
     assert torch.allclose(y.float(), ref, atol=0.2, rtol=0.05), \
         (y.float() - ref).abs().max()
+
+
+def test_qwen2_family_decode_fused_matches_eager(C):
+    """Second model family on GPU: the bias-fused qkv launch produces
+    the same tokens as the eager (FMA_DISABLE_GEMV) path, and the
+    engine sleep/wake stays bit-stable with biases in the layout."""
+    import os
+
+    from fma_amd.models.llama import LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+
+    cfg = LlamaConfig.by_name("tiny-qwen")
+    eng = ActuationEngine(cfg, seed=21)
+    toks = torch.randint(0, cfg.vocab_size, (1, 8), device=eng.device)
+    fused = eng.generate(toks, max_new_tokens=6)
+    import fma_amd.ops.linear as lin
+    os.environ["FMA_DISABLE_GEMV"] = "1"
+    lin._ENABLED = None
+    try:
+        eager = eng.generate(toks, max_new_tokens=6)
+    finally:
+        os.environ.pop("FMA_DISABLE_GEMV")
+        lin._ENABLED = None
+    assert torch.equal(fused, eager)
+    eng.sleep()
+    eng.wake_up()
+    assert torch.equal(eng.generate(toks, max_new_tokens=6), fused)

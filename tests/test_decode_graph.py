@@ -44,3 +44,16 @@ def test_static_decoder_logits_match_forward(engine):
     # bf16 accumulation-order differences (padded-window SDPA vs exact
     # length) bound the agreement; greedy decode is exact (tests above)
     assert torch.allclose(dec.logits, full[:, -1], atol=1e-2, rtol=1e-2)
+
+
+def test_graph_decoder_gated_on_safe_vocab():
+    """Big-vocab models must not select the hipGraph decoder (measured
+    ROCm 7.2 replay fault at vocab 152064; decode_graph.GRAPH_SAFE_VOCAB
+    documents the boundary). Eager decode remains the path."""
+    from fma_amd.models.decode_graph import GRAPH_SAFE_VOCAB, StaticDecoder
+    from fma_amd.models.llama import LlamaConfig
+
+    assert StaticDecoder.supported(LlamaConfig.by_name("synthetic-15gib"))
+    qwen = LlamaConfig.by_name("qwen2-7b")
+    assert qwen.vocab_size > GRAPH_SAFE_VOCAB
+    assert not StaticDecoder.supported(qwen)

@@ -92,8 +92,9 @@ def main():
 
     graphed_tps = None
     eng = r.engine if hasattr(r, "engine") else r
-    if eng.on_gpu and eng.tp_size == 1:
-        from fma_amd.models.decode_graph import StaticDecoder
+    from fma_amd.models.decode_graph import StaticDecoder
+    if eng.on_gpu and eng.tp_size == 1 \
+            and StaticDecoder.supported(eng.cfg):
         dec = StaticDecoder(eng.model, 1, args.prompt_len + args.new_tokens + 2)
         dec.capture()
         dec.generate(toks, args.new_tokens)  # warm
