@@ -29,8 +29,8 @@ from fma_amd.ops.decode_ops import (available as _fused_available,
                                     fast_rope_qkv_store,
                                     fast_rmsnorm, fast_rope1,
                                     fast_silu_mul)
-from fma_amd.ops.linear import (fast_down_proj, fast_linear,
-                                fast_linear_multi, fast_linear_residual)
+from fma_amd.ops.linear import (fast_linear, fast_linear_multi,
+                                fast_linear_residual)
 
 from fma_amd.models.llama import LlamaModel, rmsnorm
 

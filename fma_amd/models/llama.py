@@ -30,8 +30,8 @@ import torch.nn.functional as F
 from fma_amd.ops.decode_ops import (fast_attn_decode, fast_attn_prefill,
                                     fast_rmsnorm, fast_rope1,
                                     fast_rope_qkv_store, fast_silu_mul)
-from fma_amd.ops.linear import (fast_down_proj, fast_linear,
-                                fast_linear_multi, fast_linear_residual)
+from fma_amd.ops.linear import (fast_linear, fast_linear_multi,
+                                fast_linear_residual)
 
 
 @dataclass
