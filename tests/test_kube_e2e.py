@@ -98,6 +98,11 @@ def test_kube_full_actuation_hot_start_and_vap(kube_cluster):
                    json={"prompt": "hi", "max_tokens": 2}, timeout=10)
     assert r.status_code == 200
 
+    # notifier signature reflected onto the launcher Pod over the kube
+    # protocol (the informer-visible launcher-state channel)
+    wait_for(lambda: C.INSTANCE_SIGNATURE_ANNOTATION in ob.annotations_of(
+        store.get("Pod", ob.name_of(lp))), 30, desc="notifier signature")
+
     # CEL VAP: a user cannot clear the binding annotation on the provider
     hacked = store.get("Pod", ob.name_of(lp))
     hacked["metadata"]["annotations"][C.REQUESTER_ANNOTATION] = "u hack"
