@@ -351,6 +351,11 @@ class NodeQueueAndWorkers(Generic[T]):
             node = self.queue.outer.get()
             if node is None:
                 return
+            if self._mname:
+                from fma_amd.controller import metrics as _m
+                # depth = pending ITEMS across nodes (the fma_dpc_
+                # innerqueue contract), not outer-queue nodes
+                _m.queue_depth().labels(self._mname).set(len(self.queue))
             for item in self.queue.take_ready(node):
                 t0 = time.monotonic()
                 try:
