@@ -82,6 +82,30 @@ def bound_requester_policy(op: str, old: Optional[Dict[str, Any]],
             "bound server-requesting Pod")
 
 
+#: kind -> pydantic spec model: structural CRD validation, the analog of
+#: the apiserver's OpenAPI schema + listMapKey checks (the reference's
+#: e2e expects e.g. an LPP with duplicate countForLauncher keys to be
+#: REJECTED at create — test-cases.sh:266-296)
+def crd_schema_policy(op: str, old, new, actor: str) -> None:
+    if op not in ("CREATE", "UPDATE") or new is None:
+        return
+    kind = new.get("kind")
+    from fma_amd.api import types as t
+    models = {
+        "InferenceServerConfig": t.InferenceServerConfigSpec,
+        "LauncherConfig": t.LauncherConfigSpec,
+        "LauncherPopulationPolicy": t.LauncherPopulationPolicySpec,
+    }
+    model = models.get(kind)
+    if model is None:
+        return
+    try:
+        model.model_validate(new.get("spec") or {})
+    except Exception as e:  # pydantic.ValidationError
+        raise Invalid(f"{kind} spec invalid: {e}") from e
+
+
 def install_policies(store: MemStore) -> None:
+    store.add_admission_hook(crd_schema_policy)
     store.add_admission_hook(immutable_fields_policy)
     store.add_admission_hook(bound_requester_policy)

@@ -149,6 +149,10 @@ def create_app(store: Optional[MemStore] = None,
     app = FastAPI(title="fma-amd kube apiserver double")
     st = store or MemStore()
     app.state.store = st
+    # structural CRD validation, as a real apiserver's schema would
+    from fma_amd.store.admission import crd_schema_policy
+    if crd_schema_policy not in st._admission:
+        st.add_admission_hook(crd_schema_policy)
     vap = AdmissionPolicies(vap_dir)
     app.state.vap = vap
 

@@ -61,8 +61,10 @@ def _reset_backing(app):
         st._indexes.clear()
         st._indexed_keys.clear()
         st._admission.clear()
+    from fma_amd.store.admission import crd_schema_policy
     from fma_amd.store.indexes import install_pod_indexes
     install_pod_indexes(st)
+    st.add_admission_hook(crd_schema_policy)
 
 
 @pytest.fixture(autouse=True)

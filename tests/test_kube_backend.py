@@ -246,3 +246,23 @@ def test_watch_only_requested_kinds(kube):
     th.join(timeout=20)
     assert got == [("Pod", "w2")]
     stop.set()
+
+
+def test_crd_schema_rejects_duplicate_lpp_keys(kube):
+    """CRD structural validation at the apiserver: an LPP with duplicate
+    countForLauncher launcherConfigName keys is rejected at create
+    (reference test-cases.sh:266-296, the listMapKey semantics)."""
+    ks = kube["user"]
+    with pytest.raises(Invalid) as ei:
+        ks.create(ob.new_object(
+            "LauncherPopulationPolicy", "dup",
+            spec={"enhancedNodeSelector": {"labelSelector": {}},
+                  "countForLauncher": [
+                      {"launcherConfigName": "lc1", "launcherCount": 1},
+                      {"launcherConfigName": "lc1", "launcherCount": 2}]}))
+    assert "unique" in str(ei.value)
+    # malformed ISC spec rejected too
+    with pytest.raises(Invalid):
+        ks.create(ob.new_object(
+            "InferenceServerConfig", "bad",
+            spec={"modelServerConfig": {"port": "not-a-port"}}))

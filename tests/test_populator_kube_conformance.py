@@ -14,3 +14,13 @@ pytestmark = pytest.mark.timeout(300)
 def _kube_backend(monkeypatch):
     monkeypatch.setattr(_pop_mod, "MemStore", _KubeBackedStore)
     yield
+
+
+# Over the kube backend the apiserver-level CRD schema admission rejects
+# the duplicate-key LPP at create (the reference behavior,
+# test-cases.sh:266-296) — the populator-internal defense that test
+# exercises is covered on the MemStore path, and the rejection itself in
+# test_kube_backend/test_store_and_admission.
+test_lpp_status_reports_errors = pytest.mark.skip(
+    reason="kube backend rejects the bad LPP at create (CRD admission)")(
+        _pop_mod.test_lpp_status_reports_errors)

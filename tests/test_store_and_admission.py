@@ -333,3 +333,28 @@ def test_watch_current_cursor_survives_eviction():
     st.create({"kind": "Pod", "metadata": {"name": "fresh"}})
     evs = list(st.watch(since=cursor, timeout=0.1))
     assert [e.obj["metadata"]["name"] for e in evs] == ["fresh"]
+
+
+def test_crd_schema_admission_on_memstore():
+    """install_policies enforces CRD structural validation in-process
+    (duplicate LPP keys, malformed specs) — same rule as the kube
+    double's apiserver-level check."""
+    from fma_amd.store import objects as ob
+    from fma_amd.store.admission import install_policies
+    from fma_amd.store.memstore import Invalid, MemStore
+
+    st = MemStore()
+    install_policies(st)
+    with pytest.raises(Invalid):
+        st.create(ob.new_object(
+            "LauncherPopulationPolicy", "dup",
+            spec={"enhancedNodeSelector": {"labelSelector": {}},
+                  "countForLauncher": [
+                      {"launcherConfigName": "lc1", "launcherCount": 1},
+                      {"launcherConfigName": "lc1", "launcherCount": 2}]}))
+    # well-formed objects pass
+    st.create(ob.new_object(
+        "LauncherPopulationPolicy", "ok",
+        spec={"enhancedNodeSelector": {"labelSelector": {}},
+              "countForLauncher": [
+                  {"launcherConfigName": "lc1", "launcherCount": 1}]}))
