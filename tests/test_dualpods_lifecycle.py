@@ -197,3 +197,70 @@ def test_unbound_requester_not_mirrored():
     w = mk_world(with_launcher=False)
     w["ctl"]._process(infsvr_item(w["store"]))
     assert w["store"].try_get("Pod", "req1") is not None
+
+
+def test_switching_instances_in_one_launcher():
+    """Reference test-cases.sh:560: requester(iscA) unbinds (instance A
+    sleeps), requester(iscB) binds the SAME launcher — a new instance is
+    created beside the sleeping one (maxInstances 2), then a third
+    same-as-A requester hot-starts A again."""
+    import copy
+
+    w = mk_world()  # launcher holds sleeping instance for isc1 (A)
+    store, http = w["store"], w["http"]
+
+    # requester for isc1: hot start on A
+    drive(w["ctl"], infsvr_item(store, "req1"))
+    assert w["inst_srv"].wakes == 1
+
+    # unbind: A sleeps again
+    store.delete("Pod", "req1", actor="user")
+    drive(w["ctl"], infsvr_item(store, "req1"))
+    assert w["inst_srv"].sleeping
+
+    # a second ISC arrives; its requester must bind the SAME launcher
+    msc2 = copy.deepcopy(MSC)
+    msc2["port"] = 8002
+    store.create(ob.new_object(
+        "InferenceServerConfig", "isc2",
+        spec={"modelServerConfig": msc2, "launcherConfigName": "lc1"}))
+
+    def on_create(iid, inst):
+        http.register("10.0.0.2:8002", FakeInstanceServer(sleeping=False))
+
+    w["launcher"].on_create = on_create
+    req2 = ob.new_object(
+        "Pod", "req2",
+        annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: "isc2"},
+        spec={"nodeName": "node-a", "containers": [{"name": "stub"}]})
+    req2 = store.create(req2)
+    req2["status"] = {"phase": "Running", "podIP": "10.0.0.7"}
+    store.update(req2)
+    http.register("10.0.0.7:8081", Stub(["GPU-1"]))
+    drive(w["ctl"], infsvr_item(store, "req2"))
+
+    lp = store.get("Pod", "launcher1")
+    assert ob.annotations_of(lp)[C.REQUESTER_ANNOTATION].endswith(" req2")
+    # both instances live on the one launcher: A sleeping, B serving
+    assert len(w["launcher"].instances) == 2
+    assert w["inst_srv"].sleeping  # A still parked
+
+    # requester for isc1 returns; launcher is bound to req2 -> the
+    # controller must NOT steal it; a new launcher would be created
+    # (cold). Instead delete req2 first, then hot-start A.
+    store.delete("Pod", "req2", actor="user")
+    drive(w["ctl"], infsvr_item(store, "req2"))
+    req3 = ob.new_object(
+        "Pod", "req3",
+        annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: "isc1"},
+        spec={"nodeName": "node-a", "containers": [{"name": "stub"}]})
+    req3 = store.create(req3)
+    req3["status"] = {"phase": "Running", "podIP": "10.0.0.8"}
+    store.update(req3)
+    http.register("10.0.0.8:8081", Stub(["GPU-0"]))
+    drive(w["ctl"], infsvr_item(store, "req3"))
+    lp = store.get("Pod", "launcher1")
+    assert ob.annotations_of(lp)[C.REQUESTER_ANNOTATION].endswith(" req3")
+    assert ob.annotations_of(lp)[C.INSTANCE_ID_ANNOTATION] == w["iid"]
+    assert not w["inst_srv"].sleeping  # A woken again
+    assert w["inst_srv"].wakes == 2
