@@ -56,6 +56,18 @@ namespace {
 
 using Clock = std::chrono::steady_clock;
 
+// RAII join guard: a TORCH_CHECK/FMA_HIP_CHECK throw while helper
+// threads are live must not hit std::thread's terminate-on-destroy;
+// helpers here always run to completion on their own, so joining in the
+// unwind path is safe and bounded.
+struct ThreadJoiner {
+  std::thread& t;
+  explicit ThreadJoiner(std::thread& th) : t(th) {}
+  ~ThreadJoiner() {
+    if (t.joinable()) t.join();
+  }
+};
+
 double seconds_since(Clock::time_point t0) {
   return std::chrono::duration<double>(Clock::now() - t0).count();
 }
@@ -594,13 +606,14 @@ double restore_from_host_overlapped(std::vector<at::Tensor> tensors,
       cv.notify_all();
     }
   });
+  ThreadJoiner alloc_joiner(alloc_thread);
   auto wait_ready = [&](long idx) {
     std::unique_lock<std::mutex> lk(mu);
     cv.wait(lk, [&] { return failed || ready >= idx; });
     return !failed;
   };
   auto fail_out = [&]() {
-    alloc_thread.join();
+    if (alloc_thread.joinable()) alloc_thread.join();
     (void)hipDeviceSynchronize();
     TORCH_CHECK(false, "storage allocation failed during overlapped wake: ",
                 fail_msg);
@@ -617,7 +630,7 @@ double restore_from_host_overlapped(std::vector<at::Tensor> tensors,
                                    ctx.copy_streams[i % ns]));
     }
     sync_pipeline(ctx);
-    alloc_thread.join();
+    if (alloc_thread.joinable()) alloc_thread.join();
     return seconds_since(t0);
   }
   TORCH_CHECK(mode == XferMode::kStaged,
@@ -733,7 +746,7 @@ double restore_from_host_overlapped(std::vector<at::Tensor> tensors,
   }
   (void)hipFree(descs_dev);
   if (bail) return fail_out();
-  alloc_thread.join();
+  if (alloc_thread.joinable()) alloc_thread.join();
   return seconds_since(t0);
 }
 
@@ -965,9 +978,22 @@ class DeviceArena {
       };
       const int nthreads = std::max(1, alloc_threads_);
       std::vector<std::thread> workers;
+      // join-on-unwind: an FMA_HIP_CHECK throw in the copy loop must not
+      // destroy joinable threads (workers always terminate on their own)
+      struct WorkersJoiner {
+        std::vector<std::thread>& ws;
+        std::thread* self = nullptr;
+        ~WorkersJoiner() {
+          if (self && self->joinable()) self->join();
+          for (auto& w : ws) {
+            if (w.joinable()) w.join();
+          }
+        }
+      } joiner{workers};
       for (int t = 1; t < nthreads; ++t) workers.emplace_back(alloc_worker);
       // the calling thread participates too unless it must start copying
       std::thread self_worker(alloc_worker);
+      joiner.self = &self_worker;
       int c = 0;
       double alloc_wait = 0.0;
       for (size_t i = 0; i < n; ++i) {
@@ -991,7 +1017,9 @@ class DeviceArena {
         }
       }
       self_worker.join();
-      for (auto& t : workers) t.join();
+      for (auto& t : workers) {
+        if (t.joinable()) t.join();
+      }
       TORCH_CHECK(!failed.load(), "hipMalloc failed during wake");
       map_seconds_ = seconds_since(ta0);
       alloc_wait_seconds_ = alloc_wait;
