@@ -266,3 +266,45 @@ def test_crd_schema_rejects_duplicate_lpp_keys(kube):
         ks.create(ob.new_object(
             "InferenceServerConfig", "bad",
             spec={"modelServerConfig": {"port": "not-a-port"}}))
+
+
+def test_watch_survives_apiserver_restart():
+    """KubeStore's per-resource watch threads reconnect from their cursor
+    after the apiserver drops (reflector behavior): events created after
+    the restart still arrive on the same watch iterator."""
+    store = MemStore()
+    port = free_port()
+    srv = ServerThread(create_app(store), port)
+    srv.__enter__()
+    ks = KubeStore(f"http://127.0.0.1:{port}", actor="user")
+    token = ks.list_revision()
+    got = []
+    stop = threading.Event()
+
+    def consume():
+        for ev in ks.watch(since=token, kinds=["Pod"], stop=stop):
+            got.append(ob.name_of(ev.obj))
+            if len(got) >= 2:
+                return
+
+    th = threading.Thread(target=consume, daemon=True)
+    th.start()
+    time.sleep(0.3)
+    ks.create(mk_pod("before-restart"))
+    deadline = time.time() + 10
+    while time.time() < deadline and len(got) < 1:
+        time.sleep(0.05)
+    assert got == ["before-restart"]
+
+    srv.__exit__()  # apiserver down
+    time.sleep(0.5)
+    srv2 = ServerThread(create_app(store), port)  # same store, same port
+    srv2.__enter__()
+    try:
+        ks.create(mk_pod("after-restart"))
+        th.join(timeout=20)
+        assert not th.is_alive(), f"watch did not resume; got {got}"
+        assert got == ["before-restart", "after-restart"]
+    finally:
+        stop.set()
+        srv2.__exit__()
