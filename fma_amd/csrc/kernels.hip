@@ -980,7 +980,7 @@ extern "C" int fma_attn_decode_chunks(int t, int q_heads) {
   // OVERFILL the chip: ~1024 blocks puts 4 workgroups on every CU
   // (LDS/VGPR allow it), and that occupancy is what hides the HBM
   // latency of the per-row K/V streams — at 256 blocks the kernel ran
-  // at occupancy 1 and ~2% of peak at t=4K (gpurun_out/decode4k_stats).
+  // at occupancy 1 and ~2% of peak at t=4K (profiles/decode4k_kernel_stats_round2.csv).
   // Keep chunks >= 128 positions so the partials+combine overhead stays
   // amortized, and within the LDS score window.
   int chunks = 512 / q_heads;  // ~512 blocks: measured best (24.7 us at
