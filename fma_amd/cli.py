@@ -85,6 +85,10 @@ def cmd_get(client: StoreClient, kind: str, name: str, output: str) -> None:
 def main(argv=None) -> None:
     ap = argparse.ArgumentParser("fma")
     ap.add_argument("--store-url", default="http://127.0.0.1:8081")
+    ap.add_argument("--kube", default="",
+                    help="Kubernetes apiserver base URL: talk the k8s "
+                         "wire protocol (KubeStore) instead of the fma "
+                         "store server")
     ap.add_argument("--actor", default="user")
     sub = ap.add_subparsers(dest="cmd", required=True)
     p_apply = sub.add_parser("apply")
@@ -100,7 +104,11 @@ def main(argv=None) -> None:
     sub.add_parser("watch")
     args = ap.parse_args(argv)
 
-    client = StoreClient(args.store_url, actor=args.actor)
+    if args.kube:
+        from fma_amd.store.kubestore import KubeStore
+        client = KubeStore(args.kube, actor=args.actor)
+    else:
+        client = StoreClient(args.store_url, actor=args.actor)
     if args.cmd == "apply":
         cmd_apply(client, args.filename)
     elif args.cmd == "get":
@@ -109,7 +117,8 @@ def main(argv=None) -> None:
         client.delete(resolve_kind(args.kind), args.name)
         print(f"{args.kind}/{args.name} deleted")
     elif args.cmd == "watch":
-        for ev in client.watch():
+        since = client.list_revision() if args.kube else 0
+        for ev in client.watch(since=since):
             meta = ev.obj.get("metadata", {})
             print(f"{ev.revision}\t{ev.type}\t{ev.kind}/{meta.get('name')}")
             sys.stdout.flush()

@@ -58,3 +58,31 @@ def test_kubernetes_manifests_parse_and_cover_rbac():
     crd_rule = next(r for r in role["rules"]
                     if "inferenceserverconfigs" in r.get("resources", []))
     assert "watch" in crd_rule["verbs"]
+
+
+def test_cli_against_kube_apiserver(tmp_path, capsys):
+    """fma --kube drives a Kubernetes apiserver (the double here) through
+    the k8s wire protocol: apply / get / delete round-trip."""
+    from fma_amd.cli import main as cli_main
+    from fma_amd.store.kubeapiserver import create_app
+    from fma_amd.store.memstore import MemStore
+
+    from tests.test_live_servers import ServerThread, free_port
+
+    port = free_port()
+    with ServerThread(create_app(MemStore()), port):
+        base = f"http://127.0.0.1:{port}"
+        f = tmp_path / "isc.yaml"
+        f.write_text(
+            "kind: InferenceServerConfig\n"
+            "metadata: {name: cli-isc}\n"
+            "spec:\n"
+            "  modelServerConfig: {port: 8000, options: '--model tiny'}\n"
+            "  launcherConfigName: lc1\n")
+        cli_main(["--kube", base, "apply", "-f", str(f)])
+        cli_main(["--kube", base, "get", "isc", "cli-isc", "-o", "json"])
+        out = capsys.readouterr().out
+        assert "cli-isc" in out and "launcherConfigName" in out
+        cli_main(["--kube", base, "delete", "isc", "cli-isc"])
+        out = capsys.readouterr().out
+        assert "deleted" in out
