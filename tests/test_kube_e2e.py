@@ -238,3 +238,76 @@ def test_kube_stopped_instance_recovers(kube_cluster):
     # the owning ReplicaSet re-creates a fresh one)
     wait_for(lambda: store.try_get("Pod", "sreq1") is None, 120,
              desc="requester deleted after instance death")
+
+
+def test_kube_two_node_population_and_binding(tmp_path):
+    """Two node agents + both controllers all over the kube protocol:
+    the populator spans nodes (one launcher each), and a requester on
+    node-b binds node-b's launcher (binding locality) — the kind
+    multi-node analog running against the apiserver double."""
+    mem = MemStore()
+    port = free_port()
+    with ServerThread(create_app(mem), port):
+        base = f"http://127.0.0.1:{port}"
+        admin = KubeStore(base, actor="dual-pods-controller")
+        env = {
+            "PYTHONPATH": os.path.dirname(os.path.dirname(
+                os.path.abspath(__file__))),
+            "FMA_FAKE_GPU": "1",
+            "FMA_GPU_MODE": "naive",
+            "FMA_ACCELERATORS": "GPU-0",
+        }
+        agents = {}
+        for i, node in enumerate(("node-a", "node-b"), start=1):
+            n = ob.new_object("Node", node, labels={"gpu": "mi355x"})
+            n["status"] = {"allocatable": {C.GPU_RESOURCE_NAME: 8}}
+            admin.create(n)
+            agents[node] = NodeAgent(
+                KubeStore(base, actor="node-agent"), node,
+                node_index=30 + i, log_dir=str(tmp_path), extra_env=env)
+            agents[node].start()
+        ctl = DualPodsController(
+            KubeStore(base, actor="dual-pods-controller"), HttpAdapter(),
+            ControllerConfig())
+        ctl.start()
+        pop = LauncherPopulator(KubeStore(base, actor="launcher-populator"))
+        pop.start()
+        user = KubeStore(base, actor="user")
+        try:
+            mk_isc_lc_lpp(user)  # LPP matches both nodes
+            wait_for(lambda: len([
+                p for p in user.list("Pod")
+                if ob.labels_of(p).get(C.COMPONENT_LABEL) ==
+                C.LAUNCHER_COMPONENT]) == 2, 90, desc="two launchers")
+            by_node = {}
+            for p in user.list("Pod"):
+                if ob.labels_of(p).get(C.COMPONENT_LABEL) == \
+                        C.LAUNCHER_COMPONENT:
+                    by_node[ob.pod_node_name(p)] = p
+            assert set(by_node) == {"node-a", "node-b"}
+            for node, lp in by_node.items():
+                wait_for(lambda lp=lp: ob.pod_is_ready(
+                    user.get("Pod", ob.name_of(lp))), 90,
+                    desc=f"launcher on {node} ready")
+
+            pod = ob.new_object(
+                "Pod", "breq",
+                annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: "isc1"},
+                spec={"nodeName": "node-b", "containers": [{
+                    "name": "requester",
+                    "command": [sys.executable, "-m",
+                                "fma_amd.requester.server"]}]})
+            user.create(pod)
+            wait_for(lambda: requester_ready(user, agents["node-b"],
+                                             "breq"), 120,
+                     desc="node-b requester ready")
+            lp_b = user.get("Pod", ob.name_of(by_node["node-b"]))
+            assert ob.annotations_of(lp_b)[C.REQUESTER_ANNOTATION].endswith(
+                " breq")
+            lp_a = user.get("Pod", ob.name_of(by_node["node-a"]))
+            assert C.REQUESTER_ANNOTATION not in ob.annotations_of(lp_a)
+        finally:
+            ctl.stop()
+            pop.stop()
+            for a in agents.values():
+                a.stop()
