@@ -138,8 +138,12 @@ class DualPodsController:
                 self.resync()
 
     def _watch_once(self, since: int) -> None:
-        # replay everything from before start too (resync covered it)
-        for ev in self.store.watch(since=since, stop=self._stop):
+        # replay everything from before start too (resync covered it);
+        # only the kinds this controller reacts to (a KubeStore opens one
+        # watch stream per kind — unneeded kinds waste apiserver slots)
+        for ev in self.store.watch(
+                since=since, stop=self._stop,
+                kinds=["Pod", "InferenceServerConfig"]):
             if ev.kind == "Pod":
                 self._enqueue_for(ev.obj)
             elif ev.kind == "InferenceServerConfig":

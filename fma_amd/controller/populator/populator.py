@@ -156,7 +156,10 @@ class LauncherPopulator:
                 self.key_queue.queue.add(key)
 
     def _watch_once(self, since: int) -> None:
-        for ev in self.store.watch(since=since, stop=self._stop):
+        for ev in self.store.watch(
+                since=since, stop=self._stop,
+                kinds=["Pod", "LauncherConfig",
+                       "LauncherPopulationPolicy", "Node"]):
             if ev.kind == "LauncherPopulationPolicy":
                 self.digest_queue.queue.add(("lpp", ob.name_of(ev.obj)))
             elif ev.kind == "LauncherConfig":

@@ -37,6 +37,8 @@ import yaml
 from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, StreamingResponse
 
+from concurrent.futures import ThreadPoolExecutor
+
 from fma_amd.store import cel
 from fma_amd.store import objects as ob
 from fma_amd.store.memstore import ApiError, MemStore, RevisionTooOld
@@ -149,6 +151,13 @@ def create_app(store: Optional[MemStore] = None,
     app = FastAPI(title="fma-amd kube apiserver double")
     st = store or MemStore()
     app.state.store = st
+    # dedicated executor for watch polling: every live watch stream
+    # holds a worker for up to its poll interval, and the loop's default
+    # executor (~cpu+4 threads) starves under one-stream-per-kind
+    # informers from several controllers
+    watch_pool = ThreadPoolExecutor(max_workers=64,
+                                    thread_name_prefix="kube-watch")
+    app.state.watch_pool = watch_pool
     # structural CRD validation, as a real apiserver's schema would
     from fma_amd.store.admission import crd_schema_policy
     if crd_schema_policy not in st._admission:
@@ -184,7 +193,7 @@ def create_app(store: Optional[MemStore] = None,
                         return
                     try:
                         batch = await loop.run_in_executor(
-                            None, lambda: list(st.watch(
+                            watch_pool, lambda: list(st.watch(
                                 since=cursor, kinds=[kind], timeout=1.0)))
                     except RuntimeError:
                         return  # server shutting down (executor closed)
