@@ -8,4 +8,4 @@ control plane (dual-pods controllers, launcher, requester) speaks the same
 CRDs, annotations and HTTP contracts as the reference.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"

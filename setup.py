@@ -16,7 +16,7 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 
 setup(
     name="fma_amd",
-    version="0.1.0",
+    version="0.2.0",
     description="MI355X-native fast model actuation",
     packages=["fma_amd"],
     ext_modules=[
