@@ -69,6 +69,10 @@ class RateLimitingQueue(Generic[T]):
         with self._cond:
             if self._shutdown:
                 return
+            # prune fired timers so a long-running controller's retry
+            # traffic does not grow this list without bound
+            if len(self._timers) > 64:
+                self._timers = [x for x in self._timers if x.is_alive()]
             self._timers.append(t)
         t.start()
 

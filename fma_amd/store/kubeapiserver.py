@@ -158,6 +158,10 @@ def create_app(store: Optional[MemStore] = None,
     watch_pool = ThreadPoolExecutor(max_workers=64,
                                     thread_name_prefix="kube-watch")
     app.state.watch_pool = watch_pool
+
+    @app.on_event("shutdown")
+    async def _shutdown_pool():
+        watch_pool.shutdown(wait=False, cancel_futures=True)
     # structural CRD validation, as a real apiserver's schema would
     from fma_amd.store.admission import crd_schema_policy
     if crd_schema_policy not in st._admission:
