@@ -50,7 +50,8 @@ class StaticDecoder:
 
     @staticmethod
     def supported(cfg) -> bool:
-        return cfg.vocab_size <= GRAPH_SAFE_VOCAB
+        # MoE routing is data-dependent per token: un-capturable
+        return cfg.vocab_size <= GRAPH_SAFE_VOCAB and not cfg.num_experts
 
     def __init__(self, model: LlamaModel, batch: int, max_seq: int):
         assert model.tp_size == 1, "graphed decode is single-rank for now"

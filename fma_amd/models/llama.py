@@ -49,6 +49,10 @@ class LlamaConfig:
     #: Qwen2-family: biases on the q/k/v projections (the only
     #: architectural delta vs llama the actuation path sees)
     qkv_bias: bool = False
+    #: Mixtral-family: sparse MoE MLP (0 = dense). Routed top-k over
+    #: num_experts per token; each expert is a llama-shaped gated MLP.
+    num_experts: int = 0
+    num_experts_per_tok: int = 2
     dtype: torch.dtype = torch.bfloat16
 
     @property
@@ -76,6 +80,21 @@ class LlamaConfig:
         return LlamaConfig(name="tiny-qwen", vocab_size=512, hidden_size=64,
                            intermediate_size=128, num_layers=2, num_heads=4,
                            num_kv_heads=2, max_seq_len=256, qkv_bias=True)
+
+    @staticmethod
+    def mixtral_8x7b() -> "LlamaConfig":
+        return LlamaConfig(name="mixtral-8x7b", vocab_size=32000,
+                           hidden_size=4096, intermediate_size=14336,
+                           num_layers=32, num_heads=32, num_kv_heads=8,
+                           max_seq_len=8192, rope_theta=1000000.0,
+                           num_experts=8, num_experts_per_tok=2)
+
+    @staticmethod
+    def tiny_moe() -> "LlamaConfig":
+        return LlamaConfig(name="tiny-moe", vocab_size=512, hidden_size=64,
+                           intermediate_size=96, num_layers=2, num_heads=4,
+                           num_kv_heads=2, max_seq_len=256, num_experts=4,
+                           num_experts_per_tok=2)
 
     @staticmethod
     def llama3_8b() -> "LlamaConfig":
@@ -125,9 +144,11 @@ class LlamaConfig:
         presets = {
             "tiny": LlamaConfig.tiny,
             "tiny-qwen": LlamaConfig.tiny_qwen,
+            "tiny-moe": LlamaConfig.tiny_moe,
             "llama-3-8b": LlamaConfig.llama3_8b,
             "llama-3-70b": LlamaConfig.llama3_70b,
             "qwen2-7b": LlamaConfig.qwen2_7b,
+            "mixtral-8x7b": LlamaConfig.mixtral_8x7b,
         }
         if name in presets:
             return presets[name]()
@@ -170,10 +191,26 @@ class LlamaConfig:
             specs += [
                 (p + "wo.weight", (h, q_local), d),
                 (p + "mlp_norm.weight", (h,), d),
-                (p + "w_gate.weight", (i_local, h), d),
-                (p + "w_up.weight", (i_local, h), d),
-                (p + "w_down.weight", (h, i_local), d),
             ]
+            if self.num_experts:
+                # Mixtral-family: router replicated, every expert a
+                # Megatron-sharded gated MLP (leaf names w_gate/w_up/
+                # w_down keep the same col/row-parallel slicing rules)
+                specs.append((p + "router.weight",
+                              (self.num_experts, h), d))
+                for e in range(self.num_experts):
+                    ep = p + f"experts.{e}."
+                    specs += [
+                        (ep + "w_gate.weight", (i_local, h), d),
+                        (ep + "w_up.weight", (i_local, h), d),
+                        (ep + "w_down.weight", (h, i_local), d),
+                    ]
+            else:
+                specs += [
+                    (p + "w_gate.weight", (i_local, h), d),
+                    (p + "w_up.weight", (i_local, h), d),
+                    (p + "w_down.weight", (h, i_local), d),
+                ]
         specs += [
             ("final_norm.weight", (h,), d),
             ("lm_head.weight", (self.vocab_size, h), d),
@@ -271,6 +308,54 @@ class LlamaModel:
         if self.tp_size > 1:
             dist.all_reduce(x, group=self.tp_group)
         return x
+
+    def _moe_mlp(self, x: torch.Tensor, P, p: str,
+                 decode1: bool) -> torch.Tensor:
+        """Mixtral-style sparse MLP: softmax(top-k(router)) over
+        Megatron-sharded experts; ONE all-reduce after the weighted sum
+        (row-parallel partials add across ranks exactly like the dense
+        w_down). Router weights are replicated so every rank routes
+        identically. The decode path runs only the k selected experts'
+        GEMVs — at batch 1 a sparse model moves k/E of the expert bytes."""
+        cfg = self.cfg
+        h = fast_rmsnorm(x, P[p + "mlp_norm.weight"], cfg.norm_eps) \
+            if decode1 else rmsnorm(x, P[p + "mlp_norm.weight"],
+                                    cfg.norm_eps)
+        B, T, H = h.shape
+        logits = fast_linear(h, P[p + "router.weight"]).float()
+        topw, topi = torch.topk(logits, cfg.num_experts_per_tok, dim=-1)
+        topw = torch.softmax(topw, dim=-1)
+        if B * T == 1:
+            out = torch.zeros_like(x)
+            idx = topi.reshape(-1).tolist()  # k tiny ints; eager decode
+            for j, e in enumerate(idx):
+                ep = p + f"experts.{e}."
+                gate, up = fast_linear_multi(
+                    h, (P[ep + "w_gate.weight"], P[ep + "w_up.weight"]))
+                act = fast_silu_mul(gate, up) if decode1 \
+                    else F.silu(gate) * up
+                w = topw.reshape(-1)[j].to(x.dtype)
+                out = out + w * fast_linear(act, P[ep + "w_down.weight"])
+        else:
+            hf = h.view(B * T, H)
+            ti = topi.view(B * T, -1)
+            tw = topw.view(B * T, -1)
+            out = torch.zeros(B * T, H, dtype=torch.float32,
+                              device=x.device)
+            for e in range(cfg.num_experts):
+                mask = ti == e
+                tok = mask.any(-1).nonzero(as_tuple=True)[0]
+                if tok.numel() == 0:
+                    continue
+                ep = p + f"experts.{e}."
+                sub = hf[tok]
+                act = F.silu(F.linear(sub, P[ep + "w_gate.weight"])) * \
+                    F.linear(sub, P[ep + "w_up.weight"])
+                y = F.linear(act, P[ep + "w_down.weight"]).float()
+                w = (tw * mask).sum(-1)[tok].unsqueeze(-1)
+                out.index_add_(0, tok, w * y)
+            out = out.to(x.dtype).view(B, T, H)
+        return x + self._maybe_all_reduce(out)
 
     @torch.no_grad()
     def forward(self, tokens: torch.Tensor, cache: Optional[KVCache] = None,
@@ -371,6 +456,9 @@ class LlamaModel:
                 x = x + self._maybe_all_reduce(
                     fast_linear(att, P[p + "wo.weight"]))
 
+            if cfg.num_experts:
+                x = self._moe_mlp(x, P, p, decode1)
+                continue
             if decode1:
                 h = fast_rmsnorm(x, P[p + "mlp_norm.weight"],
                                  cfg.norm_eps)

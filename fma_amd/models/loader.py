@@ -45,6 +45,10 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
         max_seq_len=d.get("max_seq_len", d.get("max_position_embeddings",
                                                4096)),
         rope_theta=d.get("rope_theta", 500000.0),
+        norm_eps=d.get("norm_eps", d.get("rms_norm_eps", 1e-5)),
+        qkv_bias=d.get("qkv_bias", False),
+        num_experts=d.get("num_experts", d.get("num_local_experts", 0)),
+        num_experts_per_tok=d.get("num_experts_per_tok", 2),
     )
 
 
@@ -92,6 +96,10 @@ def save_params(params: Dict[str, torch.Tensor], path: str,
                 "num_kv_heads": cfg.num_kv_heads,
                 "max_seq_len": cfg.max_seq_len,
                 "rope_theta": cfg.rope_theta,
+                "norm_eps": cfg.norm_eps,
+                "qkv_bias": cfg.qkv_bias,
+                "num_experts": cfg.num_experts,
+                "num_experts_per_tok": cfg.num_experts_per_tok,
             }, f)
 
 

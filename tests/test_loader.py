@@ -163,3 +163,23 @@ def test_engine_tp_rank_load_checkpoint(tmp_path):
                        src.params["layers.1.w_down.weight"][:, i_local:])
     assert torch.equal(eng.params["final_norm.weight"],
                        src.params["final_norm.weight"])
+
+
+def test_config_roundtrip_preserves_family_fields(tmp_path):
+    """config.json must carry the family-defining fields (qkv_bias, MoE
+    shape, norm_eps) — a dropped field silently loads a different
+    architecture."""
+    import torch
+
+    from fma_amd.models import loader
+    from fma_amd.models.llama import LlamaConfig
+
+    for preset in ("tiny", "tiny-qwen", "tiny-moe"):
+        cfg = LlamaConfig.by_name(preset)
+        d = tmp_path / preset
+        loader.save_params({"x": torch.zeros(1)}, str(d), cfg)
+        back = loader.config_from_dir(str(d))
+        assert back.qkv_bias == cfg.qkv_bias, preset
+        assert back.num_experts == cfg.num_experts, preset
+        assert back.num_experts_per_tok == cfg.num_experts_per_tok, preset
+        assert back.norm_eps == cfg.norm_eps, preset
