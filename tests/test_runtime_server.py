@@ -63,3 +63,26 @@ def test_stats(client):
     body = client.get("/stats").json()
     assert body["model"] == "tiny"
     assert body["sleep_count"] >= 1
+
+
+def test_runtime_serves_all_three_families():
+    """ServingRuntime boots each model family and answers the actuation
+    surface (/is_sleeping semantics via the runtime object) + greedy
+    completion round-trips."""
+    import torch
+
+    from fma_amd.runtime.server import ServingRuntime, parse_options
+
+    for preset in ("tiny", "tiny-qwen", "tiny-moe"):
+        rt = ServingRuntime(parse_options(f"--model {preset} --seed 3"))
+        r = rt.rt
+        assert not r.is_sleeping()
+        eng = r.engine if hasattr(r, "engine") else r
+        toks = torch.randint(0, eng.cfg.vocab_size, (1, 5))
+        out1 = r.generate(toks, max_new_tokens=3).clone()
+        r.sleep(1)
+        assert r.is_sleeping()
+        r.wake_up()
+        assert torch.equal(r.generate(toks, max_new_tokens=3), out1), preset
+        if hasattr(r, "stop"):
+            r.stop()
