@@ -53,11 +53,28 @@ class LlamaConfig:
     #: num_experts per token; each expert is a llama-shaped gated MLP.
     num_experts: int = 0
     num_experts_per_tok: int = 2
+    #: EP: partition WHOLE experts across ranks (attention stays
+    #: Megatron-TP); non-owners contribute zero and the layer's single
+    #: all-reduce sums the routed outputs. False = Megatron-shard every
+    #: expert's MLP like the dense path.
+    expert_parallel: bool = False
     dtype: torch.dtype = torch.bfloat16
 
     @property
     def head_dim(self) -> int:
         return self.hidden_size // self.num_heads
+
+    def local_experts(self, tp_rank: int = 0, tp_size: int = 1
+                      ) -> range:
+        """Global indices of the experts this rank holds."""
+        if not self.num_experts:
+            return range(0)
+        if not (self.expert_parallel and tp_size > 1):
+            return range(self.num_experts)
+        assert self.num_experts % tp_size == 0, \
+            "EP needs num_experts divisible by tp_size"
+        per = self.num_experts // tp_size
+        return range(tp_rank * per, (tp_rank + 1) * per)
 
     # -- presets ------------------------------------------------------------
 
@@ -193,17 +210,21 @@ class LlamaConfig:
                 (p + "mlp_norm.weight", (h,), d),
             ]
             if self.num_experts:
-                # Mixtral-family: router replicated, every expert a
-                # Megatron-sharded gated MLP (leaf names w_gate/w_up/
-                # w_down keep the same col/row-parallel slicing rules)
+                # Mixtral-family: router replicated. Default: every
+                # expert Megatron-sharded (leaf names w_gate/w_up/w_down
+                # keep the col/row-parallel slicing rules). EP: this
+                # rank holds its share of WHOLE experts (global indices
+                # preserved in the names so checkpoints map naturally).
                 specs.append((p + "router.weight",
                               (self.num_experts, h), d))
-                for e in range(self.num_experts):
+                for e in self.local_experts(tp_rank, tp_size):
                     ep = p + f"experts.{e}."
+                    ei = self.intermediate_size if (
+                        self.expert_parallel and tp_size > 1) else i_local
                     specs += [
-                        (ep + "w_gate.weight", (i_local, h), d),
-                        (ep + "w_up.weight", (i_local, h), d),
-                        (ep + "w_down.weight", (h, i_local), d),
+                        (ep + "w_gate.weight", (ei, h), d),
+                        (ep + "w_up.weight", (ei, h), d),
+                        (ep + "w_down.weight", (h, ei), d),
                     ]
             else:
                 specs += [
@@ -325,10 +346,13 @@ class LlamaModel:
         logits = fast_linear(h, P[p + "router.weight"]).float()
         topw, topi = torch.topk(logits, cfg.num_experts_per_tok, dim=-1)
         topw = torch.softmax(topw, dim=-1)
+        mine = set(cfg.local_experts(self.tp_rank, self.tp_size))
         if B * T == 1:
             out = torch.zeros_like(x)
             idx = topi.reshape(-1).tolist()  # k tiny ints; eager decode
             for j, e in enumerate(idx):
+                if e not in mine:
+                    continue  # EP: another rank owns it; all-reduce sums
                 ep = p + f"experts.{e}."
                 gate, up = fast_linear_multi(
                     h, (P[ep + "w_gate.weight"], P[ep + "w_up.weight"]))
@@ -342,7 +366,7 @@ class LlamaModel:
             tw = topw.view(B * T, -1)
             out = torch.zeros(B * T, H, dtype=torch.float32,
                               device=x.device)
-            for e in range(cfg.num_experts):
+            for e in mine:
                 mask = ti == e
                 tok = mask.any(-1).nonzero(as_tuple=True)[0]
                 if tok.numel() == 0:

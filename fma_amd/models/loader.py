@@ -112,7 +112,8 @@ _ROW_PARALLEL = {"wo", "w_down"}
 
 def shard_slice(name: str, tensor: torch.Tensor, tp_rank: int,
                 tp_size: int,
-                local_rows: "int | None" = None) -> torch.Tensor:
+                local_rows: "int | None" = None,
+                local_cols: "int | None" = None) -> torch.Tensor:
     """Slice one full (unsharded) checkpoint tensor down to the Megatron
     shard that rank `tp_rank` of `tp_size` holds. Views, no copies.
 
@@ -144,8 +145,12 @@ def shard_slice(name: str, tensor: torch.Tensor, tp_rank: int,
         return tensor[tp_rank * step:(tp_rank + 1) * step]
     if leaf in _ROW_PARALLEL:
         n = tensor.shape[1]
-        if n % tp_size != 0:
-            raise ValueError(f"{name}: dim1 {n} not divisible by tp_size")
+        if local_cols is not None and local_cols == n:
+            return tensor  # expert-parallel: this rank holds it whole
+        if n % tp_size != 0 or (local_cols is not None
+                                and local_cols * tp_size != n):
+            raise ValueError(f"{name}: dim1 {n} does not match the local "
+                             f"shard at tp_size {tp_size}")
         step = n // tp_size
         return tensor[:, tp_rank * step:(tp_rank + 1) * step]
     return tensor
@@ -153,7 +158,7 @@ def shard_slice(name: str, tensor: torch.Tensor, tp_rank: int,
 
 def load_into_params(path: str, params: Dict[str, torch.Tensor],
                      strict: bool = True, tp_rank: int = 0,
-                     tp_size: int = 1) -> int:
+                     tp_size: int = 1, skip=None) -> int:
     """Copy checkpoint tensors into existing (arena-view) parameters.
 
     Returns the number of tensors loaded. Checkpoints store the full
@@ -164,13 +169,17 @@ def load_into_params(path: str, params: Dict[str, torch.Tensor],
     seen = set()
     for name, tensor in iter_safetensors(path):
         if name not in params:
+            if skip is not None and skip(name):
+                continue  # e.g. expert-parallel: another rank's expert
             if strict:
                 raise KeyError(f"checkpoint tensor {name!r} has no "
                                "matching parameter")
             continue
         p = params[name]
         tensor = shard_slice(name, tensor, tp_rank, tp_size,
-                             local_rows=p.shape[0] if p.dim() else None)
+                             local_rows=p.shape[0] if p.dim() else None,
+                             local_cols=p.shape[1] if p.dim() > 1
+                             else None)
         if tuple(tensor.shape) != tuple(p.shape):
             raise ValueError(f"shape mismatch for {name}: checkpoint "
                              f"shard {tuple(tensor.shape)} vs param "

@@ -182,13 +182,18 @@ class ActuationEngine:
         def stage(item):
             name, tensor = item
             if name not in self.layout:
+                if self.cfg.num_experts and self.cfg.expert_parallel \
+                        and ".experts." in name:
+                    return None  # another rank's whole expert (EP)
                 raise KeyError(f"checkpoint tensor {name!r} unknown")
             off, shape, dtype = self.layout[name]
             tensor = loader.shard_slice(tensor=tensor, name=name,
                                         tp_rank=self.tp_rank,
                                         tp_size=self.tp_size,
                                         local_rows=shape[0] if shape
-                                        else None)
+                                        else None,
+                                        local_cols=shape[1]
+                                        if len(shape) > 1 else None)
             if tuple(tensor.shape) != tuple(shape):
                 raise ValueError(f"shape mismatch for {name}")
             raw = tensor.to(dtype).contiguous().view(torch.uint8).view(-1)

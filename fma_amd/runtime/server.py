@@ -47,6 +47,9 @@ def parse_options(options: str) -> argparse.Namespace:
     ap.add_argument("--host", default=None)
     ap.add_argument("--tensor-parallel-size", type=int, default=1)
     ap.add_argument("--enable-sleep-mode", action="store_true")
+    ap.add_argument("--expert-parallel", action="store_true",
+                    help="MoE: partition whole experts across TP ranks "
+                         "(attention stays Megatron-TP)")
     ap.add_argument("--max-model-len", type=int, default=None)
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--served-model-name", default=None)
@@ -68,7 +71,7 @@ def _free_port() -> int:
 
 def _worker_entry(rank: int, world: int, master_port: int,
                   model_name: str, max_model_len: Optional[int],
-                  seed: int) -> None:
+                  seed: int, expert_parallel: bool = False) -> None:
     # ranks 1..N-1: build the engine shard and serve TP commands forever
     # (workers stay in the instance's process group so the launcher's
     # killpg on force-stop reaps them too)
@@ -84,6 +87,8 @@ def _worker_entry(rank: int, world: int, master_port: int,
         cfg = LlamaConfig.by_name(model_name)
     if max_model_len:
         cfg.max_seq_len = max_model_len
+    if expert_parallel:
+        cfg.expert_parallel = True
     eng = ActuationEngine(cfg, device_index=ctx.device_index,
                           tp_rank=rank, tp_size=world,
                           tp_group=ctx.device_group, seed=seed,
@@ -110,7 +115,8 @@ class ServingRuntime:
         for r in range(1, world):
             p = mp.Process(target=_worker_entry,
                            args=(r, world, master_port, args.model,
-                                 args.max_model_len, args.seed),
+                                 args.max_model_len, args.seed,
+                                 getattr(args, "expert_parallel", False)),
                            daemon=True)
             p.start()
             self.workers.append(p)
@@ -127,6 +133,8 @@ class ServingRuntime:
             cfg = LlamaConfig.by_name(args.model)
         if args.max_model_len:
             cfg.max_seq_len = args.max_model_len
+        if getattr(args, "expert_parallel", False):
+            cfg.expert_parallel = True
         engine = ActuationEngine(cfg, device_index=ctx.device_index,
                                  tp_rank=0, tp_size=world,
                                  tp_group=ctx.device_group, seed=args.seed,

@@ -248,3 +248,68 @@ def test_tp_moe_checkpoint_matches_full_model(tmp_path):
         env=dict(os.environ, PYTHONPATH=root))
     assert res.returncode == 0, f"stdout={res.stdout}\nstderr={res.stderr}"
     assert "TP_MOE_OK" in res.stdout
+
+
+EP_PROBE = r"""
+# Expert parallelism: whole experts partitioned across 2 ranks
+# (attention Megatron-TP); decode must match the full model.
+import os
+os.environ.setdefault("FMA_FAKE_GPU", "1")
+os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+import sys
+sys.path.insert(0, "__ROOT__")
+import torch
+from fma_amd.runtime.server import ServingRuntime, parse_options
+
+ref = torch.load(os.path.join("__TMP__", "ref.pt"), weights_only=True)
+toks = torch.load(os.path.join("__TMP__", "toks.pt"), weights_only=True)
+rt = ServingRuntime(parse_options(
+    "--model " + os.path.join("__TMP__", "moe-ckpt")
+    + " --tensor-parallel-size 2 --expert-parallel --seed 4"))
+eng = rt.rt.engine
+# rank 0 holds the first half of the experts, whole
+assert "layers.0.experts.0.w_gate.weight" in eng.params
+assert "layers.0.experts.3.w_gate.weight" not in eng.params
+assert eng.params["layers.0.experts.0.w_gate.weight"].shape[0] == 96
+out = rt.rt.generate(toks, max_new_tokens=3)
+assert torch.equal(out, ref), (out, ref)
+before = rt.rt.generate(toks, max_new_tokens=2).clone()
+rt.rt.sleep(1)
+rt.rt.wake_up()
+assert torch.equal(rt.rt.generate(toks, max_new_tokens=2), before)
+rt.rt.stop()
+print("TP_EP_OK")
+"""
+
+
+def test_expert_parallel_matches_full_model(tmp_path):
+    """EP=2 over a full MoE checkpoint: each rank loads only its whole
+    experts, the per-layer all-reduce sums routed outputs, and greedy
+    decode is token-identical to the single-process full model."""
+    import torch
+    from fma_amd.models import loader
+    from fma_amd.models.llama import LlamaConfig
+    from fma_amd.runtime.engine import ActuationEngine
+
+    os.environ.setdefault("FMA_FAKE_GPU", "1")
+    cfg = LlamaConfig(name="epck", vocab_size=64, hidden_size=64,
+                      intermediate_size=96, num_layers=2, num_heads=4,
+                      num_kv_heads=2, max_seq_len=32, num_experts=4,
+                      num_experts_per_tok=2)
+    src = ActuationEngine(cfg, seed=41)
+    loader.save_params(src.params, str(tmp_path / "moe-ckpt"), cfg)
+    toks = torch.randint(0, cfg.vocab_size, (1, 5),
+                         generator=torch.Generator().manual_seed(11))
+    torch.save(toks, str(tmp_path / "toks.pt"))
+    torch.save(src.model.generate(toks, max_new_tokens=3),
+               str(tmp_path / "ref.pt"))
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    probe = EP_PROBE.replace("__ROOT__", root).replace(
+        "__TMP__", str(tmp_path))
+    res = subprocess.run(
+        [sys.executable, "-c", probe],
+        capture_output=True, text=True, timeout=150,
+        env=dict(os.environ, PYTHONPATH=root))
+    assert res.returncode == 0, f"stdout={res.stdout}\nstderr={res.stderr}"
+    assert "TP_EP_OK" in res.stdout
