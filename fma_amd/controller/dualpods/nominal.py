@@ -52,7 +52,12 @@ def strategic_merge(base: Any, patch: Any) -> Any:
     if patch is None:
         return None
     if isinstance(base, dict) and isinstance(patch, dict):
-        out = dict(base)
+        # pure merge: untouched branches are COPIED, never aliased, so
+        # mutating the result can't corrupt the caller's base
+        out = {k: copy.deepcopy(v) for k, v in base.items()
+               if k not in patch}
+        out.update({k: copy.deepcopy(base[k]) for k in patch
+                    if k in base})
         for k, v in patch.items():
             if v is None:
                 out.pop(k, None)
@@ -67,7 +72,7 @@ def strategic_merge(base: Any, patch: Any) -> Any:
         return out
     if isinstance(base, list) and isinstance(patch, list):
         if all(isinstance(e, dict) and "name" in e for e in base + patch):
-            by_name = {e["name"]: e for e in base}
+            by_name = {e["name"]: copy.deepcopy(e) for e in base}
             order = [e["name"] for e in base]
             for e in patch:
                 if e["name"] in by_name:

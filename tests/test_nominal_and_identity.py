@@ -100,3 +100,66 @@ def test_instance_id_properties(opts, port, gpus):
     assert a != instance_id({"port": (port % 65534) + 1, "options": opts},
                             gpus)
     assert a != instance_id(msc, gpus + ["GPU-9"])
+
+
+def test_strategic_merge_properties_fuzz():
+    """Property sweep over random dict/list structures: (1) merging an
+    empty patch is identity; (2) patch keys win; (3) named-list merge
+    preserves base order and appends new names; (4) the result never
+    aliases mutable state of base or patch."""
+    import random
+
+    from fma_amd.controller.dualpods import nominal
+
+    rng = random.Random(13)
+
+    def rand_value(depth):
+        r = rng.random()
+        if depth >= 3 or r < 0.35:
+            return rng.choice([1, "x", "y", True, None if depth else 2])
+        if r < 0.6:
+            return {rng.choice("abcd"): rand_value(depth + 1)
+                    for _ in range(rng.randint(0, 3))}
+        if r < 0.8:
+            return [{"name": f"n{i}", "v": rand_value(depth + 1)}
+                    for i in range(rng.randint(0, 3))]
+        return [rand_value(depth + 1) for _ in range(rng.randint(0, 3))]
+
+    import copy
+    for _ in range(300):
+        base = {rng.choice("abcdef"): rand_value(0)
+                for _ in range(rng.randint(0, 4))}
+        base = {k: v for k, v in base.items() if v is not None}
+        base_snapshot = copy.deepcopy(base)
+
+        # (1) empty patch is identity
+        assert nominal.strategic_merge(base, {}) == base
+
+        patch = {rng.choice("abcdef"): rand_value(0)
+                 for _ in range(rng.randint(0, 4))}
+        patch_snapshot = copy.deepcopy(patch)
+        out = nominal.strategic_merge(base, patch)
+        # inputs never mutated
+        assert base == base_snapshot
+        assert patch == patch_snapshot
+        # (2) scalar patch keys win; None deletes
+        for k, v in patch.items():
+            if v is None:
+                assert k not in out
+            elif not isinstance(v, (dict, list)):
+                assert out[k] == v
+        # (4) no aliasing: mutating the result leaves inputs intact
+        for k in list(out):
+            if isinstance(out[k], dict):
+                out[k]["__mut__"] = 1
+            elif isinstance(out[k], list):
+                out[k].append("__mut__")
+        assert base == base_snapshot
+        assert patch == patch_snapshot
+
+    # (3) named-list semantics, explicit
+    base = {"containers": [{"name": "a", "x": 1}, {"name": "b", "x": 2}]}
+    patch = {"containers": [{"name": "b", "x": 9}, {"name": "c", "x": 3}]}
+    out = nominal.strategic_merge(base, patch)
+    assert [c["name"] for c in out["containers"]] == ["a", "b", "c"]
+    assert out["containers"][1]["x"] == 9
