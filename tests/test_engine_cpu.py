@@ -336,3 +336,58 @@ def test_moe_prefill_and_decode_paths_agree():
     out = eng.generate(toks, max_new_tokens=4)
     # re-running is deterministic
     assert torch.equal(out, eng.generate(toks, max_new_tokens=4))
+
+
+def test_plan_layout_properties_fuzz():
+    """Property sweep of the arena layout planner: 256-B alignment,
+    no overlap, no tensor straddling a slab boundary, slabs exactly
+    cover the arena, byte-count conservation."""
+    import random
+
+    import torch
+
+    from fma_amd.ops.actuation import ARENA_ALIGN, align_up, plan_layout
+
+    rng = random.Random(23)
+    dtypes = [torch.bfloat16, torch.float32, torch.float16]
+    for _ in range(200):
+        n = rng.randint(1, 40)
+        specs = []
+        for i in range(n):
+            shape = tuple(rng.randint(1, 64)
+                          for _ in range(rng.randint(1, 3)))
+            specs.append((f"t{i}", shape, rng.choice(dtypes)))
+        slab = rng.choice([0, 256, 4096, 1 << 16, 1 << 20])
+        layout, total, slabs = plan_layout(specs, slab_bytes=slab)
+
+        spans = []
+        for name, shape, dtype in specs:
+            off, lshape, ldtype = layout[name]
+            assert lshape == shape and ldtype == dtype
+            assert off % ARENA_ALIGN == 0
+            numel = 1
+            for d in shape:
+                numel *= d
+            nbytes = numel * torch.empty(0, dtype=dtype).element_size()
+            spans.append((off, off + nbytes, name))
+        spans.sort()
+        for (a0, a1, an), (b0, b1, bn) in zip(spans, spans[1:]):
+            assert a1 <= b0, f"overlap {an}/{bn}"
+        assert sum(slabs) == total
+        assert all(sz > 0 for sz in slabs)
+        # no tensor straddles a slab boundary
+        bounds = []
+        acc = 0
+        for sz in slabs:
+            acc += sz
+            bounds.append(acc)
+        for a0, a1, name in spans:
+            for b in bounds[:-1]:
+                assert not (a0 < b < a1), f"{name} straddles slab at {b}"
+        # conservation: total >= sum of aligned tensor sizes
+        assert total >= sum(align_up(max(
+            (lambda s: __import__('math').prod(s))(shape) *
+            torch.empty(0, dtype=dt).element_size(), 1))
+            for _, shape, dt in specs) - len(specs) * 0  # exact when slab=0
+        if slab == 0:
+            assert len(slabs) == 1
