@@ -442,3 +442,73 @@ def test_two_level_queue_no_item_lost_fuzz():
         assert not missing, f"{len(missing)} items never processed"
     finally:
         w.stop()
+
+
+def test_store_indexes_consistent_under_random_crud_fuzz():
+    """Property: after any random CRUD sequence (with finalizers and
+    status writes mixed in), every registered index returns exactly what
+    a brute-force scan returns."""
+    import random
+
+    from fma_amd.api import contracts as C
+    from fma_amd.store.indexes import POD_INDEXES, install_pod_indexes
+    from fma_amd.store.memstore import ApiError, MemStore
+
+    rng = random.Random(77)
+    st = MemStore()
+    install_pod_indexes(st)
+    names = [f"p{i}" for i in range(12)]
+    anns_keys = [C.REQUESTER_ANNOTATION, C.NOMINAL_ANNOTATION,
+                 C.INFERENCE_SERVER_CONFIG_ANNOTATION,
+                 C.ACCELERATORS_ANNOTATION]
+    for step in range(400):
+        name = rng.choice(names)
+        op = rng.random()
+        try:
+            if op < 0.4:
+                pod = ob.new_object(
+                    "Pod", name,
+                    labels={C.COMPONENT_LABEL: C.LAUNCHER_COMPONENT}
+                    if rng.random() < 0.4 else {},
+                    annotations={k: rng.choice(["", "a b", "h1", "GPU-0,GPU-1"])
+                                 for k in rng.sample(anns_keys,
+                                                     rng.randint(0, 3))},
+                    spec={"nodeName": rng.choice(["n1", "n2", ""]),
+                          "containers": []})
+                if rng.random() < 0.2:
+                    pod["metadata"]["finalizers"] = ["t/f"]
+                st.create(pod)
+            elif op < 0.7:
+                cur = st.try_get("Pod", name)
+                if cur is None:
+                    continue
+                k = rng.choice(anns_keys)
+                ob.annotations_of(cur)[k] = rng.choice(
+                    ["", "u2 q", "h2", "GPU-2"])
+                if rng.random() < 0.3:
+                    cur["metadata"]["finalizers"] = []
+                st.update(cur)
+            elif op < 0.9:
+                st.delete("Pod", name)
+            else:
+                cur = st.try_get("Pod", name)
+                if cur is not None:
+                    cur["status"] = {"phase": "Running"}
+                    st.update(cur, subresource="status")
+        except ApiError:
+            pass
+
+        if step % 50 == 49:
+            pods = st.list("Pod")
+            for iname, fn in POD_INDEXES.items():
+                expect = {}
+                for p in pods:
+                    for key in fn(p) or []:
+                        expect.setdefault(key, set()).add(ob.name_of(p))
+                keys = set(expect)
+                for key in keys:
+                    got = {ob.name_of(p)
+                           for p in st.index_get("Pod", iname, key)}
+                    assert got == expect[key], (iname, key, step)
+                # spot-check an absent key returns empty
+                assert st.index_get("Pod", iname, "no-such-key") == []
