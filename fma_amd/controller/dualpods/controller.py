@@ -199,6 +199,10 @@ class DualPodsController:
             return self.node_locks.setdefault(node, threading.Lock())
 
     def _process(self, item: Tuple) -> bool:
+        if self._stop.is_set():
+            # a stopped controller must not keep writing: a replacement
+            # (restart) may already be reconciling the same objects
+            return DONE
         kind, node = item[0], item[1]
         with self._node_lock(node):
             if kind == "infsvr":

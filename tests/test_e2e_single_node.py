@@ -258,6 +258,8 @@ def test_controller_restart_recovers_live_binding(cluster):
              desc="requester ready")
 
     cluster["ctl"].stop()
+    for th in cluster["ctl"].workers.threads:
+        th.join(timeout=35)  # drain in-flight reconciles before handover
     ctl2 = DualPodsController(store, HttpAdapter(), ControllerConfig())
     ctl2.start()
     try:

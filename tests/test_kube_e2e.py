@@ -142,6 +142,8 @@ def test_kube_controller_restart_recovery(kube_cluster):
              desc="requester ready")
 
     kube_cluster["ctl"].stop()
+    for th in kube_cluster["ctl"].workers.threads:
+        th.join(timeout=35)  # drain in-flight reconciles before handover
     ctl2 = DualPodsController(
         KubeStore(base, actor="dual-pods-controller"), HttpAdapter(),
         ControllerConfig())
