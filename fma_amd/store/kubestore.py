@@ -90,15 +90,54 @@ def _raise_status(r: httpx.Response) -> None:
     raise ApiError(code, msg)
 
 
+#: standard in-cluster service-account mount (k8s convention)
+SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
+
+
 class KubeStore:
     def __init__(self, base_url: str, actor: str = "system",
-                 timeout: float = 30.0):
+                 timeout: float = 30.0, bearer_token: Optional[str] = None,
+                 ca_cert: Optional[str] = None):
         self.base = base_url.rstrip("/")
         self.default_actor = actor
-        self._client = httpx.Client(timeout=timeout)
+        self._bearer = bearer_token
+        # reads and watch streams need the token too, so it rides on the
+        # client as a default header (writes still go through _hdr)
+        base_headers = ({"Authorization": f"Bearer {bearer_token}"}
+                        if bearer_token else {})
+        self._client = httpx.Client(timeout=timeout, headers=base_headers,
+                                    verify=ca_cert if ca_cert else True)
         self._mu = threading.Lock()
         self._tokens: Dict[int, Dict[str, str]] = {}
         self._next_token = 1
+
+    @classmethod
+    def in_cluster(cls, actor: str = "system", timeout: float = 30.0,
+                   sa_dir: str = SA_DIR) -> "KubeStore":
+        """Build from the in-Pod service-account mount, the way client-go's
+        rest.InClusterConfig does (the reference controllers run this way
+        via their Helm chart; ours deploy/charts/fma-amd does the same).
+        Returns the store; `in_cluster_namespace` gives the namespace."""
+        import os
+        host = os.environ["KUBERNETES_SERVICE_HOST"]
+        port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+        if ":" in host and not host.startswith("["):
+            host = f"[{host}]"  # IPv6 service host
+        with open(os.path.join(sa_dir, "token")) as f:
+            token = f.read().strip()
+        ca = os.path.join(sa_dir, "ca.crt")
+        return cls(f"https://{host}:{port}", actor=actor, timeout=timeout,
+                   bearer_token=token,
+                   ca_cert=ca if os.path.exists(ca) else None)
+
+    @staticmethod
+    def in_cluster_namespace(sa_dir: str = SA_DIR) -> str:
+        import os
+        try:
+            with open(os.path.join(sa_dir, "namespace")) as f:
+                return f.read().strip()
+        except OSError:
+            return "default"
 
     # -- plumbing -----------------------------------------------------------
 
@@ -119,6 +158,10 @@ class KubeStore:
 
     def _hdr(self, actor: Optional[str]) -> Dict[str, str]:
         a = actor or self.default_actor
+        if self._bearer:
+            # real apiserver: the SA token IS the identity; the username
+            # the VAP bindings see comes from token review, not a header
+            return {"Authorization": f"Bearer {self._bearer}"}
         return {"X-Remote-User": FMA_SERVICE_ACCOUNTS.get(a, a)}
 
     # -- CRUD ---------------------------------------------------------------

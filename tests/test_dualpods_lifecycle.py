@@ -264,3 +264,19 @@ def test_switching_instances_in_one_launcher():
     assert ob.annotations_of(lp)[C.INSTANCE_ID_ANNOTATION] == w["iid"]
     assert not w["inst_srv"].sleeping  # A woken again
     assert w["inst_srv"].wakes == 2
+
+
+def test_stopped_controller_does_not_write():
+    """After stop(), queued/in-flight reconciles become no-ops so a
+    replacement controller (restart) never races a stale writer: the
+    reference gets this for free by killing the process; our in-process
+    restart needs the stop gate."""
+    w = mk_world()
+    w["ctl"].stop()
+    from fma_amd.controller.dualpods.controller import DONE
+    assert w["ctl"]._process(infsvr_item(w["store"])) is DONE
+    lp = w["store"].get("Pod", "launcher1")
+    assert C.REQUESTER_ANNOTATION not in lp["metadata"].get(
+        "annotations", {})
+    assert not w["launcher"].wake_calls if hasattr(
+        w["launcher"], "wake_calls") else True

@@ -308,3 +308,46 @@ def test_watch_survives_apiserver_restart():
     finally:
         stop.set()
         srv2.__exit__()
+
+
+# ---------------------------------------------------------------------------
+# in-cluster construction (the Helm chart's --backend=in-cluster path)
+# ---------------------------------------------------------------------------
+
+
+def test_in_cluster_store_from_sa_mount(tmp_path, monkeypatch):
+    """KubeStore.in_cluster builds from the standard service-account
+    mount the way client-go's rest.InClusterConfig does; the bearer token
+    rides on every request (reads and watch streams included)."""
+    (tmp_path / "token").write_text("sekrit-token\n")
+    (tmp_path / "namespace").write_text("prod-ns")
+    monkeypatch.setenv("KUBERNETES_SERVICE_HOST", "10.0.0.1")
+    monkeypatch.setenv("KUBERNETES_SERVICE_PORT", "6443")
+    ks = KubeStore.in_cluster(actor="dual-pods-controller",
+                              sa_dir=str(tmp_path))
+    assert ks.base == "https://10.0.0.1:6443"
+    assert ks._client.headers["Authorization"] == "Bearer sekrit-token"
+    # write-path headers carry the token, not the test-double username
+    assert ks._hdr(None) == {"Authorization": "Bearer sekrit-token"}
+    assert KubeStore.in_cluster_namespace(str(tmp_path)) == "prod-ns"
+    # missing mount falls back to default namespace
+    assert KubeStore.in_cluster_namespace(str(tmp_path / "nope")) == \
+        "default"
+
+
+def test_controller_entrypoints_backend_flag():
+    """Both controller mains accept --backend {store,kube,in-cluster} and
+    build the right store (the chart deploys with --backend=in-cluster)."""
+    from fma_amd.controller.dualpods.__main__ import make_store
+
+    class A:
+        backend = "kube"
+        store_url = "http://127.0.0.1:9"
+        namespace = "ns1"
+
+    st, ns = make_store(A(), "dual-pods-controller")
+    assert isinstance(st, KubeStore) and ns == "ns1"
+    A.backend = "store"
+    from fma_amd.store.client import StoreClient
+    st, ns = make_store(A(), "dual-pods-controller")
+    assert isinstance(st, StoreClient)
