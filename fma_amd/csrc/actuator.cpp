@@ -687,8 +687,10 @@ double restore_from_host_overlapped(std::vector<at::Tensor> tensors,
   // device blob: all chunks' descs + the full prefix array
   const size_t db = static_cast<size_t>(total_descs) * sizeof(FmaCopyDesc);
   DevBlob dev;
-  dev.upload({}, prefix_h, ctx.kernel_stream);  // placeholder — see below
-  // DevBlob::upload sized for empty descs; allocate our own desc area:
+  // prefix array uploads whole; descriptor POINTERS cannot be known yet
+  // (storages materialize chunk-by-chunk below), so upload no descs here
+  // and stage each chunk's slice through the pinned bounce buffer instead
+  dev.upload({}, prefix_h, ctx.kernel_stream);
   FmaCopyDesc* descs_dev = nullptr;
   FMA_HIP_CHECK(hipMalloc(&descs_dev, std::max<size_t>(db, 16)));
   auto* pinned =
