@@ -251,12 +251,6 @@ class MemStore:
             cur = self._objects.get(key)
             if cur is None:
                 raise NotFound(f"{kind} {key[1]}/{key[2]} not found")
-            if expect_uid is not None and ob.uid_of(cur) != expect_uid:
-                raise Conflict(f"uid mismatch on {key}")
-            rv_expected = expect_rv if expect_rv is not None else ob.rv_of(obj)
-            if rv_expected and rv_expected != ob.rv_of(cur):
-                raise Conflict(
-                    f"resourceVersion conflict on {key}: have {ob.rv_of(cur)}, caller {rv_expected}")
             if subresource == "status":
                 new = ob.deepcopy(cur)
                 new["status"] = obj.get("status", {})
@@ -271,7 +265,19 @@ class MemStore:
                 nm["deletionTimestamp"] = cm.get("deletionTimestamp")
                 if new.get("spec") != cur.get("spec"):
                     nm["generation"] = cm.get("generation", 1) + 1
+            # admission BEFORE the optimistic-concurrency check: a real
+            # apiserver runs (validating) admission ahead of the storage
+            # commit where the RV conflict is detected, so a stale update
+            # that also violates policy is denied as Invalid, not Conflict
+            # (the apiserver double behaves the same; the differential
+            # fuzz pins the order)
             self._admit("UPDATE", cur, new, actor)
+            if expect_uid is not None and ob.uid_of(cur) != expect_uid:
+                raise Conflict(f"uid mismatch on {key}")
+            rv_expected = expect_rv if expect_rv is not None else ob.rv_of(obj)
+            if rv_expected and rv_expected != ob.rv_of(cur):
+                raise Conflict(
+                    f"resourceVersion conflict on {key}: have {ob.rv_of(cur)}, caller {rv_expected}")
             self._revision_stamp(new)
             self._objects[key] = new
             self._emit("MODIFIED", kind, new)
