@@ -70,8 +70,9 @@ class _Step:
         self.name = rng.choice(NAMES)
         self.actor = rng.choice(ACTORS)
         self.op = rng.choices(
-            ["create", "update", "stale_update", "delete", "bad_delete"],
-            weights=[30, 40, 10, 15, 5])[0]
+            ["create", "update", "stale_update", "delete", "bad_delete",
+             "status", "finalize"],
+            weights=[28, 32, 8, 13, 4, 8, 7])[0]
         self.rng_state = rng.getstate()
 
     def run(self, store, is_kube):
@@ -87,6 +88,8 @@ class _Step:
 
         if self.op == "create":
             obj = _mk_pod(rng, name) if self.kind_pod else _mk_crd(rng, name)
+            if rng.random() < 0.3:
+                obj["metadata"]["finalizers"] = ["dual-pods.llm-d.ai/test"]
             self._kind = obj["kind"]
             return call(store.create, obj)
         # the remaining ops need an existing object of SOME kind
@@ -113,6 +116,16 @@ class _Step:
             if self.op == "stale_update":
                 meta["resourceVersion"] = "1"
             return call(store.update, cur)
+        if self.op == "status":
+            cur = ob.deepcopy(target)
+            cur["status"] = {"phase": rng.choice(["Running", "Pending"]),
+                             "marker": rng.randrange(3)}
+            return call(store.update, cur, subresource="status")
+        if self.op == "finalize":
+            # drop finalizers; on a deleting object this completes removal
+            cur = ob.deepcopy(target)
+            cur["metadata"]["finalizers"] = []
+            return call(store.update, cur)
         if self.op == "delete":
             return call(store.delete, kind, name)
         return call(store.delete, kind, name, expect_uid="wrong-uid")
@@ -123,6 +136,8 @@ def _strip(obj):
     m = o.get("metadata", {})
     for k in ("uid", "creationTimestamp", "managedFields"):
         m.pop(k, None)
+    if m.get("deletionTimestamp"):
+        m["deletionTimestamp"] = "<set>"  # wall-clock; compare presence
     return o
 
 
@@ -184,3 +199,12 @@ def test_memstore_and_kube_double_agree(double, seed):
     assert set(mem_snap) == set(kube_snap)
     for key in mem_snap:
         assert mem_snap[key] == kube_snap[key], key
+
+    # the two backends must also have EMITTED the same watch-event
+    # sequence (type, kind, name) — the double is a thin wire layer, so
+    # its backing history is directly comparable
+    backing = double["backing"]
+    mem_events = [(e.type, e.kind, ob.name_of(e.obj)) for e in mem._history]
+    kube_events = [(e.type, e.kind, ob.name_of(e.obj))
+                   for e in backing._history]
+    assert mem_events == kube_events
