@@ -27,6 +27,7 @@ import yaml
 from fma_amd.api import contracts
 from fma_amd.controller.dualpods.identity import nominal_hash
 from fma_amd.store import objects as ob
+from fma_amd.store.merge import strategic_merge  # noqa: F401 (re-export)
 
 _TEMPLATE_RE = re.compile(r"\{\{\s*\.(\w+)\s*\}\}")
 
@@ -43,46 +44,6 @@ def render_template(text: str, provider_data: Dict[str, str]) -> str:
         return str(provider_data[key])
 
     return _TEMPLATE_RE.sub(sub, text)
-
-
-def strategic_merge(base: Any, patch: Any) -> Any:
-    """Strategic merge patch for the Pod-shaped subset: dicts merge
-    recursively; lists of objects with a ``name`` key merge by name;
-    other lists replace; ``None`` deletes a key."""
-    if patch is None:
-        return None
-    if isinstance(base, dict) and isinstance(patch, dict):
-        # pure merge: untouched branches are COPIED, never aliased, so
-        # mutating the result can't corrupt the caller's base
-        out = {k: copy.deepcopy(v) for k, v in base.items()
-               if k not in patch}
-        out.update({k: copy.deepcopy(base[k]) for k in patch
-                    if k in base})
-        for k, v in patch.items():
-            if v is None:
-                out.pop(k, None)
-            elif k in out:
-                merged = strategic_merge(out[k], v)
-                if merged is None:
-                    out.pop(k, None)
-                else:
-                    out[k] = merged
-            else:
-                out[k] = copy.deepcopy(v)
-        return out
-    if isinstance(base, list) and isinstance(patch, list):
-        if all(isinstance(e, dict) and "name" in e for e in base + patch):
-            by_name = {e["name"]: copy.deepcopy(e) for e in base}
-            order = [e["name"] for e in base]
-            for e in patch:
-                if e["name"] in by_name:
-                    by_name[e["name"]] = strategic_merge(by_name[e["name"]], e)
-                else:
-                    by_name[e["name"]] = copy.deepcopy(e)
-                    order.append(e["name"])
-            return [by_name[n] for n in order]
-        return copy.deepcopy(patch)
-    return copy.deepcopy(patch)
 
 
 def deindividualize(requester: Dict[str, Any]) -> Dict[str, Any]:

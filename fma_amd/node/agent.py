@@ -196,9 +196,15 @@ class NodeAgent:
             cur = self.store.try_get("Pod", pp.pod_name, self.ns)
             if cur is None or ob.uid_of(cur) != pp.uid:
                 return
-            ob.annotations_of(cur)[key] = value
             try:
-                self.store.update(cur, actor="node-agent")
+                # server-side merge patch: cannot lose a concurrent
+                # controller update the way read-modify-write can
+                # (the reference sidecar PATCHes for the same reason,
+                # launcher_pod_notifier.py:135-194)
+                self.store.patch(
+                    "Pod", pp.pod_name,
+                    {"metadata": {"annotations": {key: value}}}, self.ns,
+                    actor="node-agent")
             except (Conflict, NotFound):
                 pass
 

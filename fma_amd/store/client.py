@@ -101,6 +101,16 @@ class StoreClient:
             _raise_for(r.status_code, r.json().get("error", r.text))
         return r.json()
 
+    def patch(self, kind: str, name: str, patch: Dict[str, Any],
+              namespace: str = "default", actor: Optional[str] = None,
+              strategic: bool = False) -> Dict[str, Any]:
+        r = self._client.patch(
+            f"{self.base}/apis/{kind}/{namespace}/{name}", json=patch,
+            params={"strategic": int(strategic)}, headers=self._hdr(actor))
+        if r.status_code != 200:
+            _raise_for(r.status_code, r.json().get("error", r.text))
+        return r.json()
+
     def delete(self, kind: str, name: str, namespace: str = "default",
                actor: Optional[str] = None,
                expect_uid: Optional[str] = None,

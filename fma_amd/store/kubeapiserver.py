@@ -41,7 +41,8 @@ from concurrent.futures import ThreadPoolExecutor
 
 from fma_amd.store import cel
 from fma_amd.store import objects as ob
-from fma_amd.store.memstore import ApiError, MemStore, RevisionTooOld
+from fma_amd.store.memstore import (ApiError, Conflict, Invalid, MemStore,
+                                    RevisionTooOld)
 
 #: resource name <-> kind for everything the FMA stack touches
 CORE_RESOURCES = {
@@ -273,6 +274,43 @@ def create_app(store: Optional[MemStore] = None,
             return _err(e)
         return updated
 
+    async def handle_patch(request: Request, resource: str, group: str,
+                           ns: Optional[str], name: str):
+        """Server-side apply of merge / strategic-merge patches, content
+        type selected the k8s way. VAP admission sees the MERGED result
+        (as a real apiserver's admission chain does)."""
+        kind, namespaced = resolve(resource, group)
+        namespace = ns if namespaced else "default"
+        ctype = request.headers.get("content-type", "")
+        strategic = "strategic-merge-patch" in ctype
+        if not strategic and "merge-patch" not in ctype:
+            return JSONResponse(
+                status_body(415, f"unsupported patch type {ctype!r}",
+                            "UnsupportedMediaType"), status_code=415)
+        body = await request.json()
+        user = username_of(request)
+        from fma_amd.store.merge import merge_patch, strategic_merge
+        fn = strategic_merge if strategic else merge_patch
+        try:
+            for _ in range(16):
+                old = st.get(kind, name, namespace)
+                new = fn(old, body)
+                if not isinstance(new, dict):
+                    raise Invalid("patch must produce an object")
+                deny = vap.check_update(resource, group, old, new, user)
+                if deny:
+                    return JSONResponse(
+                        status_body(422, deny, "Invalid"), status_code=422)
+                try:
+                    return st.update(new, actor=user,
+                                     expect_rv=ob.rv_of(old))
+                except Conflict:
+                    continue
+            raise Conflict(f"patch on {kind} {namespace}/{name} kept "
+                           "conflicting after 16 attempts")
+        except ApiError as e:
+            return _err(e)
+
     async def handle_delete(request: Request, resource: str, group: str,
                             ns: Optional[str], name: str):
         kind, namespaced = resolve(resource, group)
@@ -333,6 +371,11 @@ def create_app(store: Optional[MemStore] = None,
                                     subresource="status")
         except ApiError as e:
             return _err(e)
+
+    @app.patch("/api/v1/namespaces/{ns}/{resource}/{name}")
+    async def core_patch(request: Request, ns: str, resource: str,
+                         name: str):
+        return await handle_patch(request, resource, "", ns, name)
 
     @app.delete("/api/v1/namespaces/{ns}/{resource}/{name}")
     async def core_delete(request: Request, ns: str, resource: str,
@@ -401,6 +444,11 @@ def create_app(store: Optional[MemStore] = None,
             return await handle_put(request, resource, FMA_GROUP, ns, name)
         except ApiError as e:
             return _err(e)
+
+    @app.patch(PREFIX + "/namespaces/{ns}/{resource}/{name}")
+    async def fma_patch(request: Request, ns: str, resource: str,
+                        name: str):
+        return await handle_patch(request, resource, FMA_GROUP, ns, name)
 
     @app.put(PREFIX + "/namespaces/{ns}/{resource}/{name}/status")
     async def fma_put_status(request: Request, ns: str, resource: str,

@@ -233,6 +233,24 @@ class KubeStore:
             _raise_status(r)
         return r.json()
 
+    def patch(self, kind: str, name: str, patch: Dict[str, Any],
+              namespace: str = "default", actor: str = None,
+              strategic: bool = False) -> Dict[str, Any]:
+        """Server-side merge patch with the real k8s content types
+        (application/merge-patch+json or …strategic-merge-patch+json) —
+        no resourceVersion, so concurrent patches of disjoint fields
+        cannot lose each other."""
+        ctype = ("application/strategic-merge-patch+json" if strategic
+                 else "application/merge-patch+json")
+        headers = dict(self._hdr(actor))
+        headers["Content-Type"] = ctype
+        r = self._client.request(
+            "PATCH", self.base + self._path(kind, namespace, name),
+            content=json.dumps(patch).encode(), headers=headers)
+        if r.status_code != 200:
+            _raise_status(r)
+        return r.json()
+
     def delete(self, kind: str, name: str, namespace: str = "default",
                actor: str = None, expect_uid: Optional[str] = None,
                expect_rv: Optional[str] = None) -> None:

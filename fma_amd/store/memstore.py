@@ -286,6 +286,31 @@ class MemStore:
                 self._remove_locked(key, actor)
             return ob.deepcopy(self._objects.get(key, new))
 
+    def patch(self, kind: str, name: str, patch: Dict[str, Any],
+              namespace: str = "default", actor: str = "system",
+              strategic: bool = False) -> Dict[str, Any]:
+        """Server-side merge patch (RFC 7386; ``strategic=True`` for
+        kubernetes strategic-merge list semantics). Unlike update, a
+        patch carries no resourceVersion: it applies onto whatever the
+        current object is, so concurrent patches of disjoint fields never
+        lose each other (the reference's notifier sidecar PATCHes the
+        instance-signature annotation for exactly this reason). Runs the
+        normal admission chain on the merged result."""
+        from fma_amd.store.merge import merge_patch, strategic_merge
+        fn = strategic_merge if strategic else merge_patch
+        for _ in range(16):
+            cur = self.get(kind, name, namespace)
+            new = fn(cur, patch)
+            if not isinstance(new, dict):
+                raise Invalid("patch must produce an object")
+            try:
+                return self.update(new, actor=actor,
+                                   expect_rv=ob.rv_of(cur))
+            except Conflict:
+                continue  # racer moved the object; re-read and re-merge
+        raise Conflict(f"patch on {kind} {namespace}/{name} kept "
+                       "conflicting after 16 attempts")
+
     def delete(self, kind: str, name: str, namespace: str = "default",
                actor: str = "system",
                expect_uid: Optional[str] = None,

@@ -86,6 +86,17 @@ def create_app(store: Optional[MemStore] = None) -> FastAPI:
         except ApiError as e:
             return err(e)
 
+    @app.patch("/apis/{kind}/{ns}/{name}")
+    async def patch(kind: str, ns: str, name: str, request: Request,
+                    strategic: int = 0):
+        body = await request.json()
+        try:
+            return st.patch(kind, name, body, ns,
+                            actor=actor_of(request),
+                            strategic=bool(strategic))
+        except ApiError as e:
+            return err(e)
+
     @app.delete("/apis/{kind}/{ns}/{name}")
     def delete(kind: str, ns: str, name: str, request: Request,
                uid: Optional[str] = None, rv: Optional[str] = None):

@@ -351,3 +351,53 @@ def test_controller_entrypoints_backend_flag():
     from fma_amd.store.client import StoreClient
     st, ns = make_store(A(), "dual-pods-controller")
     assert isinstance(st, StoreClient)
+
+
+# ---------------------------------------------------------------------------
+# PATCH verbs (merge + strategic-merge content types)
+# ---------------------------------------------------------------------------
+
+
+def test_patch_merge_and_strategic_over_the_wire(kube):
+    ks = kube["user"]
+    ks.create(mk_pod("pm", annotations={"keep": "1", "drop": "x"}))
+    out = ks.patch("Pod", "pm",
+                   {"metadata": {"annotations": {"drop": None, "n": "2"}}})
+    assert out["metadata"]["annotations"] == {"keep": "1", "n": "2"}
+    # strategic: containers merge by name instead of replacing
+    ks.patch("Pod", "pm", {"spec": {"containers": [
+        {"name": "c1", "image": "a"}]}}, strategic=True)
+    out = ks.patch("Pod", "pm", {"spec": {"containers": [
+        {"name": "c2", "image": "b"}]}}, strategic=True)
+    assert [c["name"] for c in out["spec"]["containers"]] == ["c1", "c2"]
+    # CRD group path works too
+    ks.create(ob.new_object(
+        "InferenceServerConfig", "iscp",
+        spec={"modelServerConfig": {"port": 8000},
+              "launcherConfigName": "lc1"}))
+    out = ks.patch("InferenceServerConfig", "iscp",
+                   {"metadata": {"labels": {"x": "y"}}})
+    assert out["metadata"]["labels"] == {"x": "y"}
+
+
+def test_patch_vap_denies_protected_mutation(kube):
+    """VAP admission sees the MERGED object: a user patching an
+    FMA-managed annotation is denied exactly like an update."""
+    ks, ctl = kube["user"], kube["ctl"]
+    ctl.create(mk_pod("pv",
+                      annotations={C.REQUESTER_ANNOTATION: "u1 req1"}))
+    with pytest.raises(Invalid) as ei:
+        ks.patch("Pod", "pv", {"metadata": {"annotations": {
+            C.REQUESTER_ANNOTATION: "evil"}}})
+    assert "fma-immutable-fields" in str(ei.value)
+    ctl.patch("Pod", "pv", {"metadata": {"annotations": {
+        C.REQUESTER_ANNOTATION: "u2 req2"}}})
+
+
+def test_patch_unsupported_content_type_415(kube):
+    ks = kube["user"]
+    ks.create(mk_pod("p415"))
+    r = httpx.patch(kube["base"] + "/api/v1/namespaces/default/pods/p415",
+                    content=b'[]',
+                    headers={"Content-Type": "application/json-patch+json"})
+    assert r.status_code == 415

@@ -358,3 +358,75 @@ def test_crd_schema_admission_on_memstore():
         spec={"enhancedNodeSelector": {"labelSelector": {}},
               "countForLauncher": [
                   {"launcherConfigName": "lc1", "launcherCount": 1}]}))
+
+
+# ---------------------------------------------------------------------------
+# server-side PATCH (merge + strategic merge)
+# ---------------------------------------------------------------------------
+
+
+def test_patch_merge_semantics():
+    st = MemStore()
+    st.create(ob.new_object("Pod", "pp", annotations={"keep": "1",
+                                                      "drop": "x"},
+                            spec={"nodeName": "n", "containers": [
+                                {"name": "c1", "image": "a"}]}))
+    out = st.patch("Pod", "pp", {
+        "metadata": {"annotations": {"drop": None, "new": "2"}}})
+    anns = out["metadata"]["annotations"]
+    assert anns == {"keep": "1", "new": "2"}
+    # RFC 7386: lists replace under plain merge
+    out = st.patch("Pod", "pp", {"spec": {"containers": [
+        {"name": "c2", "image": "b"}]}})
+    assert [c["name"] for c in out["spec"]["containers"]] == ["c2"]
+
+
+def test_patch_strategic_merges_named_lists():
+    st = MemStore()
+    st.create(ob.new_object("Pod", "ps", spec={"containers": [
+        {"name": "c1", "image": "a", "env": [
+            {"name": "E1", "value": "1"}]}]}))
+    out = st.patch("Pod", "ps", {"spec": {"containers": [
+        {"name": "c1", "env": [{"name": "E2", "value": "2"}]},
+        {"name": "c2", "image": "b"}]}}, strategic=True)
+    cs = out["spec"]["containers"]
+    assert [c["name"] for c in cs] == ["c1", "c2"]
+    assert cs[0]["image"] == "a"  # untouched field survives
+    assert [e["name"] for e in cs[0]["env"]] == ["E1", "E2"]
+
+
+def test_patch_runs_admission_on_merged_result():
+    st = MemStore()
+    install_policies(st)
+    st.create(ob.new_object(
+        "Pod", "pa",
+        annotations={C.REQUESTER_ANNOTATION: "u p"}),
+        actor="dual-pods-controller")
+    with pytest.raises(Invalid):
+        st.patch("Pod", "pa", {"metadata": {"annotations": {
+            C.REQUESTER_ANNOTATION: "evil"}}}, actor="user")
+    # FMA actor may
+    st.patch("Pod", "pa", {"metadata": {"annotations": {
+        C.REQUESTER_ANNOTATION: "u2 p2"}}},
+        actor="dual-pods-controller")
+
+
+def test_concurrent_patches_do_not_lose_updates():
+    """Two writers patching DIFFERENT annotation keys concurrently: with
+    read-modify-write one side's updates would be lost on conflict; with
+    server-side patch every write lands."""
+    import threading as th
+    st = MemStore()
+    st.create(ob.new_object("Pod", "pc"))
+    n = 30
+
+    def writer(key):
+        for i in range(n):
+            st.patch("Pod", "pc",
+                     {"metadata": {"annotations": {key: str(i)}}})
+
+    t1 = th.Thread(target=writer, args=("a",))
+    t2 = th.Thread(target=writer, args=("b",))
+    t1.start(); t2.start(); t1.join(); t2.join()
+    anns = st.get("Pod", "pc")["metadata"]["annotations"]
+    assert anns["a"] == str(n - 1) and anns["b"] == str(n - 1)

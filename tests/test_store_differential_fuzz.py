@@ -71,8 +71,8 @@ class _Step:
         self.actor = rng.choice(ACTORS)
         self.op = rng.choices(
             ["create", "update", "stale_update", "delete", "bad_delete",
-             "status", "finalize"],
-            weights=[28, 32, 8, 13, 4, 8, 7])[0]
+             "status", "finalize", "patch"],
+            weights=[26, 28, 8, 13, 4, 7, 7, 7])[0]
         self.rng_state = rng.getstate()
 
     def run(self, store, is_kube):
@@ -116,6 +116,12 @@ class _Step:
             if self.op == "stale_update":
                 meta["resourceVersion"] = "1"
             return call(store.update, cur)
+        if self.op == "patch":
+            key = rng.choice(ALL_ANN)
+            val = None if rng.random() < 0.3 else f"v{rng.randrange(3)}"
+            body = {"metadata": {"annotations": {key: val}}}
+            return call(store.patch, kind, name, body,
+                        strategic=rng.random() < 0.5)
         if self.op == "status":
             cur = ob.deepcopy(target)
             cur["status"] = {"phase": rng.choice(["Running", "Pending"]),
