@@ -189,6 +189,7 @@ def create_app(store: Optional[MemStore] = None,
         sel = _parse_label_selector(q.get("labelSelector"))
         if q.get("watch") in ("1", "true"):
             since = int(q.get("resourceVersion") or 0)
+            bookmarks = q.get("allowWatchBookmarks") in ("1", "true")
             loop = asyncio.get_running_loop()
 
             async def stream():
@@ -196,6 +197,10 @@ def create_app(store: Optional[MemStore] = None,
                 while True:
                     if await request.is_disconnected():
                         return
+                    # global revision read BEFORE polling: any event of
+                    # this kind at or below it lands in this batch, so an
+                    # empty batch makes it a safe bookmark cursor
+                    rev0 = st.list_revision()
                     try:
                         batch = await loop.run_in_executor(
                             watch_pool, lambda: list(st.watch(
@@ -208,6 +213,15 @@ def create_app(store: Optional[MemStore] = None,
                             "object": status_body(410, e.message, "Gone"),
                         }) + "\n"
                         return
+                    if bookmarks and not batch and rev0 > cursor:
+                        # other kinds advanced the store; move this
+                        # watcher past history it will never see, so a
+                        # later reconnect cannot 410 on evicted events
+                        cursor = rev0
+                        yield json.dumps({
+                            "type": "BOOKMARK",
+                            "object": {"kind": kind, "metadata": {
+                                "resourceVersion": str(rev0)}}}) + "\n"
                     for ev in batch:
                         cursor = max(cursor, ev.revision)
                         obj = ev.obj

@@ -355,7 +355,8 @@ class KubeStore:
             try:
                 with self._client.stream(
                         "GET", self.base + path,
-                        params={"watch": "1", "resourceVersion": cursor},
+                        params={"watch": "1", "resourceVersion": cursor,
+                                "allowWatchBookmarks": "true"},
                         timeout=httpx.Timeout(5.0, read=None)) as r:
                     if r.status_code == 410:
                         out.put(RevisionTooOld(f"{kind} watch expired"))
@@ -382,6 +383,8 @@ class KubeStore:
                         new_rv = ob.rv_of(obj)
                         if new_rv:
                             cursor = new_rv
+                        if ev.get("type") == "BOOKMARK":
+                            continue  # cursor keep-alive only, no event
                         out.put(ev)
             except httpx.HTTPError:
                 if stop.is_set():

@@ -401,3 +401,48 @@ def test_patch_unsupported_content_type_415(kube):
                     content=b'[]',
                     headers={"Content-Type": "application/json-patch+json"})
     assert r.status_code == 415
+
+
+def test_watch_bookmarks_keep_idle_watchers_resumable():
+    """k8s watch bookmarks: an idle Pod watcher's cursor advances past
+    history churned by OTHER kinds, so after an apiserver drop it
+    reconnects cleanly instead of hitting 410 on evicted revisions."""
+    store = MemStore()
+    store._history_cap = 50  # tiny buffer: churn evicts old revisions
+    port = free_port()
+    srv = ServerThread(create_app(store), port)
+    srv.__enter__()
+    ks = KubeStore(f"http://127.0.0.1:{port}", actor="user")
+    token = ks.list_revision()
+    got, errors = [], []
+    stop = threading.Event()
+
+    def consume():
+        try:
+            for ev in ks.watch(since=token, kinds=["Pod"], stop=stop):
+                got.append(ob.name_of(ev.obj))
+                if len(got) >= 1:
+                    return
+        except RevisionTooOld as e:
+            errors.append(e)
+
+    th = threading.Thread(target=consume, daemon=True)
+    th.start()
+    time.sleep(0.3)
+    # other-kind churn way past the history cap while the Pod watcher idles
+    for i in range(120):
+        store.create(ob.new_object("ConfigMap", f"churn-{i}"))
+    time.sleep(1.6)  # > the 1 s poll period: a BOOKMARK advances the cursor
+
+    srv.__exit__()  # drop the apiserver mid-watch
+    time.sleep(0.4)
+    srv2 = ServerThread(create_app(store), port)
+    srv2.__enter__()
+    try:
+        ks.create(mk_pod("after-bookmark"))
+        th.join(timeout=20)
+        assert not errors, f"watch 410d despite bookmarks: {errors}"
+        assert got == ["after-bookmark"], f"got {got}"
+    finally:
+        stop.set()
+        srv2.__exit__()
