@@ -1470,6 +1470,18 @@ at::Tensor attn_prefill_bf16(const at::Tensor& q, const at::Tensor& k,
                          q.options().dtype(at::kFloat));
     pptr = partials.data_ptr<float>();
   }
+  const char* p16 = std::getenv("FMA_PREFILL_16");
+  if (chunks == 1 && p16 != nullptr && std::atoi(p16) != 0) {
+    // experimental 16-row fragment variant: occupancy 4 (vs 2) and a
+    // doubled tile grid; numerics hardware-validated by
+    // tools/prefill16_probe.hip, perf not yet measured on large T
+    FMA_HIP_CHECK(fma_launch_attn_prefill16_bf16(
+        q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+        static_cast<int>(T), static_cast<int>(pos0),
+        static_cast<int>(q_heads), static_cast<int>(kv_heads),
+        static_cast<int>(hd), stream.stream()));
+    return out;
+  }
   FMA_HIP_CHECK(fma_launch_attn_prefill_bf16(
       q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
       static_cast<int>(T), static_cast<int>(pos0), static_cast<int>(q_heads),

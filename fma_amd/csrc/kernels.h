@@ -78,3 +78,9 @@ extern "C" hipError_t fma_launch_attn_prefill_bf16(
     const void* Q, const void* K, const void* V, void* O, int T, int pos0,
     int q_heads, int kv_heads, int hd, float* partials, int chunks,
     hipStream_t stream);
+
+// 16-row MFMA prefill variant (v_mfma_f32_16x16x32_bf16); single-chunk
+// only, opt-in via FMA_PREFILL_16=1
+extern "C" hipError_t fma_launch_attn_prefill16_bf16(
+    const void* Q, const void* K, const void* V, void* O, int T, int pos0,
+    int q_heads, int kv_heads, int hd, hipStream_t stream);

@@ -1541,3 +1541,214 @@ extern "C" hipError_t fma_launch_attn_prefill_bf16(
   }
   return hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// 16-row MFMA prefill: v_mfma_f32_16x16x32_bf16 fragments (lane maps
+// hardware-verified by tools/mfma_probe16.hip). Motivation (NOTES.md
+// round-3 sketch, landed early): vs the 32-row kernel the accumulator
+// footprint halves (oacc kB x 4 f32 = 32 VGPRs at HD=128) and the tile
+// grid DOUBLES (T/16 tiles), fixing the mid-T occupancy underfill. The
+// C/D layout puts each score row in ONE 16-lane fragment group, so the
+// whole online softmax runs on row-local registers + 4 shfl_xor steps —
+// no LDS transpose scratch, no cross-half handshakes at all (the 32-row
+// kernel burns two LDS round-trips per tile on exactly this).
+// Experimental: opt-in via FMA_PREFILL_16=1 (single-chunk path only).
+
+using f32x4_t = __attribute__((ext_vector_type(4))) float;
+
+template <int HD, int NW>
+__global__ __launch_bounds__(64 * NW)
+__attribute__((amdgpu_waves_per_eu(4)))
+void attn_prefill16_bf16_kernel(
+    const unsigned short* __restrict__ Q,  // [T, qH, hd]
+    const unsigned short* __restrict__ K,  // [S, kvH, hd]
+    const unsigned short* __restrict__ V,  // [S, kvH, hd]
+    unsigned short* __restrict__ O,        // [T, qH, hd]
+    int T, int pos0, int q_heads, int kv_heads) {
+  constexpr int kB = HD / 16;  // 16-column output blocks
+  const int wid = NW > 1 ? (threadIdx.x >> 6) : 0;
+  const int qh = blockIdx.x * NW + wid;
+  const int tile = gridDim.y - 1 - blockIdx.y;  // heaviest tiles first
+  const int r0 = tile * 16;
+  if (r0 >= T) return;  // uniform across the WG
+  const int kvh = qh / (q_heads / kv_heads);
+  const int l = threadIdx.x & 63;
+  const int g = l >> 4;       // 16-lane fragment group, 0..3
+  const int c = l & 15;       // position within the group
+  const int lane32 = l & 31;  // for the shared V staging helper
+  const int half = l >> 5;
+  const float scale = rsqrtf(static_cast<float>(HD));
+
+  __shared__ __bf16 s_v[32 * HD];          // V [hd][key], shared by waves
+  __shared__ __bf16 s_p_all[NW][16 * 32];  // P tile C->A relayout
+  __bf16* const s_p = s_p_all[wid];
+  constexpr int kC8PerWave = (HD / 16) / NW;
+  const int c8_lo = wid * kC8PerWave, c8_hi = c8_lo + kC8PerWave;
+
+  // Q fragments: lane holds A[c][32*ks + 8*g + i] (A row = q row = c)
+  bf16x8_t qf[HD / 32];
+  const int q_row = r0 + c;
+  const bool row_live = q_row < T;
+  {
+    const unsigned short* qp =
+        Q + (static_cast<long long>(row_live ? q_row : 0) * q_heads + qh) *
+                HD + 8 * g;
+#pragma unroll
+    for (int ks = 0; ks < HD / 32; ++ks) {
+      bf16x8_t qv;
+      if (row_live) {
+        qv = *reinterpret_cast<const bf16x8_t*>(qp + ks * 32);
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) qv[i] = static_cast<__bf16>(0.0f);
+      }
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        qf[ks][i] = static_cast<__bf16>(static_cast<float>(qv[i]) * scale);
+      }
+    }
+  }
+
+  // Per-row softmax stats in REGISTERS: this lane's reg r owns row
+  // 4*g + r, whose 16 key columns live across the lane group — row
+  // reductions are 4 shfl_xor steps (masks 1/2/4/8 stay in-group)
+  float m_row[4], l_row[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_row[r] = -1e30f;
+    l_row[r] = 0.0f;
+  }
+  f32x4_t oacc[kB];
+#pragma unroll
+  for (int b = 0; b < kB; ++b) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) oacc[b][r] = 0.0f;
+  }
+
+  const int t_kv = pos0 + T;  // causal key horizon
+  const int kt_end = min((pos0 + r0 + 15) / 32 + 1, (t_kv + 31) / 32);
+
+  for (int kt = 0; kt < kt_end; ++kt) {
+    if (NW > 1) __syncthreads();  // all waves done reading s_v
+    prefill_stage_v<HD>(V, kt * 32, t_kv, kv_heads, kvh, lane32, half, s_v,
+                        c8_lo, c8_hi);
+    if (NW > 1) __syncthreads();  // staging visible to every wave
+    __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);
+
+    // S = Q.K^T over the 32-key tile, two 16-key quadrants
+    f32x4_t sacc[2];
+#pragma unroll
+    for (int q2 = 0; q2 < 2; ++q2) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) sacc[q2][r] = 0.0f;
+      const int key = kt * 32 + 16 * q2 + c;
+      const bool klive = key < t_kv;
+      const unsigned short* kp =
+          K + (static_cast<long long>(klive ? key : 0) * kv_heads + kvh) *
+                  HD + 8 * g;
+#pragma unroll
+      for (int ks = 0; ks < HD / 32; ++ks) {
+        bf16x8_t kf;
+        if (klive) {
+          kf = *reinterpret_cast<const bf16x8_t*>(kp + ks * 32);
+        } else {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) kf[i] = static_cast<__bf16>(0.0f);
+        }
+        sacc[q2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf[ks], kf,
+                                                           sacc[q2], 0, 0, 0);
+      }
+    }
+
+    // Online softmax, all register-resident
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = 4 * g + r;
+      const int qpos = pos0 + r0 + row;
+      const int k0 = kt * 32 + c, k1 = kt * 32 + 16 + c;
+      if (k0 > qpos || k0 >= t_kv) sacc[0][r] = -1e30f;
+      if (k1 > qpos || k1 >= t_kv) sacc[1][r] = -1e30f;
+      float m = fmaxf(sacc[0][r], sacc[1][r]);
+      m = fmaxf(m, __shfl_xor(m, 1));
+      m = fmaxf(m, __shfl_xor(m, 2));
+      m = fmaxf(m, __shfl_xor(m, 4));
+      m = fmaxf(m, __shfl_xor(m, 8));
+      const float m_new = fmaxf(m_row[r], m);
+      const float alpha = __expf(m_row[r] - m_new);
+      m_row[r] = m_new;
+      l_row[r] *= alpha;
+#pragma unroll
+      for (int b = 0; b < kB; ++b) oacc[b][r] *= alpha;
+      const __bf16 p0 = static_cast<__bf16>(__expf(sacc[0][r] - m_new));
+      const __bf16 p1 = static_cast<__bf16>(__expf(sacc[1][r] - m_new));
+      s_p[row * 32 + c] = p0;
+      s_p[row * 32 + 16 + c] = p1;
+      // sum the bf16-ROUNDED probabilities: matches what the PV MFMA uses
+      float sum = static_cast<float>(p0) + static_cast<float>(p1);
+      sum += __shfl_xor(sum, 1);
+      sum += __shfl_xor(sum, 2);
+      sum += __shfl_xor(sum, 4);
+      sum += __shfl_xor(sum, 8);
+      l_row[r] += sum;
+    }
+    __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);  // s_p relayout (wave-local)
+
+    // O += P.V: K-dim 32 = exactly ONE MFMA step per 16-column block
+    bf16x8_t pf = *reinterpret_cast<const bf16x8_t*>(&s_p[c * 32 + 8 * g]);
+#pragma unroll
+    for (int b = 0; b < kB; ++b) {
+      bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+          &s_v[(b * 16 + c) * 32 + 8 * g]);
+      oacc[b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, oacc[b], 0,
+                                                        0, 0);
+    }
+    __builtin_amdgcn_s_waitcnt(FMA_WAIT_LGKM0);  // s_p/s_v reads done
+  }
+
+  // Epilogue: l_row is already uniform across the lane group — no LDS
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int orow = r0 + 4 * g + r;
+    if (orow >= T) continue;
+    const float inv = l_row[r] > 0.0f ? 1.0f / l_row[r] : 0.0f;
+    unsigned short* op =
+        O + (static_cast<long long>(orow) * q_heads + qh) * HD;
+#pragma unroll
+    for (int b = 0; b < kB; ++b) {
+      op[b * 16 + c] = f32_to_bf16(oacc[b][r] * inv);
+    }
+  }
+}
+
+extern "C" hipError_t fma_launch_attn_prefill16_bf16(
+    const void* Q, const void* K, const void* V, void* O, int T, int pos0,
+    int q_heads, int kv_heads, int hd, hipStream_t stream) {
+  if (hd != 64 && hd != 128) return hipErrorInvalidValue;
+  if (q_heads % kv_heads != 0) return hipErrorInvalidValue;
+  const int group = q_heads / kv_heads;
+  int nw = (q_heads % 4 == 0 && group % 4 == 0) ? 4 : 1;
+  if (const char* e = getenv("FMA_PREFILL_NW")) {
+    const int forced = atoi(e);
+    if (forced == 1 || (forced == 4 && q_heads % 4 == 0 && group % 4 == 0))
+      nw = forced;
+  }
+  dim3 grid(q_heads / nw, (T + 15) / 16);
+  const auto* Qp = static_cast<const unsigned short*>(Q);
+  const auto* Kp = static_cast<const unsigned short*>(K);
+  const auto* Vp = static_cast<const unsigned short*>(V);
+  auto* Op = static_cast<unsigned short*>(O);
+  if (hd == 128 && nw == 4) {
+    attn_prefill16_bf16_kernel<128, 4><<<grid, 256, 0, stream>>>(
+        Qp, Kp, Vp, Op, T, pos0, q_heads, kv_heads);
+  } else if (hd == 128) {
+    attn_prefill16_bf16_kernel<128, 1><<<grid, 64, 0, stream>>>(
+        Qp, Kp, Vp, Op, T, pos0, q_heads, kv_heads);
+  } else if (nw == 4) {
+    attn_prefill16_bf16_kernel<64, 4><<<grid, 256, 0, stream>>>(
+        Qp, Kp, Vp, Op, T, pos0, q_heads, kv_heads);
+  } else {
+    attn_prefill16_bf16_kernel<64, 1><<<grid, 64, 0, stream>>>(
+        Qp, Kp, Vp, Op, T, pos0, q_heads, kv_heads);
+  }
+  return hipGetLastError();
+}

@@ -101,7 +101,12 @@ def fast_attn_prefill(q: torch.Tensor, k_cache: torch.Tensor,
     if not available() or mode == "0":
         return None
     b, t, qh, hd = q.shape
-    if mode != "1" and t > 128:
+    # FMA_PREFILL_16=1 opts into the 16-row fragment kernel (occupancy 4,
+    # doubled tile grid — built for exactly the mid-T range the 32-row
+    # kernel loses; numerics hardware-validated, perf unmeasured), so the
+    # short-prefill cutoff does not apply to it
+    p16 = os.environ.get("FMA_PREFILL_16") == "1"
+    if mode != "1" and not p16 and t > 128:
         return None
     if (b != 1 or hd not in (64, 128) or not q.is_cuda
             or q.dtype != torch.bfloat16

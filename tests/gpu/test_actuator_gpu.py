@@ -694,3 +694,46 @@ def test_qwen2_family_decode_fused_matches_eager(C):
     eng.sleep()
     eng.wake_up()
     assert torch.equal(eng.generate(toks, max_new_tokens=6), fused)
+
+
+def test_attn_prefill16_matches_fp32_reference():
+    """16-row MFMA prefill variant (FMA_PREFILL_16=1,
+    v_mfma_f32_16x16x32_bf16 fragments) vs the same fp32 reference as
+    the 32-row kernel: both NW paths, 16-row tail tiles, pos0>0, both
+    head dims. Standalone hardware validation: tools/prefill16_probe.hip
+    (5 cases, max_err<=0.0025)."""
+    import os
+    import fma_amd._C as C
+    torch.manual_seed(17)
+    os.environ["FMA_PREFILL_16"] = "1"
+    try:
+        for (qH, kvH, hd, S, T, pos0) in (
+                (32, 8, 128, 256, 256, 0),   # NW=4, llama-8B shape
+                (16, 2, 128, 200, 100, 33),  # ragged T, chunked offset
+                (6, 2, 128, 64, 50, 0),      # group 3 -> NW=1 path
+                (8, 8, 64, 96, 33, 0),       # hd=64, tail tile
+                (4, 1, 128, 40, 1, 7),       # single query row
+        ):
+            q = torch.randn(T, qH, hd, dtype=torch.bfloat16,
+                            device="cuda:0")
+            k = torch.randn(S, kvH, hd, dtype=torch.bfloat16,
+                            device="cuda:0")
+            v = torch.randn(S, kvH, hd, dtype=torch.bfloat16,
+                            device="cuda:0")
+            out = C.attn_prefill_bf16(q, k, v, pos0, 1)  # chunks=1
+            t_kv = pos0 + T
+            rep = qH // kvH
+            kf = k[:t_kv].float().permute(1, 0, 2).repeat_interleave(rep, 0)
+            vf = v[:t_kv].float().permute(1, 0, 2).repeat_interleave(rep, 0)
+            qf = q.float().permute(1, 0, 2)
+            scores = qf @ kf.transpose(1, 2) / (hd ** 0.5)
+            qpos = pos0 + torch.arange(T, device="cuda:0")
+            kpos = torch.arange(t_kv, device="cuda:0")
+            scores.masked_fill_(kpos[None, None, :] > qpos[None, :, None],
+                                float("-inf"))
+            ref = (torch.softmax(scores, dim=-1) @ vf).permute(1, 0, 2)
+            assert torch.allclose(out.float(), ref, atol=3e-2, rtol=3e-2), \
+                (qH, kvH, hd, S, T, pos0,
+                 (out.float() - ref).abs().max().item())
+    finally:
+        os.environ.pop("FMA_PREFILL_16", None)
