@@ -44,8 +44,9 @@ def main():
         t_ms = bench_op(lambda: C.attn_prefill_bf16(q, k, v, 0), args.iters)
         print(f"T={T} mfma_ms={t_ms:.3f}")
         return
-    print(f"{'T':>6} {'mfma_ms':>9} {'sdpa_ms':>9} {'speedup':>8} {'max_err':>9}")
-    for T in (128, 512, 1024, 2048, 4096):
+    print(f"{'T':>6} {'mfma_ms':>9} {'m16_ms':>9} {'sdpa_ms':>9} "
+          f"{'sdpa/m16':>8} {'max_err16':>9}")
+    for T in (128, 256, 512, 1024, 2048, 4096):
         q = torch.randn(T, qH, hd, dtype=torch.bfloat16, device="cuda:0")
         k = torch.randn(T, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
         v = torch.randn(T, kvH, hd, dtype=torch.bfloat16, device="cuda:0")
@@ -54,16 +55,21 @@ def main():
         vh = v.permute(1, 0, 2).repeat_interleave(rep, 0).unsqueeze(0).contiguous()
 
         t_mfma = bench_op(lambda: C.attn_prefill_bf16(q, k, v, 0), args.iters)
+        # 16-row variant (chunks=1 forced so the env gate applies)
+        os.environ["FMA_PREFILL_16"] = "1"
+        t_m16 = bench_op(lambda: C.attn_prefill_bf16(q, k, v, 0, 1),
+                         args.iters)
+        out16 = C.attn_prefill_bf16(q, k, v, 0, 1).float()
+        os.environ.pop("FMA_PREFILL_16", None)
         t_sdpa = bench_op(
             lambda: torch.nn.functional.scaled_dot_product_attention(
                 qh, kh, vh, is_causal=True), args.iters)
 
-        out = C.attn_prefill_bf16(q, k, v, 0).float()
         ref = torch.nn.functional.scaled_dot_product_attention(
             qh.float(), kh.float(), vh.float(), is_causal=True)
-        err = (out - ref.squeeze(0).permute(1, 0, 2)).abs().max().item()
-        print(f"{T:>6} {t_mfma:>9.3f} {t_sdpa:>9.3f} {t_sdpa / t_mfma:>8.2f} "
-              f"{err:>9.4f}")
+        err16 = (out16 - ref.squeeze(0).permute(1, 0, 2)).abs().max().item()
+        print(f"{T:>6} {t_mfma:>9.3f} {t_m16:>9.3f} {t_sdpa:>9.3f} "
+              f"{t_sdpa / t_m16:>8.2f} {err16:>9.4f}")
 
 
 if __name__ == "__main__":
