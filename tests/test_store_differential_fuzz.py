@@ -214,3 +214,53 @@ def test_memstore_and_kube_double_agree(double, seed):
     kube_events = [(e.type, e.kind, ob.name_of(e.obj))
                    for e in backing._history]
     assert mem_events == kube_events
+
+
+def test_index_get_parity_between_backends(double):
+    """Controllers do reconcile lookups through index_get; every standard
+    Pod index must return identical result sets from MemStore (true
+    store-side indexes) and KubeStore (client-side evaluation over a
+    LIST) for the same cluster state."""
+    from fma_amd.store.indexes import POD_INDEXES
+
+    _reset(double["backing"])
+    mem = MemStore()
+    install_pod_indexes(mem)
+    kube = KubeStore(double["base"])
+
+    rng = random.Random(99)
+    for i in range(25):
+        ann, lbl = {}, {}
+        if rng.random() < 0.5:
+            ann[C.REQUESTER_ANNOTATION] = f"u{rng.randrange(3)} req{i}"
+        if rng.random() < 0.5:
+            ann[C.INFERENCE_SERVER_CONFIG_ANNOTATION] = \
+                f"isc-{rng.randrange(3)}"
+        if rng.random() < 0.4:
+            ann[C.ACCELERATORS_ANNOTATION] = ",".join(
+                f"GPU-{rng.randrange(4)}" for _ in range(rng.randrange(1, 3)))
+        if rng.random() < 0.4:
+            ann[C.NOMINAL_ANNOTATION] = f"h{rng.randrange(3)}"
+        if rng.random() < 0.5:
+            lbl[C.DUAL_LABEL] = f"l{rng.randrange(3)}"
+        if rng.random() < 0.4:
+            lbl[C.LAUNCHER_CONFIG_NAME_LABEL] = f"lc{rng.randrange(2)}"
+        pod = ob.new_object(
+            "Pod", f"ix-{i}", annotations=ann, labels=lbl,
+            spec={"nodeName": f"node-{rng.randrange(3)}",
+                  "containers": []})
+        mem.create(ob.deepcopy(pod))
+        kube.create(pod)
+
+    keys = (["u0 req1", "isc-0", "isc-1", "GPU-0", "GPU-3", "h0", "l1",
+             "lc0", "node-0", "node-2", "absent"])
+    hits = 0
+    for index_name in POD_INDEXES:
+        for key in keys:
+            m = sorted(ob.name_of(o)
+                       for o in mem.index_get("Pod", index_name, key))
+            k = sorted(ob.name_of(o)
+                       for o in kube.index_get("Pod", index_name, key))
+            assert m == k, (index_name, key, m, k)
+            hits += len(m)
+    assert hits > 10, f"parity checked mostly-empty sets (hits={hits})"
