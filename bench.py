@@ -41,7 +41,10 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--gib", type=float, default=64.0,
                     help="total parameter GiB across all ranks")
-    ap.add_argument("--mode", choices=["arena", "pack"], default="arena")
+    # pack (the gather/scatter-kernel path BASELINE.json's north star
+    # describes) reached arena parity in round 2 (1.2175 vs 1.2165 s /
+    # 64 GiB) and is now the measured default
+    ap.add_argument("--mode", choices=["arena", "pack"], default="pack")
     ap.add_argument("--pack-xfer", type=int, default=-1,
                     help="pack transfer mode: -1 auto (default; measured "
                          "fastest per direction), 0 staged-kernel, "
