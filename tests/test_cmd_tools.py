@@ -81,3 +81,42 @@ def test_store_http_roundtrip():
     assert r.status_code == 409
     assert c.delete("/apis/Pod/default/x").status_code == 200
     assert c.get("/apis/Pod/default/x").status_code == 404
+
+
+def test_ensure_nodes_mapped_fills_gpu_map():
+    """tools/ensure_nodes_mapped.py (the ensure-nodes-mapped.sh analog):
+    idempotently fills gpu-map[node] through BOTH store backends using
+    the GPU-less naive translator."""
+    import json
+    import sys
+    sys.path.insert(0, "tools")
+    from ensure_nodes_mapped import ensure_mapped
+
+    from fma_amd.api import contracts as C
+    from fma_amd.launcher.gputranslator import GpuTranslator
+    from fma_amd.store.kubeapiserver import create_app as kube_app
+    from fma_amd.store.kubestore import KubeStore
+    from fma_amd.store.memstore import MemStore
+
+    from tests.test_live_servers import ServerThread, free_port
+
+    tr = GpuTranslator(mode=GpuTranslator.MODE_NAIVE,
+                       gpu_map={"GPU-0": 0, "GPU-1": 1})
+
+    # in-process MemStore
+    st = MemStore()
+    m = ensure_mapped(st, "node-x", translator=tr)
+    assert m == {"GPU-0": 0, "GPU-1": 1}
+    # idempotent: second call returns the stored entry untouched
+    assert ensure_mapped(st, "node-x", translator=tr) == m
+    cm = st.get("ConfigMap", C.GPU_MAP_CONFIGMAP)
+    assert json.loads(cm["data"]["node-x"]) == m
+
+    # over the Kubernetes wire protocol (merge PATCH per node)
+    port = free_port()
+    with ServerThread(kube_app(MemStore()), port):
+        ks = KubeStore(f"http://127.0.0.1:{port}", actor="node-agent")
+        assert ensure_mapped(ks, "node-a", translator=tr) == m
+        assert ensure_mapped(ks, "node-b", translator=tr) == m
+        data = ks.get("ConfigMap", C.GPU_MAP_CONFIGMAP)["data"]
+        assert set(data) == {"node-a", "node-b"}
