@@ -20,7 +20,7 @@ def dump(base: str, tail: int, out=sys.stdout) -> int:
     r.raise_for_status()
     instances = r.json().get("instances", [])
     for inst in instances:
-        iid = inst.get("id")
+        iid = inst.get("instance_id")
         hdr = {"Range": f"bytes=-{tail}"} if tail else {}
         lr = httpx.get(f"{base}/v2/vllm/instances/{iid}/log",
                        headers=hdr, timeout=10)
