@@ -21,6 +21,7 @@ from __future__ import annotations
 import glob
 import json
 import os
+import re
 from typing import Dict, Iterator, List, Optional, Tuple
 
 import torch
@@ -46,7 +47,9 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
                                                4096)),
         rope_theta=d.get("rope_theta", 500000.0),
         norm_eps=d.get("norm_eps", d.get("rms_norm_eps", 1e-5)),
-        qkv_bias=d.get("qkv_bias", False),
+        # transformers configs carry model_type instead of qkv_bias;
+        # qwen2's attention projections are the biased ones
+        qkv_bias=d.get("qkv_bias", d.get("model_type") == "qwen2"),
         num_experts=d.get("num_experts", d.get("num_local_experts", 0)),
         num_experts_per_tok=d.get("num_experts_per_tok", 2),
     )
@@ -101,6 +104,79 @@ def save_params(params: Dict[str, torch.Tensor], path: str,
                 "num_experts": cfg.num_experts,
                 "num_experts_per_tok": cfg.num_experts_per_tok,
             }, f)
+
+
+# ---------------------------------------------------------------------------
+# HuggingFace checkpoint interop: a reference user's models are HF
+# safetensors (vLLM's native format, reference docs/dual-pods.md:599-608)
+# in transformers naming. Mapping + the RoPE layout fix-up below let the
+# same directories load straight into the engine.
+# ---------------------------------------------------------------------------
+
+_HF_ATTN = {"q_proj": "wq", "k_proj": "wk", "v_proj": "wv", "o_proj": "wo"}
+_HF_MLP = {"gate_proj": "w_gate", "up_proj": "w_up", "down_proj": "w_down"}
+_HF_EXPERT = {"w1": "w_gate", "w3": "w_up", "w2": "w_down"}  # Mixtral
+_HF_LAYER_RE = re.compile(r"model\.layers\.(\d+)\.(.*)$")
+_HF_ATTN_RE = re.compile(r"self_attn\.(\w+_proj)\.(weight|bias)$")
+_HF_MLP_RE = re.compile(r"mlp\.(\w+_proj)\.weight$")
+_HF_EXP_RE = re.compile(
+    r"block_sparse_moe\.experts\.(\d+)\.(w[123])\.weight$")
+
+
+def map_hf_name(name: str) -> Optional[str]:
+    """transformers parameter name -> ours, or None if unmapped
+    (Llama / Qwen2 / Mixtral families)."""
+    if name == "model.embed_tokens.weight":
+        return "embed.weight"
+    if name == "model.norm.weight":
+        return "final_norm.weight"
+    if name == "lm_head.weight":
+        return "lm_head.weight"
+    m = _HF_LAYER_RE.match(name)
+    if not m:
+        return None
+    p, leaf = f"layers.{m.group(1)}.", m.group(2)
+    if leaf == "input_layernorm.weight":
+        return p + "attn_norm.weight"
+    if leaf == "post_attention_layernorm.weight":
+        return p + "mlp_norm.weight"
+    if leaf == "block_sparse_moe.gate.weight":
+        return p + "router.weight"
+    a = _HF_ATTN_RE.match(leaf)
+    if a and a.group(1) in _HF_ATTN:
+        return p + _HF_ATTN[a.group(1)] + "." + a.group(2)
+    mm = _HF_MLP_RE.match(leaf)
+    if mm and mm.group(1) in _HF_MLP:
+        return p + _HF_MLP[mm.group(1)] + ".weight"
+    e = _HF_EXP_RE.match(leaf)
+    if e:
+        return p + f"experts.{e.group(1)}." + _HF_EXPERT[e.group(2)] + \
+            ".weight"
+    return None
+
+
+def _unrotate_half(t: torch.Tensor, heads: int, hd: int) -> torch.Tensor:
+    """transformers rope pairs dims (i, i+hd/2) per head (rotate_half);
+    our kernels/apply_rope pair adjacent dims (2i, 2i+1). Reorder q/k
+    PROJECTION ROWS so attention scores match at every frequency — the
+    standard HF->interleaved permutation."""
+    if t.dim() == 2:
+        return t.view(heads, 2, hd // 2, t.shape[1]).permute(
+            0, 2, 1, 3).reshape(t.shape)
+    return t.view(heads, 2, hd // 2).permute(0, 2, 1).reshape(t.shape)
+
+
+def hf_convert(name: str, tensor: torch.Tensor, cfg: "LlamaConfig"):
+    """(our_name, converted_tensor) for one HF tensor; None when the
+    tensor has no counterpart (e.g. rotary_emb.inv_freq buffers)."""
+    our = map_hf_name(name)
+    if our is None:
+        return None
+    leaf = our.split(".")[-2]
+    if leaf in ("wq", "wk"):
+        heads = cfg.num_heads if leaf == "wq" else cfg.num_kv_heads
+        tensor = _unrotate_half(tensor, heads, cfg.head_dim)
+    return our, tensor
 
 
 # Megatron-style sharding of the llama parameter set (models/llama.py
@@ -158,16 +234,30 @@ def shard_slice(name: str, tensor: torch.Tensor, tp_rank: int,
 
 def load_into_params(path: str, params: Dict[str, torch.Tensor],
                      strict: bool = True, tp_rank: int = 0,
-                     tp_size: int = 1, skip=None) -> int:
+                     tp_size: int = 1, skip=None, cfg=None) -> int:
     """Copy checkpoint tensors into existing (arena-view) parameters.
 
     Returns the number of tensors loaded. Checkpoints store the full
     (unsharded) tensors; with tp_size > 1 each rank slices out its
-    Megatron shard (shard_slice) before copying.
+    Megatron shard (shard_slice) before copying. HuggingFace-format
+    checkpoints (``model.*`` transformers naming) are detected per
+    tensor and converted in place (name mapping + rope-layout fix-up;
+    requires ``cfg``); tied embeddings fill ``lm_head`` from ``embed``.
     """
     loaded = 0
     seen = set()
+    hf = False
     for name, tensor in iter_safetensors(path):
+        if name.startswith("model."):
+            hf = True
+            if cfg is None:
+                raise ValueError(
+                    "HuggingFace checkpoint: pass cfg= for name mapping "
+                    "and rope conversion")
+            conv = hf_convert(name, tensor, cfg)
+            if conv is None:
+                continue  # inv_freq buffers and friends
+            name, tensor = conv
         if name not in params:
             if skip is not None and skip(name):
                 continue  # e.g. expert-parallel: another rank's expert
@@ -186,6 +276,12 @@ def load_into_params(path: str, params: Dict[str, torch.Tensor],
                              f"{tuple(p.shape)}")
         p.copy_(tensor.to(p.dtype))
         seen.add(name)
+        loaded += 1
+    if hf and "lm_head.weight" not in seen and "lm_head.weight" in params \
+            and "embed.weight" in seen:
+        # transformers tie_word_embeddings: checkpoint omits lm_head
+        params["lm_head.weight"].copy_(params["embed.weight"])
+        seen.add("lm_head.weight")
         loaded += 1
     if strict:
         missing = set(params) - seen

@@ -172,15 +172,23 @@ class ActuationEngine:
             t0 = _t.perf_counter()
             loader.load_into_params(path, self.params,
                                     tp_rank=self.tp_rank,
-                                    tp_size=self.tp_size)
+                                    tp_size=self.tp_size, cfg=self.cfg)
             return _t.perf_counter() - t0
 
         t0 = _t.perf_counter()
         from fma_amd.models import loader
         host_np = self.host.numpy()
 
+        hf_seen = [False]
+
         def stage(item):
             name, tensor = item
+            if name.startswith("model."):  # HuggingFace checkpoint
+                hf_seen[0] = True
+                conv = loader.hf_convert(name, tensor, self.cfg)
+                if conv is None:
+                    return None  # inv_freq buffers and friends
+                name, tensor = conv
             if name not in self.layout:
                 if self.cfg.num_experts and self.cfg.expert_parallel \
                         and ".experts." in name:
@@ -205,6 +213,18 @@ class ActuationEngine:
             for name in ex.map(stage, loader.iter_safetensors(path)):
                 seen.add(name)
         missing = set(self.layout) - seen
+        if hf_seen[0] and missing == {"lm_head.weight"} \
+                and "embed.weight" in seen:
+            # transformers tie_word_embeddings: mirror embed's staged
+            # bytes into lm_head's slot (both full [vocab, h])
+            eo, eshape, edt = self.layout["embed.weight"]
+            lo, lshape, ldt = self.layout["lm_head.weight"]
+            if eshape == lshape and edt == ldt:
+                import math as _m
+                nbytes = _m.prod(eshape) * torch.empty(
+                    0, dtype=edt).element_size()
+                host_np[lo:lo + nbytes] = host_np[eo:eo + nbytes]
+                missing = set()
         if missing:
             raise KeyError(f"checkpoint missing parameters: "
                            f"{sorted(missing)[:5]}...")

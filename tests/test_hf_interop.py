@@ -1,0 +1,122 @@
+"""HuggingFace checkpoint interop, cross-validated against transformers.
+
+A user of the reference serves HF safetensors models through vLLM
+(reference docs/dual-pods.md:599-608); here the loader maps transformers
+naming onto our parameter schema (loader.map_hf_name), fixes up the RoPE
+layout difference (rotate-half vs adjacent-pair, loader._unrotate_half)
+and ties lm_head when the checkpoint omits it. These tests are the
+strongest correctness statement in the repo's CPU tier: random-init
+transformers models are saved with save_pretrained and our model must
+reproduce their LOGITS — validating the converter AND our entire model
+math (rmsnorm, rope, GQA attention, silu MLP, MoE routing) against the
+canonical implementation.
+"""
+
+import pytest
+import torch
+
+from fma_amd.models import loader
+from fma_amd.models.llama import LlamaModel
+
+transformers = pytest.importorskip("transformers")
+
+pytestmark = pytest.mark.timeout(240)
+
+
+def _load_ours(ckpt_dir):
+    cfg = loader.config_from_dir(ckpt_dir)
+    cfg.dtype = torch.float32  # tight comparison against fp32 HF
+    params = {n: torch.zeros(s, dtype=d)
+              for n, s, d in cfg.param_specs()}
+    loader.load_into_params(ckpt_dir, params, cfg=cfg)
+    return cfg, LlamaModel(cfg, params, torch.device("cpu"))
+
+
+def _compare_logits(hf_model, ckpt_dir, rtol=2e-3, atol=2e-3):
+    cfg, ours = _load_ours(ckpt_dir)
+    torch.manual_seed(7)
+    tokens = torch.randint(0, cfg.vocab_size, (1, 12))
+    with torch.no_grad():
+        ref = hf_model(tokens).logits
+        got = ours.forward(tokens)
+    assert torch.allclose(got, ref, rtol=rtol, atol=atol), \
+        (got - ref).abs().max().item()
+
+
+def test_llama_hf_checkpoint_logits_match(tmp_path):
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, rope_theta=10000.0, rms_norm_eps=1e-5,
+        tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    _compare_logits(hf, str(tmp_path))
+
+
+def test_llama_hf_untied_head(tmp_path):
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=96, hidden_size=32, intermediate_size=48,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=1,
+        max_position_embeddings=32, tie_word_embeddings=False)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    _compare_logits(hf, str(tmp_path))
+
+
+def test_qwen2_hf_checkpoint_logits_match(tmp_path):
+    """Qwen2: attention-projection biases + qkv_bias inferred from
+    model_type in the HF config.json."""
+    from transformers import Qwen2Config, Qwen2ForCausalLM
+    hf = Qwen2ForCausalLM(Qwen2Config(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, rope_theta=10000.0,
+        tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    cfg = loader.config_from_dir(str(tmp_path))
+    assert cfg.qkv_bias  # inferred from model_type == "qwen2"
+    _compare_logits(hf, str(tmp_path))
+
+
+def test_mixtral_hf_checkpoint_logits_match(tmp_path):
+    """Mixtral: router (gate) + experts w1/w3/w2 naming, top-2 routing
+    with post-topk softmax."""
+    from transformers import MixtralConfig, MixtralForCausalLM
+    hf = MixtralForCausalLM(MixtralConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=4, num_experts_per_tok=2,
+        max_position_embeddings=64, tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    cfg = loader.config_from_dir(str(tmp_path))
+    assert cfg.num_experts == 4 and cfg.num_experts_per_tok == 2
+    _compare_logits(hf, str(tmp_path))
+
+
+def test_map_hf_name_table():
+    cases = {
+        "model.embed_tokens.weight": "embed.weight",
+        "model.norm.weight": "final_norm.weight",
+        "lm_head.weight": "lm_head.weight",
+        "model.layers.3.self_attn.q_proj.weight": "layers.3.wq.weight",
+        "model.layers.3.self_attn.k_proj.bias": "layers.3.wk.bias",
+        "model.layers.0.input_layernorm.weight":
+            "layers.0.attn_norm.weight",
+        "model.layers.0.post_attention_layernorm.weight":
+            "layers.0.mlp_norm.weight",
+        "model.layers.1.mlp.gate_proj.weight": "layers.1.w_gate.weight",
+        "model.layers.1.mlp.down_proj.weight": "layers.1.w_down.weight",
+        "model.layers.2.block_sparse_moe.gate.weight":
+            "layers.2.router.weight",
+        "model.layers.2.block_sparse_moe.experts.5.w1.weight":
+            "layers.2.experts.5.w_gate.weight",
+        "model.layers.2.block_sparse_moe.experts.5.w2.weight":
+            "layers.2.experts.5.w_down.weight",
+        "model.layers.2.block_sparse_moe.experts.5.w3.weight":
+            "layers.2.experts.5.w_up.weight",
+        "model.layers.0.self_attn.rotary_emb.inv_freq": None,
+        "something.else": None,
+    }
+    for hf_name, ours in cases.items():
+        assert loader.map_hf_name(hf_name) == ours, hf_name
