@@ -120,3 +120,24 @@ def test_map_hf_name_table():
     }
     for hf_name, ours in cases.items():
         assert loader.map_hf_name(hf_name) == ours, hf_name
+
+
+def test_llama_hf_greedy_generation_matches(tmp_path):
+    """Token-level equality: greedy decode through OUR KV-cache path vs
+    transformers generate on the same HF checkpoint — validates the
+    cached incremental forward (rope offsets, cache indexing) end to
+    end, not just one prefill."""
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    _, ours = _load_ours(str(tmp_path))
+    torch.manual_seed(11)
+    prompt = torch.randint(0, 128, (1, 8))
+    with torch.no_grad():
+        ref = hf.generate(prompt, max_new_tokens=12, do_sample=False,
+                          use_cache=True)
+        got = ours.generate(prompt, max_new_tokens=12)
+    assert torch.equal(got, ref[:, :got.shape[1]]), (got, ref)
