@@ -141,3 +141,46 @@ def test_llama_hf_greedy_generation_matches(tmp_path):
                           use_cache=True)
         got = ours.generate(prompt, max_new_tokens=12)
     assert torch.equal(got, ref[:, :got.shape[1]]), (got, ref)
+
+
+def test_hf_checkpoint_tp_sharding_composes(tmp_path):
+    """The HF rope permutation is per-head, so it must commute with
+    Megatron head-aligned sharding: TP shards of an HF checkpoint glued
+    back together equal the TP=1 load — including the KV-replication
+    branch (tp=4 > kv_heads=2)."""
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=96, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=1, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=32, tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    cfg = loader.config_from_dir(str(tmp_path))
+    cfg.dtype = torch.float32
+    full = {n: torch.zeros(s, dtype=d) for n, s, d in cfg.param_specs()}
+    loader.load_into_params(str(tmp_path), full, cfg=cfg)
+
+    for tp in (2, 4):  # tp=4 exercises KV replication (kv_heads=2)
+        shards = []
+        for r in range(tp):
+            p = {n: torch.zeros(s, dtype=d)
+                 for n, s, d in cfg.param_specs(r, tp)}
+            loader.load_into_params(str(tmp_path), p, tp_rank=r,
+                                    tp_size=tp, cfg=cfg)
+            shards.append(p)
+        name = "layers.0.wq.weight"
+        glued = torch.cat([s[name] for s in shards], dim=0)
+        assert torch.equal(glued, full[name]), f"tp={tp} wq"
+        name = "layers.0.wo.weight"
+        glued = torch.cat([s[name] for s in shards], dim=1)
+        assert torch.equal(glued, full[name]), f"tp={tp} wo"
+        # wk: sliced at tp=2, replicated by KV group at tp=4
+        name = "layers.0.wk.weight"
+        kvh, hd = cfg.num_kv_heads, cfg.head_dim
+        for r in range(tp):
+            if tp <= kvh:
+                lo = r * (kvh // tp) * hd
+                want = full[name][lo:lo + (kvh // tp) * hd]
+            else:
+                g = r * kvh // tp
+                want = full[name][g * hd:(g + 1) * hd]
+            assert torch.equal(shards[r][name], want), f"tp={tp} r={r} wk"
