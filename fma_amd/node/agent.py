@@ -173,9 +173,19 @@ class NodeAgent:
         log_path = os.path.join(self.log_dir,
                                 f"pod-{ob.name_of(pod)}.log")
         logf = open(log_path, "ab")
+        preexec = None
+        if os.environ.get("FMA_POD_PDEATHSIG") == "1":
+            # test harness: pods die with the agent process, so a
+            # hard-killed pytest run (timeout, SIGKILL) cannot leave
+            # orphaned servers squatting on fixed ports for later runs
+            def preexec():
+                import ctypes
+                ctypes.CDLL(None).prctl(1, signal.SIGKILL)  # PR_SET_PDEATHSIG
+                os.setsid()
         proc = subprocess.Popen(cmd, env=full_env, stdout=logf,
                                 stderr=subprocess.STDOUT,
-                                start_new_session=True)
+                                start_new_session=preexec is None,
+                                preexec_fn=preexec)
         pp = PodProcess(pod, ip, proc, probe)
         # kept for restartPolicy:Always respawns (kubelet semantics)
         pp.cmd = list(cmd)

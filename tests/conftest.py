@@ -7,6 +7,9 @@ import pytest
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
+os.environ.setdefault("FMA_POD_PDEATHSIG", "1")
+
+
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: needs a real MI355X (run via gpurun / driver)")

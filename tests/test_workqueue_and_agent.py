@@ -512,3 +512,58 @@ def test_store_indexes_consistent_under_random_crud_fuzz():
                     assert got == expect[key], (iname, key, step)
                 # spot-check an absent key returns empty
                 assert st.index_get("Pod", iname, "no-such-key") == []
+
+
+def test_pod_pdeathsig_reaps_on_hard_agent_kill(tmp_path):
+    """FMA_POD_PDEATHSIG=1 (set by the test harness): pods die when the
+    agent PROCESS is SIGKILLed, so a hard-killed pytest run cannot leave
+    orphaned servers on fixed ports poisoning later runs (observed when
+    two oversubscribed suites aborted mid-e2e)."""
+    import os
+    import signal
+    import subprocess
+    import sys
+    import time
+
+    script = r"""
+import os, sys, time
+sys.path.insert(0, os.getcwd())
+os.environ["FMA_POD_PDEATHSIG"] = "1"
+from fma_amd.node.agent import NodeAgent
+from fma_amd.store.memstore import MemStore
+from fma_amd.store import objects as ob
+st = MemStore()
+agent = NodeAgent(st, "node-t", log_dir=sys.argv[1])
+st.create(ob.new_object("Pod", "sleepy", spec={
+    "nodeName": "node-t",
+    "containers": [{"name": "main",
+                    "command": [sys.executable, "-c",
+                                "import time; time.sleep(120)"]}]}))
+agent._sync_all()
+pp = agent.pods["sleepy"]
+print(f"CHILD={pp.proc.pid}", flush=True)
+time.sleep(120)
+"""
+    proc = subprocess.Popen([sys.executable, "-c", script, str(tmp_path)],
+                            stdout=subprocess.PIPE, text=True)
+    try:
+        line = proc.stdout.readline()
+        assert line.startswith("CHILD="), line
+        child_pid = int(line.split("=")[1])
+        # child alive while the agent process lives
+        os.kill(child_pid, 0)
+        proc.kill()  # SIGKILL: no teardown runs in the agent process
+        proc.wait(timeout=10)
+        deadline = time.time() + 5
+        gone = False
+        while time.time() < deadline:
+            try:
+                os.kill(child_pid, 0)
+            except ProcessLookupError:
+                gone = True
+                break
+            time.sleep(0.1)
+        assert gone, f"pod process {child_pid} survived agent SIGKILL"
+    finally:
+        if proc.poll() is None:
+            proc.kill()
