@@ -184,3 +184,32 @@ def test_hf_checkpoint_tp_sharding_composes(tmp_path):
                 g = r * kvh // tp
                 want = full[name][g * hd:(g + 1) * hd]
             assert torch.equal(shards[r][name], want), f"tp={tp} r={r} wk"
+
+
+def test_engine_arena_fast_path_loads_hf(tmp_path):
+    """The engine's pinned fast-load path (stage -> host buffer -> arena)
+    converts HF tensors too, including mirroring embed bytes into the
+    tied lm_head slot; greedy tokens match transformers."""
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+
+    from fma_amd.runtime.engine import ActuationEngine
+
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    cfg = loader.config_from_dir(str(tmp_path))
+    eng = ActuationEngine(cfg, init_weights=False)
+    assert eng.actuation_mode == "arena" and eng.arena is not None
+    eng.load_checkpoint(str(tmp_path))
+    assert torch.equal(eng.params["lm_head.weight"],
+                       eng.params["embed.weight"])  # tied
+    torch.manual_seed(5)
+    prompt = torch.randint(0, 128, (1, 6))
+    with torch.no_grad():
+        ref = hf.generate(prompt, max_new_tokens=4, do_sample=False,
+                          use_cache=True)
+    got = eng.generate(prompt, max_new_tokens=4)
+    # engine runs bf16: agreement on prompt + first generated tokens
+    assert torch.equal(got[0, :8], ref[0, :8])
