@@ -280,3 +280,59 @@ def test_stopped_controller_does_not_write():
         "annotations", {})
     assert not w["launcher"].wake_calls if hasattr(
         w["launcher"], "wake_calls") else True
+
+
+def test_reconcile_in_non_default_namespace():
+    """Whole hot-start flow in namespace 'prod': the controllers are
+    one-namespace scoped like the reference (ControllerConfig.namespace)
+    and nothing in the reconcile path may assume 'default'."""
+    import tests.test_dualpods_controller as _c
+    FakeHttp, FakeLauncher = _c.FakeHttp, _c.FakeLauncher
+    FakeInstanceServer, Stub = _c.FakeInstanceServer, _c.Stub
+    import copy as _copy
+
+    ns = "prod"
+    store = _c.MemStore()  # conformance suite patches this to KubeStore
+    http = FakeHttp()
+    ctl = DualPodsController(store, http, ControllerConfig(namespace=ns))
+    msc = _copy.deepcopy(MSC)
+    store.create(ob.new_object(
+        "InferenceServerConfig", "isc1", namespace=ns,
+        spec={"modelServerConfig": msc, "launcherConfigName": "lc1"}))
+    store.create(ob.new_object(
+        "LauncherConfig", "lc1", namespace=ns,
+        spec={"maxInstances": 2, "podTemplate": {}}))
+    req = store.create(ob.new_object(
+        "Pod", "req1", namespace=ns,
+        annotations={C.INFERENCE_SERVER_CONFIG_ANNOTATION: "isc1"},
+        spec={"nodeName": "node-a", "containers": [{"name": "stub"}]}))
+    req["status"] = {"phase": "Running", "podIP": "10.0.9.1"}
+    store.update(req)
+    http.register("10.0.9.1:8081", Stub(["GPU-0"]))
+
+    iid = instance_id(msc, ["GPU-0"])
+    lp = store.create(ob.new_object(
+        "Pod", "launcher1", namespace=ns,
+        labels={C.COMPONENT_LABEL: C.LAUNCHER_COMPONENT,
+                C.LAUNCHER_CONFIG_NAME_LABEL: "lc1",
+                C.SLEEPING_LABEL: "true"},
+        annotations={"dual-pods.llm-d.ai/max-instances": "2"},
+        spec={"nodeName": "node-a", "containers": [{"name": "launcher"}]}))
+    lp["status"] = {"phase": "Running", "podIP": "10.0.9.2"}
+    ob.set_pod_ready(lp, True)
+    store.update(lp)
+    fl = FakeLauncher(http, "10.0.9.2")
+    http.register("10.0.9.2:8001", fl)
+    fl.instances[iid] = {
+        "instance_id": iid, "status": "running",
+        "options": "--model tiny --port 8000", "gpu_uuids": ["GPU-0"],
+        "annotations": {"isc-name": "isc1", "inference-port": "8000"}}
+    http.register("10.0.9.2:8000", FakeInstanceServer(sleeping=True))
+
+    drive(ctl, ("infsvr", "node-a", ob.uid_of(store.get(
+        "Pod", "req1", ns)), "req1"))
+    bound = store.get("Pod", "launcher1", ns)
+    assert C.REQUESTER_ANNOTATION in ob.annotations_of(bound)
+    assert ob.annotations_of(bound)[C.INSTANCE_ID_ANNOTATION] == iid
+    # nothing leaked into the default namespace
+    assert store.list("Pod", "default") == []
