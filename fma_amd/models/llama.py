@@ -58,11 +58,15 @@ class LlamaConfig:
     #: all-reduce sums the routed outputs. False = Megatron-shard every
     #: expert's MLP like the dense path.
     expert_parallel: bool = False
+    #: explicit per-head dim for models where it differs from
+    #: hidden_size // num_heads (HF configs carry "head_dim"); None
+    #: derives it
+    head_dim_override: "int | None" = None
     dtype: torch.dtype = torch.bfloat16
 
     @property
     def head_dim(self) -> int:
-        return self.hidden_size // self.num_heads
+        return self.head_dim_override or self.hidden_size // self.num_heads
 
     def local_experts(self, tp_rank: int = 0, tp_size: int = 1
                       ) -> range:

@@ -52,6 +52,7 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
         qkv_bias=d.get("qkv_bias", d.get("model_type") == "qwen2"),
         num_experts=d.get("num_experts", d.get("num_local_experts", 0)),
         num_experts_per_tok=d.get("num_experts_per_tok", 2),
+        head_dim_override=d.get("head_dim"),
     )
 
 
@@ -103,6 +104,8 @@ def save_params(params: Dict[str, torch.Tensor], path: str,
                 "qkv_bias": cfg.qkv_bias,
                 "num_experts": cfg.num_experts,
                 "num_experts_per_tok": cfg.num_experts_per_tok,
+                **({"head_dim": cfg.head_dim_override}
+                   if cfg.head_dim_override else {}),
             }, f)
 
 

@@ -213,3 +213,18 @@ def test_engine_arena_fast_path_loads_hf(tmp_path):
     got = eng.generate(prompt, max_new_tokens=4)
     # engine runs bf16: agreement on prompt + first generated tokens
     assert torch.equal(got[0, :8], ref[0, :8])
+
+
+def test_llama_hf_explicit_head_dim(tmp_path):
+    """Models whose head_dim differs from hidden//heads (HF configs
+    carry an explicit "head_dim") load and match transformers."""
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=96, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=2,
+        head_dim=16,  # derived would be 32
+        max_position_embeddings=32, tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    cfg = loader.config_from_dir(str(tmp_path))
+    assert cfg.head_dim == 16
+    _compare_logits(hf, str(tmp_path))
