@@ -62,6 +62,10 @@ class LlamaConfig:
     #: hidden_size // num_heads (HF configs carry "head_dim"); None
     #: derives it
     head_dim_override: "int | None" = None
+    #: HF-style rope_scaling dict ({"rope_type": "llama3"|"linear",
+    #: "factor": ..., ...}) — llama-3.1+ checkpoints need it for
+    #: correct frequencies; None = plain rope
+    rope_scaling: "dict | None" = None
     dtype: torch.dtype = torch.bfloat16
 
     @property
@@ -254,11 +258,39 @@ def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
     return (xf * w.float()).to(x.dtype)
 
 
+def _scale_inv_freq(inv: torch.Tensor, scaling: dict) -> torch.Tensor:
+    """HF rope_scaling: "linear" divides every frequency by factor;
+    "llama3" (llama-3.1+) keeps high frequencies, divides low ones and
+    smoothly interpolates between (the transformers rule)."""
+    rtype = scaling.get("rope_type", scaling.get("type", ""))
+    factor = float(scaling.get("factor", 1.0))
+    if rtype == "linear":
+        return inv / factor
+    if rtype == "llama3":
+        low = float(scaling.get("low_freq_factor", 1.0))
+        high = float(scaling.get("high_freq_factor", 4.0))
+        orig = float(scaling.get("original_max_position_embeddings", 8192))
+        import math as _m
+        wavelen = 2 * _m.pi / inv
+        low_wl = orig / low
+        high_wl = orig / high
+        out = torch.where(wavelen > low_wl, inv / factor, inv)
+        smooth = (orig / wavelen - low) / (high - low)
+        smoothed = (1 - smooth) * out / factor + smooth * out
+        mid = (wavelen >= high_wl) & (wavelen <= low_wl)
+        return torch.where(mid, smoothed, out)
+    if rtype in ("", "default"):
+        return inv
+    raise ValueError(f"unsupported rope_scaling type {rtype!r}")
+
+
 def precompute_rope(cfg: LlamaConfig, device) -> Tuple[torch.Tensor, torch.Tensor]:
     """Host-precomputed cos/sin tables (trig on device turns a memory-bound
     op VALU-bound — cdna_hip_programming.md Appendix B / element-wise)."""
     hd = cfg.head_dim
     inv = 1.0 / (cfg.rope_theta ** (torch.arange(0, hd, 2, dtype=torch.float32) / hd))
+    if cfg.rope_scaling:
+        inv = _scale_inv_freq(inv, cfg.rope_scaling)
     t = torch.arange(cfg.max_seq_len, dtype=torch.float32)
     freqs = torch.outer(t, inv)
     return freqs.cos().to(device), freqs.sin().to(device)

@@ -35,6 +35,13 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
         return None
     with open(cfg_path) as f:
         d = json.load(f)
+    # transformers >= 5 nests rope config under "rope_parameters"
+    # (rope_theta included); older configs use flat rope_theta +
+    # rope_scaling
+    rp = d.get("rope_parameters") or {}
+    scaling = d.get("rope_scaling")
+    if scaling is None and rp.get("rope_type", "default") != "default":
+        scaling = {k: v for k, v in rp.items() if k != "rope_theta"}
     return LlamaConfig(
         name=d.get("name", os.path.basename(path.rstrip("/"))),
         vocab_size=d.get("vocab_size", 32768),
@@ -45,7 +52,7 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
         num_kv_heads=d.get("num_kv_heads", d.get("num_key_value_heads", 8)),
         max_seq_len=d.get("max_seq_len", d.get("max_position_embeddings",
                                                4096)),
-        rope_theta=d.get("rope_theta", 500000.0),
+        rope_theta=d.get("rope_theta", rp.get("rope_theta", 500000.0)),
         norm_eps=d.get("norm_eps", d.get("rms_norm_eps", 1e-5)),
         # transformers configs carry model_type instead of qkv_bias;
         # qwen2's attention projections are the biased ones
@@ -53,6 +60,7 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
         num_experts=d.get("num_experts", d.get("num_local_experts", 0)),
         num_experts_per_tok=d.get("num_experts_per_tok", 2),
         head_dim_override=d.get("head_dim"),
+        rope_scaling=scaling,
     )
 
 
@@ -106,6 +114,8 @@ def save_params(params: Dict[str, torch.Tensor], path: str,
                 "num_experts_per_tok": cfg.num_experts_per_tok,
                 **({"head_dim": cfg.head_dim_override}
                    if cfg.head_dim_override else {}),
+                **({"rope_scaling": cfg.rope_scaling}
+                   if cfg.rope_scaling else {}),
             }, f)
 
 

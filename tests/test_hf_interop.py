@@ -228,3 +228,47 @@ def test_llama_hf_explicit_head_dim(tmp_path):
     cfg = loader.config_from_dir(str(tmp_path))
     assert cfg.head_dim == 16
     _compare_logits(hf, str(tmp_path))
+
+
+def test_llama31_rope_scaling_matches(tmp_path):
+    """llama-3.1-style rope_scaling ("llama3" type): scaled frequencies
+    must reproduce transformers' logits — plain rope would diverge."""
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=96, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=2,
+        max_position_embeddings=64, rope_theta=10000.0,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 16},
+        tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    cfg = loader.config_from_dir(str(tmp_path))
+    assert cfg.rope_scaling and cfg.rope_scaling["factor"] == 8.0
+    _compare_logits(hf, str(tmp_path))
+    # sanity: ignoring the scaling WOULD diverge (the test has teeth)
+    cfg2 = loader.config_from_dir(str(tmp_path))
+    cfg2.dtype = torch.float32
+    cfg2.rope_scaling = None
+    params = {n: torch.zeros(s, dtype=d)
+              for n, s, d in cfg2.param_specs()}
+    loader.load_into_params(str(tmp_path), params, cfg=cfg2)
+    unscaled = LlamaModel(cfg2, params, torch.device("cpu"))
+    torch.manual_seed(7)
+    tokens = torch.randint(0, cfg2.vocab_size, (1, 12))
+    with torch.no_grad():
+        ref = hf(tokens).logits
+        got = unscaled.forward(tokens)
+    assert not torch.allclose(got, ref, rtol=2e-3, atol=2e-3)
+
+
+def test_linear_rope_scaling_matches(tmp_path):
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=96, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=1,
+        max_position_embeddings=64,
+        rope_scaling={"rope_type": "linear", "factor": 2.0},
+        tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    _compare_logits(hf, str(tmp_path))
