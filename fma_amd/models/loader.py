@@ -38,6 +38,14 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
     # transformers >= 5 nests rope config under "rope_parameters"
     # (rope_theta included); older configs use flat rope_theta +
     # rope_scaling
+    sw = d.get("sliding_window")
+    if sw and (d.get("use_sliding_window", True)
+               and d.get("max_position_embeddings", 0) > sw):
+        # full-window attention only: silently truncating context would
+        # be wrong numerics, so refuse loudly
+        raise ValueError(
+            f"checkpoint uses sliding-window attention (window={sw}) "
+            "beyond its window; not supported by this engine")
     rp = d.get("rope_parameters") or {}
     scaling = d.get("rope_scaling")
     if scaling is None and rp.get("rope_type", "default") != "default":

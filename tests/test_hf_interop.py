@@ -272,3 +272,25 @@ def test_linear_rope_scaling_matches(tmp_path):
         tie_word_embeddings=True)).eval()
     hf.save_pretrained(tmp_path, safe_serialization=True)
     _compare_logits(hf, str(tmp_path))
+
+
+def test_sliding_window_checkpoint_refused(tmp_path):
+    """A config that actually relies on sliding-window attention is
+    refused loudly instead of silently serving wrong numerics."""
+    import json
+    import os
+    os.makedirs(tmp_path, exist_ok=True)
+    with open(tmp_path / "config.json", "w") as f:
+        json.dump({"model_type": "qwen2", "hidden_size": 64,
+                   "num_attention_heads": 2, "num_hidden_layers": 1,
+                   "use_sliding_window": True, "sliding_window": 1024,
+                   "max_position_embeddings": 32768}, f)
+    with pytest.raises(ValueError, match="sliding-window"):
+        loader.config_from_dir(str(tmp_path))
+    # window >= context: plain full attention is exact; accepted
+    with open(tmp_path / "config.json", "w") as f:
+        json.dump({"model_type": "qwen2", "hidden_size": 64,
+                   "num_attention_heads": 2, "num_hidden_layers": 1,
+                   "use_sliding_window": True, "sliding_window": 32768,
+                   "max_position_embeddings": 4096}, f)
+    assert loader.config_from_dir(str(tmp_path)) is not None
