@@ -121,6 +121,13 @@ class TPRuntime:
         return self.engine.generate(tokens, max_new_tokens)
 
     def generate_text(self, prompt: str, max_new_tokens: int = 16) -> str:
+        tok = getattr(self.engine, "tokenizer", None)
+        if tok is not None:
+            ids = tok.encode(prompt).ids or [1]
+            toks = torch.tensor([ids], dtype=torch.long,
+                                device=self.engine.device)
+            out = self.generate(toks, max_new_tokens)[0, len(ids):]
+            return tok.decode([int(t) for t in out.tolist()])
         ids = [b % self.engine.cfg.vocab_size for b in prompt.encode("utf-8")] \
             or [1]
         toks = torch.tensor([ids], dtype=torch.long, device=self.engine.device)

@@ -275,8 +275,16 @@ class ActuationEngine:
         return dec
 
     def generate_text(self, prompt: str, max_new_tokens: int = 16) -> str:
-        """Byte-level round trip (no tokenizer assets offline): UTF-8 bytes
-        are the token ids; generated ids map back to bytes."""
+        """Text round trip. With a real tokenizer attached (HF checkpoint
+        dirs carry tokenizer.json — see runtime/server.py), prompts
+        encode/decode through it; otherwise UTF-8 bytes are the token
+        ids (self-contained fallback for synthetic models)."""
+        tok = getattr(self, "tokenizer", None)
+        if tok is not None:
+            ids = tok.encode(prompt).ids or [1]
+            toks = torch.tensor([ids], dtype=torch.long, device=self.device)
+            out = self.generate(toks, max_new_tokens)[0, len(ids):]
+            return tok.decode([int(t) for t in out.tolist()])
         ids = [b % self.cfg.vocab_size for b in prompt.encode("utf-8")] or [1]
         toks = torch.tensor([ids], dtype=torch.long, device=self.device)
         out = self.generate(toks, max_new_tokens)[0, len(ids):]

@@ -98,6 +98,22 @@ def _worker_entry(rank: int, world: int, master_port: int,
     tp.run_worker_loop(ctx, eng)
 
 
+def load_tokenizer(checkpoint_dir: str):
+    """HF checkpoint dirs carry tokenizer.json; attach the real
+    tokenizer so /v1/completions speaks text, not byte ids. Returns
+    None (byte-level fallback) when absent or unloadable."""
+    path = os.path.join(checkpoint_dir, "tokenizer.json")
+    if not os.path.exists(path):
+        return None
+    try:
+        from tokenizers import Tokenizer
+        return Tokenizer.from_file(path)
+    except Exception as e:  # noqa: BLE001 - degraded, not fatal
+        print(f"[serve] tokenizer load failed ({e}); byte-level fallback",
+              flush=True)
+        return None
+
+
 class ServingRuntime:
     """Everything behind the HTTP app: engine (+TP fan-out) and state."""
 
@@ -143,6 +159,7 @@ class ServingRuntime:
             t_load = engine.load_checkpoint(checkpoint)
             print(f"[serve] checkpoint {checkpoint} loaded in {t_load:.2f}s",
                   flush=True)
+            engine.tokenizer = load_tokenizer(checkpoint)
         self.rt = tp.TPRuntime(ctx, engine) if world > 1 else engine
         self.model_name = args.served_model_name or args.model
         print(f"[serve] engine up in {engine.create_seconds:.2f}s "

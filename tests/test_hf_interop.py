@@ -294,3 +294,37 @@ def test_sliding_window_checkpoint_refused(tmp_path):
                    "use_sliding_window": True, "sliding_window": 32768,
                    "max_position_embeddings": 4096}, f)
     assert loader.config_from_dir(str(tmp_path)) is not None
+
+
+def test_real_tokenizer_from_checkpoint_dir(tmp_path):
+    """A tokenizer.json beside the weights makes /v1/completions speak
+    real text: encode through the checkpoint's tokenizer, decode the
+    generated ids back."""
+    pytest.importorskip("tokenizers")
+    from tokenizers import Tokenizer
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    from fma_amd.runtime.server import ServingRuntime, parse_options
+
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=32, hidden_size=32, intermediate_size=48,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=1,
+        max_position_embeddings=32, tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    vocab = {f"w{i}": i for i in range(30)}
+    vocab["[UNK]"] = 30
+    tok = Tokenizer(WordLevel(vocab, unk_token="[UNK]"))
+    tok.pre_tokenizer = Whitespace()
+    tok.save(str(tmp_path / "tokenizer.json"))
+
+    rt = ServingRuntime(parse_options(f"--model {tmp_path}"))
+    eng = rt.rt.engine if hasattr(rt.rt, "engine") else rt.rt
+    assert eng.tokenizer is not None
+    assert eng.tokenizer.encode("w1 w2 w7").ids == [1, 2, 7]
+    text = rt.rt.generate_text("w1 w2 w7", max_new_tokens=4)
+    # decoded ids are words from the vocab, not raw bytes
+    assert all(t in vocab for t in text.split()), text
+    if hasattr(rt.rt, "stop"):
+        rt.rt.stop()
