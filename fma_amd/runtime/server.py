@@ -218,6 +218,26 @@ def create_app(runtime: ServingRuntime) -> FastAPI:
                 return JSONResponse(
                     {"error": "model is sleeping"}, status_code=409)
             t0 = time.time()
+            if isinstance(prompt, list):
+                # OpenAI-compat: prompt may be token ids; respond with
+                # the generated ids so clients control detokenization
+                import torch as _torch
+                eng = rt.engine if hasattr(rt, "engine") else rt
+                ids = [int(t) % eng.cfg.vocab_size for t in prompt] or [1]
+                toks = _torch.tensor([ids], dtype=_torch.long,
+                                     device=eng.device)
+                out = rt.generate(toks, max_tokens)[0, len(ids):]
+                return {
+                    "id": f"cmpl-{int(t0*1e6)}",
+                    "object": "text_completion",
+                    "model": runtime.model_name,
+                    "choices": [{"index": 0,
+                                 "token_ids": [int(t) for t in out],
+                                 "text": "",
+                                 "finish_reason": "length"}],
+                    "usage": {"prompt_tokens": len(ids),
+                              "completion_tokens": max_tokens},
+                }
             text = rt.generate_text(prompt, max_tokens)
         return {
             "id": f"cmpl-{int(t0*1e6)}",

@@ -121,3 +121,14 @@ def test_runtime_serves_hf_checkpoint_dir(tmp_path):
     assert torch.equal(r.generate(prompt, max_new_tokens=6), got)
     if hasattr(r, "stop"):
         r.stop()
+
+
+def test_completions_accepts_token_ids(client):
+    """OpenAI-compat: a list-of-ints prompt decodes from raw ids and
+    returns generated token_ids (clients own detokenization)."""
+    r = client.post("/v1/completions",
+                    json={"prompt": [1, 2, 3], "max_tokens": 4})
+    assert r.status_code == 200
+    choice = r.json()["choices"][0]
+    assert len(choice["token_ids"]) == 4
+    assert r.json()["usage"]["prompt_tokens"] == 3
