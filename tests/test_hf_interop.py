@@ -328,3 +328,36 @@ def test_real_tokenizer_from_checkpoint_dir(tmp_path):
     assert all(t in vocab for t in text.split()), text
     if hasattr(rt.rt, "stop"):
         rt.rt.stop()
+
+
+def test_hf_mixtral_expert_parallel_slicing(tmp_path):
+    """EP over an HF Mixtral checkpoint: each rank loads only its WHOLE
+    experts (full tensors, global indices), skipping foreign ones."""
+    from transformers import MixtralConfig, MixtralForCausalLM
+    hf = MixtralForCausalLM(MixtralConfig(
+        vocab_size=64, hidden_size=32, intermediate_size=48,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=2,
+        num_local_experts=4, num_experts_per_tok=2,
+        max_position_embeddings=32, tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    cfg = loader.config_from_dir(str(tmp_path))
+    cfg.dtype = torch.float32
+    cfg.expert_parallel = True
+    for rank in (0, 1):
+        params = {n: torch.zeros(s, dtype=d)
+                  for n, s, d in cfg.param_specs(rank, 2)}
+        mine = list(cfg.local_experts(rank, 2))
+        assert mine == ([0, 1] if rank == 0 else [2, 3])
+        n = loader.load_into_params(
+            str(tmp_path), params, tp_rank=rank, tp_size=2, cfg=cfg,
+            skip=lambda name: ".experts." in name)
+        assert n == len(params)
+        from safetensors import safe_open
+        with safe_open(str(tmp_path / "model.safetensors"),
+                       framework="pt") as sf:
+            for e in mine:
+                hf_w1 = sf.get_tensor(
+                    f"model.layers.0.block_sparse_moe.experts.{e}"
+                    ".w1.weight")
+                assert torch.equal(
+                    params[f"layers.0.experts.{e}.w_gate.weight"], hf_w1)
