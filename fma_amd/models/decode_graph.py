@@ -50,8 +50,10 @@ class StaticDecoder:
 
     @staticmethod
     def supported(cfg) -> bool:
-        # MoE routing is data-dependent per token: un-capturable
-        return cfg.vocab_size <= GRAPH_SAFE_VOCAB and not cfg.num_experts
+        # MoE routing is data-dependent per token: un-capturable;
+        # qk_norm (qwen3) is outside the fused qkv+rope capture set
+        return (cfg.vocab_size <= GRAPH_SAFE_VOCAB and not cfg.num_experts
+                and not cfg.qk_norm)
 
     def __init__(self, model: LlamaModel, batch: int, max_seq: int):
         assert model.tp_size == 1, "graphed decode is single-rank for now"

@@ -49,6 +49,9 @@ class LlamaConfig:
     #: Qwen2-family: biases on the q/k/v projections (the only
     #: architectural delta vs llama the actuation path sees)
     qkv_bias: bool = False
+    #: Qwen3-family: RMSNorm over head_dim on q and k before rope
+    #: (one (hd,) gain shared across heads, per layer)
+    qk_norm: bool = False
     #: Mixtral-family: sparse MoE MLP (0 = dense). Routed top-k over
     #: num_experts per token; each expert is a llama-shaped gated MLP.
     num_experts: int = 0
@@ -212,6 +215,11 @@ class LlamaConfig:
                     (p + "wq.bias", (q_local,), d),
                     (p + "wk.bias", (kv_local,), d),
                     (p + "wv.bias", (kv_local,), d),
+                ]
+            if self.qk_norm:
+                specs += [
+                    (p + "q_norm.weight", (hd,), d),
+                    (p + "k_norm.weight", (hd,), d),
                 ]
             specs += [
                 (p + "wo.weight", (h, q_local), d),
@@ -460,8 +468,11 @@ class LlamaModel:
             q = q.view(B, T, q_heads, hd)
             k = k.view(B, T, kv_heads, hd)
             v = v.view(B, T, kv_heads, hd)
+            if cfg.qk_norm:
+                q = rmsnorm(q, P[p + "q_norm.weight"], cfg.norm_eps)
+                k = rmsnorm(k, P[p + "k_norm.weight"], cfg.norm_eps)
             stored = False
-            if decode1 and cache is not None:
+            if decode1 and cache is not None and not cfg.qk_norm:
                 # RoPE(q) + RoPE(k)->cache + v->cache in ONE launch
                 stored = fast_rope_qkv_store(
                     q, k, v, cache.data[li, 0, 0], cache.data[li, 1, 0],

@@ -65,6 +65,7 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
         # transformers configs carry model_type instead of qkv_bias;
         # qwen2's attention projections are the biased ones
         qkv_bias=d.get("qkv_bias", d.get("model_type") == "qwen2"),
+        qk_norm=d.get("qk_norm", d.get("model_type") == "qwen3"),
         num_experts=d.get("num_experts", d.get("num_local_experts", 0)),
         num_experts_per_tok=d.get("num_experts_per_tok", 2),
         head_dim_override=d.get("head_dim"),
@@ -118,6 +119,7 @@ def save_params(params: Dict[str, torch.Tensor], path: str,
                 "rope_theta": cfg.rope_theta,
                 "norm_eps": cfg.norm_eps,
                 "qkv_bias": cfg.qkv_bias,
+                "qk_norm": cfg.qk_norm,
                 "num_experts": cfg.num_experts,
                 "num_experts_per_tok": cfg.num_experts_per_tok,
                 **({"head_dim": cfg.head_dim_override}
@@ -159,6 +161,10 @@ def map_hf_name(name: str) -> Optional[str]:
     p, leaf = f"layers.{m.group(1)}.", m.group(2)
     if leaf == "input_layernorm.weight":
         return p + "attn_norm.weight"
+    if leaf == "self_attn.q_norm.weight":
+        return p + "q_norm.weight"
+    if leaf == "self_attn.k_norm.weight":
+        return p + "k_norm.weight"
     if leaf == "post_attention_layernorm.weight":
         return p + "mlp_norm.weight"
     if leaf == "block_sparse_moe.gate.weight":
@@ -197,6 +203,10 @@ def hf_convert(name: str, tensor: torch.Tensor, cfg: "LlamaConfig"):
     if leaf in ("wq", "wk"):
         heads = cfg.num_heads if leaf == "wq" else cfg.num_kv_heads
         tensor = _unrotate_half(tensor, heads, cfg.head_dim)
+    elif leaf in ("q_norm", "k_norm"):
+        # the per-head-dim gain rides BEFORE rope, so it permutes the
+        # same way the q/k projection rows do (single head's worth)
+        tensor = _unrotate_half(tensor, 1, cfg.head_dim)
     return our, tensor
 
 

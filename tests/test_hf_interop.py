@@ -361,3 +361,28 @@ def test_hf_mixtral_expert_parallel_slicing(tmp_path):
                     ".w1.weight")
                 assert torch.equal(
                     params[f"layers.0.experts.{e}.w_gate.weight"], hf_w1)
+
+
+def test_qwen3_hf_checkpoint_logits_match(tmp_path):
+    """Qwen3: per-head q/k RMSNorm before rope (the norm gain permutes
+    like a single head's projection rows), explicit head_dim, no
+    attention biases."""
+    from transformers import Qwen3Config, Qwen3ForCausalLM
+    hf = Qwen3ForCausalLM(Qwen3Config(
+        vocab_size=96, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=64,
+        tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    cfg = loader.config_from_dir(str(tmp_path))
+    assert cfg.qk_norm and not cfg.qkv_bias and cfg.head_dim == 16
+    _compare_logits(hf, str(tmp_path))
+    # greedy tokens incl. the cached decode path
+    _, ours = _load_ours(str(tmp_path))
+    torch.manual_seed(3)
+    prompt = torch.randint(0, 96, (1, 7))
+    with torch.no_grad():
+        ref = hf.generate(prompt, max_new_tokens=8, do_sample=False,
+                          use_cache=True)
+        got = ours.generate(prompt, max_new_tokens=8)
+    assert torch.equal(got, ref[:, :got.shape[1]])
