@@ -110,6 +110,22 @@ class LlamaConfig:
                            num_kv_heads=2, max_seq_len=256, qkv_bias=True)
 
     @staticmethod
+    def tiny_qwen3() -> "LlamaConfig":
+        return LlamaConfig(name="tiny-qwen3", vocab_size=512,
+                           hidden_size=64, intermediate_size=128,
+                           num_layers=2, num_heads=4, num_kv_heads=2,
+                           max_seq_len=256, qk_norm=True)
+
+    @staticmethod
+    def qwen3_8b() -> "LlamaConfig":
+        # Qwen3-8B shape: explicit head_dim 128, GQA 32/8, qk norms
+        return LlamaConfig(name="qwen3-8b", vocab_size=151936,
+                           hidden_size=4096, intermediate_size=12288,
+                           num_layers=36, num_heads=32, num_kv_heads=8,
+                           head_dim_override=128, max_seq_len=8192,
+                           rope_theta=1000000.0, qk_norm=True)
+
+    @staticmethod
     def mixtral_8x7b() -> "LlamaConfig":
         return LlamaConfig(name="mixtral-8x7b", vocab_size=32000,
                            hidden_size=4096, intermediate_size=14336,
@@ -173,6 +189,8 @@ class LlamaConfig:
             "tiny": LlamaConfig.tiny,
             "tiny-qwen": LlamaConfig.tiny_qwen,
             "tiny-moe": LlamaConfig.tiny_moe,
+            "tiny-qwen3": LlamaConfig.tiny_qwen3,
+            "qwen3-8b": LlamaConfig.qwen3_8b,
             "llama-3-8b": LlamaConfig.llama3_8b,
             "llama-3-70b": LlamaConfig.llama3_70b,
             "qwen2-7b": LlamaConfig.qwen2_7b,
