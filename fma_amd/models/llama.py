@@ -52,6 +52,14 @@ class LlamaConfig:
     #: Qwen3-family: RMSNorm over head_dim on q and k before rope
     #: (one (hd,) gain shared across heads, per layer)
     qk_norm: bool = False
+    #: MLP activation: "silu" (llama/qwen/mixtral) or "gelu_tanh"
+    #: (gemma). The fused silu kernel only engages for silu.
+    hidden_act: str = "silu"
+    #: Gemma: hidden states scale by sqrt(hidden_size) after embedding.
+    #: (Gemma's (1+w) rmsnorm offset is folded into the weights at
+    #: checkpoint load — loader.hf_convert — so the runtime norm is
+    #: the standard one.)
+    embed_scale: bool = False
     #: Mixtral-family: sparse MoE MLP (0 = dense). Routed top-k over
     #: num_experts per token; each expert is a llama-shaped gated MLP.
     num_experts: int = 0
@@ -387,6 +395,16 @@ class LlamaModel:
             else:
                 p.fill_(1.0)  # norm gains
 
+    def _act(self, gate: torch.Tensor, up: torch.Tensor,
+             decode1: bool) -> torch.Tensor:
+        """Gated-MLP activation: fused silu kernel on the decode path,
+        gelu-tanh (gemma) via torch — the GEMVs dominate either way."""
+        if self.cfg.hidden_act == "silu":
+            return fast_silu_mul(gate, up) if decode1                 else F.silu(gate) * up
+        if self.cfg.hidden_act == "gelu_tanh":
+            return F.gelu(gate, approximate="tanh") * up
+        raise ValueError(f"unknown hidden_act {self.cfg.hidden_act!r}")
+
     def _maybe_all_reduce(self, x: torch.Tensor) -> torch.Tensor:
         if self.tp_size > 1:
             dist.all_reduce(x, group=self.tp_group)
@@ -418,8 +436,7 @@ class LlamaModel:
                 ep = p + f"experts.{e}."
                 gate, up = fast_linear_multi(
                     h, (P[ep + "w_gate.weight"], P[ep + "w_up.weight"]))
-                act = fast_silu_mul(gate, up) if decode1 \
-                    else F.silu(gate) * up
+                act = self._act(gate, up, decode1)
                 w = topw.reshape(-1)[j].to(x.dtype)
                 out = out + w * fast_linear(act, P[ep + "w_down.weight"])
         else:
@@ -455,6 +472,8 @@ class LlamaModel:
         hd = cfg.head_dim
 
         x = F.embedding(tokens, P["embed.weight"])
+        if cfg.embed_scale:  # gemma: normalizer on the residual stream
+            x = x * (cfg.hidden_size ** 0.5)
         # fused single-token decode path (one kernel per elementwise op)
         decode1 = (B == 1 and T == 1 and x.is_cuda
                    and x.dtype == torch.bfloat16)
@@ -563,7 +582,7 @@ class LlamaModel:
             # on wide-K models whose x exceeds the LDS stage it
             # recomputes per dot element (70B-shape 74.7 -> 60.5).
             # Selection is measured; the separate activation launch wins.
-            act = fast_silu_mul(gate, up) if decode1 else F.silu(gate) * up
+            act = self._act(gate, up, decode1)
             if decode1 and self.tp_size == 1:
                 x = fast_linear_residual(act, P[p + "w_down.weight"], x)
             else:

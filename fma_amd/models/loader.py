@@ -50,7 +50,7 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
     scaling = d.get("rope_scaling")
     if scaling is None and rp.get("rope_type", "default") != "default":
         scaling = {k: v for k, v in rp.items() if k != "rope_theta"}
-    return LlamaConfig(
+    cfg = LlamaConfig(
         name=d.get("name", os.path.basename(path.rstrip("/"))),
         vocab_size=d.get("vocab_size", 32768),
         hidden_size=d.get("hidden_size", 4096),
@@ -66,11 +66,20 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
         # qwen2's attention projections are the biased ones
         qkv_bias=d.get("qkv_bias", d.get("model_type") == "qwen2"),
         qk_norm=d.get("qk_norm", d.get("model_type") == "qwen3"),
+        hidden_act=("gelu_tanh" if d.get(
+            "hidden_act", d.get("hidden_activation", "")).startswith("gelu")
+            else d.get("hidden_act", "silu")),
+        embed_scale=d.get("embed_scale", d.get("model_type") == "gemma"),
         num_experts=d.get("num_experts", d.get("num_local_experts", 0)),
         num_experts_per_tok=d.get("num_experts_per_tok", 2),
         head_dim_override=d.get("head_dim"),
         rope_scaling=scaling,
     )
+    if d.get("model_type") == "gemma":
+        # load-time marker: fold gemma's (1+w) rmsnorm offset into the
+        # gains during hf_convert (runtime norm stays standard)
+        cfg.norm_plus_one = True
+    return cfg
 
 
 def shard_files(path: str) -> List[str]:
@@ -120,6 +129,8 @@ def save_params(params: Dict[str, torch.Tensor], path: str,
                 "norm_eps": cfg.norm_eps,
                 "qkv_bias": cfg.qkv_bias,
                 "qk_norm": cfg.qk_norm,
+                "hidden_act": cfg.hidden_act,
+                "embed_scale": cfg.embed_scale,
                 "num_experts": cfg.num_experts,
                 "num_experts_per_tok": cfg.num_experts_per_tok,
                 **({"head_dim": cfg.head_dim_override}
@@ -207,6 +218,13 @@ def hf_convert(name: str, tensor: torch.Tensor, cfg: "LlamaConfig"):
         # the per-head-dim gain rides BEFORE rope, so it permutes the
         # same way the q/k projection rows do (single head's worth)
         tensor = _unrotate_half(tensor, 1, cfg.head_dim)
+    if getattr(cfg, "norm_plus_one", False) and (
+            our.endswith("norm.weight") and leaf not in ("q_norm",
+                                                         "k_norm")):
+        # gemma stores rmsnorm gains as offsets from 1 ((1+w)*x̂);
+        # folding the +1 here keeps the runtime norm (and its HIP
+        # kernel) standard
+        tensor = tensor.float() + 1.0
     return our, tensor
 
 

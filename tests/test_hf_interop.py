@@ -386,3 +386,27 @@ def test_qwen3_hf_checkpoint_logits_match(tmp_path):
                           use_cache=True)
         got = ours.generate(prompt, max_new_tokens=8)
     assert torch.equal(got, ref[:, :got.shape[1]])
+
+
+def test_gemma_hf_checkpoint_logits_match(tmp_path):
+    """Gemma: gelu-tanh gated MLP, sqrt(hidden) embedding scale, and the
+    (1+w) rmsnorm offset folded into the gains at load — logits match
+    transformers."""
+    from transformers import GemmaConfig as HFGemma, GemmaForCausalLM
+    hf = GemmaForCausalLM(HFGemma(
+        vocab_size=96, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=64,
+        tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    cfg = loader.config_from_dir(str(tmp_path))
+    assert cfg.hidden_act == "gelu_tanh" and cfg.embed_scale
+    _compare_logits(hf, str(tmp_path))
+    _, ours = _load_ours(str(tmp_path))
+    torch.manual_seed(3)
+    prompt = torch.randint(0, 96, (1, 7))
+    with torch.no_grad():
+        ref = hf.generate(prompt, max_new_tokens=8, do_sample=False,
+                          use_cache=True)
+        got = ours.generate(prompt, max_new_tokens=8)
+    assert torch.equal(got, ref[:, :got.shape[1]])
