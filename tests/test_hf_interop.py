@@ -410,3 +410,16 @@ def test_gemma_hf_checkpoint_logits_match(tmp_path):
                           use_cache=True)
         got = ours.generate(prompt, max_new_tokens=8)
     assert torch.equal(got, ref[:, :got.shape[1]])
+
+
+def test_mistral_hf_checkpoint_logits_match(tmp_path):
+    """Mistral (v0.3-style, no sliding window): llama-architecture with
+    its own model_type — loads and matches transformers."""
+    from transformers import MistralConfig, MistralForCausalLM
+    hf = MistralForCausalLM(MistralConfig(
+        vocab_size=96, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        sliding_window=None, max_position_embeddings=64,
+        tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    _compare_logits(hf, str(tmp_path))
