@@ -134,6 +134,14 @@ class LlamaConfig:
                            rope_theta=1000000.0, qk_norm=True)
 
     @staticmethod
+    def tiny_gemma() -> "LlamaConfig":
+        return LlamaConfig(name="tiny-gemma", vocab_size=512,
+                           hidden_size=64, intermediate_size=128,
+                           num_layers=2, num_heads=4, num_kv_heads=2,
+                           head_dim_override=16, max_seq_len=256,
+                           hidden_act="gelu_tanh", embed_scale=True)
+
+    @staticmethod
     def mixtral_8x7b() -> "LlamaConfig":
         return LlamaConfig(name="mixtral-8x7b", vocab_size=32000,
                            hidden_size=4096, intermediate_size=14336,
@@ -198,6 +206,7 @@ class LlamaConfig:
             "tiny-qwen": LlamaConfig.tiny_qwen,
             "tiny-moe": LlamaConfig.tiny_moe,
             "tiny-qwen3": LlamaConfig.tiny_qwen3,
+            "tiny-gemma": LlamaConfig.tiny_gemma,
             "qwen3-8b": LlamaConfig.qwen3_8b,
             "llama-3-8b": LlamaConfig.llama3_8b,
             "llama-3-70b": LlamaConfig.llama3_70b,

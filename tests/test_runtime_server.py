@@ -73,7 +73,8 @@ def test_runtime_serves_all_three_families():
 
     from fma_amd.runtime.server import ServingRuntime, parse_options
 
-    for preset in ("tiny", "tiny-qwen", "tiny-qwen3", "tiny-moe"):
+    for preset in ("tiny", "tiny-qwen", "tiny-qwen3", "tiny-gemma",
+                   "tiny-moe"):
         rt = ServingRuntime(parse_options(f"--model {preset} --seed 3"))
         r = rt.rt
         assert not r.is_sleeping()
