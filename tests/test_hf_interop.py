@@ -423,3 +423,19 @@ def test_mistral_hf_checkpoint_logits_match(tmp_path):
         tie_word_embeddings=True)).eval()
     hf.save_pretrained(tmp_path, safe_serialization=True)
     _compare_logits(hf, str(tmp_path))
+
+
+def test_sharded_hf_checkpoint_loads(tmp_path):
+    """Multi-file HF checkpoints (model-0000N-of-0000M.safetensors +
+    index) load the same as single-file ones."""
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=96, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True,
+                       max_shard_size="50KB")
+    import glob as _g
+    shards = _g.glob(str(tmp_path / "model-*.safetensors"))
+    assert len(shards) >= 2, "expected a sharded save"
+    _compare_logits(hf, str(tmp_path))
