@@ -604,8 +604,10 @@ class LlamaModel:
 
     @torch.no_grad()
     def generate(self, prompt: torch.Tensor, max_new_tokens: int = 16,
-                 cache: Optional[KVCache] = None) -> torch.Tensor:
-        """Greedy decode. prompt [B, T] -> [B, T + max_new_tokens]."""
+                 cache: Optional[KVCache] = None,
+                 eos_id: Optional[int] = None) -> torch.Tensor:
+        """Greedy decode. prompt [B, T] -> [B, T + up to max_new_tokens]
+        (stops early once every sequence emitted ``eos_id``)."""
         B, T = prompt.shape
         own_cache = cache is None
         if own_cache:
@@ -615,9 +617,14 @@ class LlamaModel:
         logits = self.forward(prompt, cache, 0)
         out = prompt
         pos = T
+        done = torch.zeros(B, dtype=torch.bool, device=prompt.device)
         for _ in range(max_new_tokens):
             nxt = logits[:, -1].argmax(-1, keepdim=True)
             out = torch.cat([out, nxt], dim=1)
+            if eos_id is not None:
+                done |= nxt[:, 0] == eos_id
+                if bool(done.all()):
+                    break
             logits = self.forward(nxt, cache, pos)
             pos += 1
         if own_cache:

@@ -236,16 +236,19 @@ class ActuationEngine:
     # -- serving -------------------------------------------------------------
 
     @torch.no_grad()
-    def generate(self, tokens: torch.Tensor, max_new_tokens: int = 16
-                 ) -> torch.Tensor:
+    def generate(self, tokens: torch.Tensor, max_new_tokens: int = 16,
+                 eos_id: "int | None" = None) -> torch.Tensor:
         if self.state != self.AWAKE:
             raise RuntimeError("engine is sleeping")
         tokens = tokens.to(self.device)
-        dec = self._graph_decoder(tokens.shape[0],
-                                  tokens.shape[1] + max_new_tokens + 2)
-        if dec is not None:
-            return dec.generate(tokens, max_new_tokens)
-        return self.model.generate(tokens, max_new_tokens)
+        if eos_id is None:
+            dec = self._graph_decoder(tokens.shape[0],
+                                      tokens.shape[1] + max_new_tokens + 2)
+            if dec is not None:
+                return dec.generate(tokens, max_new_tokens)
+        # early EOS stop is data-dependent: eager decode, not a fixed
+        # graph replay
+        return self.model.generate(tokens, max_new_tokens, eos_id=eos_id)
 
     def _graph_decoder(self, batch: int, need_seq: int):
         """Lazily built hipGraph decoder for batch-1 serving (2.5x on
@@ -283,7 +286,12 @@ class ActuationEngine:
         if tok is not None:
             ids = tok.encode(prompt).ids or [1]
             toks = torch.tensor([ids], dtype=torch.long, device=self.device)
-            out = self.generate(toks, max_new_tokens)[0, len(ids):]
+            out = self.generate(toks, max_new_tokens,
+                                eos_id=getattr(self, "eos_id", None)
+                                )[0, len(ids):]
+            if getattr(self, "eos_id", None) is not None:
+                keep = (out == self.eos_id).cumsum(0) == 0
+                out = out[keep]
             return tok.decode([int(t) for t in out.tolist()])
         ids = [b % self.cfg.vocab_size for b in prompt.encode("utf-8")] or [1]
         toks = torch.tensor([ids], dtype=torch.long, device=self.device)

@@ -98,6 +98,21 @@ def _worker_entry(rank: int, world: int, master_port: int,
     tp.run_worker_loop(ctx, eng)
 
 
+def load_eos_id(checkpoint_dir: str):
+    """eos_token_id from the HF config.json (first of a list), so text
+    completions stop at end-of-sequence like the upstream model."""
+    import json as _json
+    path = os.path.join(checkpoint_dir, "config.json")
+    try:
+        with open(path) as f:
+            eos = _json.load(f).get("eos_token_id")
+        if isinstance(eos, list):
+            eos = eos[0] if eos else None
+        return int(eos) if eos is not None else None
+    except (OSError, ValueError, TypeError):
+        return None
+
+
 def load_tokenizer(checkpoint_dir: str):
     """HF checkpoint dirs carry tokenizer.json; attach the real
     tokenizer so /v1/completions speaks text, not byte ids. Returns
@@ -160,6 +175,7 @@ class ServingRuntime:
             print(f"[serve] checkpoint {checkpoint} loaded in {t_load:.2f}s",
                   flush=True)
             engine.tokenizer = load_tokenizer(checkpoint)
+            engine.eos_id = load_eos_id(checkpoint)
         self.rt = tp.TPRuntime(ctx, engine) if world > 1 else engine
         self.model_name = args.served_model_name or args.model
         print(f"[serve] engine up in {engine.create_seconds:.2f}s "

@@ -439,3 +439,19 @@ def test_sharded_hf_checkpoint_loads(tmp_path):
     shards = _g.glob(str(tmp_path / "model-*.safetensors"))
     assert len(shards) >= 2, "expected a sharded save"
     _compare_logits(hf, str(tmp_path))
+
+
+def test_generate_stops_at_eos(tmp_path):
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=64, hidden_size=32, intermediate_size=48,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=1,
+        max_position_embeddings=32, tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    _, ours = _load_ours(str(tmp_path))
+    torch.manual_seed(2)
+    prompt = torch.randint(0, 64, (1, 4))
+    free_run = ours.generate(prompt, max_new_tokens=8)
+    first = int(free_run[0, 4])
+    stopped = ours.generate(prompt, max_new_tokens=8, eos_id=first)
+    assert stopped.shape[1] == 5 and int(stopped[0, -1]) == first
