@@ -295,6 +295,25 @@ class LlamaConfig:
                    self.param_specs(tp_rank, tp_size))
 
 
+def _pick_next(logits: torch.Tensor, temperature: float,
+               top_p: float) -> torch.Tensor:
+    """[B, V] -> [B, 1] next-token ids; temperature 0 = greedy, else
+    nucleus sampling."""
+    if temperature <= 0.0:
+        return logits.argmax(-1, keepdim=True)
+    probs = torch.softmax(logits.float() / temperature, dim=-1)
+    if top_p < 1.0:
+        sp, si = probs.sort(dim=-1, descending=True)
+        cum = sp.cumsum(-1)
+        # keep the smallest prefix whose mass reaches top_p
+        cut = cum - sp >= top_p
+        sp = sp.masked_fill(cut, 0.0)
+        sp = sp / sp.sum(-1, keepdim=True)
+        pick = torch.multinomial(sp, 1)
+        return si.gather(-1, pick)
+    return torch.multinomial(probs, 1)
+
+
 def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
     xf = x.float()
     xf = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
@@ -605,9 +624,13 @@ class LlamaModel:
     @torch.no_grad()
     def generate(self, prompt: torch.Tensor, max_new_tokens: int = 16,
                  cache: Optional[KVCache] = None,
-                 eos_id: Optional[int] = None) -> torch.Tensor:
-        """Greedy decode. prompt [B, T] -> [B, T + up to max_new_tokens]
-        (stops early once every sequence emitted ``eos_id``)."""
+                 eos_id: Optional[int] = None,
+                 temperature: float = 0.0,
+                 top_p: float = 1.0) -> torch.Tensor:
+        """Decode. prompt [B, T] -> [B, T + up to max_new_tokens];
+        greedy at temperature 0, else nucleus sampling (top_p) from
+        softmax(logits / temperature). Stops early once every sequence
+        emitted ``eos_id``."""
         B, T = prompt.shape
         own_cache = cache is None
         if own_cache:
@@ -619,7 +642,7 @@ class LlamaModel:
         pos = T
         done = torch.zeros(B, dtype=torch.bool, device=prompt.device)
         for _ in range(max_new_tokens):
-            nxt = logits[:, -1].argmax(-1, keepdim=True)
+            nxt = _pick_next(logits[:, -1], temperature, top_p)
             out = torch.cat([out, nxt], dim=1)
             if eos_id is not None:
                 done |= nxt[:, 0] == eos_id

@@ -237,18 +237,20 @@ class ActuationEngine:
 
     @torch.no_grad()
     def generate(self, tokens: torch.Tensor, max_new_tokens: int = 16,
-                 eos_id: "int | None" = None) -> torch.Tensor:
+                 eos_id: "int | None" = None, temperature: float = 0.0,
+                 top_p: float = 1.0) -> torch.Tensor:
         if self.state != self.AWAKE:
             raise RuntimeError("engine is sleeping")
         tokens = tokens.to(self.device)
-        if eos_id is None:
+        if eos_id is None and temperature <= 0.0:
             dec = self._graph_decoder(tokens.shape[0],
                                       tokens.shape[1] + max_new_tokens + 2)
             if dec is not None:
                 return dec.generate(tokens, max_new_tokens)
-        # early EOS stop is data-dependent: eager decode, not a fixed
-        # graph replay
-        return self.model.generate(tokens, max_new_tokens, eos_id=eos_id)
+        # EOS stop / sampling are data-dependent: eager decode, not a
+        # fixed graph replay
+        return self.model.generate(tokens, max_new_tokens, eos_id=eos_id,
+                                   temperature=temperature, top_p=top_p)
 
     def _graph_decoder(self, batch: int, need_seq: int):
         """Lazily built hipGraph decoder for batch-1 serving (2.5x on

@@ -234,6 +234,8 @@ def create_app(runtime: ServingRuntime) -> FastAPI:
                 return JSONResponse(
                     {"error": "model is sleeping"}, status_code=409)
             t0 = time.time()
+            temperature = float(body.get("temperature", 0.0))
+            top_p = float(body.get("top_p", 1.0))
             if isinstance(prompt, list):
                 # OpenAI-compat: prompt may be token ids; respond with
                 # the generated ids so clients control detokenization
@@ -242,7 +244,10 @@ def create_app(runtime: ServingRuntime) -> FastAPI:
                 ids = [int(t) % eng.cfg.vocab_size for t in prompt] or [1]
                 toks = _torch.tensor([ids], dtype=_torch.long,
                                      device=eng.device)
-                out = rt.generate(toks, max_tokens)[0, len(ids):]
+                kw = ({"temperature": temperature, "top_p": top_p}
+                      if temperature > 0 and not hasattr(rt, "ctx")
+                      else {})  # TP ranks must agree: greedy only
+                out = rt.generate(toks, max_tokens, **kw)[0, len(ids):]
                 return {
                     "id": f"cmpl-{int(t0*1e6)}",
                     "object": "text_completion",

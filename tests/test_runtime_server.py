@@ -133,3 +133,24 @@ def test_completions_accepts_token_ids(client):
     choice = r.json()["choices"][0]
     assert len(choice["token_ids"]) == 4
     assert r.json()["usage"]["prompt_tokens"] == 3
+
+
+def test_completions_sampling_params(client):
+    """temperature/top_p: sampled ids differ run to run in general but
+    temperature=0 stays greedy-deterministic; top_p=tiny collapses to
+    near-greedy. Determinism of the greedy path is asserted."""
+    import torch
+    torch.manual_seed(0)
+    r1 = client.post("/v1/completions",
+                     json={"prompt": [1, 2, 3], "max_tokens": 4,
+                           "temperature": 0.0})
+    r2 = client.post("/v1/completions",
+                     json={"prompt": [1, 2, 3], "max_tokens": 4,
+                           "temperature": 0.0})
+    assert r1.json()["choices"][0]["token_ids"] == \
+        r2.json()["choices"][0]["token_ids"]
+    r3 = client.post("/v1/completions",
+                     json={"prompt": [1, 2, 3], "max_tokens": 4,
+                           "temperature": 0.8, "top_p": 0.9})
+    assert r3.status_code == 200
+    assert len(r3.json()["choices"][0]["token_ids"]) == 4
