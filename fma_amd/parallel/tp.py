@@ -120,7 +120,9 @@ class TPRuntime:
         })
         return self.engine.generate(tokens, max_new_tokens)
 
-    def generate_text(self, prompt: str, max_new_tokens: int = 16) -> str:
+    def generate_text(self, prompt: str, max_new_tokens: int = 16,
+                      temperature: float = 0.0, top_p: float = 1.0) -> str:
+        # TP ranks must agree on every sampled token; greedy only
         tok = getattr(self.engine, "tokenizer", None)
         if tok is not None:
             ids = tok.encode(prompt).ids or [1]

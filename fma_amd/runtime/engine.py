@@ -279,7 +279,8 @@ class ActuationEngine:
             self._decoder = dec
         return dec
 
-    def generate_text(self, prompt: str, max_new_tokens: int = 16) -> str:
+    def generate_text(self, prompt: str, max_new_tokens: int = 16,
+                      temperature: float = 0.0, top_p: float = 1.0) -> str:
         """Text round trip. With a real tokenizer attached (HF checkpoint
         dirs carry tokenizer.json — see runtime/server.py), prompts
         encode/decode through it; otherwise UTF-8 bytes are the token
@@ -289,7 +290,8 @@ class ActuationEngine:
             ids = tok.encode(prompt).ids or [1]
             toks = torch.tensor([ids], dtype=torch.long, device=self.device)
             out = self.generate(toks, max_new_tokens,
-                                eos_id=getattr(self, "eos_id", None)
+                                eos_id=getattr(self, "eos_id", None),
+                                temperature=temperature, top_p=top_p
                                 )[0, len(ids):]
             if getattr(self, "eos_id", None) is not None:
                 keep = (out == self.eos_id).cumsum(0) == 0

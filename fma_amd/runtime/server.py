@@ -259,7 +259,10 @@ def create_app(runtime: ServingRuntime) -> FastAPI:
                     "usage": {"prompt_tokens": len(ids),
                               "completion_tokens": max_tokens},
                 }
-            text = rt.generate_text(prompt, max_tokens)
+            text = rt.generate_text(prompt, max_tokens,
+                                    temperature=temperature, top_p=top_p) \
+                if not hasattr(rt, "ctx") else \
+                rt.generate_text(prompt, max_tokens)
         return {
             "id": f"cmpl-{int(t0*1e6)}",
             "object": "text_completion",
