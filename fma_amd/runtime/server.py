@@ -276,6 +276,37 @@ def create_app(runtime: ServingRuntime) -> FastAPI:
     def stats():
         return rt.stats()
 
+    @app.get("/metrics")
+    def metrics():
+        """Prometheus exposition (vLLM serves /metrics too): engine
+        counters + last actuation timings as gauges, served without a
+        global registry so multiple instances coexist in one process."""
+        st = rt.stats()
+        lines = []
+
+        def g(name, val, help_):
+            if val is None:
+                return
+            lines.append(f"# HELP {name} {help_}")
+            lines.append(f"# TYPE {name} gauge")
+            lines.append(f"{name} {float(val)}")
+
+        g("fma_engine_sleep_count", st.get("sleep_count"),
+          "sleep() calls since instance start")
+        g("fma_engine_wake_count", st.get("wake_count"),
+          "wake_up() calls since instance start")
+        g("fma_engine_last_sleep_seconds", st.get("last_sleep_seconds"),
+          "duration of the most recent sleep")
+        g("fma_engine_last_wake_seconds", st.get("last_wake_seconds"),
+          "duration of the most recent wake_up")
+        g("fma_engine_param_bytes", st.get("param_bytes"),
+          "model parameter bytes resident when awake")
+        g("fma_engine_is_sleeping", 1.0 if rt.is_sleeping() else 0.0,
+          "1 while weights are offloaded to host DRAM")
+        from fastapi.responses import PlainTextResponse
+        return PlainTextResponse("\n".join(lines) + "\n",
+                                 media_type="text/plain; version=0.0.4")
+
     return app
 
 

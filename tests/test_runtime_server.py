@@ -154,3 +154,16 @@ def test_completions_sampling_params(client):
                            "temperature": 0.8, "top_p": 0.9})
     assert r3.status_code == 200
     assert len(r3.json()["choices"][0]["token_ids"]) == 4
+
+
+def test_metrics_exposition(client):
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    body = r.text
+    assert "fma_engine_is_sleeping 0.0" in body
+    assert "fma_engine_param_bytes" in body
+    client.post("/sleep?level=1")
+    assert "fma_engine_is_sleeping 1.0" in client.get("/metrics").text
+    client.post("/wake_up")
+    body = client.get("/metrics").text
+    assert "fma_engine_last_wake_seconds" in body
