@@ -272,6 +272,39 @@ def create_app(runtime: ServingRuntime) -> FastAPI:
             "usage": {"completion_tokens": max_tokens},
         }
 
+    @app.post("/v1/chat/completions")
+    def chat_completions(body: dict):
+        """Minimal chat surface (vLLM serves it; routers probe it):
+        messages are rendered with the reversible generic template
+        "<|role|>content" per turn + a final assistant header. Models
+        with their own chat template should be driven through
+        /v1/completions with pre-templated text."""
+        msgs = body.get("messages") or []
+        max_tokens = int(body.get("max_tokens", 16))
+        temperature = float(body.get("temperature", 0.0))
+        top_p = float(body.get("top_p", 1.0))
+        prompt = "".join(
+            f"<|{m.get('role', 'user')}|>{m.get('content', '')}"
+            for m in msgs) + "<|assistant|>"
+        with runtime.lock:
+            if rt.is_sleeping():
+                return JSONResponse(
+                    {"error": "model is sleeping"}, status_code=409)
+            t0 = time.time()
+            kw = ({"temperature": temperature, "top_p": top_p}
+                  if not hasattr(rt, "ctx") else {})
+            text = rt.generate_text(prompt, max_tokens, **kw)
+        return {
+            "id": f"chatcmpl-{int(t0*1e6)}",
+            "object": "chat.completion",
+            "model": runtime.model_name,
+            "choices": [{"index": 0,
+                         "message": {"role": "assistant",
+                                     "content": text},
+                         "finish_reason": "length"}],
+            "usage": {"completion_tokens": max_tokens},
+        }
+
     @app.get("/stats")
     def stats():
         return rt.stats()

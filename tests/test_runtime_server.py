@@ -167,3 +167,13 @@ def test_metrics_exposition(client):
     client.post("/wake_up")
     body = client.get("/metrics").text
     assert "fma_engine_last_wake_seconds" in body
+
+
+def test_chat_completions(client):
+    r = client.post("/v1/chat/completions", json={
+        "messages": [{"role": "system", "content": "be brief"},
+                     {"role": "user", "content": "hi"}],
+        "max_tokens": 4})
+    assert r.status_code == 200
+    msg = r.json()["choices"][0]["message"]
+    assert msg["role"] == "assistant" and isinstance(msg["content"], str)
