@@ -391,3 +391,22 @@ def test_plan_layout_properties_fuzz():
             for _, shape, dt in specs) - len(specs) * 0  # exact when slab=0
         if slab == 0:
             assert len(slabs) == 1
+
+
+def test_batched_generate_matches_single():
+    """B=2 greedy decode equals the two B=1 decodes (equal-length
+    prompts, no padding): the cache/rope/attention paths are
+    batch-correct, not just batch-1-tested."""
+    cfg = LlamaConfig(name="b2", vocab_size=128, hidden_size=64,
+                      intermediate_size=96, num_layers=2, num_heads=4,
+                      num_kv_heads=2, max_seq_len=64,
+                      dtype=torch.float32)
+    params = {n: torch.zeros(s, dtype=d) for n, s, d in cfg.param_specs()}
+    model = LlamaModel(cfg, params, torch.device("cpu"))
+    model.init_weights(5)
+    torch.manual_seed(4)
+    prompts = torch.randint(0, 128, (2, 6))
+    batched = model.generate(prompts, max_new_tokens=6)
+    for b in range(2):
+        single = model.generate(prompts[b:b + 1], max_new_tokens=6)
+        assert torch.equal(batched[b:b + 1], single), b
