@@ -64,6 +64,14 @@ class LlamaConfig:
     #: num_experts per token; each expert is a llama-shaped gated MLP.
     num_experts: int = 0
     num_experts_per_tok: int = 2
+    #: expert FFN width when it differs from intermediate_size
+    #: (qwen3-moe's moe_intermediate_size); None = intermediate_size
+    moe_intermediate_size: "int | None" = None
+    #: router weight normalization: True = softmax over the selected
+    #: top-k (mixtral; equals qwen3-moe norm_topk_prob=true), False =
+    #: softmax over ALL experts, top-k taken unrenormalized
+    #: (qwen3-moe norm_topk_prob=false)
+    moe_norm_topk: bool = True
     #: EP: partition WHOLE experts across ranks (attention stays
     #: Megatron-TP); non-owners contribute zero and the layer's single
     #: all-reduce sums the routed outputs. False = Megatron-shard every
@@ -268,10 +276,12 @@ class LlamaConfig:
                 # preserved in the names so checkpoints map naturally).
                 specs.append((p + "router.weight",
                               (self.num_experts, h), d))
+                moe_i = self.moe_intermediate_size or self.intermediate_size
                 for e in self.local_experts(tp_rank, tp_size):
                     ep = p + f"experts.{e}."
-                    ei = self.intermediate_size if (
-                        self.expert_parallel and tp_size > 1) else i_local
+                    ei = moe_i if (
+                        self.expert_parallel and tp_size > 1) \
+                        else moe_i // tp_size
                     specs += [
                         (ep + "w_gate.weight", (ei, h), d),
                         (ep + "w_up.weight", (ei, h), d),
@@ -452,8 +462,16 @@ class LlamaModel:
                                     cfg.norm_eps)
         B, T, H = h.shape
         logits = fast_linear(h, P[p + "router.weight"]).float()
-        topw, topi = torch.topk(logits, cfg.num_experts_per_tok, dim=-1)
-        topw = torch.softmax(topw, dim=-1)
+        if cfg.moe_norm_topk:
+            topw, topi = torch.topk(logits, cfg.num_experts_per_tok,
+                                    dim=-1)
+            topw = torch.softmax(topw, dim=-1)
+        else:
+            # qwen3-moe norm_topk_prob=false: global softmax, top-k
+            # probabilities used as-is (they do not sum to 1)
+            probs = torch.softmax(logits, dim=-1)
+            topw, topi = torch.topk(probs, cfg.num_experts_per_tok,
+                                    dim=-1)
         mine = set(cfg.local_experts(self.tp_rank, self.tp_size))
         if B * T == 1:
             out = torch.zeros_like(x)

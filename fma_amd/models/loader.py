@@ -65,16 +65,25 @@ def config_from_dir(path: str) -> Optional[LlamaConfig]:
         # transformers configs carry model_type instead of qkv_bias;
         # qwen2's attention projections are the biased ones
         qkv_bias=d.get("qkv_bias", d.get("model_type") == "qwen2"),
-        qk_norm=d.get("qk_norm", d.get("model_type") == "qwen3"),
+        qk_norm=d.get("qk_norm",
+                      str(d.get("model_type", "")).startswith("qwen3")),
         hidden_act=("gelu_tanh" if d.get(
             "hidden_act", d.get("hidden_activation", "")).startswith("gelu")
             else d.get("hidden_act", "silu")),
         embed_scale=d.get("embed_scale", d.get("model_type") == "gemma"),
         num_experts=d.get("num_experts", d.get("num_local_experts", 0)),
         num_experts_per_tok=d.get("num_experts_per_tok", 2),
+        moe_intermediate_size=d.get("moe_intermediate_size"),
+        moe_norm_topk=d.get("moe_norm_topk",
+                            d.get("norm_topk_prob", True)),
         head_dim_override=d.get("head_dim"),
         rope_scaling=scaling,
     )
+    if d.get("model_type") == "qwen3_moe" and (
+            d.get("decoder_sparse_step", 1) != 1 or d.get("mlp_only_layers")):
+        raise ValueError("qwen3-moe with interleaved dense layers "
+                         "(decoder_sparse_step/mlp_only_layers) is not "
+                         "supported")
     if d.get("model_type") == "gemma":
         # load-time marker: fold gemma's (1+w) rmsnorm offset into the
         # gains during hf_convert (runtime norm stays standard)
@@ -133,6 +142,9 @@ def save_params(params: Dict[str, torch.Tensor], path: str,
                 "embed_scale": cfg.embed_scale,
                 "num_experts": cfg.num_experts,
                 "num_experts_per_tok": cfg.num_experts_per_tok,
+                **({"moe_intermediate_size": cfg.moe_intermediate_size}
+                   if cfg.moe_intermediate_size else {}),
+                "moe_norm_topk": cfg.moe_norm_topk,
                 **({"head_dim": cfg.head_dim_override}
                    if cfg.head_dim_override else {}),
                 **({"rope_scaling": cfg.rope_scaling}
@@ -155,6 +167,8 @@ _HF_ATTN_RE = re.compile(r"self_attn\.(\w+_proj)\.(weight|bias)$")
 _HF_MLP_RE = re.compile(r"mlp\.(\w+_proj)\.weight$")
 _HF_EXP_RE = re.compile(
     r"block_sparse_moe\.experts\.(\d+)\.(w[123])\.weight$")
+_HF_EXP2_RE = re.compile(
+    r"mlp\.experts\.(\d+)\.(\w+_proj)\.weight$")
 
 
 def map_hf_name(name: str) -> Optional[str]:
@@ -180,6 +194,8 @@ def map_hf_name(name: str) -> Optional[str]:
         return p + "mlp_norm.weight"
     if leaf == "block_sparse_moe.gate.weight":
         return p + "router.weight"
+    if leaf == "mlp.gate.weight":
+        return p + "router.weight"  # qwen3-moe router
     a = _HF_ATTN_RE.match(leaf)
     if a and a.group(1) in _HF_ATTN:
         return p + _HF_ATTN[a.group(1)] + "." + a.group(2)
@@ -189,6 +205,10 @@ def map_hf_name(name: str) -> Optional[str]:
     e = _HF_EXP_RE.match(leaf)
     if e:
         return p + f"experts.{e.group(1)}." + _HF_EXPERT[e.group(2)] + \
+            ".weight"
+    e = _HF_EXP2_RE.match(leaf)
+    if e:  # qwen3-moe expert naming
+        return p + f"experts.{e.group(1)}." + _HF_MLP[e.group(2)] + \
             ".weight"
     return None
 

@@ -455,3 +455,23 @@ def test_generate_stops_at_eos(tmp_path):
     first = int(free_run[0, 4])
     stopped = ours.generate(prompt, max_new_tokens=8, eos_id=first)
     assert stopped.shape[1] == 5 and int(stopped[0, -1]) == first
+
+
+@pytest.mark.parametrize("norm_topk", [False, True])
+def test_qwen3_moe_hf_checkpoint_logits_match(tmp_path, norm_topk):
+    """Qwen3-MoE: qk norms + sparse MoE with its own expert width
+    (moe_intermediate_size) and qwen3-style router naming; both
+    norm_topk_prob semantics (global-softmax top-k vs renormalized)."""
+    from transformers import Qwen3MoeConfig, Qwen3MoeForCausalLM
+    hf = Qwen3MoeForCausalLM(Qwen3MoeConfig(
+        vocab_size=96, hidden_size=64, intermediate_size=96,
+        moe_intermediate_size=48, num_hidden_layers=2,
+        num_attention_heads=4, num_key_value_heads=2, head_dim=16,
+        num_experts=4, num_experts_per_tok=2, norm_topk_prob=norm_topk,
+        max_position_embeddings=64, tie_word_embeddings=True)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    cfg = loader.config_from_dir(str(tmp_path))
+    assert cfg.qk_norm and cfg.num_experts == 4
+    assert cfg.moe_intermediate_size == 48
+    assert cfg.moe_norm_topk is norm_topk
+    _compare_logits(hf, str(tmp_path))
