@@ -224,6 +224,34 @@ def _unrotate_half(t: torch.Tensor, heads: int, hd: int) -> torch.Tensor:
     return t.view(heads, 2, hd // 2).permute(0, 2, 1).reshape(t.shape)
 
 
+def hf_convert_multi(name: str, tensor: torch.Tensor, cfg: "LlamaConfig"):
+    """One HF tensor -> list of (our_name, tensor). Usually one pair;
+    phi3-style FUSED projections split into several; [] for tensors with
+    no counterpart (rotary buffers)."""
+    m = _HF_LAYER_RE.match(name)
+    if m:
+        p, leaf = f"layers.{m.group(1)}.", m.group(2)
+        if leaf == "self_attn.qkv_proj.weight":  # phi3 fused qkv
+            qn = cfg.num_heads * cfg.head_dim
+            kn = cfg.num_kv_heads * cfg.head_dim
+            q = tensor[:qn]
+            k = tensor[qn:qn + kn]
+            v = tensor[qn + kn:qn + 2 * kn]
+            return [
+                (p + "wq.weight",
+                 _unrotate_half(q, cfg.num_heads, cfg.head_dim)),
+                (p + "wk.weight",
+                 _unrotate_half(k, cfg.num_kv_heads, cfg.head_dim)),
+                (p + "wv.weight", v),
+            ]
+        if leaf == "mlp.gate_up_proj.weight":  # phi3 fused gated MLP
+            i = tensor.shape[0] // 2
+            return [(p + "w_gate.weight", tensor[:i]),
+                    (p + "w_up.weight", tensor[i:])]
+    one = hf_convert(name, tensor, cfg)
+    return [one] if one is not None else []
+
+
 def hf_convert(name: str, tensor: torch.Tensor, cfg: "LlamaConfig"):
     """(our_name, converted_tensor) for one HF tensor; None when the
     tensor has no counterpart (e.g. rotary_emb.inv_freq buffers)."""
@@ -301,6 +329,23 @@ def shard_slice(name: str, tensor: torch.Tensor, tp_rank: int,
     return tensor
 
 
+def _iter_converted(path, params, cfg):
+    """iter_safetensors with transparent HF conversion: HF names map
+    (possibly splitting fused tensors); a sentinel ("__HF__", None)
+    marks that HF format was seen."""
+    for name, tensor in iter_safetensors(path):
+        if name.startswith("model."):
+            if cfg is None:
+                raise ValueError(
+                    "HuggingFace checkpoint: pass cfg= for name mapping "
+                    "and rope conversion")
+            yield "__HF__", None
+            for pair in hf_convert_multi(name, tensor, cfg):
+                yield pair
+            continue
+        yield name, tensor
+
+
 def load_into_params(path: str, params: Dict[str, torch.Tensor],
                      strict: bool = True, tp_rank: int = 0,
                      tp_size: int = 1, skip=None, cfg=None) -> int:
@@ -316,17 +361,10 @@ def load_into_params(path: str, params: Dict[str, torch.Tensor],
     loaded = 0
     seen = set()
     hf = False
-    for name, tensor in iter_safetensors(path):
-        if name.startswith("model."):
+    for name, tensor in _iter_converted(path, params, cfg):
+        if name == "__HF__":
             hf = True
-            if cfg is None:
-                raise ValueError(
-                    "HuggingFace checkpoint: pass cfg= for name mapping "
-                    "and rope conversion")
-            conv = hf_convert(name, tensor, cfg)
-            if conv is None:
-                continue  # inv_freq buffers and friends
-            name, tensor = conv
+            continue
         if name not in params:
             if skip is not None and skip(name):
                 continue  # e.g. expert-parallel: another rank's expert

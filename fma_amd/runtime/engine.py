@@ -185,10 +185,12 @@ class ActuationEngine:
             name, tensor = item
             if name.startswith("model."):  # HuggingFace checkpoint
                 hf_seen[0] = True
-                conv = loader.hf_convert(name, tensor, self.cfg)
-                if conv is None:
+                pairs = loader.hf_convert_multi(name, tensor, self.cfg)
+                if not pairs:
                     return None  # inv_freq buffers and friends
-                name, tensor = conv
+                if len(pairs) > 1:  # fused tensor split (phi3)
+                    return [stage(pr) for pr in pairs]
+                name, tensor = pairs[0]
             if name not in self.layout:
                 if self.cfg.num_experts and self.cfg.expert_parallel \
                         and ".experts." in name:
@@ -211,7 +213,10 @@ class ActuationEngine:
         seen = set()
         with concurrent.futures.ThreadPoolExecutor(cpu_threads) as ex:
             for name in ex.map(stage, loader.iter_safetensors(path)):
-                seen.add(name)
+                if isinstance(name, list):
+                    seen.update(n for n in name if n)
+                else:
+                    seen.add(name)
         missing = set(self.layout) - seen
         if hf_seen[0] and missing == {"lm_head.weight"} \
                 and "embed.weight" in seen:

@@ -475,3 +475,28 @@ def test_qwen3_moe_hf_checkpoint_logits_match(tmp_path, norm_topk):
     assert cfg.moe_intermediate_size == 48
     assert cfg.moe_norm_topk is norm_topk
     _compare_logits(hf, str(tmp_path))
+
+
+def test_phi3_hf_checkpoint_logits_match(tmp_path):
+    """Phi-3: FUSED qkv_proj / gate_up_proj tensors split at load (with
+    the rope row-permutation applied to the q/k slices)."""
+    from transformers import Phi3Config, Phi3ForCausalLM
+    hf = Phi3ForCausalLM(Phi3Config(
+        vocab_size=96, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, eos_token_id=2, pad_token_id=0,
+        bos_token_id=1, tie_word_embeddings=False)).eval()
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    _compare_logits(hf, str(tmp_path))
+    # arena fast path splits the fused tensors too
+    from fma_amd.runtime.engine import ActuationEngine
+    cfg = loader.config_from_dir(str(tmp_path))
+    eng = ActuationEngine(cfg, init_weights=False)
+    eng.load_checkpoint(str(tmp_path))
+    torch.manual_seed(5)
+    prompt = torch.randint(0, 96, (1, 6))
+    with torch.no_grad():
+        ref = hf.generate(prompt, max_new_tokens=4, do_sample=False,
+                          use_cache=True)
+    got = eng.generate(prompt, max_new_tokens=4)
+    assert torch.equal(got[0, :8], ref[0, :8])
