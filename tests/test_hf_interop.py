@@ -500,3 +500,24 @@ def test_phi3_hf_checkpoint_logits_match(tmp_path):
                           use_cache=True)
     got = eng.generate(prompt, max_new_tokens=4)
     assert torch.equal(got[0, :8], ref[0, :8])
+
+
+def test_hf_check_tool(tmp_path, capsys):
+    """tools/hf_check.py: the operator's will-this-serve probe."""
+    import sys
+    from transformers import LlamaConfig as HFConfig, LlamaForCausalLM
+    hf = LlamaForCausalLM(HFConfig(
+        vocab_size=64, hidden_size=32, intermediate_size=48,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=1,
+        max_position_embeddings=32, tie_word_embeddings=True))
+    hf.save_pretrained(tmp_path, safe_serialization=True)
+    sys.path.insert(0, "tools")
+    import hf_check
+    old = sys.argv
+    sys.argv = ["hf_check", str(tmp_path), "--load"]
+    try:
+        assert hf_check.main() == 0
+    finally:
+        sys.argv = old
+    out = capsys.readouterr().out
+    assert "load OK" in out and "finite=True" in out
