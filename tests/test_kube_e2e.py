@@ -33,7 +33,7 @@ from tests.test_e2e_single_node import (ISC_PORT, launcher_pod,
                                         requester_ready, wait_for)
 from tests.test_live_servers import ServerThread, free_port
 
-pytestmark = pytest.mark.timeout(300)
+pytestmark = pytest.mark.timeout(700)
 
 
 @pytest.fixture()
@@ -78,12 +78,12 @@ def test_kube_full_actuation_hot_start_and_vap(kube_cluster):
     store, agent = kube_cluster["store"], kube_cluster["agent"]
     mk_isc_lc_lpp(store)
 
-    lp = wait_for(lambda: launcher_pod(store), 60, desc="launcher pod")
+    lp = wait_for(lambda: launcher_pod(store), 100, desc="launcher pod")
     wait_for(lambda: ob.pod_is_ready(store.get("Pod", ob.name_of(lp))),
              90, desc="launcher Ready")
 
     mk_requester(store, "kreq1")
-    wait_for(lambda: requester_ready(store, agent, "kreq1"), 120,
+    wait_for(lambda: requester_ready(store, agent, "kreq1"), 200,
              desc="requester /ready 200 over the kube protocol")
 
     lp = store.get("Pod", ob.name_of(lp))
@@ -112,7 +112,7 @@ def test_kube_full_actuation_hot_start_and_vap(kube_cluster):
     # delete requester -> sleep + unbind; launcher survives with the
     # sleeping instance
     store.delete("Pod", "kreq1")
-    wait_for(lambda: store.try_get("Pod", "kreq1") is None, 90,
+    wait_for(lambda: store.try_get("Pod", "kreq1") is None, 150,
              desc="requester gone")
     lp = store.get("Pod", ob.name_of(lp))
     assert C.REQUESTER_ANNOTATION not in ob.annotations_of(lp)
@@ -120,7 +120,7 @@ def test_kube_full_actuation_hot_start_and_vap(kube_cluster):
 
     # hot start: same-ISC requester rebinds the sleeping instance
     mk_requester(store, "kreq2")
-    wait_for(lambda: requester_ready(store, agent, "kreq2"), 90,
+    wait_for(lambda: requester_ready(store, agent, "kreq2"), 150,
              desc="kreq2 hot start")
     lp = store.get("Pod", ob.name_of(lp))
     assert ob.annotations_of(lp)[C.INSTANCE_ID_ANNOTATION] == iid
@@ -137,11 +137,11 @@ def test_kube_controller_restart_recovery(kube_cluster):
     store, agent = kube_cluster["store"], kube_cluster["agent"]
     base = kube_cluster["base"]
     mk_isc_lc_lpp(store)
-    lp = wait_for(lambda: launcher_pod(store), 90, desc="launcher pod")
+    lp = wait_for(lambda: launcher_pod(store), 150, desc="launcher pod")
     wait_for(lambda: ob.pod_is_ready(store.get("Pod", ob.name_of(lp))),
              120, desc="launcher Ready")
     mk_requester(store, "rreq1")
-    wait_for(lambda: requester_ready(store, agent, "rreq1"), 180,
+    wait_for(lambda: requester_ready(store, agent, "rreq1"), 300,
              desc="requester ready")
 
     kube_cluster["ctl"].stop()
@@ -163,7 +163,7 @@ def test_kube_controller_restart_recovery(kube_cluster):
             return f"requester gone via recovered controller; pods={pods}"
 
         try:
-            wait_for(_gone, 150, desc="requester gone")
+            wait_for(_gone, 250, desc="requester gone")
         except AssertionError:
             raise AssertionError(f"timed out: {_state()}") from None
         lp2 = store.get("Pod", ob.name_of(lp))
@@ -231,11 +231,11 @@ def test_kube_stopped_instance_recovers(kube_cluster):
 
     store, agent = kube_cluster["store"], kube_cluster["agent"]
     mk_isc_lc_lpp(store)
-    lp = wait_for(lambda: launcher_pod(store), 60, desc="launcher pod")
+    lp = wait_for(lambda: launcher_pod(store), 100, desc="launcher pod")
     wait_for(lambda: ob.pod_is_ready(store.get("Pod", ob.name_of(lp))),
              90, desc="launcher Ready")
     mk_requester(store, "sreq1")
-    wait_for(lambda: requester_ready(store, agent, "sreq1"), 120,
+    wait_for(lambda: requester_ready(store, agent, "sreq1"), 200,
              desc="requester ready")
 
     # SIGKILL the forked serving-instance process: it is the launcher
@@ -246,13 +246,13 @@ def test_kube_stopped_instance_recovers(kube_cluster):
 
     lname = ob.name_of(lp)
     launcher_proc = psutil.Process(agent.pods[lname].proc.pid)
-    kids = wait_for(lambda: launcher_proc.children() or None, 30,
+    kids = wait_for(lambda: launcher_proc.children() or None, 50,
                     desc="instance child process")
     os.kill(kids[0].pid, signal.SIGKILL)
 
     # controller reacts: requester Pod deleted (reference deletes it so
     # the owning ReplicaSet re-creates a fresh one)
-    wait_for(lambda: store.try_get("Pod", "sreq1") is None, 120,
+    wait_for(lambda: store.try_get("Pod", "sreq1") is None, 200,
              desc="requester deleted after instance death")
 
 
@@ -303,7 +303,7 @@ def test_kube_two_node_population_and_binding(tmp_path):
             assert set(by_node) == {"node-a", "node-b"}
             for node, lp in by_node.items():
                 wait_for(lambda lp=lp: ob.pod_is_ready(
-                    user.get("Pod", ob.name_of(lp))), 90,
+                    user.get("Pod", ob.name_of(lp))), 150,
                     desc=f"launcher on {node} ready")
 
             pod = ob.new_object(
@@ -315,7 +315,7 @@ def test_kube_two_node_population_and_binding(tmp_path):
                                 "fma_amd.requester.server"]}]})
             user.create(pod)
             wait_for(lambda: requester_ready(user, agents["node-b"],
-                                             "breq"), 120,
+                                             "breq"), 200,
                      desc="node-b requester ready")
             lp_b = user.get("Pod", ob.name_of(by_node["node-b"]))
             assert ob.annotations_of(lp_b)[C.REQUESTER_ANNOTATION].endswith(
