@@ -163,6 +163,12 @@ def create_app(store: Optional[MemStore] = None,
     @app.on_event("shutdown")
     async def _shutdown_pool():
         watch_pool.shutdown(wait=False, cancel_futures=True)
+
+    # explicit close for test servers whose lifespan shutdown may not
+    # fire (thread-leak fix: 64 idle pool workers per double otherwise
+    # outlive their server for the whole pytest session)
+    app.state.close = lambda: watch_pool.shutdown(wait=False,
+                                                  cancel_futures=True)
     # structural CRD validation, as a real apiserver's schema would
     from fma_amd.store.admission import crd_schema_policy
     if crd_schema_policy not in st._admission:
@@ -213,15 +219,16 @@ def create_app(store: Optional[MemStore] = None,
                             "object": status_body(410, e.message, "Gone"),
                         }) + "\n"
                         return
-                    if bookmarks and not batch and rev0 > cursor:
-                        # other kinds advanced the store; move this
-                        # watcher past history it will never see, so a
-                        # later reconnect cannot 410 on evicted events
-                        cursor = rev0
+                    if bookmarks and not batch:
+                        # keep-alive + cursor advance: moves idle
+                        # watchers past history churned by other kinds
+                        # (no 410 on reconnect) and keeps the stream
+                        # fed so clients can use bounded read timeouts
+                        cursor = max(cursor, rev0)
                         yield json.dumps({
                             "type": "BOOKMARK",
                             "object": {"kind": kind, "metadata": {
-                                "resourceVersion": str(rev0)}}}) + "\n"
+                                "resourceVersion": str(cursor)}}}) + "\n"
                     for ev in batch:
                         cursor = max(cursor, ev.revision)
                         obj = ev.obj

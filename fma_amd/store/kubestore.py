@@ -357,7 +357,12 @@ class KubeStore:
                         "GET", self.base + path,
                         params={"watch": "1", "resourceVersion": cursor,
                                 "allowWatchBookmarks": "true"},
-                        timeout=httpx.Timeout(5.0, read=None)) as r:
+                        # bounded read: bookmarks keep a live stream fed,
+                        # so a timeout means the server idled/died — we
+                        # reconnect from the cursor, and an abandoned
+                        # watcher can observe its stop event instead of
+                        # blocking forever (thread-leak fix)
+                        timeout=httpx.Timeout(5.0, read=8.0)) as r:
                     if r.status_code == 410:
                         out.put(RevisionTooOld(f"{kind} watch expired"))
                         return

@@ -41,6 +41,10 @@ class ServerThread:
     def __exit__(self, *a):
         self.server.should_exit = True
         self.thread.join(timeout=5)
+        close = getattr(self.config.app, "state", None)
+        close = getattr(close, "close", None) if close else None
+        if close:
+            close()  # reap the app's watch pool (thread-leak fix)
 
 
 def test_store_client_over_live_http():
