@@ -446,3 +446,36 @@ def test_watch_bookmarks_keep_idle_watchers_resumable():
     finally:
         stop.set()
         srv2.__exit__()
+
+
+def test_watch_threads_do_not_leak():
+    """Controller+server lifecycles leave no lingering watch threads:
+    bounded read timeouts let abandoned per-kind watchers observe their
+    stop event, and ServerThread closes the double's watch pool. (The
+    leak — ~6 threads per cycle, hundreds over a full suite — was the
+    root cause of load-correlated e2e flakes.)"""
+    import gc
+
+    from fma_amd.controller.dualpods.controller import (ControllerConfig,
+                                                        DualPodsController)
+    from fma_amd.controller.httpadapter import HttpAdapter
+
+    before = threading.active_count()
+    for i in range(3):
+        port = free_port()
+        with ServerThread(create_app(MemStore()), port):
+            ctl = DualPodsController(
+                KubeStore(f"http://127.0.0.1:{port}",
+                          actor="dual-pods-controller"),
+                HttpAdapter(), ControllerConfig())
+            ctl.start()
+            time.sleep(0.4)
+            ctl.stop()
+    gc.collect()
+    deadline = time.time() + 20
+    while time.time() < deadline:
+        if threading.active_count() <= before + 2:
+            break
+        time.sleep(0.5)
+    assert threading.active_count() <= before + 2, \
+        [t.name for t in threading.enumerate()]
